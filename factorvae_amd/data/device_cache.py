@@ -1,0 +1,82 @@
+"""Device-resident epoch cache.
+
+The reference re-gathers + H2D-copies every day-batch every epoch
+(/root/reference/train_model.py:17-24). On MI355X, 288 GB HBM3E makes the
+whole training range trivially device-resident (full A-share: 3500 x 60 x
+159 fp32 = 133 MB/day; a decade of days is a few hundred GB only at the
+extreme — CSI300 is ~2 GB): gather each day ONCE, ship it ONCE, and
+iterate epochs with zero host traffic.
+
+Each cached day holds the fp32 feature block x (N, T, C) and label
+y (N, 1) already split (the slicing of train_model.py:18-24 done once).
+"""
+
+from __future__ import annotations
+
+from typing import Iterator, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+
+class DeviceEpochCache:
+    """Gathers all day-batches of a loader once and keeps them on device.
+
+    Iteration order support: `order(epoch, shuffle, rank, world_size)`
+    yields (x, y) day tensors, matching DateGroupedBatchSampler's
+    seeded shuffle + round-robin DP sharding.
+    """
+
+    def __init__(self, dataloader, device: torch.device, dtype: torch.dtype = torch.float32,
+                 seed: int = 0):
+        self.device = device
+        self.dtype = dtype
+        self.seed = seed
+        self.days: List[Tuple[torch.Tensor, torch.Tensor]] = []
+        for char_with_label, _ in dataloader:
+            x = char_with_label[:, :, :-1].to(device=device, dtype=dtype, non_blocking=True)
+            y = char_with_label[:, -1, -1].reshape(-1, 1).to(device=device, dtype=dtype,
+                                                             non_blocking=True)
+            self.days.append((x.contiguous(), y.contiguous()))
+        if device.type == "cuda":
+            torch.cuda.synchronize(device)
+
+    def __len__(self) -> int:
+        return len(self.days)
+
+    @property
+    def max_stocks(self) -> int:
+        return max((x.shape[0] for x, _ in self.days), default=0)
+
+    def order(self, epoch: int = 0, shuffle: bool = False,
+              rank: int = 0, world_size: int = 1) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:
+        n = len(self.days)
+        idx = np.arange(n)
+        if shuffle:
+            rng = np.random.default_rng(self.seed + epoch)
+            rng.shuffle(idx)
+        for k in range(rank, n, world_size):
+            yield self.days[idx[k]]
+
+    def num_batches(self, rank: int = 0, world_size: int = 1) -> int:
+        n = len(self.days)
+        return (n - rank + world_size - 1) // world_size
+
+
+def synthetic_device_days(n_days: int, n_stocks: int, seq_len: int, n_features: int,
+                          device: torch.device, seed: int = 0,
+                          dtype: torch.dtype = torch.float32,
+                          generator: Optional[torch.Generator] = None
+                          ) -> List[Tuple[torch.Tensor, torch.Tensor]]:
+    """Random-init synthetic day tensors generated directly on device
+    (bench path: no host data, no network)."""
+    g = generator
+    if g is None:
+        g = torch.Generator(device=device)
+        g.manual_seed(seed)
+    days = []
+    for _ in range(n_days):
+        x = torch.randn(n_stocks, seq_len, n_features, device=device, dtype=dtype, generator=g)
+        y = torch.randn(n_stocks, 1, device=device, dtype=dtype, generator=g)
+        days.append((x, y))
+    return days
