@@ -1,0 +1,178 @@
+#!/usr/bin/env python3
+"""Benchmark: training cross-sections/sec (BASELINE.json headline metric).
+
+Measures the flagship FactorVAE training step — forward + backward +
+Adam + LR-scheduler per trading-day cross-section — on synthetic
+Alpha158-shaped data (N=300 stocks, T=20, C=158 features, K=20 factors;
+BASELINE.json config) with random-init weights.
+
+Usage:
+  python bench.py [--gpus N] [--steps K] [--warmup W] [--engine fused|eager]
+Multi-GPU (driver contract): launched under torch.distributed.run with
+one rank per GPU; ranks read RANK/LOCAL_RANK/WORLD_SIZE from the env.
+Weak scaling: each rank trains its own shard of day cross-sections
+(day-level DP, gradient all-reduced over RCCL each step), so the
+whole-job aggregate is steps * N_gpus cross-sections.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from factorvae_amd.models.modules import build_factorvae
+from factorvae_amd.parallel.ddp import (
+    FlatGradBucket,
+    get_rank,
+    get_world_size,
+    init_distributed,
+    is_distributed,
+)
+from factorvae_amd.utils import set_seed
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=200)
+    p.add_argument("--warmup", type=int, default=20)
+    p.add_argument("--engine", type=str, default="auto",
+                   choices=["auto", "fused", "eager"])
+    p.add_argument("--n_stocks", type=int, default=300)
+    p.add_argument("--seq_len", type=int, default=20)
+    p.add_argument("--num_latent", type=int, default=158)
+    p.add_argument("--num_factor", type=int, default=20)
+    p.add_argument("--hidden_size", type=int, default=64)
+    p.add_argument("--num_portfolio", type=int, default=128)
+    p.add_argument("--dtype", type=str, default="fp32", choices=["fp32"])
+    p.add_argument("--lr", type=float, default=1e-4)
+    p.add_argument("--seed", type=int, default=42)
+    p.add_argument("--n_days", type=int, default=32,
+                   help="distinct synthetic day tensors cycled through")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    rank = init_distributed()
+    world = get_world_size()
+    set_seed(args.seed + rank)
+
+    if torch.cuda.is_available():
+        device = torch.device("cuda", torch.cuda.current_device())
+    else:
+        device = torch.device("cpu")
+
+    N, T, C = args.n_stocks, args.seq_len, args.num_latent
+    H, M, K = args.hidden_size, args.num_portfolio, args.num_factor
+
+    model = build_factorvae(num_latent=C, hidden_size=H, num_portfolio=M,
+                            num_factor=K).to(device)
+
+    engine_name = args.engine
+    if engine_name == "auto":
+        engine_name = "fused" if device.type == "cuda" else "eager"
+
+    # synthetic device-resident day tensors (weak scaling: each rank its own days)
+    g = torch.Generator(device=device)
+    g.manual_seed(args.seed * 1000 + rank)
+    days = [(torch.randn(N, T, C, device=device, generator=g),
+             torch.randn(N, 1, device=device, generator=g))
+            for _ in range(args.n_days)]
+
+    total_steps = args.warmup + args.steps
+
+    if engine_name == "fused":
+        from factorvae_amd.engine.fused import FusedTrainer
+        trainer = FusedTrainer(model, lr=args.lr, t_max=total_steps, device=device)
+        step_fn = trainer.step
+    else:
+        bucket = FlatGradBucket(model.parameters()) if is_distributed() else None
+        opt = torch.optim.Adam(model.parameters(), lr=args.lr)
+        sched = torch.optim.lr_scheduler.CosineAnnealingLR(opt, T_max=total_steps)
+        model.train()
+
+        def step_fn(x, y):
+            if bucket is not None:
+                bucket.zero_()
+            else:
+                opt.zero_grad(set_to_none=True)
+            loss, *_ = model(x, y)
+            loss.backward()
+            if bucket is not None:
+                bucket.all_reduce_()
+            opt.step()
+            sched.step()
+            return loss
+
+    def sync():
+        if device.type == "cuda":
+            torch.cuda.synchronize(device)
+
+    def barrier():
+        if is_distributed():
+            torch.distributed.barrier()
+
+    # warmup
+    for i in range(args.warmup):
+        step_fn(*days[i % len(days)])
+    barrier()
+    sync()
+
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step_fn(*days[(args.warmup + i) % len(days)])
+    barrier()
+    sync()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    # max over ranks (report the slowest rank's wall time)
+    if is_distributed():
+        t = torch.tensor([elapsed], device=device if device.type == "cuda" else None,
+                         dtype=torch.float64)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    # whole-job aggregate: each rank processes one cross-section per step
+    value = args.steps * world / elapsed
+    baseline = 6.54  # cross-sections/sec, reference's only published speed datum
+    result = {
+        "metric": "training cross-sections/sec",
+        "value": value,
+        "unit": "cross-sections/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": ms_per_step,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": value / baseline,
+        "dtype": args.dtype,
+        "data": "synthetic",
+        "config": {
+            "model": "FactorVAE",
+            "global_batch": N * world,
+            "seq_len": T,
+            "parallelism": f"dp{world}",
+            "n_stocks": N,
+            "num_latent": C,
+            "num_factor": K,
+            "hidden_size": H,
+            "num_portfolio": M,
+            "engine": engine_name,
+        },
+    }
+    if rank == 0:
+        print(json.dumps(result))
+    if is_distributed():
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,343 @@
+// K-head cross-sectional attention (FactorPredictor) kernels.
+// Reference math: /root/reference/module.py:125-188 — scores
+// q·K^T/sqrt(H+1e-6) -> Dropout(0.1) pre-softmax -> ReLU -> softmax over
+// stocks (dim=0) -> NaN/Inf guard -> context a^T V, then shared MLP heads.
+//
+// MI355X-first restructure (same math, associativity aside): per-head key
+// matrices are NEVER materialized —
+//   scores[:,k] = h @ (Wk_k^T q_k) + q_k·bk_k      (one gemm_nt over all heads)
+//   ctx_k      = Wv_k (a_k^T h) + bv_k             (u = gemm_tn(a, h), tiny matvecs)
+// This removes the reference's 2K serialized (N,H)x(H,H) GEMMs per step.
+
+#include "common.h"
+
+// qk[k][i] = sum_j q[k][j] * Wk[k][j][i];  c[k] = sum_j q[k][j]*bk[k][j]
+// grid K, 64 threads (one wave).
+__global__ __launch_bounds__(64) void attn_qk_fwd_kernel(
+    const float* __restrict__ q, const float* __restrict__ Wk,
+    const float* __restrict__ bk, float* __restrict__ qk,
+    float* __restrict__ c, int K, int H) {
+  const int k = blockIdx.x;
+  const int lane = threadIdx.x;
+  const float* qh = q + (long)k * H;
+  const float* W = Wk + (long)k * H * H;
+
+  float acc = 0.0f;
+  if (lane < H) {
+    for (int j = 0; j < H; ++j) acc = fmaf(qh[j], W[(long)j * H + lane], acc);
+    qk[(long)k * H + lane] = acc;
+  }
+  float cv = (lane < H) ? qh[lane] * bk[(long)k * H + lane] : 0.0f;
+  cv = wave_reduce_sum(cv);
+  if (lane == 0) c[k] = cv;
+}
+
+// Per-head column: dropout (train) -> relu -> softmax over N -> NaN guard.
+// s (N,K) = (h@qk^T + c) * invsqrt scale (from gemm_nt's alpha).
+// Saves sd (post-dropout, pre-relu) for backward; guard[k]=1 zeroes the head.
+__global__ __launch_bounds__(256) void attn_softmax_fwd_kernel(
+    const float* __restrict__ s, const float* __restrict__ mask,  // mask may be null
+    float* __restrict__ a, float* __restrict__ sd_out,
+    int* __restrict__ guard, int N, int K, float keep_inv) {
+  __shared__ float scratch[8];
+  __shared__ int bad_s;
+  const int k = blockIdx.x;
+  const int tid = threadIdx.x;
+  if (tid == 0) bad_s = 0;
+  __syncthreads();
+
+  // Guard semantics match torch: softmax output contains NaN iff the
+  // post-relu input has NaN or +inf (relu(-inf)=0 is benign). fmaxf would
+  // silently swallow NaN, so detect on sd directly.
+  int bad = 0;
+  float mx = -INFINITY;
+  for (int n = tid; n < N; n += 256) {
+    float v = s[(long)n * K + k];
+    if (mask) v *= mask[(long)n * K + k] * keep_inv;
+    sd_out[(long)n * K + k] = v;
+    if (isnan(v) || v == INFINITY) bad = 1;
+    mx = fmaxf(mx, (v > 0.0f) ? v : 0.0f);
+  }
+  if (bad) atomicOr(&bad_s, 1);
+  mx = block_reduce_max(mx, scratch);
+
+  float sum = 0.0f;
+  for (int n = tid; n < N; n += 256) {
+    const float v = sd_out[(long)n * K + k];
+    sum += __expf(((v > 0.0f) ? v : 0.0f) - mx);
+  }
+  sum = block_reduce_sum(sum, scratch);
+  const float inv = 1.0f / sum;
+
+  for (int n = tid; n < N; n += 256) {
+    const float v = sd_out[(long)n * K + k];
+    a[(long)n * K + k] = __expf(((v > 0.0f) ? v : 0.0f) - mx) * inv;
+  }
+  __syncthreads();
+  if (tid == 0) guard[k] = bad_s;
+}
+
+// ctx[k][j] = sum_i Wv[k][j][i] * u[k][i] + bv[k][j]; guarded heads -> 0.
+__global__ __launch_bounds__(64) void attn_ctx_fwd_kernel(
+    const float* __restrict__ u, const float* __restrict__ Wv,
+    const float* __restrict__ bv, const int* __restrict__ guard,
+    float* __restrict__ ctx, int K, int H) {
+  const int k = blockIdx.x;
+  const int j = threadIdx.x;
+  if (j >= H) return;
+  if (guard[k]) {
+    ctx[(long)k * H + j] = 0.0f;
+    return;
+  }
+  const float* W = Wv + ((long)k * H + j) * H;
+  const float* uh = u + (long)k * H;
+  float acc = bv[(long)k * H + j];
+  for (int i = 0; i < H; ++i) acc = fmaf(W[i], uh[i], acc);
+  ctx[(long)k * H + j] = acc;
+}
+
+// Backward of the value path: du[k][i] = sum_j Wv[k][j][i]*dctx[k][j];
+// dWv[k][j][i] = dctx[k][j]*u[k][i]; dbv[k][j] = dctx[k][j].
+// Guarded heads contribute zero (reference returns a detached zeros
+// context). Param grads are plain += (each element owned by one WG).
+__global__ __launch_bounds__(64) void attn_head_bwd_kernel(
+    const float* __restrict__ dctx_in, const float* __restrict__ u,
+    const float* __restrict__ Wv, const int* __restrict__ guard,
+    float* __restrict__ du, float* __restrict__ dWv, float* __restrict__ dbv,
+    int K, int H) {
+  const int k = blockIdx.x;
+  const int lane = threadIdx.x;
+  if (lane >= H) return;
+  const bool g = guard[k] != 0;
+  const float* dctx = dctx_in + (long)k * H;
+  const float* uh = u + (long)k * H;
+
+  // dWv row j = lane: dctx[lane] * u[:]
+  const float dcj = g ? 0.0f : dctx[lane];
+  float* dWrow = dWv + ((long)k * H + lane) * H;
+  for (int i = 0; i < H; ++i) dWrow[i] += dcj * uh[i];
+  dbv[(long)k * H + lane] += dcj;
+
+  // du[i = lane] = sum_j Wv[j][i] * dctx[j]
+  float acc = 0.0f;
+  if (!g) {
+    const float* W = Wv + (long)k * H * H;
+    for (int j = 0; j < H; ++j) acc = fmaf(W[(long)j * H + lane], dctx[j], acc);
+  }
+  du[(long)k * H + lane] = acc;
+}
+
+// Softmax+relu+dropout backward per head column; also emits
+// dc[k] = sum_n ds and applies the score scale alpha so the output is the
+// gradient w.r.t. (h@qk^T + c).
+__global__ __launch_bounds__(256) void attn_softmax_bwd_kernel(
+    const float* __restrict__ da, const float* __restrict__ a,
+    const float* __restrict__ sd, const float* __restrict__ mask,
+    const int* __restrict__ guard, float* __restrict__ ds,
+    float* __restrict__ dc, int N, int K, float keep_inv, float alpha) {
+  __shared__ float scratch[8];
+  const int k = blockIdx.x;
+  const int tid = threadIdx.x;
+
+  if (guard[k]) {
+    for (int n = tid; n < N; n += 256) ds[(long)n * K + k] = 0.0f;
+    if (tid == 0) dc[k] = 0.0f;
+    return;
+  }
+
+  float t = 0.0f;
+  for (int n = tid; n < N; n += 256) {
+    const long i = (long)n * K + k;
+    t = fmaf(a[i], da[i], t);
+  }
+  t = block_reduce_sum(t, scratch);
+
+  float csum = 0.0f;
+  for (int n = tid; n < N; n += 256) {
+    const long i = (long)n * K + k;
+    float dr = a[i] * (da[i] - t);          // softmax bwd
+    dr = (sd[i] > 0.0f) ? dr : 0.0f;        // relu bwd
+    if (mask) dr *= mask[i] * keep_inv;     // dropout bwd
+    dr *= alpha;                            // score scale
+    ds[i] = dr;
+    csum += dr;
+  }
+  csum = block_reduce_sum(csum, scratch);
+  if (tid == 0) dc[k] = csum;
+}
+
+// dq[k][j] = sum_i Wk[k][j][i]*dqk[k][i] + dc[k]*bk[k][j];
+// dWk[k][j][i] = q[k][j]*dqk[k][i]; dbk[k][j] = dc[k]*q[k][j].
+__global__ __launch_bounds__(64) void attn_qk_bwd_kernel(
+    const float* __restrict__ dqk, const float* __restrict__ dc,
+    const float* __restrict__ q, const float* __restrict__ Wk,
+    const float* __restrict__ bk, float* __restrict__ dq,
+    float* __restrict__ dWk, float* __restrict__ dbk, int K, int H) {
+  const int k = blockIdx.x;
+  const int lane = threadIdx.x;
+  if (lane >= H) return;
+  const float* dqkh = dqk + (long)k * H;
+  const float dck = dc[k];
+  const float qj = q[(long)k * H + lane];
+
+  float* dWrow = dWk + ((long)k * H + lane) * H;
+  for (int i = 0; i < H; ++i) dWrow[i] += qj * dqkh[i];
+  dbk[(long)k * H + lane] += dck * qj;
+
+  const float* W = Wk + (long)k * H * H;
+  float acc = dck * bk[(long)k * H + lane];
+  const float* Wrow = W + (long)lane * H;
+  for (int i = 0; i < H; ++i) acc = fmaf(Wrow[i], dqkh[i], acc);
+  dq[(long)k * H + lane] += acc;
+}
+
+// Predictor shared MLP forward: hm2 = lrelu(ctx@Wl^T + bl);
+// pmu = hm2·wmu + bmu; psig = softplus(hm2·wsig + bsig); clamp ==0 -> 1e-6
+// (module.py:264-265). One wave per head row.
+__global__ __launch_bounds__(64) void pred_mlp_fwd_kernel(
+    const float* __restrict__ ctx, const float* __restrict__ Wl,
+    const float* __restrict__ bl, const float* __restrict__ wmu,
+    const float* __restrict__ bmu, const float* __restrict__ wsig,
+    const float* __restrict__ bsig, float* __restrict__ hm2,
+    float* __restrict__ pmu, float* __restrict__ psig_pre,
+    float* __restrict__ psig, float* __restrict__ psig_c, int K, int H) {
+  const int k = blockIdx.x;
+  const int lane = threadIdx.x;
+  const float* cr = ctx + (long)k * H;
+
+  float z = 0.0f;
+  if (lane < H) {
+    z = bl[lane];
+    const float* Wrow = Wl + (long)lane * H;
+    for (int i = 0; i < H; ++i) z = fmaf(cr[i], Wrow[i], z);
+    z = lrelu_(z);
+    hm2[(long)k * H + lane] = z;
+  }
+  float pm = (lane < H) ? z * wmu[lane] : 0.0f;
+  float ps = (lane < H) ? z * wsig[lane] : 0.0f;
+  pm = wave_reduce_sum(pm);
+  ps = wave_reduce_sum(ps);
+  if (lane == 0) {
+    pmu[k] = pm + bmu[0];
+    const float pre = ps + bsig[0];
+    psig_pre[k] = pre;
+    const float s = softplusf_(pre);
+    psig[k] = s;
+    psig_c[k] = (s == 0.0f) ? 1e-6f : s;
+  }
+}
+
+// Predictor MLP backward: dpmu, dpsig_c -> dz2 (K,H) (grad at pre-lrelu)
+// + head param grads (wmu/bmu/wsig/bsig; atomic across head WGs).
+__global__ __launch_bounds__(64) void pred_mlp_bwd_kernel(
+    const float* __restrict__ dpmu, const float* __restrict__ dpsig_c,
+    const float* __restrict__ psig, const float* __restrict__ psig_pre,
+    const float* __restrict__ hm2, const float* __restrict__ wmu,
+    const float* __restrict__ wsig, float* __restrict__ dz2,
+    float* __restrict__ dwmu, float* __restrict__ dbmu,
+    float* __restrict__ dwsig, float* __restrict__ dbsig, int K, int H) {
+  const int k = blockIdx.x;
+  const int lane = threadIdx.x;
+  if (lane >= H) return;
+  const float dm = dpmu[k];
+  const float dsc = (psig[k] == 0.0f) ? 0.0f : dpsig_c[k];
+  const float dsp = dsc * softplus_gradf_(psig_pre[k]);
+  const float h2 = hm2[(long)k * H + lane];
+
+  const float dh2 = dm * wmu[lane] + dsp * wsig[lane];
+  dz2[(long)k * H + lane] = dh2 * lrelu_grad_from_out_(h2);
+
+  atomicAdd(&dwmu[lane], dm * h2);
+  atomicAdd(&dwsig[lane], dsp * h2);
+  if (lane == 0) {
+    atomicAdd(&dbmu[0], dm);
+    atomicAdd(&dbsig[0], dsp);
+  }
+}
+
+extern "C" {
+
+hipError_t fv_attn_qk_fwd(const float* q, const float* Wk, const float* bk,
+                          float* qk, float* c, int K, int H, hipStream_t s) {
+  if (H > 64) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(attn_qk_fwd_kernel, dim3(K), dim3(64), 0, s, q, Wk, bk, qk, c, K, H);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_attn_softmax_fwd(const float* sc, const float* mask, float* a,
+                               float* sd, int* guard, int N, int K,
+                               float keep_inv, hipStream_t s) {
+  hipLaunchKernelGGL(attn_softmax_fwd_kernel, dim3(K), dim3(256), 0, s,
+                     sc, mask, a, sd, guard, N, K, keep_inv);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_attn_ctx_fwd(const float* u, const float* Wv, const float* bv,
+                           const int* guard, float* ctx, int K, int H,
+                           hipStream_t s) {
+  if (H > 64) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(attn_ctx_fwd_kernel, dim3(K), dim3(64), 0, s,
+                     u, Wv, bv, guard, ctx, K, H);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_attn_head_bwd(const float* dctx, const float* u, const float* Wv,
+                            const int* guard, float* du, float* dWv, float* dbv,
+                            int K, int H, hipStream_t s) {
+  if (H > 64) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(attn_head_bwd_kernel, dim3(K), dim3(64), 0, s,
+                     dctx, u, Wv, guard, du, dWv, dbv, K, H);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_attn_softmax_bwd(const float* da, const float* a, const float* sd,
+                               const float* mask, const int* guard, float* ds,
+                               float* dc, int N, int K, float keep_inv,
+                               float alpha, hipStream_t s) {
+  hipLaunchKernelGGL(attn_softmax_bwd_kernel, dim3(K), dim3(256), 0, s,
+                     da, a, sd, mask, guard, ds, dc, N, K, keep_inv, alpha);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_attn_qk_bwd(const float* dqk, const float* dc, const float* q,
+                          const float* Wk, const float* bk, float* dq,
+                          float* dWk, float* dbk, int K, int H, hipStream_t s) {
+  if (H > 64) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(attn_qk_bwd_kernel, dim3(K), dim3(64), 0, s,
+                     dqk, dc, q, Wk, bk, dq, dWk, dbk, K, H);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_pred_mlp_fwd(const float* ctx, const float* Wl, const float* bl,
+                           const float* wmu, const float* bmu, const float* wsig,
+                           const float* bsig, float* hm2, float* pmu,
+                           float* psig_pre, float* psig, float* psig_c,
+                           int K, int H, hipStream_t s) {
+  if (H > 64) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(pred_mlp_fwd_kernel, dim3(K), dim3(64), 0, s,
+                     ctx, Wl, bl, wmu, bmu, wsig, bsig, hm2, pmu, psig_pre,
+                     psig, psig_c, K, H);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_pred_mlp_bwd(const float* dpmu, const float* dpsig_c,
+                           const float* psig, const float* psig_pre,
+                           const float* hm2, const float* wmu, const float* wsig,
+                           float* dz2, float* dwmu, float* dbmu, float* dwsig,
+                           float* dbsig, int K, int H, hipStream_t s) {
+  if (H > 64) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(pred_mlp_bwd_kernel, dim3(K), dim3(64), 0, s,
+                     dpmu, dpsig_c, psig, psig_pre, hm2, wmu, wsig,
+                     dz2, dwmu, dbmu, dwsig, dbsig, K, H);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+}  // extern "C"

@@ -1,0 +1,405 @@
+// Torch extension bindings for the FactorVAE gfx950 kernels.
+// Thin layer: validate tensors, extract raw pointers, forward to the
+// extern "C" launchers defined in the .hip translation units.
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include <hip/hip_runtime_api.h>
+
+extern "C" {
+hipError_t fv_gemm_nt(const float*, const float*, const float*, float*, int,
+                      int, int, float, int, int, hipStream_t);
+hipError_t fv_gemm_nn(const float*, const float*, const float*, float*, int,
+                      int, int, float, int, int, hipStream_t);
+hipError_t fv_gemm_tn(const float*, const float*, float*, int, int, int, int,
+                      int, hipStream_t);
+hipError_t fv_colsum(const float*, float*, int, int, int, hipStream_t);
+hipError_t fv_lrelu_bwd(const float*, const float*, float*, long, hipStream_t);
+hipError_t fv_ln_fwd(const float*, const float*, const float*, float*, float*,
+                     float*, long, int, float, hipStream_t);
+hipError_t fv_ln_bwd_params(const float*, const float*, const float*,
+                            const float*, float*, float*, long, int, int,
+                            hipStream_t);
+hipError_t fv_gru_fwd(const float*, const float*, const float*, float*, float*,
+                      float*, float*, int, int, int, hipStream_t);
+hipError_t fv_gru_bwd(const float*, const float*, const float*, const float*,
+                      float*, float*, int, int, int, hipStream_t);
+hipError_t fv_enc_softmax_fwd(const float*, const float*, float*, float*, int,
+                              int, hipStream_t);
+hipError_t fv_enc_softmax_bwd(const float*, const float*, const float*, float*,
+                              int, int, hipStream_t);
+hipError_t fv_enc_heads_fwd(const float*, const float*, const float*,
+                            const float*, const float*, float*, float*, float*,
+                            float*, int, int, hipStream_t);
+hipError_t fv_enc_heads_bwd(const float*, const float*, const float*,
+                            const float*, const float*, const float*,
+                            const float*, float*, float*, float*, float*,
+                            float*, int, int, hipStream_t);
+hipError_t fv_attn_qk_fwd(const float*, const float*, const float*, float*,
+                          float*, int, int, hipStream_t);
+hipError_t fv_attn_softmax_fwd(const float*, const float*, float*, float*,
+                               int*, int, int, float, hipStream_t);
+hipError_t fv_attn_ctx_fwd(const float*, const float*, const float*,
+                           const int*, float*, int, int, hipStream_t);
+hipError_t fv_attn_head_bwd(const float*, const float*, const float*,
+                            const int*, float*, float*, float*, int, int,
+                            hipStream_t);
+hipError_t fv_attn_softmax_bwd(const float*, const float*, const float*,
+                               const float*, const int*, float*, float*, int,
+                               int, float, float, hipStream_t);
+hipError_t fv_attn_qk_bwd(const float*, const float*, const float*,
+                          const float*, const float*, float*, float*, float*,
+                          int, int, hipStream_t);
+hipError_t fv_pred_mlp_fwd(const float*, const float*, const float*,
+                           const float*, const float*, const float*,
+                           const float*, float*, float*, float*, float*,
+                           float*, int, int, hipStream_t);
+hipError_t fv_pred_mlp_bwd(const float*, const float*, const float*,
+                           const float*, const float*, const float*,
+                           const float*, float*, float*, float*, float*,
+                           float*, int, int, hipStream_t);
+hipError_t fv_dec_fwd(const float*, const float*, const float*, const float*,
+                      const float*, const float*, const float*, const float*,
+                      const float*, const float*, const float*, const float*,
+                      float*, float*, float*, float*, float*, int, int, int,
+                      hipStream_t);
+hipError_t fv_dec_bwd(const float*, const float*, const float*, const float*,
+                      const float*, const float*, const float*, const float*,
+                      const float*, const float*, const float*, const float*,
+                      const float*, float*, float*, float*, float*, float*,
+                      float*, float*, float*, float*, int, int, int,
+                      hipStream_t);
+hipError_t fv_loss_fwd(const float*, const float*, const float*, const float*,
+                       const float*, const float*, float*, float*, float*, int,
+                       int, hipStream_t);
+hipError_t fv_loss_bwd(const float*, const float*, const float*, const float*,
+                       const float*, const float*, float*, float*, float*,
+                       float*, float*, int, int, float, hipStream_t);
+hipError_t fv_step_inc(int*, hipStream_t);
+hipError_t fv_adam(float*, const float*, float*, float*, const int*, long,
+                   float, float, float, float, float, float, hipStream_t);
+}
+
+namespace {
+
+inline hipStream_t cur_stream() {
+  return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+}
+
+inline const float* fp(const torch::Tensor& t) { return t.data_ptr<float>(); }
+inline float* fpm(torch::Tensor& t) { return t.data_ptr<float>(); }
+
+void check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+}
+
+#define CK(t) check_f32(t, #t)
+#define RUN(call)                                                      \
+  do {                                                                 \
+    hipError_t e_ = (call);                                            \
+    TORCH_CHECK(e_ == hipSuccess, "HIP kernel failed: ", #call, " (", \
+                (int)e_, ")");                                         \
+  } while (0)
+
+void gemm_nt(torch::Tensor A, torch::Tensor W,
+             c10::optional<torch::Tensor> bias, torch::Tensor out,
+             double alpha, bool accumulate, bool act_lrelu) {
+  CK(A); CK(W); CK(out);
+  const int R = A.size(0), Ci = A.size(1), Co = W.size(0);
+  TORCH_CHECK(W.size(1) == Ci && out.size(0) == R && out.size(1) == Co);
+  const float* b = nullptr;
+  if (bias.has_value()) { CK(*bias); b = fp(*bias); }
+  RUN(fv_gemm_nt(fp(A), fp(W), b, fpm(out), R, Ci, Co, (float)alpha,
+                 accumulate, act_lrelu, cur_stream()));
+}
+
+void gemm_nn(torch::Tensor A, torch::Tensor B,
+             c10::optional<torch::Tensor> bias, torch::Tensor out,
+             double alpha, bool accumulate, bool act_lrelu) {
+  CK(A); CK(B); CK(out);
+  const int R = A.size(0), Ci = A.size(1), Co = B.size(1);
+  TORCH_CHECK(B.size(0) == Ci && out.size(0) == R && out.size(1) == Co);
+  const float* b = nullptr;
+  if (bias.has_value()) { CK(*bias); b = fp(*bias); }
+  RUN(fv_gemm_nn(fp(A), fp(B), b, fpm(out), R, Ci, Co, (float)alpha,
+                 accumulate, act_lrelu, cur_stream()));
+}
+
+void gemm_tn(torch::Tensor A, torch::Tensor B, torch::Tensor out, long r_chunks,
+             bool use_atomic) {
+  CK(A); CK(B); CK(out);
+  const int R = A.size(0), M = A.size(1), N = B.size(1);
+  TORCH_CHECK(B.size(0) == R && out.size(0) == M && out.size(1) == N);
+  RUN(fv_gemm_tn(fp(A), fp(B), fpm(out), R, M, N, (int)r_chunks, use_atomic,
+                 cur_stream()));
+}
+
+void colsum(torch::Tensor A, torch::Tensor out, long r_chunks) {
+  CK(A); CK(out);
+  const int R = A.size(0), C = A.size(1);
+  TORCH_CHECK(out.numel() == C);
+  RUN(fv_colsum(fp(A), fpm(out), R, C, (int)r_chunks, cur_stream()));
+}
+
+void lrelu_bwd(torch::Tensor dY, torch::Tensor Y, torch::Tensor dZ) {
+  CK(dY); CK(Y); CK(dZ);
+  RUN(fv_lrelu_bwd(fp(dY), fp(Y), fpm(dZ), dY.numel(), cur_stream()));
+}
+
+void ln_fwd(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
+            torch::Tensor xln, torch::Tensor mean, torch::Tensor rstd,
+            double eps) {
+  CK(x); CK(gamma); CK(beta); CK(xln); CK(mean); CK(rstd);
+  const long R = x.numel() / x.size(-1);
+  const int C = x.size(-1);
+  RUN(fv_ln_fwd(fp(x), fp(gamma), fp(beta), fpm(xln), fpm(mean), fpm(rstd), R,
+                C, (float)eps, cur_stream()));
+}
+
+void ln_bwd_params(torch::Tensor x, torch::Tensor dxln, torch::Tensor mean,
+                   torch::Tensor rstd, torch::Tensor dgamma,
+                   torch::Tensor dbeta, long r_chunks) {
+  CK(x); CK(dxln); CK(mean); CK(rstd); CK(dgamma); CK(dbeta);
+  const long R = x.numel() / x.size(-1);
+  const int C = x.size(-1);
+  RUN(fv_ln_bwd_params(fp(x), fp(dxln), fp(mean), fp(rstd), fpm(dgamma),
+                       fpm(dbeta), R, C, (int)r_chunks, cur_stream()));
+}
+
+void gru_fwd(torch::Tensor gi, torch::Tensor Whh, torch::Tensor bhh,
+             torch::Tensor h_final, torch::Tensor h_seq, torch::Tensor h_prev,
+             torch::Tensor gates4, long N, long T, long H) {
+  CK(gi); CK(Whh); CK(bhh); CK(h_final); CK(h_seq); CK(h_prev); CK(gates4);
+  RUN(fv_gru_fwd(fp(gi), fp(Whh), fp(bhh), fpm(h_final), fpm(h_seq),
+                 fpm(h_prev), fpm(gates4), (int)N, (int)T, (int)H,
+                 cur_stream()));
+}
+
+void gru_bwd(torch::Tensor dh_final, torch::Tensor h_prev, torch::Tensor gates4,
+             torch::Tensor Whh, torch::Tensor dgi, torch::Tensor dgh, long N,
+             long T, long H) {
+  CK(dh_final); CK(h_prev); CK(gates4); CK(Whh); CK(dgi); CK(dgh);
+  RUN(fv_gru_bwd(fp(dh_final), fp(h_prev), fp(gates4), fp(Whh), fpm(dgi),
+                 fpm(dgh), (int)N, (int)T, (int)H, cur_stream()));
+}
+
+void enc_softmax_fwd(torch::Tensor scores, torch::Tensor y, torch::Tensor a,
+                     torch::Tensor yp) {
+  CK(scores); CK(y); CK(a); CK(yp);
+  RUN(fv_enc_softmax_fwd(fp(scores), fp(y), fpm(a), fpm(yp),
+                         scores.size(0), scores.size(1), cur_stream()));
+}
+
+void enc_softmax_bwd(torch::Tensor dyp, torch::Tensor a, torch::Tensor y,
+                     torch::Tensor dscores) {
+  CK(dyp); CK(a); CK(y); CK(dscores);
+  RUN(fv_enc_softmax_bwd(fp(dyp), fp(a), fp(y), fpm(dscores), a.size(0),
+                         a.size(1), cur_stream()));
+}
+
+void enc_heads_fwd(torch::Tensor yp, torch::Tensor Wmu, torch::Tensor bmu,
+                   torch::Tensor Wsig, torch::Tensor bsig, torch::Tensor fmu,
+                   torch::Tensor fsig_pre, torch::Tensor fsig,
+                   torch::Tensor fsig_c) {
+  CK(yp); CK(Wmu); CK(bmu); CK(Wsig); CK(bsig); CK(fmu); CK(fsig_pre);
+  CK(fsig); CK(fsig_c);
+  RUN(fv_enc_heads_fwd(fp(yp), fp(Wmu), fp(bmu), fp(Wsig), fp(bsig), fpm(fmu),
+                       fpm(fsig_pre), fpm(fsig), fpm(fsig_c), yp.numel(),
+                       Wmu.size(0), cur_stream()));
+}
+
+void enc_heads_bwd(torch::Tensor dfmu, torch::Tensor dfsig_c,
+                   torch::Tensor fsig, torch::Tensor fsig_pre,
+                   torch::Tensor yp, torch::Tensor Wmu, torch::Tensor Wsig,
+                   torch::Tensor dyp, torch::Tensor dWmu, torch::Tensor dbmu,
+                   torch::Tensor dWsig, torch::Tensor dbsig) {
+  CK(dfmu); CK(dfsig_c); CK(fsig); CK(fsig_pre); CK(yp); CK(Wmu); CK(Wsig);
+  CK(dyp); CK(dWmu); CK(dbmu); CK(dWsig); CK(dbsig);
+  RUN(fv_enc_heads_bwd(fp(dfmu), fp(dfsig_c), fp(fsig), fp(fsig_pre), fp(yp),
+                       fp(Wmu), fp(Wsig), fpm(dyp), fpm(dWmu), fpm(dbmu),
+                       fpm(dWsig), fpm(dbsig), yp.numel(), Wmu.size(0),
+                       cur_stream()));
+}
+
+void attn_qk_fwd(torch::Tensor q, torch::Tensor Wk, torch::Tensor bk,
+                 torch::Tensor qk, torch::Tensor c) {
+  CK(q); CK(Wk); CK(bk); CK(qk); CK(c);
+  RUN(fv_attn_qk_fwd(fp(q), fp(Wk), fp(bk), fpm(qk), fpm(c), q.size(0),
+                     q.size(1), cur_stream()));
+}
+
+void attn_softmax_fwd(torch::Tensor s, c10::optional<torch::Tensor> mask,
+                      torch::Tensor a, torch::Tensor sd, torch::Tensor guard,
+                      double keep_inv) {
+  CK(s); CK(a); CK(sd);
+  TORCH_CHECK(guard.scalar_type() == torch::kInt32 && guard.is_cuda());
+  const float* mp = nullptr;
+  if (mask.has_value()) { CK(*mask); mp = fp(*mask); }
+  RUN(fv_attn_softmax_fwd(fp(s), mp, fpm(a), fpm(sd), guard.data_ptr<int>(),
+                          s.size(0), s.size(1), (float)keep_inv,
+                          cur_stream()));
+}
+
+void attn_ctx_fwd(torch::Tensor u, torch::Tensor Wv, torch::Tensor bv,
+                  torch::Tensor guard, torch::Tensor ctx) {
+  CK(u); CK(Wv); CK(bv); CK(ctx);
+  RUN(fv_attn_ctx_fwd(fp(u), fp(Wv), fp(bv), guard.data_ptr<int>(), fpm(ctx),
+                      u.size(0), u.size(1), cur_stream()));
+}
+
+void attn_head_bwd(torch::Tensor dctx, torch::Tensor u, torch::Tensor Wv,
+                   torch::Tensor guard, torch::Tensor du, torch::Tensor dWv,
+                   torch::Tensor dbv) {
+  CK(dctx); CK(u); CK(Wv); CK(du); CK(dWv); CK(dbv);
+  RUN(fv_attn_head_bwd(fp(dctx), fp(u), fp(Wv), guard.data_ptr<int>(),
+                       fpm(du), fpm(dWv), fpm(dbv), u.size(0), u.size(1),
+                       cur_stream()));
+}
+
+void attn_softmax_bwd(torch::Tensor da, torch::Tensor a, torch::Tensor sd,
+                      c10::optional<torch::Tensor> mask, torch::Tensor guard,
+                      torch::Tensor ds, torch::Tensor dc, double keep_inv,
+                      double alpha) {
+  CK(da); CK(a); CK(sd); CK(ds); CK(dc);
+  const float* mp = nullptr;
+  if (mask.has_value()) { CK(*mask); mp = fp(*mask); }
+  RUN(fv_attn_softmax_bwd(fp(da), fp(a), fp(sd), mp, guard.data_ptr<int>(),
+                          fpm(ds), fpm(dc), a.size(0), a.size(1),
+                          (float)keep_inv, (float)alpha, cur_stream()));
+}
+
+void attn_qk_bwd(torch::Tensor dqk, torch::Tensor dc, torch::Tensor q,
+                 torch::Tensor Wk, torch::Tensor bk, torch::Tensor dq,
+                 torch::Tensor dWk, torch::Tensor dbk) {
+  CK(dqk); CK(dc); CK(q); CK(Wk); CK(bk); CK(dq); CK(dWk); CK(dbk);
+  RUN(fv_attn_qk_bwd(fp(dqk), fp(dc), fp(q), fp(Wk), fp(bk), fpm(dq),
+                     fpm(dWk), fpm(dbk), q.size(0), q.size(1), cur_stream()));
+}
+
+void pred_mlp_fwd(torch::Tensor ctx, torch::Tensor Wl, torch::Tensor bl,
+                  torch::Tensor wmu, torch::Tensor bmu, torch::Tensor wsig,
+                  torch::Tensor bsig, torch::Tensor hm2, torch::Tensor pmu,
+                  torch::Tensor psig_pre, torch::Tensor psig,
+                  torch::Tensor psig_c) {
+  CK(ctx); CK(Wl); CK(bl); CK(wmu); CK(bmu); CK(wsig); CK(bsig); CK(hm2);
+  CK(pmu); CK(psig_pre); CK(psig); CK(psig_c);
+  RUN(fv_pred_mlp_fwd(fp(ctx), fp(Wl), fp(bl), fp(wmu), fp(bmu), fp(wsig),
+                      fp(bsig), fpm(hm2), fpm(pmu), fpm(psig_pre), fpm(psig),
+                      fpm(psig_c), ctx.size(0), ctx.size(1), cur_stream()));
+}
+
+void pred_mlp_bwd(torch::Tensor dpmu, torch::Tensor dpsig_c, torch::Tensor psig,
+                  torch::Tensor psig_pre, torch::Tensor hm2, torch::Tensor wmu,
+                  torch::Tensor wsig, torch::Tensor dz2, torch::Tensor dwmu,
+                  torch::Tensor dbmu, torch::Tensor dwsig,
+                  torch::Tensor dbsig) {
+  CK(dpmu); CK(dpsig_c); CK(psig); CK(psig_pre); CK(hm2); CK(wmu); CK(wsig);
+  CK(dz2); CK(dwmu); CK(dbmu); CK(dwsig); CK(dbsig);
+  RUN(fv_pred_mlp_bwd(fp(dpmu), fp(dpsig_c), fp(psig), fp(psig_pre), fp(hm2),
+                      fp(wmu), fp(wsig), fpm(dz2), fpm(dwmu), fpm(dbmu),
+                      fpm(dwsig), fpm(dbsig), hm2.size(0), hm2.size(1),
+                      cur_stream()));
+}
+
+void dec_fwd(torch::Tensor h, torch::Tensor W1, torch::Tensor b1,
+             torch::Tensor wmu, torch::Tensor bmu, torch::Tensor wsig,
+             torch::Tensor bsig, torch::Tensor Wb, torch::Tensor bb,
+             torch::Tensor fmu, torch::Tensor fsig_c, torch::Tensor eps,
+             torch::Tensor recon, torch::Tensor a1, torch::Tensor beta,
+             torch::Tensor asig_pre, torch::Tensor sigma) {
+  CK(h); CK(W1); CK(b1); CK(Wb); CK(bb); CK(fmu); CK(fsig_c); CK(eps);
+  CK(recon); CK(a1); CK(beta); CK(asig_pre); CK(sigma);
+  RUN(fv_dec_fwd(fp(h), fp(W1), fp(b1), fp(wmu), fp(bmu), fp(wsig), fp(bsig),
+                 fp(Wb), fp(bb), fp(fmu), fp(fsig_c), fp(eps), fpm(recon),
+                 fpm(a1), fpm(beta), fpm(asig_pre), fpm(sigma), h.size(0),
+                 Wb.size(0), h.size(1), cur_stream()));
+}
+
+void dec_bwd(torch::Tensor drecon, torch::Tensor h, torch::Tensor a1,
+             torch::Tensor beta, torch::Tensor asig_pre, torch::Tensor sigma,
+             torch::Tensor eps, torch::Tensor fmu, torch::Tensor fsig_c,
+             torch::Tensor W1, torch::Tensor wmu, torch::Tensor wsig,
+             torch::Tensor Wb, torch::Tensor dh, torch::Tensor dz1,
+             torch::Tensor dbeta, torch::Tensor dfmu, torch::Tensor dfsig_c,
+             torch::Tensor dwmu, torch::Tensor dbmu, torch::Tensor dwsig,
+             torch::Tensor dbsig) {
+  CK(drecon); CK(h); CK(a1); CK(beta); CK(asig_pre); CK(sigma); CK(eps);
+  CK(fmu); CK(fsig_c); CK(W1); CK(wmu); CK(wsig); CK(Wb); CK(dh); CK(dz1);
+  CK(dbeta); CK(dfmu); CK(dfsig_c); CK(dwmu); CK(dbmu); CK(dwsig); CK(dbsig);
+  RUN(fv_dec_bwd(fp(drecon), fp(h), fp(a1), fp(beta), fp(asig_pre), fp(sigma),
+                 fp(eps), fp(fmu), fp(fsig_c), fp(W1), fp(wmu), fp(wsig),
+                 fp(Wb), fpm(dh), fpm(dz1), fpm(dbeta), fpm(dfmu),
+                 fpm(dfsig_c), fpm(dwmu), fpm(dbmu), fpm(dwsig), fpm(dbsig),
+                 h.size(0), Wb.size(0), h.size(1), cur_stream()));
+}
+
+void loss_fwd(torch::Tensor recon, torch::Tensor y, torch::Tensor fmu,
+              torch::Tensor fsig_c, torch::Tensor pmu, torch::Tensor psig_c,
+              torch::Tensor loss, torch::Tensor mse, torch::Tensor kl) {
+  CK(recon); CK(y); CK(fmu); CK(fsig_c); CK(pmu); CK(psig_c); CK(loss);
+  CK(mse); CK(kl);
+  RUN(fv_loss_fwd(fp(recon), fp(y), fp(fmu), fp(fsig_c), fp(pmu), fp(psig_c),
+                  fpm(loss), fpm(mse), fpm(kl), recon.numel(), fmu.numel(),
+                  cur_stream()));
+}
+
+void loss_bwd(torch::Tensor recon, torch::Tensor y, torch::Tensor fmu,
+              torch::Tensor fsig_c, torch::Tensor pmu, torch::Tensor psig_c,
+              torch::Tensor drecon, torch::Tensor dfmu, torch::Tensor dfsig_c,
+              torch::Tensor dpmu, torch::Tensor dpsig_c, double gscale) {
+  CK(recon); CK(y); CK(drecon); CK(dfmu); CK(dfsig_c); CK(dpmu); CK(dpsig_c);
+  RUN(fv_loss_bwd(fp(recon), fp(y), fp(fmu), fp(fsig_c), fp(pmu), fp(psig_c),
+                  fpm(drecon), fpm(dfmu), fpm(dfsig_c), fpm(dpmu),
+                  fpm(dpsig_c), recon.numel(), fmu.numel(), (float)gscale,
+                  cur_stream()));
+}
+
+void step_inc(torch::Tensor step_t) {
+  TORCH_CHECK(step_t.scalar_type() == torch::kInt32 && step_t.is_cuda());
+  RUN(fv_step_inc(step_t.data_ptr<int>(), cur_stream()));
+}
+
+void adam(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
+          torch::Tensor step_t, double lr0, double eta_min, double t_max,
+          double beta1, double beta2, double eps) {
+  CK(p); CK(g); CK(m); CK(v);
+  TORCH_CHECK(step_t.scalar_type() == torch::kInt32 && step_t.is_cuda());
+  RUN(fv_adam(fpm(p), fp(g), fpm(m), fpm(v), step_t.data_ptr<int>(), p.numel(),
+              (float)lr0, (float)eta_min, (float)t_max, (float)beta1,
+              (float)beta2, (float)eps, cur_stream()));
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("gemm_nt", &gemm_nt);
+  mod.def("gemm_nn", &gemm_nn);
+  mod.def("gemm_tn", &gemm_tn);
+  mod.def("colsum", &colsum);
+  mod.def("lrelu_bwd", &lrelu_bwd);
+  mod.def("ln_fwd", &ln_fwd);
+  mod.def("ln_bwd_params", &ln_bwd_params);
+  mod.def("gru_fwd", &gru_fwd);
+  mod.def("gru_bwd", &gru_bwd);
+  mod.def("enc_softmax_fwd", &enc_softmax_fwd);
+  mod.def("enc_softmax_bwd", &enc_softmax_bwd);
+  mod.def("enc_heads_fwd", &enc_heads_fwd);
+  mod.def("enc_heads_bwd", &enc_heads_bwd);
+  mod.def("attn_qk_fwd", &attn_qk_fwd);
+  mod.def("attn_softmax_fwd", &attn_softmax_fwd);
+  mod.def("attn_ctx_fwd", &attn_ctx_fwd);
+  mod.def("attn_head_bwd", &attn_head_bwd);
+  mod.def("attn_softmax_bwd", &attn_softmax_bwd);
+  mod.def("attn_qk_bwd", &attn_qk_bwd);
+  mod.def("pred_mlp_fwd", &pred_mlp_fwd);
+  mod.def("pred_mlp_bwd", &pred_mlp_bwd);
+  mod.def("dec_fwd", &dec_fwd);
+  mod.def("dec_bwd", &dec_bwd);
+  mod.def("loss_fwd", &loss_fwd);
+  mod.def("loss_bwd", &loss_bwd);
+  mod.def("step_inc", &step_inc);
+  mod.def("adam", &adam);
+}

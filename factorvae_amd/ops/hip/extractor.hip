@@ -1,0 +1,92 @@
+// Feature-extractor front kernels: LayerNorm forward + parameter backward.
+// (The C->C projection and the GRU input projection are gemm_nt calls;
+// the GRU recurrence is gru.hip.)
+
+#include "common.h"
+
+// One wave per row: x(R,C) -> xln = gamma * (x-mean)*rstd + beta.
+// Saves mean & rstd per row for the backward recompute.
+__global__ __launch_bounds__(256) void ln_fwd_kernel(
+    const float* __restrict__ x, const float* __restrict__ gamma,
+    const float* __restrict__ beta, float* __restrict__ xln,
+    float* __restrict__ mean, float* __restrict__ rstd,
+    long R, int C, float eps) {
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const long row = (long)blockIdx.x * 4 + wid;
+  if (row >= R) return;
+  const float* xr = x + row * C;
+
+  float s = 0.0f;
+  for (int c = lane; c < C; c += 64) s += xr[c];
+  s = wave_reduce_sum(s);
+  s = __shfl(s, 0, 64);
+  const float mu = s / C;
+
+  float v = 0.0f;
+  for (int c = lane; c < C; c += 64) {
+    const float d = xr[c] - mu;
+    v = fmaf(d, d, v);
+  }
+  v = wave_reduce_sum(v);
+  v = __shfl(v, 0, 64);
+  const float rs = rsqrtf(v / C + eps);
+
+  if (lane == 0) {
+    mean[row] = mu;
+    rstd[row] = rs;
+  }
+  float* o = xln + row * C;
+  for (int c = lane; c < C; c += 64)
+    o[c] = fmaf((xr[c] - mu) * rs, gamma[c], beta[c]);
+}
+
+// Parameter grads only (x is input data, no dx needed):
+//   dgamma[c] = sum_r dxln[r][c] * xhat[r][c],  dbeta[c] = sum_r dxln[r][c]
+// xhat recomputed from x, mean, rstd. Grid: (ceil(C/256), 1, r_chunks);
+// atomicAdd into pre-zeroed grad slots.
+__global__ __launch_bounds__(256) void ln_bwd_params_kernel(
+    const float* __restrict__ x, const float* __restrict__ dxln,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    float* __restrict__ dgamma, float* __restrict__ dbeta, long R, int C) {
+  const int c = blockIdx.x * 256 + threadIdx.x;
+  if (c >= C) return;
+  const long chunk = (R + gridDim.z - 1) / gridDim.z;
+  const long rbeg = (long)blockIdx.z * chunk;
+  const long rend = min(rbeg + chunk, R);
+  float dg = 0.0f, db = 0.0f;
+  for (long r = rbeg; r < rend; ++r) {
+    const float g = dxln[r * C + c];
+    const float xh = (x[r * C + c] - mean[r]) * rstd[r];
+    dg = fmaf(g, xh, dg);
+    db += g;
+  }
+  atomicAdd(&dgamma[c], dg);
+  atomicAdd(&dbeta[c], db);
+}
+
+extern "C" {
+
+hipError_t fv_ln_fwd(const float* x, const float* gamma, const float* beta,
+                     float* xln, float* mean, float* rstd, long R, int C,
+                     float eps, hipStream_t stream) {
+  dim3 grid((unsigned)((R + 3) / 4));
+  hipLaunchKernelGGL(ln_fwd_kernel, grid, dim3(256), 0, stream,
+                     x, gamma, beta, xln, mean, rstd, R, C, eps);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_ln_bwd_params(const float* x, const float* dxln,
+                            const float* mean, const float* rstd,
+                            float* dgamma, float* dbeta, long R, int C,
+                            int r_chunks, hipStream_t stream) {
+  if (r_chunks < 1) r_chunks = 1;
+  dim3 grid((C + 255) / 256, 1, r_chunks);
+  hipLaunchKernelGGL(ln_bwd_params_kernel, grid, dim3(256), 0, stream,
+                     x, dxln, mean, rstd, dgamma, dbeta, R, C);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+}  // extern "C"

@@ -1,0 +1,194 @@
+// Persistent GRU recurrence kernels (forward + BPTT backward).
+//
+// The reference's nn.GRU (/root/reference/module.py:20,30) maps to cuDNN
+// RNN kernels; here the whole T-step recurrence is ONE kernel launch:
+// the input projection gi = xp @ W_ih^T + b_ih for ALL timesteps is a
+// single gemm_nt beforehand, and this kernel walks the T-sequential
+// h-dependence with W_hh staged in LDS (transposed for conflict-free
+// lane reads). PyTorch gate order r,z,n preserved:
+//   r = sigmoid(gi_r + gh_r); z = sigmoid(gi_z + gh_z)
+//   q = gh_n (+ b_hh_n);      n = tanh(gi_n + r*q)
+//   h' = (1-z)*n + z*h
+//
+// Geometry: 256 threads = 4 waves; wave <-> stock, lane <-> hidden unit
+// (H <= 64). Grid = ceil(N/4). Stocks are independent, so there is no
+// cross-workgroup traffic at any t.
+//
+// Saved for backward: h_seq (N,T,H), h_prev (N,T,H), gates4 (N,T,4H) =
+// [r, z, n, q].
+
+#include "common.h"
+
+#define GRU_SPW 4  // stocks per workgroup (= waves)
+
+__global__ __launch_bounds__(256) void gru_fwd_kernel(
+    const float* __restrict__ gi,     // (N,T,3H)
+    const float* __restrict__ Whh,    // (3H,H)
+    const float* __restrict__ bhh,    // (3H)
+    float* __restrict__ h_final,      // (N,H)
+    float* __restrict__ h_seq,        // (N,T,H)
+    float* __restrict__ h_prev_out,   // (N,T,H)
+    float* __restrict__ gates4,       // (N,T,4H)
+    int N, int T, int H) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* WhhT = (float*)smem;                   // [H][3H] transposed
+  float* hprev = WhhT + (size_t)H * 3 * H;      // [GRU_SPW][H]
+
+  const int tid = threadIdx.x;
+  const int w = tid >> 6;
+  const int lane = tid & 63;
+  const int s = blockIdx.x * GRU_SPW + w;       // global stock
+  const bool live = (s < N) && (lane < H);
+
+  // stage Whh transposed: WhhT[i][j] = Whh[j][i], conflict-free lane reads
+  for (int idx = tid; idx < 3 * H * H; idx += 256) {
+    const int j = idx / H, i = idx % H;
+    WhhT[(size_t)i * 3 * H + j] = Whh[idx];
+  }
+  if (lane < H) hprev[w * H + lane] = 0.0f;
+  __syncthreads();
+
+  const float br = bhh[lane < H ? lane : 0];
+  const float bz = bhh[lane < H ? H + lane : 0];
+  const float bn = bhh[lane < H ? 2 * H + lane : 0];
+
+  for (int t = 0; t < T; ++t) {
+    float hn = 0.0f;
+    if (live) {
+      float ghr = br, ghz = bz, q = bn;
+      const float* hp = &hprev[w * H];
+      for (int i = 0; i < H; ++i) {
+        const float hv = hp[i];
+        const float* wrow = &WhhT[(size_t)i * 3 * H];
+        ghr = fmaf(hv, wrow[lane], ghr);
+        ghz = fmaf(hv, wrow[H + lane], ghz);
+        q = fmaf(hv, wrow[2 * H + lane], q);
+      }
+      const long base = ((long)s * T + t) * 3 * H;
+      const float gir = gi[base + lane];
+      const float giz = gi[base + H + lane];
+      const float gin = gi[base + 2 * H + lane];
+      const float r = sigmoidf_(gir + ghr);
+      const float z = sigmoidf_(giz + ghz);
+      const float n = tanhf(fmaf(r, q, gin));
+      const float hp_l = hprev[w * H + lane];
+      hn = fmaf(z, hp_l - n, n);  // (1-z)*n + z*hp
+
+      const long ob = ((long)s * T + t) * H + lane;
+      h_seq[ob] = hn;
+      h_prev_out[ob] = hp_l;
+      const long gb = ((long)s * T + t) * 4 * H + lane;
+      gates4[gb] = r;
+      gates4[gb + H] = z;
+      gates4[gb + 2 * H] = n;
+      gates4[gb + 3 * H] = q;
+      if (t == T - 1) h_final[(long)s * H + lane] = hn;
+    }
+    // hprev[w][*] is private to wave w: in-wave LDS dependency only,
+    // no barrier needed between timesteps.
+    if (live) hprev[w * H + lane] = hn;
+  }
+}
+
+// BPTT backward. Inputs: dh_final (N,H) = dL/dh_T, saved tensors.
+// Outputs: dgi (N,T,3H) for the input-projection grads, dgh (N,T,3H) for
+// dW_hh = sum dgh^T h_prev (gemm_tn) and db_hh (colsum).
+__global__ __launch_bounds__(256) void gru_bwd_kernel(
+    const float* __restrict__ dh_final,   // (N,H)
+    const float* __restrict__ h_prev_in,  // (N,T,H)
+    const float* __restrict__ gates4,     // (N,T,4H)
+    const float* __restrict__ Whh,        // (3H,H)
+    float* __restrict__ dgi,              // (N,T,3H)
+    float* __restrict__ dgh,              // (N,T,3H)
+    int N, int T, int H) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* WhhS = (float*)smem;                    // [3H][H] as-is
+  float* dghS = WhhS + (size_t)3 * H * H;        // [GRU_SPW][3H]
+  float* dhS = dghS + (size_t)GRU_SPW * 3 * H;   // [GRU_SPW][H]
+
+  const int tid = threadIdx.x;
+  const int w = tid >> 6;
+  const int lane = tid & 63;
+  const int s = blockIdx.x * GRU_SPW + w;
+  const bool live = (s < N) && (lane < H);
+
+  for (int idx = tid; idx < 3 * H * H; idx += 256) WhhS[idx] = Whh[idx];
+  if (live) dhS[w * H + lane] = dh_final[(long)s * H + lane];
+  __syncthreads();
+
+  for (int t = T - 1; t >= 0; --t) {
+    float zv = 0.0f;
+    if (live) {
+      const long gb = ((long)s * T + t) * 4 * H + lane;
+      const float r = gates4[gb];
+      const float z = gates4[gb + H];
+      const float n = gates4[gb + 2 * H];
+      const float q = gates4[gb + 3 * H];
+      const long hb = ((long)s * T + t) * H + lane;
+      const float hp = h_prev_in[hb];
+      const float dh = dhS[w * H + lane];
+
+      const float dz = dh * (hp - n);
+      const float dn = dh * (1.0f - z);
+      const float da = dn * (1.0f - n * n);
+      const float dgi_n = da;
+      const float dgh_n = da * r;
+      const float dr = da * q;
+      const float dgate_r = dr * r * (1.0f - r);
+      const float dgate_z = dz * z * (1.0f - z);
+
+      const long ob = ((long)s * T + t) * 3 * H + lane;
+      dgi[ob] = dgate_r;
+      dgi[ob + H] = dgate_z;
+      dgi[ob + 2 * H] = dgi_n;
+      dgh[ob] = dgate_r;
+      dgh[ob + H] = dgate_z;
+      dgh[ob + 2 * H] = dgh_n;
+
+      dghS[w * 3 * H + lane] = dgate_r;
+      dghS[w * 3 * H + H + lane] = dgate_z;
+      dghS[w * 3 * H + 2 * H + lane] = dgh_n;
+      zv = z;
+    }
+    // dghS/dhS slices are per-wave private: no barriers in the t-loop.
+    float acc = 0.0f;
+    if (live) {
+      // dh_prev = dh*z + sum_j dgh[j] * Whh[j][lane]
+      acc = dhS[w * H + lane] * zv;
+      const float* dg = &dghS[w * 3 * H];
+      for (int j = 0; j < 3 * H; ++j)
+        acc = fmaf(dg[j], WhhS[(size_t)j * H + lane], acc);
+    }
+    if (live) dhS[w * H + lane] = acc;
+  }
+}
+
+extern "C" {
+
+hipError_t fv_gru_fwd(const float* gi, const float* Whh, const float* bhh,
+                      float* h_final, float* h_seq, float* h_prev,
+                      float* gates4, int N, int T, int H, hipStream_t stream) {
+  if (H > 64) return hipErrorInvalidValue;
+  const size_t lds = ((size_t)H * 3 * H + GRU_SPW * H) * sizeof(float);
+  dim3 grid((N + GRU_SPW - 1) / GRU_SPW);
+  hipLaunchKernelGGL(gru_fwd_kernel, grid, dim3(256), lds, stream,
+                     gi, Whh, bhh, h_final, h_seq, h_prev, gates4, N, T, H);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_gru_bwd(const float* dh_final, const float* h_prev,
+                      const float* gates4, const float* Whh,
+                      float* dgi, float* dgh, int N, int T, int H,
+                      hipStream_t stream) {
+  if (H > 64) return hipErrorInvalidValue;
+  const size_t lds =
+      ((size_t)3 * H * H + GRU_SPW * 3 * H + GRU_SPW * H) * sizeof(float);
+  dim3 grid((N + GRU_SPW - 1) / GRU_SPW);
+  hipLaunchKernelGGL(gru_bwd_kernel, grid, dim3(256), lds, stream,
+                     dh_final, h_prev, gates4, Whh, dgi, dgh, N, T, H);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+}  // extern "C"

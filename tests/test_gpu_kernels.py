@@ -1,0 +1,406 @@
+"""GPU kernel numerics tests: every HIP kernel against a plain PyTorch
+fp32 eager reference (SURVEY.md §4 test strategy (a)/(b))."""
+
+import math
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from factorvae_amd.ops import get_extension
+    ext = get_extension()
+DEV = torch.device("cuda:0")
+
+
+def t(*shape, seed=0, scale=1.0):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    return (torch.randn(*shape, generator=g) * scale).to(DEV)
+
+
+def assert_close(a, b, atol=1e-4, rtol=1e-4, what=""):
+    torch.testing.assert_close(a, b, atol=atol, rtol=rtol, msg=lambda m: f"{what}: {m}")
+
+
+# ------------------------------------------------------------------ GEMM
+@pytest.mark.parametrize("R,Ci,Co", [(300, 158, 192), (37, 33, 20), (64, 64, 64), (6000, 158, 158)])
+def test_gemm_nt(R, Ci, Co):
+    A, W, b = t(R, Ci, seed=1), t(Co, Ci, seed=2), t(Co, seed=3)
+    out = torch.empty(R, Co, device=DEV)
+    ext.gemm_nt(A, W, b, out, 1.0, False, False)
+    torch.cuda.synchronize()
+    assert_close(out, A @ W.t() + b, what="gemm_nt")
+    # lrelu + alpha + accumulate
+    out2 = out.clone()
+    ext.gemm_nt(A, W, b, out2, 0.5, True, True)
+    torch.cuda.synchronize()
+    ref = out + F.leaky_relu(0.5 * (A @ W.t() + b), 0.01)
+    assert_close(out2, ref, what="gemm_nt acc+lrelu")
+
+
+@pytest.mark.parametrize("R,Ci,Co", [(300, 192, 158), (41, 20, 37)])
+def test_gemm_nn(R, Ci, Co):
+    A, B = t(R, Ci, seed=4), t(Ci, Co, seed=5)
+    out = torch.empty(R, Co, device=DEV)
+    ext.gemm_nn(A, B, None, out, 1.0, False, False)
+    torch.cuda.synchronize()
+    assert_close(out, A @ B, what="gemm_nn")
+
+
+@pytest.mark.parametrize("R,M,N,chunks", [(300, 20, 64, 1), (6000, 192, 64, 8), (77, 33, 17, 3)])
+def test_gemm_tn(R, M, N, chunks):
+    A, B = t(R, M, seed=6), t(R, N, seed=7)
+    out = torch.zeros(M, N, device=DEV)
+    ext.gemm_tn(A, B, out, chunks, chunks > 1)
+    torch.cuda.synchronize()
+    assert_close(out, A.t() @ B, atol=5e-4, rtol=5e-4, what="gemm_tn")
+
+
+def test_colsum():
+    A = t(1234, 77, seed=8)
+    out = torch.zeros(77, device=DEV)
+    ext.colsum(A, out, 4)
+    torch.cuda.synchronize()
+    assert_close(out, A.sum(dim=0), atol=5e-4, rtol=5e-4, what="colsum")
+
+
+def test_lrelu_bwd():
+    Y = F.leaky_relu(t(100, 50, seed=9), 0.01)
+    dY = t(100, 50, seed=10)
+    dZ = torch.empty_like(dY)
+    ext.lrelu_bwd(dY, Y, dZ)
+    torch.cuda.synchronize()
+    assert_close(dZ, dY * torch.where(Y > 0, 1.0, 0.01), what="lrelu_bwd")
+
+
+# ------------------------------------------------------------------ LN
+def test_ln_fwd():
+    R, C = 500, 158
+    x, g_, b_ = t(R, C, seed=11), t(C, seed=12), t(C, seed=13)
+    xln = torch.empty(R, C, device=DEV)
+    mean = torch.empty(R, device=DEV)
+    rstd = torch.empty(R, device=DEV)
+    ext.ln_fwd(x, g_, b_, xln, mean, rstd, 1e-5)
+    torch.cuda.synchronize()
+    ref = F.layer_norm(x, (C,), g_, b_, 1e-5)
+    assert_close(xln, ref, what="ln_fwd")
+    assert_close(mean, x.mean(dim=1), what="ln mean")
+
+
+def test_ln_bwd_params():
+    R, C = 400, 63
+    x = t(R, C, seed=14)
+    dxln = t(R, C, seed=15)
+    xr = x.clone().requires_grad_(True)
+    g_ = t(C, seed=16).requires_grad_(True)
+    b_ = t(C, seed=17).requires_grad_(True)
+    F.layer_norm(xr, (C,), g_, b_, 1e-5).backward(dxln)
+
+    xln = torch.empty(R, C, device=DEV)
+    mean = torch.empty(R, device=DEV)
+    rstd = torch.empty(R, device=DEV)
+    ext.ln_fwd(x, g_.detach(), b_.detach(), xln, mean, rstd, 1e-5)
+    dg = torch.zeros(C, device=DEV)
+    db = torch.zeros(C, device=DEV)
+    ext.ln_bwd_params(x, dxln, mean, rstd, dg, db, 3)
+    torch.cuda.synchronize()
+    assert_close(dg, g_.grad, atol=5e-4, rtol=5e-4, what="dgamma")
+    assert_close(db, b_.grad, atol=5e-4, rtol=5e-4, what="dbeta")
+
+
+# ------------------------------------------------------------------ GRU
+@pytest.mark.parametrize("N,T,H,C", [(300, 20, 64, 158), (13, 5, 48, 33)])
+def test_gru_fwd_bwd(N, T, H, C):
+    torch.manual_seed(20)
+    gru = torch.nn.GRU(C, H, 1, batch_first=True).to(DEV)
+    xp = t(N, T, C, seed=21)
+
+    # reference fwd/bwd via autograd
+    xp_r = xp.clone().requires_grad_(True)
+    out, _ = gru(xp_r)
+    href = out[:, -1, :]
+    dh = t(N, H, seed=22)
+    href.backward(dh)
+
+    # fused: gi via gemm, then recurrence kernel
+    gi = torch.empty(N * T, 3 * H, device=DEV)
+    ext.gemm_nt(xp.view(N * T, C), gru.weight_ih_l0.detach(),
+                gru.bias_ih_l0.detach(), gi, 1.0, False, False)
+    h_final = torch.empty(N, H, device=DEV)
+    h_seq = torch.empty(N, T, H, device=DEV)
+    h_prev = torch.empty(N, T, H, device=DEV)
+    gates4 = torch.empty(N, T, 4 * H, device=DEV)
+    ext.gru_fwd(gi.view(N, T, 3 * H), gru.weight_hh_l0.detach(),
+                gru.bias_hh_l0.detach(), h_final, h_seq, h_prev, gates4, N, T, H)
+    torch.cuda.synchronize()
+    assert_close(h_final, href.detach(), atol=2e-5, rtol=2e-5, what="gru h_final")
+
+    dgi = torch.empty(N, T, 3 * H, device=DEV)
+    dgh = torch.empty(N, T, 3 * H, device=DEV)
+    ext.gru_bwd(dh, h_prev, gates4, gru.weight_hh_l0.detach(), dgi, dgh, N, T, H)
+    # dWhh, dbhh, dWih, dbih, dxp via gemms
+    dWhh = torch.zeros(3 * H, H, device=DEV)
+    ext.gemm_tn(dgh.view(N * T, 3 * H), h_prev.view(N * T, H), dWhh, 4, True)
+    dbhh = torch.zeros(3 * H, device=DEV)
+    ext.colsum(dgh.view(N * T, 3 * H), dbhh, 4)
+    dWih = torch.zeros(3 * H, C, device=DEV)
+    ext.gemm_tn(dgi.view(N * T, 3 * H), xp.view(N * T, C), dWih, 4, True)
+    dbih = torch.zeros(3 * H, device=DEV)
+    ext.colsum(dgi.view(N * T, 3 * H), dbih, 4)
+    dxp = torch.empty(N * T, C, device=DEV)
+    ext.gemm_nn(dgi.view(N * T, 3 * H), gru.weight_ih_l0.detach(), None, dxp,
+                1.0, False, False)
+    torch.cuda.synchronize()
+
+    assert_close(dWhh, gru.weight_hh_l0.grad, atol=1e-3, rtol=1e-3, what="dWhh")
+    assert_close(dbhh, gru.bias_hh_l0.grad, atol=1e-3, rtol=1e-3, what="dbhh")
+    assert_close(dWih, gru.weight_ih_l0.grad, atol=1e-3, rtol=1e-3, what="dWih")
+    assert_close(dbih, gru.bias_ih_l0.grad, atol=1e-3, rtol=1e-3, what="dbih")
+    assert_close(dxp.view(N, T, C), xp_r.grad, atol=1e-3, rtol=1e-3, what="dxp")
+
+
+# ------------------------------------------------------------------ encoder
+def test_enc_softmax_and_heads():
+    N, M, K = 300, 128, 20
+    scores, y = t(N, M, seed=30), t(N, 1, seed=31)
+    a = torch.empty(N, M, device=DEV)
+    yp = torch.empty(M, device=DEV)
+    ext.enc_softmax_fwd(scores, y, a, yp)
+    torch.cuda.synchronize()
+    a_ref = torch.softmax(scores, dim=0)
+    assert_close(a, a_ref, what="enc softmax")
+    assert_close(yp, (a_ref.t() @ y).squeeze(1), atol=2e-5, rtol=2e-5, what="yp")
+
+    Wmu, bmu = t(K, M, seed=32), t(K, seed=33)
+    Wsig, bsig = t(K, M, seed=34), t(K, seed=35)
+    fmu = torch.empty(K, device=DEV)
+    fsig_pre = torch.empty(K, device=DEV)
+    fsig = torch.empty(K, device=DEV)
+    fsig_c = torch.empty(K, device=DEV)
+    ext.enc_heads_fwd(yp, Wmu, bmu, Wsig, bsig, fmu, fsig_pre, fsig, fsig_c)
+    torch.cuda.synchronize()
+    assert_close(fmu, yp @ Wmu.t() + bmu, atol=2e-5, rtol=2e-5, what="fmu")
+    assert_close(fsig, F.softplus(yp @ Wsig.t() + bsig), atol=2e-5, rtol=2e-5, what="fsig")
+    assert (fsig_c > 0).all()
+
+
+def test_enc_bwd():
+    N, M, K = 120, 64, 16
+    scores = t(N, M, seed=36).requires_grad_(True)
+    y = t(N, 1, seed=37)
+    Wmu = t(K, M, seed=38).requires_grad_(True)
+    bmu = t(K, seed=39).requires_grad_(True)
+    Wsig = t(K, M, seed=40).requires_grad_(True)
+    bsig = t(K, seed=41).requires_grad_(True)
+
+    a_ref = torch.softmax(scores, dim=0)
+    yp_ref = (a_ref.t() @ y).squeeze(1)
+    fmu_ref = yp_ref @ Wmu.t() + bmu
+    fsig_ref = F.softplus(yp_ref @ Wsig.t() + bsig)
+    dfmu, dfsig = t(K, seed=42), t(K, seed=43)
+    (fmu_ref * dfmu + fsig_ref * dfsig).sum().backward()
+
+    # fused path
+    a = torch.empty(N, M, device=DEV)
+    yp = torch.empty(M, device=DEV)
+    ext.enc_softmax_fwd(scores.detach(), y, a, yp)
+    fmu = torch.empty(K, device=DEV)
+    fsig_pre = torch.empty(K, device=DEV)
+    fsig = torch.empty(K, device=DEV)
+    fsig_c = torch.empty(K, device=DEV)
+    ext.enc_heads_fwd(yp, Wmu.detach(), bmu.detach(), Wsig.detach(),
+                      bsig.detach(), fmu, fsig_pre, fsig, fsig_c)
+    dyp = torch.empty(M, device=DEV)
+    gWmu = torch.zeros(K, M, device=DEV)
+    gbmu = torch.zeros(K, device=DEV)
+    gWsig = torch.zeros(K, M, device=DEV)
+    gbsig = torch.zeros(K, device=DEV)
+    ext.enc_heads_bwd(dfmu, dfsig, fsig, fsig_pre, yp, Wmu.detach(),
+                      Wsig.detach(), dyp, gWmu, gbmu, gWsig, gbsig)
+    dscores = torch.empty(N, M, device=DEV)
+    ext.enc_softmax_bwd(dyp, a, y, dscores)
+    torch.cuda.synchronize()
+
+    assert_close(gWmu, Wmu.grad, atol=5e-5, rtol=5e-5, what="dWmu")
+    assert_close(gbmu, bmu.grad, atol=5e-5, rtol=5e-5, what="dbmu")
+    assert_close(gWsig, Wsig.grad, atol=5e-5, rtol=5e-5, what="dWsig")
+    assert_close(dscores, scores.grad, atol=5e-5, rtol=5e-5, what="dscores")
+
+
+# ------------------------------------------------------------------ attention + MLP + decoder + loss: full-step parity
+def eager_forward_explicit_noise(model, x, y, eps, mask, training):
+    """Reference-math forward with EXPLICIT noise tensors (same draws the
+    fused engine consumes), so fused vs eager compare exactly."""
+    m = model
+    h = m.feature_extractor(x)
+    fmu, fsig = m.factor_encoder(h, y)
+    fsig_c = torch.where(fsig == 0, torch.full_like(fsig, 1e-6), fsig)
+
+    # decoder with explicit eps
+    dec = m.factor_decoder
+    amu, asig = dec.alpha_layer(h)
+    beta = dec.beta_layer(h)
+    mu = amu + beta @ fmu.view(-1, 1)
+    sig = torch.sqrt(asig ** 2 + (beta ** 2) @ (fsig_c.view(-1, 1) ** 2) + 1e-6)
+    recon = mu + eps.view(-1, 1) * sig
+
+    # predictor with explicit dropout mask
+    pred = m.factor_predictor
+    K, H = pred.num_factor, pred.hidden_size
+    q = torch.stack([l.query for l in pred.attention_layers])
+    Wk = torch.stack([l.key_layer.weight for l in pred.attention_layers])
+    bk = torch.stack([l.key_layer.bias for l in pred.attention_layers])
+    Wv = torch.stack([l.value_layer.weight for l in pred.attention_layers])
+    bv = torch.stack([l.value_layer.bias for l in pred.attention_layers])
+    qk = torch.einsum("khj,kh->kj", Wk, q)
+    c = (q * bk).sum(dim=1)
+    scale = math.sqrt(H + 1e-6)
+    s = (h @ qk.t() + c) / scale
+    if training:
+        s = s * mask / 0.9
+    a = torch.softmax(F.relu(s), dim=0)
+    u = a.t() @ h
+    ctx = torch.einsum("kij,kj->ki", Wv, u) + bv
+    hm2 = F.leaky_relu(pred.linear(ctx), 0.01)
+    pmu = pred.mu_layer(hm2).view(-1)
+    psig = F.softplus(pred.sigma_layer(hm2)).view(-1)
+    psig_c = torch.where(psig == 0, torch.full_like(psig, 1e-6), psig)
+
+    mse = F.mse_loss(recon, y)
+    kl = (torch.log(psig_c / fsig_c)
+          + (fsig_c ** 2 + (fmu - pmu) ** 2) / (2 * psig_c ** 2) - 0.5).sum()
+    return mse + kl, recon
+
+
+@pytest.mark.parametrize("N,T,C,H,M,K,training", [
+    (300, 20, 158, 64, 128, 20, True),
+    (300, 20, 158, 64, 128, 20, False),
+    (37, 7, 33, 48, 24, 12, True),
+])
+def test_full_step_grad_parity(N, T, C, H, M, K, training):
+    from factorvae_amd.engine.fused import FusedTrainer
+    from factorvae_amd.models.modules import build_factorvae
+    from factorvae_amd.utils import set_seed
+
+    set_seed(0)
+    model = build_factorvae(num_latent=C, hidden_size=H, num_portfolio=M,
+                            num_factor=K).to(DEV)
+    trainer = FusedTrainer(model, lr=1e-4, t_max=100, device=DEV,
+                           use_graph=False, train=training)
+    x = t(N, T, C, seed=50)
+    y = t(N, 1, seed=51)
+
+    trainer._ensure_ws(N, T)
+    w = trainer.ws
+    w["x"].copy_(x)
+    w["y"].copy_(y)
+    trainer._fill_rng(N)
+    eps = w["eps"].clone()
+    mask = w["mask"].clone()
+
+    trainer.grads.zero_()
+    trainer._launch_forward(N, T)
+    trainer._launch_backward(N, T)
+    torch.cuda.synchronize()
+    fused_loss = w["loss"].item()
+
+    # eager oracle with the same noise
+    for p in model.parameters():
+        p.grad = None
+    loss, recon = eager_forward_explicit_noise(model, x, y, eps, mask, training)
+    loss.backward()
+
+    assert abs(fused_loss - loss.item()) < 1e-3 * max(1.0, abs(loss.item())), \
+        f"loss mismatch: fused={fused_loss} eager={loss.item()}"
+    assert_close(w["recon"].view(N, 1), recon.detach(), atol=1e-4, rtol=1e-4,
+                 what="recon")
+
+    for name, param, _ in trainer._param_specs():
+        gk = trainer.g(name)
+        ref = param.grad
+        assert ref is not None, name
+        torch.testing.assert_close(gk, ref, atol=2e-3, rtol=2e-3,
+                                   msg=lambda m, n=name: f"grad {n}: {m}")
+
+
+def test_fused_step_runs_and_graph_matches_eager_launch():
+    from factorvae_amd.engine.fused import FusedTrainer
+    from factorvae_amd.models.modules import build_factorvae
+    from factorvae_amd.utils import set_seed
+
+    N, T, C, H, M, K = 300, 20, 158, 64, 128, 20
+
+    def run(use_graph, steps=5):
+        set_seed(0)
+        model = build_factorvae(num_latent=C, hidden_size=H, num_portfolio=M,
+                                num_factor=K).to(DEV)
+        trainer = FusedTrainer(model, lr=1e-3, t_max=steps, device=DEV,
+                               use_graph=use_graph)
+        torch.manual_seed(7)
+        losses = []
+        for i in range(steps):
+            x = torch.randn(N, T, C, device=DEV)
+            y = torch.randn(N, 1, device=DEV)
+            loss = trainer.step(x, y)
+            losses.append(loss.item())
+        return losses, trainer.params.flat.clone()
+
+    l1, p1 = run(False)
+    l2, p2 = run(True)
+    for a, b in zip(l1, l2):
+        assert abs(a - b) < 1e-3 * max(1.0, abs(a)), (l1, l2)
+    torch.testing.assert_close(p1, p2, atol=1e-5, rtol=1e-5)
+    assert all(v == v for v in l1), "NaN loss"
+
+
+def test_fused_adam_matches_torch_adam():
+    from factorvae_amd.engine.fused import FusedTrainer
+    from factorvae_amd.models.modules import build_factorvae
+    from factorvae_amd.utils import set_seed
+
+    N, T, C, H, M, K = 64, 6, 30, 32, 16, 8
+    set_seed(3)
+    model_f = build_factorvae(num_latent=C, hidden_size=H, num_portfolio=M,
+                              num_factor=K).to(DEV)
+    set_seed(3)
+    model_e = build_factorvae(num_latent=C, hidden_size=H, num_portfolio=M,
+                              num_factor=K).to(DEV)
+    for (n1, p1_), (n2, p2_) in zip(model_f.named_parameters(),
+                                    model_e.named_parameters()):
+        assert torch.equal(p1_, p2_), n1
+
+    steps = 4
+    trainer = FusedTrainer(model_f, lr=1e-3, t_max=steps, device=DEV,
+                           use_graph=False, train=False)  # no dropout
+    opt = torch.optim.Adam(model_e.parameters(), lr=1e-3)
+    sched = torch.optim.lr_scheduler.CosineAnnealingLR(opt, T_max=steps)
+
+    for i in range(steps):
+        x = t(N, T, C, seed=100 + i)
+        y = t(N, 1, seed=200 + i)
+        trainer._ensure_ws(N, T)
+        w = trainer.ws
+        w["x"].copy_(x)
+        w["y"].copy_(y)
+        trainer._fill_rng(N)
+        eps = w["eps"].clone()
+        trainer.grads.zero_()
+        trainer._launch_forward(N, T)
+        trainer._launch_backward(N, T)
+        trainer._launch_optimizer()
+
+        model_e.eval()
+        opt.zero_grad(set_to_none=True)
+        loss, _ = eager_forward_explicit_noise(model_e, x, y, eps, None, False)
+        loss.backward()
+        opt.step()
+        sched.step()
+
+    torch.cuda.synchronize()
+    fused_sd = model_f.state_dict()
+    eager_sd = model_e.state_dict()
+    for k in fused_sd:
+        torch.testing.assert_close(fused_sd[k], eager_sd[k], atol=5e-4,
+                                   rtol=5e-4, msg=lambda m, kk=k: f"{kk}: {m}")
