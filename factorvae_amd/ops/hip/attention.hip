@@ -13,23 +13,33 @@
 
 // qk[k][i] = sum_j q[k][j] * Wk[k][j][i];  c[k] = sum_j q[k][j]*bk[k][j]
 // grid K, 64 threads (one wave).
-__global__ __launch_bounds__(64) void attn_qk_fwd_kernel(
+__global__ __launch_bounds__(256) void attn_qk_fwd_kernel(
     const float* __restrict__ q, const float* __restrict__ Wk,
     const float* __restrict__ bk, float* __restrict__ qk,
     float* __restrict__ c, int K, int H) {
+  __shared__ float part[4][64];
   const int k = blockIdx.x;
-  const int lane = threadIdx.x;
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
   const float* qh = q + (long)k * H;
   const float* W = Wk + (long)k * H * H;
 
+  // 4 waves split the j-reduction; lane <-> output column i (coalesced)
   float acc = 0.0f;
   if (lane < H) {
-    for (int j = 0; j < H; ++j) acc = fmaf(qh[j], W[(long)j * H + lane], acc);
-    qk[(long)k * H + lane] = acc;
+    for (int j = w; j < H; j += 4)
+      acc = fmaf(qh[j], W[(long)j * H + lane], acc);
   }
-  float cv = (lane < H) ? qh[lane] * bk[(long)k * H + lane] : 0.0f;
-  cv = wave_reduce_sum(cv);
-  if (lane == 0) c[k] = cv;
+  part[w][lane] = acc;
+  __syncthreads();
+  if (w == 0) {
+    if (lane < H)
+      qk[(long)k * H + lane] =
+          part[0][lane] + part[1][lane] + part[2][lane] + part[3][lane];
+    float cv = (lane < H) ? qh[lane] * bk[(long)k * H + lane] : 0.0f;
+    cv = wave_reduce_sum(cv);
+    if (lane == 0) c[k] = cv;
+  }
 }
 
 // Per-head column: dropout (train) -> relu -> softmax over N -> NaN guard.
@@ -78,53 +88,66 @@ __global__ __launch_bounds__(256) void attn_softmax_fwd_kernel(
 }
 
 // ctx[k][j] = sum_i Wv[k][j][i] * u[k][i] + bv[k][j]; guarded heads -> 0.
-__global__ __launch_bounds__(64) void attn_ctx_fwd_kernel(
+__global__ __launch_bounds__(256) void attn_ctx_fwd_kernel(
     const float* __restrict__ u, const float* __restrict__ Wv,
     const float* __restrict__ bv, const int* __restrict__ guard,
     float* __restrict__ ctx, int K, int H) {
   const int k = blockIdx.x;
-  const int j = threadIdx.x;
-  if (j >= H) return;
-  if (guard[k]) {
-    ctx[(long)k * H + j] = 0.0f;
-    return;
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const bool g = guard[k] != 0;
+  const float ul = (lane < H) ? u[(long)k * H + lane] : 0.0f;
+  const int jpw = (H + 3) / 4;  // output rows per wave
+  // wave w computes ctx[j] for its j-chunk: lane <-> i (coalesced Wv row)
+  for (int j = w * jpw; j < min((w + 1) * jpw, H); ++j) {
+    float v = (!g && lane < H) ? Wv[((long)k * H + j) * H + lane] * ul : 0.0f;
+    v = wave_reduce_sum(v);
+    if (lane == 0)
+      ctx[(long)k * H + j] = g ? 0.0f : v + bv[(long)k * H + j];
   }
-  const float* W = Wv + ((long)k * H + j) * H;
-  const float* uh = u + (long)k * H;
-  float acc = bv[(long)k * H + j];
-  for (int i = 0; i < H; ++i) acc = fmaf(W[i], uh[i], acc);
-  ctx[(long)k * H + j] = acc;
 }
 
 // Backward of the value path: du[k][i] = sum_j Wv[k][j][i]*dctx[k][j];
 // dWv[k][j][i] = dctx[k][j]*u[k][i]; dbv[k][j] = dctx[k][j].
 // Guarded heads contribute zero (reference returns a detached zeros
 // context). Param grads are plain += (each element owned by one WG).
-__global__ __launch_bounds__(64) void attn_head_bwd_kernel(
+__global__ __launch_bounds__(256) void attn_head_bwd_kernel(
     const float* __restrict__ dctx_in, const float* __restrict__ u,
     const float* __restrict__ Wv, const int* __restrict__ guard,
     float* __restrict__ du, float* __restrict__ dWv, float* __restrict__ dbv,
     int K, int H) {
+  __shared__ float part[4][64];
+  __shared__ float dcS[64];
+  __shared__ float uS[64];
   const int k = blockIdx.x;
-  const int lane = threadIdx.x;
-  if (lane >= H) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
   const bool g = guard[k] != 0;
-  const float* dctx = dctx_in + (long)k * H;
-  const float* uh = u + (long)k * H;
-
-  // dWv row j = lane: dctx[lane] * u[:]
-  const float dcj = g ? 0.0f : dctx[lane];
-  float* dWrow = dWv + ((long)k * H + lane) * H;
-  for (int i = 0; i < H; ++i) dWrow[i] += dcj * uh[i];
-  dbv[(long)k * H + lane] += dcj;
-
-  // du[i = lane] = sum_j Wv[j][i] * dctx[j]
-  float acc = 0.0f;
-  if (!g) {
-    const float* W = Wv + (long)k * H * H;
-    for (int j = 0; j < H; ++j) acc = fmaf(W[(long)j * H + lane], dctx[j], acc);
+  if (tid < H) {
+    dcS[tid] = g ? 0.0f : dctx_in[(long)k * H + tid];
+    uS[tid] = u[(long)k * H + tid];
   }
-  du[(long)k * H + lane] = acc;
+  __syncthreads();
+
+  // dWv[j][i] += dctx[j]*u[i] — flat coalesced sweep
+  float* dW = dWv + (long)k * H * H;
+  for (int idx = tid; idx < H * H; idx += 256)
+    dW[idx] += dcS[idx / H] * uS[idx % H];
+  if (tid < H) dbv[(long)k * H + tid] += dcS[tid];
+
+  // du[i] = sum_j Wv[j][i]*dctx[j] — 4 waves split j, lane <-> i
+  float acc = 0.0f;
+  if (!g && lane < H) {
+    const float* W = Wv + (long)k * H * H;
+    for (int j = w; j < H; j += 4)
+      acc = fmaf(W[(long)j * H + lane], dcS[j], acc);
+  }
+  part[w][lane] = acc;
+  __syncthreads();
+  if (w == 0 && lane < H)
+    du[(long)k * H + lane] =
+        part[0][lane] + part[1][lane] + part[2][lane] + part[3][lane];
 }
 
 // Softmax+relu+dropout backward per head column; also emits
@@ -168,62 +191,81 @@ __global__ __launch_bounds__(256) void attn_softmax_bwd_kernel(
 
 // dq[k][j] = sum_i Wk[k][j][i]*dqk[k][i] + dc[k]*bk[k][j];
 // dWk[k][j][i] = q[k][j]*dqk[k][i]; dbk[k][j] = dc[k]*q[k][j].
-__global__ __launch_bounds__(64) void attn_qk_bwd_kernel(
+__global__ __launch_bounds__(256) void attn_qk_bwd_kernel(
     const float* __restrict__ dqk, const float* __restrict__ dc,
     const float* __restrict__ q, const float* __restrict__ Wk,
     const float* __restrict__ bk, float* __restrict__ dq,
     float* __restrict__ dWk, float* __restrict__ dbk, int K, int H) {
+  __shared__ float dqkS[64];
+  __shared__ float qS[64];
   const int k = blockIdx.x;
-  const int lane = threadIdx.x;
-  if (lane >= H) return;
-  const float* dqkh = dqk + (long)k * H;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
   const float dck = dc[k];
-  const float qj = q[(long)k * H + lane];
+  if (tid < H) {
+    dqkS[tid] = dqk[(long)k * H + tid];
+    qS[tid] = q[(long)k * H + tid];
+  }
+  __syncthreads();
 
-  float* dWrow = dWk + ((long)k * H + lane) * H;
-  for (int i = 0; i < H; ++i) dWrow[i] += qj * dqkh[i];
-  dbk[(long)k * H + lane] += dck * qj;
+  // dWk[j][i] += q[j]*dqk[i] — flat coalesced sweep
+  float* dW = dWk + (long)k * H * H;
+  for (int idx = tid; idx < H * H; idx += 256)
+    dW[idx] += qS[idx / H] * dqkS[idx % H];
+  if (tid < H) dbk[(long)k * H + tid] += dck * qS[tid];
 
-  const float* W = Wk + (long)k * H * H;
-  float acc = dck * bk[(long)k * H + lane];
-  const float* Wrow = W + (long)lane * H;
-  for (int i = 0; i < H; ++i) acc = fmaf(Wrow[i], dqkh[i], acc);
-  dq[(long)k * H + lane] += acc;
+  // dq[j] = dc*bk[j] + sum_i Wk[j][i]*dqk[i] — wave per j-chunk, lane <-> i
+  const int jpw = (H + 3) / 4;
+  for (int j = w * jpw; j < min((w + 1) * jpw, H); ++j) {
+    float v = (lane < H) ? Wk[((long)k * H + j) * H + lane] * dqkS[lane] : 0.0f;
+    v = wave_reduce_sum(v);
+    if (lane == 0) dq[(long)k * H + j] += dck * bk[(long)k * H + j] + v;
+  }
 }
 
 // Predictor shared MLP forward: hm2 = lrelu(ctx@Wl^T + bl);
 // pmu = hm2·wmu + bmu; psig = softplus(hm2·wsig + bsig); clamp ==0 -> 1e-6
 // (module.py:264-265). One wave per head row.
-__global__ __launch_bounds__(64) void pred_mlp_fwd_kernel(
+__global__ __launch_bounds__(256) void pred_mlp_fwd_kernel(
     const float* __restrict__ ctx, const float* __restrict__ Wl,
     const float* __restrict__ bl, const float* __restrict__ wmu,
     const float* __restrict__ bmu, const float* __restrict__ wsig,
     const float* __restrict__ bsig, float* __restrict__ hm2,
     float* __restrict__ pmu, float* __restrict__ psig_pre,
     float* __restrict__ psig, float* __restrict__ psig_c, int K, int H) {
+  __shared__ float hm2S[64];
   const int k = blockIdx.x;
-  const int lane = threadIdx.x;
-  const float* cr = ctx + (long)k * H;
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const float cl = (lane < H) ? ctx[(long)k * H + lane] : 0.0f;
 
-  float z = 0.0f;
-  if (lane < H) {
-    z = bl[lane];
-    const float* Wrow = Wl + (long)lane * H;
-    for (int i = 0; i < H; ++i) z = fmaf(cr[i], Wrow[i], z);
-    z = lrelu_(z);
-    hm2[(long)k * H + lane] = z;
+  // z2[j] = ctx · Wl[j] + bl[j]: wave per j-chunk, lane <-> i (coalesced)
+  const int jpw = (H + 3) / 4;
+  for (int j = w * jpw; j < min((w + 1) * jpw, H); ++j) {
+    float v = (lane < H) ? Wl[(long)j * H + lane] * cl : 0.0f;
+    v = wave_reduce_sum(v);
+    if (lane == 0) {
+      const float z = lrelu_(v + bl[j]);
+      hm2S[j] = z;
+      hm2[(long)k * H + j] = z;
+    }
   }
-  float pm = (lane < H) ? z * wmu[lane] : 0.0f;
-  float ps = (lane < H) ? z * wsig[lane] : 0.0f;
-  pm = wave_reduce_sum(pm);
-  ps = wave_reduce_sum(ps);
-  if (lane == 0) {
-    pmu[k] = pm + bmu[0];
-    const float pre = ps + bsig[0];
-    psig_pre[k] = pre;
-    const float s = softplusf_(pre);
-    psig[k] = s;
-    psig_c[k] = (s == 0.0f) ? 1e-6f : s;
+  __syncthreads();
+  if (w == 0) {
+    const float z = (lane < H) ? hm2S[lane] : 0.0f;
+    float pm = (lane < H) ? z * wmu[lane] : 0.0f;
+    float ps = (lane < H) ? z * wsig[lane] : 0.0f;
+    pm = wave_reduce_sum(pm);
+    ps = wave_reduce_sum(ps);
+    if (lane == 0) {
+      pmu[k] = pm + bmu[0];
+      const float pre = ps + bsig[0];
+      psig_pre[k] = pre;
+      const float sp = softplusf_(pre);
+      psig[k] = sp;
+      psig_c[k] = (sp == 0.0f) ? 1e-6f : sp;
+    }
   }
 }
 
@@ -260,7 +302,7 @@ extern "C" {
 hipError_t fv_attn_qk_fwd(const float* q, const float* Wk, const float* bk,
                           float* qk, float* c, int K, int H, hipStream_t s) {
   if (H > 64) return hipErrorInvalidValue;
-  hipLaunchKernelGGL(attn_qk_fwd_kernel, dim3(K), dim3(64), 0, s, q, Wk, bk, qk, c, K, H);
+  hipLaunchKernelGGL(attn_qk_fwd_kernel, dim3(K), dim3(256), 0, s, q, Wk, bk, qk, c, K, H);
   HIP_CHECK_LAST();
   return hipSuccess;
 }
@@ -278,7 +320,7 @@ hipError_t fv_attn_ctx_fwd(const float* u, const float* Wv, const float* bv,
                            const int* guard, float* ctx, int K, int H,
                            hipStream_t s) {
   if (H > 64) return hipErrorInvalidValue;
-  hipLaunchKernelGGL(attn_ctx_fwd_kernel, dim3(K), dim3(64), 0, s,
+  hipLaunchKernelGGL(attn_ctx_fwd_kernel, dim3(K), dim3(256), 0, s,
                      u, Wv, bv, guard, ctx, K, H);
   HIP_CHECK_LAST();
   return hipSuccess;
@@ -288,7 +330,7 @@ hipError_t fv_attn_head_bwd(const float* dctx, const float* u, const float* Wv,
                             const int* guard, float* du, float* dWv, float* dbv,
                             int K, int H, hipStream_t s) {
   if (H > 64) return hipErrorInvalidValue;
-  hipLaunchKernelGGL(attn_head_bwd_kernel, dim3(K), dim3(64), 0, s,
+  hipLaunchKernelGGL(attn_head_bwd_kernel, dim3(K), dim3(256), 0, s,
                      dctx, u, Wv, guard, du, dWv, dbv, K, H);
   HIP_CHECK_LAST();
   return hipSuccess;
@@ -308,7 +350,7 @@ hipError_t fv_attn_qk_bwd(const float* dqk, const float* dc, const float* q,
                           const float* Wk, const float* bk, float* dq,
                           float* dWk, float* dbk, int K, int H, hipStream_t s) {
   if (H > 64) return hipErrorInvalidValue;
-  hipLaunchKernelGGL(attn_qk_bwd_kernel, dim3(K), dim3(64), 0, s,
+  hipLaunchKernelGGL(attn_qk_bwd_kernel, dim3(K), dim3(256), 0, s,
                      dqk, dc, q, Wk, bk, dq, dWk, dbk, K, H);
   HIP_CHECK_LAST();
   return hipSuccess;
@@ -320,7 +362,7 @@ hipError_t fv_pred_mlp_fwd(const float* ctx, const float* Wl, const float* bl,
                            float* psig_pre, float* psig, float* psig_c,
                            int K, int H, hipStream_t s) {
   if (H > 64) return hipErrorInvalidValue;
-  hipLaunchKernelGGL(pred_mlp_fwd_kernel, dim3(K), dim3(64), 0, s,
+  hipLaunchKernelGGL(pred_mlp_fwd_kernel, dim3(K), dim3(256), 0, s,
                      ctx, Wl, bl, wmu, bmu, wsig, bsig, hm2, pmu, psig_pre,
                      psig, psig_c, K, H);
   HIP_CHECK_LAST();

@@ -43,26 +43,36 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
 
 // Parameter grads only (x is input data, no dx needed):
 //   dgamma[c] = sum_r dxln[r][c] * xhat[r][c],  dbeta[c] = sum_r dxln[r][c]
-// xhat recomputed from x, mean, rstd. Grid: (ceil(C/256), 1, r_chunks);
-// atomicAdd into pre-zeroed grad slots.
+// xhat recomputed from x, mean, rstd. Grid: (ceil(C/64), ceil(R/LNB_ROWS));
+// 4 waves stripe the row chunk, LDS-reduced, one atomicAdd per column/WG.
+#define LNB_ROWS 256
 __global__ __launch_bounds__(256) void ln_bwd_params_kernel(
     const float* __restrict__ x, const float* __restrict__ dxln,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     float* __restrict__ dgamma, float* __restrict__ dbeta, long R, int C) {
-  const int c = blockIdx.x * 256 + threadIdx.x;
-  if (c >= C) return;
-  const long chunk = (R + gridDim.z - 1) / gridDim.z;
-  const long rbeg = (long)blockIdx.z * chunk;
-  const long rend = min(rbeg + chunk, R);
+  __shared__ float pg[4][64];
+  __shared__ float pb[4][64];
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int c = blockIdx.x * 64 + lane;
+  const long rbeg = (long)blockIdx.y * LNB_ROWS;
+  const long rend = min(rbeg + LNB_ROWS, R);
   float dg = 0.0f, db = 0.0f;
-  for (long r = rbeg; r < rend; ++r) {
-    const float g = dxln[r * C + c];
-    const float xh = (x[r * C + c] - mean[r]) * rstd[r];
-    dg = fmaf(g, xh, dg);
-    db += g;
+  if (c < C) {
+    for (long r = rbeg + w; r < rend; r += 4) {
+      const float g = dxln[r * C + c];
+      const float xh = (x[r * C + c] - mean[r]) * rstd[r];
+      dg = fmaf(g, xh, dg);
+      db += g;
+    }
   }
-  atomicAdd(&dgamma[c], dg);
-  atomicAdd(&dbeta[c], db);
+  pg[w][lane] = dg;
+  pb[w][lane] = db;
+  __syncthreads();
+  if (w == 0 && c < C) {
+    atomicAdd(&dgamma[c], pg[0][lane] + pg[1][lane] + pg[2][lane] + pg[3][lane]);
+    atomicAdd(&dbeta[c], pb[0][lane] + pb[1][lane] + pb[2][lane] + pb[3][lane]);
+  }
 }
 
 extern "C" {
@@ -81,8 +91,8 @@ hipError_t fv_ln_bwd_params(const float* x, const float* dxln,
                             const float* mean, const float* rstd,
                             float* dgamma, float* dbeta, long R, int C,
                             int r_chunks, hipStream_t stream) {
-  if (r_chunks < 1) r_chunks = 1;
-  dim3 grid((C + 255) / 256, 1, r_chunks);
+  (void)r_chunks;
+  dim3 grid((C + 63) / 64, (unsigned)((R + LNB_ROWS - 1) / LNB_ROWS));
   hipLaunchKernelGGL(ln_bwd_params_kernel, grid, dim3(256), 0, stream,
                      x, dxln, mean, rstd, dgamma, dbeta, R, C);
   HIP_CHECK_LAST();

@@ -211,18 +211,28 @@ __global__ __launch_bounds__(256) void gemm_tn_kernel(
   }
 }
 
-// out(C) (+)= sum_r A(R,C): grid (ceil(C/256), z R-chunks); atomic adds
-// (out is a pre-zeroed grad slot).
+// out(C) (+)= sum_r A(R,C): grid (ceil(C/64), ceil(R/CS_ROWS)); 256
+// threads = 4 waves striping the row chunk, LDS-reduced, one atomicAdd
+// per column per WG (out is a pre-zeroed grad slot).
+#define CS_ROWS 256
 __global__ __launch_bounds__(256) void colsum_kernel(
     const float* __restrict__ A, float* __restrict__ out, int R, int C) {
-  const int c = blockIdx.x * 256 + threadIdx.x;
-  if (c >= C) return;
-  const int chunk = (R + gridDim.z - 1) / gridDim.z;
-  const int rbeg = blockIdx.z * chunk;
-  const int rend = min(rbeg + chunk, R);
+  __shared__ float part[4][64];
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int c = blockIdx.x * 64 + lane;
+  const int rbeg = blockIdx.y * CS_ROWS;
+  const int rend = min(rbeg + CS_ROWS, R);
   float s = 0.0f;
-  for (int r = rbeg; r < rend; ++r) s += A[(long)r * C + c];
-  atomicAdd(&out[c], s);
+  if (c < C) {
+    for (int r = rbeg + w; r < rend; r += 4) s += A[(long)r * C + c];
+  }
+  part[w][lane] = s;
+  __syncthreads();
+  if (w == 0 && c < C) {
+    const float v = part[0][lane] + part[1][lane] + part[2][lane] + part[3][lane];
+    atomicAdd(&out[c], v);
+  }
 }
 
 // dZ = dY * lrelu'(Y) elementwise, where Y is the post-activation output
@@ -263,6 +273,12 @@ hipError_t fv_gemm_tn(const float* A, const float* B, float* out,
                       hipStream_t stream) {
   if (r_chunks < 1) r_chunks = 1;
   if (r_chunks > 1) use_atomic = 1;
+  if (use_atomic) {
+    // fill the chip: enough R-chunks that grid >> CUs (atomic accumulate)
+    int auto_chunks = (R + 255) / 256;
+    if (auto_chunks > 64) auto_chunks = 64;
+    if (auto_chunks > r_chunks) r_chunks = auto_chunks;
+  }
   dim3 grid((M + TM - 1) / TM, (N + TN_ - 1) / TN_, r_chunks);
   hipLaunchKernelGGL(gemm_tn_kernel, grid, dim3(256), 0, stream,
                      A, B, out, R, M, N, use_atomic);
@@ -272,8 +288,8 @@ hipError_t fv_gemm_tn(const float* A, const float* B, float* out,
 
 hipError_t fv_colsum(const float* A, float* out, int R, int C, int r_chunks,
                      hipStream_t stream) {
-  if (r_chunks < 1) r_chunks = 1;
-  dim3 grid((C + 255) / 256, 1, r_chunks);
+  (void)r_chunks;
+  dim3 grid((C + 63) / 64, (R + CS_ROWS - 1) / CS_ROWS);
   hipLaunchKernelGGL(colsum_kernel, grid, dim3(256), 0, stream, A, out, R, C);
   HIP_CHECK_LAST();
   return hipSuccess;
