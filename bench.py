@@ -89,8 +89,23 @@ def main():
     if engine_name == "fused":
         from factorvae_amd.engine.fused import FusedTrainer
         trainer = FusedTrainer(model, lr=args.lr, t_max=total_steps, device=device)
-        step_fn = trainer.step
+        # multi-step graph: capture G = gcd-ish steps per replay so the
+        # timed region is exactly `steps` cross-sections
+        G = 1
+        for d in range(min(args.n_days, args.steps), 0, -1):
+            if args.steps % d == 0 and args.warmup % d == 0:
+                G = d
+                break
+        runner, _ = trainer.make_bench_runner(days[:G])
+
+        class _Multi:
+            pass
+
+        multi = _Multi()
+        multi.runner, multi.G = runner, G
+        step_fn = None
     else:
+        multi = None
         bucket = FlatGradBucket(model.parameters()) if is_distributed() else None
         opt = torch.optim.Adam(model.parameters(), lr=args.lr)
         sched = torch.optim.lr_scheduler.CosineAnnealingLR(opt, T_max=total_steps)
@@ -118,14 +133,22 @@ def main():
             torch.distributed.barrier()
 
     # warmup
-    for i in range(args.warmup):
-        step_fn(*days[i % len(days)])
+    if multi is not None:
+        for _ in range(max(1, args.warmup // multi.G)):
+            multi.runner()
+    else:
+        for i in range(args.warmup):
+            step_fn(*days[i % len(days)])
     barrier()
     sync()
 
     t0 = time.perf_counter()
-    for i in range(args.steps):
-        step_fn(*days[(args.warmup + i) % len(days)])
+    if multi is not None:
+        for _ in range(args.steps // multi.G):
+            multi.runner()
+    else:
+        for i in range(args.steps):
+            step_fn(*days[(args.warmup + i) % len(days)])
     barrier()
     sync()
     t1 = time.perf_counter()

@@ -259,6 +259,8 @@ class FusedTrainer:
         w["dgi"] = f(R, 3 * H)
         w["dgh"] = f(R, 3 * H)
         w["dxp"] = f(R, C)
+        max_mn = max(3 * H * max(H, C), C * C, M * H, K * max(H, M))
+        w["tn_part"] = f(32 * max_mn)
         w["dzx"] = f(R, C)
         w["dxln"] = f(R, C)
         self.ws = w
@@ -266,20 +268,23 @@ class FusedTrainer:
         self._ws_t = T
 
     # ------------------------------------------------------------ the step
-    def _launch_forward(self, N: int, T: int, with_loss: bool = True):
+    def _launch_forward(self, N: int, T: int, with_loss: bool = True,
+                        x=None, y=None):
         ext, w, p = self.ext, self.ws, self.p
         C, H, M, K = self.C, self.H, self.M, self.K
         R = N * T
+        x2d = (w["x"] if x is None else x).view(R, C)
+        yv = w["y"] if y is None else y
         alpha = 1.0 / math.sqrt(float(H) + 1e-6)
 
-        ext.ln_fwd(w["x"].view(R, C), p("ln_g"), p("ln_b"), w["xln"],
+        ext.ln_fwd(x2d, p("ln_g"), p("ln_b"), w["xln"],
                    w["mean"], w["rstd"], 1e-5)
         ext.gemm_nt(w["xln"], p("W1x"), p("b1x"), w["xp"], 1.0, False, True)
         ext.gemm_nt(w["xp"], p("Wih"), p("bih"), w["gi"], 1.0, False, False)
         ext.gru_fwd(w["gi"], p("Whh"), p("bhh"), w["h"], w["h_seq"],
                     w["h_prev"], w["gates4"], N, T, H)
         ext.gemm_nt(w["h"], p("Wenc"), p("benc"), w["scores_enc"], 1.0, False, False)
-        ext.enc_softmax_fwd(w["scores_enc"], w["y"], w["a_enc"], w["yp"])
+        ext.enc_softmax_fwd(w["scores_enc"], yv, w["a_enc"], w["yp"])
         ext.enc_heads_fwd(w["yp"], p("Wmu_e"), p("bmu_e"), p("Wsig_e"),
                           p("bsig_e"), w["fmu"], w["fsig_pre"], w["fsig"],
                           w["fsig_c"])
@@ -289,7 +294,7 @@ class FusedTrainer:
         keep_inv = 1.0 / (1.0 - self.DROPOUT_P)
         ext.attn_softmax_fwd(w["s_att"], mask, w["a_att"], w["sd"],
                              w["guard"], keep_inv)
-        ext.gemm_tn(w["a_att"], w["h"], w["u"], 1, False)
+        ext.gemm_tn(w["a_att"], w["h"], w["u"], None, 1, False)
         ext.attn_ctx_fwd(w["u"], self.p_Wv, self.p_bv, w["guard"], w["ctx"])
         ext.pred_mlp_fwd(w["ctx"], p("Wl"), p("bl"), p("wmu_p"), p("bmu_p"),
                          p("wsig_p"), p("bsig_p"), w["hm2"], w["pmu"],
@@ -299,18 +304,20 @@ class FusedTrainer:
                     w["fsig_c"], w["eps"], w["recon"], w["a1"], w["beta"],
                     w["asig_pre"], w["sigma"])
         if with_loss:
-            ext.loss_fwd(w["recon"], w["y"], w["fmu"], w["fsig_c"], w["pmu"],
+            ext.loss_fwd(w["recon"], yv, w["fmu"], w["fsig_c"], w["pmu"],
                          w["psig_c"], w["loss"], w["mse"], w["kl"])
 
-    def _launch_backward(self, N: int, T: int):
+    def _launch_backward(self, N: int, T: int, x=None, y=None):
         ext, w, p, g = self.ext, self.ws, self.p, self.g
         C, H, M, K = self.C, self.H, self.M, self.K
         R = N * T
+        x2d = (w["x"] if x is None else x).view(R, C)
+        yv = w["y"] if y is None else y
         alpha = 1.0 / math.sqrt(float(H) + 1e-6)
         keep_inv = 1.0 / (1.0 - self.DROPOUT_P)
         chunks = max(1, min(32, R // 1024))
 
-        ext.loss_bwd(w["recon"], w["y"], w["fmu"], w["fsig_c"], w["pmu"],
+        ext.loss_bwd(w["recon"], yv, w["fmu"], w["fsig_c"], w["pmu"],
                      w["psig_c"], w["drecon"], w["dfmu"], w["dfsig_c"],
                      w["dpmu"], w["dpsig_c"], 1.0)
         ext.dec_bwd(w["drecon"], w["h"], w["a1"], w["beta"], w["asig_pre"],
@@ -318,16 +325,16 @@ class FusedTrainer:
                     p("wmu_d"), p("wsig_d"), p("Wb"), w["dh"], w["dz1"],
                     w["dbeta"], w["dfmu"], w["dfsig_c"], g("wmu_d"),
                     g("bmu_d"), g("wsig_d"), g("bsig_d"))
-        ext.gemm_tn(w["dz1"], w["h"], g("W1d"), 1, True)
+        ext.gemm_tn(w["dz1"], w["h"], g("W1d"), None, 1, True)
         ext.colsum(w["dz1"], g("b1d"), 1)
-        ext.gemm_tn(w["dbeta"], w["h"], g("Wb"), 1, True)
+        ext.gemm_tn(w["dbeta"], w["h"], g("Wb"), None, 1, True)
         ext.colsum(w["dbeta"], g("bb"), 1)
 
         # predictor MLP + attention backward
         ext.pred_mlp_bwd(w["dpmu"], w["dpsig_c"], w["psig"], w["psig_pre"],
                          w["hm2"], p("wmu_p"), p("wsig_p"), w["dz2"],
                          g("wmu_p"), g("bmu_p"), g("wsig_p"), g("bsig_p"))
-        ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), 1, True)
+        ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True)
         ext.colsum(w["dz2"], g("bl"), 1)
         ext.gemm_nn(w["dz2"], p("Wl"), None, w["dctx"], 1.0, False, False)
         gWv = self._gstack("Wv.0", (K, H, H))
@@ -340,7 +347,7 @@ class FusedTrainer:
                              w["ds"], w["dc"], keep_inv, alpha)
         ext.gemm_nn(w["ds"], w["qk"], None, w["dh"], 1.0, True, False)
         ext.gemm_nn(w["a_att"], w["du"], None, w["dh"], 1.0, True, False)
-        ext.gemm_tn(w["ds"], w["h"], w["dqk"], 1, False)
+        ext.gemm_tn(w["ds"], w["h"], w["dqk"], None, 1, False)
         gq = self._gstack("q_att.0", (K, H))
         gWk = self._gstack("Wk.0", (K, H, H))
         gbk = self._gstack("bk.0", (K, H))
@@ -351,26 +358,26 @@ class FusedTrainer:
         ext.enc_heads_bwd(w["dfmu"], w["dfsig_c"], w["fsig"], w["fsig_pre"],
                           w["yp"], p("Wmu_e"), p("Wsig_e"), w["dyp"],
                           g("Wmu_e"), g("bmu_e"), g("Wsig_e"), g("bsig_e"))
-        ext.enc_softmax_bwd(w["dyp"], w["a_enc"], w["y"], w["dscores"])
+        ext.enc_softmax_bwd(w["dyp"], w["a_enc"], yv, w["dscores"])
         ext.gemm_nn(w["dscores"], p("Wenc"), None, w["dh"], 1.0, True, False)
-        ext.gemm_tn(w["dscores"], w["h"], g("Wenc"), 1, True)
+        ext.gemm_tn(w["dscores"], w["h"], g("Wenc"), None, 1, True)
         ext.colsum(w["dscores"], g("benc"), 1)
 
         # extractor backward
         ext.gru_bwd(w["dh"], w["h_prev"], w["gates4"], p("Whh"), w["dgi"],
                     w["dgh"], N, T, H)
         ext.gemm_tn(w["dgh"].view(R, 3 * H), w["h_prev"].view(R, H), g("Whh"),
-                    chunks, True)
+                    w["tn_part"], chunks, True)
         ext.colsum(w["dgh"].view(R, 3 * H), g("bhh"), chunks)
         ext.gemm_nn(w["dgi"].view(R, 3 * H), p("Wih"), None, w["dxp"], 1.0,
                     False, False)
-        ext.gemm_tn(w["dgi"].view(R, 3 * H), w["xp"], g("Wih"), chunks, True)
+        ext.gemm_tn(w["dgi"].view(R, 3 * H), w["xp"], g("Wih"), w["tn_part"], chunks, True)
         ext.colsum(w["dgi"].view(R, 3 * H), g("bih"), chunks)
         ext.lrelu_bwd(w["dxp"], w["xp"], w["dzx"])
         ext.gemm_nn(w["dzx"], p("W1x"), None, w["dxln"], 1.0, False, False)
-        ext.gemm_tn(w["dzx"], w["xln"], g("W1x"), chunks, True)
+        ext.gemm_tn(w["dzx"], w["xln"], g("W1x"), w["tn_part"], chunks, True)
         ext.colsum(w["dzx"], g("b1x"), chunks)
-        ext.ln_bwd_params(w["x"].view(R, C), w["dxln"], w["mean"], w["rstd"],
+        ext.ln_bwd_params(x2d, w["dxln"], w["mean"], w["rstd"],
                           g("ln_g"), g("ln_b"), chunks)
 
     def _launch_optimizer(self):
@@ -389,6 +396,58 @@ class FusedTrainer:
             self._graphs.clear()
             self._alloc_ws(N, T)
 
+    # -------------------------------------------------- graph capability probe
+    _caps = None
+
+    def _probe_caps(self):
+        """Can torch RNG ops / RCCL collectives be captured in a hipGraph?
+        Probed once; capture plans adapt (fallbacks keep correctness)."""
+        if self._caps is not None:
+            return self._caps
+        rng_ok = True
+        try:
+            t_ = torch.zeros(8, device=self.device)
+            torch.cuda.synchronize()
+            gp = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(gp):
+                t_.normal_()
+            gp.replay()
+            torch.cuda.synchronize()
+        except Exception:
+            rng_ok = False
+            torch.cuda.synchronize()
+        comm_ok = False
+        if is_distributed():
+            try:
+                t_ = torch.ones(8, device=self.device)
+                torch.distributed.all_reduce(t_)  # eager warmup of the PG
+                torch.cuda.synchronize()
+                gp = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(gp):
+                    torch.distributed.all_reduce(t_)
+                gp.replay()
+                torch.cuda.synchronize()
+                comm_ok = True
+            except Exception:
+                comm_ok = False
+                torch.cuda.synchronize()
+        self._caps = (rng_ok, comm_ok)
+        return self._caps
+
+    def _graph_step_body(self, x, y, N, T, rng_in_graph: bool,
+                         comm_in_graph: bool, with_opt: bool = True):
+        """The full training step as a capturable kernel sequence."""
+        self.grads.zero_()
+        if rng_in_graph:
+            self._fill_rng(N)
+        self._launch_forward(N, T, x=x, y=y)
+        self._launch_backward(N, T, x=x, y=y)
+        if comm_in_graph:
+            self.grads.div_(get_world_size())
+            torch.distributed.all_reduce(self.grads)
+        if with_opt:
+            self._launch_optimizer()
+
     def step(self, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
         """One training step on a day cross-section; returns device loss."""
         N, T, C = x.shape
@@ -397,19 +456,9 @@ class FusedTrainer:
         w = self.ws
         w["x"].copy_(x)
         w["y"].copy_(y.view(N, 1))
-        self._fill_rng(N)
 
-        if self.use_graph:
-            key = ("train", N, T)
-            if key not in self._graphs:
-                self._capture(key, N, T)
-            gr_fb, gr_opt = self._graphs[key]
-            gr_fb.replay()
-            if is_distributed():
-                self.grads.div_(get_world_size())
-                torch.distributed.all_reduce(self.grads)
-            gr_opt.replay()
-        else:
+        if not self.use_graph:
+            self._fill_rng(N)
             self.grads.zero_()
             self._launch_forward(N, T)
             self._launch_backward(N, T)
@@ -417,7 +466,104 @@ class FusedTrainer:
                 self.grads.div_(get_world_size())
                 torch.distributed.all_reduce(self.grads)
             self._launch_optimizer()
+            return w["loss"]
+
+        rng_ok, comm_ok = self._probe_caps()
+        key = ("train", N, T)
+        if key not in self._graphs:
+            self._capture(key, N, T, rng_ok, comm_ok)
+        plan = self._graphs[key]
+        if not rng_ok:
+            self._fill_rng(N)
+        if plan["split"]:
+            plan["g_fb"].replay()
+            self.grads.div_(get_world_size())
+            torch.distributed.all_reduce(self.grads)
+            plan["g_opt"].replay()
+        else:
+            plan["g"].replay()
         return w["loss"]
+
+    def _capture(self, key, N: int, T: int, rng_ok: bool, comm_ok: bool):
+        # warmup fwd+bwd (settles lazy state; params/step counter untouched)
+        torch.cuda.synchronize()
+        if not rng_ok:
+            self._fill_rng(N)
+        self.grads.zero_()
+        self._launch_forward(N, T)
+        self._launch_backward(N, T)
+        torch.cuda.synchronize()
+
+        split = is_distributed() and not comm_ok
+        if split:
+            g1 = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g1):
+                self._graph_step_body(None, None, N, T, rng_ok, False,
+                                      with_opt=False)
+            g2 = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g2):
+                self._launch_optimizer()
+            self._graphs[key] = {"split": True, "g_fb": g1, "g_opt": g2}
+        else:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._graph_step_body(None, None, N, T, rng_ok,
+                                      is_distributed() and comm_ok)
+            self._graphs[key] = {"split": False, "g": g}
+
+    # -------------------------------------------------------- bench fast path
+    def make_bench_runner(self, days):
+        """Capture len(days) full training steps into ONE graph that reads
+        the resident day tensors directly (no copies, no per-step host
+        work). Returns (replay_fn, steps_per_replay). Falls back to the
+        per-step path when collectives cannot be captured."""
+        G = len(days)
+        N, T, C = days[0][0].shape
+        self._ensure_ws(N, T)
+        rng_ok, comm_ok = self._probe_caps()
+        days = [(x, y.view(-1, 1)) for x, y in days]
+
+        if is_distributed() and not comm_ok:
+            def run_fallback():
+                for x, y in days:
+                    self.step(x, y)
+            return run_fallback, G
+
+        # warmup
+        torch.cuda.synchronize()
+        if not rng_ok:
+            self._fill_rng(N)
+        self.grads.zero_()
+        self._launch_forward(N, T, x=days[0][0], y=days[0][1])
+        self._launch_backward(N, T, x=days[0][0], y=days[0][1])
+        torch.cuda.synchronize()
+
+        # per-day RNG buffers when RNG can't live in the graph
+        rng_bufs = None
+        if not rng_ok:
+            rng_bufs = [(torch.empty(N, device=self.device),
+                         torch.empty(N, self.K, device=self.device))
+                        for _ in range(G)]
+
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            for i, (x, y) in enumerate(days):
+                if rng_bufs is not None:
+                    self.ws["eps"].copy_(rng_bufs[i][0])
+                    if self.training:
+                        self.ws["mask"].copy_(rng_bufs[i][1])
+                self._graph_step_body(x, y, N, T, rng_ok,
+                                      is_distributed() and comm_ok)
+
+        def run():
+            if rng_bufs is not None:
+                for e, m in rng_bufs:
+                    e.normal_()
+                    if self.training:
+                        m.bernoulli_(1.0 - self.DROPOUT_P)
+            g.replay()
+
+        return run, G
 
     def forward_only(self, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
         """Fused validation forward (no grad); returns device loss."""
@@ -439,41 +585,19 @@ class FusedTrainer:
         w = self.ws
         w["x"].copy_(x)
         w["y"].zero_()
-        self._fill_rng(N)
         was_training = self.training
         self.training = False  # prediction: dropout off
         try:
+            self._fill_rng(N)
             self._launch_forward(N, T, with_loss=False)
         finally:
             self.training = was_training
         # decoder with PRIOR mu/sigma (module.py:273-278): psig clamp applies
-        H, K = self.H, self.K
         ext.dec_fwd(w["h"], p("W1d"), p("b1d"), p("wmu_d"), p("bmu_d"),
                     p("wsig_d"), p("bsig_d"), p("Wb"), p("bb"), w["pmu"],
                     w["psig_c"], w["eps"], w["recon"], w["a1"], w["beta"],
                     w["asig_pre"], w["sigma"])
         return w["recon"].view(N, 1).clone()
-
-    def _capture(self, key, N: int, T: int):
-        # Warmup fwd+bwd only (settles any lazy allocator state without
-        # mutating params/step counter — grads are re-zeroed inside the
-        # captured graph anyway); the adam kernels allocate nothing, so
-        # the optimizer graph captures cold.
-        torch.cuda.synchronize()
-        self.grads.zero_()
-        self._launch_forward(N, T)
-        self._launch_backward(N, T)
-        torch.cuda.synchronize()
-
-        g1 = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g1):
-            self.grads.zero_()
-            self._launch_forward(N, T)
-            self._launch_backward(N, T)
-        g2 = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g2):
-            self._launch_optimizer()
-        self._graphs[key] = (g1, g2)
 
     # ------------------------------------------------------------- epochs
     def train_epoch(self, days, shuffle_order=None) -> float:

@@ -12,8 +12,8 @@ hipError_t fv_gemm_nt(const float*, const float*, const float*, float*, int,
                       int, int, float, int, int, hipStream_t);
 hipError_t fv_gemm_nn(const float*, const float*, const float*, float*, int,
                       int, int, float, int, int, hipStream_t);
-hipError_t fv_gemm_tn(const float*, const float*, float*, int, int, int, int,
-                      int, hipStream_t);
+hipError_t fv_gemm_tn(const float*, const float*, float*, float*, int, int,
+                      int, int, int, hipStream_t);
 hipError_t fv_colsum(const float*, float*, int, int, int, hipStream_t);
 hipError_t fv_lrelu_bwd(const float*, const float*, float*, long, hipStream_t);
 hipError_t fv_ln_fwd(const float*, const float*, const float*, float*, float*,
@@ -128,13 +128,21 @@ void gemm_nn(torch::Tensor A, torch::Tensor B,
                  accumulate, act_lrelu, cur_stream()));
 }
 
-void gemm_tn(torch::Tensor A, torch::Tensor B, torch::Tensor out, long r_chunks,
-             bool use_atomic) {
+void gemm_tn(torch::Tensor A, torch::Tensor B, torch::Tensor out,
+             c10::optional<torch::Tensor> part, long r_chunks,
+             bool accumulate) {
   CK(A); CK(B); CK(out);
   const int R = A.size(0), M = A.size(1), N = B.size(1);
   TORCH_CHECK(B.size(0) == R && out.size(0) == M && out.size(1) == N);
-  RUN(fv_gemm_tn(fp(A), fp(B), fpm(out), R, M, N, (int)r_chunks, use_atomic,
-                 cur_stream()));
+  float* pp = nullptr;
+  if (part.has_value()) {
+    CK(*part);
+    TORCH_CHECK(part->numel() >= (long)32 * M * N,
+                "tn partial workspace too small");
+    pp = fpm(*part);
+  }
+  RUN(fv_gemm_tn(fp(A), fp(B), fpm(out), pp, R, M, N, (int)r_chunks,
+                 accumulate, cur_stream()));
 }
 
 void colsum(torch::Tensor A, torch::Tensor out, long r_chunks) {

@@ -7,8 +7,8 @@
 //
 //   gemm_nt:  out(R,Co)  = act(alpha * (A(R,Ci) @ W(Co,Ci)^T + bias))   [+=]
 //   gemm_nn:  out(R,Co)  = act(alpha * (A(R,Ci) @ B(Ci,Co) + bias))    [+=]
-//   gemm_tn:  out(M,N)  (+)= A(R,M)^T @ B(R,N)    (weight grads; atomic
-//             accumulation when the R-chunk grid has >1 z-slice)
+//   gemm_tn:  out(M,N)  (+)= A(R,M)^T @ B(R,N)    (weight grads; R-chunked
+//             slices write partials, reduced in fixed order: deterministic)
 //   colsum:   out(C)    (+)= sum_r A(R,C)         (bias grads)
 
 #include "common.h"
@@ -151,15 +151,16 @@ __global__ __launch_bounds__(256) void gemm_nn_kernel(
 }
 
 // out(M,N) (+)= A(R,M)^T @ B(R,N); R-chunked over gridDim.z.
-// When gridDim.z == 1 and !accumulate, writes directly; otherwise
-// atomically adds (out must be pre-zeroed unless it is a grad arena that
-// accumulates by design).
+// gridDim.z == 1: plain write to out. gridDim.z > 1: each z-slice writes
+// its partial tile to part[z][M][N] (plain stores); tn_reduce_kernel then
+// sums slices in FIXED order into out — deterministic, no atomics.
 #define TM 32
 #define TN_ 32
 #define TKR 32
 __global__ __launch_bounds__(256) void gemm_tn_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
-    float* __restrict__ out, int R, int M, int N, int use_atomic) {
+    float* __restrict__ out, float* __restrict__ part,
+    int R, int M, int N, int accumulate) {
   __shared__ float As[TKR][TM + 1];
   __shared__ float Bs[TKR][TN_ + 1];
 
@@ -197,18 +198,37 @@ __global__ __launch_bounds__(256) void gemm_tn_kernel(
   }
 
   const int gm = m0 + i;
-  if (gm < M) {
+  if (gm >= M) return;
+  if (gridDim.z == 1) {
 #pragma unroll
     for (int jj = 0; jj < 4; ++jj) {
       const int gn = n0 + j4 * 4 + jj;
       if (gn < N) {
-        if (use_atomic)
-          atomicAdd(&out[(long)gm * N + gn], acc[jj]);
+        if (accumulate)
+          out[(long)gm * N + gn] += acc[jj];
         else
           out[(long)gm * N + gn] = acc[jj];
       }
     }
+  } else {
+    float* po = part + (long)blockIdx.z * M * N;
+#pragma unroll
+    for (int jj = 0; jj < 4; ++jj) {
+      const int gn = n0 + j4 * 4 + jj;
+      if (gn < N) po[(long)gm * N + gn] = acc[jj];
+    }
   }
+}
+
+// out[e] += sum_z part[z][e] in fixed z order (deterministic reduce).
+__global__ __launch_bounds__(256) void tn_reduce_kernel(
+    const float* __restrict__ part, float* __restrict__ out, long elems,
+    int z) {
+  const long e = (long)blockIdx.x * 256 + threadIdx.x;
+  if (e >= elems) return;
+  float s = 0.0f;
+  for (int c = 0; c < z; ++c) s += part[(long)c * elems + e];
+  out[e] += s;
 }
 
 // out(C) (+)= sum_r A(R,C): grid (ceil(C/64), ceil(R/CS_ROWS)); 256
@@ -269,20 +289,28 @@ hipError_t fv_gemm_nn(const float* A, const float* B, const float* bias,
 }
 
 hipError_t fv_gemm_tn(const float* A, const float* B, float* out,
-                      int R, int M, int N, int r_chunks, int use_atomic,
-                      hipStream_t stream) {
+                      float* part, int R, int M, int N, int r_chunks,
+                      int accumulate, hipStream_t stream) {
+  // r_chunks > 1 is a request to chunk the R-reduction; the launcher
+  // picks the actual split (>=512 rows per chunk, <=32 slices).
   if (r_chunks < 1) r_chunks = 1;
-  if (r_chunks > 1) use_atomic = 1;
-  if (use_atomic) {
-    // fill the chip: enough R-chunks that grid >> CUs (atomic accumulate)
-    int auto_chunks = (R + 255) / 256;
-    if (auto_chunks > 64) auto_chunks = 64;
-    if (auto_chunks > r_chunks) r_chunks = auto_chunks;
+  if (!part) r_chunks = 1;
+  if (r_chunks > 1) {
+    r_chunks = (R + 511) / 512;
+    if (r_chunks > 32) r_chunks = 32;
+    if (r_chunks < 1) r_chunks = 1;
   }
   dim3 grid((M + TM - 1) / TM, (N + TN_ - 1) / TN_, r_chunks);
   hipLaunchKernelGGL(gemm_tn_kernel, grid, dim3(256), 0, stream,
-                     A, B, out, R, M, N, use_atomic);
+                     A, B, out, part, R, M, N, accumulate);
   HIP_CHECK_LAST();
+  if (r_chunks > 1) {
+    const long elems = (long)M * N;
+    dim3 rgrid((unsigned)((elems + 255) / 256));
+    hipLaunchKernelGGL(tn_reduce_kernel, rgrid, dim3(256), 0, stream,
+                       part, out, elems, r_chunks);
+    HIP_CHECK_LAST();
+  }
   return hipSuccess;
 }
 

@@ -207,8 +207,23 @@ __global__ __launch_bounds__(512) void gru_fwd_fast_kernel(
   const float bz = bhh[lane < H ? H + lane : 0];
   const float bn = bhh[lane < H ? 2 * H + lane : 0];
 
+  // prefetch t=0 gate-input rows so global latency hides under the dot
+  float gir = 0.f, giz = 0.f, gin = 0.f;
+  if (live) {
+    const long b0 = (long)s * T * 3 * H;
+    gir = gi[b0 + lane];
+    giz = gi[b0 + H + lane];
+    gin = gi[b0 + 2 * H + lane];
+  }
   for (int t = 0; t < T; ++t) {
     float hn = 0.0f;
+    float nir = 0.f, niz = 0.f, nin = 0.f;
+    if (live && t + 1 < T) {
+      const long bnx = ((long)s * T + t + 1) * 3 * H;
+      nir = gi[bnx + lane];
+      niz = gi[bnx + H + lane];
+      nin = gi[bnx + 2 * H + lane];
+    }
     if (live) {
       float ghr = br, ghz = bz, q = bn;
       const float4* hp4 = (const float4*)&hprev[w * 64];
@@ -221,10 +236,9 @@ __global__ __launch_bounds__(512) void gru_fwd_fast_kernel(
         ghz = fmaf(hv.x, wz.x, fmaf(hv.y, wz.y, fmaf(hv.z, wz.z, fmaf(hv.w, wz.w, ghz))));
         q   = fmaf(hv.x, wn.x, fmaf(hv.y, wn.y, fmaf(hv.z, wn.z, fmaf(hv.w, wn.w, q))));
       }
-      const long base = ((long)s * T + t) * 3 * H;
-      const float r = sigmoidf_(gi[base + lane] + ghr);
-      const float z = sigmoidf_(gi[base + H + lane] + ghz);
-      const float n = tanhf(fmaf(r, q, gi[base + 2 * H + lane]));
+      const float r = sigmoidf_(gir + ghr);
+      const float z = sigmoidf_(giz + ghz);
+      const float n = tanhf(fmaf(r, q, gin));
       const float hp_l = hprev[w * 64 + lane];
       hn = fmaf(z, hp_l - n, n);
 
@@ -238,6 +252,7 @@ __global__ __launch_bounds__(512) void gru_fwd_fast_kernel(
       gates4[gb + 3 * H] = q;
       if (t == T - 1) h_final[(long)s * H + lane] = hn;
     }
+    gir = nir; giz = niz; gin = nin;
     // per-wave-private slice: no barrier
     if (s < N && lane < H) hprev[w * 64 + lane] = hn;
   }
@@ -269,15 +284,33 @@ __global__ __launch_bounds__(512) void gru_bwd_fast_kernel(
   if (live) dhS[w * 64 + lane] = dh_final[(long)s * H + lane];
   __syncthreads();
 
+  // software prefetch of the (t-1) gate rows across the dot product
+  float pr = 0.f, pz = 0.f, pn = 0.f, pq = 0.f, php = 0.f;
+  if (live) {
+    const long gb = ((long)s * T + (T - 1)) * 4 * H + lane;
+    pr = gates4[gb];
+    pz = gates4[gb + H];
+    pn = gates4[gb + 2 * H];
+    pq = gates4[gb + 3 * H];
+    php = h_prev_in[((long)s * T + (T - 1)) * H + lane];
+  }
   for (int t = T - 1; t >= 0; --t) {
     float zv = 0.0f;
+    float xr = 0.f, xz = 0.f, xn = 0.f, xq = 0.f, xhp = 0.f;
+    if (live && t > 0) {
+      const long gb = ((long)s * T + t - 1) * 4 * H + lane;
+      xr = gates4[gb];
+      xz = gates4[gb + H];
+      xn = gates4[gb + 2 * H];
+      xq = gates4[gb + 3 * H];
+      xhp = h_prev_in[((long)s * T + t - 1) * H + lane];
+    }
     if (live) {
-      const long gb = ((long)s * T + t) * 4 * H + lane;
-      const float r = gates4[gb];
-      const float z = gates4[gb + H];
-      const float n = gates4[gb + 2 * H];
-      const float q = gates4[gb + 3 * H];
-      const float hp = h_prev_in[((long)s * T + t) * H + lane];
+      const float r = pr;
+      const float z = pz;
+      const float n = pn;
+      const float q = pq;
+      const float hp = php;
       const float dh = dhS[w * 64 + lane];
 
       const float dz = dh * (hp - n);
@@ -301,6 +334,7 @@ __global__ __launch_bounds__(512) void gru_bwd_fast_kernel(
       dghS[w * 3 * H + 2 * H + lane] = dgh_n;
       zv = z;
     }
+    pr = xr; pz = xz; pn = xn; pq = xq; php = xhp;
     float acc = 0.0f;
     if (live) {
       acc = dhS[w * 64 + lane] * zv;
