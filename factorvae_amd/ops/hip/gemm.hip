@@ -13,6 +13,8 @@
 
 #include "common.h"
 
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
 #define BR 64
 #define BC 64
 #define BK 32
@@ -29,14 +31,15 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
   const int r0 = blockIdx.x * BR;
   const int c0 = blockIdx.y * BC;
   const int tid = threadIdx.x;
-  const int tx = tid & 15;        // 16 col-groups of 4
-  const int ty = tid >> 4;        // 16 row-groups of 4
+  const int lane = tid & 63;
+  const int wv = tid >> 6;        // wave -> 16-row strip
+  const int fi = lane & 15;       // fragment row/col index
+  const int fk = lane >> 4;       // fragment k index (0..3)
 
-  float acc[4][4];
-#pragma unroll
-  for (int i = 0; i < 4; ++i)
-#pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = 0.0f;
+  // MFMA f32 16x16x4: wave wv computes rows [wv*16, wv*16+16) x all 64
+  // cols as 4 independent 16x16 accumulators (hides the 40-cyc
+  // dependent-accumulator latency at the 32-cyc issue rate).
+  f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
 
   for (int k0 = 0; k0 < Ci; k0 += BK) {
     // stage A tile (BR x BK) and W tile (BC x BK), zero-padded at edges
@@ -51,30 +54,28 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
       Ws[j][k] = (gc < Co && gk < Ci) ? W[(long)gc * Ci + gk] : 0.0f;
     }
     __syncthreads();
-#pragma unroll 4
-    for (int kk = 0; kk < BK; ++kk) {
-      float a[4], w[4];
 #pragma unroll
-      for (int i = 0; i < 4; ++i) a[i] = As[ty * 4 + i][kk];
+    for (int k4 = 0; k4 < BK; k4 += 4) {
+      // A fragment: A[i=fi][k=fk]; B fragment: B[k][j] = W[j][k]
+      const float a = As[wv * 16 + fi][k4 + fk];
 #pragma unroll
-      for (int j = 0; j < 4; ++j) w[j] = Ws[tx * 4 + j][kk];
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j) acc[i][j] = fmaf(a[i], w[j], acc[i][j]);
+      for (int jt = 0; jt < 4; ++jt) {
+        const float b = Ws[jt * 16 + fi][k4 + fk];
+        acc[jt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[jt], 0, 0, 0);
+      }
     }
     __syncthreads();
   }
 
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    const int gr = r0 + ty * 4 + i;
-    if (gr >= R) continue;
+  for (int jt = 0; jt < 4; ++jt) {
+    const int gc = c0 + jt * 16 + fi;
+    if (gc >= Co) continue;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const int gc = c0 + tx * 4 + j;
-      if (gc >= Co) continue;
-      float v = acc[i][j];
+    for (int rr = 0; rr < 4; ++rr) {
+      const int gr = r0 + wv * 16 + (lane >> 4) * 4 + rr;
+      if (gr >= R) continue;
+      float v = acc[jt][rr];
       if (flags & 4) v += bias[gc];
       v *= alpha;
       if (flags & 2) v = lrelu_(v);
@@ -95,14 +96,12 @@ __global__ __launch_bounds__(256) void gemm_nn_kernel(
   const int r0 = blockIdx.x * BR;
   const int c0 = blockIdx.y * BC;
   const int tid = threadIdx.x;
-  const int tx = tid & 15;
-  const int ty = tid >> 4;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const int fi = lane & 15;
+  const int fk = lane >> 4;
 
-  float acc[4][4];
-#pragma unroll
-  for (int i = 0; i < 4; ++i)
-#pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = 0.0f;
+  f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
 
   for (int k0 = 0; k0 < Ci; k0 += BK) {
     for (int idx = tid; idx < BR * BK; idx += 256) {
@@ -116,30 +115,27 @@ __global__ __launch_bounds__(256) void gemm_nn_kernel(
       Bs[k][j] = (gk < Ci && gc < Co) ? B[(long)gk * Co + gc] : 0.0f;
     }
     __syncthreads();
-#pragma unroll 4
-    for (int kk = 0; kk < BK; ++kk) {
-      float a[4], b[4];
 #pragma unroll
-      for (int i = 0; i < 4; ++i) a[i] = As[ty * 4 + i][kk];
+    for (int k4 = 0; k4 < BK; k4 += 4) {
+      const float a = As[wv * 16 + fi][k4 + fk];
 #pragma unroll
-      for (int j = 0; j < 4; ++j) b[j] = Bs[kk][tx * 4 + j];
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j) acc[i][j] = fmaf(a[i], b[j], acc[i][j]);
+      for (int jt = 0; jt < 4; ++jt) {
+        const float b = Bs[k4 + fk][jt * 16 + fi];
+        acc[jt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[jt], 0, 0, 0);
+      }
     }
     __syncthreads();
   }
 
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    const int gr = r0 + ty * 4 + i;
-    if (gr >= R) continue;
+  for (int jt = 0; jt < 4; ++jt) {
+    const int gc = c0 + jt * 16 + fi;
+    if (gc >= Co) continue;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const int gc = c0 + tx * 4 + j;
-      if (gc >= Co) continue;
-      float v = acc[i][j];
+    for (int rr = 0; rr < 4; ++rr) {
+      const int gr = r0 + wv * 16 + (lane >> 4) * 4 + rr;
+      if (gr >= R) continue;
+      float v = acc[jt][rr];
       if (flags & 4) v += bias[gc];
       v *= alpha;
       if (flags & 2) v = lrelu_(v);
@@ -171,9 +167,14 @@ __global__ __launch_bounds__(256) void gemm_tn_kernel(
   const int rend = min(rbeg + chunk, R);
 
   const int tid = threadIdx.x;
-  const int i = tid & 31;         // m within tile
-  const int j4 = tid >> 5;        // 8 groups of 4 n-columns
-  float acc[4] = {0.f, 0.f, 0.f, 0.f};
+  const int lane = tid & 63;
+  const int wv = tid >> 6;        // wave -> one 16x16 out tile (2x2 grid)
+  const int fi = lane & 15;
+  const int fk = lane >> 4;
+  const int mt = (wv & 1) * 16;   // m-tile offset
+  const int nt = (wv >> 1) * 16;  // n-tile offset
+
+  f32x4 acc = {0, 0, 0, 0};
 
   for (int r0 = rbeg; r0 < rend; r0 += TKR) {
     for (int idx = tid; idx < TKR * TM; idx += 256) {
@@ -187,36 +188,30 @@ __global__ __launch_bounds__(256) void gemm_tn_kernel(
       Bs[rr][nn] = (gr < rend && gn < N) ? B[(long)gr * N + gn] : 0.0f;
     }
     __syncthreads();
-#pragma unroll 8
-    for (int rr = 0; rr < TKR; ++rr) {
-      const float a = As[rr][i];
+    // out[m][n] = sum_r A[r][m]*B[r][n]: MFMA with k = r
 #pragma unroll
-      for (int jj = 0; jj < 4; ++jj)
-        acc[jj] = fmaf(a, Bs[rr][j4 * 4 + jj], acc[jj]);
+    for (int r4 = 0; r4 < TKR; r4 += 4) {
+      const float a = As[r4 + fk][mt + fi];
+      const float b = Bs[r4 + fk][nt + fi];
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
     }
     __syncthreads();
   }
 
-  const int gm = m0 + i;
-  if (gm >= M) return;
-  if (gridDim.z == 1) {
+  const int gn = n0 + nt + fi;
+  if (gn >= N) return;
+  float* dst;
+  const bool direct = (gridDim.z == 1);
+  float* po = direct ? out : part + (long)blockIdx.z * M * N;
 #pragma unroll
-    for (int jj = 0; jj < 4; ++jj) {
-      const int gn = n0 + j4 * 4 + jj;
-      if (gn < N) {
-        if (accumulate)
-          out[(long)gm * N + gn] += acc[jj];
-        else
-          out[(long)gm * N + gn] = acc[jj];
-      }
-    }
-  } else {
-    float* po = part + (long)blockIdx.z * M * N;
-#pragma unroll
-    for (int jj = 0; jj < 4; ++jj) {
-      const int gn = n0 + j4 * 4 + jj;
-      if (gn < N) po[(long)gm * N + gn] = acc[jj];
-    }
+  for (int rr = 0; rr < 4; ++rr) {
+    const int gm = m0 + mt + (lane >> 4) * 4 + rr;
+    if (gm >= M) continue;
+    dst = &po[(long)gm * N + gn];
+    if (direct && accumulate)
+      *dst += acc[rr];
+    else
+      *dst = acc[rr];
   }
 }
 
