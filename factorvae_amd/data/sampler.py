@@ -82,7 +82,8 @@ class TSDataSampler:
             self.flt_rows = None
 
         self.start_idx, self.end_idx = self.data_index.slice_locs(
-            start=pd.Timestamp(start), end=pd.Timestamp(end)
+            start=pd.Timestamp(start) if start is not None else None,
+            end=pd.Timestamp(end) if end is not None else None,
         )
 
     def get_index(self) -> pd.MultiIndex:
