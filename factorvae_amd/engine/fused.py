@@ -90,8 +90,14 @@ class FusedTrainer:
         self.step_t = torch.zeros(1, device=self.device, dtype=torch.int32)
 
         self._ws_n = 0
+        self._ws_cache: Dict = {}
         self._graphs: Dict = {}
         self._g_inputs = None
+        # side stream: weight-gradient GEMMs/colsums run here, overlapped
+        # with the activation-gradient critical path (fork/join via events;
+        # the dependencies are recorded into the captured hipGraph)
+        self.s_side = (torch.cuda.Stream(device=self.device)
+                       if self.device.type == "cuda" else None)
 
     # ---------------------------------------------------------------- params
     _STACKED = ("q_att", "Wk", "bk", "Wv", "bv")
@@ -261,8 +267,11 @@ class FusedTrainer:
         w["dxp"] = f(R, C)
         max_mn = max(3 * H * max(H, C), C * C, M * H, K * max(H, M))
         w["tn_part"] = f(32 * max_mn)
+        w["tn_part2"] = f(32 * max_mn)
+        w["tn_part3"] = f(32 * max_mn)
         w["dzx"] = f(R, C)
         w["dxln"] = f(R, C)
+        self._ws_cache[(N, T)] = w
         self.ws = w
         self._ws_n = N
         self._ws_t = T
@@ -308,6 +317,11 @@ class FusedTrainer:
                          w["psig_c"], w["loss"], w["mse"], w["kl"])
 
     def _launch_backward(self, N: int, T: int, x=None, y=None):
+        """Backward pass. The activation-gradient chain runs on the main
+        stream; every weight-gradient GEMM / colsum forks onto the side
+        stream as soon as its producer is done (fork/join events become
+        hipGraph edges under capture), so wgrad work overlaps the dgrad
+        critical path and joins before the optimizer."""
         ext, w, p, g = self.ext, self.ws, self.p, self.g
         C, H, M, K = self.C, self.H, self.M, self.K
         R = N * T
@@ -317,6 +331,28 @@ class FusedTrainer:
         keep_inv = 1.0 / (1.0 - self.DROPOUT_P)
         chunks = max(1, min(32, R // 1024))
 
+        main = torch.cuda.current_stream(self.device) if self.s_side else None
+        side = self.s_side
+
+        def fork():
+            """Make the side stream wait for everything issued on main."""
+            if side is None:
+                return
+            e = torch.cuda.Event()
+            e.record(main)
+            side.wait_event(e)
+
+        class _on_side:
+            def __enter__(_s):
+                if side is not None:
+                    _s.ctx = torch.cuda.stream(side)
+                    _s.ctx.__enter__()
+                return _s
+
+            def __exit__(_s, *a):
+                if side is not None:
+                    _s.ctx.__exit__(*a)
+
         ext.loss_bwd(w["recon"], yv, w["fmu"], w["fsig_c"], w["pmu"],
                      w["psig_c"], w["drecon"], w["dfmu"], w["dfsig_c"],
                      w["dpmu"], w["dpsig_c"], 1.0)
@@ -325,17 +361,21 @@ class FusedTrainer:
                     p("wmu_d"), p("wsig_d"), p("Wb"), w["dh"], w["dz1"],
                     w["dbeta"], w["dfmu"], w["dfsig_c"], g("wmu_d"),
                     g("bmu_d"), g("wsig_d"), g("bsig_d"))
-        ext.gemm_tn(w["dz1"], w["h"], g("W1d"), None, 1, True)
-        ext.colsum(w["dz1"], g("b1d"), 1)
-        ext.gemm_tn(w["dbeta"], w["h"], g("Wb"), None, 1, True)
-        ext.colsum(w["dbeta"], g("bb"), 1)
+        fork()
+        with _on_side():
+            ext.gemm_tn(w["dz1"], w["h"], g("W1d"), None, 1, True)
+            ext.colsum(w["dz1"], g("b1d"), 1)
+            ext.gemm_tn(w["dbeta"], w["h"], g("Wb"), None, 1, True)
+            ext.colsum(w["dbeta"], g("bb"), 1)
 
         # predictor MLP + attention backward
         ext.pred_mlp_bwd(w["dpmu"], w["dpsig_c"], w["psig"], w["psig_pre"],
                          w["hm2"], p("wmu_p"), p("wsig_p"), w["dz2"],
                          g("wmu_p"), g("bmu_p"), g("wsig_p"), g("bsig_p"))
-        ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True)
-        ext.colsum(w["dz2"], g("bl"), 1)
+        fork()
+        with _on_side():
+            ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True)
+            ext.colsum(w["dz2"], g("bl"), 1)
         ext.gemm_nn(w["dz2"], p("Wl"), None, w["dctx"], 1.0, False, False)
         gWv = self._gstack("Wv.0", (K, H, H))
         gbv = self._gstack("bv.0", (K, H))
@@ -345,43 +385,59 @@ class FusedTrainer:
         mask = w["mask"] if self.training else None
         ext.attn_softmax_bwd(w["da"], w["a_att"], w["sd"], mask, w["guard"],
                              w["ds"], w["dc"], keep_inv, alpha)
+        fork()
+        with _on_side():
+            ext.gemm_tn(w["ds"], w["h"], w["dqk"], None, 1, False)
+            gq = self._gstack("q_att.0", (K, H))
+            gWk = self._gstack("Wk.0", (K, H, H))
+            gbk = self._gstack("bk.0", (K, H))
+            ext.attn_qk_bwd(w["dqk"], w["dc"], self.p_q, self.p_Wk, self.p_bk,
+                            gq, gWk, gbk)
         ext.gemm_nn(w["ds"], w["qk"], None, w["dh"], 1.0, True, False)
         ext.gemm_nn(w["a_att"], w["du"], None, w["dh"], 1.0, True, False)
-        ext.gemm_tn(w["ds"], w["h"], w["dqk"], None, 1, False)
-        gq = self._gstack("q_att.0", (K, H))
-        gWk = self._gstack("Wk.0", (K, H, H))
-        gbk = self._gstack("bk.0", (K, H))
-        ext.attn_qk_bwd(w["dqk"], w["dc"], self.p_q, self.p_Wk, self.p_bk,
-                        gq, gWk, gbk)
 
         # encoder backward
         ext.enc_heads_bwd(w["dfmu"], w["dfsig_c"], w["fsig"], w["fsig_pre"],
                           w["yp"], p("Wmu_e"), p("Wsig_e"), w["dyp"],
                           g("Wmu_e"), g("bmu_e"), g("Wsig_e"), g("bsig_e"))
         ext.enc_softmax_bwd(w["dyp"], w["a_enc"], yv, w["dscores"])
+        fork()
+        with _on_side():
+            ext.gemm_tn(w["dscores"], w["h"], g("Wenc"), None, 1, True)
+            ext.colsum(w["dscores"], g("benc"), 1)
         ext.gemm_nn(w["dscores"], p("Wenc"), None, w["dh"], 1.0, True, False)
-        ext.gemm_tn(w["dscores"], w["h"], g("Wenc"), None, 1, True)
-        ext.colsum(w["dscores"], g("benc"), 1)
 
         # extractor backward
         ext.gru_bwd(w["dh"], w["h_prev"], w["gates4"], p("Whh"), w["dgi"],
                     w["dgh"], N, T, H)
-        ext.gemm_tn(w["dgh"].view(R, 3 * H), w["h_prev"].view(R, H), g("Whh"),
-                    w["tn_part"], chunks, True)
-        ext.colsum(w["dgh"].view(R, 3 * H), g("bhh"), chunks)
+        fork()
+        with _on_side():
+            ext.gemm_tn(w["dgh"].view(R, 3 * H), w["h_prev"].view(R, H),
+                        g("Whh"), w["tn_part"], chunks, True)
+            ext.colsum(w["dgh"].view(R, 3 * H), g("bhh"), chunks)
+            ext.gemm_tn(w["dgi"].view(R, 3 * H), w["xp"], g("Wih"),
+                        w["tn_part2"], chunks, True)
+            ext.colsum(w["dgi"].view(R, 3 * H), g("bih"), chunks)
         ext.gemm_nn(w["dgi"].view(R, 3 * H), p("Wih"), None, w["dxp"], 1.0,
                     False, False)
-        ext.gemm_tn(w["dgi"].view(R, 3 * H), w["xp"], g("Wih"), w["tn_part"], chunks, True)
-        ext.colsum(w["dgi"].view(R, 3 * H), g("bih"), chunks)
         ext.lrelu_bwd(w["dxp"], w["xp"], w["dzx"])
+        fork()
+        with _on_side():
+            ext.gemm_tn(w["dzx"], w["xln"], g("W1x"), w["tn_part3"], chunks,
+                        True)
+            ext.colsum(w["dzx"], g("b1x"), chunks)
         ext.gemm_nn(w["dzx"], p("W1x"), None, w["dxln"], 1.0, False, False)
-        ext.gemm_tn(w["dzx"], w["xln"], g("W1x"), w["tn_part"], chunks, True)
-        ext.colsum(w["dzx"], g("b1x"), chunks)
         ext.ln_bwd_params(x2d, w["dxln"], w["mean"], w["rstd"],
                           g("ln_g"), g("ln_b"), chunks)
+        # join: main waits for all side-stream wgrad work
+        if side is not None:
+            e = torch.cuda.Event()
+            e.record(side)
+            main.wait_event(e)
 
-    def _launch_optimizer(self):
-        self.ext.step_inc(self.step_t)
+    def _launch_optimizer(self, inc: bool = True):
+        if inc:
+            self.ext.step_inc(self.step_t)
         self.ext.adam(self.params.flat, self.grads, self.adam_m, self.adam_v,
                       self.step_t, self.lr, self.eta_min, float(self.t_max),
                       0.9, 0.999, 1e-8)
@@ -392,8 +448,18 @@ class FusedTrainer:
             self.ws["mask"].bernoulli_(1.0 - self.DROPOUT_P)
 
     def _ensure_ws(self, N: int, T: int):
-        if self._ws_n != N or getattr(self, "_ws_t", None) != T:
-            self._graphs.clear()
+        """Workspaces (and captured graphs) are cached per (N, T): real
+        universes have a different stock count N every day, and 288 GB
+        HBM3E makes keeping one ~60 MB workspace per distinct shape far
+        cheaper than re-capturing the step graph each day."""
+        if self._ws_n == N and getattr(self, "_ws_t", None) == T:
+            return
+        w = self._ws_cache.get((N, T))
+        if w is not None:
+            self.ws = w
+            self._ws_n = N
+            self._ws_t = T
+        else:
             self._alloc_ws(N, T)
 
     # -------------------------------------------------- graph capability probe
@@ -438,6 +504,17 @@ class FusedTrainer:
                          comm_in_graph: bool, with_opt: bool = True):
         """The full training step as a capturable kernel sequence."""
         self.grads.zero_()
+        inc_early = False
+        if with_opt and self.s_side is not None:
+            # LR-step counter increment runs on the side stream, overlapped
+            # with forward (it must still follow the previous step's adam,
+            # hence the fork event)
+            e = torch.cuda.Event()
+            e.record(torch.cuda.current_stream(self.device))
+            self.s_side.wait_event(e)
+            with torch.cuda.stream(self.s_side):
+                self.ext.step_inc(self.step_t)
+            inc_early = True
         if rng_in_graph:
             self._fill_rng(N)
         self._launch_forward(N, T, x=x, y=y)
@@ -446,7 +523,7 @@ class FusedTrainer:
             self.grads.div_(get_world_size())
             torch.distributed.all_reduce(self.grads)
         if with_opt:
-            self._launch_optimizer()
+            self._launch_optimizer(inc=not inc_early)
 
     def step(self, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
         """One training step on a day cross-section; returns device loss."""
@@ -608,6 +685,30 @@ class FusedTrainer:
             total += loss[0]
             n += 1
         return (total / max(n, 1)).item()
+
+    # ------------------------------------------------- optimizer side-car
+    def opt_state_dict(self) -> Dict:
+        """Optimizer/scheduler state for true resume (the reference saves
+        only model weights, /root/reference/main.py:79 — SURVEY.md §5.4
+        calls out the side-car as a new-engine addition)."""
+        return {
+            "adam_m": self.adam_m.detach().cpu(),
+            "adam_v": self.adam_v.detach().cpu(),
+            "step_t": self.step_t.detach().cpu(),
+            "lr": self.lr,
+            "eta_min": self.eta_min,
+            "t_max": self.t_max,
+        }
+
+    def load_opt_state_dict(self, sd: Dict) -> None:
+        self.adam_m.copy_(sd["adam_m"].to(self.device))
+        self.adam_v.copy_(sd["adam_v"].to(self.device))
+        self.step_t.copy_(sd["step_t"].to(self.device))
+        self.lr = float(sd.get("lr", self.lr))
+        self.eta_min = float(sd.get("eta_min", self.eta_min))
+        self.t_max = int(sd.get("t_max", self.t_max))
+        # lr/t_max are baked into captured graphs as kernel args
+        self._graphs.clear()
 
     @torch.no_grad()
     def validate_epoch(self, days) -> float:

@@ -122,10 +122,20 @@ def train_main(args, data_args, df=None) -> float:
     best-val state_dict checkpointing. Distributed-aware: day-sharded
     loaders, flat-bucket gradient all-reduce, val-loss all-reduce,
     rank-0-only checkpointing. Returns best validation loss.
+
+    Engines (args.engine, default "auto" = fused on GPU, eager on CPU):
+    - "fused": the MI355X production path — device-resident epoch cache
+      (one H2D per day ever), hipGraph-captured HIP-kernel step, fused
+      Adam, one-bucket RCCL all-reduce.
+    - "eager": the PyTorch-ROCm oracle path (same math via modules.py).
+    New-engine additions over the reference: JSONL/wandb metrics, a
+    cross-sections/sec step timer, and a true-resume side-car (optimizer
+    + epoch state next to the weights-only reference checkpoint).
     """
     import pandas as pd
 
     from ..data.sampler import init_data_loader
+    from ..observability import MetricsLogger, StepTimer
 
     rank = init_distributed()
     world_size = get_world_size()
@@ -145,47 +155,124 @@ def train_main(args, data_args, df=None) -> float:
         df = pd.read_pickle(args.dataset).iloc[:, :159]
         df = df.rename(columns={df.columns[-1]: "LABEL0"})
 
-    train_dataloader = init_data_loader(
-        df, shuffle=True, step_len=data_args.seq_len,
-        start=data_args.start_time, end=data_args.fit_end_time,
-        select_feature=data_args.select_feature,
-        rank=rank, world_size=world_size, seed=args.seed,
-    )
-    valid_dataloader = init_data_loader(
-        df, shuffle=False, step_len=data_args.seq_len,
-        start=data_args.val_start_time, end=data_args.val_end_time,
-        select_feature=data_args.select_feature,
-        rank=rank, world_size=world_size, seed=args.seed,
-    )
-
-    T_max = len(train_dataloader) * args.num_epochs
     device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
-    factorVAE.to(device)
+    engine = getattr(args, "engine", "auto") or "auto"
+    if engine == "auto":
+        engine = "fused" if device.type == "cuda" else "eager"
 
-    grad_bucket = FlatGradBucket(factorVAE.parameters()) if is_distributed() else None
-    optimizer = optim.Adam(factorVAE.parameters(), lr=args.lr)
-    scheduler = optim.lr_scheduler.CosineAnnealingLR(optimizer, T_max=T_max)
+    logger = MetricsLogger(args.run_name, out_dir=args.save_dir,
+                           use_wandb=getattr(args, "wandb", False),
+                           config={k: v for k, v in vars(args).items()
+                                   if isinstance(v, (int, float, str, bool))},
+                           rank=rank)
+    timer = StepTimer(device)
+    save_root = checkpoint_path(args.save_dir, args.run_name,
+                                args.num_factor, args.hidden_size,
+                                args.num_portfolio, args.seed)
+    sidecar = save_root + ".opt"
+    resume = bool(getattr(args, "resume", False))
 
+    if engine == "fused":
+        from ..data.device_cache import DeviceEpochCache
+        from .fused import FusedTrainer
+
+        # unsharded loaders -> every rank caches all days once (288 GB
+        # HBM3E), then each epoch does the sampler-equivalent global
+        # shuffle + round-robin shard on device-resident tensors
+        train_cache = DeviceEpochCache(init_data_loader(
+            df, shuffle=False, step_len=data_args.seq_len,
+            start=data_args.start_time, end=data_args.fit_end_time,
+            select_feature=data_args.select_feature), device, seed=args.seed)
+        valid_cache = DeviceEpochCache(init_data_loader(
+            df, shuffle=False, step_len=data_args.seq_len,
+            start=data_args.val_start_time, end=data_args.val_end_time,
+            select_feature=data_args.select_feature), device, seed=args.seed)
+
+        steps_per_epoch = train_cache.num_batches(rank, world_size)
+        trainer = FusedTrainer(factorVAE, lr=args.lr,
+                               t_max=steps_per_epoch * args.num_epochs,
+                               device=device)
+        optimizer = scheduler = grad_bucket = None
+    else:
+        train_dataloader = init_data_loader(
+            df, shuffle=True, step_len=data_args.seq_len,
+            start=data_args.start_time, end=data_args.fit_end_time,
+            select_feature=data_args.select_feature,
+            rank=rank, world_size=world_size, seed=args.seed,
+        )
+        valid_dataloader = init_data_loader(
+            df, shuffle=False, step_len=data_args.seq_len,
+            start=data_args.val_start_time, end=data_args.val_end_time,
+            select_feature=data_args.select_feature,
+            rank=rank, world_size=world_size, seed=args.seed,
+        )
+        T_max = len(train_dataloader) * args.num_epochs
+        factorVAE.to(device)
+        grad_bucket = (FlatGradBucket(factorVAE.parameters())
+                       if is_distributed() else None)
+        optimizer = optim.Adam(factorVAE.parameters(), lr=args.lr)
+        scheduler = optim.lr_scheduler.CosineAnnealingLR(optimizer, T_max=T_max)
+        trainer = None
+
+    start_epoch = 0
     best_val_loss = float("inf")
-    for epoch in range(args.num_epochs):
-        if hasattr(train_dataloader.batch_sampler, "set_epoch"):
-            train_dataloader.batch_sampler.set_epoch(epoch)
-        train_loss = train(factorVAE, train_dataloader, optimizer, scheduler,
-                           args, grad_bucket=grad_bucket, device=device)
-        val_loss = validate(factorVAE, valid_dataloader, args, device=device)
+    if resume and os.path.exists(save_root) and os.path.exists(sidecar):
+        state = torch.load(save_root, map_location=device, weights_only=True)
+        factorVAE.load_state_dict(state)
+        side = torch.load(sidecar, map_location="cpu", weights_only=True)
+        start_epoch = int(side.get("epoch", 0))
+        best_val_loss = float(side.get("best_val_loss", float("inf")))
+        if trainer is not None and "fused_opt" in side:
+            trainer.load_opt_state_dict(side["fused_opt"])
+        elif optimizer is not None and "optimizer" in side:
+            optimizer.load_state_dict(side["optimizer"])
+            scheduler.load_state_dict(side["scheduler"])
+        if rank == 0:
+            print(f"Resumed from {save_root} at epoch {start_epoch}")
+
+    for epoch in range(start_epoch, args.num_epochs):
+        timer.start()
+        if engine == "fused":
+            days = list(train_cache.order(epoch, shuffle=True, rank=rank,
+                                          world_size=world_size))
+            train_loss = trainer.train_epoch(days)
+            timer.tick(len(days))
+            val_days = list(valid_cache.order(0, shuffle=False, rank=rank,
+                                              world_size=world_size))
+            rate = timer.rate()
+            val_loss = trainer.validate_epoch(val_days)
+        else:
+            if hasattr(train_dataloader.batch_sampler, "set_epoch"):
+                train_dataloader.batch_sampler.set_epoch(epoch)
+            train_loss = train(factorVAE, train_dataloader, optimizer,
+                               scheduler, args, grad_bucket=grad_bucket,
+                               device=device)
+            timer.tick(len(train_dataloader))
+            rate = timer.rate()
+            val_loss = validate(factorVAE, valid_dataloader, args,
+                                device=device)
         val_loss = all_reduce_scalar(val_loss, device=device)
         train_loss = all_reduce_scalar(train_loss, device=device)
+        rate_total = all_reduce_scalar(rate, device=device, average=False)
 
         if rank == 0:
             print(f"Epoch {epoch + 1}: Train Loss: {train_loss:.4f}, "
-                  f"Validation Loss: {val_loss:.4f}")
+                  f"Validation Loss: {val_loss:.4f}, "
+                  f"{rate_total:.1f} cross-sections/s")
+        logger.log({"Train Loss": train_loss, "Validation Loss": val_loss,
+                    "cross_sections_per_sec": rate_total}, step=epoch)
         if val_loss < best_val_loss:
             best_val_loss = val_loss
             if rank == 0:
-                save_root = checkpoint_path(args.save_dir, args.run_name,
-                                            args.num_factor, args.hidden_size,
-                                            args.num_portfolio, args.seed)
                 torch.save(factorVAE.state_dict(), save_root)
+                side = {"epoch": epoch + 1, "best_val_loss": best_val_loss}
+                if trainer is not None:
+                    side["fused_opt"] = trainer.opt_state_dict()
+                else:
+                    side["optimizer"] = optimizer.state_dict()
+                    side["scheduler"] = scheduler.state_dict()
+                torch.save(side, sidecar)
                 print(f"Model saved at {save_root}")
 
+    logger.finish({"Best Validation Loss": best_val_loss})
     return best_val_loss
