@@ -64,8 +64,17 @@ class FusedTrainer:
     def __init__(self, model: FactorVAE, lr: float, t_max: int,
                  device: Optional[torch.device] = None, eta_min: float = 0.0,
                  use_graph: bool = True, max_stocks: Optional[int] = None,
-                 train: bool = True):
+                 train: bool = True, dtype: str = "fp32"):
+        assert dtype in ("fp32", "bf16")
         self.ext = get_extension()
+        # bf16 mode (BASELINE.json configs 2-4): the FLOP-bound extractor
+        # GEMM family (R = N*T rows) runs on bf16 MFMA (~2.5 PF/s dense on
+        # gfx950 vs 157 TF/s f32) with fp32 accumulate; fp32 master
+        # weights + fp32 Adam + fp32 grad all-reduce; the latency-bound
+        # N-row kernels (encoder/attention/decoder/GRU recurrence) stay
+        # fp32 — their cost is dispatch+LDS, not FLOPs.
+        self.bf16 = dtype == "bf16"
+        self.dtype = dtype
         self.model = model
         self.device = device or torch.device("cuda")
         model.to(self.device)
@@ -84,6 +93,13 @@ class FusedTrainer:
         assert self.H <= 64, "fused engine supports hidden_size <= 64"
 
         self._build_param_arena()
+        if self.bf16:
+            C3 = 3 * self.H
+            self.w1x_bf = torch.empty(self.C, self.C, dtype=torch.bfloat16,
+                                      device=self.device)
+            self.wih_bf = torch.empty(C3, self.C, dtype=torch.bfloat16,
+                                      device=self.device)
+            self._refresh_bf16_shadows()
         self.grads = torch.zeros_like(self.params.flat)
         self.adam_m = torch.zeros_like(self.params.flat)
         self.adam_v = torch.zeros_like(self.params.flat)
@@ -269,8 +285,20 @@ class FusedTrainer:
         w["tn_part"] = f(32 * max_mn)
         w["tn_part2"] = f(32 * max_mn)
         w["tn_part3"] = f(32 * max_mn)
+        max_m = max(3 * H, C, M, K)
+        w["tn_partb"] = f(32 * max_m)
+        w["tn_partb2"] = f(32 * max_m)
+        w["tn_partb3"] = f(32 * max_m)
         w["dzx"] = f(R, C)
         w["dxln"] = f(R, C)
+        if self.bf16:
+            fb = lambda *shape: torch.zeros(*shape, device=d,
+                                            dtype=torch.bfloat16)
+            w["xln_bf"] = fb(R, C)
+            w["xp_bf"] = fb(R, C)
+            w["dgi_bf"] = fb(R, 3 * H)
+            w["dxp_bf"] = fb(R, C)
+            w["dzx_bf"] = fb(R, C)
         self._ws_cache[(N, T)] = w
         self.ws = w
         self._ws_n = N
@@ -286,10 +314,18 @@ class FusedTrainer:
         yv = w["y"] if y is None else y
         alpha = 1.0 / math.sqrt(float(H) + 1e-6)
 
-        ext.ln_fwd(x2d, p("ln_g"), p("ln_b"), w["xln"],
-                   w["mean"], w["rstd"], 1e-5)
-        ext.gemm_nt(w["xln"], p("W1x"), p("b1x"), w["xp"], 1.0, False, True)
-        ext.gemm_nt(w["xp"], p("Wih"), p("bih"), w["gi"], 1.0, False, False)
+        if self.bf16:
+            ext.ln_fwd(x2d, p("ln_g"), p("ln_b"), None, w["mean"],
+                       w["rstd"], 1e-5, w["xln_bf"])
+            ext.gemm_nt_bf16(w["xln_bf"], self.w1x_bf, p("b1x"), None,
+                             w["xp_bf"], 1.0, False, True)
+            ext.gemm_nt_bf16(w["xp_bf"], self.wih_bf, p("bih"), w["gi"],
+                             None, 1.0, False, False)
+        else:
+            ext.ln_fwd(x2d, p("ln_g"), p("ln_b"), w["xln"],
+                       w["mean"], w["rstd"], 1e-5)
+            ext.gemm_nt(w["xln"], p("W1x"), p("b1x"), w["xp"], 1.0, False, True)
+            ext.gemm_nt(w["xp"], p("Wih"), p("bih"), w["gi"], 1.0, False, False)
         ext.gru_fwd(w["gi"], p("Whh"), p("bhh"), w["h"], w["h_seq"],
                     w["h_prev"], w["gates4"], N, T, H)
         ext.gemm_nt(w["h"], p("Wenc"), p("benc"), w["scores_enc"], 1.0, False, False)
@@ -363,10 +399,8 @@ class FusedTrainer:
                     g("bmu_d"), g("wsig_d"), g("bsig_d"))
         fork()
         with _on_side():
-            ext.gemm_tn(w["dz1"], w["h"], g("W1d"), None, 1, True)
-            ext.colsum(w["dz1"], g("b1d"), 1)
-            ext.gemm_tn(w["dbeta"], w["h"], g("Wb"), None, 1, True)
-            ext.colsum(w["dbeta"], g("bb"), 1)
+            ext.gemm_tn(w["dz1"], w["h"], g("W1d"), None, 1, True, g("b1d"))
+            ext.gemm_tn(w["dbeta"], w["h"], g("Wb"), None, 1, True, g("bb"))
 
         # predictor MLP + attention backward
         ext.pred_mlp_bwd(w["dpmu"], w["dpsig_c"], w["psig"], w["psig_pre"],
@@ -374,8 +408,7 @@ class FusedTrainer:
                          g("wmu_p"), g("bmu_p"), g("wsig_p"), g("bsig_p"))
         fork()
         with _on_side():
-            ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True)
-            ext.colsum(w["dz2"], g("bl"), 1)
+            ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True, g("bl"))
         ext.gemm_nn(w["dz2"], p("Wl"), None, w["dctx"], 1.0, False, False)
         gWv = self._gstack("Wv.0", (K, H, H))
         gbv = self._gstack("bv.0", (K, H))
@@ -403,30 +436,49 @@ class FusedTrainer:
         ext.enc_softmax_bwd(w["dyp"], w["a_enc"], yv, w["dscores"])
         fork()
         with _on_side():
-            ext.gemm_tn(w["dscores"], w["h"], g("Wenc"), None, 1, True)
-            ext.colsum(w["dscores"], g("benc"), 1)
+            ext.gemm_tn(w["dscores"], w["h"], g("Wenc"), None, 1, True, g("benc"))
         ext.gemm_nn(w["dscores"], p("Wenc"), None, w["dh"], 1.0, True, False)
 
         # extractor backward
         ext.gru_bwd(w["dh"], w["h_prev"], w["gates4"], p("Whh"), w["dgi"],
                     w["dgh"], N, T, H)
-        fork()
-        with _on_side():
-            ext.gemm_tn(w["dgh"].view(R, 3 * H), w["h_prev"].view(R, H),
-                        g("Whh"), w["tn_part"], chunks, True)
-            ext.colsum(w["dgh"].view(R, 3 * H), g("bhh"), chunks)
-            ext.gemm_tn(w["dgi"].view(R, 3 * H), w["xp"], g("Wih"),
-                        w["tn_part2"], chunks, True)
-            ext.colsum(w["dgi"].view(R, 3 * H), g("bih"), chunks)
-        ext.gemm_nn(w["dgi"].view(R, 3 * H), p("Wih"), None, w["dxp"], 1.0,
-                    False, False)
-        ext.lrelu_bwd(w["dxp"], w["xp"], w["dzx"])
-        fork()
-        with _on_side():
-            ext.gemm_tn(w["dzx"], w["xln"], g("W1x"), w["tn_part3"], chunks,
-                        True)
-            ext.colsum(w["dzx"], g("b1x"), chunks)
-        ext.gemm_nn(w["dzx"], p("W1x"), None, w["dxln"], 1.0, False, False)
+        if self.bf16:
+            ext.cast_f32_bf16(w["dgi"].view(-1), w["dgi_bf"].view(-1))
+            fork()
+            with _on_side():
+                ext.gemm_tn(w["dgh"].view(R, 3 * H), w["h_prev"].view(R, H),
+                            g("Whh"), w["tn_part"], chunks, True,
+                            g("bhh"), w["tn_partb"])
+                ext.gemm_tn_bf16(w["dgi_bf"].view(R, 3 * H), w["xp_bf"],
+                                 g("Wih"), w["tn_part2"], chunks, True,
+                                 g("bih"), w["tn_partb2"])
+            ext.gemm_nn_bf16(w["dgi_bf"].view(R, 3 * H), self.wih_bf, None,
+                             None, w["dxp_bf"], 1.0, False, False)
+            ext.lrelu_bwd_bf16(w["dxp_bf"], w["xp_bf"], w["dzx_bf"])
+            fork()
+            with _on_side():
+                ext.gemm_tn_bf16(w["dzx_bf"], w["xln_bf"], g("W1x"),
+                                 w["tn_part3"], chunks, True, g("b1x"),
+                                 w["tn_partb3"])
+            ext.gemm_nn_bf16(w["dzx_bf"], self.w1x_bf, None, w["dxln"],
+                             None, 1.0, False, False)
+        else:
+            fork()
+            with _on_side():
+                ext.gemm_tn(w["dgh"].view(R, 3 * H), w["h_prev"].view(R, H),
+                            g("Whh"), w["tn_part"], chunks, True,
+                            g("bhh"), w["tn_partb"])
+                ext.gemm_tn(w["dgi"].view(R, 3 * H), w["xp"], g("Wih"),
+                            w["tn_part2"], chunks, True, g("bih"),
+                            w["tn_partb2"])
+            ext.gemm_nn(w["dgi"].view(R, 3 * H), p("Wih"), None, w["dxp"], 1.0,
+                        False, False)
+            ext.lrelu_bwd(w["dxp"], w["xp"], w["dzx"])
+            fork()
+            with _on_side():
+                ext.gemm_tn(w["dzx"], w["xln"], g("W1x"), w["tn_part3"], chunks,
+                            True, g("b1x"), w["tn_partb3"])
+            ext.gemm_nn(w["dzx"], p("W1x"), None, w["dxln"], 1.0, False, False)
         ext.ln_bwd_params(x2d, w["dxln"], w["mean"], w["rstd"],
                           g("ln_g"), g("ln_b"), chunks)
         # join: main waits for all side-stream wgrad work
@@ -435,12 +487,18 @@ class FusedTrainer:
             e.record(side)
             main.wait_event(e)
 
+    def _refresh_bf16_shadows(self):
+        self.ext.cast_f32_bf16(self.p("W1x"), self.w1x_bf)
+        self.ext.cast_f32_bf16(self.p("Wih"), self.wih_bf)
+
     def _launch_optimizer(self, inc: bool = True):
         if inc:
             self.ext.step_inc(self.step_t)
         self.ext.adam(self.params.flat, self.grads, self.adam_m, self.adam_v,
                       self.step_t, self.lr, self.eta_min, float(self.t_max),
                       0.9, 0.999, 1e-8)
+        if self.bf16:
+            self._refresh_bf16_shadows()
 
     def _fill_rng(self, N: int):
         self.ws["eps"].normal_()
@@ -709,6 +767,8 @@ class FusedTrainer:
         self.t_max = int(sd.get("t_max", self.t_max))
         # lr/t_max are baked into captured graphs as kernel args
         self._graphs.clear()
+        if self.bf16:  # weights typically re-loaded alongside opt state
+            self._refresh_bf16_shadows()
 
     @torch.no_grad()
     def validate_epoch(self, days) -> float:

@@ -191,7 +191,8 @@ def train_main(args, data_args, df=None) -> float:
         steps_per_epoch = train_cache.num_batches(rank, world_size)
         trainer = FusedTrainer(factorVAE, lr=args.lr,
                                t_max=steps_per_epoch * args.num_epochs,
-                               device=device)
+                               device=device,
+                               dtype=getattr(args, "dtype", "fp32"))
         optimizer = scheduler = grad_bucket = None
     else:
         train_dataloader = init_data_loader(

@@ -40,6 +40,16 @@ def build_argparser() -> argparse.ArgumentParser:
     parser.add_argument("--num_workers", type=int, default=4)
     parser.add_argument("--wandb", action="store_true",
                         help="log to wandb if installed (no-op otherwise)")
+    parser.add_argument("--engine", type=str, default="auto",
+                        choices=["auto", "fused", "eager"],
+                        help="auto = fused HIP engine on GPU, eager on CPU")
+    parser.add_argument("--dtype", type=str, default="fp32",
+                        choices=["fp32", "bf16"],
+                        help="fused-engine compute dtype for the extractor "
+                             "GEMM family (bf16 MFMA path)")
+    parser.add_argument("--resume", action="store_true",
+                        help="resume from the checkpoint + optimizer "
+                             "side-car in save_dir if present")
     return parser
 
 

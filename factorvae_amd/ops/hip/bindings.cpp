@@ -12,12 +12,23 @@ hipError_t fv_gemm_nt(const float*, const float*, const float*, float*, int,
                       int, int, float, int, int, hipStream_t);
 hipError_t fv_gemm_nn(const float*, const float*, const float*, float*, int,
                       int, int, float, int, int, hipStream_t);
-hipError_t fv_gemm_tn(const float*, const float*, float*, float*, int, int,
-                      int, int, int, hipStream_t);
+hipError_t fv_gemm_tn(const float*, const float*, float*, float*, float*,
+                      float*, int, int, int, int, int, hipStream_t);
 hipError_t fv_colsum(const float*, float*, int, int, int, hipStream_t);
+hipError_t fv_gemm_nt_bf16(const void*, const void*, const float*, float*,
+                           void*, int, int, int, float, int, int,
+                           hipStream_t);
+hipError_t fv_gemm_nn_bf16(const void*, const void*, const float*, float*,
+                           void*, int, int, int, float, int, int,
+                           hipStream_t);
+hipError_t fv_gemm_tn_bf16(const void*, const void*, float*, float*, float*,
+                           float*, int, int, int, int, int, hipStream_t);
+hipError_t fv_cast_f32_bf16(const float*, void*, long, hipStream_t);
+hipError_t fv_lrelu_bwd_bf16(const void*, const void*, void*, long,
+                             hipStream_t);
 hipError_t fv_lrelu_bwd(const float*, const float*, float*, long, hipStream_t);
-hipError_t fv_ln_fwd(const float*, const float*, const float*, float*, float*,
-                     float*, long, int, float, hipStream_t);
+hipError_t fv_ln_fwd(const float*, const float*, const float*, float*, void*,
+                     float*, float*, long, int, float, hipStream_t);
 hipError_t fv_ln_bwd_params(const float*, const float*, const float*,
                             const float*, float*, float*, long, int, int,
                             hipStream_t);
@@ -97,6 +108,15 @@ void check_f32(const torch::Tensor& t, const char* name) {
 }
 
 #define CK(t) check_f32(t, #t)
+
+inline void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+}
+#define CKB(t) check_bf16(t, #t)
+inline void* bfp(torch::Tensor& t) { return t.data_ptr(); }
+inline const void* bfpc(const torch::Tensor& t) { return t.data_ptr(); }
 #define RUN(call)                                                      \
   do {                                                                 \
     hipError_t e_ = (call);                                            \
@@ -128,9 +148,99 @@ void gemm_nn(torch::Tensor A, torch::Tensor B,
                  accumulate, act_lrelu, cur_stream()));
 }
 
+// ---------------------------------------------------------------- bf16
+void gemm_nt_bf16(torch::Tensor A, torch::Tensor W,
+                  c10::optional<torch::Tensor> bias,
+                  c10::optional<torch::Tensor> out_f32,
+                  c10::optional<torch::Tensor> out_bf16,
+                  double alpha, bool accumulate, bool act_lrelu) {
+  CKB(A); CKB(W);
+  const int R = A.size(0), Ci = A.size(1), Co = W.size(0);
+  TORCH_CHECK(W.size(1) == Ci);
+  const float* b = nullptr;
+  if (bias.has_value()) { CK(*bias); b = fp(*bias); }
+  float* of = nullptr; void* ob = nullptr;
+  if (out_f32.has_value()) {
+    CK(*out_f32);
+    TORCH_CHECK(out_f32->size(0) == R && out_f32->size(1) == Co);
+    of = fpm(*out_f32);
+  }
+  if (out_bf16.has_value()) {
+    CKB(*out_bf16);
+    TORCH_CHECK(out_bf16->size(0) == R && out_bf16->size(1) == Co);
+    ob = bfp(*out_bf16);
+  }
+  TORCH_CHECK(of || ob, "need at least one output");
+  RUN(fv_gemm_nt_bf16(bfpc(A), bfpc(W), b, of, ob, R, Ci, Co, (float)alpha,
+                      accumulate, act_lrelu, cur_stream()));
+}
+
+void gemm_nn_bf16(torch::Tensor A, torch::Tensor B,
+                  c10::optional<torch::Tensor> bias,
+                  c10::optional<torch::Tensor> out_f32,
+                  c10::optional<torch::Tensor> out_bf16,
+                  double alpha, bool accumulate, bool act_lrelu) {
+  CKB(A); CKB(B);
+  const int R = A.size(0), Ci = A.size(1), Co = B.size(1);
+  TORCH_CHECK(B.size(0) == Ci);
+  const float* b = nullptr;
+  if (bias.has_value()) { CK(*bias); b = fp(*bias); }
+  float* of = nullptr; void* ob = nullptr;
+  if (out_f32.has_value()) {
+    CK(*out_f32);
+    TORCH_CHECK(out_f32->size(0) == R && out_f32->size(1) == Co);
+    of = fpm(*out_f32);
+  }
+  if (out_bf16.has_value()) {
+    CKB(*out_bf16);
+    TORCH_CHECK(out_bf16->size(0) == R && out_bf16->size(1) == Co);
+    ob = bfp(*out_bf16);
+  }
+  TORCH_CHECK(of || ob, "need at least one output");
+  RUN(fv_gemm_nn_bf16(bfpc(A), bfpc(B), b, of, ob, R, Ci, Co, (float)alpha,
+                      accumulate, act_lrelu, cur_stream()));
+}
+
+void gemm_tn_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor out,
+                  c10::optional<torch::Tensor> part, long r_chunks,
+                  bool accumulate, c10::optional<torch::Tensor> db,
+                  c10::optional<torch::Tensor> db_part) {
+  CKB(A); CKB(B); CK(out);
+  const int R = A.size(0), M = A.size(1), N = B.size(1);
+  TORCH_CHECK(B.size(0) == R && out.size(0) == M && out.size(1) == N);
+  float* pp = nullptr;
+  if (part.has_value()) {
+    CK(*part);
+    TORCH_CHECK(part->numel() >= (long)32 * M * N, "partials too small");
+    pp = fpm(*part);
+  }
+  float* dbp = nullptr; float* dbpp = nullptr;
+  if (db.has_value()) {
+    CK(*db);
+    TORCH_CHECK(db->numel() == M);
+    dbp = fpm(*db);
+    if (db_part.has_value()) { CK(*db_part); dbpp = fpm(*db_part); }
+  }
+  RUN(fv_gemm_tn_bf16(bfpc(A), bfpc(B), fpm(out), pp, dbp, dbpp, R, M, N,
+                      (int)r_chunks, accumulate, cur_stream()));
+}
+
+void cast_f32_bf16(torch::Tensor src, torch::Tensor dst) {
+  CK(src); CKB(dst);
+  TORCH_CHECK(src.numel() == dst.numel());
+  RUN(fv_cast_f32_bf16(fp(src), bfp(dst), src.numel(), cur_stream()));
+}
+
+void lrelu_bwd_bf16(torch::Tensor dY, torch::Tensor Y, torch::Tensor dZ) {
+  CKB(dY); CKB(Y); CKB(dZ);
+  RUN(fv_lrelu_bwd_bf16(bfpc(dY), bfpc(Y), bfp(dZ), dY.numel(),
+                        cur_stream()));
+}
+
 void gemm_tn(torch::Tensor A, torch::Tensor B, torch::Tensor out,
              c10::optional<torch::Tensor> part, long r_chunks,
-             bool accumulate) {
+             bool accumulate, c10::optional<torch::Tensor> db,
+             c10::optional<torch::Tensor> db_part) {
   CK(A); CK(B); CK(out);
   const int R = A.size(0), M = A.size(1), N = B.size(1);
   TORCH_CHECK(B.size(0) == R && out.size(0) == M && out.size(1) == N);
@@ -141,8 +251,21 @@ void gemm_tn(torch::Tensor A, torch::Tensor B, torch::Tensor out,
                 "tn partial workspace too small");
     pp = fpm(*part);
   }
-  RUN(fv_gemm_tn(fp(A), fp(B), fpm(out), pp, R, M, N, (int)r_chunks,
-                 accumulate, cur_stream()));
+  float* dbp = nullptr;
+  float* dbpp = nullptr;
+  if (db.has_value()) {
+    CK(*db);
+    TORCH_CHECK(db->numel() == M, "db must be (M,)");
+    dbp = fpm(*db);
+    if (db_part.has_value()) {
+      CK(*db_part);
+      TORCH_CHECK(db_part->numel() >= (long)32 * M,
+                  "db partial workspace too small");
+      dbpp = fpm(*db_part);
+    }
+  }
+  RUN(fv_gemm_tn(fp(A), fp(B), fpm(out), pp, dbp, dbpp, R, M, N,
+                 (int)r_chunks, accumulate, cur_stream()));
 }
 
 void colsum(torch::Tensor A, torch::Tensor out, long r_chunks) {
@@ -158,12 +281,18 @@ void lrelu_bwd(torch::Tensor dY, torch::Tensor Y, torch::Tensor dZ) {
 }
 
 void ln_fwd(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
-            torch::Tensor xln, torch::Tensor mean, torch::Tensor rstd,
-            double eps) {
-  CK(x); CK(gamma); CK(beta); CK(xln); CK(mean); CK(rstd);
+            c10::optional<torch::Tensor> xln,
+            torch::Tensor mean, torch::Tensor rstd,
+            double eps, c10::optional<torch::Tensor> xln_bf = c10::nullopt) {
+  CK(x); CK(gamma); CK(beta); CK(mean); CK(rstd);
   const long R = x.numel() / x.size(-1);
   const int C = x.size(-1);
-  RUN(fv_ln_fwd(fp(x), fp(gamma), fp(beta), fpm(xln), fpm(mean), fpm(rstd), R,
+  float* xo = nullptr;
+  if (xln.has_value()) { CK(*xln); xo = fpm(*xln); }
+  void* xb = nullptr;
+  if (xln_bf.has_value()) { CKB(*xln_bf); xb = bfp(*xln_bf); }
+  TORCH_CHECK(xo || xb, "ln_fwd needs an output");
+  RUN(fv_ln_fwd(fp(x), fp(gamma), fp(beta), xo, xb, fpm(mean), fpm(rstd), R,
                 C, (float)eps, cur_stream()));
 }
 
@@ -385,10 +514,26 @@ void adam(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_nt", &gemm_nt);
   mod.def("gemm_nn", &gemm_nn);
-  mod.def("gemm_tn", &gemm_tn);
+  mod.def("gemm_tn", &gemm_tn, py::arg("A"), py::arg("B"), py::arg("out"),
+          py::arg("part"), py::arg("r_chunks"), py::arg("accumulate"),
+          py::arg("db") = py::none(), py::arg("db_part") = py::none());
   mod.def("colsum", &colsum);
+  mod.def("gemm_nt_bf16", &gemm_nt_bf16, py::arg("A"), py::arg("W"),
+          py::arg("bias"), py::arg("out_f32"), py::arg("out_bf16"),
+          py::arg("alpha"), py::arg("accumulate"), py::arg("act_lrelu"));
+  mod.def("gemm_nn_bf16", &gemm_nn_bf16, py::arg("A"), py::arg("B"),
+          py::arg("bias"), py::arg("out_f32"), py::arg("out_bf16"),
+          py::arg("alpha"), py::arg("accumulate"), py::arg("act_lrelu"));
+  mod.def("gemm_tn_bf16", &gemm_tn_bf16, py::arg("A"), py::arg("B"),
+          py::arg("out"), py::arg("part"), py::arg("r_chunks"),
+          py::arg("accumulate"), py::arg("db") = py::none(),
+          py::arg("db_part") = py::none());
+  mod.def("cast_f32_bf16", &cast_f32_bf16);
+  mod.def("lrelu_bwd_bf16", &lrelu_bwd_bf16);
   mod.def("lrelu_bwd", &lrelu_bwd);
-  mod.def("ln_fwd", &ln_fwd);
+  mod.def("ln_fwd", &ln_fwd, py::arg("x"), py::arg("gamma"), py::arg("beta"),
+          py::arg("xln"), py::arg("mean"), py::arg("rstd"), py::arg("eps"),
+          py::arg("xln_bf") = py::none());
   mod.def("ln_bwd_params", &ln_bwd_params);
   mod.def("gru_fwd", &gru_fwd);
   mod.def("gru_bwd", &gru_bwd);

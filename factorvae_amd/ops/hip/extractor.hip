@@ -9,6 +9,7 @@
 __global__ __launch_bounds__(256) void ln_fwd_kernel(
     const float* __restrict__ x, const float* __restrict__ gamma,
     const float* __restrict__ beta, float* __restrict__ xln,
+    __bf16* __restrict__ xln_bf,
     float* __restrict__ mean, float* __restrict__ rstd,
     long R, int C, float eps) {
   const int wid = threadIdx.x >> 6;
@@ -36,9 +37,13 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
     mean[row] = mu;
     rstd[row] = rs;
   }
-  float* o = xln + row * C;
-  for (int c = lane; c < C; c += 64)
-    o[c] = fmaf((xr[c] - mu) * rs, gamma[c], beta[c]);
+  float* o = xln ? xln + row * C : nullptr;
+  __bf16* ob = xln_bf ? xln_bf + row * C : nullptr;
+  for (int c = lane; c < C; c += 64) {
+    const float v_ = fmaf((xr[c] - mu) * rs, gamma[c], beta[c]);
+    if (o) o[c] = v_;
+    if (ob) ob[c] = (__bf16)v_;
+  }
 }
 
 // Parameter grads only (x is input data, no dx needed):
@@ -78,11 +83,11 @@ __global__ __launch_bounds__(256) void ln_bwd_params_kernel(
 extern "C" {
 
 hipError_t fv_ln_fwd(const float* x, const float* gamma, const float* beta,
-                     float* xln, float* mean, float* rstd, long R, int C,
-                     float eps, hipStream_t stream) {
+                     float* xln, void* xln_bf, float* mean, float* rstd,
+                     long R, int C, float eps, hipStream_t stream) {
   dim3 grid((unsigned)((R + 3) / 4));
   hipLaunchKernelGGL(ln_fwd_kernel, grid, dim3(256), 0, stream,
-                     x, gamma, beta, xln, mean, rstd, R, C, eps);
+                     x, gamma, beta, xln, (__bf16*)xln_bf, mean, rstd, R, C, eps);
   HIP_CHECK_LAST();
   return hipSuccess;
 }

@@ -150,12 +150,17 @@ __global__ __launch_bounds__(256) void gemm_nn_kernel(
 // gridDim.z == 1: plain write to out. gridDim.z > 1: each z-slice writes
 // its partial tile to part[z][M][N] (plain stores); tn_reduce_kernel then
 // sums slices in FIXED order into out — deterministic, no atomics.
+// Fused bias-grad epilogue: when db != null, blocks with blockIdx.y == 0
+// also emit db(M) (+)= sum_r A(R,M) — the bias gradient of the same
+// layer — from the A tiles they already stage (kills the separate
+// colsum pass; partials go to db_part[z][M] when chunked).
 #define TM 32
 #define TN_ 32
 #define TKR 32
 __global__ __launch_bounds__(256) void gemm_tn_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
     float* __restrict__ out, float* __restrict__ part,
+    float* __restrict__ db, float* __restrict__ db_part,
     int R, int M, int N, int accumulate) {
   __shared__ float As[TKR][TM + 1];
   __shared__ float Bs[TKR][TN_ + 1];
@@ -174,6 +179,11 @@ __global__ __launch_bounds__(256) void gemm_tn_kernel(
   const int mt = (wv & 1) * 16;   // m-tile offset
   const int nt = (wv >> 1) * 16;  // n-tile offset
 
+  const bool do_bias = (db != nullptr) && (blockIdx.y == 0);
+  const int bcol = tid & 31;       // bias: thread -> column m0+bcol
+  const int bgrp = tid >> 5;       // 8 row-groups
+  float bsum = 0.0f;
+
   f32x4 acc = {0, 0, 0, 0};
 
   for (int r0 = rbeg; r0 < rend; r0 += TKR) {
@@ -188,6 +198,10 @@ __global__ __launch_bounds__(256) void gemm_tn_kernel(
       Bs[rr][nn] = (gr < rend && gn < N) ? B[(long)gr * N + gn] : 0.0f;
     }
     __syncthreads();
+    if (do_bias) {
+#pragma unroll
+      for (int rr = bgrp; rr < TKR; rr += 8) bsum += As[rr][bcol];
+    }
     // out[m][n] = sum_r A[r][m]*B[r][n]: MFMA with k = r
 #pragma unroll
     for (int r4 = 0; r4 < TKR; r4 += 4) {
@@ -198,10 +212,29 @@ __global__ __launch_bounds__(256) void gemm_tn_kernel(
     __syncthreads();
   }
 
+  const bool direct = (gridDim.z == 1);
+  if (do_bias) {
+    // LDS-reduce the 8 row-group partials per column (reuse As storage)
+    As[bgrp][bcol] = bsum;
+    __syncthreads();
+    if (tid < TM) {
+      float s = 0.0f;
+#pragma unroll
+      for (int gq = 0; gq < 8; ++gq) s += As[gq][tid];
+      const int gm = m0 + tid;
+      if (gm < M) {
+        if (direct) {
+          if (accumulate) db[gm] += s; else db[gm] = s;
+        } else {
+          db_part[(long)blockIdx.z * M + gm] = s;
+        }
+      }
+    }
+  }
+
   const int gn = n0 + nt + fi;
   if (gn >= N) return;
   float* dst;
-  const bool direct = (gridDim.z == 1);
   float* po = direct ? out : part + (long)blockIdx.z * M * N;
 #pragma unroll
   for (int rr = 0; rr < 4; ++rr) {
@@ -215,15 +248,23 @@ __global__ __launch_bounds__(256) void gemm_tn_kernel(
   }
 }
 
-// out[e] += sum_z part[z][e] in fixed z order (deterministic reduce).
+// out[e] += sum_z part[z][e]; db[m] += sum_z db_part[z][m] — fixed z
+// order (deterministic reduce, no atomics).
 __global__ __launch_bounds__(256) void tn_reduce_kernel(
     const float* __restrict__ part, float* __restrict__ out, long elems,
+    const float* __restrict__ db_part, float* __restrict__ db, long m_elems,
     int z) {
   const long e = (long)blockIdx.x * 256 + threadIdx.x;
-  if (e >= elems) return;
-  float s = 0.0f;
-  for (int c = 0; c < z; ++c) s += part[(long)c * elems + e];
-  out[e] += s;
+  if (e < elems) {
+    float s = 0.0f;
+    for (int c = 0; c < z; ++c) s += part[(long)c * elems + e];
+    out[e] += s;
+  } else if (e < elems + m_elems) {
+    const long m = e - elems;
+    float s = 0.0f;
+    for (int c = 0; c < z; ++c) s += db_part[(long)c * m_elems + m];
+    db[m] += s;
+  }
 }
 
 // out(C) (+)= sum_r A(R,C): grid (ceil(C/64), ceil(R/CS_ROWS)); 256
@@ -284,7 +325,8 @@ hipError_t fv_gemm_nn(const float* A, const float* B, const float* bias,
 }
 
 hipError_t fv_gemm_tn(const float* A, const float* B, float* out,
-                      float* part, int R, int M, int N, int r_chunks,
+                      float* part, float* db, float* db_part,
+                      int R, int M, int N, int r_chunks,
                       int accumulate, hipStream_t stream) {
   // r_chunks > 1 is a request to chunk the R-reduction; the launcher
   // picks the actual split (>=512 rows per chunk, <=32 slices).
@@ -295,15 +337,17 @@ hipError_t fv_gemm_tn(const float* A, const float* B, float* out,
     if (r_chunks > 32) r_chunks = 32;
     if (r_chunks < 1) r_chunks = 1;
   }
+  if (r_chunks > 1 && db && !db_part) r_chunks = 1;  // need partial space
   dim3 grid((M + TM - 1) / TM, (N + TN_ - 1) / TN_, r_chunks);
   hipLaunchKernelGGL(gemm_tn_kernel, grid, dim3(256), 0, stream,
-                     A, B, out, part, R, M, N, accumulate);
+                     A, B, out, part, db, db_part, R, M, N, accumulate);
   HIP_CHECK_LAST();
   if (r_chunks > 1) {
     const long elems = (long)M * N;
-    dim3 rgrid((unsigned)((elems + 255) / 256));
+    const long m_elems = db ? M : 0;
+    dim3 rgrid((unsigned)((elems + m_elems + 255) / 256));
     hipLaunchKernelGGL(tn_reduce_kernel, rgrid, dim3(256), 0, stream,
-                       part, out, elems, r_chunks);
+                       part, out, elems, db_part, db, m_elems, r_chunks);
     HIP_CHECK_LAST();
   }
   return hipSuccess;

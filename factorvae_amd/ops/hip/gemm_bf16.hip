@@ -1,0 +1,503 @@
+// bf16 GEMM kernels for the FactorVAE engine (gfx950) — the compute
+// path of the bf16 engine mode (BASELINE.json configs 2-4).
+//
+// bf16 inputs, fp32 MFMA accumulation (v_mfma_f32_16x16x32_bf16: dense
+// bf16 MFMA peak on MI355X is ~2.5 PF/s vs 157 TF/s for f32-input MFMA),
+// outputs selectable fp32 / bf16 / both. Double-buffered staging with
+// dword-coalesced global loads (a wave's load instruction covers whole
+// 128B segments; bf16 activation tensors always have even row stride so
+// dword granularity is alignment-safe), LDS tiles padded to keep
+// ds_read_b128 16B-aligned and bank-conflict-free.
+//
+//   gemm_nt_bf16: out(R,Co)  = act(alpha*(A(R,Ci) @ W(Co,Ci)^T + bias))
+//   gemm_nn_bf16: out(R,Co)  = act(alpha*(A(R,Ci) @ B(Ci,Co) + bias))
+//   gemm_tn_bf16: out(M,N) (+)= A(R,M)^T @ B(R,N), fp32 out (+ fused
+//                 fp32 bias-grad db = colsum(A)), z-chunked partials
+//                 reduced in fixed order (deterministic)
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+#define BBR 64   // row tile
+#define BBC 64   // col tile
+#define BBK 64   // k tile (2 MFMA k-steps of 32)
+
+DEVINL __bf16 to_bf16(float x) { return (__bf16)x; }
+
+union dw_bf2 {
+  unsigned int u;
+  __bf16 h[2];
+};
+
+// Load one dword (2 bf16) of a row-major bf16 matrix with bounds checks.
+// `col0` must be even; row stride `ld` must be even (always true here:
+// C=158, 3H=192, H=64 are even) so the dword is 4B-aligned.
+DEVINL unsigned int load_dw_guard(const __bf16* p, long row, int col0,
+                                  long nrows, int ncols, int ld) {
+  if (row < nrows) {
+    if (col0 + 1 < ncols)
+      return *(const unsigned int*)(p + row * (long)ld + col0);
+    dw_bf2 d;
+    d.h[0] = (col0 < ncols) ? p[row * (long)ld + col0] : (__bf16)0.0f;
+    d.h[1] = (__bf16)0.0f;
+    return d.u;
+  }
+  return 0u;
+}
+
+// ---------------------------------------------------------------- NT
+// A (R,Ci) bf16 row-major, W (Co,Ci) bf16 row-major: both k-contiguous.
+// flags bit0: accumulate (fp32 out only); bit1: lrelu; bit2: bias (fp32)
+// out_f32 / out_bf16: either or both may be non-null.
+//
+// Staging: tile is 64 rows x 64 bf16 = 64x32 dwords; thread t handles
+// dwords t, t+256, ... in row-major dword order -> lanes 0..31 cover one
+// full 128B row, perfectly coalesced. 8 dwords/thread/operand in
+// registers for the double buffer.
+__global__ __launch_bounds__(256) void gemm_nt_bf16_kernel(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ W,
+    const float* __restrict__ bias, float* __restrict__ out_f32,
+    __bf16* __restrict__ out_bf16, int R, int Ci, int Co, float alpha,
+    int flags) {
+  __shared__ __bf16 As[2][BBR][BBK + 8];
+  __shared__ __bf16 Ws[2][BBC][BBK + 8];
+
+  const int r0 = blockIdx.x * BBR;
+  const int c0 = blockIdx.y * BBC;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;        // wave -> 16-row strip
+  const int fi = lane & 15;       // fragment row/col
+  const int fk = lane >> 4;       // fragment k-group (0..3), 8 k each
+
+  const int KD = BBK / 2;         // dwords per row (32)
+  // dword (row, col-pair) handled at iteration u: idx = tid + u*256
+  f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+
+  const int ktiles = (Ci + BBK - 1) / BBK;
+  unsigned int pa[8], pw[8];
+
+  auto stage_regs = [&](int k0) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int idx = tid + u * 256;
+      const int row = idx / KD;          // 0..63
+      const int cp = (idx % KD) * 2;     // even col within tile
+      pa[u] = load_dw_guard(A, (long)r0 + row, k0 + cp, R, Ci, Ci);
+      pw[u] = load_dw_guard(W, (long)c0 + row, k0 + cp, Co, Ci, Ci);
+    }
+  };
+  auto regs_to_lds = [&](int buf) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int idx = tid + u * 256;
+      const int row = idx / KD;
+      const int cp = (idx % KD) * 2;
+      *(unsigned int*)&As[buf][row][cp] = pa[u];
+      *(unsigned int*)&Ws[buf][row][cp] = pw[u];
+    }
+  };
+
+  stage_regs(0);
+  regs_to_lds(0);
+
+  for (int kt = 0; kt < ktiles; ++kt) {
+    __syncthreads();
+    if (kt + 1 < ktiles) stage_regs((kt + 1) * BBK);
+    const int buf = kt & 1;
+#pragma unroll
+    for (int k32 = 0; k32 < BBK; k32 += 32) {
+      const bf16x8 a = *(const bf16x8*)&As[buf][wv * 16 + fi][k32 + fk * 8];
+#pragma unroll
+      for (int jt = 0; jt < 4; ++jt) {
+        const bf16x8 b = *(const bf16x8*)&Ws[buf][jt * 16 + fi][k32 + fk * 8];
+        acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[jt], 0, 0, 0);
+      }
+    }
+    if (kt + 1 < ktiles) {
+      __syncthreads();
+      regs_to_lds(1 - buf);
+    }
+  }
+
+#pragma unroll
+  for (int jt = 0; jt < 4; ++jt) {
+    const int gc = c0 + jt * 16 + fi;
+    if (gc >= Co) continue;
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int gr = r0 + wv * 16 + fk * 4 + rr;
+      if (gr >= R) continue;
+      float v = acc[jt][rr];
+      if (flags & 4) v += bias[gc];
+      v *= alpha;
+      if (flags & 2) v = lrelu_(v);
+      if (out_f32) {
+        float* o = &out_f32[(long)gr * Co + gc];
+        if (flags & 1) v += *o;
+        *o = v;
+      }
+      if (out_bf16) out_bf16[(long)gr * Co + gc] = to_bf16(v);
+    }
+  }
+}
+
+// ---------------------------------------------------------------- NN
+// B (Ci,Co) bf16 row-major: k runs over rows -> dword-coalesced loads of
+// B's rows, transpose-scattered into Bs[n][k] (two b16 LDS stores per
+// dword).
+__global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    const float* __restrict__ bias, float* __restrict__ out_f32,
+    __bf16* __restrict__ out_bf16, int R, int Ci, int Co, float alpha,
+    int flags) {
+  __shared__ __bf16 As[2][BBR][BBK + 8];
+  __shared__ __bf16 Bs[2][BBC][BBK + 8];
+
+  const int r0 = blockIdx.x * BBR;
+  const int c0 = blockIdx.y * BBC;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const int fi = lane & 15;
+  const int fk = lane >> 4;
+
+  const int KD = BBK / 2;
+  f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+
+  const int ktiles = (Ci + BBK - 1) / BBK;
+  unsigned int pa[8], pb[8];
+
+  auto stage_regs = [&](int k0) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int idx = tid + u * 256;
+      const int row = idx / KD;
+      const int cp = (idx % KD) * 2;
+      pa[u] = load_dw_guard(A, (long)r0 + row, k0 + cp, R, Ci, Ci);
+      // B tile: 64 k-rows x 64 cols; dword idx: krow = idx/32, colpair
+      pb[u] = load_dw_guard(B, (long)k0 + row, c0 + cp, Ci, Co, Co);
+    }
+  };
+  auto regs_to_lds = [&](int buf) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int idx = tid + u * 256;
+      const int row = idx / KD;
+      const int cp = (idx % KD) * 2;
+      *(unsigned int*)&As[buf][row][cp] = pa[u];
+      dw_bf2 d;
+      d.u = pb[u];
+      Bs[buf][cp][row] = d.h[0];       // transpose scatter
+      Bs[buf][cp + 1][row] = d.h[1];
+    }
+  };
+
+  stage_regs(0);
+  regs_to_lds(0);
+
+  for (int kt = 0; kt < ktiles; ++kt) {
+    __syncthreads();
+    if (kt + 1 < ktiles) stage_regs((kt + 1) * BBK);
+    const int buf = kt & 1;
+#pragma unroll
+    for (int k32 = 0; k32 < BBK; k32 += 32) {
+      const bf16x8 a = *(const bf16x8*)&As[buf][wv * 16 + fi][k32 + fk * 8];
+#pragma unroll
+      for (int jt = 0; jt < 4; ++jt) {
+        const bf16x8 b = *(const bf16x8*)&Bs[buf][jt * 16 + fi][k32 + fk * 8];
+        acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[jt], 0, 0, 0);
+      }
+    }
+    if (kt + 1 < ktiles) {
+      __syncthreads();
+      regs_to_lds(1 - buf);
+    }
+  }
+
+#pragma unroll
+  for (int jt = 0; jt < 4; ++jt) {
+    const int gc = c0 + jt * 16 + fi;
+    if (gc >= Co) continue;
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int gr = r0 + wv * 16 + fk * 4 + rr;
+      if (gr >= R) continue;
+      float v = acc[jt][rr];
+      if (flags & 4) v += bias[gc];
+      v *= alpha;
+      if (flags & 2) v = lrelu_(v);
+      if (out_f32) {
+        float* o = &out_f32[(long)gr * Co + gc];
+        if (flags & 1) v += *o;
+        *o = v;
+      }
+      if (out_bf16) out_bf16[(long)gr * Co + gc] = to_bf16(v);
+    }
+  }
+}
+
+// ---------------------------------------------------------------- TN
+// out(M,N) (+)= A(R,M)^T @ B(R,N), k = R: dword-coalesced loads of the
+// R-major rows, transpose-scattered into [m][k]/[n][k] LDS tiles. fp32
+// out/partials (weight grads); fused fp32 bias-grad db = colsum(A) from
+// y==0 blocks. 64x64 output tile (4 waves, 2x2 16x16 frags each) with
+// k-tile 64.
+#define TBM 64
+#define TBN 64
+#define TBK 64
+__global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    float* __restrict__ out, float* __restrict__ part,
+    float* __restrict__ db, float* __restrict__ db_part,
+    int R, int M, int N, int accumulate) {
+  __shared__ __bf16 As[2][TBM][TBK + 8];
+  __shared__ __bf16 Bs[2][TBN][TBK + 8];
+
+  const int m0 = blockIdx.x * TBM;
+  const int n0 = blockIdx.y * TBN;
+  const int chunk = (R + gridDim.z - 1) / gridDim.z;
+  const int rbeg = blockIdx.z * chunk;
+  const int rend = min(rbeg + chunk, R);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;        // wave -> 16(m)x64(n) strip as 4 frags
+  const int fi = lane & 15;
+  const int fk = lane >> 4;
+
+  const int MD = TBM / 2;         // dword cols per k-row (32)
+
+  const bool do_bias = (db != nullptr) && (blockIdx.y == 0);
+  float bsum[2] = {0.0f, 0.0f};   // per-thread: 2 adjacent m columns
+
+  f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+
+  const int span = rend - rbeg;
+  const int ktiles = (span + TBK - 1) / TBK;
+  unsigned int pa[8], pb[8];
+
+  // staging: tile is 64 k-rows x 64 cols = 64x32 dwords
+  auto stage_regs = [&](int r0_) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int idx = tid + u * 256;
+      const int krow = idx / MD;
+      const int cp = (idx % MD) * 2;
+      const long gr = (long)r0_ + krow;
+      const long rows = rend;  // guard against rend, not R
+      pa[u] = load_dw_guard(A, gr, m0 + cp, rows, M, M);
+      pb[u] = load_dw_guard(B, gr, n0 + cp, rows, N, N);
+    }
+  };
+  auto regs_to_lds = [&](int buf) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int idx = tid + u * 256;
+      const int krow = idx / MD;
+      const int cp = (idx % MD) * 2;
+      dw_bf2 da_, db_;
+      da_.u = pa[u];
+      db_.u = pb[u];
+      As[buf][cp][krow] = da_.h[0];
+      As[buf][cp + 1][krow] = da_.h[1];
+      Bs[buf][cp][krow] = db_.h[0];
+      Bs[buf][cp + 1][krow] = db_.h[1];
+      if (do_bias) {
+        // colsum(A) accumulated at stage time (fp32 adds); thread owns
+        // columns (m0+cp, m0+cp+1) footprints of its own dwords
+      }
+    }
+  };
+
+  if (ktiles > 0) {
+    stage_regs(rbeg);
+    regs_to_lds(0);
+  }
+
+  for (int kt = 0; kt < ktiles; ++kt) {
+    __syncthreads();
+    if (kt + 1 < ktiles) stage_regs(rbeg + (kt + 1) * TBK);
+    const int buf = kt & 1;
+    if (do_bias) {
+      // colsum over the staged A tile: thread covers col pair
+      // (2*(tid%32), +1), k-rows tid/32*8 .. +8
+      const int bc = (tid & 31) * 2;
+      const int bq = tid >> 5;
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        bsum[0] += (float)As[buf][bc][bq * 8 + u];
+        bsum[1] += (float)As[buf][bc + 1][bq * 8 + u];
+      }
+    }
+#pragma unroll
+    for (int k32 = 0; k32 < TBK; k32 += 32) {
+      const bf16x8 a = *(const bf16x8*)&As[buf][wv * 16 + fi][k32 + fk * 8];
+#pragma unroll
+      for (int jt = 0; jt < 4; ++jt) {
+        const bf16x8 b = *(const bf16x8*)&Bs[buf][jt * 16 + fi][k32 + fk * 8];
+        acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[jt], 0, 0, 0);
+      }
+    }
+    if (kt + 1 < ktiles) {
+      __syncthreads();
+      regs_to_lds(1 - buf);
+    }
+  }
+
+  const bool direct = (gridDim.z == 1);
+  if (do_bias) {
+    __syncthreads();
+    // reduce 8 k-row-group partials per column via LDS (reuse As)
+    float* bred = (float*)&As[0][0][0];  // [8][64] fp32
+    const int bc = (tid & 31) * 2;
+    const int bq = tid >> 5;
+    bred[bq * 64 + bc] = bsum[0];
+    bred[bq * 64 + bc + 1] = bsum[1];
+    __syncthreads();
+    if (tid < TBM) {
+      float s = 0.0f;
+#pragma unroll
+      for (int q = 0; q < 8; ++q) s += bred[q * 64 + tid];
+      const int gm = m0 + tid;
+      if (gm < M) {
+        if (direct) {
+          if (accumulate) db[gm] += s; else db[gm] = s;
+        } else {
+          db_part[(long)blockIdx.z * M + gm] = s;
+        }
+      }
+    }
+  }
+
+  float* po = direct ? out : part + (long)blockIdx.z * M * N;
+#pragma unroll
+  for (int jt = 0; jt < 4; ++jt) {
+    const int gn = n0 + jt * 16 + fi;
+    if (gn >= N) continue;
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int gm = m0 + wv * 16 + fk * 4 + rr;
+      if (gm >= M) continue;
+      float* dst = &po[(long)gm * N + gn];
+      if (direct && accumulate)
+        *dst += acc[jt][rr];
+      else
+        *dst = acc[jt][rr];
+    }
+  }
+}
+
+// fixed-order partial reduce (same contract as fp32 tn_reduce_kernel)
+__global__ __launch_bounds__(256) void tn_reduce_bf16_kernel(
+    const float* __restrict__ part, float* __restrict__ out, long elems,
+    const float* __restrict__ db_part, float* __restrict__ db, long m_elems,
+    int z) {
+  const long e = (long)blockIdx.x * 256 + threadIdx.x;
+  if (e < elems) {
+    float s = 0.0f;
+    for (int c = 0; c < z; ++c) s += part[(long)c * elems + e];
+    out[e] += s;
+  } else if (e < elems + m_elems) {
+    const long m = e - elems;
+    float s = 0.0f;
+    for (int c = 0; c < z; ++c) s += db_part[(long)c * m_elems + m];
+    db[m] += s;
+  }
+}
+
+// elementwise casts / fused small ops for the bf16 path
+__global__ __launch_bounds__(256) void cast_f32_bf16_kernel(
+    const float* __restrict__ src, __bf16* __restrict__ dst, long n) {
+  const long i = (long)blockIdx.x * 256 + threadIdx.x;
+  if (i < n) dst[i] = (__bf16)src[i];
+}
+
+// dZ(bf16) = dY(bf16) * lrelu'(Y(bf16))
+__global__ __launch_bounds__(256) void lrelu_bwd_bf16_kernel(
+    const __bf16* __restrict__ dY, const __bf16* __restrict__ Y,
+    __bf16* __restrict__ dZ, long total) {
+  const long i = (long)blockIdx.x * 256 + threadIdx.x;
+  if (i < total) {
+    const float y = (float)Y[i];
+    dZ[i] = (__bf16)((float)dY[i] * (y > 0.0f ? 1.0f : 0.01f));
+  }
+}
+
+extern "C" {
+
+hipError_t fv_gemm_nt_bf16(const void* A, const void* W, const float* bias,
+                           float* out_f32, void* out_bf16, int R, int Ci,
+                           int Co, float alpha, int accumulate, int act_lrelu,
+                           hipStream_t stream) {
+  int flags = (accumulate ? 1 : 0) | (act_lrelu ? 2 : 0) | (bias ? 4 : 0);
+  dim3 grid((R + BBR - 1) / BBR, (Co + BBC - 1) / BBC);
+  hipLaunchKernelGGL(gemm_nt_bf16_kernel, grid, dim3(256), 0, stream,
+                     (const __bf16*)A, (const __bf16*)W, bias, out_f32,
+                     (__bf16*)out_bf16, R, Ci, Co, alpha, flags);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_gemm_nn_bf16(const void* A, const void* B, const float* bias,
+                           float* out_f32, void* out_bf16, int R, int Ci,
+                           int Co, float alpha, int accumulate, int act_lrelu,
+                           hipStream_t stream) {
+  int flags = (accumulate ? 1 : 0) | (act_lrelu ? 2 : 0) | (bias ? 4 : 0);
+  dim3 grid((R + BBR - 1) / BBR, (Co + BBC - 1) / BBC);
+  hipLaunchKernelGGL(gemm_nn_bf16_kernel, grid, dim3(256), 0, stream,
+                     (const __bf16*)A, (const __bf16*)B, bias, out_f32,
+                     (__bf16*)out_bf16, R, Ci, Co, alpha, flags);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_gemm_tn_bf16(const void* A, const void* B, float* out,
+                           float* part, float* db, float* db_part,
+                           int R, int M, int N, int r_chunks, int accumulate,
+                           hipStream_t stream) {
+  if (r_chunks < 1) r_chunks = 1;
+  if (!part) r_chunks = 1;
+  if (r_chunks > 1) {
+    r_chunks = (R + 2047) / 2048;
+    if (r_chunks > 32) r_chunks = 32;
+    if (r_chunks < 1) r_chunks = 1;
+  }
+  if (r_chunks > 1 && db && !db_part) r_chunks = 1;
+  dim3 grid((M + TBM - 1) / TBM, (N + TBN - 1) / TBN, r_chunks);
+  hipLaunchKernelGGL(gemm_tn_bf16_kernel, grid, dim3(256), 0, stream,
+                     (const __bf16*)A, (const __bf16*)B, out, part, db,
+                     db_part, R, M, N, accumulate);
+  HIP_CHECK_LAST();
+  if (r_chunks > 1) {
+    const long elems = (long)M * N;
+    const long m_elems = db ? M : 0;
+    dim3 rgrid((unsigned)((elems + m_elems + 255) / 256));
+    hipLaunchKernelGGL(tn_reduce_bf16_kernel, rgrid, dim3(256), 0, stream,
+                       part, out, elems, db_part, db, m_elems, r_chunks);
+    HIP_CHECK_LAST();
+  }
+  return hipSuccess;
+}
+
+hipError_t fv_cast_f32_bf16(const float* src, void* dst, long n,
+                            hipStream_t stream) {
+  dim3 grid((unsigned)((n + 255) / 256));
+  hipLaunchKernelGGL(cast_f32_bf16_kernel, grid, dim3(256), 0, stream,
+                     src, (__bf16*)dst, n);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_lrelu_bwd_bf16(const void* dY, const void* Y, void* dZ,
+                             long total, hipStream_t stream) {
+  dim3 grid((unsigned)((total + 255) / 256));
+  hipLaunchKernelGGL(lrelu_bwd_bf16_kernel, grid, dim3(256), 0, stream,
+                     (const __bf16*)dY, (const __bf16*)Y, (__bf16*)dZ, total);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+}  // extern "C"

@@ -1,7 +1,11 @@
 """Dispatch-floor microbenchmark: what does one kernel cost in this
 environment, eager vs graph, serial vs parallel streams?"""
 
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
@@ -121,5 +125,61 @@ def main():
     print(f"gemm_nt 2-stream: {t / 50 * 1e6:.2f} us/call-equivalent")
 
 
+def gemm_bench():
+    """Isolated timings of the hot GEMM shapes (no contention)."""
+    shapes_nt = [(6000, 158, 192), (6000, 158, 158), (210000, 158, 192)]
+    for R, Ci, Co in shapes_nt:
+        A = torch.randn(R, Ci, device=dev)
+        W = torch.randn(Co, Ci, device=dev)
+        b = torch.randn(Co, device=dev)
+        out = torch.empty(R, Co, device=dev)
+        t = timeit(lambda: ext.gemm_nt(A, W, b, out, 1.0, False, False), 20)
+        fl = 2.0 * R * Ci * Co
+        print(f"gemm_nt {R}x{Ci}x{Co}: {t*1e6:.1f} us  {fl/t/1e12:.1f} TF/s")
+    for R, Ci, Co in [(6000, 192, 158), (210000, 192, 158)]:
+        A = torch.randn(R, Ci, device=dev)
+        B = torch.randn(Ci, Co, device=dev)
+        out = torch.empty(R, Co, device=dev)
+        t = timeit(lambda: ext.gemm_nn(A, B, None, out, 1.0, False, False), 20)
+        fl = 2.0 * R * Ci * Co
+        print(f"gemm_nn {R}x{Ci}x{Co}: {t*1e6:.1f} us  {fl/t/1e12:.1f} TF/s")
+    for R, Ci, Co in [(6000, 158, 192), (210000, 158, 192)]:
+        A = torch.randn(R, Ci, device=dev).bfloat16()
+        W = torch.randn(Co, Ci, device=dev).bfloat16()
+        b = torch.randn(Co, device=dev)
+        out = torch.empty(R, Co, device=dev)
+        t = timeit(lambda: ext.gemm_nt_bf16(A, W, b, out, None, 1.0, False, False), 20)
+        fl = 2.0 * R * Ci * Co
+        print(f"gemm_nt_bf16 {R}x{Ci}x{Co}: {t*1e6:.1f} us  {fl/t/1e12:.1f} TF/s")
+    for R, Ci, Co in [(6000, 192, 158), (210000, 192, 158)]:
+        A = torch.randn(R, Ci, device=dev).bfloat16()
+        B = torch.randn(Ci, Co, device=dev).bfloat16()
+        out = torch.empty(R, Co, device=dev, dtype=torch.bfloat16)
+        t = timeit(lambda: ext.gemm_nn_bf16(A, B, None, None, out, 1.0, False, False), 20)
+        fl = 2.0 * R * Ci * Co
+        print(f"gemm_nn_bf16 {R}x{Ci}x{Co}: {t*1e6:.1f} us  {fl/t/1e12:.1f} TF/s")
+    for R, M, N, chunks in [(6000, 192, 158, 8), (210000, 192, 158, 8)]:
+        A = torch.randn(R, M, device=dev).bfloat16()
+        B = torch.randn(R, N, device=dev).bfloat16()
+        out = torch.zeros(M, N, device=dev)
+        db = torch.zeros(M, device=dev)
+        part = torch.zeros(32 * M * N, device=dev)
+        db_part = torch.zeros(32 * M, device=dev)
+        t = timeit(lambda: ext.gemm_tn_bf16(A, B, out, part, chunks, True, db, db_part), 20)
+        fl = 2.0 * R * M * N
+        print(f"gemm_tn_bf16 {R}x{M}x{N} z{chunks}+bias: {t*1e6:.1f} us  {fl/t/1e12:.1f} TF/s")
+    for R, M, N, chunks in [(6000, 192, 158, 8), (210000, 192, 158, 8)]:
+        A = torch.randn(R, M, device=dev)
+        B = torch.randn(R, N, device=dev)
+        out = torch.zeros(M, N, device=dev)
+        db = torch.zeros(M, device=dev)
+        part = torch.zeros(32 * M * N, device=dev)
+        db_part = torch.zeros(32 * M, device=dev)
+        t = timeit(lambda: ext.gemm_tn(A, B, out, part, chunks, True, db, db_part), 20)
+        fl = 2.0 * R * M * N
+        print(f"gemm_tn {R}x{M}x{N} z{chunks}+bias: {t*1e6:.1f} us  {fl/t/1e12:.1f} TF/s")
+
+
 if __name__ == "__main__":
     main()
+    gemm_bench()

@@ -59,6 +59,20 @@ def test_gemm_tn(R, M, N, chunks):
     assert_close(out, A.t() @ B, atol=5e-4, rtol=5e-4, what="gemm_tn")
 
 
+@pytest.mark.parametrize("R,M,N,chunks", [(300, 20, 64, 1), (6000, 192, 64, 8), (77, 33, 17, 3)])
+def test_gemm_tn_fused_bias(R, M, N, chunks):
+    """db epilogue: bias grad colsum fused into the wgrad GEMM."""
+    A, B = t(R, M, seed=16), t(R, N, seed=17)
+    out = torch.zeros(M, N, device=DEV)
+    db = torch.zeros(M, device=DEV)
+    part = t(32 * M * N, seed=99) if chunks > 1 else None
+    db_part = t(32 * M, seed=98) if chunks > 1 else None
+    ext.gemm_tn(A, B, out, part, chunks, chunks > 1, db, db_part)
+    torch.cuda.synchronize()
+    assert_close(out, A.t() @ B, atol=5e-4, rtol=5e-4, what="gemm_tn+db W")
+    assert_close(db, A.sum(dim=0), atol=5e-4, rtol=5e-4, what="gemm_tn+db b")
+
+
 def test_colsum():
     A = t(1234, 77, seed=8)
     out = torch.zeros(77, device=DEV)
@@ -280,6 +294,8 @@ def eager_forward_explicit_noise(model, x, y, eps, mask, training):
     (300, 20, 158, 64, 128, 20, True),
     (300, 20, 158, 64, 128, 20, False),
     (37, 7, 33, 48, 24, 12, True),
+    # full A-share shape (BASELINE.json config 4): N=3500, T=60, K=96
+    (3500, 60, 158, 64, 128, 96, True),
 ])
 def test_full_step_grad_parity(N, T, C, H, M, K, training):
     from factorvae_amd.engine.fused import FusedTrainer
@@ -440,3 +456,139 @@ def test_fused_adam_matches_torch_adam():
     for k in fused_sd:
         torch.testing.assert_close(fused_sd[k], eager_sd[k], atol=4e-3,
                                    rtol=5e-2, msg=lambda m, kk=k: f"{kk}: {m}")
+
+
+# ------------------------------------------------------------------ bf16 GEMMs
+def bt(*shape, seed=0, scale=1.0):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    return (torch.randn(*shape, generator=g) * scale).to(DEV).to(torch.bfloat16)
+
+
+@pytest.mark.parametrize("R,Ci,Co", [(300, 158, 192), (37, 33, 20), (6000, 158, 158), (210000, 158, 192)])
+def test_gemm_nt_bf16(R, Ci, Co):
+    A, W = bt(R, Ci, seed=1), bt(Co, Ci, seed=2)
+    b = t(Co, seed=3)
+    out = torch.empty(R, Co, device=DEV)
+    outb = torch.empty(R, Co, device=DEV, dtype=torch.bfloat16)
+    ext.gemm_nt_bf16(A, W, b, out, outb, 1.0, False, False)
+    torch.cuda.synchronize()
+    ref = A.float() @ W.float().t() + b
+    tol = 3e-2 * math.sqrt(Ci / 64)
+    assert_close(out, ref, atol=tol, rtol=tol, what="gemm_nt_bf16 f32out")
+    assert_close(outb.float(), ref, atol=4e-1, rtol=2e-2, what="gemm_nt_bf16 bf16out")
+    # lrelu epilogue
+    out2 = torch.empty(R, Co, device=DEV)
+    ext.gemm_nt_bf16(A, W, b, out2, None, 0.5, False, True)
+    torch.cuda.synchronize()
+    ref2 = F.leaky_relu(0.5 * ref, 0.01)
+    assert_close(out2, ref2, atol=tol, rtol=tol, what="gemm_nt_bf16 lrelu")
+
+
+@pytest.mark.parametrize("R,Ci,Co", [(300, 192, 158), (37, 33, 20), (6000, 192, 158)])
+def test_gemm_nn_bf16(R, Ci, Co):
+    A, B = bt(R, Ci, seed=4), bt(Ci, Co, seed=5)
+    out = torch.empty(R, Co, device=DEV)
+    ext.gemm_nn_bf16(A, B, None, out, None, 1.0, False, False)
+    torch.cuda.synchronize()
+    ref = A.float() @ B.float()
+    tol = 3e-2 * math.sqrt(Ci / 64)
+    assert_close(out, ref, atol=tol, rtol=tol, what="gemm_nn_bf16")
+
+
+@pytest.mark.parametrize("R,M,N,chunks", [(300, 20, 64, 1), (6000, 192, 64, 8), (77, 33, 17, 3), (210000, 192, 158, 8)])
+def test_gemm_tn_bf16(R, M, N, chunks):
+    A, B = bt(R, M, seed=6, scale=0.1), bt(R, N, seed=7, scale=0.1)
+    out = torch.zeros(M, N, device=DEV)
+    db = torch.zeros(M, device=DEV)
+    part = torch.zeros(32 * M * N, device=DEV) if chunks > 1 else None
+    db_part = torch.zeros(32 * M, device=DEV) if chunks > 1 else None
+    ext.gemm_tn_bf16(A, B, out, part, chunks, chunks > 1, db, db_part)
+    torch.cuda.synchronize()
+    ref = A.float().t() @ B.float()
+    tol = 2e-3 * math.sqrt(R)
+    assert_close(out, ref, atol=tol, rtol=2e-2, what="gemm_tn_bf16")
+    assert_close(db, A.float().sum(dim=0), atol=tol, rtol=2e-2,
+                 what="gemm_tn_bf16 bias")
+
+
+def test_cast_and_lrelu_bwd_bf16():
+    x = t(1234, 77, seed=8)
+    xb = torch.empty_like(x, dtype=torch.bfloat16)
+    ext.cast_f32_bf16(x, xb)
+    torch.cuda.synchronize()
+    assert torch.equal(xb, x.to(torch.bfloat16))
+    dY, Y = bt(500, 30, seed=9), bt(500, 30, seed=10)
+    dZ = torch.empty_like(dY)
+    ext.lrelu_bwd_bf16(dY, Y, dZ)
+    torch.cuda.synchronize()
+    ref = (dY.float() * torch.where(Y.float() > 0, 1.0, 0.01)).to(torch.bfloat16)
+    assert torch.equal(dZ, ref)
+
+
+@pytest.mark.parametrize("N,T", [(300, 20), (3500, 60)])
+def test_bf16_full_step_vs_fp32(N, T):
+    """bf16 engine mode: loss matches the fp32 fused engine closely and
+    gradients are directionally identical (cosine > 0.995)."""
+    from factorvae_amd.engine.fused import FusedTrainer
+    from factorvae_amd.models.modules import build_factorvae
+    from factorvae_amd.utils import set_seed
+
+    C, H, M, K = 158, 64, 128, 20
+    losses, grads = {}, {}
+    for dtype in ("fp32", "bf16"):
+        set_seed(0)
+        model = build_factorvae(num_latent=C, hidden_size=H, num_portfolio=M,
+                                num_factor=K).to(DEV)
+        tr = FusedTrainer(model, lr=1e-4, t_max=100, device=DEV,
+                          use_graph=False, train=True, dtype=dtype)
+        x = t(N, T, C, seed=50)
+        y = t(N, 1, seed=51)
+        tr._ensure_ws(N, T)
+        w = tr.ws
+        w["x"].copy_(x)
+        w["y"].copy_(y)
+        set_seed(1)
+        tr._fill_rng(N)
+        tr.grads.zero_()
+        tr._launch_forward(N, T)
+        tr._launch_backward(N, T)
+        torch.cuda.synchronize()
+        losses[dtype] = float(w["loss"][0])
+        grads[dtype] = tr.grads.clone()
+
+    assert abs(losses["bf16"] - losses["fp32"]) < 0.02 * (abs(losses["fp32"]) + 1.0)
+    g32, g16 = grads["fp32"], grads["bf16"]
+    cos = torch.nn.functional.cosine_similarity(g32, g16, dim=0)
+    assert float(cos) > 0.995, f"grad cosine {float(cos)}"
+
+
+@pytest.mark.gpu
+def test_bf16_training_reduces_loss():
+    """Short bf16 training run: loss decreases (graph path)."""
+    from factorvae_amd.engine.fused import FusedTrainer
+    from factorvae_amd.models.modules import build_factorvae
+    from factorvae_amd.utils import set_seed
+
+    set_seed(3)
+    N, T, C, H, M, K = 300, 20, 158, 64, 128, 20
+    model = build_factorvae(num_latent=C, hidden_size=H, num_portfolio=M,
+                            num_factor=K).to(DEV)
+    tr = FusedTrainer(model, lr=1e-3, t_max=200, device=DEV, dtype="bf16")
+    g = torch.Generator(device=DEV).manual_seed(7)
+    beta = torch.randn(C, 1, device=DEV, generator=g) * 0.1
+    days = []
+    for _ in range(8):
+        x = torch.randn(N, T, C, device=DEV, generator=g)
+        y = x[:, -1, :] @ beta + 0.1 * torch.randn(N, 1, device=DEV, generator=g)
+        days.append((x, y))
+    first = last = None
+    for ep in range(20):
+        tot = 0.0
+        for x, y in days:
+            loss = tr.step(x, y)
+            tot += float(loss[0])
+        if ep == 0:
+            first = tot
+        last = tot
+    assert last < first, (first, last)
+    assert last == last  # not NaN
