@@ -1,0 +1,212 @@
+// fp8 (e4m3) forward path for the FactorVAE engine (gfx950) —
+// BASELINE.json config 5: "fp8 weights+activations (CDNA4 fp8 MFMA)".
+//
+// Design: the FLOP-bound extractor forward GEMMs run on fp8 MFMA
+// (v_mfma_f32_16x16x32_fp8_fp8, fp32 accumulate) with e4m3 weights and
+// activations; per-tensor weight scales (absmax -> 448 full-scale) are
+// applied in the fp32 epilogue; activations are LN-normalized so they
+// use unit scale. Backward runs on the bf16 kernels (the fp8 GEMMs also
+// emit the bf16 activation copies backward needs). fp8 rows are padded
+// to a multiple of 4 bytes so staging can use dword-coalesced loads.
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define FBR 64   // row tile
+#define FBC 64   // col tile
+#define FBK 64   // k tile (2 MFMA k-steps of 32)
+
+union dw_f8 {
+  unsigned int u;
+  unsigned char b[4];
+};
+
+DEVINL unsigned char f32_to_e4m3(float x) {
+  unsigned int v = 0;
+  v = __builtin_amdgcn_cvt_pk_fp8_f32(x, 0.0f, v, false);
+  return (unsigned char)(v & 0xff);
+}
+
+DEVINL float e4m3_to_f32(unsigned char b) {
+  return __builtin_amdgcn_cvt_f32_fp8((unsigned int)b, 0);
+}
+
+// bounds-checked dword (4 fp8) load from a row-major padded matrix
+DEVINL unsigned int load_dw_f8(const unsigned char* p, long row, int col0,
+                               long nrows, int ncols, int ld) {
+  if (row >= nrows) return 0u;
+  const unsigned char* rp = p + row * (long)ld;
+  if (col0 + 4 <= ncols) return *(const unsigned int*)(rp + col0);
+  dw_f8 d;
+  d.u = 0u;
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+    d.b[i] = (col0 + i < ncols) ? rp[col0 + i] : 0;
+  return d.u;
+}
+
+// ---------------------------------------------------------------- NT
+// out = act(alpha * sa*sb * (A(R,Ci)@W(Co,Ci)^T) + bias), A/W fp8 e4m3
+// with padded leading dims lda/ldw (bytes). Outputs: fp32 and/or bf16
+// and/or fp8 (fp8 out uses out scale so8, padded ldo).
+__global__ __launch_bounds__(256) void gemm_nt_fp8_kernel(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ W,
+    const float* __restrict__ bias, const float* __restrict__ inv_sw,
+    float* __restrict__ out_f32, __bf16* __restrict__ out_bf16,
+    unsigned char* __restrict__ out_fp8, int ldo,
+    int R, int Ci, int Co, int lda, int ldw, float alpha, int flags) {
+  __shared__ unsigned char As[2][FBR][FBK + 16];
+  __shared__ unsigned char Ws[2][FBC][FBK + 16];
+
+  const int r0 = blockIdx.x * FBR;
+  const int c0 = blockIdx.y * FBC;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const int fi = lane & 15;
+  const int fk = lane >> 4;       // k-group: 8 consecutive k each
+
+  const int KD = FBK / 4;         // dwords per tile row (16)
+  f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+
+  const int ktiles = (Ci + FBK - 1) / FBK;
+  unsigned int pa[4], pw[4];
+
+  auto stage_regs = [&](int k0) {
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int idx = tid + u * 256;
+      const int row = idx / KD;
+      const int cp = (idx % KD) * 4;
+      pa[u] = load_dw_f8(A, (long)r0 + row, k0 + cp, R, Ci, lda);
+      pw[u] = load_dw_f8(W, (long)c0 + row, k0 + cp, Co, Ci, ldw);
+    }
+  };
+  auto regs_to_lds = [&](int buf) {
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int idx = tid + u * 256;
+      const int row = idx / KD;
+      const int cp = (idx % KD) * 4;
+      *(unsigned int*)&As[buf][row][cp] = pa[u];
+      *(unsigned int*)&Ws[buf][row][cp] = pw[u];
+    }
+  };
+
+  stage_regs(0);
+  regs_to_lds(0);
+
+  for (int kt = 0; kt < ktiles; ++kt) {
+    __syncthreads();
+    if (kt + 1 < ktiles) stage_regs((kt + 1) * FBK);
+    const int buf = kt & 1;
+#pragma unroll
+    for (int k32 = 0; k32 < FBK; k32 += 32) {
+      const long a = *(const long*)&As[buf][wv * 16 + fi][k32 + fk * 8];
+#pragma unroll
+      for (int jt = 0; jt < 4; ++jt) {
+        const long b = *(const long*)&Ws[buf][jt * 16 + fi][k32 + fk * 8];
+        acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a, b, acc[jt],
+                                                             0, 0, 0);
+      }
+    }
+    if (kt + 1 < ktiles) {
+      __syncthreads();
+      regs_to_lds(1 - buf);
+    }
+  }
+
+  const float sw = inv_sw ? *inv_sw : 1.0f;  // undo the weight scale
+#pragma unroll
+  for (int jt = 0; jt < 4; ++jt) {
+    const int gc = c0 + jt * 16 + fi;
+    if (gc >= Co) continue;
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int gr = r0 + wv * 16 + fk * 4 + rr;
+      if (gr >= R) continue;
+      float v = acc[jt][rr] * sw;
+      if (flags & 4) v += bias[gc];
+      v *= alpha;
+      if (flags & 2) v = lrelu_(v);
+      if (out_f32) out_f32[(long)gr * Co + gc] = v;
+      if (out_bf16) out_bf16[(long)gr * Co + gc] = (__bf16)v;
+      if (out_fp8) out_fp8[(long)gr * ldo + gc] = f32_to_e4m3(v);
+    }
+  }
+}
+
+// scale[0] = 448 / max(absmax(src), 1e-8): per-tensor e4m3 full-scale.
+// Single workgroup (params are <=128k elements); deterministic.
+__global__ __launch_bounds__(1024) void absmax_scale_kernel(
+    const float* __restrict__ src, long n, float* __restrict__ scale,
+    float* __restrict__ inv_scale) {
+  __shared__ float red[16];
+  float m = 0.0f;
+  for (long i = threadIdx.x; i < n; i += 1024)
+    m = fmaxf(m, fabsf(src[i]));
+  m = wave_reduce_max(m);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = m;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float mm = 0.0f;
+#pragma unroll
+    for (int i = 0; i < 16; ++i) mm = fmaxf(mm, red[i]);
+    mm = fmaxf(mm, 1e-8f);
+    scale[0] = 448.0f / mm;
+    inv_scale[0] = mm / 448.0f;
+  }
+}
+
+// dst(rows, ldp bytes) = e4m3(src(rows, cols) * scale[0])
+__global__ __launch_bounds__(256) void cast_f32_fp8_scaled_kernel(
+    const float* __restrict__ src, unsigned char* __restrict__ dst,
+    const float* __restrict__ scale, long rows, int cols, int ldp) {
+  const long i = (long)blockIdx.x * 256 + threadIdx.x;
+  const long r = i / cols;
+  const int c = (int)(i % cols);
+  if (r < rows) {
+    const float s = scale ? scale[0] : 1.0f;
+    dst[r * (long)ldp + c] = f32_to_e4m3(src[r * (long)cols + c] * s);
+  }
+}
+
+extern "C" {
+
+hipError_t fv_gemm_nt_fp8(const void* A, const void* W, const float* bias,
+                          const float* inv_sw, float* out_f32, void* out_bf16,
+                          void* out_fp8, int ldo, int R, int Ci, int Co,
+                          int lda, int ldw, float alpha, int act_lrelu,
+                          int has_bias, hipStream_t stream) {
+  int flags = (act_lrelu ? 2 : 0) | (has_bias ? 4 : 0);
+  dim3 grid((R + FBR - 1) / FBR, (Co + FBC - 1) / FBC);
+  hipLaunchKernelGGL(gemm_nt_fp8_kernel, grid, dim3(256), 0, stream,
+                     (const unsigned char*)A, (const unsigned char*)W, bias,
+                     inv_sw, out_f32, (__bf16*)out_bf16,
+                     (unsigned char*)out_fp8, ldo, R, Ci, Co, lda, ldw,
+                     alpha, flags);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_absmax_scale(const float* src, long n, float* scale,
+                           float* inv_scale, hipStream_t stream) {
+  hipLaunchKernelGGL(absmax_scale_kernel, dim3(1), dim3(1024), 0, stream,
+                     src, n, scale, inv_scale);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_cast_f32_fp8_scaled(const float* src, void* dst,
+                                  const float* scale, long rows, int cols,
+                                  int ldp, hipStream_t stream) {
+  const long total = rows * cols;
+  dim3 grid((unsigned)((total + 255) / 256));
+  hipLaunchKernelGGL(cast_f32_fp8_scaled_kernel, grid, dim3(256), 0, stream,
+                     src, (unsigned char*)dst, scale, rows, cols, ldp);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+}  // extern "C"

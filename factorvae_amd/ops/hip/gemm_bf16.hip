@@ -80,13 +80,26 @@ __global__ __launch_bounds__(256) void gemm_nt_bf16_kernel(
   unsigned int pa[8], pw[8];
 
   auto stage_regs = [&](int k0) {
+    const bool interior = (r0 + BBR <= R) && (c0 + BBC <= Co) &&
+                          (k0 + BBK <= Ci);
+    if (interior) {
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      const int idx = tid + u * 256;
-      const int row = idx / KD;          // 0..63
-      const int cp = (idx % KD) * 2;     // even col within tile
-      pa[u] = load_dw_guard(A, (long)r0 + row, k0 + cp, R, Ci, Ci);
-      pw[u] = load_dw_guard(W, (long)c0 + row, k0 + cp, Co, Ci, Ci);
+      for (int u = 0; u < 8; ++u) {
+        const int idx = tid + u * 256;
+        const int row = idx / KD;
+        const int cp = (idx % KD) * 2;
+        pa[u] = *(const unsigned int*)(A + ((long)r0 + row) * Ci + k0 + cp);
+        pw[u] = *(const unsigned int*)(W + ((long)c0 + row) * Ci + k0 + cp);
+      }
+    } else {
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const int idx = tid + u * 256;
+        const int row = idx / KD;          // 0..63
+        const int cp = (idx % KD) * 2;     // even col within tile
+        pa[u] = load_dw_guard(A, (long)r0 + row, k0 + cp, R, Ci, Ci);
+        pw[u] = load_dw_guard(W, (long)c0 + row, k0 + cp, Co, Ci, Ci);
+      }
     }
   };
   auto regs_to_lds = [&](int buf) {
@@ -171,14 +184,26 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
   unsigned int pa[8], pb[8];
 
   auto stage_regs = [&](int k0) {
+    const bool interior = (r0 + BBR <= R) && (c0 + BBC <= Co) &&
+                          (k0 + BBK <= Ci);
+    if (interior) {
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      const int idx = tid + u * 256;
-      const int row = idx / KD;
-      const int cp = (idx % KD) * 2;
-      pa[u] = load_dw_guard(A, (long)r0 + row, k0 + cp, R, Ci, Ci);
-      // B tile: 64 k-rows x 64 cols; dword idx: krow = idx/32, colpair
-      pb[u] = load_dw_guard(B, (long)k0 + row, c0 + cp, Ci, Co, Co);
+      for (int u = 0; u < 8; ++u) {
+        const int idx = tid + u * 256;
+        const int row = idx / KD;
+        const int cp = (idx % KD) * 2;
+        pa[u] = *(const unsigned int*)(A + ((long)r0 + row) * Ci + k0 + cp);
+        pb[u] = *(const unsigned int*)(B + ((long)k0 + row) * Co + c0 + cp);
+      }
+    } else {
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const int idx = tid + u * 256;
+        const int row = idx / KD;
+        const int cp = (idx % KD) * 2;
+        pa[u] = load_dw_guard(A, (long)r0 + row, k0 + cp, R, Ci, Ci);
+        pb[u] = load_dw_guard(B, (long)k0 + row, c0 + cp, Ci, Co, Co);
+      }
     }
   };
   auto regs_to_lds = [&](int buf) {
@@ -281,15 +306,29 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
 
   // staging: tile is 64 k-rows x 64 cols = 64x32 dwords
   auto stage_regs = [&](int r0_) {
+    const bool interior = (r0_ + TBK <= rend) && (m0 + TBM <= M) &&
+                          (n0 + TBN <= N);
+    if (interior) {
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      const int idx = tid + u * 256;
-      const int krow = idx / MD;
-      const int cp = (idx % MD) * 2;
-      const long gr = (long)r0_ + krow;
-      const long rows = rend;  // guard against rend, not R
-      pa[u] = load_dw_guard(A, gr, m0 + cp, rows, M, M);
-      pb[u] = load_dw_guard(B, gr, n0 + cp, rows, N, N);
+      for (int u = 0; u < 8; ++u) {
+        const int idx = tid + u * 256;
+        const int krow = idx / MD;
+        const int cp = (idx % MD) * 2;
+        const long gr = (long)r0_ + krow;
+        pa[u] = *(const unsigned int*)(A + gr * M + m0 + cp);
+        pb[u] = *(const unsigned int*)(B + gr * N + n0 + cp);
+      }
+    } else {
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const int idx = tid + u * 256;
+        const int krow = idx / MD;
+        const int cp = (idx % MD) * 2;
+        const long gr = (long)r0_ + krow;
+        const long rows = rend;  // guard against rend, not R
+        pa[u] = load_dw_guard(A, gr, m0 + cp, rows, M, M);
+        pb[u] = load_dw_guard(B, gr, n0 + cp, rows, N, N);
+      }
     }
   };
   auto regs_to_lds = [&](int buf) {
@@ -461,7 +500,7 @@ hipError_t fv_gemm_tn_bf16(const void* A, const void* B, float* out,
   if (r_chunks < 1) r_chunks = 1;
   if (!part) r_chunks = 1;
   if (r_chunks > 1) {
-    r_chunks = (R + 2047) / 2048;
+    r_chunks = (R + 511) / 512;
     if (r_chunks > 32) r_chunks = 32;
     if (r_chunks < 1) r_chunks = 1;
   }
