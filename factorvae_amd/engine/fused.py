@@ -289,6 +289,12 @@ class FusedTrainer:
         w["tn_partb"] = f(32 * max_m)
         w["tn_partb2"] = f(32 * max_m)
         w["tn_partb3"] = f(32 * max_m)
+        # N-row reductions (attention/encoder/decoder wgrads + fwd u):
+        # chunked over z so a 3500-stock day doesn't serialize 6 blocks
+        small_mn = 32 * max(M, K, H) * H
+        w["tn_part_u"] = f(small_mn)
+        w["tn_part_s"] = f(small_mn)
+        w["tn_partb_s"] = f(32 * max_m)
         w["dzx"] = f(R, C)
         w["dxln"] = f(R, C)
         if self.bf16:
@@ -299,6 +305,8 @@ class FusedTrainer:
             w["dgi_bf"] = fb(R, 3 * H)
             w["dxp_bf"] = fb(R, C)
             w["dzx_bf"] = fb(R, C)
+            w["h_prev_bf"] = fb(R, H)
+            w["dgh_bf"] = fb(R, 3 * H)
         self._ws_cache[(N, T)] = w
         self.ws = w
         self._ws_n = N
@@ -328,6 +336,15 @@ class FusedTrainer:
             ext.gemm_nt(w["xp"], p("Wih"), p("bih"), w["gi"], 1.0, False, False)
         ext.gru_fwd(w["gi"], p("Whh"), p("bhh"), w["h"], w["h_seq"],
                     w["h_prev"], w["gates4"], N, T, H)
+        if self.bf16 and self.s_side is not None:
+            e_ = torch.cuda.Event()
+            e_.record(torch.cuda.current_stream(self.device))
+            self.s_side.wait_event(e_)
+            with torch.cuda.stream(self.s_side):
+                ext.cast_f32_bf16(w["h_prev"].view(-1),
+                                  w["h_prev_bf"].view(-1))
+        elif self.bf16:
+            ext.cast_f32_bf16(w["h_prev"].view(-1), w["h_prev_bf"].view(-1))
         ext.gemm_nt(w["h"], p("Wenc"), p("benc"), w["scores_enc"], 1.0, False, False)
         ext.enc_softmax_fwd(w["scores_enc"], yv, w["a_enc"], w["yp"])
         ext.enc_heads_fwd(w["yp"], p("Wmu_e"), p("bmu_e"), p("Wsig_e"),
@@ -339,7 +356,7 @@ class FusedTrainer:
         keep_inv = 1.0 / (1.0 - self.DROPOUT_P)
         ext.attn_softmax_fwd(w["s_att"], mask, w["a_att"], w["sd"],
                              w["guard"], keep_inv)
-        ext.gemm_tn(w["a_att"], w["h"], w["u"], None, 1, False)
+        ext.gemm_tn(w["a_att"], w["h"], w["u"], w["tn_part_u"], 2, False)
         ext.attn_ctx_fwd(w["u"], self.p_Wv, self.p_bv, w["guard"], w["ctx"])
         ext.pred_mlp_fwd(w["ctx"], p("Wl"), p("bl"), p("wmu_p"), p("bmu_p"),
                          p("wsig_p"), p("bsig_p"), w["hm2"], w["pmu"],
@@ -399,8 +416,10 @@ class FusedTrainer:
                     g("bmu_d"), g("wsig_d"), g("bsig_d"))
         fork()
         with _on_side():
-            ext.gemm_tn(w["dz1"], w["h"], g("W1d"), None, 1, True, g("b1d"))
-            ext.gemm_tn(w["dbeta"], w["h"], g("Wb"), None, 1, True, g("bb"))
+            ext.gemm_tn(w["dz1"], w["h"], g("W1d"), w["tn_part_s"], 2, True,
+                        g("b1d"), w["tn_partb_s"])
+            ext.gemm_tn(w["dbeta"], w["h"], g("Wb"), w["tn_part_s"], 2, True,
+                        g("bb"), w["tn_partb_s"])
 
         # predictor MLP + attention backward
         ext.pred_mlp_bwd(w["dpmu"], w["dpsig_c"], w["psig"], w["psig_pre"],
@@ -420,7 +439,7 @@ class FusedTrainer:
                              w["ds"], w["dc"], keep_inv, alpha)
         fork()
         with _on_side():
-            ext.gemm_tn(w["ds"], w["h"], w["dqk"], None, 1, False)
+            ext.gemm_tn(w["ds"], w["h"], w["dqk"], w["tn_part_s"], 2, False)
             gq = self._gstack("q_att.0", (K, H))
             gWk = self._gstack("Wk.0", (K, H, H))
             gbk = self._gstack("bk.0", (K, H))
@@ -436,7 +455,8 @@ class FusedTrainer:
         ext.enc_softmax_bwd(w["dyp"], w["a_enc"], yv, w["dscores"])
         fork()
         with _on_side():
-            ext.gemm_tn(w["dscores"], w["h"], g("Wenc"), None, 1, True, g("benc"))
+            ext.gemm_tn(w["dscores"], w["h"], g("Wenc"), w["tn_part_s"], 2,
+                        True, g("benc"), w["tn_partb_s"])
         ext.gemm_nn(w["dscores"], p("Wenc"), None, w["dh"], 1.0, True, False)
 
         # extractor backward
@@ -446,9 +466,10 @@ class FusedTrainer:
             ext.cast_f32_bf16(w["dgi"].view(-1), w["dgi_bf"].view(-1))
             fork()
             with _on_side():
-                ext.gemm_tn(w["dgh"].view(R, 3 * H), w["h_prev"].view(R, H),
-                            g("Whh"), w["tn_part"], chunks, True,
-                            g("bhh"), w["tn_partb"])
+                ext.cast_f32_bf16(w["dgh"].view(-1), w["dgh_bf"].view(-1))
+                ext.gemm_tn_bf16(w["dgh_bf"].view(R, 3 * H), w["h_prev_bf"],
+                                 g("Whh"), w["tn_part"], chunks, True,
+                                 g("bhh"), w["tn_partb"])
                 ext.gemm_tn_bf16(w["dgi_bf"].view(R, 3 * H), w["xp_bf"],
                                  g("Wih"), w["tn_part2"], chunks, True,
                                  g("bih"), w["tn_partb2"])

@@ -28,7 +28,14 @@ hipError_t fv_lrelu_bwd_bf16(const void*, const void*, void*, long,
                              hipStream_t);
 hipError_t fv_lrelu_bwd(const float*, const float*, float*, long, hipStream_t);
 hipError_t fv_ln_fwd(const float*, const float*, const float*, float*, void*,
-                     float*, float*, long, int, float, hipStream_t);
+                     void*, int, float*, float*, long, int, float,
+                     hipStream_t);
+hipError_t fv_gemm_nt_fp8(const void*, const void*, const float*,
+                          const float*, float*, void*, void*, int, int, int,
+                          int, int, int, float, int, int, hipStream_t);
+hipError_t fv_absmax_scale(const float*, long, float*, float*, hipStream_t);
+hipError_t fv_cast_f32_fp8_scaled(const float*, void*, const float*, long,
+                                  int, int, hipStream_t);
 hipError_t fv_ln_bwd_params(const float*, const float*, const float*,
                             const float*, float*, float*, long, int, int,
                             hipStream_t);
@@ -283,7 +290,8 @@ void lrelu_bwd(torch::Tensor dY, torch::Tensor Y, torch::Tensor dZ) {
 void ln_fwd(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
             c10::optional<torch::Tensor> xln,
             torch::Tensor mean, torch::Tensor rstd,
-            double eps, c10::optional<torch::Tensor> xln_bf = c10::nullopt) {
+            double eps, c10::optional<torch::Tensor> xln_bf = c10::nullopt,
+            c10::optional<torch::Tensor> xln_f8 = c10::nullopt) {
   CK(x); CK(gamma); CK(beta); CK(mean); CK(rstd);
   const long R = x.numel() / x.size(-1);
   const int C = x.size(-1);
@@ -291,9 +299,73 @@ void ln_fwd(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
   if (xln.has_value()) { CK(*xln); xo = fpm(*xln); }
   void* xb = nullptr;
   if (xln_bf.has_value()) { CKB(*xln_bf); xb = bfp(*xln_bf); }
-  TORCH_CHECK(xo || xb, "ln_fwd needs an output");
-  RUN(fv_ln_fwd(fp(x), fp(gamma), fp(beta), xo, xb, fpm(mean), fpm(rstd), R,
-                C, (float)eps, cur_stream()));
+  void* x8 = nullptr;
+  int f8_ld = 0;
+  if (xln_f8.has_value()) {
+    TORCH_CHECK(xln_f8->is_cuda() && xln_f8->is_contiguous());
+    TORCH_CHECK(xln_f8->scalar_type() == torch::kFloat8_e4m3fn);
+    x8 = xln_f8->data_ptr();
+    f8_ld = xln_f8->size(-1);
+  }
+  TORCH_CHECK(xo || xb || x8, "ln_fwd needs an output");
+  RUN(fv_ln_fwd(fp(x), fp(gamma), fp(beta), xo, xb, x8, f8_ld, fpm(mean),
+                fpm(rstd), R, C, (float)eps, cur_stream()));
+}
+
+inline void check_fp8(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat8_e4m3fn, name,
+              " must be float8_e4m3fn");
+}
+#define CK8(t) check_fp8(t, #t)
+
+// A/W fp8 (R, lda)/(Co, ldw) padded; logical K = Ci. Outputs optional.
+void gemm_nt_fp8(torch::Tensor A, torch::Tensor W,
+                 c10::optional<torch::Tensor> bias,
+                 c10::optional<torch::Tensor> inv_sw,
+                 c10::optional<torch::Tensor> out_f32,
+                 c10::optional<torch::Tensor> out_bf16,
+                 c10::optional<torch::Tensor> out_fp8,
+                 long R, long Ci, long Co, double alpha, bool act_lrelu) {
+  CK8(A); CK8(W);
+  const int lda = A.size(1), ldw = W.size(1);
+  TORCH_CHECK(A.size(0) >= R && lda >= Ci && W.size(0) >= Co && ldw >= Ci);
+  const float* b = nullptr;
+  if (bias.has_value()) { CK(*bias); b = fp(*bias); }
+  const float* isw = nullptr;
+  if (inv_sw.has_value()) { CK(*inv_sw); isw = fp(*inv_sw); }
+  float* of = nullptr; void* ob = nullptr; void* o8 = nullptr; int ldo = 0;
+  if (out_f32.has_value()) { CK(*out_f32); of = fpm(*out_f32); }
+  if (out_bf16.has_value()) { CKB(*out_bf16); ob = bfp(*out_bf16); }
+  if (out_fp8.has_value()) {
+    CK8(*out_fp8);
+    o8 = out_fp8->data_ptr();
+    ldo = out_fp8->size(-1);
+  }
+  TORCH_CHECK(of || ob || o8, "need at least one output");
+  RUN(fv_gemm_nt_fp8(A.data_ptr(), W.data_ptr(), b, isw, of, ob, o8, ldo,
+                     (int)R, (int)Ci, (int)Co, lda, ldw, (float)alpha,
+                     act_lrelu, b != nullptr, cur_stream()));
+}
+
+void absmax_scale(torch::Tensor src, torch::Tensor scale,
+                  torch::Tensor inv_scale) {
+  CK(src); CK(scale); CK(inv_scale);
+  RUN(fv_absmax_scale(fp(src), src.numel(), fpm(scale), fpm(inv_scale),
+                      cur_stream()));
+}
+
+void cast_f32_fp8_scaled(torch::Tensor src, torch::Tensor dst,
+                         c10::optional<torch::Tensor> scale) {
+  CK(src); CK8(dst);
+  const long rows = src.numel() / src.size(-1);
+  const int cols = src.size(-1);
+  TORCH_CHECK(dst.size(-1) >= cols && dst.numel() / dst.size(-1) >= rows);
+  const float* sp = nullptr;
+  if (scale.has_value()) { CK(*scale); sp = fp(*scale); }
+  RUN(fv_cast_f32_fp8_scaled(fp(src), dst.data_ptr(), sp, rows, cols,
+                             (int)dst.size(-1), cur_stream()));
 }
 
 void ln_bwd_params(torch::Tensor x, torch::Tensor dxln, torch::Tensor mean,
@@ -533,7 +605,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("lrelu_bwd", &lrelu_bwd);
   mod.def("ln_fwd", &ln_fwd, py::arg("x"), py::arg("gamma"), py::arg("beta"),
           py::arg("xln"), py::arg("mean"), py::arg("rstd"), py::arg("eps"),
-          py::arg("xln_bf") = py::none());
+          py::arg("xln_bf") = py::none(), py::arg("xln_f8") = py::none());
+  mod.def("gemm_nt_fp8", &gemm_nt_fp8, py::arg("A"), py::arg("W"),
+          py::arg("bias"), py::arg("inv_sw"), py::arg("out_f32"),
+          py::arg("out_bf16"), py::arg("out_fp8"), py::arg("R"),
+          py::arg("Ci"), py::arg("Co"), py::arg("alpha"),
+          py::arg("act_lrelu"));
+  mod.def("absmax_scale", &absmax_scale);
+  mod.def("cast_f32_fp8_scaled", &cast_f32_fp8_scaled);
   mod.def("ln_bwd_params", &ln_bwd_params);
   mod.def("gru_fwd", &gru_fwd);
   mod.def("gru_bwd", &gru_bwd);

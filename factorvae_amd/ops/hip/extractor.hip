@@ -9,7 +9,8 @@
 __global__ __launch_bounds__(256) void ln_fwd_kernel(
     const float* __restrict__ x, const float* __restrict__ gamma,
     const float* __restrict__ beta, float* __restrict__ xln,
-    __bf16* __restrict__ xln_bf,
+    __bf16* __restrict__ xln_bf, unsigned char* __restrict__ xln_f8,
+    int f8_ld,
     float* __restrict__ mean, float* __restrict__ rstd,
     long R, int C, float eps) {
   const int wid = threadIdx.x >> 6;
@@ -39,10 +40,16 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
   }
   float* o = xln ? xln + row * C : nullptr;
   __bf16* ob = xln_bf ? xln_bf + row * C : nullptr;
+  unsigned char* o8 = xln_f8 ? xln_f8 + row * (long)f8_ld : nullptr;
   for (int c = lane; c < C; c += 64) {
     const float v_ = fmaf((xr[c] - mu) * rs, gamma[c], beta[c]);
     if (o) o[c] = v_;
     if (ob) ob[c] = (__bf16)v_;
+    if (o8) {
+      unsigned int u = 0;
+      u = __builtin_amdgcn_cvt_pk_fp8_f32(v_, 0.0f, u, false);
+      o8[c] = (unsigned char)(u & 0xff);
+    }
   }
 }
 
@@ -83,11 +90,13 @@ __global__ __launch_bounds__(256) void ln_bwd_params_kernel(
 extern "C" {
 
 hipError_t fv_ln_fwd(const float* x, const float* gamma, const float* beta,
-                     float* xln, void* xln_bf, float* mean, float* rstd,
+                     float* xln, void* xln_bf, void* xln_f8, int f8_ld,
+                     float* mean, float* rstd,
                      long R, int C, float eps, hipStream_t stream) {
   dim3 grid((unsigned)((R + 3) / 4));
   hipLaunchKernelGGL(ln_fwd_kernel, grid, dim3(256), 0, stream,
-                     x, gamma, beta, xln, (__bf16*)xln_bf, mean, rstd, R, C, eps);
+                     x, gamma, beta, xln, (__bf16*)xln_bf,
+                     (unsigned char*)xln_f8, f8_ld, mean, rstd, R, C, eps);
   HIP_CHECK_LAST();
   return hipSuccess;
 }

@@ -253,17 +253,17 @@ __global__ __launch_bounds__(256) void gemm_tn_kernel(
 __global__ __launch_bounds__(256) void tn_reduce_kernel(
     const float* __restrict__ part, float* __restrict__ out, long elems,
     const float* __restrict__ db_part, float* __restrict__ db, long m_elems,
-    int z) {
+    int z, int accumulate) {
   const long e = (long)blockIdx.x * 256 + threadIdx.x;
   if (e < elems) {
-    float s = 0.0f;
+    float s = accumulate ? out[e] : 0.0f;
     for (int c = 0; c < z; ++c) s += part[(long)c * elems + e];
-    out[e] += s;
+    out[e] = s;
   } else if (e < elems + m_elems) {
     const long m = e - elems;
-    float s = 0.0f;
+    float s = accumulate ? db[m] : 0.0f;
     for (int c = 0; c < z; ++c) s += db_part[(long)c * m_elems + m];
-    db[m] += s;
+    db[m] = s;
   }
 }
 
@@ -347,7 +347,8 @@ hipError_t fv_gemm_tn(const float* A, const float* B, float* out,
     const long m_elems = db ? M : 0;
     dim3 rgrid((unsigned)((elems + m_elems + 255) / 256));
     hipLaunchKernelGGL(tn_reduce_kernel, rgrid, dim3(256), 0, stream,
-                       part, out, elems, db_part, db, m_elems, r_chunks);
+                       part, out, elems, db_part, db, m_elems, r_chunks,
+                       accumulate);
     HIP_CHECK_LAST();
   }
   return hipSuccess;

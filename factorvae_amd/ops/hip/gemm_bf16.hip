@@ -265,21 +265,43 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
 }
 
 // ---------------------------------------------------------------- TN
-// out(M,N) (+)= A(R,M)^T @ B(R,N), k = R: dword-coalesced loads of the
-// R-major rows, transpose-scattered into [m][k]/[n][k] LDS tiles. fp32
-// out/partials (weight grads); fused fp32 bias-grad db = colsum(A) from
-// y==0 blocks. 64x64 output tile (4 waves, 2x2 16x16 frags each) with
-// k-tile 64.
+// out(M,N) (+)= A(R,M)^T @ B(R,N), k = R. Both operands arrive k-major
+// (row-major over R), but MFMA fragments need 8 k-contiguous elements
+// per lane. Instead of a bank-conflicted transpose-scatter, tiles are
+// stored in a blocked [4k x 16col] image written with conflict-free b32
+// stores and read back with gfx950's ds_read_b64_tr_b16 hardware
+// transpose (guide T10: the attention-V recipe): lane l of each 16-lane
+// group receives column (l&15) of a 4x16 row-major bf16 block, so two
+// tr reads assemble the 8-k fragment. Block order is permuted so the
+// instruction's fixed group-stride (+64 elems) lands each lane group on
+// its own k-range. The fused fp32 bias-grad (db = colsum(A)) falls out
+// of the staging registers for free.
 #define TBM 64
 #define TBN 64
 #define TBK 64
+typedef __attribute__((ext_vector_type(4))) short s16x4;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+typedef __attribute__((address_space(3))) s16x4* lds_v4p;
+
+// image position of k-block kb (k = kb*4..+4) within one 16-col block
+// column nb: two tr reads at pos base {0,4} (+group g) hit kb = 2g and
+// kb = 2g+1 of the same 32-k half.
+DEVINL int tn_blockpos(int kb, int nb) {
+  const int r = kb & 7;
+  return nb * 16 + (kb >> 3) * 8 + (r & 1) * 4 + (r >> 1);
+}
+
 __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     float* __restrict__ out, float* __restrict__ part,
     float* __restrict__ db, float* __restrict__ db_part,
     int R, int M, int N, int accumulate) {
-  __shared__ __bf16 As[2][TBM][TBK + 8];
-  __shared__ __bf16 Bs[2][TBN][TBK + 8];
+  // blocked images: 64 blocks x 64 elems (no extra padding needed: the
+  // b32 store pattern below touches all 32 banks exactly once per
+  // 32-lane group)
+  __shared__ __bf16 As[2][TBM * TBK];
+  __shared__ __bf16 Bs[2][TBN * TBK];
+  __shared__ float bred[8][TBM];
 
   const int m0 = blockIdx.x * TBM;
   const int n0 = blockIdx.y * TBN;
@@ -289,14 +311,21 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wv = tid >> 6;        // wave -> 16(m)x64(n) strip as 4 frags
+  const int wv = tid >> 6;        // wave -> m-strip wv*16, all 64 n
   const int fi = lane & 15;
-  const int fk = lane >> 4;
 
-  const int MD = TBM / 2;         // dword cols per k-row (32)
+  // staging assignment (idx = tid + u*256, u in [0,8)):
+  //   dk   = idx & 3              (k within 4-block)
+  //   cp   = ((idx >> 2) & 31)*2  (even col)           [fixed per thread]
+  //   kb   = (idx >> 7) & 15      (k-block)            [b7 + u bits]
+  // -> one b32 store per dword at elem addr pos(kb)*64 + dk*16 + (cp&15),
+  //    banks = dk*8 + (cp&15)/2: all 32 distinct per 32-lane group.
+  const int s_dk = tid & 3;
+  const int s_cp = ((tid >> 2) & 31) * 2;
+  const int s_kb0 = (tid >> 7) & 1;  // + 2*u
 
   const bool do_bias = (db != nullptr) && (blockIdx.y == 0);
-  float bsum[2] = {0.0f, 0.0f};   // per-thread: 2 adjacent m columns
+  float bsum0 = 0.0f, bsum1 = 0.0f;
 
   f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
 
@@ -304,79 +333,83 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
   const int ktiles = (span + TBK - 1) / TBK;
   unsigned int pa[8], pb[8];
 
-  // staging: tile is 64 k-rows x 64 cols = 64x32 dwords
   auto stage_regs = [&](int r0_) {
     const bool interior = (r0_ + TBK <= rend) && (m0 + TBM <= M) &&
                           (n0 + TBN <= N);
     if (interior) {
 #pragma unroll
       for (int u = 0; u < 8; ++u) {
-        const int idx = tid + u * 256;
-        const int krow = idx / MD;
-        const int cp = (idx % MD) * 2;
+        const int krow = s_dk + 4 * (s_kb0 + 2 * u);
         const long gr = (long)r0_ + krow;
-        pa[u] = *(const unsigned int*)(A + gr * M + m0 + cp);
-        pb[u] = *(const unsigned int*)(B + gr * N + n0 + cp);
+        pa[u] = *(const unsigned int*)(A + gr * M + m0 + s_cp);
+        pb[u] = *(const unsigned int*)(B + gr * N + n0 + s_cp);
       }
     } else {
 #pragma unroll
       for (int u = 0; u < 8; ++u) {
-        const int idx = tid + u * 256;
-        const int krow = idx / MD;
-        const int cp = (idx % MD) * 2;
+        const int krow = s_dk + 4 * (s_kb0 + 2 * u);
         const long gr = (long)r0_ + krow;
-        const long rows = rend;  // guard against rend, not R
-        pa[u] = load_dw_guard(A, gr, m0 + cp, rows, M, M);
-        pb[u] = load_dw_guard(B, gr, n0 + cp, rows, N, N);
+        pa[u] = load_dw_guard(A, gr, m0 + s_cp, rend, M, M);
+        pb[u] = load_dw_guard(B, gr, n0 + s_cp, rend, N, N);
       }
     }
   };
   auto regs_to_lds = [&](int buf) {
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
-      const int idx = tid + u * 256;
-      const int krow = idx / MD;
-      const int cp = (idx % MD) * 2;
-      dw_bf2 da_, db_;
-      da_.u = pa[u];
-      db_.u = pb[u];
-      As[buf][cp][krow] = da_.h[0];
-      As[buf][cp + 1][krow] = da_.h[1];
-      Bs[buf][cp][krow] = db_.h[0];
-      Bs[buf][cp + 1][krow] = db_.h[1];
-      if (do_bias) {
-        // colsum(A) accumulated at stage time (fp32 adds); thread owns
-        // columns (m0+cp, m0+cp+1) footprints of its own dwords
-      }
+      const int kb = s_kb0 + 2 * u;
+      const int nb = s_cp >> 4;
+      const int dn = s_cp & 15;
+      const int ea = tn_blockpos(kb, nb) * 64 + s_dk * 16 + dn;
+      *(unsigned int*)&As[buf][ea] = pa[u];
+      *(unsigned int*)&Bs[buf][ea] = pb[u];
+    }
+  };
+  auto bias_from_regs = [&]() {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      dw_bf2 d;
+      d.u = pa[u];
+      bsum0 += (float)d.h[0];
+      bsum1 += (float)d.h[1];
     }
   };
 
   if (ktiles > 0) {
     stage_regs(rbeg);
+    if (do_bias) bias_from_regs();
     regs_to_lds(0);
   }
 
   for (int kt = 0; kt < ktiles; ++kt) {
     __syncthreads();
-    if (kt + 1 < ktiles) stage_regs(rbeg + (kt + 1) * TBK);
-    const int buf = kt & 1;
-    if (do_bias) {
-      // colsum over the staged A tile: thread covers col pair
-      // (2*(tid%32), +1), k-rows tid/32*8 .. +8
-      const int bc = (tid & 31) * 2;
-      const int bq = tid >> 5;
-#pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        bsum[0] += (float)As[buf][bc][bq * 8 + u];
-        bsum[1] += (float)As[buf][bc + 1][bq * 8 + u];
-      }
+    if (kt + 1 < ktiles) {
+      stage_regs(rbeg + (kt + 1) * TBK);
+      if (do_bias) bias_from_regs();
     }
+    const int buf = kt & 1;
 #pragma unroll
     for (int k32 = 0; k32 < TBK; k32 += 32) {
-      const bf16x8 a = *(const bf16x8*)&As[buf][wv * 16 + fi][k32 + fk * 8];
+      const int b32 = k32 >> 5;
+      // A fragment: column (m) fi of the wave's m-strip
+      const int abase = tn_blockpos(b32 * 8, wv) * 64;
+      s16x4 a0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+          (lds_v4p)&As[buf][abase]);
+      s16x4 a1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+          (lds_v4p)&As[buf][abase + 4 * 64]);
+      bf16x8 a;
+      *(bf16x4*)&a = *(bf16x4*)&a0;
+      *(((bf16x4*)&a) + 1) = *(bf16x4*)&a1;
 #pragma unroll
       for (int jt = 0; jt < 4; ++jt) {
-        const bf16x8 b = *(const bf16x8*)&Bs[buf][jt * 16 + fi][k32 + fk * 8];
+        const int bbase = tn_blockpos(b32 * 8, jt) * 64;
+        s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4p)&Bs[buf][bbase]);
+        s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4p)&Bs[buf][bbase + 4 * 64]);
+        bf16x8 b;
+        *(bf16x4*)&b = *(bf16x4*)&b0;
+        *(((bf16x4*)&b) + 1) = *(bf16x4*)&b1;
         acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[jt], 0, 0, 0);
       }
     }
@@ -388,18 +421,17 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
 
   const bool direct = (gridDim.z == 1);
   if (do_bias) {
+    // per-thread partials cover columns (s_cp, s_cp+1) over 8 k-rows
+    // per tile; 8 thread-groups (s_dk x s_kb0) share each column pair
     __syncthreads();
-    // reduce 8 k-row-group partials per column via LDS (reuse As)
-    float* bred = (float*)&As[0][0][0];  // [8][64] fp32
-    const int bc = (tid & 31) * 2;
-    const int bq = tid >> 5;
-    bred[bq * 64 + bc] = bsum[0];
-    bred[bq * 64 + bc + 1] = bsum[1];
+    const int grp = s_dk | (s_kb0 << 2);  // wait: 4*2 = 8 groups
+    bred[grp][s_cp] = bsum0;
+    bred[grp][s_cp + 1] = bsum1;
     __syncthreads();
     if (tid < TBM) {
       float s = 0.0f;
 #pragma unroll
-      for (int q = 0; q < 8; ++q) s += bred[q * 64 + tid];
+      for (int q = 0; q < 8; ++q) s += bred[q][tid];
       const int gm = m0 + tid;
       if (gm < M) {
         if (direct) {
@@ -411,6 +443,7 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
     }
   }
 
+  const int fk = lane >> 4;
   float* po = direct ? out : part + (long)blockIdx.z * M * N;
 #pragma unroll
   for (int jt = 0; jt < 4; ++jt) {
@@ -433,17 +466,17 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
 __global__ __launch_bounds__(256) void tn_reduce_bf16_kernel(
     const float* __restrict__ part, float* __restrict__ out, long elems,
     const float* __restrict__ db_part, float* __restrict__ db, long m_elems,
-    int z) {
+    int z, int accumulate) {
   const long e = (long)blockIdx.x * 256 + threadIdx.x;
   if (e < elems) {
-    float s = 0.0f;
+    float s = accumulate ? out[e] : 0.0f;
     for (int c = 0; c < z; ++c) s += part[(long)c * elems + e];
-    out[e] += s;
+    out[e] = s;
   } else if (e < elems + m_elems) {
     const long m = e - elems;
-    float s = 0.0f;
+    float s = accumulate ? db[m] : 0.0f;
     for (int c = 0; c < z; ++c) s += db_part[(long)c * m_elems + m];
-    db[m] += s;
+    db[m] = s;
   }
 }
 
@@ -515,7 +548,8 @@ hipError_t fv_gemm_tn_bf16(const void* A, const void* B, float* out,
     const long m_elems = db ? M : 0;
     dim3 rgrid((unsigned)((elems + m_elems + 255) / 256));
     hipLaunchKernelGGL(tn_reduce_bf16_kernel, rgrid, dim3(256), 0, stream,
-                       part, out, elems, db_part, db, m_elems, r_chunks);
+                       part, out, elems, db_part, db, m_elems, r_chunks,
+                       accumulate);
     HIP_CHECK_LAST();
   }
   return hipSuccess;
