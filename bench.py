@@ -49,7 +49,7 @@ def parse_args():
     p.add_argument("--hidden_size", type=int, default=64)
     p.add_argument("--num_portfolio", type=int, default=128)
     p.add_argument("--dtype", type=str, default="fp32",
-                   choices=["fp32", "bf16"])
+                   choices=["fp32", "bf16", "fp8"])
     p.add_argument("--lr", type=float, default=1e-4)
     p.add_argument("--seed", type=int, default=42)
     p.add_argument("--n_days", type=int, default=32,

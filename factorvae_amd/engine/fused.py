@@ -65,7 +65,7 @@ class FusedTrainer:
                  device: Optional[torch.device] = None, eta_min: float = 0.0,
                  use_graph: bool = True, max_stocks: Optional[int] = None,
                  train: bool = True, dtype: str = "fp32"):
-        assert dtype in ("fp32", "bf16")
+        assert dtype in ("fp32", "bf16", "fp8")
         self.ext = get_extension()
         # bf16 mode (BASELINE.json configs 2-4): the FLOP-bound extractor
         # GEMM family (R = N*T rows) runs on bf16 MFMA (~2.5 PF/s dense on
@@ -73,7 +73,11 @@ class FusedTrainer:
         # weights + fp32 Adam + fp32 grad all-reduce; the latency-bound
         # N-row kernels (encoder/attention/decoder/GRU recurrence) stay
         # fp32 — their cost is dispatch+LDS, not FLOPs.
-        self.bf16 = dtype == "bf16"
+        # fp8 mode (config 5): forward extractor GEMMs on fp8 e4m3 MFMA
+        # (weights+activations; per-tensor weight scales, LN-normalized
+        # activations at unit scale); backward reuses the bf16 path.
+        self.bf16 = dtype in ("bf16", "fp8")
+        self.fp8 = dtype == "fp8"
         self.dtype = dtype
         self.model = model
         self.device = device or torch.device("cuda")
@@ -99,6 +103,17 @@ class FusedTrainer:
                                       device=self.device)
             self.wih_bf = torch.empty(C3, self.C, dtype=torch.bfloat16,
                                       device=self.device)
+            if self.fp8:
+                ldp = (self.C + 3) & ~3
+                f8 = torch.float8_e4m3fn
+                self.w1x_f8 = torch.empty(self.C, ldp, dtype=f8,
+                                          device=self.device)
+                self.wih_f8 = torch.empty(C3, ldp, dtype=f8,
+                                          device=self.device)
+                self.s_w1x = torch.ones(1, device=self.device)
+                self.is_w1x = torch.ones(1, device=self.device)
+                self.s_wih = torch.ones(1, device=self.device)
+                self.is_wih = torch.ones(1, device=self.device)
             self._refresh_bf16_shadows()
         self.grads = torch.zeros_like(self.params.flat)
         self.adam_m = torch.zeros_like(self.params.flat)
@@ -307,6 +322,12 @@ class FusedTrainer:
             w["dzx_bf"] = fb(R, C)
             w["h_prev_bf"] = fb(R, H)
             w["dgh_bf"] = fb(R, 3 * H)
+            if self.fp8:
+                ldp = (C + 3) & ~3
+                f8t = lambda *shape: torch.zeros(
+                    *shape, device=d, dtype=torch.float8_e4m3fn)
+                w["xln_f8"] = f8t(R, ldp)
+                w["xp_f8"] = f8t(R, ldp)
         self._ws_cache[(N, T)] = w
         self.ws = w
         self._ws_n = N
@@ -322,7 +343,18 @@ class FusedTrainer:
         yv = w["y"] if y is None else y
         alpha = 1.0 / math.sqrt(float(H) + 1e-6)
 
-        if self.bf16:
+        if self.fp8:
+            # fwd in e4m3 (weights+activations); also emits the bf16
+            # activation copies the bf16 backward consumes
+            ext.ln_fwd(x2d, p("ln_g"), p("ln_b"), None, w["mean"],
+                       w["rstd"], 1e-5, w["xln_bf"], w["xln_f8"])
+            ext.gemm_nt_fp8(w["xln_f8"], self.w1x_f8, p("b1x"), self.is_w1x,
+                            None, w["xp_bf"], w["xp_f8"], R, self.C, self.C,
+                            1.0, True)
+            ext.gemm_nt_fp8(w["xp_f8"], self.wih_f8, p("bih"), self.is_wih,
+                            w["gi"].view(R, 3 * H), None, None, R, self.C,
+                            3 * H, 1.0, False)
+        elif self.bf16:
             ext.ln_fwd(x2d, p("ln_g"), p("ln_b"), None, w["mean"],
                        w["rstd"], 1e-5, w["xln_bf"])
             ext.gemm_nt_bf16(w["xln_bf"], self.w1x_bf, p("b1x"), None,
@@ -511,6 +543,13 @@ class FusedTrainer:
     def _refresh_bf16_shadows(self):
         self.ext.cast_f32_bf16(self.p("W1x"), self.w1x_bf)
         self.ext.cast_f32_bf16(self.p("Wih"), self.wih_bf)
+        if self.fp8:
+            self.ext.absmax_scale(self.p("W1x"), self.s_w1x, self.is_w1x)
+            self.ext.cast_f32_fp8_scaled(self.p("W1x"), self.w1x_f8,
+                                         self.s_w1x)
+            self.ext.absmax_scale(self.p("Wih"), self.s_wih, self.is_wih)
+            self.ext.cast_f32_fp8_scaled(self.p("Wih"), self.wih_f8,
+                                         self.s_wih)
 
     def _launch_optimizer(self, inc: bool = True):
         if inc:

@@ -44,7 +44,7 @@ def build_argparser() -> argparse.ArgumentParser:
                         choices=["auto", "fused", "eager"],
                         help="auto = fused HIP engine on GPU, eager on CPU")
     parser.add_argument("--dtype", type=str, default="fp32",
-                        choices=["fp32", "bf16"],
+                        choices=["fp32", "bf16", "fp8"],
                         help="fused-engine compute dtype for the extractor "
                              "GEMM family (bf16 MFMA path)")
     parser.add_argument("--resume", action="store_true",

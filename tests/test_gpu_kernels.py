@@ -592,3 +592,88 @@ def test_bf16_training_reduces_loss():
         last = tot
     assert last < first, (first, last)
     assert last == last  # not NaN
+
+
+# ------------------------------------------------------------------ fp8
+def test_fp8_cast_and_absmax():
+    x = t(300, 158, seed=20)
+    scale = torch.zeros(1, device=DEV)
+    inv = torch.zeros(1, device=DEV)
+    ext.absmax_scale(x.reshape(-1), scale, inv)
+    torch.cuda.synchronize()
+    amax = float(x.abs().max())
+    assert float(scale) == pytest.approx(448.0 / amax, rel=1e-5)
+    assert float(inv) == pytest.approx(amax / 448.0, rel=1e-5)
+    ldp = (158 + 3) & ~3
+    x8 = torch.zeros(300, ldp, device=DEV, dtype=torch.float8_e4m3fn)
+    ext.cast_f32_fp8_scaled(x, x8, scale)
+    torch.cuda.synchronize()
+    back = x8[:, :158].float() * float(inv)
+    # e4m3 has ~2^-3 relative resolution
+    assert_close(back, x, atol=3e-3 * amax, rtol=0.08, what="fp8 roundtrip")
+
+
+@pytest.mark.parametrize("R,Ci,Co", [(300, 158, 192), (37, 33, 20), (6000, 158, 158)])
+def test_gemm_nt_fp8(R, Ci, Co):
+    """fp8 MFMA GEMM == matmul of the e4m3-quantized operands (fp32 acc)."""
+    ldp = (Ci + 3) & ~3
+    A = t(R, Ci, seed=21, scale=1.0)
+    W = t(Co, Ci, seed=22, scale=0.1)
+    sw = torch.zeros(1, device=DEV)
+    isw = torch.zeros(1, device=DEV)
+    ext.absmax_scale(W.reshape(-1), sw, isw)
+    A8 = torch.zeros(R, ldp, device=DEV, dtype=torch.float8_e4m3fn)
+    W8 = torch.zeros(Co, ldp, device=DEV, dtype=torch.float8_e4m3fn)
+    ext.cast_f32_fp8_scaled(A, A8, None)
+    ext.cast_f32_fp8_scaled(W, W8, sw)
+    b = t(Co, seed=23)
+    out = torch.empty(R, Co, device=DEV)
+    outb = torch.empty(R, Co, device=DEV, dtype=torch.bfloat16)
+    ext.gemm_nt_fp8(A8, W8, b, isw, out, outb, None, R, Ci, Co, 1.0, False)
+    torch.cuda.synchronize()
+    ref = (A8[:, :Ci].float() @ W8[:, :Ci].float().t()) * float(isw) + b
+    assert_close(out, ref, atol=2e-2 * math.sqrt(Ci), rtol=2e-2,
+                 what="gemm_nt_fp8")
+    assert_close(outb.float(), ref, atol=3e-2 * math.sqrt(Ci), rtol=3e-2,
+                 what="gemm_nt_fp8 bf16out")
+    # lrelu epilogue + fp8 out
+    out8 = torch.zeros(R, ldp, device=DEV, dtype=torch.float8_e4m3fn)
+    ext.gemm_nt_fp8(A8, W8, b, isw, None, None, out8, R, Ci, Co, 1.0, True)
+    torch.cuda.synchronize()
+    ref2 = F.leaky_relu(ref, 0.01)
+    assert_close(out8[:, :Co].float(), ref2, atol=4e-2 * math.sqrt(Ci),
+                 rtol=0.1, what="gemm_nt_fp8 fp8out lrelu")
+
+
+def test_fp8_full_step_vs_fp32():
+    """fp8 engine mode: loss near fp32, gradients directionally right."""
+    from factorvae_amd.engine.fused import FusedTrainer
+    from factorvae_amd.models.modules import build_factorvae
+    from factorvae_amd.utils import set_seed
+
+    N, T, C, H, M, K = 300, 20, 158, 64, 128, 20
+    losses, grads = {}, {}
+    for dtype in ("fp32", "fp8"):
+        set_seed(0)
+        model = build_factorvae(num_latent=C, hidden_size=H, num_portfolio=M,
+                                num_factor=K).to(DEV)
+        tr = FusedTrainer(model, lr=1e-4, t_max=100, device=DEV,
+                          use_graph=False, train=True, dtype=dtype)
+        x = t(N, T, C, seed=50)
+        y = t(N, 1, seed=51)
+        tr._ensure_ws(N, T)
+        w = tr.ws
+        w["x"].copy_(x)
+        w["y"].copy_(y)
+        set_seed(1)
+        tr._fill_rng(N)
+        tr.grads.zero_()
+        tr._launch_forward(N, T)
+        tr._launch_backward(N, T)
+        torch.cuda.synchronize()
+        losses[dtype] = float(w["loss"][0])
+        grads[dtype] = tr.grads.clone()
+
+    assert abs(losses["fp8"] - losses["fp32"]) < 0.1 * (abs(losses["fp32"]) + 1.0)
+    cos = torch.nn.functional.cosine_similarity(grads["fp32"], grads["fp8"], dim=0)
+    assert float(cos) > 0.97, f"grad cosine {float(cos)}"
