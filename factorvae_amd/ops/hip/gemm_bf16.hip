@@ -266,41 +266,31 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
 
 // ---------------------------------------------------------------- TN
 // out(M,N) (+)= A(R,M)^T @ B(R,N), k = R. Both operands arrive k-major
-// (row-major over R), but MFMA fragments need 8 k-contiguous elements
-// per lane. Instead of a bank-conflicted transpose-scatter, tiles are
-// stored in a blocked [4k x 16col] image written with conflict-free b32
-// stores and read back with gfx950's ds_read_b64_tr_b16 hardware
-// transpose (guide T10: the attention-V recipe): lane l of each 16-lane
-// group receives column (l&15) of a 4x16 row-major bf16 block, so two
-// tr reads assemble the 8-k fragment. Block order is permuted so the
-// instruction's fixed group-stride (+64 elems) lands each lane group on
-// its own k-range. The fused fp32 bias-grad (db = colsum(A)) falls out
-// of the staging registers for free.
+// (row-major over R), and MFMA fragments need 8 k-contiguous elements
+// per lane. The tile stays ROW-MAJOR in LDS (same as global -> fully
+// coalesced dword stores, zero transpose work); the transpose happens
+// in the READ via gfx950's ds_read_b64_tr_b16: within each 4-lane quad,
+// member m supplies a 64-bit load of 4 consecutive columns at k-row
+// (g*8 + m), and the hardware hands lane l element (l&3) of each
+// member's load — i.e. lane l receives column (l&15) at k = g*8+j
+// (verified on hardware by scripts/probe/tr16_probe.hip). Two tr reads
+// assemble the 8-k fragment. The fused fp32 bias-grad (db = colsum(A))
+// falls out of the staging registers for free.
 #define TBM 64
 #define TBN 64
 #define TBK 64
+#define TSA (TBM + 4)   // row stride (elems): 8B-aligned, conflict-free
 typedef __attribute__((ext_vector_type(4))) short s16x4;
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 typedef __attribute__((address_space(3))) s16x4* lds_v4p;
-
-// image position of k-block kb (k = kb*4..+4) within one 16-col block
-// column nb: two tr reads at pos base {0,4} (+group g) hit kb = 2g and
-// kb = 2g+1 of the same 32-k half.
-DEVINL int tn_blockpos(int kb, int nb) {
-  const int r = kb & 7;
-  return nb * 16 + (kb >> 3) * 8 + (r & 1) * 4 + (r >> 1);
-}
 
 __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     float* __restrict__ out, float* __restrict__ part,
     float* __restrict__ db, float* __restrict__ db_part,
     int R, int M, int N, int accumulate) {
-  // blocked images: 64 blocks x 64 elems (no extra padding needed: the
-  // b32 store pattern below touches all 32 banks exactly once per
-  // 32-lane group)
-  __shared__ __bf16 As[2][TBM * TBK];
-  __shared__ __bf16 Bs[2][TBN * TBK];
+  __shared__ __bf16 As[2][TBK][TSA];
+  __shared__ __bf16 Bs[2][TBK][TSA];
   __shared__ float bred[8][TBM];
 
   const int m0 = blockIdx.x * TBM;
@@ -313,16 +303,14 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
   const int lane = tid & 63;
   const int wv = tid >> 6;        // wave -> m-strip wv*16, all 64 n
   const int fi = lane & 15;
+  const int fk = lane >> 4;       // k-group g
+  const int qm = lane & 3;        // quad member -> k-row offset
+  const int nq = fi & ~3;         // column-quad base
 
-  // staging assignment (idx = tid + u*256, u in [0,8)):
-  //   dk   = idx & 3              (k within 4-block)
-  //   cp   = ((idx >> 2) & 31)*2  (even col)           [fixed per thread]
-  //   kb   = (idx >> 7) & 15      (k-block)            [b7 + u bits]
-  // -> one b32 store per dword at elem addr pos(kb)*64 + dk*16 + (cp&15),
-  //    banks = dk*8 + (cp&15)/2: all 32 distinct per 32-lane group.
-  const int s_dk = tid & 3;
-  const int s_cp = ((tid >> 2) & 31) * 2;
-  const int s_kb0 = (tid >> 7) & 1;  // + 2*u
+  // staging: row-major dwords; thread t -> (krow = t/32 + 8u, colpair
+  // cp = (t%32)*2): coalesced global loads, conflict-free b32 stores
+  const int s_cp = (tid & 31) * 2;
+  const int s_kr0 = tid >> 5;     // + 8*u
 
   const bool do_bias = (db != nullptr) && (blockIdx.y == 0);
   float bsum0 = 0.0f, bsum1 = 0.0f;
@@ -339,16 +327,14 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
     if (interior) {
 #pragma unroll
       for (int u = 0; u < 8; ++u) {
-        const int krow = s_dk + 4 * (s_kb0 + 2 * u);
-        const long gr = (long)r0_ + krow;
+        const long gr = (long)r0_ + s_kr0 + 8 * u;
         pa[u] = *(const unsigned int*)(A + gr * M + m0 + s_cp);
         pb[u] = *(const unsigned int*)(B + gr * N + n0 + s_cp);
       }
     } else {
 #pragma unroll
       for (int u = 0; u < 8; ++u) {
-        const int krow = s_dk + 4 * (s_kb0 + 2 * u);
-        const long gr = (long)r0_ + krow;
+        const long gr = (long)r0_ + s_kr0 + 8 * u;
         pa[u] = load_dw_guard(A, gr, m0 + s_cp, rend, M, M);
         pb[u] = load_dw_guard(B, gr, n0 + s_cp, rend, N, N);
       }
@@ -357,12 +343,9 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
   auto regs_to_lds = [&](int buf) {
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
-      const int kb = s_kb0 + 2 * u;
-      const int nb = s_cp >> 4;
-      const int dn = s_cp & 15;
-      const int ea = tn_blockpos(kb, nb) * 64 + s_dk * 16 + dn;
-      *(unsigned int*)&As[buf][ea] = pa[u];
-      *(unsigned int*)&Bs[buf][ea] = pb[u];
+      const int kr = s_kr0 + 8 * u;
+      *(unsigned int*)&As[buf][kr][s_cp] = pa[u];
+      *(unsigned int*)&Bs[buf][kr][s_cp] = pb[u];
     }
   };
   auto bias_from_regs = [&]() {
@@ -390,23 +373,20 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
     const int buf = kt & 1;
 #pragma unroll
     for (int k32 = 0; k32 < TBK; k32 += 32) {
-      const int b32 = k32 >> 5;
-      // A fragment: column (m) fi of the wave's m-strip
-      const int abase = tn_blockpos(b32 * 8, wv) * 64;
+      const int kb = k32 + fk * 8 + qm;
       s16x4 a0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-          (lds_v4p)&As[buf][abase]);
+          (lds_v4p)&As[buf][kb][wv * 16 + nq]);
       s16x4 a1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-          (lds_v4p)&As[buf][abase + 4 * 64]);
+          (lds_v4p)&As[buf][kb + 4][wv * 16 + nq]);
       bf16x8 a;
       *(bf16x4*)&a = *(bf16x4*)&a0;
       *(((bf16x4*)&a) + 1) = *(bf16x4*)&a1;
 #pragma unroll
       for (int jt = 0; jt < 4; ++jt) {
-        const int bbase = tn_blockpos(b32 * 8, jt) * 64;
         s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (lds_v4p)&Bs[buf][bbase]);
+            (lds_v4p)&Bs[buf][kb][jt * 16 + nq]);
         s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (lds_v4p)&Bs[buf][bbase + 4 * 64]);
+            (lds_v4p)&Bs[buf][kb + 4][jt * 16 + nq]);
         bf16x8 b;
         *(bf16x4*)&b = *(bf16x4*)&b0;
         *(((bf16x4*)&b) + 1) = *(bf16x4*)&b1;
@@ -421,12 +401,11 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
 
   const bool direct = (gridDim.z == 1);
   if (do_bias) {
-    // per-thread partials cover columns (s_cp, s_cp+1) over 8 k-rows
-    // per tile; 8 thread-groups (s_dk x s_kb0) share each column pair
+    // per-thread partials cover columns (s_cp, s_cp+1); the 8 s_kr0
+    // groups sharing a column pair reduce through LDS
     __syncthreads();
-    const int grp = s_dk | (s_kb0 << 2);  // wait: 4*2 = 8 groups
-    bred[grp][s_cp] = bsum0;
-    bred[grp][s_cp + 1] = bsum1;
+    bred[s_kr0][s_cp] = bsum0;
+    bred[s_kr0][s_cp + 1] = bsum1;
     __syncthreads();
     if (tid < TBM) {
       float s = 0.0f;
@@ -443,7 +422,6 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
     }
   }
 
-  const int fk = lane >> 4;
   float* po = direct ? out : part + (long)blockIdx.z * M * N;
 #pragma unroll
   for (int jt = 0; jt < 4; ++jt) {
