@@ -492,13 +492,17 @@ class FusedTrainer:
         ext.gemm_nn(w["dscores"], p("Wenc"), None, w["dh"], 1.0, True, False)
 
         # extractor backward
-        ext.gru_bwd(w["dh"], w["h_prev"], w["gates4"], p("Whh"), w["dgi"],
-                    w["dgh"], N, T, H)
         if self.bf16:
-            ext.cast_f32_bf16(w["dgi"].view(-1), w["dgi_bf"].view(-1))
+            ext.gru_bwd(w["dh"], w["h_prev"], w["gates4"], p("Whh"),
+                        w["dgi"], w["dgh"], N, T, H,
+                        w["dgi_bf"].view(N, T, 3 * H),
+                        w["dgh_bf"].view(N, T, 3 * H))
+        else:
+            ext.gru_bwd(w["dh"], w["h_prev"], w["gates4"], p("Whh"),
+                        w["dgi"], w["dgh"], N, T, H)
+        if self.bf16:
             fork()
             with _on_side():
-                ext.cast_f32_bf16(w["dgh"].view(-1), w["dgh_bf"].view(-1))
                 ext.gemm_tn_bf16(w["dgh_bf"].view(R, 3 * H), w["h_prev_bf"],
                                  g("Whh"), w["tn_part"], chunks, True,
                                  g("bhh"), w["tn_partb"])

@@ -42,7 +42,8 @@ hipError_t fv_ln_bwd_params(const float*, const float*, const float*,
 hipError_t fv_gru_fwd(const float*, const float*, const float*, float*, float*,
                       float*, float*, int, int, int, hipStream_t);
 hipError_t fv_gru_bwd(const float*, const float*, const float*, const float*,
-                      float*, float*, int, int, int, hipStream_t);
+                      float*, float*, void*, void*, int, int, int,
+                      hipStream_t);
 hipError_t fv_enc_softmax_fwd(const float*, const float*, float*, float*, int,
                               int, hipStream_t);
 hipError_t fv_enc_softmax_bwd(const float*, const float*, const float*, float*,
@@ -389,10 +390,15 @@ void gru_fwd(torch::Tensor gi, torch::Tensor Whh, torch::Tensor bhh,
 
 void gru_bwd(torch::Tensor dh_final, torch::Tensor h_prev, torch::Tensor gates4,
              torch::Tensor Whh, torch::Tensor dgi, torch::Tensor dgh, long N,
-             long T, long H) {
+             long T, long H,
+             c10::optional<torch::Tensor> dgi_bf = c10::nullopt,
+             c10::optional<torch::Tensor> dgh_bf = c10::nullopt) {
   CK(dh_final); CK(h_prev); CK(gates4); CK(Whh); CK(dgi); CK(dgh);
+  void* gib = nullptr; void* ghb = nullptr;
+  if (dgi_bf.has_value()) { CKB(*dgi_bf); gib = bfp(*dgi_bf); }
+  if (dgh_bf.has_value()) { CKB(*dgh_bf); ghb = bfp(*dgh_bf); }
   RUN(fv_gru_bwd(fp(dh_final), fp(h_prev), fp(gates4), fp(Whh), fpm(dgi),
-                 fpm(dgh), (int)N, (int)T, (int)H, cur_stream()));
+                 fpm(dgh), gib, ghb, (int)N, (int)T, (int)H, cur_stream()));
 }
 
 void enc_softmax_fwd(torch::Tensor scores, torch::Tensor y, torch::Tensor a,
@@ -615,7 +621,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("cast_f32_fp8_scaled", &cast_f32_fp8_scaled);
   mod.def("ln_bwd_params", &ln_bwd_params);
   mod.def("gru_fwd", &gru_fwd);
-  mod.def("gru_bwd", &gru_bwd);
+  mod.def("gru_bwd", &gru_bwd, py::arg("dh_final"), py::arg("h_prev"),
+          py::arg("gates4"), py::arg("Whh"), py::arg("dgi"), py::arg("dgh"),
+          py::arg("N"), py::arg("T"), py::arg("H"),
+          py::arg("dgi_bf") = py::none(), py::arg("dgh_bf") = py::none());
   mod.def("enc_softmax_fwd", &enc_softmax_fwd);
   mod.def("enc_softmax_bwd", &enc_softmax_bwd);
   mod.def("enc_heads_fwd", &enc_heads_fwd);

@@ -100,6 +100,7 @@ __global__ __launch_bounds__(256) void gru_bwd_generic_kernel(
     const float* __restrict__ Whh,        // (3H,H)
     float* __restrict__ dgi,              // (N,T,3H)
     float* __restrict__ dgh,              // (N,T,3H)
+    __bf16* __restrict__ dgi_bf, __bf16* __restrict__ dgh_bf,
     int N, int T, int H) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* WhhS = (float*)smem;                    // [3H][H] as-is
@@ -144,6 +145,16 @@ __global__ __launch_bounds__(256) void gru_bwd_generic_kernel(
       dgh[ob] = dgate_r;
       dgh[ob + H] = dgate_z;
       dgh[ob + 2 * H] = dgh_n;
+      if (dgi_bf) {
+        dgi_bf[ob] = (__bf16)dgate_r;
+        dgi_bf[ob + H] = (__bf16)dgate_z;
+        dgi_bf[ob + 2 * H] = (__bf16)dgi_n;
+      }
+      if (dgh_bf) {
+        dgh_bf[ob] = (__bf16)dgate_r;
+        dgh_bf[ob + H] = (__bf16)dgate_z;
+        dgh_bf[ob + 2 * H] = (__bf16)dgh_n;
+      }
 
       dghS[w * 3 * H + lane] = dgate_r;
       dghS[w * 3 * H + H + lane] = dgate_z;
@@ -262,7 +273,9 @@ __global__ __launch_bounds__(512) void gru_fwd_fast_kernel(
 __global__ __launch_bounds__(512) void gru_bwd_fast_kernel(
     const float* __restrict__ dh_final, const float* __restrict__ h_prev_in,
     const float* __restrict__ gates4, const float* __restrict__ Whh,
-    float* __restrict__ dgi, float* __restrict__ dgh, int N, int T, int H) {
+    float* __restrict__ dgi, float* __restrict__ dgh,
+    __bf16* __restrict__ dgi_bf, __bf16* __restrict__ dgh_bf,
+    int N, int T, int H) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* WB4 = (float*)smem;                        // [3H/4][64][4]
   float* dghS = WB4 + (size_t)3 * H * 64;           // [GRU_SPW_F][3H]
@@ -328,6 +341,16 @@ __global__ __launch_bounds__(512) void gru_bwd_fast_kernel(
       dgh[ob] = dgate_r;
       dgh[ob + H] = dgate_z;
       dgh[ob + 2 * H] = dgh_n;
+      if (dgi_bf) {
+        dgi_bf[ob] = (__bf16)dgate_r;
+        dgi_bf[ob + H] = (__bf16)dgate_z;
+        dgi_bf[ob + 2 * H] = (__bf16)da;
+      }
+      if (dgh_bf) {
+        dgh_bf[ob] = (__bf16)dgate_r;
+        dgh_bf[ob + H] = (__bf16)dgate_z;
+        dgh_bf[ob + 2 * H] = (__bf16)dgh_n;
+      }
 
       dghS[w * 3 * H + lane] = dgate_r;
       dghS[w * 3 * H + H + lane] = dgate_z;
@@ -372,21 +395,23 @@ hipError_t fv_gru_fwd(const float* gi, const float* Whh, const float* bhh,
 
 hipError_t fv_gru_bwd(const float* dh_final, const float* h_prev,
                       const float* gates4, const float* Whh,
-                      float* dgi, float* dgh, int N, int T, int H,
-                      hipStream_t stream) {
+                      float* dgi, float* dgh, void* dgi_bf, void* dgh_bf,
+                      int N, int T, int H, hipStream_t stream) {
   if (H > 64) return hipErrorInvalidValue;
   if ((H & 3) == 0) {
     const size_t lds = ((size_t)3 * H * 64 + GRU_SPW_F * 3 * H +
                         (size_t)GRU_SPW_F * 64) * sizeof(float);
     dim3 grid((N + GRU_SPW_F - 1) / GRU_SPW_F);
     hipLaunchKernelGGL(gru_bwd_fast_kernel, grid, dim3(512), lds, stream,
-                       dh_final, h_prev, gates4, Whh, dgi, dgh, N, T, H);
+                       dh_final, h_prev, gates4, Whh, dgi, dgh,
+                       (__bf16*)dgi_bf, (__bf16*)dgh_bf, N, T, H);
   } else {
     const size_t lds =
         ((size_t)3 * H * H + GRU_SPW * 3 * H + GRU_SPW * H) * sizeof(float);
     dim3 grid((N + GRU_SPW - 1) / GRU_SPW);
     hipLaunchKernelGGL(gru_bwd_generic_kernel, grid, dim3(256), lds, stream,
-                       dh_final, h_prev, gates4, Whh, dgi, dgh, N, T, H);
+                       dh_final, h_prev, gates4, Whh, dgi, dgh,
+                       (__bf16*)dgi_bf, (__bf16*)dgh_bf, N, T, H);
   }
   HIP_CHECK_LAST();
   return hipSuccess;
