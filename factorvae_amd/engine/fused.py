@@ -510,8 +510,8 @@ class FusedTrainer:
                                  g("Wih"), w["tn_part2"], chunks, True,
                                  g("bih"), w["tn_partb2"])
             ext.gemm_nn_bf16(w["dgi_bf"].view(R, 3 * H), self.wih_bf, None,
-                             None, w["dxp_bf"], 1.0, False, False)
-            ext.lrelu_bwd_bf16(w["dxp_bf"], w["xp_bf"], w["dzx_bf"])
+                             None, w["dzx_bf"], 1.0, False, False,
+                             w["xp_bf"])
             fork()
             with _on_side():
                 ext.gemm_tn_bf16(w["dzx_bf"], w["xln_bf"], g("W1x"),

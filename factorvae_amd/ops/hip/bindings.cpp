@@ -19,8 +19,8 @@ hipError_t fv_gemm_nt_bf16(const void*, const void*, const float*, float*,
                            void*, int, int, int, float, int, int,
                            hipStream_t);
 hipError_t fv_gemm_nn_bf16(const void*, const void*, const float*, float*,
-                           void*, int, int, int, float, int, int,
-                           hipStream_t);
+                           void*, const void*, int, int, int, float, int,
+                           int, hipStream_t);
 hipError_t fv_gemm_tn_bf16(const void*, const void*, float*, float*, float*,
                            float*, int, int, int, int, int, hipStream_t);
 hipError_t fv_cast_f32_bf16(const float*, void*, long, hipStream_t);
@@ -187,7 +187,8 @@ void gemm_nn_bf16(torch::Tensor A, torch::Tensor B,
                   c10::optional<torch::Tensor> bias,
                   c10::optional<torch::Tensor> out_f32,
                   c10::optional<torch::Tensor> out_bf16,
-                  double alpha, bool accumulate, bool act_lrelu) {
+                  double alpha, bool accumulate, bool act_lrelu,
+                  c10::optional<torch::Tensor> lrelu_bwd_of = c10::nullopt) {
   CKB(A); CKB(B);
   const int R = A.size(0), Ci = A.size(1), Co = B.size(1);
   TORCH_CHECK(B.size(0) == Ci);
@@ -205,8 +206,14 @@ void gemm_nn_bf16(torch::Tensor A, torch::Tensor B,
     ob = bfp(*out_bf16);
   }
   TORCH_CHECK(of || ob, "need at least one output");
-  RUN(fv_gemm_nn_bf16(bfpc(A), bfpc(B), b, of, ob, R, Ci, Co, (float)alpha,
-                      accumulate, act_lrelu, cur_stream()));
+  const void* yp = nullptr;
+  if (lrelu_bwd_of.has_value()) {
+    CKB(*lrelu_bwd_of);
+    TORCH_CHECK(lrelu_bwd_of->size(0) == R && lrelu_bwd_of->size(1) == Co);
+    yp = bfpc(*lrelu_bwd_of);
+  }
+  RUN(fv_gemm_nn_bf16(bfpc(A), bfpc(B), b, of, ob, yp, R, Ci, Co,
+                      (float)alpha, accumulate, act_lrelu, cur_stream()));
 }
 
 void gemm_tn_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor out,
@@ -601,7 +608,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           py::arg("alpha"), py::arg("accumulate"), py::arg("act_lrelu"));
   mod.def("gemm_nn_bf16", &gemm_nn_bf16, py::arg("A"), py::arg("B"),
           py::arg("bias"), py::arg("out_f32"), py::arg("out_bf16"),
-          py::arg("alpha"), py::arg("accumulate"), py::arg("act_lrelu"));
+          py::arg("alpha"), py::arg("accumulate"), py::arg("act_lrelu"),
+          py::arg("lrelu_bwd_of") = py::none());
   mod.def("gemm_tn_bf16", &gemm_tn_bf16, py::arg("A"), py::arg("B"),
           py::arg("out"), py::arg("part"), py::arg("r_chunks"),
           py::arg("accumulate"), py::arg("db") = py::none(),

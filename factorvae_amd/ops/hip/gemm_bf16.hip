@@ -161,11 +161,13 @@ __global__ __launch_bounds__(256) void gemm_nt_bf16_kernel(
 // B (Ci,Co) bf16 row-major: k runs over rows -> dword-coalesced loads of
 // B's rows, transpose-scattered into Bs[n][k] (two b16 LDS stores per
 // dword).
+// flags bit3: multiply the result by lrelu'(Y[gr][gc]) — fuses the
+// elementwise lrelu-backward into the producing dgrad GEMM.
 __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     const float* __restrict__ bias, float* __restrict__ out_f32,
-    __bf16* __restrict__ out_bf16, int R, int Ci, int Co, float alpha,
-    int flags) {
+    __bf16* __restrict__ out_bf16, const __bf16* __restrict__ Y,
+    int R, int Ci, int Co, float alpha, int flags) {
   __shared__ __bf16 As[2][BBR][BBK + 8];
   __shared__ __bf16 Bs[2][BBC][BBK + 8];
 
@@ -254,6 +256,10 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
       if (flags & 4) v += bias[gc];
       v *= alpha;
       if (flags & 2) v = lrelu_(v);
+      if (flags & 8) {
+        const float y = (float)Y[(long)gr * Co + gc];
+        v *= (y > 0.0f ? 1.0f : 0.01f);
+      }
       if (out_f32) {
         float* o = &out_f32[(long)gr * Co + gc];
         if (flags & 1) v += *o;
@@ -271,10 +277,10 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
 // coalesced dword stores, zero transpose work); the transpose happens
 // in the READ via gfx950's ds_read_b64_tr_b16: within each 4-lane quad,
 // member m supplies a 64-bit load of 4 consecutive columns at k-row
-// (g*8 + m), and the hardware hands lane l element (l&3) of each
-// member's load — i.e. lane l receives column (l&15) at k = g*8+j
-// (verified on hardware by scripts/probe/tr16_probe.hip). Two tr reads
-// assemble the 8-k fragment. The fused fp32 bias-grad (db = colsum(A))
+// (g*8 + (m>>2)), and the hardware transposes the group's 16x4 loads so
+// lane l receives column (l&15) with k ascending over the result
+// elements (verified on hardware by scripts/probe/tr16_probe2.hip).
+// Two tr reads assemble the 8-k fragment. The fused fp32 bias-grad (db = colsum(A))
 // falls out of the staging registers for free.
 #define TBM 64
 #define TBN 64
@@ -304,8 +310,12 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
   const int wv = tid >> 6;        // wave -> m-strip wv*16, all 64 n
   const int fi = lane & 15;
   const int fk = lane >> 4;       // k-group g
-  const int qm = lane & 3;        // quad member -> k-row offset
-  const int nq = fi & ~3;         // column-quad base
+  // tr16 supplier role (hardware-verified, scripts/probe/tr16_probe2):
+  // within each 16-lane group, lane m loads 4 consecutive COLUMNS at
+  // k-row offset (m>>2); the transpose hands lane l column l&15 with
+  // k ascending over the 4 result elements.
+  const int qm = (fi >> 2);       // -> k-row offset
+  const int nq = (lane & 3) * 4;  // -> column-quad base
 
   // staging: row-major dwords; thread t -> (krow = t/32 + 8u, colpair
   // cp = (t%32)*2): coalesced global loads, conflict-free b32 stores
@@ -492,14 +502,16 @@ hipError_t fv_gemm_nt_bf16(const void* A, const void* W, const float* bias,
 }
 
 hipError_t fv_gemm_nn_bf16(const void* A, const void* B, const float* bias,
-                           float* out_f32, void* out_bf16, int R, int Ci,
-                           int Co, float alpha, int accumulate, int act_lrelu,
-                           hipStream_t stream) {
-  int flags = (accumulate ? 1 : 0) | (act_lrelu ? 2 : 0) | (bias ? 4 : 0);
+                           float* out_f32, void* out_bf16, const void* Y,
+                           int R, int Ci, int Co, float alpha, int accumulate,
+                           int act_lrelu, hipStream_t stream) {
+  int flags = (accumulate ? 1 : 0) | (act_lrelu ? 2 : 0) | (bias ? 4 : 0) |
+              (Y ? 8 : 0);
   dim3 grid((R + BBR - 1) / BBR, (Co + BBC - 1) / BBC);
   hipLaunchKernelGGL(gemm_nn_bf16_kernel, grid, dim3(256), 0, stream,
                      (const __bf16*)A, (const __bf16*)B, bias, out_f32,
-                     (__bf16*)out_bf16, R, Ci, Co, alpha, flags);
+                     (__bf16*)out_bf16, (const __bf16*)Y, R, Ci, Co, alpha,
+                     flags);
   HIP_CHECK_LAST();
   return hipSuccess;
 }
