@@ -92,9 +92,13 @@ __global__ __launch_bounds__(256) void dec_fwd_kernel(
 }
 
 // Backward. Writes dh (N,H) directly (FIRST dh contributor — plain
-// store), dz1 (N,H) and dbeta (N,K) for the weight-grad gemm_tn calls,
-// and accumulates dfmu/dfsig_c (K), dwmu/dwsig (H), dbmu/dbsig (1)
-// atomically across row workgroups.
+// store), dz1 (N,H) and dbeta (N,K) for the weight-grad gemm_tn calls.
+// The shared grads dfmu/dfsig_c (K), dwmu/dwsig (H), dbmu/dbsig (1) are
+// accumulated in REGISTERS over DEC_ITERS rows per wave and flushed with
+// ONE atomicAdd per element per block — N per-row atomic storms (the
+// profile showed 1.1M global atomics at N=3500) collapse to
+// N/(DEC_RPW*DEC_ITERS) per element.
+#define DEC_ITERS 8
 __global__ __launch_bounds__(256) void dec_bwd_kernel(
     const float* __restrict__ drecon, const float* __restrict__ h,
     const float* __restrict__ a1, const float* __restrict__ beta,
@@ -113,67 +117,131 @@ __global__ __launch_bounds__(256) void dec_bwd_kernel(
   float* WbS = W1S + (size_t)H * H;          // [K][H] as-is
   float* dz1S = WbS + (size_t)K * H;         // [DEC_RPW][H]
   float* dbS = dz1S + (size_t)DEC_RPW * H;   // [DEC_RPW][K]
-  float* red = dbS + (size_t)DEC_RPW * K;    // [DEC_RPW] scratch
+  float* red = dbS + (size_t)DEC_RPW * K;    // [4][max(K,2H)] reduce pad
 
   const int tid = threadIdx.x;
   const int w = tid >> 6;
   const int lane = tid & 63;
-  const int row = blockIdx.x * DEC_RPW + w;
-  const bool live = row < N;
+  const bool kh2 = (lane + 64) < K;
 
   for (int idx = tid; idx < H * H; idx += 256) W1S[idx] = W1[idx];
   for (int idx = tid; idx < K * H; idx += 256) WbS[idx] = Wb[idx];
   __syncthreads();
 
-  float dmu = 0.f, dvar = 0.f, dasig_pre = 0.f, a1v = 0.f;
-  if (live) {
-    dmu = drecon[row];
-    const float sig = sigma[row];
-    const float dsig = dmu * eps[row];
-    dvar = dsig / (2.0f * sig);
-    const float ap = asig_pre[row];
-    const float asig = softplusf_(ap);
-    const float dasig = 2.0f * asig * dvar;
-    dasig_pre = dasig * softplus_gradf_(ap);
+  // register accumulators for the block-shared grads
+  float rfmu0 = 0.f, rfsig0 = 0.f, rfmu1 = 0.f, rfsig1 = 0.f;  // k=lane,+64
+  float rwmu = 0.f, rwsig = 0.f;                               // i=lane
+  float rbmu = 0.f, rbsig = 0.f;
 
-    if (lane < H) a1v = a1[(long)row * H + lane];
+  for (int it = 0; it < DEC_ITERS; ++it) {
+    const int row = (blockIdx.x * DEC_ITERS + it) * DEC_RPW + w;
+    const bool live = row < N;
+    float dmu = 0.f, dvar = 0.f, dasig_pre = 0.f, a1v = 0.f;
+    if (live) {
+      dmu = drecon[row];
+      const float sig = sigma[row];
+      const float dsig = dmu * eps[row];
+      dvar = dsig / (2.0f * sig);
+      const float ap = asig_pre[row];
+      const float asig = softplusf_(ap);
+      const float dasig = 2.0f * asig * dvar;
+      dasig_pre = dasig * softplus_gradf_(ap);
 
-    // per-k grads + factor grads
-    for (int k = lane; k < K; k += 64) {
-      const float bv = beta[(long)row * K + k];
-      const float fs = fsig_c[k];
-      const float db = dmu * fmu[k] + dvar * 2.0f * bv * fs * fs;
-      dbeta_out[(long)row * K + k] = db;
-      dbS[w * K + k] = db;
-      atomicAdd(&dfmu[k], dmu * bv);
-      atomicAdd(&dfsig_c[k], dvar * bv * bv * 2.0f * fs);
+      if (lane < H) a1v = a1[(long)row * H + lane];
+
+      // per-k grads + factor-grad register accumulation
+      for (int k = lane; k < K; k += 64) {
+        const float bv = beta[(long)row * K + k];
+        const float fs = fsig_c[k];
+        const float db = dmu * fmu[k] + dvar * 2.0f * bv * fs * fs;
+        dbeta_out[(long)row * K + k] = db;
+        dbS[w * K + k] = db;
+        if (k == lane) {
+          rfmu0 += dmu * bv;
+          rfsig0 += dvar * bv * bv * 2.0f * fs;
+        } else {
+          rfmu1 += dmu * bv;
+          rfsig1 += dvar * bv * bv * 2.0f * fs;
+        }
+      }
+      // alpha-head grads
+      if (lane < H) {
+        const float da1 = dmu * wmu[lane] + dasig_pre * wsig[lane];
+        const float dz = da1 * lrelu_grad_from_out_(a1v);
+        dz1[(long)row * H + lane] = dz;
+        dz1S[w * H + lane] = dz;
+        rwmu += dmu * a1v;
+        rwsig += dasig_pre * a1v;
+      }
+      if (lane == 0) {
+        rbmu += dmu;
+        rbsig += dasig_pre;
+      }
     }
-    // alpha-head grads
-    if (lane < H) {
-      const float da1 = dmu * wmu[lane] + dasig_pre * wsig[lane];
-      const float dz = da1 * lrelu_grad_from_out_(a1v);
-      dz1[(long)row * H + lane] = dz;
-      dz1S[w * H + lane] = dz;
-      atomicAdd(&dwmu[lane], dmu * a1v);
-      atomicAdd(&dwsig[lane], dasig_pre * a1v);
+
+    if (live && lane < H) {
+      // dh[i] = sum_j dz1[j]*W1[j][i] + sum_k dbeta[k]*Wb[k][i]
+      float acc = 0.0f;
+      const float* dz = &dz1S[w * H];
+      for (int j = 0; j < H; ++j)
+        acc = fmaf(dz[j], W1S[(size_t)j * H + lane], acc);
+      const float* db = &dbS[w * K];
+      for (int k = 0; k < K; ++k)
+        acc = fmaf(db[k], WbS[(size_t)k * H + lane], acc);
+      dh[(long)row * H + lane] = acc;
     }
-    if (lane == 0) {
-      atomicAdd(&dbmu[0], dmu);
-      atomicAdd(&dbsig[0], dasig_pre);
-    }
+  }
+
+  // cross-wave reduce of the shared-grad registers, one atomic/element
+  __syncthreads();
+  red[w * 64 + lane] = rfmu0;
+  __syncthreads();
+  if (w == 0 && lane < K && lane < 64)
+    atomicAdd(&dfmu[lane],
+              red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane]);
+  __syncthreads();
+  red[w * 64 + lane] = rfsig0;
+  __syncthreads();
+  if (w == 0 && lane < K && lane < 64)
+    atomicAdd(&dfsig_c[lane],
+              red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane]);
+  if (K > 64) {
+    __syncthreads();
+    red[w * 64 + lane] = kh2 ? rfmu1 : 0.0f;
+    __syncthreads();
+    if (w == 0 && lane + 64 < K)
+      atomicAdd(&dfmu[lane + 64],
+                red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane]);
+    __syncthreads();
+    red[w * 64 + lane] = kh2 ? rfsig1 : 0.0f;
+    __syncthreads();
+    if (w == 0 && lane + 64 < K)
+      atomicAdd(&dfsig_c[lane + 64],
+                red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane]);
   }
   __syncthreads();
-
-  if (live && lane < H) {
-    // dh[i] = sum_j dz1[j]*W1[j][i] + sum_k dbeta[k]*Wb[k][i]
-    float acc = 0.0f;
-    const float* dz = &dz1S[w * H];
-    for (int j = 0; j < H; ++j) acc = fmaf(dz[j], W1S[(size_t)j * H + lane], acc);
-    const float* db = &dbS[w * K];
-    for (int k = 0; k < K; ++k) acc = fmaf(db[k], WbS[(size_t)k * H + lane], acc);
-    dh[(long)row * H + lane] = acc;
+  red[w * 64 + lane] = rwmu;
+  __syncthreads();
+  if (w == 0 && lane < H)
+    atomicAdd(&dwmu[lane],
+              red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane]);
+  __syncthreads();
+  red[w * 64 + lane] = rwsig;
+  __syncthreads();
+  if (w == 0 && lane < H)
+    atomicAdd(&dwsig[lane],
+              red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane]);
+  // scalar bias grads (lane-0 registers only)
+  __syncthreads();
+  if (lane == 0) {
+    red[w] = rbmu;
+    red[4 + w] = rbsig;
   }
-  (void)red;
+  __syncthreads();
+  if (tid == 0) {
+    atomicAdd(&dbmu[0], red[0] + red[1] + red[2] + red[3]);
+    atomicAdd(&dbsig[0], red[4] + red[5] + red[6] + red[7]);
+  }
 }
 
 extern "C" {
@@ -202,10 +270,10 @@ hipError_t fv_dec_bwd(const float* drecon, const float* h, const float* a1,
                       float* dz1, float* dbeta, float* dfmu, float* dfsig_c,
                       float* dwmu, float* dbmu, float* dwsig, float* dbsig,
                       int N, int K, int H, hipStream_t s) {
-  if (H > 64) return hipErrorInvalidValue;
+  if (H > 64 || K > 128) return hipErrorInvalidValue;
   const size_t lds = ((size_t)H * H + (size_t)K * H + DEC_RPW * H +
-                      (size_t)DEC_RPW * K + DEC_RPW) * sizeof(float);
-  dim3 grid((N + DEC_RPW - 1) / DEC_RPW);
+                      (size_t)DEC_RPW * K + 4 * 64) * sizeof(float);
+  dim3 grid((N + DEC_RPW * DEC_ITERS - 1) / (DEC_RPW * DEC_ITERS));
   hipLaunchKernelGGL(dec_bwd_kernel, grid, dim3(256), lds, s,
                      drecon, h, a1, beta, asig_pre, sigma, eps, fmu, fsig_c,
                      W1, wmu, wsig, Wb, dh, dz1, dbeta, dfmu, dfsig_c,

@@ -523,8 +523,10 @@ hipError_t fv_gemm_tn_bf16(const void* A, const void* B, float* out,
   if (r_chunks < 1) r_chunks = 1;
   if (!part) r_chunks = 1;
   if (r_chunks > 1) {
+    // fill the chip: at TBM=TBN=64 the x/y grid is tiny (<=9 blocks for
+    // this model), so the z split is the only parallelism lever
     r_chunks = (R + 511) / 512;
-    if (r_chunks > 32) r_chunks = 32;
+    if (r_chunks > 64) r_chunks = 64;
     if (r_chunks < 1) r_chunks = 1;
   }
   if (r_chunks > 1 && db && !db_part) r_chunks = 1;

@@ -500,8 +500,8 @@ def test_gemm_tn_bf16(R, M, N, chunks):
     A, B = bt(R, M, seed=6, scale=0.1), bt(R, N, seed=7, scale=0.1)
     out = torch.zeros(M, N, device=DEV)
     db = torch.zeros(M, device=DEV)
-    part = torch.zeros(32 * M * N, device=DEV) if chunks > 1 else None
-    db_part = torch.zeros(32 * M, device=DEV) if chunks > 1 else None
+    part = torch.zeros(64 * M * N, device=DEV) if chunks > 1 else None
+    db_part = torch.zeros(64 * M, device=DEV) if chunks > 1 else None
     ext.gemm_tn_bf16(A, B, out, part, chunks, chunks > 1, db, db_part)
     torch.cuda.synchronize()
     ref = A.float().t() @ B.float()
@@ -637,7 +637,8 @@ def test_gemm_nt_fp8(R, Ci, Co):
     assert_close(outb.float(), ref, atol=3e-2 * math.sqrt(Ci), rtol=3e-2,
                  what="gemm_nt_fp8 bf16out")
     # lrelu epilogue + fp8 out
-    out8 = torch.zeros(R, ldp, device=DEV, dtype=torch.float8_e4m3fn)
+    ldo = (Co + 3) & ~3
+    out8 = torch.zeros(R, ldo, device=DEV, dtype=torch.float8_e4m3fn)
     ext.gemm_nt_fp8(A8, W8, b, isw, None, None, out8, R, Ci, Co, 1.0, True)
     torch.cuda.synchronize()
     ref2 = F.leaky_relu(ref, 0.01)
