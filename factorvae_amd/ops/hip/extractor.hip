@@ -19,20 +19,29 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
   if (row >= R) return;
   const float* xr = x + row * C;
 
-  float s = 0.0f;
-  for (int c = lane; c < C; c += 64) s += xr[c];
-  s = wave_reduce_sum(s);
-  s = __shfl(s, 0, 64);
-  const float mu = s / C;
-
-  float v = 0.0f;
-  for (int c = lane; c < C; c += 64) {
-    const float d = xr[c] - mu;
-    v = fmaf(d, d, v);
+  // single pass: per-lane sum and sum-of-squares (keeps x in registers
+  // for the normalize write — one HBM read of x instead of three)
+  float xv[8];  // C <= 512 per lane budget: ceil(C/64) values
+  const int nv = (C - lane + 63) / 64;
+  float s = 0.0f, sq = 0.0f;
+  for (int i = 0; i < nv && i < 8; ++i) {
+    const float xx = xr[lane + i * 64];
+    xv[i] = xx;
+    s += xx;
+    sq = fmaf(xx, xx, sq);
   }
-  v = wave_reduce_sum(v);
-  v = __shfl(v, 0, 64);
-  const float rs = rsqrtf(v / C + eps);
+  for (int c = lane + 8 * 64; c < C; c += 64) {  // spill tail (C > 512)
+    const float xx = xr[c];
+    s += xx;
+    sq = fmaf(xx, xx, sq);
+  }
+  s = wave_reduce_sum(s);
+  sq = wave_reduce_sum(sq);
+  s = __shfl(s, 0, 64);
+  sq = __shfl(sq, 0, 64);
+  const float mu = s / C;
+  const float var = fmaxf(sq / C - mu * mu, 0.0f);
+  const float rs = rsqrtf(var + eps);
 
   if (lane == 0) {
     mean[row] = mu;
@@ -41,8 +50,9 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
   float* o = xln ? xln + row * C : nullptr;
   __bf16* ob = xln_bf ? xln_bf + row * C : nullptr;
   unsigned char* o8 = xln_f8 ? xln_f8 + row * (long)f8_ld : nullptr;
-  for (int c = lane; c < C; c += 64) {
-    const float v_ = fmaf((xr[c] - mu) * rs, gamma[c], beta[c]);
+  for (int c = lane, i = 0; c < C; c += 64, ++i) {
+    const float xx = (i < 8) ? xv[i] : xr[c];
+    const float v_ = fmaf((xx - mu) * rs, gamma[c], beta[c]);
     if (o) o[c] = v_;
     if (ob) ob[c] = (__bf16)v_;
     if (o8) {
