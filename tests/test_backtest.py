@@ -122,3 +122,36 @@ def test_pipeline_end_to_end(tmp_path):
     rep = out["report"]["excess_return_without_cost"]
     assert np.isfinite(rep.loc["information_ratio", "risk"]) or True
     assert len(out["result"].daily_return) > 0
+
+
+def test_score_cli_end_to_end(tmp_path):
+    """python -m factorvae_amd.score: checkpoint -> scores CSV (reference
+    artifact schema) -> backtest report."""
+    import torch
+
+    from factorvae_amd import score as score_mod
+    from factorvae_amd.data.synthetic import make_synthetic_frame
+    from factorvae_amd.models.modules import build_factorvae
+
+    df = make_synthetic_frame(n_days=30, n_stocks=25, seed=9)
+    pkl = tmp_path / "d.pkl"
+    df.to_pickle(pkl)
+    model = build_factorvae(num_latent=158, hidden_size=16, num_portfolio=8,
+                            num_factor=4)
+    ckpt = tmp_path / "m.pt"
+    torch.save(model.state_dict(), ckpt)
+
+    out = score_mod.main([
+        "--checkpoint", str(ckpt), "--dataset", str(pkl),
+        "--run_name", "t", "--num_factor", "4", "--hidden_size", "16",
+        "--num_latent", "158", "--num_portfolio", "8",
+        "--seq_length", "8", "--out_dir", str(tmp_path / "scores"),
+        "--backtest", "--topk", "5", "--n_drop", "2",
+    ])
+    csvs = list((tmp_path / "scores").glob("*.csv"))
+    assert len(csvs) == 1
+    assert csvs[0].name == "t_4_True_False_158_16.csv"
+    import pandas as pd
+    back = pd.read_csv(csvs[0])
+    assert list(back.columns) == ["datetime", "instrument", "score"]
+    assert len(back) == len(out)
