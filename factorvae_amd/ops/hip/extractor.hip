@@ -67,7 +67,7 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
 //   dgamma[c] = sum_r dxln[r][c] * xhat[r][c],  dbeta[c] = sum_r dxln[r][c]
 // xhat recomputed from x, mean, rstd. Grid: (ceil(C/64), ceil(R/LNB_ROWS));
 // 4 waves stripe the row chunk, LDS-reduced, one atomicAdd per column/WG.
-#define LNB_ROWS 256
+#define LNB_ROWS 2048
 __global__ __launch_bounds__(256) void ln_bwd_params_kernel(
     const float* __restrict__ x, const float* __restrict__ dxln,
     const float* __restrict__ mean, const float* __restrict__ rstd,

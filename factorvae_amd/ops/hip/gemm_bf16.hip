@@ -19,6 +19,9 @@
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) short s16x4;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+typedef __attribute__((address_space(3))) s16x4* lds_v4p;
 
 #define BBR 64   // row tile
 #define BBC 64   // col tile
@@ -158,9 +161,10 @@ __global__ __launch_bounds__(256) void gemm_nt_bf16_kernel(
 }
 
 // ---------------------------------------------------------------- NN
-// B (Ci,Co) bf16 row-major: k runs over rows -> dword-coalesced loads of
-// B's rows, transpose-scattered into Bs[n][k] (two b16 LDS stores per
-// dword).
+// B (Ci,Co) bf16 row-major: k runs over rows. The B tile stays
+// row-major in LDS ([k][n], coalesced dword stores) and the fragments
+// are read with ds_read_b64_tr_b16 per the hardware-verified supplier
+// mapping (see the TN kernel below).
 // flags bit3: multiply the result by lrelu'(Y[gr][gc]) — fuses the
 // elementwise lrelu-backward into the producing dgrad GEMM.
 __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
@@ -169,7 +173,7 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
     __bf16* __restrict__ out_bf16, const __bf16* __restrict__ Y,
     int R, int Ci, int Co, float alpha, int flags) {
   __shared__ __bf16 As[2][BBR][BBK + 8];
-  __shared__ __bf16 Bs[2][BBC][BBK + 8];
+  __shared__ __bf16 Bs[2][BBK][BBC + 4];   // row-major [k][n], tr16-read
 
   const int r0 = blockIdx.x * BBR;
   const int c0 = blockIdx.y * BBC;
@@ -178,6 +182,8 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
   const int wv = tid >> 6;
   const int fi = lane & 15;
   const int fk = lane >> 4;
+  const int b_kofs = fi >> 2;       // tr16 supplier: k-row offset
+  const int b_nq = (lane & 3) * 4;  // tr16 supplier: column-quad base
 
   const int KD = BBK / 2;
   f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
@@ -215,10 +221,7 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
       const int row = idx / KD;
       const int cp = (idx % KD) * 2;
       *(unsigned int*)&As[buf][row][cp] = pa[u];
-      dw_bf2 d;
-      d.u = pb[u];
-      Bs[buf][cp][row] = d.h[0];       // transpose scatter
-      Bs[buf][cp + 1][row] = d.h[1];
+      *(unsigned int*)&Bs[buf][row][cp] = pb[u];   // row-major, coalesced
     }
   };
 
@@ -232,9 +235,16 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
 #pragma unroll
     for (int k32 = 0; k32 < BBK; k32 += 32) {
       const bf16x8 a = *(const bf16x8*)&As[buf][wv * 16 + fi][k32 + fk * 8];
+      const int kb = k32 + fk * 8 + b_kofs;
 #pragma unroll
       for (int jt = 0; jt < 4; ++jt) {
-        const bf16x8 b = *(const bf16x8*)&Bs[buf][jt * 16 + fi][k32 + fk * 8];
+        s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4p)&Bs[buf][kb][jt * 16 + b_nq]);
+        s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4p)&Bs[buf][kb + 4][jt * 16 + b_nq]);
+        bf16x8 b;
+        *(bf16x4*)&b = *(bf16x4*)&b0;
+        *(((bf16x4*)&b) + 1) = *(bf16x4*)&b1;
         acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[jt], 0, 0, 0);
       }
     }
@@ -286,9 +296,6 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
 #define TBN 64
 #define TBK 64
 #define TSA (TBM + 4)   // row stride (elems): 8B-aligned, conflict-free
-typedef __attribute__((ext_vector_type(4))) short s16x4;
-typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
-typedef __attribute__((address_space(3))) s16x4* lds_v4p;
 
 __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
