@@ -181,5 +181,9 @@ def gemm_bench():
 
 
 if __name__ == "__main__":
-    main()
-    gemm_bench()
+    import os as _os
+    if _os.environ.get("MICROBENCH_ONLY") == "gemm":
+        gemm_bench()
+    else:
+        main()
+        gemm_bench()
