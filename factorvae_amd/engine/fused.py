@@ -103,6 +103,8 @@ class FusedTrainer:
                                       device=self.device)
             self.wih_bf = torch.empty(C3, self.C, dtype=torch.bfloat16,
                                       device=self.device)
+            self.whh_bf = torch.empty(C3, self.H, dtype=torch.bfloat16,
+                                      device=self.device)
             if self.fp8:
                 ldp = (self.C + 3) & ~3
                 f8 = torch.float8_e4m3fn
@@ -366,8 +368,12 @@ class FusedTrainer:
                        w["mean"], w["rstd"], 1e-5)
             ext.gemm_nt(w["xln"], p("W1x"), p("b1x"), w["xp"], 1.0, False, True)
             ext.gemm_nt(w["xp"], p("Wih"), p("bih"), w["gi"], 1.0, False, False)
-        ext.gru_fwd(w["gi"], p("Whh"), p("bhh"), w["h"], w["h_seq"],
-                    w["h_prev"], w["gates4"], N, T, H)
+        if self.bf16 and H == 64:
+            ext.gru_fwd_mfma(w["gi"], self.whh_bf, p("bhh"), w["h"],
+                             w["h_seq"], w["h_prev"], w["gates4"], N, T, H)
+        else:
+            ext.gru_fwd(w["gi"], p("Whh"), p("bhh"), w["h"], w["h_seq"],
+                        w["h_prev"], w["gates4"], N, T, H)
         if self.bf16 and self.s_side is not None:
             e_ = torch.cuda.Event()
             e_.record(torch.cuda.current_stream(self.device))
@@ -492,7 +498,12 @@ class FusedTrainer:
         ext.gemm_nn(w["dscores"], p("Wenc"), None, w["dh"], 1.0, True, False)
 
         # extractor backward
-        if self.bf16:
+        if self.bf16 and H == 64:
+            ext.gru_bwd_mfma(w["dh"], w["h_prev"], w["gates4"], self.whh_bf,
+                             w["dgi"], w["dgh"], N, T, H,
+                             w["dgi_bf"].view(N, T, 3 * H),
+                             w["dgh_bf"].view(N, T, 3 * H))
+        elif self.bf16:
             ext.gru_bwd(w["dh"], w["h_prev"], w["gates4"], p("Whh"),
                         w["dgi"], w["dgh"], N, T, H,
                         w["dgi_bf"].view(N, T, 3 * H),
@@ -547,6 +558,7 @@ class FusedTrainer:
     def _refresh_bf16_shadows(self):
         self.ext.cast_f32_bf16(self.p("W1x"), self.w1x_bf)
         self.ext.cast_f32_bf16(self.p("Wih"), self.wih_bf)
+        self.ext.cast_f32_bf16(self.p("Whh"), self.whh_bf)
         if self.fp8:
             self.ext.absmax_scale(self.p("W1x"), self.s_w1x, self.is_w1x)
             self.ext.cast_f32_fp8_scaled(self.p("W1x"), self.w1x_f8,

@@ -44,6 +44,12 @@ hipError_t fv_gru_fwd(const float*, const float*, const float*, float*, float*,
 hipError_t fv_gru_bwd(const float*, const float*, const float*, const float*,
                       float*, float*, void*, void*, int, int, int,
                       hipStream_t);
+hipError_t fv_gru_fwd_mfma(const float*, const void*, const float*, float*,
+                           float*, float*, float*, int, int, int,
+                           hipStream_t);
+hipError_t fv_gru_bwd_mfma(const float*, const float*, const float*,
+                           const void*, float*, float*, void*, void*, int,
+                           int, int, hipStream_t);
 hipError_t fv_enc_softmax_fwd(const float*, const float*, float*, float*, int,
                               int, hipStream_t);
 hipError_t fv_enc_softmax_bwd(const float*, const float*, const float*, float*,
@@ -411,6 +417,32 @@ void gru_bwd(torch::Tensor dh_final, torch::Tensor h_prev, torch::Tensor gates4,
                  fpm(dgh), gib, ghb, (int)N, (int)T, (int)H, cur_stream()));
 }
 
+void gru_fwd_mfma(torch::Tensor gi, torch::Tensor whh_bf, torch::Tensor bhh,
+                  torch::Tensor h_final, torch::Tensor h_seq,
+                  torch::Tensor h_prev, torch::Tensor gates4, long N, long T,
+                  long H) {
+  CK(gi); CKB(whh_bf); CK(bhh); CK(h_final); CK(h_seq); CK(h_prev);
+  CK(gates4);
+  RUN(fv_gru_fwd_mfma(fp(gi), bfpc(whh_bf), fp(bhh), fpm(h_final),
+                      fpm(h_seq), fpm(h_prev), fpm(gates4), (int)N, (int)T,
+                      (int)H, cur_stream()));
+}
+
+void gru_bwd_mfma(torch::Tensor dh_final, torch::Tensor h_prev,
+                  torch::Tensor gates4, torch::Tensor whh_bf,
+                  torch::Tensor dgi, torch::Tensor dgh, long N, long T,
+                  long H,
+                  c10::optional<torch::Tensor> dgi_bf = c10::nullopt,
+                  c10::optional<torch::Tensor> dgh_bf = c10::nullopt) {
+  CK(dh_final); CK(h_prev); CK(gates4); CKB(whh_bf); CK(dgi); CK(dgh);
+  void* gib = nullptr; void* ghb = nullptr;
+  if (dgi_bf.has_value()) { CKB(*dgi_bf); gib = bfp(*dgi_bf); }
+  if (dgh_bf.has_value()) { CKB(*dgh_bf); ghb = bfp(*dgh_bf); }
+  RUN(fv_gru_bwd_mfma(fp(dh_final), fp(h_prev), fp(gates4), bfpc(whh_bf),
+                      fpm(dgi), fpm(dgh), gib, ghb, (int)N, (int)T, (int)H,
+                      cur_stream()));
+}
+
 void enc_softmax_fwd(torch::Tensor scores, torch::Tensor y, torch::Tensor a,
                      torch::Tensor yp) {
   CK(scores); CK(y); CK(a); CK(yp);
@@ -632,6 +664,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("cast_f32_fp8_scaled", &cast_f32_fp8_scaled);
   mod.def("ln_bwd_params", &ln_bwd_params);
   mod.def("gru_fwd", &gru_fwd);
+  mod.def("gru_fwd_mfma", &gru_fwd_mfma);
+  mod.def("gru_bwd_mfma", &gru_bwd_mfma, py::arg("dh_final"),
+          py::arg("h_prev"), py::arg("gates4"), py::arg("whh_bf"),
+          py::arg("dgi"), py::arg("dgh"), py::arg("N"), py::arg("T"),
+          py::arg("H"), py::arg("dgi_bf") = py::none(),
+          py::arg("dgh_bf") = py::none());
   mod.def("gru_bwd", &gru_bwd, py::arg("dh_final"), py::arg("h_prev"),
           py::arg("gates4"), py::arg("Whh"), py::arg("dgi"), py::arg("dgh"),
           py::arg("N"), py::arg("T"), py::arg("H"),

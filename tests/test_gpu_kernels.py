@@ -678,3 +678,54 @@ def test_fp8_full_step_vs_fp32():
     assert abs(losses["fp8"] - losses["fp32"]) < 0.1 * (abs(losses["fp32"]) + 1.0)
     cos = torch.nn.functional.cosine_similarity(grads["fp32"], grads["fp8"], dim=0)
     assert float(cos) > 0.97, f"grad cosine {float(cos)}"
+
+
+# ------------------------------------------------------------ MFMA GRU
+@pytest.mark.parametrize("N,T", [(300, 20), (37, 7), (3500, 60)])
+def test_gru_mfma_vs_fp32(N, T):
+    """bf16 MFMA GRU vs the fp32 recurrence kernel (bf16 tolerances)."""
+    H, C = 64, 158
+    torch.manual_seed(30)
+    gru = torch.nn.GRU(C, H, 1, batch_first=True).to(DEV)
+    xp = t(N, T, C, seed=31)
+    gi = torch.empty(N * T, 3 * H, device=DEV)
+    ext.gemm_nt(xp.view(N * T, C), gru.weight_ih_l0.detach(),
+                gru.bias_ih_l0.detach(), gi, 1.0, False, False)
+
+    def run(fn, whh):
+        h_final = torch.empty(N, H, device=DEV)
+        h_seq = torch.empty(N, T, H, device=DEV)
+        h_prev = torch.empty(N, T, H, device=DEV)
+        gates4 = torch.empty(N, T, 4 * H, device=DEV)
+        fn(gi.view(N, T, 3 * H), whh, gru.bias_hh_l0.detach(),
+           h_final, h_seq, h_prev, gates4, N, T, H)
+        torch.cuda.synchronize()
+        return h_final, h_seq, h_prev, gates4
+
+    whh = gru.weight_hh_l0.detach().contiguous()
+    whh_bf = whh.to(torch.bfloat16)
+    hf32, hs32, hp32, g32 = run(ext.gru_fwd, whh)
+    hfm, hsm, hpm, gm = run(ext.gru_fwd_mfma, whh_bf)
+    tol = 0.03 * math.sqrt(T / 20)
+    assert_close(hfm, hf32, atol=tol, rtol=0.05, what="mfma gru h_final")
+    assert_close(hpm, hp32, atol=tol, rtol=0.05, what="mfma gru h_prev")
+
+    # backward: same upstream grad through both
+    dh = t(N, H, seed=32)
+    dgi32 = torch.empty(N, T, 3 * H, device=DEV)
+    dgh32 = torch.empty(N, T, 3 * H, device=DEV)
+    ext.gru_bwd(dh, hp32, g32, whh, dgi32, dgh32, N, T, H)
+    dgim = torch.empty(N, T, 3 * H, device=DEV)
+    dghm = torch.empty(N, T, 3 * H, device=DEV)
+    dgim_bf = torch.empty(N, T, 3 * H, device=DEV, dtype=torch.bfloat16)
+    dghm_bf = torch.empty(N, T, 3 * H, device=DEV, dtype=torch.bfloat16)
+    ext.gru_bwd_mfma(dh, hpm, gm, whh_bf, dgim, dghm, N, T, H,
+                     dgim_bf, dghm_bf)
+    torch.cuda.synchronize()
+    cos = torch.nn.functional.cosine_similarity(
+        dgi32.reshape(-1), dgim.reshape(-1), dim=0)
+    assert float(cos) > 0.995, f"dgi cosine {float(cos)}"
+    cos2 = torch.nn.functional.cosine_similarity(
+        dgh32.reshape(-1), dghm.reshape(-1), dim=0)
+    assert float(cos2) > 0.995, f"dgh cosine {float(cos2)}"
+    assert torch.equal(dgim_bf.float().to(torch.bfloat16), dgim_bf)
