@@ -87,23 +87,41 @@ def checkpoint_path(save_dir: str, run_name: str, num_factor: int,
 
 
 @torch.no_grad()
-def generate_prediction_scores(model, test_dataloader, test_dataset, args):
+def generate_prediction_scores(model, test_dataloader, test_dataset, args,
+                               engine: str = "auto"):
     """Score every test cross-section with model.prediction; returns a
     MultiIndex (datetime, instrument) DataFrame['score'].
 
     Matches the operative scorer of the reference (backtest.ipynb cell 1,
     which slices the label column off the (N,T,C+1) block; the
-    utils.py:70-93 variant is shadowed by it)."""
+    utils.py:70-93 variant is shadowed by it).
+
+    engine="auto" uses the fused HIP predict path on GPU (extractor ->
+    prior predictor -> stochastic decoder, same math incl. the eval-time
+    reparameterized sample); "eager" forces the module path."""
     device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
     model.to(device)
     model.eval()
     ls = []
 
+    trainer = None
+    if engine == "auto" and device.type == "cuda":
+        try:
+            from .engine.fused import FusedTrainer
+
+            trainer = FusedTrainer(model, lr=0.0, t_max=1, device=device,
+                                   use_graph=False, train=False)
+        except Exception:
+            trainer = None
+
     for char_with_label, _ in test_dataloader:
         if char_with_label.shape[1] != args.seq_length:
             continue
         char = char_with_label[:, :, :-1].to(device)
-        predictions = model.prediction(char.float())
+        if trainer is not None:
+            predictions = trainer.predict(char.float())
+        else:
+            predictions = model.prediction(char.float())
         ls.append(predictions.detach().cpu())
 
     ls = torch.cat(ls, dim=0)
