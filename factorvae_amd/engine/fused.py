@@ -682,7 +682,15 @@ class FusedTrainer:
         rng_ok, comm_ok = self._probe_caps()
         key = ("train", N, T)
         if key not in self._graphs:
-            self._capture(key, N, T, rng_ok, comm_ok)
+            try:
+                self._capture(key, N, T, rng_ok, comm_ok)
+            except Exception:
+                # capture unavailable (driver box quirk): permanent
+                # fallback to the eager-launch path — same kernels, same
+                # numerics, just per-launch dispatch cost
+                torch.cuda.synchronize(self.device)
+                self.use_graph = False
+                return self.step(x, y)
         plan = self._graphs[key]
         if not rng_ok:
             self._fill_rng(N)
@@ -756,15 +764,23 @@ class FusedTrainer:
                          torch.empty(N, self.K, device=self.device))
                         for _ in range(G)]
 
-        g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
-            for i, (x, y) in enumerate(days):
-                if rng_bufs is not None:
-                    self.ws["eps"].copy_(rng_bufs[i][0])
-                    if self.training:
-                        self.ws["mask"].copy_(rng_bufs[i][1])
-                self._graph_step_body(x, y, N, T, rng_ok,
-                                      is_distributed() and comm_ok)
+        try:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                for i, (x, y) in enumerate(days):
+                    if rng_bufs is not None:
+                        self.ws["eps"].copy_(rng_bufs[i][0])
+                        if self.training:
+                            self.ws["mask"].copy_(rng_bufs[i][1])
+                    self._graph_step_body(x, y, N, T, rng_ok,
+                                          is_distributed() and comm_ok)
+        except Exception:
+            torch.cuda.synchronize()
+
+            def run_nograph():
+                for x, y in days:
+                    self.step(x, y)
+            return run_nograph, G
 
         def run():
             if rng_bufs is not None:
