@@ -389,16 +389,27 @@ class FusedTrainer:
                           p("bsig_e"), w["fmu"], w["fsig_pre"], w["fsig"],
                           w["fsig_c"])
         ext.attn_qk_fwd(self.p_q, self.p_Wk, self.p_bk, w["qk"], w["cb"])
-        ext.gemm_nt(w["h"], w["qk"], w["cb"], w["s_att"], alpha, False, False)
         mask = w["mask"] if self.training else None
         keep_inv = 1.0 / (1.0 - self.DROPOUT_P)
-        ext.attn_softmax_fwd(w["s_att"], mask, w["a_att"], w["sd"],
-                             w["guard"], keep_inv)
-        ext.gemm_tn(w["a_att"], w["h"], w["u"], w["tn_part_u"], 2, False)
-        ext.attn_ctx_fwd(w["u"], self.p_Wv, self.p_bv, w["guard"], w["ctx"])
-        ext.pred_mlp_fwd(w["ctx"], p("Wl"), p("bl"), p("wmu_p"), p("bmu_p"),
-                         p("wsig_p"), p("bsig_p"), w["hm2"], w["pmu"],
-                         w["psig_pre"], w["psig"], w["psig_c"])
+        if N <= 448:
+            # whole per-head chain in ONE kernel (h staged in LDS)
+            ext.attn_fused_fwd(w["h"], w["qk"], w["cb"], mask, self.p_Wv,
+                               self.p_bv, p("Wl"), p("bl"), p("wmu_p"),
+                               p("bmu_p"), p("wsig_p"), p("bsig_p"),
+                               w["a_att"], w["sd"], w["guard"], w["u"],
+                               w["ctx"], w["hm2"], w["pmu"], w["psig_pre"],
+                               w["psig"], w["psig_c"], alpha, keep_inv)
+        else:
+            ext.gemm_nt(w["h"], w["qk"], w["cb"], w["s_att"], alpha, False,
+                        False)
+            ext.attn_softmax_fwd(w["s_att"], mask, w["a_att"], w["sd"],
+                                 w["guard"], keep_inv)
+            ext.gemm_tn(w["a_att"], w["h"], w["u"], w["tn_part_u"], 2, False)
+            ext.attn_ctx_fwd(w["u"], self.p_Wv, self.p_bv, w["guard"],
+                             w["ctx"])
+            ext.pred_mlp_fwd(w["ctx"], p("Wl"), p("bl"), p("wmu_p"),
+                             p("bmu_p"), p("wsig_p"), p("bsig_p"), w["hm2"],
+                             w["pmu"], w["psig_pre"], w["psig"], w["psig_c"])
         ext.dec_fwd(w["h"], p("W1d"), p("b1d"), p("wmu_d"), p("bmu_d"),
                     p("wsig_d"), p("bsig_d"), p("Wb"), p("bb"), w["fmu"],
                     w["fsig_c"], w["eps"], w["recon"], w["a1"], w["beta"],

@@ -297,7 +297,168 @@ __global__ __launch_bounds__(64) void pred_mlp_bwd_kernel(
   }
 }
 
+// ---------------------------------------------------------------------
+// Fused forward attention megakernel (small/medium N): the whole per-head
+// chain — scores = alpha*(h@qk + c) -> dropout -> relu -> softmax(dim 0)
+// -> u = a^T h -> ctx = Wv u + bv (NaN guard) -> shared MLP heads — as
+// ONE kernel, one workgroup per head, h staged once in LDS (N*(H+1)
+// floats; used when that fits). Replaces 5 launches on the critical
+// path. Writes exactly the buffers the backward chain consumes
+// (a, sd, guard, u, ctx, hm2, pmu, psig*).
+__global__ __launch_bounds__(256) void attn_fused_fwd_kernel(
+    const float* __restrict__ h, const float* __restrict__ qk,
+    const float* __restrict__ cb, const float* __restrict__ mask,
+    const float* __restrict__ Wv, const float* __restrict__ bv,
+    const float* __restrict__ Wl, const float* __restrict__ bl,
+    const float* __restrict__ wmu, const float* __restrict__ bmu,
+    const float* __restrict__ wsig, const float* __restrict__ bsig,
+    float* __restrict__ a_out, float* __restrict__ sd_out,
+    int* __restrict__ guard, float* __restrict__ u_out,
+    float* __restrict__ ctx_out, float* __restrict__ hm2_out,
+    float* __restrict__ pmu, float* __restrict__ psig_pre,
+    float* __restrict__ psig, float* __restrict__ psig_c,
+    int N, int K, int H, float alpha, float keep_inv) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* hS = (float*)smem;              // [N][H+1] (stride 65: no bank dup)
+  float* aS = hS + (size_t)N * (H + 1);  // [N]
+  float* scratch = aS + N;               // [8] block reduces
+  float* part = scratch + 8;             // [4][64]
+  float* qkS = part + 256;               // [64]
+  float* uS = qkS + 64;                  // [64]
+  float* hm2S = uS + 64;                 // [64]
+  __shared__ int bad_s;
+
+  const int k = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int SH = H + 1;
+
+  if (tid == 0) bad_s = 0;
+  if (tid < H) qkS[tid] = qk[(long)k * H + tid];
+  for (int idx = tid; idx < N * H; idx += 256)
+    hS[(idx / H) * SH + (idx % H)] = h[idx];
+  __syncthreads();
+  const float ck = cb[k];
+
+  // scores + dropout + NaN scan + max (thread <-> stock rows)
+  int bad = 0;
+  float mx = -INFINITY;
+  for (int n = tid; n < N; n += 256) {
+    const float* hr = &hS[(size_t)n * SH];
+    float s = 0.0f;
+    for (int c = 0; c < H; ++c) s = fmaf(hr[c], qkS[c], s);
+    s = (s + ck) * alpha;
+    if (mask) s *= mask[(long)n * K + k] * keep_inv;
+    sd_out[(long)n * K + k] = s;
+    aS[n] = s;
+    if (isnan(s) || s == INFINITY) bad = 1;
+    mx = fmaxf(mx, (s > 0.0f) ? s : 0.0f);
+  }
+  if (bad) atomicOr(&bad_s, 1);
+  mx = block_reduce_max(mx, scratch);
+
+  float sum = 0.0f;
+  for (int n = tid; n < N; n += 256) {
+    const float v = aS[n];
+    sum += __expf(((v > 0.0f) ? v : 0.0f) - mx);
+  }
+  sum = block_reduce_sum(sum, scratch);
+  const float inv = 1.0f / sum;
+  for (int n = tid; n < N; n += 256) {
+    const float v = aS[n];
+    const float av = __expf(((v > 0.0f) ? v : 0.0f) - mx) * inv;
+    aS[n] = av;
+    a_out[(long)n * K + k] = av;
+  }
+  __syncthreads();
+  const bool g = bad_s != 0;
+  if (tid == 0) guard[k] = bad_s;
+
+  // u[c] = sum_n a[n] * h[n][c]  (4 waves split the stock range)
+  float acc = 0.0f;
+  if (lane < H) {
+    for (int n = w; n < N; n += 4)
+      acc = fmaf(aS[n], hS[(size_t)n * SH + lane], acc);
+  }
+  part[w * 64 + lane] = acc;
+  __syncthreads();
+  if (w == 0 && lane < H) {
+    const float uv = part[lane] + part[64 + lane] + part[128 + lane] +
+                     part[192 + lane];
+    uS[lane] = uv;
+    u_out[(long)k * H + lane] = uv;
+  }
+  __syncthreads();
+
+  // ctx[j] = Wv[k][j]·u + bv (guarded -> 0); wave per j-chunk, lane <-> i
+  const float ul = (lane < H) ? uS[lane] : 0.0f;
+  const int jpw = (H + 3) / 4;
+  __shared__ float ctxS[64];
+  for (int j = w * jpw; j < min((w + 1) * jpw, H); ++j) {
+    float v = (!g && lane < H)
+                  ? Wv[((long)k * H + j) * H + lane] * ul : 0.0f;
+    v = wave_reduce_sum(v);
+    if (lane == 0) {
+      const float cv = g ? 0.0f : v + bv[(long)k * H + j];
+      ctxS[j] = cv;
+      ctx_out[(long)k * H + j] = cv;
+    }
+  }
+  __syncthreads();
+
+  // shared MLP: hm2 = lrelu(ctx@Wl^T + bl); mu/sig heads
+  const float cl = (lane < H) ? ctxS[lane] : 0.0f;
+  for (int j = w * jpw; j < min((w + 1) * jpw, H); ++j) {
+    float v = (lane < H) ? Wl[(long)j * H + lane] * cl : 0.0f;
+    v = wave_reduce_sum(v);
+    if (lane == 0) {
+      const float z = lrelu_(v + bl[j]);
+      hm2S[j] = z;
+      hm2_out[(long)k * H + j] = z;
+    }
+  }
+  __syncthreads();
+  if (w == 0) {
+    const float z = (lane < H) ? hm2S[lane] : 0.0f;
+    float pm = (lane < H) ? z * wmu[lane] : 0.0f;
+    float ps = (lane < H) ? z * wsig[lane] : 0.0f;
+    pm = wave_reduce_sum(pm);
+    ps = wave_reduce_sum(ps);
+    if (lane == 0) {
+      pmu[k] = pm + bmu[0];
+      const float pre = ps + bsig[0];
+      psig_pre[k] = pre;
+      const float sp = softplusf_(pre);
+      psig[k] = sp;
+      psig_c[k] = (sp == 0.0f) ? 1e-6f : sp;
+    }
+  }
+}
+
 extern "C" {
+
+hipError_t fv_attn_fused_fwd(const float* h, const float* qk, const float* cb,
+                             const float* mask, const float* Wv,
+                             const float* bv, const float* Wl,
+                             const float* bl, const float* wmu,
+                             const float* bmu, const float* wsig,
+                             const float* bsig, float* a, float* sd,
+                             int* guard, float* u, float* ctx, float* hm2,
+                             float* pmu, float* psig_pre, float* psig,
+                             float* psig_c, int N, int K, int H, float alpha,
+                             float keep_inv, hipStream_t s) {
+  if (H > 64) return hipErrorInvalidValue;
+  const size_t lds =
+      ((size_t)N * (H + 1) + N + 8 + 256 + 64 + 64 + 64) * sizeof(float);
+  if (lds > 128 * 1024) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(attn_fused_fwd_kernel, dim3(K), dim3(256), lds, s,
+                     h, qk, cb, mask, Wv, bv, Wl, bl, wmu, bmu, wsig, bsig,
+                     a, sd, guard, u, ctx, hm2, pmu, psig_pre, psig, psig_c,
+                     N, K, H, alpha, keep_inv);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
 
 hipError_t fv_attn_qk_fwd(const float* q, const float* Wk, const float* bk,
                           float* qk, float* c, int K, int H, hipStream_t s) {
@@ -383,3 +544,4 @@ hipError_t fv_pred_mlp_bwd(const float* dpmu, const float* dpsig_c,
 }
 
 }  // extern "C"
+

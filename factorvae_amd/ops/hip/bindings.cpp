@@ -63,6 +63,13 @@ hipError_t fv_enc_heads_bwd(const float*, const float*, const float*,
                             float*, int, int, hipStream_t);
 hipError_t fv_attn_qk_fwd(const float*, const float*, const float*, float*,
                           float*, int, int, hipStream_t);
+hipError_t fv_attn_fused_fwd(const float*, const float*, const float*,
+                             const float*, const float*, const float*,
+                             const float*, const float*, const float*,
+                             const float*, const float*, const float*,
+                             float*, float*, int*, float*, float*, float*,
+                             float*, float*, float*, float*, int, int, int,
+                             float, float, hipStream_t);
 hipError_t fv_attn_softmax_fwd(const float*, const float*, float*, float*,
                                int*, int, int, float, hipStream_t);
 hipError_t fv_attn_ctx_fwd(const float*, const float*, const float*,
@@ -443,6 +450,30 @@ void gru_bwd_mfma(torch::Tensor dh_final, torch::Tensor h_prev,
                       cur_stream()));
 }
 
+void attn_fused_fwd(torch::Tensor h, torch::Tensor qk, torch::Tensor cb,
+                    c10::optional<torch::Tensor> mask, torch::Tensor Wv,
+                    torch::Tensor bv, torch::Tensor Wl, torch::Tensor bl,
+                    torch::Tensor wmu, torch::Tensor bmu, torch::Tensor wsig,
+                    torch::Tensor bsig, torch::Tensor a, torch::Tensor sd,
+                    torch::Tensor guard, torch::Tensor u, torch::Tensor ctx,
+                    torch::Tensor hm2, torch::Tensor pmu,
+                    torch::Tensor psig_pre, torch::Tensor psig,
+                    torch::Tensor psig_c, double alpha, double keep_inv) {
+  CK(h); CK(qk); CK(cb); CK(Wv); CK(bv); CK(Wl); CK(bl); CK(wmu); CK(bmu);
+  CK(wsig); CK(bsig); CK(a); CK(sd); CK(u); CK(ctx); CK(hm2); CK(pmu);
+  CK(psig_pre); CK(psig); CK(psig_c);
+  TORCH_CHECK(guard.scalar_type() == torch::kInt32);
+  const int N = h.size(0), H = h.size(1), K = qk.size(0);
+  const float* mp = nullptr;
+  if (mask.has_value()) { CK(*mask); mp = fp(*mask); }
+  RUN(fv_attn_fused_fwd(fp(h), fp(qk), fp(cb), mp, fp(Wv), fp(bv), fp(Wl),
+                        fp(bl), fp(wmu), fp(bmu), fp(wsig), fp(bsig),
+                        fpm(a), fpm(sd), guard.data_ptr<int>(), fpm(u),
+                        fpm(ctx), fpm(hm2), fpm(pmu), fpm(psig_pre),
+                        fpm(psig), fpm(psig_c), N, K, H, (float)alpha,
+                        (float)keep_inv, cur_stream()));
+}
+
 void enc_softmax_fwd(torch::Tensor scores, torch::Tensor y, torch::Tensor a,
                      torch::Tensor yp) {
   CK(scores); CK(y); CK(a); CK(yp);
@@ -665,6 +696,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ln_bwd_params", &ln_bwd_params);
   mod.def("gru_fwd", &gru_fwd);
   mod.def("gru_fwd_mfma", &gru_fwd_mfma);
+  mod.def("attn_fused_fwd", &attn_fused_fwd);
   mod.def("gru_bwd_mfma", &gru_bwd_mfma, py::arg("dh_final"),
           py::arg("h_prev"), py::arg("gates4"), py::arg("whh_bf"),
           py::arg("dgi"), py::arg("dgh"), py::arg("N"), py::arg("T"),
