@@ -383,8 +383,13 @@ class FusedTrainer:
                                   w["h_prev_bf"].view(-1))
         elif self.bf16:
             ext.cast_f32_bf16(w["h_prev"].view(-1), w["h_prev_bf"].view(-1))
-        ext.gemm_nt(w["h"], p("Wenc"), p("benc"), w["scores_enc"], 1.0, False, False)
-        ext.enc_softmax_fwd(w["scores_enc"], yv, w["a_enc"], w["yp"])
+        if N <= 448:
+            ext.enc_fused_fwd(w["h"], p("Wenc"), p("benc"), yv,
+                              w["scores_enc"], w["a_enc"], w["yp"])
+        else:
+            ext.gemm_nt(w["h"], p("Wenc"), p("benc"), w["scores_enc"], 1.0,
+                        False, False)
+            ext.enc_softmax_fwd(w["scores_enc"], yv, w["a_enc"], w["yp"])
         ext.enc_heads_fwd(w["yp"], p("Wmu_e"), p("bmu_e"), p("Wsig_e"),
                           p("bsig_e"), w["fmu"], w["fsig_pre"], w["fsig"],
                           w["fsig_c"])

@@ -52,6 +52,9 @@ hipError_t fv_gru_bwd_mfma(const float*, const float*, const float*,
                            int, int, hipStream_t);
 hipError_t fv_enc_softmax_fwd(const float*, const float*, float*, float*, int,
                               int, hipStream_t);
+hipError_t fv_enc_fused_fwd(const float*, const float*, const float*,
+                            const float*, float*, float*, float*, int, int,
+                            int, hipStream_t);
 hipError_t fv_enc_softmax_bwd(const float*, const float*, const float*, float*,
                               int, int, hipStream_t);
 hipError_t fv_enc_heads_fwd(const float*, const float*, const float*,
@@ -474,6 +477,15 @@ void attn_fused_fwd(torch::Tensor h, torch::Tensor qk, torch::Tensor cb,
                         (float)keep_inv, cur_stream()));
 }
 
+void enc_fused_fwd(torch::Tensor h, torch::Tensor Wenc, torch::Tensor benc,
+                   torch::Tensor y, torch::Tensor scores, torch::Tensor a,
+                   torch::Tensor yp) {
+  CK(h); CK(Wenc); CK(benc); CK(y); CK(scores); CK(a); CK(yp);
+  const int N = h.size(0), H = h.size(1), M = Wenc.size(0);
+  RUN(fv_enc_fused_fwd(fp(h), fp(Wenc), fp(benc), fp(y), fpm(scores),
+                       fpm(a), fpm(yp), N, M, H, cur_stream()));
+}
+
 void enc_softmax_fwd(torch::Tensor scores, torch::Tensor y, torch::Tensor a,
                      torch::Tensor yp) {
   CK(scores); CK(y); CK(a); CK(yp);
@@ -697,6 +709,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gru_fwd", &gru_fwd);
   mod.def("gru_fwd_mfma", &gru_fwd_mfma);
   mod.def("attn_fused_fwd", &attn_fused_fwd);
+  mod.def("enc_fused_fwd", &enc_fused_fwd);
   mod.def("gru_bwd_mfma", &gru_bwd_mfma, py::arg("dh_final"),
           py::arg("h_prev"), py::arg("gates4"), py::arg("whh_bf"),
           py::arg("dgi"), py::arg("dgh"), py::arg("N"), py::arg("T"),

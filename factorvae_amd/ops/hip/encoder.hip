@@ -36,6 +36,62 @@ __global__ __launch_bounds__(256) void enc_softmax_fwd_kernel(
   if (tid == 0) yp[m] = wy;
 }
 
+
+// Fused encoder forward (small/medium N): scores column GEMV + stock-axis
+// softmax + portfolio return in ONE kernel — one workgroup per portfolio
+// m, h staged in LDS. Folds the (N,H)x(H,M) gemm_nt into the softmax
+// pass (the column read of scores becomes a LDS-resident dot).
+__global__ __launch_bounds__(256) void enc_fused_fwd_kernel(
+    const float* __restrict__ h,      // (N,H)
+    const float* __restrict__ Wenc,   // (M,H)
+    const float* __restrict__ benc,   // (M)
+    const float* __restrict__ y,      // (N,1)
+    float* __restrict__ scores_out,   // (N,M) saved for backward
+    float* __restrict__ a,            // (N,M)
+    float* __restrict__ yp,           // (M)
+    int N, int M, int H) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* hS = (float*)smem;               // [N][H+1]
+  float* sS = hS + (size_t)N * (H + 1);   // [N]
+  float* scratch = sS + N;                // [8]
+  float* wS = scratch + 8;                // [64]
+
+  const int m = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int SH = H + 1;
+
+  if (tid < H) wS[tid] = Wenc[(long)m * H + tid];
+  for (int idx = tid; idx < N * H; idx += 256)
+    hS[(idx / H) * SH + (idx % H)] = h[idx];
+  __syncthreads();
+  const float bm = benc[m];
+
+  float mx = -INFINITY;
+  for (int n = tid; n < N; n += 256) {
+    const float* hr = &hS[(size_t)n * SH];
+    float sv = bm;
+    for (int c = 0; c < H; ++c) sv = fmaf(hr[c], wS[c], sv);
+    sS[n] = sv;
+    scores_out[(long)n * M + m] = sv;
+    mx = fmaxf(mx, sv);
+  }
+  mx = block_reduce_max(mx, scratch);
+
+  float sum = 0.0f;
+  for (int n = tid; n < N; n += 256) sum += __expf(sS[n] - mx);
+  sum = block_reduce_sum(sum, scratch);
+  const float inv = 1.0f / sum;
+
+  float wy = 0.0f;
+  for (int n = tid; n < N; n += 256) {
+    const float an = __expf(sS[n] - mx) * inv;
+    a[(long)n * M + m] = an;
+    wy = fmaf(an, y[n], wy);
+  }
+  wy = block_reduce_sum(wy, scratch);
+  if (tid == 0) yp[m] = wy;
+}
+
 // dyp (M) -> dscores (N,M):  da[n] = dyp[m]*y[n];
 // softmax bwd: ds = a * (da - sum_n a*da)
 __global__ __launch_bounds__(256) void enc_softmax_bwd_kernel(
@@ -125,6 +181,20 @@ __global__ __launch_bounds__(256) void enc_heads_bwd_kernel(
 }
 
 extern "C" {
+
+hipError_t fv_enc_fused_fwd(const float* h, const float* Wenc,
+                            const float* benc, const float* y,
+                            float* scores, float* a, float* yp, int N, int M,
+                            int H, hipStream_t s) {
+  if (H > 64) return hipErrorInvalidValue;
+  const size_t lds = ((size_t)N * (H + 1) + N + 8 + 64) * sizeof(float);
+  if (lds > 128 * 1024) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(enc_fused_fwd_kernel, dim3(M), dim3(256), lds, s,
+                     h, Wenc, benc, y, scores, a, yp, N, M, H);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
 
 hipError_t fv_enc_softmax_fwd(const float* scores, const float* y, float* a,
                               float* yp, int N, int M, hipStream_t stream) {
