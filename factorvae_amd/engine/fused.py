@@ -476,29 +476,52 @@ class FusedTrainer:
                         g("bb"), w["tn_partb_s"])
 
         # predictor MLP + attention backward
-        ext.pred_mlp_bwd(w["dpmu"], w["dpsig_c"], w["psig"], w["psig_pre"],
-                         w["hm2"], p("wmu_p"), p("wsig_p"), w["dz2"],
-                         g("wmu_p"), g("bmu_p"), g("wsig_p"), g("bsig_p"))
-        fork()
-        with _on_side():
-            ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True, g("bl"))
-        ext.gemm_nn(w["dz2"], p("Wl"), None, w["dctx"], 1.0, False, False)
+        mask = w["mask"] if self.training else None
         gWv = self._gstack("Wv.0", (K, H, H))
         gbv = self._gstack("bv.0", (K, H))
-        ext.attn_head_bwd(w["dctx"], w["u"], self.p_Wv, w["guard"], w["du"],
-                          gWv, gbv)
-        ext.gemm_nt(w["h"], w["du"], None, w["da"], 1.0, False, False)
-        mask = w["mask"] if self.training else None
-        ext.attn_softmax_bwd(w["da"], w["a_att"], w["sd"], mask, w["guard"],
-                             w["ds"], w["dc"], keep_inv, alpha)
-        fork()
-        with _on_side():
-            ext.gemm_tn(w["ds"], w["h"], w["dqk"], w["tn_part_s"], 2, False)
-            gq = self._gstack("q_att.0", (K, H))
-            gWk = self._gstack("Wk.0", (K, H, H))
-            gbk = self._gstack("bk.0", (K, H))
-            ext.attn_qk_bwd(w["dqk"], w["dc"], self.p_q, self.p_Wk, self.p_bk,
-                            gq, gWk, gbk)
+        gq = self._gstack("q_att.0", (K, H))
+        gWk = self._gstack("Wk.0", (K, H, H))
+        gbk = self._gstack("bk.0", (K, H))
+        if N <= 384:
+            # whole per-head backward chain in ONE kernel (incl. the
+            # query/key/value wgrads); only the Wl wgrad (cross-head)
+            # stays on the side stream and dh accumulation stays as the
+            # two deterministic GEMMs below
+            ext.attn_fused_bwd(w["dpmu"], w["dpsig_c"], w["psig"],
+                               w["psig_pre"], w["hm2"], p("wmu_p"),
+                               p("wsig_p"), p("Wl"), w["h"], w["a_att"],
+                               w["sd"], mask, w["guard"], w["u"], self.p_Wv,
+                               self.p_q, self.p_Wk, self.p_bk, w["dz2"],
+                               w["du"], w["ds"], w["dc"], gWv, gbv, gq, gWk,
+                               gbk, g("wmu_p"), g("bmu_p"), g("wsig_p"),
+                               g("bsig_p"), alpha, keep_inv)
+            fork()
+            with _on_side():
+                ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True,
+                            g("bl"))
+        else:
+            ext.pred_mlp_bwd(w["dpmu"], w["dpsig_c"], w["psig"],
+                             w["psig_pre"], w["hm2"], p("wmu_p"),
+                             p("wsig_p"), w["dz2"], g("wmu_p"), g("bmu_p"),
+                             g("wsig_p"), g("bsig_p"))
+            fork()
+            with _on_side():
+                ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True,
+                            g("bl"))
+            ext.gemm_nn(w["dz2"], p("Wl"), None, w["dctx"], 1.0, False,
+                        False)
+            ext.attn_head_bwd(w["dctx"], w["u"], self.p_Wv, w["guard"],
+                              w["du"], gWv, gbv)
+            ext.gemm_nt(w["h"], w["du"], None, w["da"], 1.0, False, False)
+            ext.attn_softmax_bwd(w["da"], w["a_att"], w["sd"], mask,
+                                 w["guard"], w["ds"], w["dc"], keep_inv,
+                                 alpha)
+            fork()
+            with _on_side():
+                ext.gemm_tn(w["ds"], w["h"], w["dqk"], w["tn_part_s"], 2,
+                            False)
+                ext.attn_qk_bwd(w["dqk"], w["dc"], self.p_q, self.p_Wk,
+                                self.p_bk, gq, gWk, gbk)
         ext.gemm_nn(w["ds"], w["qk"], None, w["dh"], 1.0, True, False)
         ext.gemm_nn(w["a_att"], w["du"], None, w["dh"], 1.0, True, False)
 

@@ -436,7 +436,206 @@ __global__ __launch_bounds__(256) void attn_fused_fwd_kernel(
   }
 }
 
+
+// Fused backward attention megakernel (N <= ~384): the whole per-head
+// backward chain — shared-MLP bwd, dctx = dz2@Wl, value-path bwd
+// (du + dWv/dbv), da = h@du, softmax/relu/dropout bwd (ds, dc),
+// dqk = ds^T h, and the query/key wgrads — as ONE kernel, one workgroup
+// per head, h and Wl staged in LDS. The two dh accumulations
+// (a_att@du and ds@qk) remain the existing gemm_nn calls, so dh stays
+// deterministic. dwmu/dbmu/dwsig/dbsig use cross-head atomics exactly
+// like pred_mlp_bwd did.
+__global__ __launch_bounds__(256) void attn_fused_bwd_kernel(
+    const float* __restrict__ dpmu, const float* __restrict__ dpsig_c,
+    const float* __restrict__ psig, const float* __restrict__ psig_pre,
+    const float* __restrict__ hm2, const float* __restrict__ wmu,
+    const float* __restrict__ wsig, const float* __restrict__ Wl,
+    const float* __restrict__ h, const float* __restrict__ a,
+    const float* __restrict__ sd, const float* __restrict__ mask,
+    const int* __restrict__ guard, const float* __restrict__ u,
+    const float* __restrict__ Wv, const float* __restrict__ q,
+    const float* __restrict__ Wk, const float* __restrict__ bk,
+    float* __restrict__ dz2_out, float* __restrict__ du_out,
+    float* __restrict__ ds_out, float* __restrict__ dc_out,
+    float* __restrict__ dWv, float* __restrict__ dbv,
+    float* __restrict__ dq, float* __restrict__ dWk,
+    float* __restrict__ dbk, float* __restrict__ dwmu,
+    float* __restrict__ dbmu, float* __restrict__ dwsig,
+    float* __restrict__ dbsig,
+    int N, int K, int H, float alpha, float keep_inv) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* hS = (float*)smem;               // [N][H+1]
+  float* WlS = hS + (size_t)N * (H + 1);  // [H][H+1] (column reads)
+  float* dsS = WlS + (size_t)H * (H + 1); // [N]
+  float* scratch = dsS + N;               // [8]
+  float* part = scratch + 8;              // [4][64]
+  float* dz2S = part + 256;               // [64]
+  float* dcS = dz2S + 64;                 // [64] dctx
+  float* duS = dcS + 64;                  // [64]
+  float* dqkS = duS + 64;                 // [64]
+  float* uS = dqkS + 64;                  // [64]
+
+  const int k = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int SH = H + 1;
+  const bool g = guard[k] != 0;
+
+  for (int idx = tid; idx < N * H; idx += 256)
+    hS[(idx / H) * SH + (idx % H)] = h[idx];
+  for (int idx = tid; idx < H * H; idx += 256)
+    WlS[(idx / H) * SH + (idx % H)] = Wl[idx];
+  if (tid < H) uS[tid] = u[(long)k * H + tid];
+  __syncthreads();
+
+  // ---- shared-MLP backward (pred_mlp_bwd math)
+  const float dm = dpmu[k];
+  const float dsc = (psig[k] == 0.0f) ? 0.0f : dpsig_c[k];
+  const float dsp = dsc * softplus_gradf_(psig_pre[k]);
+  if (tid < H) {
+    const float h2 = hm2[(long)k * H + tid];
+    const float dh2 = dm * wmu[tid] + dsp * wsig[tid];
+    const float dz = dh2 * lrelu_grad_from_out_(h2);
+    dz2S[tid] = dz;
+    dz2_out[(long)k * H + tid] = dz;
+    atomicAdd(&dwmu[tid], dm * h2);
+    atomicAdd(&dwsig[tid], dsp * h2);
+  }
+  if (tid == 0) {
+    atomicAdd(&dbmu[0], dm);
+    atomicAdd(&dbsig[0], dsp);
+  }
+  __syncthreads();
+
+  // ---- dctx[j] = sum_i dz2[i] * Wl[i][j]  (LDS column reads, padded)
+  const int jpw = (H + 3) / 4;
+  for (int j = w * jpw; j < min((w + 1) * jpw, H); ++j) {
+    float v = (lane < H) ? dz2S[lane] * WlS[(size_t)lane * SH + j] : 0.0f;
+    v = wave_reduce_sum(v);
+    if (lane == 0) dcS[j] = g ? 0.0f : v;
+  }
+  __syncthreads();
+
+  // ---- value-path backward: dWv/dbv (+=, head-owned), du
+  float* dW = dWv + (long)k * H * H;
+  for (int idx = tid; idx < H * H; idx += 256)
+    dW[idx] += dcS[idx / H] * uS[idx % H];
+  if (tid < H) dbv[(long)k * H + tid] += dcS[tid];
+  {
+    float acc = 0.0f;
+    if (!g && lane < H) {
+      const float* W = Wv + (long)k * H * H;
+      for (int j = w; j < H; j += 4)
+        acc = fmaf(W[(long)j * H + lane], dcS[j], acc);
+    }
+    part[w * 64 + lane] = acc;
+  }
+  __syncthreads();
+  if (w == 0 && lane < H) {
+    const float d = part[lane] + part[64 + lane] + part[128 + lane] +
+                    part[192 + lane];
+    duS[lane] = d;
+    du_out[(long)k * H + lane] = d;
+  }
+  __syncthreads();
+
+  // ---- da[n] = h[n]·du; softmax/relu/dropout bwd -> ds, dc
+  if (g) {
+    for (int n = tid; n < N; n += 256) {
+      ds_out[(long)n * K + k] = 0.0f;
+      dsS[n] = 0.0f;
+    }
+    if (tid == 0) dc_out[k] = 0.0f;
+    __syncthreads();
+  } else {
+    float t = 0.0f;
+    for (int n = tid; n < N; n += 256) {
+      const float* hr = &hS[(size_t)n * SH];
+      float da = 0.0f;
+      for (int c = 0; c < H; ++c) da = fmaf(hr[c], duS[c], da);
+      dsS[n] = da;                       // stash da
+      t = fmaf(a[(long)n * K + k], da, t);
+    }
+    t = block_reduce_sum(t, scratch);
+    float csum = 0.0f;
+    for (int n = tid; n < N; n += 256) {
+      const long i = (long)n * K + k;
+      float dr = a[i] * (dsS[n] - t);
+      dr = (sd[i] > 0.0f) ? dr : 0.0f;
+      if (mask) dr *= mask[i] * keep_inv;
+      dr *= alpha;
+      dsS[n] = dr;
+      ds_out[i] = dr;
+      csum += dr;
+    }
+    csum = block_reduce_sum(csum, scratch);
+    if (tid == 0) {
+      dc_out[k] = csum;
+      scratch[0] = csum;
+    }
+    __syncthreads();
+  }
+  const float dck = g ? 0.0f : scratch[0];
+
+  // ---- dqk[i] = sum_n ds[n] * h[n][i]
+  {
+    float acc = 0.0f;
+    if (lane < H) {
+      for (int n = w; n < N; n += 4)
+        acc = fmaf(dsS[n], hS[(size_t)n * SH + lane], acc);
+    }
+    __syncthreads();
+    part[w * 64 + lane] = acc;
+  }
+  __syncthreads();
+  if (w == 0 && lane < H)
+    dqkS[lane] = part[lane] + part[64 + lane] + part[128 + lane] +
+                 part[192 + lane];
+  __syncthreads();
+
+  // ---- query/key wgrads (attn_qk_bwd math; += head-owned)
+  const float* qh = q + (long)k * H;
+  float* dWkh = dWk + (long)k * H * H;
+  for (int idx = tid; idx < H * H; idx += 256)
+    dWkh[idx] += qh[idx / H] * dqkS[idx % H];
+  if (tid < H) dbk[(long)k * H + tid] += dck * qh[tid];
+  for (int j = w * jpw; j < min((w + 1) * jpw, H); ++j) {
+    float v = (lane < H)
+                  ? Wk[((long)k * H + j) * H + lane] * dqkS[lane] : 0.0f;
+    v = wave_reduce_sum(v);
+    if (lane == 0)
+      dq[(long)k * H + j] += dck * bk[(long)k * H + j] + v;
+  }
+}
+
 extern "C" {
+
+hipError_t fv_attn_fused_bwd(const float* dpmu, const float* dpsig_c,
+                             const float* psig, const float* psig_pre,
+                             const float* hm2, const float* wmu,
+                             const float* wsig, const float* Wl,
+                             const float* h, const float* a, const float* sd,
+                             const float* mask, const int* guard,
+                             const float* u, const float* Wv, const float* q,
+                             const float* Wk, const float* bk, float* dz2,
+                             float* du, float* ds, float* dc, float* dWv,
+                             float* dbv, float* dq, float* dWk, float* dbk,
+                             float* dwmu, float* dbmu, float* dwsig,
+                             float* dbsig, int N, int K, int H, float alpha,
+                             float keep_inv, hipStream_t s) {
+  if (H > 64) return hipErrorInvalidValue;
+  const size_t lds = ((size_t)N * (H + 1) + (size_t)H * (H + 1) + N + 8 +
+                      256 + 5 * 64) * sizeof(float);
+  if (lds > 128 * 1024) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(attn_fused_bwd_kernel, dim3(K), dim3(256), lds, s,
+                     dpmu, dpsig_c, psig, psig_pre, hm2, wmu, wsig, Wl, h, a,
+                     sd, mask, guard, u, Wv, q, Wk, bk, dz2, du, ds, dc, dWv,
+                     dbv, dq, dWk, dbk, dwmu, dbmu, dwsig, dbsig, N, K, H,
+                     alpha, keep_inv);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
 
 hipError_t fv_attn_fused_fwd(const float* h, const float* qk, const float* cb,
                              const float* mask, const float* Wv,

@@ -66,6 +66,16 @@ hipError_t fv_enc_heads_bwd(const float*, const float*, const float*,
                             float*, int, int, hipStream_t);
 hipError_t fv_attn_qk_fwd(const float*, const float*, const float*, float*,
                           float*, int, int, hipStream_t);
+hipError_t fv_attn_fused_bwd(const float*, const float*, const float*,
+                             const float*, const float*, const float*,
+                             const float*, const float*, const float*,
+                             const float*, const float*, const float*,
+                             const int*, const float*, const float*,
+                             const float*, const float*, const float*,
+                             float*, float*, float*, float*, float*, float*,
+                             float*, float*, float*, float*, float*, float*,
+                             float*, int, int, int, float, float,
+                             hipStream_t);
 hipError_t fv_attn_fused_fwd(const float*, const float*, const float*,
                              const float*, const float*, const float*,
                              const float*, const float*, const float*,
@@ -477,6 +487,34 @@ void attn_fused_fwd(torch::Tensor h, torch::Tensor qk, torch::Tensor cb,
                         (float)keep_inv, cur_stream()));
 }
 
+void attn_fused_bwd(torch::Tensor dpmu, torch::Tensor dpsig_c,
+                    torch::Tensor psig, torch::Tensor psig_pre,
+                    torch::Tensor hm2, torch::Tensor wmu, torch::Tensor wsig,
+                    torch::Tensor Wl, torch::Tensor h, torch::Tensor a,
+                    torch::Tensor sd, c10::optional<torch::Tensor> mask,
+                    torch::Tensor guard, torch::Tensor u, torch::Tensor Wv,
+                    torch::Tensor q, torch::Tensor Wk, torch::Tensor bk,
+                    torch::Tensor dz2, torch::Tensor du, torch::Tensor ds,
+                    torch::Tensor dc, torch::Tensor dWv, torch::Tensor dbv,
+                    torch::Tensor dq, torch::Tensor dWk, torch::Tensor dbk,
+                    torch::Tensor dwmu, torch::Tensor dbmu,
+                    torch::Tensor dwsig, torch::Tensor dbsig,
+                    double alpha, double keep_inv) {
+  CK(h); CK(a); CK(sd); CK(u); CK(Wv); CK(q); CK(Wk); CK(bk); CK(dz2);
+  CK(du); CK(ds); CK(dc); CK(dWv); CK(dbv); CK(dq); CK(dWk); CK(dbk);
+  const int N = h.size(0), H = h.size(1), K = q.size(0);
+  const float* mp = nullptr;
+  if (mask.has_value()) { CK(*mask); mp = fp(*mask); }
+  RUN(fv_attn_fused_bwd(fp(dpmu), fp(dpsig_c), fp(psig), fp(psig_pre),
+                        fp(hm2), fp(wmu), fp(wsig), fp(Wl), fp(h), fp(a),
+                        fp(sd), mp, guard.data_ptr<int>(), fp(u), fp(Wv),
+                        fp(q), fp(Wk), fp(bk), fpm(dz2), fpm(du), fpm(ds),
+                        fpm(dc), fpm(dWv), fpm(dbv), fpm(dq), fpm(dWk),
+                        fpm(dbk), fpm(dwmu), fpm(dbmu), fpm(dwsig),
+                        fpm(dbsig), N, K, H, (float)alpha, (float)keep_inv,
+                        cur_stream()));
+}
+
 void enc_fused_fwd(torch::Tensor h, torch::Tensor Wenc, torch::Tensor benc,
                    torch::Tensor y, torch::Tensor scores, torch::Tensor a,
                    torch::Tensor yp) {
@@ -710,6 +748,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gru_fwd_mfma", &gru_fwd_mfma);
   mod.def("attn_fused_fwd", &attn_fused_fwd);
   mod.def("enc_fused_fwd", &enc_fused_fwd);
+  mod.def("attn_fused_bwd", &attn_fused_bwd);
   mod.def("gru_bwd_mfma", &gru_bwd_mfma, py::arg("dh_final"),
           py::arg("h_prev"), py::arg("gates4"), py::arg("whh_bf"),
           py::arg("dgi"), py::arg("dgh"), py::arg("N"), py::arg("T"),
