@@ -326,6 +326,7 @@ __global__ __launch_bounds__(256) void attn_fused_fwd_kernel(
   float* qkS = part + 256;               // [64]
   float* uS = qkS + 64;                  // [64]
   float* hm2S = uS + 64;                 // [64]
+  float* WS = hm2S + 64;                 // [H][H+1]: Wv then Wl staged
   __shared__ int bad_s;
 
   const int k = blockIdx.x;
@@ -391,13 +392,18 @@ __global__ __launch_bounds__(256) void attn_fused_fwd_kernel(
   }
   __syncthreads();
 
-  // ctx[j] = Wv[k][j]·u + bv (guarded -> 0); wave per j-chunk, lane <-> i
+  // ctx[j] = Wv[k][j]·u + bv (guarded -> 0); Wv staged in LDS first
+  // (a cooperative coalesced load beats 16 latency-serialized global
+  // row reads under the wave_reduce chain)
+  const int SW = H + 1;
+  for (int idx = tid; idx < H * H; idx += 256)
+    WS[(idx / H) * SW + (idx % H)] = Wv[(long)k * H * H + idx];
+  __syncthreads();
   const float ul = (lane < H) ? uS[lane] : 0.0f;
   const int jpw = (H + 3) / 4;
   __shared__ float ctxS[64];
   for (int j = w * jpw; j < min((w + 1) * jpw, H); ++j) {
-    float v = (!g && lane < H)
-                  ? Wv[((long)k * H + j) * H + lane] * ul : 0.0f;
+    float v = (!g && lane < H) ? WS[(size_t)j * SW + lane] * ul : 0.0f;
     v = wave_reduce_sum(v);
     if (lane == 0) {
       const float cv = g ? 0.0f : v + bv[(long)k * H + j];
@@ -407,10 +413,13 @@ __global__ __launch_bounds__(256) void attn_fused_fwd_kernel(
   }
   __syncthreads();
 
-  // shared MLP: hm2 = lrelu(ctx@Wl^T + bl); mu/sig heads
+  // shared MLP: hm2 = lrelu(ctx@Wl^T + bl); Wl staged over Wv's slot
+  for (int idx = tid; idx < H * H; idx += 256)
+    WS[(idx / H) * SW + (idx % H)] = Wl[idx];
+  __syncthreads();
   const float cl = (lane < H) ? ctxS[lane] : 0.0f;
   for (int j = w * jpw; j < min((w + 1) * jpw, H); ++j) {
-    float v = (lane < H) ? Wl[(long)j * H + lane] * cl : 0.0f;
+    float v = (lane < H) ? WS[(size_t)j * SW + lane] * cl : 0.0f;
     v = wave_reduce_sum(v);
     if (lane == 0) {
       const float z = lrelu_(v + bl[j]);
@@ -518,16 +527,21 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_kernel(
   __syncthreads();
 
   // ---- value-path backward: dWv/dbv (+=, head-owned), du
+  // stage Wv[k] over the Wl slot (Wl no longer needed)
   float* dW = dWv + (long)k * H * H;
-  for (int idx = tid; idx < H * H; idx += 256)
+  const float* Wvk = Wv + (long)k * H * H;
+  for (int idx = tid; idx < H * H; idx += 256) {
+    const float wv_ = Wvk[idx];
+    WlS[(idx / H) * SH + (idx % H)] = wv_;
     dW[idx] += dcS[idx / H] * uS[idx % H];
+  }
   if (tid < H) dbv[(long)k * H + tid] += dcS[tid];
+  __syncthreads();
   {
     float acc = 0.0f;
     if (!g && lane < H) {
-      const float* W = Wv + (long)k * H * H;
       for (int j = w; j < H; j += 4)
-        acc = fmaf(W[(long)j * H + lane], dcS[j], acc);
+        acc = fmaf(WlS[(size_t)j * SH + lane], dcS[j], acc);
     }
     part[w * 64 + lane] = acc;
   }
@@ -594,15 +608,18 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_kernel(
                  part[192 + lane];
   __syncthreads();
 
-  // ---- query/key wgrads (attn_qk_bwd math; += head-owned)
+  // ---- query/key wgrads (attn_qk_bwd math; += head-owned); Wk staged
   const float* qh = q + (long)k * H;
+  const float* Wkk = Wk + (long)k * H * H;
   float* dWkh = dWk + (long)k * H * H;
-  for (int idx = tid; idx < H * H; idx += 256)
+  for (int idx = tid; idx < H * H; idx += 256) {
+    WlS[(idx / H) * SH + (idx % H)] = Wkk[idx];
     dWkh[idx] += qh[idx / H] * dqkS[idx % H];
+  }
   if (tid < H) dbk[(long)k * H + tid] += dck * qh[tid];
+  __syncthreads();
   for (int j = w * jpw; j < min((w + 1) * jpw, H); ++j) {
-    float v = (lane < H)
-                  ? Wk[((long)k * H + j) * H + lane] * dqkS[lane] : 0.0f;
+    float v = (lane < H) ? WlS[(size_t)j * SH + lane] * dqkS[lane] : 0.0f;
     v = wave_reduce_sum(v);
     if (lane == 0)
       dq[(long)k * H + j] += dck * bk[(long)k * H + j] + v;
@@ -648,9 +665,9 @@ hipError_t fv_attn_fused_fwd(const float* h, const float* qk, const float* cb,
                              float* psig_c, int N, int K, int H, float alpha,
                              float keep_inv, hipStream_t s) {
   if (H > 64) return hipErrorInvalidValue;
-  const size_t lds =
-      ((size_t)N * (H + 1) + N + 8 + 256 + 64 + 64 + 64) * sizeof(float);
-  if (lds > 128 * 1024) return hipErrorInvalidValue;
+  const size_t lds = ((size_t)N * (H + 1) + N + 8 + 256 + 3 * 64 +
+                      (size_t)H * (H + 1)) * sizeof(float);
+  if (lds > 144 * 1024) return hipErrorInvalidValue;
   hipLaunchKernelGGL(attn_fused_fwd_kernel, dim3(K), dim3(256), lds, s,
                      h, qk, cb, mask, Wv, bv, Wl, bl, wmu, bmu, wsig, bsig,
                      a, sd, guard, u, ctx, hm2, pmu, psig_pre, psig, psig_c,

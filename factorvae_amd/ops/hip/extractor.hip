@@ -67,18 +67,21 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
 //   dgamma[c] = sum_r dxln[r][c] * xhat[r][c],  dbeta[c] = sum_r dxln[r][c]
 // xhat recomputed from x, mean, rstd. Grid: (ceil(C/64), ceil(R/LNB_ROWS));
 // 4 waves stripe the row chunk, LDS-reduced, one atomicAdd per column/WG.
-#define LNB_ROWS 2048
+// rows per block chosen by the launcher: enough blocks to fill the
+// chip, few enough that the one-atomic-per-column-per-block flush stays
+// cheap at large R
 __global__ __launch_bounds__(256) void ln_bwd_params_kernel(
     const float* __restrict__ x, const float* __restrict__ dxln,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    float* __restrict__ dgamma, float* __restrict__ dbeta, long R, int C) {
+    float* __restrict__ dgamma, float* __restrict__ dbeta, long R, int C,
+    int rows_per_block) {
   __shared__ float pg[4][64];
   __shared__ float pb[4][64];
   const int lane = threadIdx.x & 63;
   const int w = threadIdx.x >> 6;
   const int c = blockIdx.x * 64 + lane;
-  const long rbeg = (long)blockIdx.y * LNB_ROWS;
-  const long rend = min(rbeg + LNB_ROWS, R);
+  const long rbeg = (long)blockIdx.y * rows_per_block;
+  const long rend = min(rbeg + (long)rows_per_block, R);
   float dg = 0.0f, db = 0.0f;
   if (c < C) {
     for (long r = rbeg + w; r < rend; r += 4) {
@@ -116,9 +119,15 @@ hipError_t fv_ln_bwd_params(const float* x, const float* dxln,
                             float* dgamma, float* dbeta, long R, int C,
                             int r_chunks, hipStream_t stream) {
   (void)r_chunks;
-  dim3 grid((C + 63) / 64, (unsigned)((R + LNB_ROWS - 1) / LNB_ROWS));
+  // ~400 blocks fills the chip; cap atomics at large R with bigger rows
+  const int cblocks = (C + 63) / 64;
+  long target_y = (400 + cblocks - 1) / cblocks;
+  long rpb = (R + target_y - 1) / target_y;
+  if (rpb < 64) rpb = 64;
+  if (rpb > 8192) rpb = 8192;
+  dim3 grid(cblocks, (unsigned)((R + rpb - 1) / rpb));
   hipLaunchKernelGGL(ln_bwd_params_kernel, grid, dim3(256), 0, stream,
-                     x, dxln, mean, rstd, dgamma, dbeta, R, C);
+                     x, dxln, mean, rstd, dgamma, dbeta, R, C, (int)rpb);
   HIP_CHECK_LAST();
   return hipSuccess;
 }
