@@ -708,14 +708,20 @@ class FusedTrainer:
         w["y"].copy_(y.view(N, 1))
 
         if not self.use_graph:
+            from ..observability import roctx_range
+
             self._fill_rng(N)
             self.grads.zero_()
-            self._launch_forward(N, T)
-            self._launch_backward(N, T)
+            with roctx_range("fv_forward"):
+                self._launch_forward(N, T)
+            with roctx_range("fv_backward"):
+                self._launch_backward(N, T)
             if is_distributed():
-                self.grads.div_(get_world_size())
-                torch.distributed.all_reduce(self.grads)
-            self._launch_optimizer()
+                with roctx_range("fv_allreduce"):
+                    self.grads.div_(get_world_size())
+                    torch.distributed.all_reduce(self.grads)
+            with roctx_range("fv_adam"):
+                self._launch_optimizer()
             return w["loss"]
 
         rng_ok, comm_ok = self._probe_caps()
