@@ -131,6 +131,12 @@ class FusedTrainer:
         # the dependencies are recorded into the captured hipGraph)
         self.s_side = (torch.cuda.Stream(device=self.device)
                        if self.device.type == "cuda" else None)
+        # second side stream: the three big extractor wgrad reductions
+        # run here so they overlap BOTH the main path and the other
+        # (attention/encoder) wgrads — at A-share shapes the side work
+        # exceeds the main path, so one side stream becomes the bottleneck
+        self.s_side2 = (torch.cuda.Stream(device=self.device)
+                        if self.device.type == "cuda" else None)
 
     # ---------------------------------------------------------------- params
     _STACKED = ("q_att", "Wk", "bk", "Wv", "bv")
@@ -374,11 +380,11 @@ class FusedTrainer:
         else:
             ext.gru_fwd(w["gi"], p("Whh"), p("bhh"), w["h"], w["h_seq"],
                         w["h_prev"], w["gates4"], N, T, H)
-        if self.bf16 and self.s_side is not None:
+        if self.bf16 and self.s_side2 is not None:
             e_ = torch.cuda.Event()
             e_.record(torch.cuda.current_stream(self.device))
-            self.s_side.wait_event(e_)
-            with torch.cuda.stream(self.s_side):
+            self.s_side2.wait_event(e_)
+            with torch.cuda.stream(self.s_side2):
                 ext.cast_f32_bf16(w["h_prev"].view(-1),
                                   w["h_prev_bf"].view(-1))
         elif self.bf16:
@@ -439,25 +445,28 @@ class FusedTrainer:
         chunks = max(1, min(32, R // 1024))
 
         main = torch.cuda.current_stream(self.device) if self.s_side else None
-        side = self.s_side
+        sides = (self.s_side, self.s_side2)
 
-        def fork():
-            """Make the side stream wait for everything issued on main."""
-            if side is None:
+        def fork(si=0):
+            """Make side stream si wait for everything issued on main."""
+            if sides[si] is None:
                 return
             e = torch.cuda.Event()
             e.record(main)
-            side.wait_event(e)
+            sides[si].wait_event(e)
 
         class _on_side:
+            def __init__(_s, si=0):
+                _s.si = si
+
             def __enter__(_s):
-                if side is not None:
-                    _s.ctx = torch.cuda.stream(side)
+                if sides[_s.si] is not None:
+                    _s.ctx = torch.cuda.stream(sides[_s.si])
                     _s.ctx.__enter__()
                 return _s
 
             def __exit__(_s, *a):
-                if side is not None:
+                if sides[_s.si] is not None:
                     _s.ctx.__exit__(*a)
 
         ext.loss_bwd(w["recon"], yv, w["fmu"], w["fsig_c"], w["pmu"],
@@ -551,8 +560,8 @@ class FusedTrainer:
             ext.gru_bwd(w["dh"], w["h_prev"], w["gates4"], p("Whh"),
                         w["dgi"], w["dgh"], N, T, H)
         if self.bf16:
-            fork()
-            with _on_side():
+            fork(1)
+            with _on_side(1):
                 ext.gemm_tn_bf16(w["dgh_bf"].view(R, 3 * H), w["h_prev_bf"],
                                  g("Whh"), w["tn_part"], chunks, True,
                                  g("bhh"), w["tn_partb"])
@@ -562,16 +571,16 @@ class FusedTrainer:
             ext.gemm_nn_bf16(w["dgi_bf"].view(R, 3 * H), self.wih_bf, None,
                              None, w["dzx_bf"], 1.0, False, False,
                              w["xp_bf"])
-            fork()
-            with _on_side():
+            fork(1)
+            with _on_side(1):
                 ext.gemm_tn_bf16(w["dzx_bf"], w["xln_bf"], g("W1x"),
                                  w["tn_part3"], chunks, True, g("b1x"),
                                  w["tn_partb3"])
             ext.gemm_nn_bf16(w["dzx_bf"], self.w1x_bf, None, w["dxln"],
                              None, 1.0, False, False)
         else:
-            fork()
-            with _on_side():
+            fork(1)
+            with _on_side(1):
                 ext.gemm_tn(w["dgh"].view(R, 3 * H), w["h_prev"].view(R, H),
                             g("Whh"), w["tn_part"], chunks, True,
                             g("bhh"), w["tn_partb"])
@@ -581,18 +590,21 @@ class FusedTrainer:
             ext.gemm_nn(w["dgi"].view(R, 3 * H), p("Wih"), None, w["dxp"], 1.0,
                         False, False)
             ext.lrelu_bwd(w["dxp"], w["xp"], w["dzx"])
-            fork()
-            with _on_side():
+            fork(1)
+            with _on_side(1):
                 ext.gemm_tn(w["dzx"], w["xln"], g("W1x"), w["tn_part3"], chunks,
                             True, g("b1x"), w["tn_partb3"])
             ext.gemm_nn(w["dzx"], p("W1x"), None, w["dxln"], 1.0, False, False)
-        ext.ln_bwd_params(x2d, w["dxln"], w["mean"], w["rstd"],
-                          g("ln_g"), g("ln_b"), chunks)
+        fork()
+        with _on_side():
+            ext.ln_bwd_params(x2d, w["dxln"], w["mean"], w["rstd"],
+                              g("ln_g"), g("ln_b"), chunks)
         # join: main waits for all side-stream wgrad work
-        if side is not None:
-            e = torch.cuda.Event()
-            e.record(side)
-            main.wait_event(e)
+        for sstream in sides:
+            if sstream is not None:
+                e = torch.cuda.Event()
+                e.record(sstream)
+                main.wait_event(e)
 
     def _refresh_bf16_shadows(self):
         self.ext.cast_f32_bf16(self.p("W1x"), self.w1x_bf)

@@ -119,9 +119,10 @@ hipError_t fv_ln_bwd_params(const float* x, const float* dxln,
                             float* dgamma, float* dbeta, long R, int C,
                             int r_chunks, hipStream_t stream) {
   (void)r_chunks;
-  // ~400 blocks fills the chip; cap atomics at large R with bigger rows
+  // ~2000 blocks: this reduction is latency-bound per block (two global
+  // reads per row-iteration), so parallelism beats atomic thrift
   const int cblocks = (C + 63) / 64;
-  long target_y = (400 + cblocks - 1) / cblocks;
+  long target_y = (2048 + cblocks - 1) / cblocks;
   long rpb = (R + target_y - 1) / target_y;
   if (rpb < 64) rpb = 64;
   if (rpb > 8192) rpb = 8192;
