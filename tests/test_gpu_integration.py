@@ -1,0 +1,86 @@
+"""GPU integration tests: the production train_main driver end-to-end on
+the fused HIP engine, and scoring throughput vs the reference's 6.54
+cross-sections/s baseline."""
+
+import os
+import time
+from types import SimpleNamespace
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = torch.device("cuda:0") if torch.cuda.is_available() else None
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("dtype", ["fp32", "bf16"])
+def test_train_main_fused_end_to_end(tmp_path, dtype):
+    """Full driver: synthetic pickle -> device epoch cache -> fused
+    hipGraph engine -> best-val checkpoint + side-car; loss finite and
+    improving."""
+    import json
+
+    from factorvae_amd.data.synthetic import make_synthetic_frame
+    from factorvae_amd.engine.trainer import train_main
+    from factorvae_amd.utils import DataArgument, checkpoint_path
+
+    df = make_synthetic_frame(n_days=60, n_stocks=64, seed=11,
+                              signal_strength=0.5)
+    args = SimpleNamespace(
+        num_epochs=4, lr=1e-3, num_latent=158, num_portfolio=32,
+        seq_len=8, num_factor=8, hidden_size=64, seed=0,
+        run_name=f"gpu_e2e_{dtype}", save_dir=str(tmp_path), dataset=None,
+        engine="fused", dtype=dtype, wandb=False, resume=False,
+    )
+    data_args = DataArgument(
+        start_time="2015-01-01", end_time="2015-12-31",
+        fit_end_time="2015-03-01", val_start_time="2015-03-02",
+        val_end_time="2015-03-25", seq_len=8,
+    )
+    best = train_main(args, data_args, df=df)
+    assert best == best and best < 1e6  # finite
+    ckpt = checkpoint_path(str(tmp_path), args.run_name, 8, 64, 32, 0)
+    assert os.path.exists(ckpt) and os.path.exists(ckpt + ".opt")
+    # metrics JSONL written with the reference's key names
+    mpath = os.path.join(str(tmp_path), f"{args.run_name}_metrics.jsonl")
+    recs = [json.loads(l) for l in open(mpath)]
+    assert any("Train Loss" in r for r in recs)
+    # checkpoint round-trips into a fresh reference-API model
+    from factorvae_amd.models.modules import build_factorvae
+    m2 = build_factorvae(num_latent=158, hidden_size=64, num_portfolio=32,
+                         num_factor=8)
+    m2.load_state_dict(torch.load(ckpt, map_location="cpu",
+                                  weights_only=True))
+
+
+@pytest.mark.timeout(300)
+def test_scoring_throughput_beats_reference():
+    """generate_prediction_scores (fused predict path) must beat the
+    reference's published 6.54 cross-sections/s by a wide margin."""
+    from factorvae_amd.data.sampler import init_data_loader
+    from factorvae_amd.data.synthetic import make_synthetic_frame
+    from factorvae_amd.models.modules import build_factorvae
+    from factorvae_amd.utils import generate_prediction_scores, test_args
+
+    df = make_synthetic_frame(n_days=120, n_stocks=300, seed=4)
+    model = build_factorvae(num_latent=158, hidden_size=64,
+                            num_portfolio=128, num_factor=20).to(DEV)
+    loader = init_data_loader(df, step_len=20, shuffle=False,
+                              start=None, end=None)
+    args = test_args(run_name="t", num_factor=20, hidden_size=64,
+                     num_latent=158, num_portfolio=128, seq_length=20)
+    # warm up (engine build + graphless predict)
+    generate_prediction_scores(model, loader, loader.dataset, args)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    scores = generate_prediction_scores(model, loader, loader.dataset, args)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    n_days = df.index.get_level_values(0).nunique()
+    rate = n_days / dt
+    print(f"\nscoring throughput: {rate:.1f} cross-sections/s "
+          f"({n_days} days in {dt:.2f}s)")
+    assert len(scores) > 0
+    assert rate > 6.54 * 5, f"only {rate:.1f} cs/s"
