@@ -294,7 +294,7 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
 // falls out of the staging registers for free.
 #define TBM 64
 #define TBN 64
-#define TBK 128
+#define TBK 64
 #define TSA (TBM + 4)   // row stride (elems): 8B-aligned, conflict-free
 
 __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
@@ -336,21 +336,21 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
 
   const int span = rend - rbeg;
   const int ktiles = (span + TBK - 1) / TBK;
-  unsigned int pa[16], pb[16];
+  unsigned int pa[8], pb[8];
 
   auto stage_regs = [&](int r0_) {
     const bool interior = (r0_ + TBK <= rend) && (m0 + TBM <= M) &&
                           (n0 + TBN <= N);
     if (interior) {
 #pragma unroll
-      for (int u = 0; u < 16; ++u) {
+      for (int u = 0; u < 8; ++u) {
         const long gr = (long)r0_ + s_kr0 + 8 * u;
         pa[u] = *(const unsigned int*)(A + gr * M + m0 + s_cp);
         pb[u] = *(const unsigned int*)(B + gr * N + n0 + s_cp);
       }
     } else {
 #pragma unroll
-      for (int u = 0; u < 16; ++u) {
+      for (int u = 0; u < 8; ++u) {
         const long gr = (long)r0_ + s_kr0 + 8 * u;
         pa[u] = load_dw_guard(A, gr, m0 + s_cp, rend, M, M);
         pb[u] = load_dw_guard(B, gr, n0 + s_cp, rend, N, N);
@@ -359,7 +359,7 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
   };
   auto regs_to_lds = [&](int buf) {
 #pragma unroll
-    for (int u = 0; u < 16; ++u) {
+    for (int u = 0; u < 8; ++u) {
       const int kr = s_kr0 + 8 * u;
       *(unsigned int*)&As[buf][kr][s_cp] = pa[u];
       *(unsigned int*)&Bs[buf][kr][s_cp] = pb[u];
@@ -367,7 +367,7 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
   };
   auto bias_from_regs = [&]() {
 #pragma unroll
-    for (int u = 0; u < 16; ++u) {
+    for (int u = 0; u < 8; ++u) {
       dw_bf2 d;
       d.u = pa[u];
       bsum0 += (float)d.h[0];
