@@ -128,7 +128,12 @@ class TSDataSampler:
 
     def __getitem__(self, idx):
         if isinstance(idx, (list, np.ndarray)):
-            rows = np.asarray([self._resolve_pos(i) for i in idx], dtype=np.int64)
+            pos = np.asarray(idx, dtype=np.int64) + self.start_idx
+            if len(pos) and (pos.min() < self.start_idx or
+                             pos.max() >= self.end_idx):
+                raise KeyError("index out of bounds")
+            rows = (self.flt_rows[pos] if self.flt_rows is not None
+                    else pos)
             ids = self._window_ids(rows)
             data = self.data_arr[ids]                      # (B, T, C+1)
             actual = self.data.index[rows]
@@ -152,6 +157,15 @@ class TSDatasetH(Dataset):
 
     def __getitem__(self, idx):
         return self.sampler[idx]
+
+    def __getitems__(self, indices):
+        """torch DataLoader batched-fetch protocol: gather the whole
+        day-batch with ONE vectorized sampler call (the per-sample path
+        pays pandas MultiIndex costs per row — ~15 days/s; this path
+        does ~500+ days/s). Returns a single pre-batched element that
+        custom_collate_fn recognizes."""
+        data, actual = self.sampler[list(indices)]
+        return [(data, actual)]
 
     def __len__(self):
         return len(self.sampler)
@@ -211,7 +225,11 @@ class DateGroupedBatchSampler(Sampler):
 
 def custom_collate_fn(batch):
     """(data, MultiIndex) pairs -> (tensor (N,T,C+1), list of index lists)
-    (/root/reference/dataset.py:242-249)."""
+    (/root/reference/dataset.py:242-249). Also accepts the pre-batched
+    single element produced by TSDatasetH.__getitems__."""
+    if len(batch) == 1 and getattr(batch[0][0], "ndim", 0) == 3:
+        data, index = batch[0]
+        return torch.as_tensor(data), [list(index)]
     data, indices = zip(*batch)
     data = torch.utils.data.dataloader.default_collate(data)
     indices = [list(index) for index in indices]
