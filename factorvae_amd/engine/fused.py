@@ -318,6 +318,13 @@ class FusedTrainer:
         w["tn_part_u"] = f(small_mn)
         w["tn_part_s"] = f(small_mn)
         w["tn_partb_s"] = f(32 * max_m)
+        # deterministic shared-grad partials (no float atomics anywhere)
+        w["dec_part"] = f(((N + 31) // 32 + 1) * (2 * K + 2 * H + 2))
+        w["hpart"] = f(K * (2 * H + 2))
+        cb = (C + 63) // 64
+        ty = (2048 + cb - 1) // cb
+        rpb = max(64, (R + ty - 1) // ty)
+        w["ln_part"] = f(((R + rpb - 1) // rpb + 1) * 2 * C)
         w["dzx"] = f(R, C)
         w["dxln"] = f(R, C)
         if self.bf16:
@@ -475,8 +482,8 @@ class FusedTrainer:
         ext.dec_bwd(w["drecon"], w["h"], w["a1"], w["beta"], w["asig_pre"],
                     w["sigma"], w["eps"], w["fmu"], w["fsig_c"], p("W1d"),
                     p("wmu_d"), p("wsig_d"), p("Wb"), w["dh"], w["dz1"],
-                    w["dbeta"], w["dfmu"], w["dfsig_c"], g("wmu_d"),
-                    g("bmu_d"), g("wsig_d"), g("bsig_d"))
+                    w["dbeta"], w["dec_part"], w["dfmu"], w["dfsig_c"],
+                    g("wmu_d"), g("bmu_d"), g("wsig_d"), g("bsig_d"))
         fork()
         with _on_side():
             ext.gemm_tn(w["dz1"], w["h"], g("W1d"), w["tn_part_s"], 2, True,
@@ -502,8 +509,8 @@ class FusedTrainer:
                                w["sd"], mask, w["guard"], w["u"], self.p_Wv,
                                self.p_q, self.p_Wk, self.p_bk, w["dz2"],
                                w["du"], w["ds"], w["dc"], gWv, gbv, gq, gWk,
-                               gbk, g("wmu_p"), g("bmu_p"), g("wsig_p"),
-                               g("bsig_p"), alpha, keep_inv)
+                               gbk, w["hpart"], g("wmu_p"), g("bmu_p"),
+                               g("wsig_p"), g("bsig_p"), alpha, keep_inv)
             fork()
             with _on_side():
                 ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True,
@@ -511,8 +518,8 @@ class FusedTrainer:
         else:
             ext.pred_mlp_bwd(w["dpmu"], w["dpsig_c"], w["psig"],
                              w["psig_pre"], w["hm2"], p("wmu_p"),
-                             p("wsig_p"), w["dz2"], g("wmu_p"), g("bmu_p"),
-                             g("wsig_p"), g("bsig_p"))
+                             p("wsig_p"), w["dz2"], w["hpart"], g("wmu_p"),
+                             g("bmu_p"), g("wsig_p"), g("bsig_p"))
             fork()
             with _on_side():
                 ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True,
@@ -598,7 +605,7 @@ class FusedTrainer:
         fork()
         with _on_side():
             ext.ln_bwd_params(x2d, w["dxln"], w["mean"], w["rstd"],
-                              g("ln_g"), g("ln_b"), chunks)
+                              w["ln_part"], g("ln_g"), g("ln_b"), chunks)
         # join: main waits for all side-stream wgrad work
         for sstream in sides:
             if sstream is not None:

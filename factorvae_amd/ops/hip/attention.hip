@@ -270,14 +270,15 @@ __global__ __launch_bounds__(256) void pred_mlp_fwd_kernel(
 }
 
 // Predictor MLP backward: dpmu, dpsig_c -> dz2 (K,H) (grad at pre-lrelu)
-// + head param grads (wmu/bmu/wsig/bsig; atomic across head WGs).
+// + per-head PARTIALS of the shared head-param grads ([dwmu H][dwsig H]
+// [dbmu][dbsig] per head), reduced in fixed head order by
+// pred_head_reduce_kernel (bit-deterministic, no float atomics).
 __global__ __launch_bounds__(64) void pred_mlp_bwd_kernel(
     const float* __restrict__ dpmu, const float* __restrict__ dpsig_c,
     const float* __restrict__ psig, const float* __restrict__ psig_pre,
     const float* __restrict__ hm2, const float* __restrict__ wmu,
     const float* __restrict__ wsig, float* __restrict__ dz2,
-    float* __restrict__ dwmu, float* __restrict__ dbmu,
-    float* __restrict__ dwsig, float* __restrict__ dbsig, int K, int H) {
+    float* __restrict__ hpart, int K, int H) {
   const int k = blockIdx.x;
   const int lane = threadIdx.x;
   if (lane >= H) return;
@@ -289,12 +290,29 @@ __global__ __launch_bounds__(64) void pred_mlp_bwd_kernel(
   const float dh2 = dm * wmu[lane] + dsp * wsig[lane];
   dz2[(long)k * H + lane] = dh2 * lrelu_grad_from_out_(h2);
 
-  atomicAdd(&dwmu[lane], dm * h2);
-  atomicAdd(&dwsig[lane], dsp * h2);
+  float* po = hpart + (long)k * (2 * H + 2);
+  po[lane] = dm * h2;
+  po[H + lane] = dsp * h2;
   if (lane == 0) {
-    atomicAdd(&dbmu[0], dm);
-    atomicAdd(&dbsig[0], dsp);
+    po[2 * H] = dm;
+    po[2 * H + 1] = dsp;
   }
+}
+
+// fixed-head-order reduce of the shared head-param grad partials
+__global__ __launch_bounds__(256) void pred_head_reduce_kernel(
+    const float* __restrict__ hpart, float* __restrict__ dwmu,
+    float* __restrict__ dbmu, float* __restrict__ dwsig,
+    float* __restrict__ dbsig, int K, int H) {
+  const int E = 2 * H + 2;
+  const int e = blockIdx.x * 256 + threadIdx.x;
+  if (e >= E) return;
+  float s = 0.0f;
+  for (int k = 0; k < K; ++k) s += hpart[(long)k * E + e];
+  if (e < H) dwmu[e] += s;
+  else if (e < 2 * H) dwsig[e - H] += s;
+  else if (e == 2 * H) dbmu[0] += s;
+  else dbsig[0] += s;
 }
 
 // ---------------------------------------------------------------------
@@ -468,9 +486,7 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_kernel(
     float* __restrict__ ds_out, float* __restrict__ dc_out,
     float* __restrict__ dWv, float* __restrict__ dbv,
     float* __restrict__ dq, float* __restrict__ dWk,
-    float* __restrict__ dbk, float* __restrict__ dwmu,
-    float* __restrict__ dbmu, float* __restrict__ dwsig,
-    float* __restrict__ dbsig,
+    float* __restrict__ dbk, float* __restrict__ hpart,
     int N, int K, int H, float alpha, float keep_inv) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* hS = (float*)smem;               // [N][H+1]
@@ -508,12 +524,13 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_kernel(
     const float dz = dh2 * lrelu_grad_from_out_(h2);
     dz2S[tid] = dz;
     dz2_out[(long)k * H + tid] = dz;
-    atomicAdd(&dwmu[tid], dm * h2);
-    atomicAdd(&dwsig[tid], dsp * h2);
-  }
-  if (tid == 0) {
-    atomicAdd(&dbmu[0], dm);
-    atomicAdd(&dbsig[0], dsp);
+    float* po = hpart + (long)k * (2 * H + 2);
+    po[tid] = dm * h2;
+    po[H + tid] = dsp * h2;
+    if (tid == 0) {
+      po[2 * H] = dm;
+      po[2 * H + 1] = dsp;
+    }
   }
   __syncthreads();
 
@@ -638,9 +655,9 @@ hipError_t fv_attn_fused_bwd(const float* dpmu, const float* dpsig_c,
                              const float* Wk, const float* bk, float* dz2,
                              float* du, float* ds, float* dc, float* dWv,
                              float* dbv, float* dq, float* dWk, float* dbk,
-                             float* dwmu, float* dbmu, float* dwsig,
-                             float* dbsig, int N, int K, int H, float alpha,
-                             float keep_inv, hipStream_t s) {
+                             float* hpart, float* dwmu, float* dbmu,
+                             float* dwsig, float* dbsig, int N, int K, int H,
+                             float alpha, float keep_inv, hipStream_t s) {
   if (H > 64) return hipErrorInvalidValue;
   const size_t lds = ((size_t)N * (H + 1) + (size_t)H * (H + 1) + N + 8 +
                       256 + 5 * 64) * sizeof(float);
@@ -648,8 +665,11 @@ hipError_t fv_attn_fused_bwd(const float* dpmu, const float* dpsig_c,
   hipLaunchKernelGGL(attn_fused_bwd_kernel, dim3(K), dim3(256), lds, s,
                      dpmu, dpsig_c, psig, psig_pre, hm2, wmu, wsig, Wl, h, a,
                      sd, mask, guard, u, Wv, q, Wk, bk, dz2, du, ds, dc, dWv,
-                     dbv, dq, dWk, dbk, dwmu, dbmu, dwsig, dbsig, N, K, H,
-                     alpha, keep_inv);
+                     dbv, dq, dWk, dbk, hpart, N, K, H, alpha, keep_inv);
+  HIP_CHECK_LAST();
+  const int E = 2 * H + 2;
+  hipLaunchKernelGGL(pred_head_reduce_kernel, dim3((E + 255) / 256),
+                     dim3(256), 0, s, hpart, dwmu, dbmu, dwsig, dbsig, K, H);
   HIP_CHECK_LAST();
   return hipSuccess;
 }
@@ -749,12 +769,17 @@ hipError_t fv_pred_mlp_fwd(const float* ctx, const float* Wl, const float* bl,
 hipError_t fv_pred_mlp_bwd(const float* dpmu, const float* dpsig_c,
                            const float* psig, const float* psig_pre,
                            const float* hm2, const float* wmu, const float* wsig,
-                           float* dz2, float* dwmu, float* dbmu, float* dwsig,
-                           float* dbsig, int K, int H, hipStream_t s) {
+                           float* dz2, float* hpart, float* dwmu, float* dbmu,
+                           float* dwsig, float* dbsig, int K, int H,
+                           hipStream_t s) {
   if (H > 64) return hipErrorInvalidValue;
   hipLaunchKernelGGL(pred_mlp_bwd_kernel, dim3(K), dim3(64), 0, s,
                      dpmu, dpsig_c, psig, psig_pre, hm2, wmu, wsig,
-                     dz2, dwmu, dbmu, dwsig, dbsig, K, H);
+                     dz2, hpart, K, H);
+  HIP_CHECK_LAST();
+  const int E = 2 * H + 2;
+  hipLaunchKernelGGL(pred_head_reduce_kernel, dim3((E + 255) / 256),
+                     dim3(256), 0, s, hpart, dwmu, dbmu, dwsig, dbsig, K, H);
   HIP_CHECK_LAST();
   return hipSuccess;
 }

@@ -37,8 +37,8 @@ hipError_t fv_absmax_scale(const float*, long, float*, float*, hipStream_t);
 hipError_t fv_cast_f32_fp8_scaled(const float*, void*, const float*, long,
                                   int, int, hipStream_t);
 hipError_t fv_ln_bwd_params(const float*, const float*, const float*,
-                            const float*, float*, float*, long, int, int,
-                            hipStream_t);
+                            const float*, float*, float*, float*, long, int,
+                            int, hipStream_t);
 hipError_t fv_gru_fwd(const float*, const float*, const float*, float*, float*,
                       float*, float*, int, int, int, hipStream_t);
 hipError_t fv_gru_bwd(const float*, const float*, const float*, const float*,
@@ -74,7 +74,7 @@ hipError_t fv_attn_fused_bwd(const float*, const float*, const float*,
                              const float*, const float*, const float*,
                              float*, float*, float*, float*, float*, float*,
                              float*, float*, float*, float*, float*, float*,
-                             float*, int, int, int, float, float,
+                             float*, float*, int, int, int, float, float,
                              hipStream_t);
 hipError_t fv_attn_fused_fwd(const float*, const float*, const float*,
                              const float*, const float*, const float*,
@@ -103,7 +103,7 @@ hipError_t fv_pred_mlp_fwd(const float*, const float*, const float*,
 hipError_t fv_pred_mlp_bwd(const float*, const float*, const float*,
                            const float*, const float*, const float*,
                            const float*, float*, float*, float*, float*,
-                           float*, int, int, hipStream_t);
+                           float*, float*, int, int, hipStream_t);
 hipError_t fv_dec_fwd(const float*, const float*, const float*, const float*,
                       const float*, const float*, const float*, const float*,
                       const float*, const float*, const float*, const float*,
@@ -113,7 +113,7 @@ hipError_t fv_dec_bwd(const float*, const float*, const float*, const float*,
                       const float*, const float*, const float*, const float*,
                       const float*, const float*, const float*, const float*,
                       const float*, float*, float*, float*, float*, float*,
-                      float*, float*, float*, float*, int, int, int,
+                      float*, float*, float*, float*, float*, int, int, int,
                       hipStream_t);
 hipError_t fv_loss_fwd(const float*, const float*, const float*, const float*,
                        const float*, const float*, float*, float*, float*, int,
@@ -406,13 +406,14 @@ void cast_f32_fp8_scaled(torch::Tensor src, torch::Tensor dst,
 }
 
 void ln_bwd_params(torch::Tensor x, torch::Tensor dxln, torch::Tensor mean,
-                   torch::Tensor rstd, torch::Tensor dgamma,
-                   torch::Tensor dbeta, long r_chunks) {
-  CK(x); CK(dxln); CK(mean); CK(rstd); CK(dgamma); CK(dbeta);
+                   torch::Tensor rstd, torch::Tensor part,
+                   torch::Tensor dgamma, torch::Tensor dbeta, long r_chunks) {
+  CK(x); CK(dxln); CK(mean); CK(rstd); CK(part); CK(dgamma); CK(dbeta);
   const long R = x.numel() / x.size(-1);
   const int C = x.size(-1);
-  RUN(fv_ln_bwd_params(fp(x), fp(dxln), fp(mean), fp(rstd), fpm(dgamma),
-                       fpm(dbeta), R, C, (int)r_chunks, cur_stream()));
+  RUN(fv_ln_bwd_params(fp(x), fp(dxln), fp(mean), fp(rstd), fpm(part),
+                       fpm(dgamma), fpm(dbeta), R, C, (int)r_chunks,
+                       cur_stream()));
 }
 
 void gru_fwd(torch::Tensor gi, torch::Tensor Whh, torch::Tensor bhh,
@@ -497,12 +498,15 @@ void attn_fused_bwd(torch::Tensor dpmu, torch::Tensor dpsig_c,
                     torch::Tensor dz2, torch::Tensor du, torch::Tensor ds,
                     torch::Tensor dc, torch::Tensor dWv, torch::Tensor dbv,
                     torch::Tensor dq, torch::Tensor dWk, torch::Tensor dbk,
+                    torch::Tensor hpart,
                     torch::Tensor dwmu, torch::Tensor dbmu,
                     torch::Tensor dwsig, torch::Tensor dbsig,
                     double alpha, double keep_inv) {
   CK(h); CK(a); CK(sd); CK(u); CK(Wv); CK(q); CK(Wk); CK(bk); CK(dz2);
   CK(du); CK(ds); CK(dc); CK(dWv); CK(dbv); CK(dq); CK(dWk); CK(dbk);
+  CK(hpart);
   const int N = h.size(0), H = h.size(1), K = q.size(0);
+  TORCH_CHECK(hpart.numel() >= (long)K * (2 * H + 2));
   const float* mp = nullptr;
   if (mask.has_value()) { CK(*mask); mp = fp(*mask); }
   RUN(fv_attn_fused_bwd(fp(dpmu), fp(dpsig_c), fp(psig), fp(psig_pre),
@@ -510,9 +514,9 @@ void attn_fused_bwd(torch::Tensor dpmu, torch::Tensor dpsig_c,
                         fp(sd), mp, guard.data_ptr<int>(), fp(u), fp(Wv),
                         fp(q), fp(Wk), fp(bk), fpm(dz2), fpm(du), fpm(ds),
                         fpm(dc), fpm(dWv), fpm(dbv), fpm(dq), fpm(dWk),
-                        fpm(dbk), fpm(dwmu), fpm(dbmu), fpm(dwsig),
-                        fpm(dbsig), N, K, H, (float)alpha, (float)keep_inv,
-                        cur_stream()));
+                        fpm(dbk), fpm(hpart), fpm(dwmu), fpm(dbmu),
+                        fpm(dwsig), fpm(dbsig), N, K, H, (float)alpha,
+                        (float)keep_inv, cur_stream()));
 }
 
 void enc_fused_fwd(torch::Tensor h, torch::Tensor Wenc, torch::Tensor benc,
@@ -631,15 +635,16 @@ void pred_mlp_fwd(torch::Tensor ctx, torch::Tensor Wl, torch::Tensor bl,
 
 void pred_mlp_bwd(torch::Tensor dpmu, torch::Tensor dpsig_c, torch::Tensor psig,
                   torch::Tensor psig_pre, torch::Tensor hm2, torch::Tensor wmu,
-                  torch::Tensor wsig, torch::Tensor dz2, torch::Tensor dwmu,
-                  torch::Tensor dbmu, torch::Tensor dwsig,
+                  torch::Tensor wsig, torch::Tensor dz2, torch::Tensor hpart,
+                  torch::Tensor dwmu, torch::Tensor dbmu, torch::Tensor dwsig,
                   torch::Tensor dbsig) {
   CK(dpmu); CK(dpsig_c); CK(psig); CK(psig_pre); CK(hm2); CK(wmu); CK(wsig);
-  CK(dz2); CK(dwmu); CK(dbmu); CK(dwsig); CK(dbsig);
+  CK(dz2); CK(hpart); CK(dwmu); CK(dbmu); CK(dwsig); CK(dbsig);
+  TORCH_CHECK(hpart.numel() >= (long)hm2.size(0) * (2 * hm2.size(1) + 2));
   RUN(fv_pred_mlp_bwd(fp(dpmu), fp(dpsig_c), fp(psig), fp(psig_pre), fp(hm2),
-                      fp(wmu), fp(wsig), fpm(dz2), fpm(dwmu), fpm(dbmu),
-                      fpm(dwsig), fpm(dbsig), hm2.size(0), hm2.size(1),
-                      cur_stream()));
+                      fp(wmu), fp(wsig), fpm(dz2), fpm(hpart), fpm(dwmu),
+                      fpm(dbmu), fpm(dwsig), fpm(dbsig), hm2.size(0),
+                      hm2.size(1), cur_stream()));
 }
 
 void dec_fwd(torch::Tensor h, torch::Tensor W1, torch::Tensor b1,
@@ -661,17 +666,21 @@ void dec_bwd(torch::Tensor drecon, torch::Tensor h, torch::Tensor a1,
              torch::Tensor eps, torch::Tensor fmu, torch::Tensor fsig_c,
              torch::Tensor W1, torch::Tensor wmu, torch::Tensor wsig,
              torch::Tensor Wb, torch::Tensor dh, torch::Tensor dz1,
-             torch::Tensor dbeta, torch::Tensor dfmu, torch::Tensor dfsig_c,
-             torch::Tensor dwmu, torch::Tensor dbmu, torch::Tensor dwsig,
-             torch::Tensor dbsig) {
+             torch::Tensor dbeta, torch::Tensor part, torch::Tensor dfmu,
+             torch::Tensor dfsig_c, torch::Tensor dwmu, torch::Tensor dbmu,
+             torch::Tensor dwsig, torch::Tensor dbsig) {
   CK(drecon); CK(h); CK(a1); CK(beta); CK(asig_pre); CK(sigma); CK(eps);
   CK(fmu); CK(fsig_c); CK(W1); CK(wmu); CK(wsig); CK(Wb); CK(dh); CK(dz1);
-  CK(dbeta); CK(dfmu); CK(dfsig_c); CK(dwmu); CK(dbmu); CK(dwsig); CK(dbsig);
+  CK(dbeta); CK(part); CK(dfmu); CK(dfsig_c); CK(dwmu); CK(dbmu); CK(dwsig);
+  CK(dbsig);
+  const int N_ = h.size(0), K_ = Wb.size(0), H_ = h.size(1);
+  const long nblk = (N_ + 31) / 32;
+  TORCH_CHECK(part.numel() >= nblk * (2 * K_ + 2 * H_ + 2));
   RUN(fv_dec_bwd(fp(drecon), fp(h), fp(a1), fp(beta), fp(asig_pre), fp(sigma),
                  fp(eps), fp(fmu), fp(fsig_c), fp(W1), fp(wmu), fp(wsig),
-                 fp(Wb), fpm(dh), fpm(dz1), fpm(dbeta), fpm(dfmu),
+                 fp(Wb), fpm(dh), fpm(dz1), fpm(dbeta), fpm(part), fpm(dfmu),
                  fpm(dfsig_c), fpm(dwmu), fpm(dbmu), fpm(dwsig), fpm(dbsig),
-                 h.size(0), Wb.size(0), h.size(1), cur_stream()));
+                 N_, K_, H_, cur_stream()));
 }
 
 void loss_fwd(torch::Tensor recon, torch::Tensor y, torch::Tensor fmu,

@@ -94,10 +94,10 @@ __global__ __launch_bounds__(256) void dec_fwd_kernel(
 // Backward. Writes dh (N,H) directly (FIRST dh contributor — plain
 // store), dz1 (N,H) and dbeta (N,K) for the weight-grad gemm_tn calls.
 // The shared grads dfmu/dfsig_c (K), dwmu/dwsig (H), dbmu/dbsig (1) are
-// accumulated in REGISTERS over DEC_ITERS rows per wave and flushed with
-// ONE atomicAdd per element per block — N per-row atomic storms (the
-// profile showed 1.1M global atomics at N=3500) collapse to
-// N/(DEC_RPW*DEC_ITERS) per element.
+// accumulated in REGISTERS over DEC_ITERS rows per wave and flushed as
+// PER-BLOCK PARTIALS (layout [dfmu K][dfsig K][dwmu H][dwsig H][2]),
+// summed in fixed block order by dec_bwd_reduce_kernel — bit-exact
+// run-to-run determinism, no float atomics.
 #define DEC_ITERS 8
 __global__ __launch_bounds__(256) void dec_bwd_kernel(
     const float* __restrict__ drecon, const float* __restrict__ h,
@@ -108,9 +108,7 @@ __global__ __launch_bounds__(256) void dec_bwd_kernel(
     const float* __restrict__ wmu, const float* __restrict__ wsig,
     const float* __restrict__ Wb, float* __restrict__ dh,
     float* __restrict__ dz1, float* __restrict__ dbeta_out,
-    float* __restrict__ dfmu, float* __restrict__ dfsig_c,
-    float* __restrict__ dwmu, float* __restrict__ dbmu,
-    float* __restrict__ dwsig, float* __restrict__ dbsig,
+    float* __restrict__ part,
     int N, int K, int H) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* W1S = (float*)smem;                 // [H][H] as-is
@@ -192,45 +190,47 @@ __global__ __launch_bounds__(256) void dec_bwd_kernel(
     }
   }
 
-  // cross-wave reduce of the shared-grad registers, one atomic/element
+  // cross-wave reduce of the shared-grad registers -> per-block partial
+  const int E = 2 * K + 2 * H + 2;
+  float* po = part + (long)blockIdx.x * E;
   __syncthreads();
   red[w * 64 + lane] = rfmu0;
   __syncthreads();
   if (w == 0 && lane < K && lane < 64)
-    atomicAdd(&dfmu[lane],
-              red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane]);
+    po[lane] =
+        red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
   __syncthreads();
   red[w * 64 + lane] = rfsig0;
   __syncthreads();
   if (w == 0 && lane < K && lane < 64)
-    atomicAdd(&dfsig_c[lane],
-              red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane]);
+    po[K + lane] =
+        red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
   if (K > 64) {
     __syncthreads();
     red[w * 64 + lane] = kh2 ? rfmu1 : 0.0f;
     __syncthreads();
     if (w == 0 && lane + 64 < K)
-      atomicAdd(&dfmu[lane + 64],
-                red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane]);
+      po[lane + 64] =
+          red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
     __syncthreads();
     red[w * 64 + lane] = kh2 ? rfsig1 : 0.0f;
     __syncthreads();
     if (w == 0 && lane + 64 < K)
-      atomicAdd(&dfsig_c[lane + 64],
-                red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane]);
+      po[K + lane + 64] =
+          red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
   }
   __syncthreads();
   red[w * 64 + lane] = rwmu;
   __syncthreads();
   if (w == 0 && lane < H)
-    atomicAdd(&dwmu[lane],
-              red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane]);
+    po[2 * K + lane] =
+        red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
   __syncthreads();
   red[w * 64 + lane] = rwsig;
   __syncthreads();
   if (w == 0 && lane < H)
-    atomicAdd(&dwsig[lane],
-              red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane]);
+    po[2 * K + H + lane] =
+        red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
   // scalar bias grads (lane-0 registers only)
   __syncthreads();
   if (lane == 0) {
@@ -239,9 +239,28 @@ __global__ __launch_bounds__(256) void dec_bwd_kernel(
   }
   __syncthreads();
   if (tid == 0) {
-    atomicAdd(&dbmu[0], red[0] + red[1] + red[2] + red[3]);
-    atomicAdd(&dbsig[0], red[4] + red[5] + red[6] + red[7]);
+    po[2 * K + 2 * H] = red[0] + red[1] + red[2] + red[3];
+    po[2 * K + 2 * H + 1] = red[4] + red[5] + red[6] + red[7];
   }
+}
+
+// fixed-order reduce of the per-block partials into the grad slots
+__global__ __launch_bounds__(256) void dec_bwd_reduce_kernel(
+    const float* __restrict__ part, float* __restrict__ dfmu,
+    float* __restrict__ dfsig_c, float* __restrict__ dwmu,
+    float* __restrict__ dbmu, float* __restrict__ dwsig,
+    float* __restrict__ dbsig, int nblk, int K, int H) {
+  const int E = 2 * K + 2 * H + 2;
+  const int e = blockIdx.x * 256 + threadIdx.x;
+  if (e >= E) return;
+  float s = 0.0f;
+  for (int z = 0; z < nblk; ++z) s += part[(long)z * E + e];
+  if (e < K) dfmu[e] += s;
+  else if (e < 2 * K) dfsig_c[e - K] += s;
+  else if (e < 2 * K + H) dwmu[e - 2 * K] += s;
+  else if (e < 2 * K + 2 * H) dwsig[e - 2 * K - H] += s;
+  else if (e == 2 * K + 2 * H) dbmu[0] += s;
+  else dbsig[0] += s;
 }
 
 extern "C" {
@@ -267,17 +286,21 @@ hipError_t fv_dec_bwd(const float* drecon, const float* h, const float* a1,
                       const float* sigma, const float* eps, const float* fmu,
                       const float* fsig_c, const float* W1, const float* wmu,
                       const float* wsig, const float* Wb, float* dh,
-                      float* dz1, float* dbeta, float* dfmu, float* dfsig_c,
-                      float* dwmu, float* dbmu, float* dwsig, float* dbsig,
-                      int N, int K, int H, hipStream_t s) {
+                      float* dz1, float* dbeta, float* part, float* dfmu,
+                      float* dfsig_c, float* dwmu, float* dbmu, float* dwsig,
+                      float* dbsig, int N, int K, int H, hipStream_t s) {
   if (H > 64 || K > 128) return hipErrorInvalidValue;
   const size_t lds = ((size_t)H * H + (size_t)K * H + DEC_RPW * H +
                       (size_t)DEC_RPW * K + 4 * 64) * sizeof(float);
-  dim3 grid((N + DEC_RPW * DEC_ITERS - 1) / (DEC_RPW * DEC_ITERS));
-  hipLaunchKernelGGL(dec_bwd_kernel, grid, dim3(256), lds, s,
+  const int nblk = (N + DEC_RPW * DEC_ITERS - 1) / (DEC_RPW * DEC_ITERS);
+  hipLaunchKernelGGL(dec_bwd_kernel, dim3(nblk), dim3(256), lds, s,
                      drecon, h, a1, beta, asig_pre, sigma, eps, fmu, fsig_c,
-                     W1, wmu, wsig, Wb, dh, dz1, dbeta, dfmu, dfsig_c,
-                     dwmu, dbmu, dwsig, dbsig, N, K, H);
+                     W1, wmu, wsig, Wb, dh, dz1, dbeta, part, N, K, H);
+  HIP_CHECK_LAST();
+  const int E = 2 * K + 2 * H + 2;
+  hipLaunchKernelGGL(dec_bwd_reduce_kernel, dim3((E + 255) / 256), dim3(256),
+                     0, s, part, dfmu, dfsig_c, dwmu, dbmu, dwsig, dbsig,
+                     nblk, K, H);
   HIP_CHECK_LAST();
   return hipSuccess;
 }

@@ -70,10 +70,12 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
 // rows per block chosen by the launcher: enough blocks to fill the
 // chip, few enough that the one-atomic-per-column-per-block flush stays
 // cheap at large R
+// per-(row-chunk) partials, layout part[yblock][2C] ([dgamma C][dbeta C]),
+// reduced in fixed yblock order by ln_bwd_reduce_kernel (deterministic).
 __global__ __launch_bounds__(256) void ln_bwd_params_kernel(
     const float* __restrict__ x, const float* __restrict__ dxln,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    float* __restrict__ dgamma, float* __restrict__ dbeta, long R, int C,
+    float* __restrict__ part, long R, int C,
     int rows_per_block) {
   __shared__ float pg[4][64];
   __shared__ float pb[4][64];
@@ -95,9 +97,21 @@ __global__ __launch_bounds__(256) void ln_bwd_params_kernel(
   pb[w][lane] = db;
   __syncthreads();
   if (w == 0 && c < C) {
-    atomicAdd(&dgamma[c], pg[0][lane] + pg[1][lane] + pg[2][lane] + pg[3][lane]);
-    atomicAdd(&dbeta[c], pb[0][lane] + pb[1][lane] + pb[2][lane] + pb[3][lane]);
+    float* po = part + (long)blockIdx.y * 2 * C;
+    po[c] = pg[0][lane] + pg[1][lane] + pg[2][lane] + pg[3][lane];
+    po[C + c] = pb[0][lane] + pb[1][lane] + pb[2][lane] + pb[3][lane];
   }
+}
+
+__global__ __launch_bounds__(256) void ln_bwd_reduce_kernel(
+    const float* __restrict__ part, float* __restrict__ dgamma,
+    float* __restrict__ dbeta, int yblocks, int C) {
+  const int e = blockIdx.x * 256 + threadIdx.x;
+  if (e >= 2 * C) return;
+  float s = 0.0f;
+  for (int z = 0; z < yblocks; ++z) s += part[(long)z * 2 * C + e];
+  if (e < C) dgamma[e] += s;
+  else dbeta[e - C] += s;
 }
 
 extern "C" {
@@ -116,19 +130,25 @@ hipError_t fv_ln_fwd(const float* x, const float* gamma, const float* beta,
 
 hipError_t fv_ln_bwd_params(const float* x, const float* dxln,
                             const float* mean, const float* rstd,
-                            float* dgamma, float* dbeta, long R, int C,
-                            int r_chunks, hipStream_t stream) {
+                            float* part, float* dgamma, float* dbeta,
+                            long R, int C, int r_chunks,
+                            hipStream_t stream) {
   (void)r_chunks;
   // ~2000 blocks: this reduction is latency-bound per block (two global
-  // reads per row-iteration), so parallelism beats atomic thrift
+  // reads per row-iteration), so parallelism beats reduce thrift
   const int cblocks = (C + 63) / 64;
   long target_y = (2048 + cblocks - 1) / cblocks;
   long rpb = (R + target_y - 1) / target_y;
   if (rpb < 64) rpb = 64;
   if (rpb > 8192) rpb = 8192;
-  dim3 grid(cblocks, (unsigned)((R + rpb - 1) / rpb));
+  const unsigned yblocks = (unsigned)((R + rpb - 1) / rpb);
+  dim3 grid(cblocks, yblocks);
   hipLaunchKernelGGL(ln_bwd_params_kernel, grid, dim3(256), 0, stream,
-                     x, dxln, mean, rstd, dgamma, dbeta, R, C, (int)rpb);
+                     x, dxln, mean, rstd, part, R, C, (int)rpb);
+  HIP_CHECK_LAST();
+  hipLaunchKernelGGL(ln_bwd_reduce_kernel, dim3((2 * C + 255) / 256),
+                     dim3(256), 0, stream, part, dgamma, dbeta,
+                     (int)yblocks, C);
   HIP_CHECK_LAST();
   return hipSuccess;
 }

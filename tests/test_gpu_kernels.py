@@ -119,7 +119,8 @@ def test_ln_bwd_params():
     ext.ln_fwd(x, g_.detach(), b_.detach(), xln, mean, rstd, 1e-5)
     dg = torch.zeros(C, device=DEV)
     db = torch.zeros(C, device=DEV)
-    ext.ln_bwd_params(x, dxln, mean, rstd, dg, db, 3)
+    part = torch.zeros(4096 * 2 * C, device=DEV)
+    ext.ln_bwd_params(x, dxln, mean, rstd, part, dg, db, 3)
     torch.cuda.synchronize()
     assert_close(dg, g_.grad, atol=5e-4, rtol=5e-4, what="dgamma")
     assert_close(db, b_.grad, atol=5e-4, rtol=5e-4, what="dbeta")
@@ -729,3 +730,36 @@ def test_gru_mfma_vs_fp32(N, T):
         dgh32.reshape(-1), dghm.reshape(-1), dim=0)
     assert float(cos2) > 0.995, f"dgh cosine {float(cos2)}"
     assert torch.equal(dgim_bf.float().to(torch.bfloat16), dgim_bf)
+
+
+def test_full_step_bit_determinism():
+    """set_seed contract (reference utils.py:10-17): two identically
+    seeded runs must produce BIT-IDENTICAL parameters after training.
+    All cross-workgroup reductions must therefore be fixed-order."""
+    from factorvae_amd.engine.fused import FusedTrainer
+    from factorvae_amd.models.modules import build_factorvae
+    from factorvae_amd.utils import set_seed
+
+    N, T, C, H, M, K = 300, 20, 158, 64, 128, 20
+
+    def run():
+        set_seed(7)
+        model = build_factorvae(num_latent=C, hidden_size=H, num_portfolio=M,
+                                num_factor=K).to(DEV)
+        tr = FusedTrainer(model, lr=1e-3, t_max=40, device=DEV,
+                          use_graph=False)
+        g = torch.Generator(device=DEV).manual_seed(3)
+        days = [(torch.randn(N, T, C, device=DEV, generator=g),
+                 torch.randn(N, 1, device=DEV, generator=g))
+                for _ in range(4)]
+        set_seed(11)
+        for _ in range(5):
+            for x, y in days:
+                tr.step(x, y)
+        torch.cuda.synchronize()
+        return tr.params.flat.clone()
+
+    p1 = run()
+    p2 = run()
+    diff = (p1 != p2).sum().item()
+    assert diff == 0, f"{diff}/{p1.numel()} parameter words differ between identically-seeded runs"
