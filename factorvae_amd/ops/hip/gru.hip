@@ -374,10 +374,21 @@ __global__ __launch_bounds__(512) void gru_bwd_fast_kernel(
 
 extern "C" {
 
+// f32 MFMA recurrence (gru_mfma.hip) — used when H == 64 (exact f32)
+hipError_t fv_gru_fwd_mfma_f32(const float*, const float*, const float*,
+                               float*, float*, float*, float*, int, int, int,
+                               hipStream_t);
+hipError_t fv_gru_bwd_mfma_f32(const float*, const float*, const float*,
+                               const float*, float*, float*, int, int, int,
+                               hipStream_t);
+
 hipError_t fv_gru_fwd(const float* gi, const float* Whh, const float* bhh,
                       float* h_final, float* h_seq, float* h_prev,
                       float* gates4, int N, int T, int H, hipStream_t stream) {
   if (H > 64) return hipErrorInvalidValue;
+  if (H == 64)
+    return fv_gru_fwd_mfma_f32(gi, Whh, bhh, h_final, h_seq, h_prev, gates4,
+                               N, T, H, stream);
   if ((H & 3) == 0) {
     const size_t lds = ((size_t)3 * 64 * H + GRU_SPW_F * 64) * sizeof(float);
     dim3 grid((N + GRU_SPW_F - 1) / GRU_SPW_F);
@@ -398,6 +409,9 @@ hipError_t fv_gru_bwd(const float* dh_final, const float* h_prev,
                       float* dgi, float* dgh, void* dgi_bf, void* dgh_bf,
                       int N, int T, int H, hipStream_t stream) {
   if (H > 64) return hipErrorInvalidValue;
+  if (H == 64 && !dgi_bf && !dgh_bf)
+    return fv_gru_bwd_mfma_f32(dh_final, h_prev, gates4, Whh, dgi, dgh,
+                               N, T, H, stream);
   if ((H & 3) == 0) {
     const size_t lds = ((size_t)3 * H * 64 + GRU_SPW_F * 3 * H +
                         (size_t)GRU_SPW_F * 64) * sizeof(float);

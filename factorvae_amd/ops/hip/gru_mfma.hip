@@ -274,6 +274,213 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
   }
 }
 
+// ---------------------------------------------------------------------
+// f32 variants (fp32 engine mode): identical phase structure, but the
+// recurrence matmul uses v_mfma_f32_16x16x4_f32 — exact fp32 (the f32
+// MFMA path computes bit-exact f32 products), so these replace the
+// float4-LDS kernels wholesale when H = 64. B-fragments (Whh) live in
+// registers: 48 floats/lane fwd (3 n-tiles x 16 k-steps), 48 bwd.
+#define GMF_SH 68   // fp32 h image stride (2-way worst-case banks)
+
+__global__ __launch_bounds__(256) void gru_fwd_mfma_f32_kernel(
+    const float* __restrict__ gi, const float* __restrict__ Whh,
+    const float* __restrict__ bhh, float* __restrict__ h_final,
+    float* __restrict__ h_seq, float* __restrict__ h_prev_out,
+    float* __restrict__ gates4, int N, int T) {
+  __shared__ float hS[GM_S][GMF_SH];
+  __shared__ float ghS[GM_S][192];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const int fi = lane & 15;
+  const int fk = lane >> 4;       // k-offset within each 4-step
+  const int s0 = blockIdx.x * GM_S;
+
+  // B frags: b[t3][ks] = Whh[wv*48 + t3*16 + fi][ks*4 + fk]
+  float bfr[3][16];
+#pragma unroll
+  for (int t3 = 0; t3 < 3; ++t3) {
+    const int n = wv * 48 + t3 * 16 + fi;
+#pragma unroll
+    for (int ks = 0; ks < 16; ++ks)
+      bfr[t3][ks] = Whh[(long)n * 64 + ks * 4 + fk];
+  }
+
+  for (int idx = tid; idx < GM_S * 64; idx += 256)
+    hS[idx >> 6][idx & 63] = 0.0f;
+  __syncthreads();
+
+  const int es = tid >> 4;
+  const int ej = (tid & 15) * 4;
+  const bool elive = (s0 + es) < N;
+  const long erow = (long)(s0 + es) * T;
+
+  float pgr[4], pgz[4], pgn[4];
+  auto gi_load = [&](int t) {
+    if (elive) {
+      const float* g = gi + (erow + t) * 192;
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        pgr[u] = g[ej + u];
+        pgz[u] = g[64 + ej + u];
+        pgn[u] = g[128 + ej + u];
+      }
+    }
+  };
+  gi_load(0);
+
+  float bh_r[4], bh_z[4], bh_n[4];
+#pragma unroll
+  for (int u = 0; u < 4; ++u) {
+    bh_r[u] = bhh[ej + u];
+    bh_z[u] = bhh[64 + ej + u];
+    bh_n[u] = bhh[128 + ej + u];
+  }
+
+  for (int t = 0; t < T; ++t) {
+    f32x4 acc[3] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+#pragma unroll
+    for (int ks = 0; ks < 16; ++ks) {
+      const float a = hS[fi][ks * 4 + fk];
+#pragma unroll
+      for (int t3 = 0; t3 < 3; ++t3)
+        acc[t3] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bfr[t3][ks],
+                                                       acc[t3], 0, 0, 0);
+    }
+#pragma unroll
+    for (int t3 = 0; t3 < 3; ++t3) {
+      const int n = wv * 48 + t3 * 16 + fi;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) ghS[fk * 4 + rr][n] = acc[t3][rr];
+    }
+    __syncthreads();
+
+    float hn[4];
+    float gr4[4], gz4[4], gn4[4], gq4[4], hp4[4];
+    if (elive) {
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const int j = ej + u;
+        const float hp = hS[es][j];
+        const float r = sigmoidf_(pgr[u] + ghS[es][j] + bh_r[u]);
+        const float z = sigmoidf_(pgz[u] + ghS[es][64 + j] + bh_z[u]);
+        const float q = ghS[es][128 + j] + bh_n[u];
+        const float n = tanhf(fmaf(r, q, pgn[u]));
+        hn[u] = fmaf(z, hp - n, n);
+        gr4[u] = r; gz4[u] = z; gn4[u] = n; gq4[u] = q; hp4[u] = hp;
+      }
+      const long tb = (erow + t);
+      *(f32x4*)&h_seq[tb * 64 + ej] = *(f32x4*)hn;
+      *(f32x4*)&h_prev_out[tb * 64 + ej] = *(f32x4*)hp4;
+      float* g4 = &gates4[tb * 256];
+      *(f32x4*)&g4[ej] = *(f32x4*)gr4;
+      *(f32x4*)&g4[64 + ej] = *(f32x4*)gz4;
+      *(f32x4*)&g4[128 + ej] = *(f32x4*)gn4;
+      *(f32x4*)&g4[192 + ej] = *(f32x4*)gq4;
+      if (t == T - 1)
+        *(f32x4*)&h_final[(long)(s0 + es) * 64 + ej] = *(f32x4*)hn;
+    }
+    if (t + 1 < T) gi_load(t + 1);
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+      hS[es][ej + u] = elive ? hn[u] : 0.0f;
+    __syncthreads();
+  }
+}
+
+__global__ __launch_bounds__(256) void gru_bwd_mfma_f32_kernel(
+    const float* __restrict__ dh_final, const float* __restrict__ h_prev_in,
+    const float* __restrict__ gates4, const float* __restrict__ Whh,
+    float* __restrict__ dgi, float* __restrict__ dgh, int N, int T) {
+  __shared__ float dgS[GM_S][200];
+  __shared__ float dhS[GM_S][64];
+  __shared__ float zdh[GM_S][64];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const int fi = lane & 15;
+  const int fk = lane >> 4;
+  const int s0 = blockIdx.x * GM_S;
+
+  // B frags: Whh^T — wave's 16 h-cols: b[ks] = Whh[ks*4 + fk][wv*16 + fi]
+  float bfr[48];
+#pragma unroll
+  for (int ks = 0; ks < 48; ++ks)
+    bfr[ks] = Whh[(long)(ks * 4 + fk) * 64 + wv * 16 + fi];
+
+  const int es = tid >> 4;
+  const int ej = (tid & 15) * 4;
+  const bool elive = (s0 + es) < N;
+  const long erow = (long)(s0 + es) * T;
+
+  for (int idx = tid; idx < GM_S * 200; idx += 256)
+    dgS[idx / 200][idx % 200] = 0.0f;
+  if (elive) {
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+      dhS[es][ej + u] = dh_final[(long)(s0 + es) * 64 + ej + u];
+  } else {
+#pragma unroll
+    for (int u = 0; u < 4; ++u) dhS[es][ej + u] = 0.0f;
+  }
+  __syncthreads();
+
+  for (int t = T - 1; t >= 0; --t) {
+    if (elive) {
+      const long tb = erow + t;
+      const float* g4 = &gates4[tb * 256];
+      float dgr4[4], dgz4[4], da4[4], dghn4[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const int j = ej + u;
+        const float r = g4[j];
+        const float z = g4[64 + j];
+        const float n = g4[128 + j];
+        const float q = g4[192 + j];
+        const float hp = h_prev_in[tb * 64 + j];
+        const float dh = dhS[es][j];
+        const float dz = dh * (hp - n);
+        const float dn = dh * (1.0f - z);
+        const float da = dn * (1.0f - n * n);
+        const float dgh_n = da * r;
+        const float dr = da * q;
+        const float dgate_r = dr * r * (1.0f - r);
+        const float dgate_z = dz * z * (1.0f - z);
+        dgr4[u] = dgate_r; dgz4[u] = dgate_z; da4[u] = da; dghn4[u] = dgh_n;
+        zdh[es][j] = dh * z;
+        dgS[es][j] = dgate_r;
+        dgS[es][64 + j] = dgate_z;
+        dgS[es][128 + j] = dgh_n;
+      }
+      float* di = &dgi[tb * 192];
+      float* dg = &dgh[tb * 192];
+      *(f32x4*)&di[ej] = *(f32x4*)dgr4;
+      *(f32x4*)&di[64 + ej] = *(f32x4*)dgz4;
+      *(f32x4*)&di[128 + ej] = *(f32x4*)da4;
+      *(f32x4*)&dg[ej] = *(f32x4*)dgr4;
+      *(f32x4*)&dg[64 + ej] = *(f32x4*)dgz4;
+      *(f32x4*)&dg[128 + ej] = *(f32x4*)dghn4;
+    }
+    __syncthreads();
+
+    f32x4 acc = {0, 0, 0, 0};
+#pragma unroll
+    for (int ks = 0; ks < 48; ++ks) {
+      const float a = dgS[fi][ks * 4 + fk];
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bfr[ks], acc, 0, 0, 0);
+    }
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int m = fk * 4 + rr;
+      const int i = wv * 16 + fi;
+      dhS[m][i] = acc[rr] + zdh[m][i];
+    }
+    __syncthreads();
+  }
+}
+
 extern "C" {
 
 hipError_t fv_gru_fwd_mfma(const float* gi, const void* whh_bf,
@@ -284,6 +491,30 @@ hipError_t fv_gru_fwd_mfma(const float* gi, const void* whh_bf,
   dim3 grid((N + GM_S - 1) / GM_S);
   hipLaunchKernelGGL(gru_fwd_mfma_kernel, grid, dim3(256), 0, stream,
                      gi, whh_bf, bhh, h_final, h_seq, h_prev, gates4, N, T);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_gru_fwd_mfma_f32(const float* gi, const float* Whh,
+                               const float* bhh, float* h_final,
+                               float* h_seq, float* h_prev, float* gates4,
+                               int N, int T, int H, hipStream_t stream) {
+  if (H != 64) return hipErrorInvalidValue;
+  dim3 grid((N + GM_S - 1) / GM_S);
+  hipLaunchKernelGGL(gru_fwd_mfma_f32_kernel, grid, dim3(256), 0, stream,
+                     gi, Whh, bhh, h_final, h_seq, h_prev, gates4, N, T);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_gru_bwd_mfma_f32(const float* dh_final, const float* h_prev,
+                               const float* gates4, const float* Whh,
+                               float* dgi, float* dgh, int N, int T, int H,
+                               hipStream_t stream) {
+  if (H != 64) return hipErrorInvalidValue;
+  dim3 grid((N + GM_S - 1) / GM_S);
+  hipLaunchKernelGGL(gru_bwd_mfma_f32_kernel, grid, dim3(256), 0, stream,
+                     dh_final, h_prev, gates4, Whh, dgi, dgh, N, T);
   HIP_CHECK_LAST();
   return hipSuccess;
 }
@@ -303,3 +534,4 @@ hipError_t fv_gru_bwd_mfma(const float* dh_final, const float* h_prev,
 }
 
 }  // extern "C"
+
