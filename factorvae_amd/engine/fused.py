@@ -333,7 +333,6 @@ class FusedTrainer:
             w["xln_bf"] = fb(R, C)
             w["xp_bf"] = fb(R, C)
             w["dgi_bf"] = fb(R, 3 * H)
-            w["dxp_bf"] = fb(R, C)
             w["dzx_bf"] = fb(R, C)
             w["h_prev_bf"] = fb(R, H)
             w["dgh_bf"] = fb(R, 3 * H)
