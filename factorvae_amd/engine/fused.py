@@ -137,6 +137,14 @@ class FusedTrainer:
         # exceeds the main path, so one side stream becomes the bottleneck
         self.s_side2 = (torch.cuda.Stream(device=self.device)
                         if self.device.type == "cuda" else None)
+        # comm stream: in DP runs the attention/encoder/decoder slice of
+        # the gradient arena (everything packed after the extractor) is
+        # all-reduced here as soon as its last producer finishes,
+        # overlapped with the extractor backward; the extractor slice
+        # follows after the join (xGMI ring latency hides under ~40% of
+        # the backward)
+        self.s_comm = (torch.cuda.Stream(device=self.device)
+                       if self.device.type == "cuda" else None)
 
     # ---------------------------------------------------------------- params
     _STACKED = ("q_att", "Wk", "bk", "Wv", "bv")
@@ -435,7 +443,8 @@ class FusedTrainer:
             ext.loss_fwd(w["recon"], yv, w["fmu"], w["fsig_c"], w["pmu"],
                          w["psig_c"], w["loss"], w["mse"], w["kl"])
 
-    def _launch_backward(self, N: int, T: int, x=None, y=None):
+    def _launch_backward(self, N: int, T: int, x=None, y=None,
+                         comm_overlap: bool = False):
         """Backward pass. The activation-gradient chain runs on the main
         stream; every weight-gradient GEMM / colsum forks onto the side
         stream as soon as its producer is done (fork/join events become
@@ -551,6 +560,21 @@ class FusedTrainer:
                         True, g("benc"), w["tn_partb_s"])
         ext.gemm_nn(w["dscores"], p("Wenc"), None, w["dh"], 1.0, True, False)
 
+        if comm_overlap:
+            # every non-extractor gradient (arena tail from Wenc on) is
+            # final: reduce it now, overlapped with the extractor bwd
+            e1 = torch.cuda.Event()
+            e1.record(main)
+            self.s_comm.wait_event(e1)
+            if sides[0] is not None:
+                e2 = torch.cuda.Event()
+                e2.record(sides[0])
+                self.s_comm.wait_event(e2)
+            with torch.cuda.stream(self.s_comm):
+                tail = self.grads[self.params.offsets["Wenc"][0]:]
+                tail.div_(get_world_size())
+                torch.distributed.all_reduce(tail)
+
         # extractor backward
         if self.bf16 and H == 64:
             ext.gru_bwd_mfma(w["dh"], w["h_prev"], w["gates4"], self.whh_bf,
@@ -605,8 +629,12 @@ class FusedTrainer:
         with _on_side():
             ext.ln_bwd_params(x2d, w["dxln"], w["mean"], w["rstd"],
                               w["ln_part"], g("ln_g"), g("ln_b"), chunks)
-        # join: main waits for all side-stream wgrad work
-        for sstream in sides:
+        # join: main waits for all side-stream wgrad work (+ the early
+        # comm slice, so the final slice reduce and adam are ordered)
+        join_streams = list(sides)
+        if comm_overlap:
+            join_streams.append(self.s_comm)
+        for sstream in join_streams:
             if sstream is not None:
                 e = torch.cuda.Event()
                 e.record(sstream)
@@ -709,10 +737,13 @@ class FusedTrainer:
         if rng_in_graph:
             self._fill_rng(N)
         self._launch_forward(N, T, x=x, y=y)
-        self._launch_backward(N, T, x=x, y=y)
+        self._launch_backward(N, T, x=x, y=y, comm_overlap=comm_in_graph)
         if comm_in_graph:
-            self.grads.div_(get_world_size())
-            torch.distributed.all_reduce(self.grads)
+            # the non-extractor slice was reduced inside backward,
+            # overlapped; only the extractor slice remains
+            head = self.grads[:self.params.offsets["Wenc"][0]]
+            head.div_(get_world_size())
+            torch.distributed.all_reduce(head)
         if with_opt:
             self._launch_optimizer(inc=not inc_early)
 
