@@ -1,0 +1,54 @@
+"""Serving daemon tests (FastAPI TestClient, CPU eager path here; the
+fused HIP path activates on GPU)."""
+
+import numpy as np
+import pytest
+import torch
+
+fastapi = pytest.importorskip("fastapi")
+from fastapi.testclient import TestClient
+
+from factorvae_amd.serve import ScoringEngine, build_app
+
+
+@pytest.fixture(scope="module")
+def client(tmp_path_factory):
+    from factorvae_amd.models.modules import build_factorvae
+
+    model = build_factorvae(num_latent=12, hidden_size=8, num_portfolio=4,
+                            num_factor=3)
+    ckpt = tmp_path_factory.mktemp("srv") / "m.pt"
+    torch.save(model.state_dict(), ckpt)
+    eng = ScoringEngine(str(ckpt), num_latent=12, hidden_size=8,
+                        num_portfolio=4, num_factor=3, seq_length=5,
+                        device="cpu")
+    return TestClient(build_app(eng))
+
+
+def test_health_and_model(client):
+    r = client.get("/health")
+    assert r.status_code == 200 and r.json()["status"] == "ok"
+    m = client.get("/model").json()
+    assert m["num_factor"] == 3 and m["seq_length"] == 5
+
+
+def test_score_roundtrip(client):
+    x = np.random.default_rng(0).standard_normal((7, 5, 12)).tolist()
+    r = client.post("/score", json={"x": x})
+    assert r.status_code == 200
+    scores = r.json()["scores"]
+    assert len(scores) == 7
+    assert all(np.isfinite(scores))
+
+
+def test_score_batch_and_validation(client):
+    rng = np.random.default_rng(1)
+    days = [rng.standard_normal((n, 5, 12)).tolist() for n in (3, 9)]
+    r = client.post("/score_batch", json={"days": days})
+    assert r.status_code == 200
+    out = r.json()["scores"]
+    assert [len(o) for o in out] == [3, 9]
+    # wrong T -> 422
+    bad = rng.standard_normal((4, 6, 12)).tolist()
+    r = client.post("/score", json={"x": bad})
+    assert r.status_code == 422
