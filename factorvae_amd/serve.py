@@ -82,17 +82,22 @@ class ScoringEngine:
         return out.reshape(-1).cpu()
 
 
-def build_app(engine: ScoringEngine):
-    from fastapi import FastAPI, HTTPException
+try:  # module-level request models (FastAPI resolves annotations here)
     from pydantic import BaseModel
-
-    app = FastAPI(title="FactorVAE MI355X scoring daemon")
 
     class ScoreRequest(BaseModel):
         x: List[List[List[float]]]  # (N, T, C)
 
     class BatchRequest(BaseModel):
         days: List[List[List[List[float]]]]
+except ImportError:  # serving extras absent: ScoringEngine still usable
+    ScoreRequest = BatchRequest = None
+
+
+def build_app(engine: ScoringEngine):
+    from fastapi import FastAPI, HTTPException
+
+    app = FastAPI(title="FactorVAE MI355X scoring daemon")
 
     @app.get("/health")
     def health():
