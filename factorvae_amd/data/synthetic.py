@@ -27,6 +27,7 @@ def make_synthetic_frame(
     ragged: bool = False,
     signal_rank: int = 8,
     signal_strength: float = 0.15,
+    label_from_features: bool = False,
 ) -> pd.DataFrame:
     """Build a synthetic (datetime, instrument)-indexed frame.
 
@@ -37,6 +38,12 @@ def make_synthetic_frame(
 
     ragged=True drops a random ~10% of stocks on each day (variable N per
     day, like real universes).
+
+    label_from_features=True makes the return a fixed hidden linear
+    function of the day's OWN feature vector (plus noise) — i.e. a
+    signal a cross-sectional model can actually learn from x_t, unlike
+    the default latent-factor returns (whose day factor is zero-mean
+    given the features). Use it for end-to-end learning tests/demos.
     """
     rng = np.random.default_rng(seed)
     dates = pd.bdate_range(start, periods=n_days)
@@ -50,6 +57,8 @@ def make_synthetic_frame(
     # per-stock loadings onto latent factors and onto features
     loadings = rng.standard_normal((n_stocks, signal_rank)) * 0.5
     feat_mix = rng.standard_normal((signal_rank, n_features)) * 0.3
+    # hidden feature->return weights for the learnable-label mode
+    w_hidden = rng.standard_normal(n_features) / np.sqrt(n_features)
 
     rows = []
     index = []
@@ -65,8 +74,12 @@ def make_synthetic_frame(
         feats = loadings[keep] @ feat_mix + rng.standard_normal((len(keep), n_features))
         feats = feats.astype(np.float32)
 
-        # next-period "return": signal from features + idiosyncratic noise
-        raw_ret = signal_strength * common + rng.standard_normal(len(keep))
+        # next-period "return": signal + idiosyncratic noise
+        if label_from_features:
+            raw_ret = (signal_strength * (feats @ w_hidden)
+                       + 0.3 * rng.standard_normal(len(keep)))
+        else:
+            raw_ret = signal_strength * common + rng.standard_normal(len(keep))
         # CSRankNorm: rank -> centered/scaled (qlib convention: (rank-0.5)*sqrt(12))
         rank = pd.Series(raw_ret).rank(pct=True).to_numpy()
         label = ((rank - 0.5) * np.sqrt(12.0)).astype(np.float32)
