@@ -27,7 +27,7 @@ def test_train_main_fused_end_to_end(tmp_path, dtype):
     from factorvae_amd.utils import DataArgument, checkpoint_path
 
     df = make_synthetic_frame(n_days=60, n_stocks=64, seed=11,
-                              signal_strength=0.5)
+                              signal_strength=1.0, label_from_features=True)
     args = SimpleNamespace(
         num_epochs=4, lr=1e-3, num_latent=158, num_portfolio=32,
         seq_len=8, num_factor=8, hidden_size=64, seed=0,
@@ -53,6 +53,19 @@ def test_train_main_fused_end_to_end(tmp_path, dtype):
                          num_factor=8)
     m2.load_state_dict(torch.load(ckpt, map_location="cpu",
                                   weights_only=True))
+    # the trained model must have learned the planted signal
+    from factorvae_amd.data.sampler import init_data_loader
+    from factorvae_amd.utils import RankIC, generate_prediction_scores, test_args
+
+    loader = init_data_loader(df, step_len=8, shuffle=False,
+                              start="2015-03-02", end=None)
+    targs = test_args(run_name="t", num_factor=8, hidden_size=64,
+                      num_latent=158, num_portfolio=32, seq_length=8)
+    scores = generate_prediction_scores(m2.to(DEV), loader, loader.dataset,
+                                        targs)
+    merged = scores.join(df[["LABEL0"]], how="inner")
+    ic = float(RankIC(merged, column1="LABEL0", column2="score")["RankIC"].iloc[0])
+    assert abs(ic) == ic and ic > 0.25, f"RankIC {ic} after training"
 
 
 @pytest.mark.timeout(300)
