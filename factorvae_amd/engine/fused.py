@@ -313,13 +313,13 @@ class FusedTrainer:
         w["dgh"] = f(R, 3 * H)
         w["dxp"] = f(R, C)
         max_mn = max(3 * H * max(H, C), C * C, M * H, K * max(H, M))
-        w["tn_part"] = f(64 * max_mn)
-        w["tn_part2"] = f(64 * max_mn)
-        w["tn_part3"] = f(64 * max_mn)
+        w["tn_part"] = f(128 * max_mn)
+        w["tn_part2"] = f(128 * max_mn)
+        w["tn_part3"] = f(128 * max_mn)
         max_m = max(3 * H, C, M, K)
-        w["tn_partb"] = f(64 * max_m)
-        w["tn_partb2"] = f(64 * max_m)
-        w["tn_partb3"] = f(64 * max_m)
+        w["tn_partb"] = f(128 * max_m)
+        w["tn_partb2"] = f(128 * max_m)
+        w["tn_partb3"] = f(128 * max_m)
         # N-row reductions (attention/encoder/decoder wgrads + fwd u):
         # chunked over z so a 3500-stock day doesn't serialize 6 blocks
         small_mn = 32 * max(M, K, H) * H

@@ -250,7 +250,7 @@ void gemm_tn_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor out,
   const int R = A.size(0), M = A.size(1), N = B.size(1);
   TORCH_CHECK(B.size(0) == R && out.size(0) == M && out.size(1) == N);
   long zmax = (R + 511) / 512;
-  if (zmax > 64) zmax = 64;
+  if (zmax > 128) zmax = 128;
   if (zmax < 1 || r_chunks <= 1) zmax = 1;
   float* pp = nullptr;
   if (part.has_value()) {

@@ -464,9 +464,19 @@ __global__ __launch_bounds__(256) void tn_reduce_bf16_kernel(
     int z, int accumulate) {
   const long e = (long)blockIdx.x * 256 + threadIdx.x;
   if (e < elems) {
-    float s = accumulate ? out[e] : 0.0f;
-    for (int c = 0; c < z; ++c) s += part[(long)c * elems + e];
-    out[e] = s;
+    // 4 independent partial sums keep ~4 loads in flight per thread
+    // (the z-strided reads are latency-bound), summed in FIXED order
+    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    int c = 0;
+    for (; c + 4 <= z; c += 4) {
+      s0 += part[(long)c * elems + e];
+      s1 += part[(long)(c + 1) * elems + e];
+      s2 += part[(long)(c + 2) * elems + e];
+      s3 += part[(long)(c + 3) * elems + e];
+    }
+    float s = ((s0 + s1) + (s2 + s3));
+    for (; c < z; ++c) s += part[(long)c * elems + e];
+    out[e] = (accumulate ? out[e] : 0.0f) + s;
   } else if (e < elems + m_elems) {
     const long m = e - elems;
     float s = accumulate ? db[m] : 0.0f;
@@ -531,9 +541,10 @@ hipError_t fv_gemm_tn_bf16(const void* A, const void* B, float* out,
   if (!part) r_chunks = 1;
   if (r_chunks > 1) {
     // fill the chip: at TBM=TBN=64 the x/y grid is tiny (<=9 blocks for
-    // this model), so the z split is the only parallelism lever
+    // this model), so the z split is the only parallelism lever (4+
+    // resident blocks/CU hide the staging latency)
     r_chunks = (R + 511) / 512;
-    if (r_chunks > 64) r_chunks = 64;
+    if (r_chunks > 128) r_chunks = 128;
     if (r_chunks < 1) r_chunks = 1;
   }
   if (r_chunks > 1 && db && !db_part) r_chunks = 1;

@@ -163,8 +163,8 @@ def gemm_bench():
         B = torch.randn(R, N, device=dev).bfloat16()
         out = torch.zeros(M, N, device=dev)
         db = torch.zeros(M, device=dev)
-        part = torch.zeros(64 * M * N, device=dev)
-        db_part = torch.zeros(64 * M, device=dev)
+        part = torch.zeros(128 * M * N, device=dev)
+        db_part = torch.zeros(128 * M, device=dev)
         t = timeit(lambda: ext.gemm_tn_bf16(A, B, out, part, chunks, True, db, db_part), 20)
         fl = 2.0 * R * M * N
         print(f"gemm_tn_bf16 {R}x{M}x{N} z{chunks}+bias: {t*1e6:.1f} us  {fl/t/1e12:.1f} TF/s")

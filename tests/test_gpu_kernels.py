@@ -501,8 +501,8 @@ def test_gemm_tn_bf16(R, M, N, chunks):
     A, B = bt(R, M, seed=6, scale=0.1), bt(R, N, seed=7, scale=0.1)
     out = torch.zeros(M, N, device=DEV)
     db = torch.zeros(M, device=DEV)
-    part = torch.zeros(64 * M * N, device=DEV) if chunks > 1 else None
-    db_part = torch.zeros(64 * M, device=DEV) if chunks > 1 else None
+    part = torch.zeros(128 * M * N, device=DEV) if chunks > 1 else None
+    db_part = torch.zeros(128 * M, device=DEV) if chunks > 1 else None
     ext.gemm_tn_bf16(A, B, out, part, chunks, chunks > 1, db, db_part)
     torch.cuda.synchronize()
     ref = A.float().t() @ B.float()
