@@ -897,28 +897,53 @@ class FusedTrainer:
         self._launch_forward(N, T)
         return w["loss"]
 
-    def predict(self, x: torch.Tensor) -> torch.Tensor:
-        """model.prediction through the fused kernels: extractor ->
-        predictor (prior) -> decoder with prior mu/sigma."""
-        ext, p = self.ext, self.p
-        N, T, C = x.shape
-        self._ensure_ws(N, T)
-        w = self.ws
-        w["x"].copy_(x)
-        w["y"].zero_()
-        was_training = self.training
-        self.training = False  # prediction: dropout off
-        try:
-            self._fill_rng(N)
-            self._launch_forward(N, T, with_loss=False)
-        finally:
-            self.training = was_training
-        # decoder with PRIOR mu/sigma (module.py:273-278): psig clamp applies
+    def _launch_predict(self, N: int, T: int):
+        """Prediction kernel sequence: extractor -> predictor (prior) ->
+        decoder with PRIOR mu/sigma (module.py:273-278)."""
+        ext, p, w = self.ext, self.p, self.ws
+        self._launch_forward(N, T, with_loss=False)
         ext.dec_fwd(w["h"], p("W1d"), p("b1d"), p("wmu_d"), p("bmu_d"),
                     p("wsig_d"), p("bsig_d"), p("Wb"), p("bb"), w["pmu"],
                     w["psig_c"], w["eps"], w["recon"], w["a1"], w["beta"],
                     w["asig_pre"], w["sigma"])
-        return w["recon"].view(N, 1).clone()
+
+    def predict(self, x: torch.Tensor) -> torch.Tensor:
+        """model.prediction through the fused kernels; the sequence is
+        hipGraph-captured per (N, T) shape (eps refilled outside)."""
+        N, T, C = x.shape
+        self._ensure_ws(N, T)
+        w = self.ws
+        w["x"].copy_(x)
+        was_training = self.training
+        self.training = False  # prediction: dropout off
+        try:
+            self._fill_rng(N)
+            if not self.use_graph:
+                w["y"].zero_()
+                self._launch_predict(N, T)
+                return w["recon"].view(N, 1).clone()
+            key = ("predict", N, T)
+            if key not in self._graphs:
+                try:
+                    w["y"].zero_()
+                    torch.cuda.synchronize(self.device)
+                    self._launch_predict(N, T)  # warmup
+                    torch.cuda.synchronize(self.device)
+                    gp = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(gp):
+                        self._launch_predict(N, T)
+                    self._graphs[key] = {"g": gp}
+                except Exception:
+                    torch.cuda.synchronize(self.device)
+                    self._graphs[key] = {"g": None}
+            gp = self._graphs[key]["g"]
+            if gp is None:
+                self._launch_predict(N, T)
+            else:
+                gp.replay()
+            return w["recon"].view(N, 1).clone()
+        finally:
+            self.training = was_training
 
     # ------------------------------------------------------------- epochs
     def train_epoch(self, days, shuffle_order=None) -> float:

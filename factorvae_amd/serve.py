@@ -60,8 +60,7 @@ class ScoringEngine:
                 from .engine.fused import FusedTrainer
 
                 self.trainer = FusedTrainer(self.model, lr=0.0, t_max=1,
-                                            device=self.device,
-                                            use_graph=False, train=False)
+                                            device=self.device, train=False)
                 self.engine = "fused"
             except Exception:
                 self.trainer = None

@@ -110,7 +110,7 @@ def generate_prediction_scores(model, test_dataloader, test_dataset, args,
             from .engine.fused import FusedTrainer
 
             trainer = FusedTrainer(model, lr=0.0, t_max=1, device=device,
-                                   use_graph=False, train=False)
+                                   train=False)
         except Exception:
             trainer = None
 

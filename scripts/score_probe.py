@@ -16,7 +16,7 @@ batches = [(x[:, :, :-1].contiguous()) for x,_ in loader]
 t1=time.perf_counter()
 print(f"loader+slice: {len(batches)} days in {t1-t0:.2f}s = {len(batches)/(t1-t0):.0f}/s")
 
-trainer = FusedTrainer(model, lr=0.0, t_max=1, device=DEV, use_graph=False, train=False)
+trainer = FusedTrainer(model, lr=0.0, t_max=1, device=DEV, train=False)
 x0 = batches[0].to(DEV)
 trainer.predict(x0); torch.cuda.synchronize()
 t0=time.perf_counter()
