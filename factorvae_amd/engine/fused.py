@@ -549,11 +549,11 @@ class FusedTrainer:
         ext.gemm_nn(w["ds"], w["qk"], None, w["dh"], 1.0, True, False)
         ext.gemm_nn(w["a_att"], w["du"], None, w["dh"], 1.0, True, False)
 
-        # encoder backward
-        ext.enc_heads_bwd(w["dfmu"], w["dfsig_c"], w["fsig"], w["fsig_pre"],
-                          w["yp"], p("Wmu_e"), p("Wsig_e"), w["dyp"],
-                          g("Wmu_e"), g("bmu_e"), g("Wsig_e"), g("bsig_e"))
-        ext.enc_softmax_bwd(w["dyp"], w["a_enc"], yv, w["dscores"])
+        # encoder backward (heads + stock-axis softmax bwd, one kernel)
+        ext.enc_bwd_fused(w["dfmu"], w["dfsig_c"], w["fsig"], w["fsig_pre"],
+                          w["yp"], p("Wmu_e"), p("Wsig_e"), w["a_enc"], yv,
+                          w["dscores"], g("Wmu_e"), g("bmu_e"), g("Wsig_e"),
+                          g("bsig_e"))
         fork()
         with _on_side():
             ext.gemm_tn(w["dscores"], w["h"], g("Wenc"), w["tn_part_s"], 2,

@@ -55,6 +55,11 @@ hipError_t fv_enc_softmax_fwd(const float*, const float*, float*, float*, int,
 hipError_t fv_enc_fused_fwd(const float*, const float*, const float*,
                             const float*, float*, float*, float*, int, int,
                             int, hipStream_t);
+hipError_t fv_enc_bwd_fused(const float*, const float*, const float*,
+                            const float*, const float*, const float*,
+                            const float*, const float*, const float*,
+                            float*, float*, float*, float*, float*, int,
+                            int, int, hipStream_t);
 hipError_t fv_enc_softmax_bwd(const float*, const float*, const float*, float*,
                               int, int, hipStream_t);
 hipError_t fv_enc_heads_fwd(const float*, const float*, const float*,
@@ -528,6 +533,21 @@ void enc_fused_fwd(torch::Tensor h, torch::Tensor Wenc, torch::Tensor benc,
                        fpm(a), fpm(yp), N, M, H, cur_stream()));
 }
 
+void enc_bwd_fused(torch::Tensor dfmu, torch::Tensor dfsig_c,
+                   torch::Tensor fsig, torch::Tensor fsig_pre,
+                   torch::Tensor yp, torch::Tensor Wmu, torch::Tensor Wsig,
+                   torch::Tensor a, torch::Tensor y, torch::Tensor dscores,
+                   torch::Tensor dWmu, torch::Tensor dbmu,
+                   torch::Tensor dWsig, torch::Tensor dbsig) {
+  CK(dfmu); CK(fsig); CK(yp); CK(Wmu); CK(Wsig); CK(a); CK(y); CK(dscores);
+  CK(dWmu); CK(dbmu); CK(dWsig); CK(dbsig);
+  const int N = a.size(0), M = a.size(1), K = Wmu.size(0);
+  RUN(fv_enc_bwd_fused(fp(dfmu), fp(dfsig_c), fp(fsig), fp(fsig_pre),
+                       fp(yp), fp(Wmu), fp(Wsig), fp(a), fp(y),
+                       fpm(dscores), fpm(dWmu), fpm(dbmu), fpm(dWsig),
+                       fpm(dbsig), N, M, K, cur_stream()));
+}
+
 void enc_softmax_fwd(torch::Tensor scores, torch::Tensor y, torch::Tensor a,
                      torch::Tensor yp) {
   CK(scores); CK(y); CK(a); CK(yp);
@@ -757,6 +777,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gru_fwd_mfma", &gru_fwd_mfma);
   mod.def("attn_fused_fwd", &attn_fused_fwd);
   mod.def("enc_fused_fwd", &enc_fused_fwd);
+  mod.def("enc_bwd_fused", &enc_bwd_fused);
   mod.def("attn_fused_bwd", &attn_fused_bwd);
   mod.def("gru_bwd_mfma", &gru_bwd_mfma, py::arg("dh_final"),
           py::arg("h_prev"), py::arg("gates4"), py::arg("whh_bf"),

@@ -115,6 +115,69 @@ __global__ __launch_bounds__(256) void enc_softmax_bwd_kernel(
   }
 }
 
+
+// Fused encoder backward: head grads + dyp + stock-axis softmax backward
+// in ONE kernel. One workgroup per portfolio column m; the K-sized head
+// chain (dmu/dsig/dyp[m]) is recomputed per WG (K*2 fma — free), the
+// m==0 workgroup additionally emits the bias grads, and each WG writes
+// its own column of dWmu/dWsig (plain stores, deterministic).
+__global__ __launch_bounds__(256) void enc_bwd_fused_kernel(
+    const float* __restrict__ dfmu, const float* __restrict__ dfsig_c,
+    const float* __restrict__ fsig, const float* __restrict__ fsig_pre,
+    const float* __restrict__ yp, const float* __restrict__ Wmu,
+    const float* __restrict__ Wsig, const float* __restrict__ a,
+    const float* __restrict__ y, float* __restrict__ dscores,
+    float* __restrict__ dWmu, float* __restrict__ dbmu,
+    float* __restrict__ dWsig, float* __restrict__ dbsig,
+    int N, int M, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* dmuS = (float*)smem;      // [K]
+  float* dsigS = dmuS + K;         // [K]
+  float* scratch = dsigS + K;      // [8]
+  __shared__ float dypS;
+
+  const int m = blockIdx.x;
+  const int tid = threadIdx.x;
+
+  for (int k = tid; k < K; k += 256) {
+    const float dm = dfmu[k];
+    const float dsc = (fsig[k] == 0.0f) ? 0.0f : dfsig_c[k];
+    const float dsg = dsc * softplus_gradf_(fsig_pre[k]);
+    dmuS[k] = dm;
+    dsigS[k] = dsg;
+    // column m of the head weight grads (plain +=, column-owned)
+    dWmu[(long)k * M + m] += dm * yp[m];
+    dWsig[(long)k * M + m] += dsg * yp[m];
+    if (m == 0) {
+      dbmu[k] += dm;
+      dbsig[k] += dsg;
+    }
+  }
+  __syncthreads();
+
+  // dyp[m] = sum_k dmu[k]*Wmu[k][m] + dsig[k]*Wsig[k][m]
+  if (tid == 0) {
+    float acc = 0.0f;
+    for (int k = 0; k < K; ++k)
+      acc += dmuS[k] * Wmu[(long)k * M + m] + dsigS[k] * Wsig[(long)k * M + m];
+    dypS = acc;
+  }
+  __syncthreads();
+  const float g = dypS;
+
+  float t = 0.0f;
+  for (int n = tid; n < N; n += 256) {
+    const float an = a[(long)n * M + m];
+    t = fmaf(an, g * y[n], t);
+  }
+  t = block_reduce_sum(t, scratch);
+
+  for (int n = tid; n < N; n += 256) {
+    const float an = a[(long)n * M + m];
+    dscores[(long)n * M + m] = an * (g * y[n] - t);
+  }
+}
+
 // yp (M) -> fmu (K), fsig_pre (K), fsig (K), fsig_c (K) (the decoder's
 // in-place ==0 -> 1e-6 clamp, module.py:117). One workgroup.
 __global__ __launch_bounds__(256) void enc_heads_fwd_kernel(
@@ -181,6 +244,21 @@ __global__ __launch_bounds__(256) void enc_heads_bwd_kernel(
 }
 
 extern "C" {
+
+hipError_t fv_enc_bwd_fused(const float* dfmu, const float* dfsig_c,
+                            const float* fsig, const float* fsig_pre,
+                            const float* yp, const float* Wmu,
+                            const float* Wsig, const float* a, const float* y,
+                            float* dscores, float* dWmu, float* dbmu,
+                            float* dWsig, float* dbsig, int N, int M, int K,
+                            hipStream_t s) {
+  const size_t lds = ((size_t)2 * K + 8) * sizeof(float);
+  hipLaunchKernelGGL(enc_bwd_fused_kernel, dim3(M), dim3(256), lds, s,
+                     dfmu, dfsig_c, fsig, fsig_pre, yp, Wmu, Wsig, a, y,
+                     dscores, dWmu, dbmu, dWsig, dbsig, N, M, K);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
 
 hipError_t fv_enc_fused_fwd(const float* h, const float* Wenc,
                             const float* benc, const float* y,
