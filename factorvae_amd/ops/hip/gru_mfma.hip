@@ -72,22 +72,19 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_kernel(
   const long erow = (long)(s0 + es) * T;
 
   // per-thread gi prefetch registers (12 floats: r,z,n x 4 units)
-  // double-buffered gi prefetch: the NEXT step's rows are issued at the
-  // top of the current iteration, hiding the global latency under the
-  // whole MFMA+gates pipeline of this step
-  float pgr[2][4], pgz[2][4], pgn[2][4];
-  auto gi_load = [&](int t, int buf) {
+  float pgr[4], pgz[4], pgn[4];
+  auto gi_load = [&](int t) {
     if (elive) {
       const float* g = gi + (erow + t) * 192;
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
-        pgr[buf][u] = g[ej + u];
-        pgz[buf][u] = g[64 + ej + u];
-        pgn[buf][u] = g[128 + ej + u];
+        pgr[u] = g[ej + u];
+        pgz[u] = g[64 + ej + u];
+        pgn[u] = g[128 + ej + u];
       }
     }
   };
-  gi_load(0, 0);
+  gi_load(0);
 
   float bh_r[4], bh_z[4], bh_n[4];
 #pragma unroll
@@ -98,8 +95,6 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_kernel(
   }
 
   for (int t = 0; t < T; ++t) {
-    const int pb = t & 1;
-    if (t + 1 < T) gi_load(t + 1, 1 - pb);
     // ---- MFMA phase: gh = h @ Whh^T for this wave's 48 columns
     f32x4 acc[3] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
 #pragma unroll
@@ -126,10 +121,10 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_kernel(
       for (int u = 0; u < 4; ++u) {
         const int j = ej + u;
         const float hp = hS[es][j];
-        const float r = sigmoidf_(pgr[pb][u] + ghS[es][j] + bh_r[u]);
-        const float z = sigmoidf_(pgz[pb][u] + ghS[es][64 + j] + bh_z[u]);
+        const float r = sigmoidf_(pgr[u] + ghS[es][j] + bh_r[u]);
+        const float z = sigmoidf_(pgz[u] + ghS[es][64 + j] + bh_z[u]);
         const float q = ghS[es][128 + j] + bh_n[u];
-        const float n = tanhf(fmaf(r, q, pgn[pb][u]));
+        const float n = tanhf(fmaf(r, q, pgn[u]));
         hn[u] = fmaf(z, hp - n, n);
         gr4[u] = r; gz4[u] = z; gn4[u] = n; gq4[u] = q; hp4[u] = hp;
       }
@@ -144,6 +139,7 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_kernel(
       *(f32x4*)&g4[192 + ej] = *(f32x4*)gq4;
       if (t == T - 1) *(f32x4*)&h_final[(long)(s0 + es) * 64 + ej] = *(f32x4*)hn;
     }
+    if (t + 1 < T) gi_load(t + 1);
     // state update (each (stock, unit) owned by one thread)
 #pragma unroll
     for (int u = 0; u < 4; ++u) {
@@ -206,38 +202,20 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
   }
   __syncthreads();
 
-  // double-buffered prefetch of this thread's gates4/h_prev rows: step
-  // t-1's 20 floats are issued at the top of iteration t, hiding the
-  // global latency under the gate-grad math + MFMA of step t
-  f32x4 pr_[2], pz_[2], pn_[2], pq_[2], ph_[2];
-  auto pre_load = [&](int t, int buf) {
-    if (elive) {
-      const long tb = erow + t;
-      const float* g4 = &gates4[tb * 256];
-      pr_[buf] = *(const f32x4*)&g4[ej];
-      pz_[buf] = *(const f32x4*)&g4[64 + ej];
-      pn_[buf] = *(const f32x4*)&g4[128 + ej];
-      pq_[buf] = *(const f32x4*)&g4[192 + ej];
-      ph_[buf] = *(const f32x4*)&h_prev_in[tb * 64 + ej];
-    }
-  };
-  pre_load(T - 1, (T - 1) & 1);
-
   for (int t = T - 1; t >= 0; --t) {
-    const int pb = t & 1;
-    if (t > 0) pre_load(t - 1, 1 - pb);
     // ---- elementwise phase: gate grads for this step
     if (elive) {
       const long tb = erow + t;
+      const float* g4 = &gates4[tb * 256];
       float dgr4[4], dgz4[4], da4[4], dghn4[4];
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         const int j = ej + u;
-        const float r = pr_[pb][u];
-        const float z = pz_[pb][u];
-        const float n = pn_[pb][u];
-        const float q = pq_[pb][u];
-        const float hp = ph_[pb][u];
+        const float r = g4[j];
+        const float z = g4[64 + j];
+        const float n = g4[128 + j];
+        const float q = g4[192 + j];
+        const float hp = h_prev_in[tb * 64 + j];
         const float dh = dhS[es][j];
         const float dz = dh * (hp - n);
         const float dn = dh * (1.0f - z);
@@ -338,19 +316,19 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_f32_kernel(
   const bool elive = (s0 + es) < N;
   const long erow = (long)(s0 + es) * T;
 
-  float pgr[2][4], pgz[2][4], pgn[2][4];
-  auto gi_load = [&](int t, int buf) {
+  float pgr[4], pgz[4], pgn[4];
+  auto gi_load = [&](int t) {
     if (elive) {
       const float* g = gi + (erow + t) * 192;
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
-        pgr[buf][u] = g[ej + u];
-        pgz[buf][u] = g[64 + ej + u];
-        pgn[buf][u] = g[128 + ej + u];
+        pgr[u] = g[ej + u];
+        pgz[u] = g[64 + ej + u];
+        pgn[u] = g[128 + ej + u];
       }
     }
   };
-  gi_load(0, 0);
+  gi_load(0);
 
   float bh_r[4], bh_z[4], bh_n[4];
 #pragma unroll
@@ -361,8 +339,6 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_f32_kernel(
   }
 
   for (int t = 0; t < T; ++t) {
-    const int pb = t & 1;
-    if (t + 1 < T) gi_load(t + 1, 1 - pb);
     f32x4 acc[3] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
 #pragma unroll
     for (int ks = 0; ks < 16; ++ks) {
@@ -387,10 +363,10 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_f32_kernel(
       for (int u = 0; u < 4; ++u) {
         const int j = ej + u;
         const float hp = hS[es][j];
-        const float r = sigmoidf_(pgr[pb][u] + ghS[es][j] + bh_r[u]);
-        const float z = sigmoidf_(pgz[pb][u] + ghS[es][64 + j] + bh_z[u]);
+        const float r = sigmoidf_(pgr[u] + ghS[es][j] + bh_r[u]);
+        const float z = sigmoidf_(pgz[u] + ghS[es][64 + j] + bh_z[u]);
         const float q = ghS[es][128 + j] + bh_n[u];
-        const float n = tanhf(fmaf(r, q, pgn[pb][u]));
+        const float n = tanhf(fmaf(r, q, pgn[u]));
         hn[u] = fmaf(z, hp - n, n);
         gr4[u] = r; gz4[u] = z; gn4[u] = n; gq4[u] = q; hp4[u] = hp;
       }
@@ -405,6 +381,7 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_f32_kernel(
       if (t == T - 1)
         *(f32x4*)&h_final[(long)(s0 + es) * 64 + ej] = *(f32x4*)hn;
     }
+    if (t + 1 < T) gi_load(t + 1);
 #pragma unroll
     for (int u = 0; u < 4; ++u)
       hS[es][ej + u] = elive ? hn[u] : 0.0f;
@@ -450,34 +427,19 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_f32_kernel(
   }
   __syncthreads();
 
-  f32x4 pr_[2], pz_[2], pn_[2], pq_[2], ph_[2];
-  auto pre_load = [&](int t, int buf) {
+  for (int t = T - 1; t >= 0; --t) {
     if (elive) {
       const long tb = erow + t;
       const float* g4 = &gates4[tb * 256];
-      pr_[buf] = *(const f32x4*)&g4[ej];
-      pz_[buf] = *(const f32x4*)&g4[64 + ej];
-      pn_[buf] = *(const f32x4*)&g4[128 + ej];
-      pq_[buf] = *(const f32x4*)&g4[192 + ej];
-      ph_[buf] = *(const f32x4*)&h_prev_in[tb * 64 + ej];
-    }
-  };
-  pre_load(T - 1, (T - 1) & 1);
-
-  for (int t = T - 1; t >= 0; --t) {
-    const int pb = t & 1;
-    if (t > 0) pre_load(t - 1, 1 - pb);
-    if (elive) {
-      const long tb = erow + t;
       float dgr4[4], dgz4[4], da4[4], dghn4[4];
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         const int j = ej + u;
-        const float r = pr_[pb][u];
-        const float z = pz_[pb][u];
-        const float n = pn_[pb][u];
-        const float q = pq_[pb][u];
-        const float hp = ph_[pb][u];
+        const float r = g4[j];
+        const float z = g4[64 + j];
+        const float n = g4[128 + j];
+        const float q = g4[192 + j];
+        const float hp = h_prev_in[tb * 64 + j];
         const float dh = dhS[es][j];
         const float dz = dh * (hp - n);
         const float dn = dh * (1.0f - z);
