@@ -179,14 +179,17 @@ def train_main(args, data_args, df=None) -> float:
         # unsharded loaders -> every rank caches all days once (288 GB
         # HBM3E), then each epoch does the sampler-equivalent global
         # shuffle + round-robin shard on device-resident tensors
+        nw = int(getattr(args, "num_workers", 0) or 0)
         train_cache = DeviceEpochCache(init_data_loader(
             df, shuffle=False, step_len=data_args.seq_len,
             start=data_args.start_time, end=data_args.fit_end_time,
-            select_feature=data_args.select_feature), device, seed=args.seed)
+            select_feature=data_args.select_feature, num_workers=nw),
+            device, seed=args.seed)
         valid_cache = DeviceEpochCache(init_data_loader(
             df, shuffle=False, step_len=data_args.seq_len,
             start=data_args.val_start_time, end=data_args.val_end_time,
-            select_feature=data_args.select_feature), device, seed=args.seed)
+            select_feature=data_args.select_feature, num_workers=nw),
+            device, seed=args.seed)
 
         steps_per_epoch = train_cache.num_batches(rank, world_size)
         trainer = FusedTrainer(factorVAE, lr=args.lr,
@@ -200,6 +203,7 @@ def train_main(args, data_args, df=None) -> float:
             start=data_args.start_time, end=data_args.fit_end_time,
             select_feature=data_args.select_feature,
             rank=rank, world_size=world_size, seed=args.seed,
+            num_workers=int(getattr(args, "num_workers", 0) or 0),
         )
         valid_dataloader = init_data_loader(
             df, shuffle=False, step_len=data_args.seq_len,
