@@ -97,9 +97,12 @@ __global__ __launch_bounds__(256) void ln_bwd_params_kernel(
   pb[w][lane] = db;
   __syncthreads();
   if (w == 0 && c < C) {
-    float* po = part + (long)blockIdx.y * 2 * C;
-    po[c] = pg[0][lane] + pg[1][lane] + pg[2][lane] + pg[3][lane];
-    po[C + c] = pb[0][lane] + pb[1][lane] + pb[2][lane] + pb[3][lane];
+    // TRANSPOSED partial layout part[e][yblock]: the reduce kernel then
+    // reads each column's partials contiguously instead of strided
+    part[(long)c * gridDim.y + blockIdx.y] =
+        pg[0][lane] + pg[1][lane] + pg[2][lane] + pg[3][lane];
+    part[((long)C + c) * gridDim.y + blockIdx.y] =
+        pb[0][lane] + pb[1][lane] + pb[2][lane] + pb[3][lane];
   }
 }
 
@@ -108,8 +111,14 @@ __global__ __launch_bounds__(256) void ln_bwd_reduce_kernel(
     float* __restrict__ dbeta, int yblocks, int C) {
   const int e = blockIdx.x * 256 + threadIdx.x;
   if (e >= 2 * C) return;
-  float s = 0.0f;
-  for (int z = 0; z < yblocks; ++z) s += part[(long)z * 2 * C + e];
+  const float* pe = part + (long)e * yblocks;
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+  int z = 0;
+  for (; z + 4 <= yblocks; z += 4) {
+    s0 += pe[z]; s1 += pe[z + 1]; s2 += pe[z + 2]; s3 += pe[z + 3];
+  }
+  float s = (s0 + s1) + (s2 + s3);
+  for (; z < yblocks; ++z) s += pe[z];
   if (e < C) dgamma[e] += s;
   else dbeta[e - C] += s;
 }
