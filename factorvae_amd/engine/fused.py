@@ -327,7 +327,9 @@ class FusedTrainer:
         w["tn_part_s"] = f(small_mn)
         w["tn_partb_s"] = f(32 * max_m)
         # deterministic shared-grad partials (no float atomics anywhere)
-        w["dec_part"] = f(((N + 31) // 32 + 1) * (2 * K + 2 * H + 2))
+        dec_iters = min(8, max(1, (N + 511) // 512))
+        dec_nblk = (N + 4 * dec_iters - 1) // (4 * dec_iters) + 1
+        w["dec_part"] = f(dec_nblk * (2 * K + 2 * H + 2))
         w["hpart"] = f(K * (2 * H + 2))
         cb = (C + 63) // 64
         ty = (2048 + cb - 1) // cb
@@ -641,9 +643,9 @@ class FusedTrainer:
                 main.wait_event(e)
 
     def _refresh_bf16_shadows(self):
-        self.ext.cast_f32_bf16(self.p("W1x"), self.w1x_bf)
-        self.ext.cast_f32_bf16(self.p("Wih"), self.wih_bf)
-        self.ext.cast_f32_bf16(self.p("Whh"), self.whh_bf)
+        self.ext.cast3_f32_bf16(self.p("W1x"), self.w1x_bf,
+                                self.p("Wih"), self.wih_bf,
+                                self.p("Whh"), self.whh_bf)
         if self.fp8:
             self.ext.absmax_scale(self.p("W1x"), self.s_w1x, self.is_w1x)
             self.ext.cast_f32_fp8_scaled(self.p("W1x"), self.w1x_f8,

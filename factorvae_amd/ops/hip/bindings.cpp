@@ -24,6 +24,8 @@ hipError_t fv_gemm_nn_bf16(const void*, const void*, const float*, float*,
 hipError_t fv_gemm_tn_bf16(const void*, const void*, float*, float*, float*,
                            float*, int, int, int, int, int, hipStream_t);
 hipError_t fv_cast_f32_bf16(const float*, void*, long, hipStream_t);
+hipError_t fv_cast3_f32_bf16(const float*, void*, long, const float*, void*,
+                             long, const float*, void*, long, hipStream_t);
 hipError_t fv_lrelu_bwd_bf16(const void*, const void*, void*, long,
                              hipStream_t);
 hipError_t fv_lrelu_bwd(const float*, const float*, float*, long, hipStream_t);
@@ -272,6 +274,14 @@ void gemm_tn_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor out,
   }
   RUN(fv_gemm_tn_bf16(bfpc(A), bfpc(B), fpm(out), pp, dbp, dbpp, R, M, N,
                       (int)r_chunks, accumulate, cur_stream()));
+}
+
+void cast3_f32_bf16(torch::Tensor s0, torch::Tensor d0, torch::Tensor s1,
+                    torch::Tensor d1, torch::Tensor s2, torch::Tensor d2) {
+  CK(s0); CKB(d0); CK(s1); CKB(d1); CK(s2); CKB(d2);
+  RUN(fv_cast3_f32_bf16(fp(s0), bfp(d0), s0.numel(), fp(s1), bfp(d1),
+                        s1.numel(), fp(s2), bfp(d2), s2.numel(),
+                        cur_stream()));
 }
 
 void cast_f32_bf16(torch::Tensor src, torch::Tensor dst) {
@@ -694,7 +704,10 @@ void dec_bwd(torch::Tensor drecon, torch::Tensor h, torch::Tensor a1,
   CK(dbeta); CK(part); CK(dfmu); CK(dfsig_c); CK(dwmu); CK(dbmu); CK(dwsig);
   CK(dbsig);
   const int N_ = h.size(0), K_ = Wb.size(0), H_ = h.size(1);
-  const long nblk = (N_ + 31) / 32;
+  long iters = (N_ + 4 * 128 - 1) / (4 * 128);
+  if (iters > 8) iters = 8;
+  if (iters < 1) iters = 1;
+  const long nblk = (N_ + 4 * iters - 1) / (4 * iters);
   TORCH_CHECK(part.numel() >= nblk * (2 * K_ + 2 * H_ + 2));
   RUN(fv_dec_bwd(fp(drecon), fp(h), fp(a1), fp(beta), fp(asig_pre), fp(sigma),
                  fp(eps), fp(fmu), fp(fsig_c), fp(W1), fp(wmu), fp(wsig),
@@ -760,6 +773,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           py::arg("accumulate"), py::arg("db") = py::none(),
           py::arg("db_part") = py::none());
   mod.def("cast_f32_bf16", &cast_f32_bf16);
+  mod.def("cast3_f32_bf16", &cast3_f32_bf16);
   mod.def("lrelu_bwd_bf16", &lrelu_bwd_bf16);
   mod.def("lrelu_bwd", &lrelu_bwd);
   mod.def("ln_fwd", &ln_fwd, py::arg("x"), py::arg("gamma"), py::arg("beta"),

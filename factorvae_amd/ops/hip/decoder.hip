@@ -98,7 +98,8 @@ __global__ __launch_bounds__(256) void dec_fwd_kernel(
 // PER-BLOCK PARTIALS (layout [dfmu K][dfsig K][dwmu H][dwsig H][2]),
 // summed in fixed block order by dec_bwd_reduce_kernel — bit-exact
 // run-to-run determinism, no float atomics.
-#define DEC_ITERS 8
+// rows-per-block iteration count is a runtime arg: small days want
+// many blocks (fill the CUs), big days want fewer partial slices
 __global__ __launch_bounds__(256) void dec_bwd_kernel(
     const float* __restrict__ drecon, const float* __restrict__ h,
     const float* __restrict__ a1, const float* __restrict__ beta,
@@ -109,7 +110,7 @@ __global__ __launch_bounds__(256) void dec_bwd_kernel(
     const float* __restrict__ Wb, float* __restrict__ dh,
     float* __restrict__ dz1, float* __restrict__ dbeta_out,
     float* __restrict__ part,
-    int N, int K, int H) {
+    int N, int K, int H, int iters) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* W1S = (float*)smem;                 // [H][H] as-is
   float* WbS = W1S + (size_t)H * H;          // [K][H] as-is
@@ -131,8 +132,8 @@ __global__ __launch_bounds__(256) void dec_bwd_kernel(
   float rwmu = 0.f, rwsig = 0.f;                               // i=lane
   float rbmu = 0.f, rbsig = 0.f;
 
-  for (int it = 0; it < DEC_ITERS; ++it) {
-    const int row = (blockIdx.x * DEC_ITERS + it) * DEC_RPW + w;
+  for (int it = 0; it < iters; ++it) {
+    const int row = (blockIdx.x * iters + it) * DEC_RPW + w;
     const bool live = row < N;
     float dmu = 0.f, dvar = 0.f, dasig_pre = 0.f, a1v = 0.f;
     if (live) {
@@ -292,10 +293,14 @@ hipError_t fv_dec_bwd(const float* drecon, const float* h, const float* a1,
   if (H > 64 || K > 128) return hipErrorInvalidValue;
   const size_t lds = ((size_t)H * H + (size_t)K * H + DEC_RPW * H +
                       (size_t)DEC_RPW * K + 4 * 64) * sizeof(float);
-  const int nblk = (N + DEC_RPW * DEC_ITERS - 1) / (DEC_RPW * DEC_ITERS);
+  // ~128+ blocks to fill the chip; <=8 row-iterations per block
+  int iters = (N + DEC_RPW * 128 - 1) / (DEC_RPW * 128);
+  if (iters > 8) iters = 8;
+  if (iters < 1) iters = 1;
+  const int nblk = (N + DEC_RPW * iters - 1) / (DEC_RPW * iters);
   hipLaunchKernelGGL(dec_bwd_kernel, dim3(nblk), dim3(256), lds, s,
                      drecon, h, a1, beta, asig_pre, sigma, eps, fmu, fsig_c,
-                     W1, wmu, wsig, Wb, dh, dz1, dbeta, part, N, K, H);
+                     W1, wmu, wsig, Wb, dh, dz1, dbeta, part, N, K, H, iters);
   HIP_CHECK_LAST();
   const int E = 2 * K + 2 * H + 2;
   hipLaunchKernelGGL(dec_bwd_reduce_kernel, dim3((E + 255) / 256), dim3(256),

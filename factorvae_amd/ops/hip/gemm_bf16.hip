@@ -492,6 +492,21 @@ __global__ __launch_bounds__(256) void cast_f32_bf16_kernel(
   if (i < n) dst[i] = (__bf16)src[i];
 }
 
+// refresh all three extractor weight shadows in ONE launch (they are
+// re-cast after every Adam step; 3 tiny kernels were ~15 us of pure
+// launch+latency)
+__global__ __launch_bounds__(256) void cast3_f32_bf16_kernel(
+    const float* __restrict__ s0, __bf16* __restrict__ d0, long n0,
+    const float* __restrict__ s1, __bf16* __restrict__ d1, long n1,
+    const float* __restrict__ s2, __bf16* __restrict__ d2, long n2) {
+  long i = (long)blockIdx.x * 256 + threadIdx.x;
+  if (i < n0) { d0[i] = (__bf16)s0[i]; return; }
+  i -= n0;
+  if (i < n1) { d1[i] = (__bf16)s1[i]; return; }
+  i -= n1;
+  if (i < n2) d2[i] = (__bf16)s2[i];
+}
+
 // dZ(bf16) = dY(bf16) * lrelu'(Y(bf16))
 __global__ __launch_bounds__(256) void lrelu_bwd_bf16_kernel(
     const __bf16* __restrict__ dY, const __bf16* __restrict__ Y,
@@ -562,6 +577,19 @@ hipError_t fv_gemm_tn_bf16(const void* A, const void* B, float* out,
                        accumulate);
     HIP_CHECK_LAST();
   }
+  return hipSuccess;
+}
+
+hipError_t fv_cast3_f32_bf16(const float* s0, void* d0, long n0,
+                             const float* s1, void* d1, long n1,
+                             const float* s2, void* d2, long n2,
+                             hipStream_t stream) {
+  const long total = n0 + n1 + n2;
+  dim3 grid((unsigned)((total + 255) / 256));
+  hipLaunchKernelGGL(cast3_f32_bf16_kernel, grid, dim3(256), 0, stream,
+                     s0, (__bf16*)d0, n0, s1, (__bf16*)d1, n1, s2,
+                     (__bf16*)d2, n2);
+  HIP_CHECK_LAST();
   return hipSuccess;
 }
 
