@@ -18,6 +18,10 @@ Design:
   Adam is one kernel over the arena;
 - eps / dropout-mask RNG uses torch's generator OUTSIDE the graph (two
   tiny kernels per step), keeping set_seed reproducibility;
+- four HIP streams per step: the main activation chain, two side
+  streams for the weight-gradient reductions, and (in DP runs) a comm
+  stream whose early all-reduce of the non-extractor gradient slice
+  overlaps the extractor backward;
 - loss stays on device; reading it is the caller's (async) choice.
 """
 
@@ -147,8 +151,6 @@ class FusedTrainer:
                        if self.device.type == "cuda" else None)
 
     # ---------------------------------------------------------------- params
-    _STACKED = ("q_att", "Wk", "bk", "Wv", "bv")
-
     def _param_specs(self):
         m = self.model
         C, H, M, K = self.C, self.H, self.M, self.K
@@ -307,7 +309,6 @@ class FusedTrainer:
         w["ds"] = f(N, K)
         w["dc"] = f(K)
         w["dqk"] = f(K, H)
-        w["dyp"] = f(M)
         w["dscores"] = f(N, M)
         w["dgi"] = f(R, 3 * H)
         w["dgh"] = f(R, 3 * H)
