@@ -21,12 +21,16 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 // act: 0 = none, 1 = leaky_relu(0.01)
 // flags bit0: accumulate into out; bit1: apply act; bit2: bias present
+// Register double buffer: the next k-tile's global loads issue while
+// the current tile's MFMAs run (f32 MFMA is only 1/16 of bf16 peak, but
+// without the prefetch these mid-size GEMMs were global-latency-bound
+// at ~12 TF/s).
 __global__ __launch_bounds__(256) void gemm_nt_kernel(
     const float* __restrict__ A, const float* __restrict__ W,
     const float* __restrict__ bias, float* __restrict__ out,
     int R, int Ci, int Co, float alpha, int flags) {
-  __shared__ float As[BR][BK + 1];
-  __shared__ float Ws[BC][BK + 1];
+  __shared__ float As[2][BR][BK + 1];
+  __shared__ float Ws[2][BC][BK + 1];
 
   const int r0 = blockIdx.x * BR;
   const int c0 = blockIdx.y * BC;
@@ -36,35 +40,65 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
   const int fi = lane & 15;       // fragment row/col index
   const int fk = lane >> 4;       // fragment k index (0..3)
 
-  // MFMA f32 16x16x4: wave wv computes rows [wv*16, wv*16+16) x all 64
-  // cols as 4 independent 16x16 accumulators (hides the 40-cyc
-  // dependent-accumulator latency at the 32-cyc issue rate).
+  // staging: thread -> (row = tid/32 + 8u, k-pair kp = (tid%32))
+  const int s_row0 = tid >> 5;     // + 8*u
+  const int s_k = tid & 31;        // BK = 32 columns
+
   f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
 
-  for (int k0 = 0; k0 < Ci; k0 += BK) {
-    // stage A tile (BR x BK) and W tile (BC x BK), zero-padded at edges
-    for (int idx = tid; idx < BR * BK; idx += 256) {
-      const int i = idx / BK, k = idx % BK;
-      const int gr = r0 + i, gk = k0 + k;
-      As[i][k] = (gr < R && gk < Ci) ? A[(long)gr * Ci + gk] : 0.0f;
+  const int ktiles = (Ci + BK - 1) / BK;
+  float pa[8], pw[8];
+  auto stage_regs = [&](int k0) {
+    const bool interior = (r0 + BR <= R) && (c0 + BC <= Co) &&
+                          (k0 + BK <= Ci);
+    if (interior) {
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const int row = s_row0 + 8 * u;
+        pa[u] = A[(long)(r0 + row) * Ci + k0 + s_k];
+        pw[u] = W[(long)(c0 + row) * Ci + k0 + s_k];
+      }
+    } else {
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const int row = s_row0 + 8 * u;
+        const int gk = k0 + s_k;
+        pa[u] = (r0 + row < R && gk < Ci)
+                    ? A[(long)(r0 + row) * Ci + gk] : 0.0f;
+        pw[u] = (c0 + row < Co && gk < Ci)
+                    ? W[(long)(c0 + row) * Ci + gk] : 0.0f;
+      }
     }
-    for (int idx = tid; idx < BC * BK; idx += 256) {
-      const int j = idx / BK, k = idx % BK;
-      const int gc = c0 + j, gk = k0 + k;
-      Ws[j][k] = (gc < Co && gk < Ci) ? W[(long)gc * Ci + gk] : 0.0f;
+  };
+  auto regs_to_lds = [&](int buf) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int row = s_row0 + 8 * u;
+      As[buf][row][s_k] = pa[u];
+      Ws[buf][row][s_k] = pw[u];
     }
+  };
+
+  stage_regs(0);
+  regs_to_lds(0);
+
+  for (int kt = 0; kt < ktiles; ++kt) {
     __syncthreads();
+    if (kt + 1 < ktiles) stage_regs((kt + 1) * BK);
+    const int buf = kt & 1;
 #pragma unroll
     for (int k4 = 0; k4 < BK; k4 += 4) {
-      // A fragment: A[i=fi][k=fk]; B fragment: B[k][j] = W[j][k]
-      const float a = As[wv * 16 + fi][k4 + fk];
+      const float a = As[buf][wv * 16 + fi][k4 + fk];
 #pragma unroll
       for (int jt = 0; jt < 4; ++jt) {
-        const float b = Ws[jt * 16 + fi][k4 + fk];
+        const float b = Ws[buf][jt * 16 + fi][k4 + fk];
         acc[jt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[jt], 0, 0, 0);
       }
     }
-    __syncthreads();
+    if (kt + 1 < ktiles) {
+      __syncthreads();
+      regs_to_lds(1 - buf);
+    }
   }
 
 #pragma unroll
@@ -90,8 +124,8 @@ __global__ __launch_bounds__(256) void gemm_nn_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
     const float* __restrict__ bias, float* __restrict__ out,
     int R, int Ci, int Co, float alpha, int flags) {
-  __shared__ float As[BR][BK + 1];
-  __shared__ float Bs[BK][BC + 1];
+  __shared__ float As[2][BR][BK + 1];
+  __shared__ float Bs[2][BK][BC + 1];
 
   const int r0 = blockIdx.x * BR;
   const int c0 = blockIdx.y * BC;
@@ -101,30 +135,65 @@ __global__ __launch_bounds__(256) void gemm_nn_kernel(
   const int fi = lane & 15;
   const int fk = lane >> 4;
 
+  const int s_row0 = tid >> 5;     // A: + 8*u ; B: k-row = tid>>6 + 4*u
+  const int s_k = tid & 31;
+  const int b_k0 = tid >> 6;       // B staging: k-row + 4*u
+  const int b_c = tid & 63;        // B column
+
   f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
 
-  for (int k0 = 0; k0 < Ci; k0 += BK) {
-    for (int idx = tid; idx < BR * BK; idx += 256) {
-      const int i = idx / BK, k = idx % BK;
-      const int gr = r0 + i, gk = k0 + k;
-      As[i][k] = (gr < R && gk < Ci) ? A[(long)gr * Ci + gk] : 0.0f;
+  const int ktiles = (Ci + BK - 1) / BK;
+  float pa[8], pb[8];
+  auto stage_regs = [&](int k0) {
+    const bool interior = (r0 + BR <= R) && (c0 + BC <= Co) &&
+                          (k0 + BK <= Ci);
+    if (interior) {
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        pa[u] = A[(long)(r0 + s_row0 + 8 * u) * Ci + k0 + s_k];
+        pb[u] = B[(long)(k0 + b_k0 + 4 * u) * Co + c0 + b_c];
+      }
+    } else {
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const int row = s_row0 + 8 * u;
+        const int gk = k0 + s_k;
+        pa[u] = (r0 + row < R && gk < Ci)
+                    ? A[(long)(r0 + row) * Ci + gk] : 0.0f;
+        const int bk = k0 + b_k0 + 4 * u;
+        pb[u] = (bk < Ci && c0 + b_c < Co)
+                    ? B[(long)bk * Co + c0 + b_c] : 0.0f;
+      }
     }
-    for (int idx = tid; idx < BK * BC; idx += 256) {
-      const int k = idx / BC, j = idx % BC;
-      const int gk = k0 + k, gc = c0 + j;
-      Bs[k][j] = (gk < Ci && gc < Co) ? B[(long)gk * Co + gc] : 0.0f;
+  };
+  auto regs_to_lds = [&](int buf) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      As[buf][s_row0 + 8 * u][s_k] = pa[u];
+      Bs[buf][b_k0 + 4 * u][b_c] = pb[u];
     }
+  };
+
+  stage_regs(0);
+  regs_to_lds(0);
+
+  for (int kt = 0; kt < ktiles; ++kt) {
     __syncthreads();
+    if (kt + 1 < ktiles) stage_regs((kt + 1) * BK);
+    const int buf = kt & 1;
 #pragma unroll
     for (int k4 = 0; k4 < BK; k4 += 4) {
-      const float a = As[wv * 16 + fi][k4 + fk];
+      const float a = As[buf][wv * 16 + fi][k4 + fk];
 #pragma unroll
       for (int jt = 0; jt < 4; ++jt) {
-        const float b = Bs[k4 + fk][jt * 16 + fi];
+        const float b = Bs[buf][k4 + fk][jt * 16 + fi];
         acc[jt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[jt], 0, 0, 0);
       }
     }
-    __syncthreads();
+    if (kt + 1 < ktiles) {
+      __syncthreads();
+      regs_to_lds(1 - buf);
+    }
   }
 
 #pragma unroll
@@ -162,8 +231,8 @@ __global__ __launch_bounds__(256) void gemm_tn_kernel(
     float* __restrict__ out, float* __restrict__ part,
     float* __restrict__ db, float* __restrict__ db_part,
     int R, int M, int N, int accumulate) {
-  __shared__ float As[TKR][TM + 1];
-  __shared__ float Bs[TKR][TN_ + 1];
+  __shared__ float As[2][TKR][TM + 1];
+  __shared__ float Bs[2][TKR][TN_ + 1];
 
   const int m0 = blockIdx.x * TM;
   const int n0 = blockIdx.y * TN_;
@@ -186,41 +255,67 @@ __global__ __launch_bounds__(256) void gemm_tn_kernel(
 
   f32x4 acc = {0, 0, 0, 0};
 
-  for (int r0 = rbeg; r0 < rend; r0 += TKR) {
-    for (int idx = tid; idx < TKR * TM; idx += 256) {
-      const int rr = idx / TM, mm = idx % TM;
-      const int gr = r0 + rr, gm = m0 + mm;
-      As[rr][mm] = (gr < rend && gm < M) ? A[(long)gr * M + gm] : 0.0f;
-    }
-    for (int idx = tid; idx < TKR * TN_; idx += 256) {
-      const int rr = idx / TN_, nn = idx % TN_;
-      const int gr = r0 + rr, gn = n0 + nn;
-      Bs[rr][nn] = (gr < rend && gn < N) ? B[(long)gr * N + gn] : 0.0f;
-    }
-    __syncthreads();
-    if (do_bias) {
+  // register double buffer (4 floats per operand per thread):
+  // thread -> (k-row kr = tid/32 + 8u, col = tid%32)
+  const int s_kr0 = tid >> 5;
+  const int s_c = tid & 31;
+  const int ktiles = (rend - rbeg + TKR - 1) / TKR;
+  float pa[4], pb[4];
+  auto stage_regs = [&](int r0_) {
 #pragma unroll
-      for (int rr = bgrp; rr < TKR; rr += 8) bsum += As[rr][bcol];
+    for (int u = 0; u < 4; ++u) {
+      const int gr = r0_ + s_kr0 + 8 * u;
+      pa[u] = (gr < rend && m0 + s_c < M) ? A[(long)gr * M + m0 + s_c] : 0.0f;
+      pb[u] = (gr < rend && n0 + s_c < N) ? B[(long)gr * N + n0 + s_c] : 0.0f;
     }
+  };
+  auto regs_to_lds = [&](int buf) {
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      As[buf][s_kr0 + 8 * u][s_c] = pa[u];
+      Bs[buf][s_kr0 + 8 * u][s_c] = pb[u];
+    }
+  };
+  auto bias_from_regs = [&]() {
+#pragma unroll
+    for (int u = 0; u < 4; ++u) bsum += pa[u];
+  };
+
+  if (ktiles > 0) {
+    stage_regs(rbeg);
+    if (do_bias) bias_from_regs();
+    regs_to_lds(0);
+  }
+  for (int kt = 0; kt < ktiles; ++kt) {
+    __syncthreads();
+    if (kt + 1 < ktiles) {
+      stage_regs(rbeg + (kt + 1) * TKR);
+      if (do_bias) bias_from_regs();
+    }
+    const int buf = kt & 1;
     // out[m][n] = sum_r A[r][m]*B[r][n]: MFMA with k = r
 #pragma unroll
     for (int r4 = 0; r4 < TKR; r4 += 4) {
-      const float a = As[r4 + fk][mt + fi];
-      const float b = Bs[r4 + fk][nt + fi];
+      const float a = As[buf][r4 + fk][mt + fi];
+      const float b = Bs[buf][r4 + fk][nt + fi];
       acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
     }
-    __syncthreads();
+    if (kt + 1 < ktiles) {
+      __syncthreads();
+      regs_to_lds(1 - buf);
+    }
   }
 
   const bool direct = (gridDim.z == 1);
   if (do_bias) {
     // LDS-reduce the 8 row-group partials per column (reuse As storage)
-    As[bgrp][bcol] = bsum;
+    __syncthreads();
+    As[0][s_kr0][s_c] = bsum;
     __syncthreads();
     if (tid < TM) {
       float s = 0.0f;
 #pragma unroll
-      for (int gq = 0; gq < 8; ++gq) s += As[gq][tid];
+      for (int gq = 0; gq < 8; ++gq) s += As[0][gq][tid];
       const int gm = m0 + tid;
       if (gm < M) {
         if (direct) {
