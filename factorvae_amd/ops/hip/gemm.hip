@@ -32,8 +32,8 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
   __shared__ float As[2][BR][BK + 1];
   __shared__ float Ws[2][BC][BK + 1];
 
-  const int r0 = blockIdx.x * BR;
-  const int c0 = blockIdx.y * BC;
+  const int r0 = blockIdx.y * BR;  // y-major rows: A stays L2-resident
+  const int c0 = blockIdx.x * BC;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wv = tid >> 6;        // wave -> 16-row strip
@@ -127,8 +127,8 @@ __global__ __launch_bounds__(256) void gemm_nn_kernel(
   __shared__ float As[2][BR][BK + 1];
   __shared__ float Bs[2][BK][BC + 1];
 
-  const int r0 = blockIdx.x * BR;
-  const int c0 = blockIdx.y * BC;
+  const int r0 = blockIdx.y * BR;  // y-major rows: A stays L2-resident
+  const int c0 = blockIdx.x * BC;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wv = tid >> 6;
@@ -401,7 +401,7 @@ hipError_t fv_gemm_nt(const float* A, const float* W, const float* bias,
                       float* out, int R, int Ci, int Co, float alpha,
                       int accumulate, int act_lrelu, hipStream_t stream) {
   int flags = (accumulate ? 1 : 0) | (act_lrelu ? 2 : 0) | (bias ? 4 : 0);
-  dim3 grid((R + BR - 1) / BR, (Co + BC - 1) / BC);
+  dim3 grid((Co + BC - 1) / BC, (R + BR - 1) / BR);
   hipLaunchKernelGGL(gemm_nt_kernel, grid, dim3(256), 0, stream,
                      A, W, bias, out, R, Ci, Co, alpha, flags);
   HIP_CHECK_LAST();
@@ -412,7 +412,7 @@ hipError_t fv_gemm_nn(const float* A, const float* B, const float* bias,
                       float* out, int R, int Ci, int Co, float alpha,
                       int accumulate, int act_lrelu, hipStream_t stream) {
   int flags = (accumulate ? 1 : 0) | (act_lrelu ? 2 : 0) | (bias ? 4 : 0);
-  dim3 grid((R + BR - 1) / BR, (Co + BC - 1) / BC);
+  dim3 grid((Co + BC - 1) / BC, (R + BR - 1) / BR);
   hipLaunchKernelGGL(gemm_nn_kernel, grid, dim3(256), 0, stream,
                      A, B, bias, out, R, Ci, Co, alpha, flags);
   HIP_CHECK_LAST();

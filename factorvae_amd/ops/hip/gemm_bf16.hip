@@ -67,8 +67,10 @@ __global__ __launch_bounds__(256) void gemm_nt_bf16_kernel(
   __shared__ __bf16 As[2][BBR][BBK + 8];
   __shared__ __bf16 Ws[2][BBC][BBK + 8];
 
-  const int r0 = blockIdx.x * BBR;
-  const int c0 = blockIdx.y * BBC;
+  // x = column tile: consecutive block IDs share the row range so the
+  // streamed A tiles stay L2-resident across the (few) column tiles
+  const int r0 = blockIdx.y * BBR;
+  const int c0 = blockIdx.x * BBC;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wv = tid >> 6;        // wave -> 16-row strip
@@ -175,8 +177,10 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
   __shared__ __bf16 As[2][BBR][BBK + 8];
   __shared__ __bf16 Bs[2][BBK][BBC + 4];   // row-major [k][n], tr16-read
 
-  const int r0 = blockIdx.x * BBR;
-  const int c0 = blockIdx.y * BBC;
+  // x = column tile: consecutive block IDs share the row range so the
+  // streamed A tiles stay L2-resident across the (few) column tiles
+  const int r0 = blockIdx.y * BBR;
+  const int c0 = blockIdx.x * BBC;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wv = tid >> 6;
@@ -525,7 +529,7 @@ hipError_t fv_gemm_nt_bf16(const void* A, const void* W, const float* bias,
                            int Co, float alpha, int accumulate, int act_lrelu,
                            hipStream_t stream) {
   int flags = (accumulate ? 1 : 0) | (act_lrelu ? 2 : 0) | (bias ? 4 : 0);
-  dim3 grid((R + BBR - 1) / BBR, (Co + BBC - 1) / BBC);
+  dim3 grid((Co + BBC - 1) / BBC, (R + BBR - 1) / BBR);
   hipLaunchKernelGGL(gemm_nt_bf16_kernel, grid, dim3(256), 0, stream,
                      (const __bf16*)A, (const __bf16*)W, bias, out_f32,
                      (__bf16*)out_bf16, R, Ci, Co, alpha, flags);
@@ -539,7 +543,7 @@ hipError_t fv_gemm_nn_bf16(const void* A, const void* B, const float* bias,
                            int act_lrelu, hipStream_t stream) {
   int flags = (accumulate ? 1 : 0) | (act_lrelu ? 2 : 0) | (bias ? 4 : 0) |
               (Y ? 8 : 0);
-  dim3 grid((R + BBR - 1) / BBR, (Co + BBC - 1) / BBC);
+  dim3 grid((Co + BBC - 1) / BBC, (R + BBR - 1) / BBR);
   hipLaunchKernelGGL(gemm_nn_bf16_kernel, grid, dim3(256), 0, stream,
                      (const __bf16*)A, (const __bf16*)B, bias, out_f32,
                      (__bf16*)out_bf16, (const __bf16*)Y, R, Ci, Co, alpha,
