@@ -4,6 +4,8 @@
 
 #include "common.h"
 
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
 // One wave per row: x(R,C) -> xln = gamma * (x-mean)*rstd + beta.
 // Saves mean & rstd per row for the backward recompute.
 __global__ __launch_bounds__(256) void ln_fwd_kernel(
@@ -19,19 +21,34 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
   if (row >= R) return;
   const float* xr = x + row * C;
 
-  // single pass: per-lane sum and sum-of-squares (keeps x in registers
-  // for the normalize write — one HBM read of x instead of three)
-  float xv[8];  // C <= 512 per lane budget: ceil(C/64) values
-  const int nv = (C - lane + 63) / 64;
+  // single pass, vectorized: lane l owns the f32x4 chunk at 4*l (plus a
+  // strided tail for C > 256); the row stays in registers for the
+  // normalize write — one HBM read of x instead of three
+  float xv[8];
   float s = 0.0f, sq = 0.0f;
-  for (int i = 0; i < nv && i < 8; ++i) {
-    const float xx = xr[lane + i * 64];
-    xv[i] = xx;
-    s += xx;
-    sq = fmaf(xx, xx, sq);
+  const int c4 = lane * 4;
+  if (c4 + 4 <= C) {
+    const f32x4 v = *(const f32x4*)&xr[c4];
+    xv[0] = v.x; xv[1] = v.y; xv[2] = v.z; xv[3] = v.w;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      s += xv[i];
+      sq = fmaf(xv[i], xv[i], sq);
+    }
+  } else {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int c = c4 + i;
+      const float xx = (c < C) ? xr[c] : 0.0f;
+      xv[i] = xx;
+      s += xx;
+      sq = fmaf(xx, xx, sq);
+    }
   }
-  for (int c = lane + 8 * 64; c < C; c += 64) {  // spill tail (C > 512)
+  int tail = 0;
+  for (int c = 256 + lane; c < C; c += 64, ++tail) {  // C > 256
     const float xx = xr[c];
+    if (tail < 4) xv[4 + tail] = xx;
     s += xx;
     sq = fmaf(xx, xx, sq);
   }
@@ -50,8 +67,22 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
   float* o = xln ? xln + row * C : nullptr;
   __bf16* ob = xln_bf ? xln_bf + row * C : nullptr;
   unsigned char* o8 = xln_f8 ? xln_f8 + row * (long)f8_ld : nullptr;
-  for (int c = lane, i = 0; c < C; c += 64, ++i) {
-    const float xx = (i < 8) ? xv[i] : xr[c];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int c = c4 + i;
+    if (c >= C) break;
+    const float v_ = fmaf((xv[i] - mu) * rs, gamma[c], beta[c]);
+    if (o) o[c] = v_;
+    if (ob) ob[c] = (__bf16)v_;
+    if (o8) {
+      unsigned int u = 0;
+      u = __builtin_amdgcn_cvt_pk_fp8_f32(v_, 0.0f, u, false);
+      o8[c] = (unsigned char)(u & 0xff);
+    }
+  }
+  tail = 0;
+  for (int c = 256 + lane; c < C; c += 64, ++tail) {
+    const float xx = (tail < 4) ? xv[4 + tail] : xr[c];
     const float v_ = fmaf((xx - mu) * rs, gamma[c], beta[c]);
     if (o) o[c] = v_;
     if (ob) ob[c] = (__bf16)v_;
