@@ -192,45 +192,48 @@ __global__ __launch_bounds__(256) void dec_bwd_kernel(
   }
 
   // cross-wave reduce of the shared-grad registers -> per-block partial
-  const int E = 2 * K + 2 * H + 2;
-  float* po = part + (long)blockIdx.x * E;
+  // (TRANSPOSED layout part[e][nblk]: the reduce kernel reads each
+  // element's partials contiguously)
+  const int NB = gridDim.x;
+  float* po = part;
+#define DP(e) po[(long)(e) * NB + blockIdx.x]
   __syncthreads();
   red[w * 64 + lane] = rfmu0;
   __syncthreads();
   if (w == 0 && lane < K && lane < 64)
-    po[lane] =
+    DP(lane) =
         red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
   __syncthreads();
   red[w * 64 + lane] = rfsig0;
   __syncthreads();
   if (w == 0 && lane < K && lane < 64)
-    po[K + lane] =
+    DP(K + lane) =
         red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
   if (K > 64) {
     __syncthreads();
     red[w * 64 + lane] = kh2 ? rfmu1 : 0.0f;
     __syncthreads();
     if (w == 0 && lane + 64 < K)
-      po[lane + 64] =
+      DP(lane + 64) =
           red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
     __syncthreads();
     red[w * 64 + lane] = kh2 ? rfsig1 : 0.0f;
     __syncthreads();
     if (w == 0 && lane + 64 < K)
-      po[K + lane + 64] =
+      DP(K + lane + 64) =
           red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
   }
   __syncthreads();
   red[w * 64 + lane] = rwmu;
   __syncthreads();
   if (w == 0 && lane < H)
-    po[2 * K + lane] =
+    DP(2 * K + lane) =
         red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
   __syncthreads();
   red[w * 64 + lane] = rwsig;
   __syncthreads();
   if (w == 0 && lane < H)
-    po[2 * K + H + lane] =
+    DP(2 * K + H + lane) =
         red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
   // scalar bias grads (lane-0 registers only)
   __syncthreads();
@@ -240,9 +243,10 @@ __global__ __launch_bounds__(256) void dec_bwd_kernel(
   }
   __syncthreads();
   if (tid == 0) {
-    po[2 * K + 2 * H] = red[0] + red[1] + red[2] + red[3];
-    po[2 * K + 2 * H + 1] = red[4] + red[5] + red[6] + red[7];
+    DP(2 * K + 2 * H) = red[0] + red[1] + red[2] + red[3];
+    DP(2 * K + 2 * H + 1) = red[4] + red[5] + red[6] + red[7];
   }
+#undef DP
 }
 
 // fixed-order reduce of the per-block partials into the grad slots
@@ -254,8 +258,14 @@ __global__ __launch_bounds__(256) void dec_bwd_reduce_kernel(
   const int E = 2 * K + 2 * H + 2;
   const int e = blockIdx.x * 256 + threadIdx.x;
   if (e >= E) return;
-  float s = 0.0f;
-  for (int z = 0; z < nblk; ++z) s += part[(long)z * E + e];
+  const float* pe = part + (long)e * nblk;
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+  int z = 0;
+  for (; z + 4 <= nblk; z += 4) {
+    s0 += pe[z]; s1 += pe[z + 1]; s2 += pe[z + 2]; s3 += pe[z + 3];
+  }
+  float s = (s0 + s1) + (s2 + s3);
+  for (; z < nblk; ++z) s += pe[z];
   if (e < K) dfmu[e] += s;
   else if (e < 2 * K) dfsig_c[e - K] += s;
   else if (e < 2 * K + H) dwmu[e - 2 * K] += s;
