@@ -175,7 +175,12 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
     __bf16* __restrict__ out_bf16, const __bf16* __restrict__ Y,
     int R, int Ci, int Co, float alpha, int flags) {
   __shared__ __bf16 As[2][BBR][BBK + 8];
-  __shared__ __bf16 Bs[2][BBK][BBC + 4];   // row-major [k][n], tr16-read
+  // row-major [k][n], tr16-read. Stride 72 elems = 36 dwords: the four
+  // tr16 k-row offsets land on disjoint bank ranges ({0-7},{36-43},
+  // {8-15},{44-51} dwords mod 64, and +16 for the kb+4 read) — stride
+  // 68 (34 dw) made rows 0/2 and 1/3 collide (2-way conflict on every
+  // fragment read, ~36% LDS overhead in PMC).
+  __shared__ __bf16 Bs[2][BBK][BBC + 8];
 
   // x = column tile: consecutive block IDs share the row range so the
   // streamed A tiles stay L2-resident across the (few) column tiles
@@ -299,7 +304,8 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
 #define TBM 64
 #define TBN 64
 #define TBK 64
-#define TSA (TBM + 4)   // row stride (elems): 8B-aligned, conflict-free
+#define TSA (TBM + 8)   // row stride: 36 dwords -> conflict-free tr16 reads
+                        // (34 dw collided k-row offsets 0/2 and 1/3)
 
 __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
