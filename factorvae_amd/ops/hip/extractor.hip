@@ -137,21 +137,30 @@ __global__ __launch_bounds__(256) void ln_bwd_params_kernel(
   }
 }
 
+// One WAVE per output element: lane l sums partials l, l+64, ... then a
+// fixed-tree wave reduction combines lanes. The schedule is a fixed
+// function of (e, yblocks), so results stay bit-identical run to run.
+// (The old thread-per-element form was 2-3 workgroups on 256 CUs, a
+// ~55 us dependent-load chain at the very tail of the backward.)
 __global__ __launch_bounds__(256) void ln_bwd_reduce_kernel(
     const float* __restrict__ part, float* __restrict__ dgamma,
     float* __restrict__ dbeta, int yblocks, int C) {
-  const int e = blockIdx.x * 256 + threadIdx.x;
+  const int e = blockIdx.x * 4 + (threadIdx.x >> 6);
   if (e >= 2 * C) return;
+  const int lane = threadIdx.x & 63;
   const float* pe = part + (long)e * yblocks;
-  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
-  int z = 0;
-  for (; z + 4 <= yblocks; z += 4) {
-    s0 += pe[z]; s1 += pe[z + 1]; s2 += pe[z + 2]; s3 += pe[z + 3];
+  float s0 = 0.f, s1 = 0.f;
+  int z = lane;
+  for (; z + 64 < yblocks; z += 128) {
+    s0 += pe[z];
+    s1 += pe[z + 64];
   }
-  float s = (s0 + s1) + (s2 + s3);
-  for (; z < yblocks; ++z) s += pe[z];
-  if (e < C) dgamma[e] += s;
-  else dbeta[e - C] += s;
+  if (z < yblocks) s0 += pe[z];
+  float s = wave_reduce_sum(s0 + s1);
+  if (lane == 0) {
+    if (e < C) dgamma[e] += s;
+    else dbeta[e - C] += s;
+  }
 }
 
 extern "C" {
@@ -186,7 +195,7 @@ hipError_t fv_ln_bwd_params(const float* x, const float* dxln,
   hipLaunchKernelGGL(ln_bwd_params_kernel, grid, dim3(256), 0, stream,
                      x, dxln, mean, rstd, part, R, C, (int)rpb);
   HIP_CHECK_LAST();
-  hipLaunchKernelGGL(ln_bwd_reduce_kernel, dim3((2 * C + 255) / 256),
+  hipLaunchKernelGGL(ln_bwd_reduce_kernel, dim3((2 * C + 3) / 4),
                      dim3(256), 0, stream, part, dgamma, dbeta,
                      (int)yblocks, C);
   HIP_CHECK_LAST();

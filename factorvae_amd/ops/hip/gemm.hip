@@ -345,20 +345,38 @@ __global__ __launch_bounds__(256) void gemm_tn_kernel(
 
 // out[e] += sum_z part[z][e]; db[m] += sum_z db_part[z][m] — fixed z
 // order (deterministic reduce, no atomics).
+// 64 output elements per workgroup; the 4 waves each sum a fixed
+// quarter of the z slices (coalesced across elements) and combine
+// through LDS in fixed order - deterministic, 4x the wave parallelism
+// of a thread-per-element loop (see tn_reduce_bf16_kernel).
 __global__ __launch_bounds__(256) void tn_reduce_kernel(
     const float* __restrict__ part, float* __restrict__ out, long elems,
     const float* __restrict__ db_part, float* __restrict__ db, long m_elems,
     int z, int accumulate) {
-  const long e = (long)blockIdx.x * 256 + threadIdx.x;
+  __shared__ float ps[4][64];
+  const int el = threadIdx.x & 63;
+  const int zg = threadIdx.x >> 6;
+  const long e = (long)blockIdx.x * 64 + el;
+  const int zchunk = (z + 3) / 4;
+  const int zbeg = zg * zchunk;
+  const int zend = min(zbeg + zchunk, z);
+  float s = 0.0f;
   if (e < elems) {
-    float s = accumulate ? out[e] : 0.0f;
-    for (int c = 0; c < z; ++c) s += part[(long)c * elems + e];
-    out[e] = s;
+    for (int c = zbeg; c < zend; ++c) s += part[(long)c * elems + e];
   } else if (e < elems + m_elems) {
     const long m = e - elems;
-    float s = accumulate ? db[m] : 0.0f;
-    for (int c = 0; c < z; ++c) s += db_part[(long)c * m_elems + m];
-    db[m] = s;
+    for (int c = zbeg; c < zend; ++c) s += db_part[(long)c * m_elems + m];
+  }
+  ps[zg][el] = s;
+  __syncthreads();
+  if (zg == 0) {
+    const float t = ((ps[0][el] + ps[1][el]) + (ps[2][el] + ps[3][el]));
+    if (e < elems) {
+      out[e] = (accumulate ? out[e] : 0.0f) + t;
+    } else if (e < elems + m_elems) {
+      const long m = e - elems;
+      db[m] = (accumulate ? db[m] : 0.0f) + t;
+    }
   }
 }
 
@@ -440,7 +458,7 @@ hipError_t fv_gemm_tn(const float* A, const float* B, float* out,
   if (r_chunks > 1) {
     const long elems = (long)M * N;
     const long m_elems = db ? M : 0;
-    dim3 rgrid((unsigned)((elems + m_elems + 255) / 256));
+    dim3 rgrid((unsigned)((elems + m_elems + 63) / 64));
     hipLaunchKernelGGL(tn_reduce_kernel, rgrid, dim3(256), 0, stream,
                        part, out, elems, db_part, db, m_elems, r_chunks,
                        accumulate);

@@ -467,31 +467,51 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
   }
 }
 
-// fixed-order partial reduce (same contract as fp32 tn_reduce_kernel)
+// fixed-order partial reduce (same contract as fp32 tn_reduce_kernel).
+// Each workgroup covers 64 output elements; its 4 waves each sum a
+// fixed quarter of the z slices (reads stay coalesced across the 64
+// consecutive elements) and the quarters combine through LDS in fixed
+// order - deterministic, with 4x the wave parallelism of the old
+// thread-per-element form (whose z-strided dependent loads were the
+// latency chain at the backward tail).
 __global__ __launch_bounds__(256) void tn_reduce_bf16_kernel(
     const float* __restrict__ part, float* __restrict__ out, long elems,
     const float* __restrict__ db_part, float* __restrict__ db, long m_elems,
     int z, int accumulate) {
-  const long e = (long)blockIdx.x * 256 + threadIdx.x;
+  __shared__ float ps[4][64];
+  const int el = threadIdx.x & 63;
+  const int zg = threadIdx.x >> 6;
+  const long e = (long)blockIdx.x * 64 + el;
+  const int zchunk = (z + 3) / 4;
+  const int zbeg = zg * zchunk;
+  const int zend = min(zbeg + zchunk, z);
+  float s0 = 0.f, s1 = 0.f;
   if (e < elems) {
-    // 4 independent partial sums keep ~4 loads in flight per thread
-    // (the z-strided reads are latency-bound), summed in FIXED order
-    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
-    int c = 0;
-    for (; c + 4 <= z; c += 4) {
+    int c = zbeg;
+    for (; c + 2 <= zend; c += 2) {
       s0 += part[(long)c * elems + e];
       s1 += part[(long)(c + 1) * elems + e];
-      s2 += part[(long)(c + 2) * elems + e];
-      s3 += part[(long)(c + 3) * elems + e];
     }
-    float s = ((s0 + s1) + (s2 + s3));
-    for (; c < z; ++c) s += part[(long)c * elems + e];
-    out[e] = (accumulate ? out[e] : 0.0f) + s;
+    if (c < zend) s0 += part[(long)c * elems + e];
   } else if (e < elems + m_elems) {
     const long m = e - elems;
-    float s = accumulate ? db[m] : 0.0f;
-    for (int c = 0; c < z; ++c) s += db_part[(long)c * m_elems + m];
-    db[m] = s;
+    int c = zbeg;
+    for (; c + 2 <= zend; c += 2) {
+      s0 += db_part[(long)c * m_elems + m];
+      s1 += db_part[(long)(c + 1) * m_elems + m];
+    }
+    if (c < zend) s0 += db_part[(long)c * m_elems + m];
+  }
+  ps[zg][el] = s0 + s1;
+  __syncthreads();
+  if (zg == 0) {
+    const float s = ((ps[0][el] + ps[1][el]) + (ps[2][el] + ps[3][el]));
+    if (e < elems) {
+      out[e] = (accumulate ? out[e] : 0.0f) + s;
+    } else if (e < elems + m_elems) {
+      const long m = e - elems;
+      db[m] = (accumulate ? db[m] : 0.0f) + s;
+    }
   }
 }
 
@@ -581,7 +601,7 @@ hipError_t fv_gemm_tn_bf16(const void* A, const void* B, float* out,
   if (r_chunks > 1) {
     const long elems = (long)M * N;
     const long m_elems = db ? M : 0;
-    dim3 rgrid((unsigned)((elems + m_elems + 255) / 256));
+    dim3 rgrid((unsigned)((elems + m_elems + 63) / 64));
     hipLaunchKernelGGL(tn_reduce_bf16_kernel, rgrid, dim3(256), 0, stream,
                        part, out, elems, db_part, db, m_elems, r_chunks,
                        accumulate);
