@@ -300,15 +300,27 @@ __global__ __launch_bounds__(64) void pred_mlp_bwd_kernel(
 }
 
 // fixed-head-order reduce of the shared head-param grad partials
+// One WAVE per element (E = 2H+2 was a single workgroup as a
+// thread-per-element loop); hpart is freshly written and L2-resident,
+// so the per-lane strided reads are cheap. Fixed-tree wave reduction:
+// still bit-deterministic run to run.
 __global__ __launch_bounds__(256) void pred_head_reduce_kernel(
     const float* __restrict__ hpart, float* __restrict__ dwmu,
     float* __restrict__ dbmu, float* __restrict__ dwsig,
     float* __restrict__ dbsig, int K, int H) {
   const int E = 2 * H + 2;
-  const int e = blockIdx.x * 256 + threadIdx.x;
+  const int e = blockIdx.x * 4 + (threadIdx.x >> 6);
   if (e >= E) return;
-  float s = 0.0f;
-  for (int k = 0; k < K; ++k) s += hpart[(long)k * E + e];
+  const int lane = threadIdx.x & 63;
+  float s0 = 0.0f, s1 = 0.0f;
+  int k = lane;
+  for (; k + 64 < K; k += 128) {
+    s0 += hpart[(long)k * E + e];
+    s1 += hpart[(long)(k + 64) * E + e];
+  }
+  if (k < K) s0 += hpart[(long)k * E + e];
+  const float s = wave_reduce_sum(s0 + s1);
+  if (lane != 0) return;
   if (e < H) dwmu[e] += s;
   else if (e < 2 * H) dwsig[e - H] += s;
   else if (e == 2 * H) dbmu[0] += s;
@@ -668,7 +680,7 @@ hipError_t fv_attn_fused_bwd(const float* dpmu, const float* dpsig_c,
                      dbv, dq, dWk, dbk, hpart, N, K, H, alpha, keep_inv);
   HIP_CHECK_LAST();
   const int E = 2 * H + 2;
-  hipLaunchKernelGGL(pred_head_reduce_kernel, dim3((E + 255) / 256),
+  hipLaunchKernelGGL(pred_head_reduce_kernel, dim3((E + 3) / 4),
                      dim3(256), 0, s, hpart, dwmu, dbmu, dwsig, dbsig, K, H);
   HIP_CHECK_LAST();
   return hipSuccess;
@@ -778,7 +790,7 @@ hipError_t fv_pred_mlp_bwd(const float* dpmu, const float* dpsig_c,
                      dz2, hpart, K, H);
   HIP_CHECK_LAST();
   const int E = 2 * H + 2;
-  hipLaunchKernelGGL(pred_head_reduce_kernel, dim3((E + 255) / 256),
+  hipLaunchKernelGGL(pred_head_reduce_kernel, dim3((E + 3) / 4),
                      dim3(256), 0, s, hpart, dwmu, dbmu, dwsig, dbsig, K, H);
   HIP_CHECK_LAST();
   return hipSuccess;

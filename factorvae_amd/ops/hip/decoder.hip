@@ -250,22 +250,29 @@ __global__ __launch_bounds__(256) void dec_bwd_kernel(
 }
 
 // fixed-order reduce of the per-block partials into the grad slots
+// One WAVE per element (E = 2K+2H+2 was only 1-2 workgroups as a
+// thread-per-element loop): lane l sums partials l, l+64, ... and a
+// fixed-tree wave reduction combines lanes - schedule is a fixed
+// function of (e, nblk), so still bit-deterministic run to run.
 __global__ __launch_bounds__(256) void dec_bwd_reduce_kernel(
     const float* __restrict__ part, float* __restrict__ dfmu,
     float* __restrict__ dfsig_c, float* __restrict__ dwmu,
     float* __restrict__ dbmu, float* __restrict__ dwsig,
     float* __restrict__ dbsig, int nblk, int K, int H) {
   const int E = 2 * K + 2 * H + 2;
-  const int e = blockIdx.x * 256 + threadIdx.x;
+  const int e = blockIdx.x * 4 + (threadIdx.x >> 6);
   if (e >= E) return;
+  const int lane = threadIdx.x & 63;
   const float* pe = part + (long)e * nblk;
-  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
-  int z = 0;
-  for (; z + 4 <= nblk; z += 4) {
-    s0 += pe[z]; s1 += pe[z + 1]; s2 += pe[z + 2]; s3 += pe[z + 3];
+  float s0 = 0.f, s1 = 0.f;
+  int z = lane;
+  for (; z + 64 < nblk; z += 128) {
+    s0 += pe[z];
+    s1 += pe[z + 64];
   }
-  float s = (s0 + s1) + (s2 + s3);
-  for (; z < nblk; ++z) s += pe[z];
+  if (z < nblk) s0 += pe[z];
+  const float s = wave_reduce_sum(s0 + s1);
+  if (lane != 0) return;
   if (e < K) dfmu[e] += s;
   else if (e < 2 * K) dfsig_c[e - K] += s;
   else if (e < 2 * K + H) dwmu[e - 2 * K] += s;
@@ -313,7 +320,7 @@ hipError_t fv_dec_bwd(const float* drecon, const float* h, const float* a1,
                      W1, wmu, wsig, Wb, dh, dz1, dbeta, part, N, K, H, iters);
   HIP_CHECK_LAST();
   const int E = 2 * K + 2 * H + 2;
-  hipLaunchKernelGGL(dec_bwd_reduce_kernel, dim3((E + 255) / 256), dim3(256),
+  hipLaunchKernelGGL(dec_bwd_reduce_kernel, dim3((E + 3) / 4), dim3(256),
                      0, s, part, dfmu, dfsig_c, dwmu, dbmu, dwsig, dbsig,
                      nblk, K, H);
   HIP_CHECK_LAST();
