@@ -460,7 +460,9 @@ class FusedTrainer:
         yv = w["y"] if y is None else y
         alpha = 1.0 / math.sqrt(float(H) + 1e-6)
         keep_inv = 1.0 / (1.0 - self.DROPOUT_P)
-        chunks = max(1, min(32, R // 1024))
+        import os as _os
+        chunks = (int(_os.environ.get("FV_TN_CHUNKS", "0"))
+                  or max(1, min(32, R // 1024)))
 
         main = torch.cuda.current_stream(self.device) if self.s_side else None
         sides = (self.s_side, self.s_side2)
