@@ -6,8 +6,9 @@
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
-// One wave per row: x(R,C) -> xln = gamma * (x-mean)*rstd + beta.
+// x(R,C) -> xln = gamma * (x-mean)*rstd + beta; one wave per row batch.
 // Saves mean & rstd per row for the backward recompute.
+#define LNF_ROWS 4
 __global__ __launch_bounds__(256) void ln_fwd_kernel(
     const float* __restrict__ x, const float* __restrict__ gamma,
     const float* __restrict__ beta, float* __restrict__ xln,
@@ -17,79 +18,96 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
     long R, int C, float eps) {
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const long row = (long)blockIdx.x * 4 + wid;
-  if (row >= R) return;
-  const float* xr = x + row * C;
-
-  // single pass, vectorized: lane l owns the f32x4 chunk at 4*l (plus a
-  // strided tail for C > 256); the row stays in registers for the
-  // normalize write — one HBM read of x instead of three
-  float xv[8];
-  float s = 0.0f, sq = 0.0f;
+  // LNF_ROWS sequential rows per wave: 1-row waves left the kernel
+  // launch/dispatch-bound (52k tiny workgroups at A-share, 96.6 us);
+  // 4 rows amortize the workgroup and the gamma/beta loads -> 58.9 us,
+  // bit-identical (scripts/probe/ln_probe.hip A/B).
+  const long row0 = ((long)blockIdx.x * 4 + wid) * LNF_ROWS;
   const int c4 = lane * 4;
-  if (c4 + 4 <= C) {
-    const f32x4 v = *(const f32x4*)&xr[c4];
-    xv[0] = v.x; xv[1] = v.y; xv[2] = v.z; xv[3] = v.w;
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      s += xv[i];
-      sq = fmaf(xv[i], xv[i], sq);
-    }
-  } else {
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int c = c4 + i;
-      const float xx = (c < C) ? xr[c] : 0.0f;
-      xv[i] = xx;
-      s += xx;
-      sq = fmaf(xx, xx, sq);
-    }
-  }
-  int tail = 0;
-  for (int c = 256 + lane; c < C; c += 64, ++tail) {  // C > 256
-    const float xx = xr[c];
-    if (tail < 4) xv[4 + tail] = xx;
-    s += xx;
-    sq = fmaf(xx, xx, sq);
-  }
-  s = wave_reduce_sum(s);
-  sq = wave_reduce_sum(sq);
-  s = __shfl(s, 0, 64);
-  sq = __shfl(sq, 0, 64);
-  const float mu = s / C;
-  const float var = fmaxf(sq / C - mu * mu, 0.0f);
-  const float rs = rsqrtf(var + eps);
 
-  if (lane == 0) {
-    mean[row] = mu;
-    rstd[row] = rs;
-  }
-  float* o = xln ? xln + row * C : nullptr;
-  __bf16* ob = xln_bf ? xln_bf + row * C : nullptr;
-  unsigned char* o8 = xln_f8 ? xln_f8 + row * (long)f8_ld : nullptr;
+  // gamma/beta for this lane's main chunk, loaded once per wave
+  float gv[4], bvv[4];
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     const int c = c4 + i;
-    if (c >= C) break;
-    const float v_ = fmaf((xv[i] - mu) * rs, gamma[c], beta[c]);
-    if (o) o[c] = v_;
-    if (ob) ob[c] = (__bf16)v_;
-    if (o8) {
-      unsigned int u = 0;
-      u = __builtin_amdgcn_cvt_pk_fp8_f32(v_, 0.0f, u, false);
-      o8[c] = (unsigned char)(u & 0xff);
-    }
+    gv[i] = (c < C) ? gamma[c] : 0.0f;
+    bvv[i] = (c < C) ? beta[c] : 0.0f;
   }
-  tail = 0;
-  for (int c = 256 + lane; c < C; c += 64, ++tail) {
-    const float xx = (tail < 4) ? xv[4 + tail] : xr[c];
-    const float v_ = fmaf((xx - mu) * rs, gamma[c], beta[c]);
-    if (o) o[c] = v_;
-    if (ob) ob[c] = (__bf16)v_;
-    if (o8) {
-      unsigned int u = 0;
-      u = __builtin_amdgcn_cvt_pk_fp8_f32(v_, 0.0f, u, false);
-      o8[c] = (unsigned char)(u & 0xff);
+
+  for (int r = 0; r < LNF_ROWS; ++r) {
+    const long row = row0 + r;
+    if (row >= R) return;
+    const float* xr = x + row * C;
+
+    // single pass, vectorized: lane l owns the f32x4 chunk at 4*l (plus
+    // a strided tail for C > 256); the row stays in registers for the
+    // normalize write — one HBM read of x instead of three
+    float xv[8];
+    float s = 0.0f, sq = 0.0f;
+    if (c4 + 4 <= C) {
+      const f32x4 v = *(const f32x4*)&xr[c4];
+      xv[0] = v.x; xv[1] = v.y; xv[2] = v.z; xv[3] = v.w;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        s += xv[i];
+        sq = fmaf(xv[i], xv[i], sq);
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int c = c4 + i;
+        const float xx = (c < C) ? xr[c] : 0.0f;
+        xv[i] = xx;
+        s += xx;
+        sq = fmaf(xx, xx, sq);
+      }
+    }
+    int tail = 0;
+    for (int c = 256 + lane; c < C; c += 64, ++tail) {  // C > 256
+      const float xx = xr[c];
+      if (tail < 4) xv[4 + tail] = xx;
+      s += xx;
+      sq = fmaf(xx, xx, sq);
+    }
+    s = wave_reduce_sum(s);
+    sq = wave_reduce_sum(sq);
+    s = __shfl(s, 0, 64);
+    sq = __shfl(sq, 0, 64);
+    const float mu = s / C;
+    const float var = fmaxf(sq / C - mu * mu, 0.0f);
+    const float rs = rsqrtf(var + eps);
+
+    if (lane == 0) {
+      mean[row] = mu;
+      rstd[row] = rs;
+    }
+    float* o = xln ? xln + row * C : nullptr;
+    __bf16* ob = xln_bf ? xln_bf + row * C : nullptr;
+    unsigned char* o8 = xln_f8 ? xln_f8 + row * (long)f8_ld : nullptr;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int c = c4 + i;
+      if (c >= C) break;
+      const float v_ = fmaf((xv[i] - mu) * rs, gv[i], bvv[i]);
+      if (o) o[c] = v_;
+      if (ob) ob[c] = (__bf16)v_;
+      if (o8) {
+        unsigned int u = 0;
+        u = __builtin_amdgcn_cvt_pk_fp8_f32(v_, 0.0f, u, false);
+        o8[c] = (unsigned char)(u & 0xff);
+      }
+    }
+    tail = 0;
+    for (int c = 256 + lane; c < C; c += 64, ++tail) {
+      const float xx = (tail < 4) ? xv[4 + tail] : xr[c];
+      const float v_ = fmaf((xx - mu) * rs, gamma[c], beta[c]);
+      if (o) o[c] = v_;
+      if (ob) ob[c] = (__bf16)v_;
+      if (o8) {
+        unsigned int u = 0;
+        u = __builtin_amdgcn_cvt_pk_fp8_f32(v_, 0.0f, u, false);
+        o8[c] = (unsigned char)(u & 0xff);
+      }
     }
   }
 }
@@ -169,7 +187,8 @@ hipError_t fv_ln_fwd(const float* x, const float* gamma, const float* beta,
                      float* xln, void* xln_bf, void* xln_f8, int f8_ld,
                      float* mean, float* rstd,
                      long R, int C, float eps, hipStream_t stream) {
-  dim3 grid((unsigned)((R + 3) / 4));
+  const long wgrows = 4L * LNF_ROWS;
+  dim3 grid((unsigned)((R + wgrows - 1) / wgrows));
   hipLaunchKernelGGL(ln_fwd_kernel, grid, dim3(256), 0, stream,
                      x, gamma, beta, xln, (__bf16*)xln_bf,
                      (unsigned char*)xln_f8, f8_ld, mean, rstd, R, C, eps);
