@@ -55,12 +55,23 @@ class DeviceEpochCache:
         if shuffle:
             rng = np.random.default_rng(self.seed + epoch)
             rng.shuffle(idx)
-        for k in range(rank, n, world_size):
+        # pad (wrap-around) to a multiple of world_size: every rank must
+        # run the SAME number of steps per epoch or the per-step gradient
+        # all-reduces go out of lockstep at the epoch tail (matches
+        # DateGroupedBatchSampler._padded_order)
+        if world_size > 1 and n % world_size != 0 and n > 0:
+            pad = world_size - n % world_size
+            reps = (pad + n - 1) // n
+            idx = np.concatenate([idx] + [idx] * reps)[:n + pad]
+        for k in range(rank, len(idx), world_size):
             yield self.days[idx[k]]
 
     def num_batches(self, rank: int = 0, world_size: int = 1) -> int:
+        """Per-rank steps per epoch — identical on every rank (padded)."""
         n = len(self.days)
-        return (n - rank + world_size - 1) // world_size
+        if n == 0:
+            return 0
+        return (n + world_size - 1) // world_size
 
 
 def synthetic_device_days(n_days: int, n_stocks: int, seq_len: int, n_features: int,

@@ -213,14 +213,30 @@ class DateGroupedBatchSampler(Sampler):
             rng.shuffle(order)
         return order
 
-    def __iter__(self):
+    def _padded_order(self):
+        """Day order padded (wrap-around) to a multiple of world_size so
+        every rank sees the SAME number of days per epoch: unequal counts
+        would make ranks issue different numbers of per-step gradient
+        all-reduces (hang / silent corruption at the epoch tail) and
+        derive diverging cosine-LR t_max values."""
         order = self._day_order()
+        n = len(order)
+        if self.world_size > 1 and n % self.world_size != 0 and n > 0:
+            pad = self.world_size - n % self.world_size
+            reps = (pad + n - 1) // n
+            order = np.concatenate([order] + [order] * reps)[:n + pad]
+        return order
+
+    def __iter__(self):
+        order = self._padded_order()
         for k in range(self.rank, len(order), self.world_size):
             yield self.grouped_indices[order[k]]
 
     def __len__(self):
         n = len(self.grouped_indices)
-        return (n - self.rank + self.world_size - 1) // self.world_size
+        if n == 0:
+            return 0
+        return (n + self.world_size - 1) // self.world_size
 
 
 def custom_collate_fn(batch):

@@ -157,3 +157,98 @@ def test_train_main_two_rank_end_to_end(tmp_path):
     # identical best-val on both ranks (all-reduced), rank 0 wrote files
     assert results[0][0] == pytest.approx(results[1][0], rel=1e-5)
     assert results[0][1] and results[0][2]
+
+
+# ---------------------------------------------------------------- odd days
+class _FakeLoader:
+    """Minimal stand-in yielding (N,T,C+1) day blocks for DeviceEpochCache."""
+
+    def __init__(self, n_days, n=4, t=3, c=5):
+        self.blocks = [(torch.randn(n, t, c + 1), None) for _ in range(n_days)]
+
+    def __iter__(self):
+        return iter(self.blocks)
+
+
+def test_device_cache_padded_sharding_equal_counts():
+    """n_days % world_size != 0: every rank still gets the same step count
+    (padded wrap-around), so per-step collectives stay in lockstep and
+    t_max = num_batches * epochs agrees across ranks."""
+    from factorvae_amd.data.device_cache import DeviceEpochCache
+
+    for n_days, ws in [(5, 2), (3, 2), (7, 4), (2, 4), (6, 3)]:
+        cache = DeviceEpochCache(_FakeLoader(n_days), torch.device("cpu"))
+        counts = []
+        seen = set()
+        for r in range(ws):
+            days = list(cache.order(epoch=1, shuffle=True, rank=r, world_size=ws))
+            counts.append(len(days))
+            assert len(days) == cache.num_batches(r, ws)
+            for x, _ in days:
+                seen.add(x.data_ptr())
+        assert len(set(counts)) == 1, (n_days, ws, counts)
+        assert counts[0] == (n_days + ws - 1) // ws
+        # union of shards covers every day exactly (padding repeats some)
+        assert len(seen) == n_days
+
+
+def test_batch_sampler_padded_sharding_equal_counts():
+    from factorvae_amd.data.sampler import init_data_loader
+    from factorvae_amd.data.synthetic import make_synthetic_frame
+
+    df = make_synthetic_frame(n_days=15, n_stocks=6, seed=11)
+    lens = []
+    for r in range(2):
+        dl = init_data_loader(df, step_len=4, shuffle=True, start=None,
+                              end=None, rank=r, world_size=2, seed=3)
+        batches = list(dl.batch_sampler)
+        lens.append(len(batches))
+        assert len(batches) == len(dl.batch_sampler)
+    assert lens[0] == lens[1]
+
+
+def _odd_day_worker(rank, world_size, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        from factorvae_amd.data.device_cache import DeviceEpochCache
+        
+        set_seed(0)
+        model = build_factorvae(num_latent=C, hidden_size=H,
+                                num_portfolio=M, num_factor=K)
+        bucket = FlatGradBucket(model.parameters())
+        opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+        cache = DeviceEpochCache(_FakeLoader(3, n=N, t=T, c=C),
+                                 torch.device("cpu"))
+        # 3 days, 2 ranks: without padding rank 1 would run 1 step while
+        # rank 0 runs 2 -> mismatched all_reduce count -> hang here
+        steps = 0
+        for inputs, labels in cache.order(0, shuffle=True, rank=rank,
+                                          world_size=world_size):
+            bucket.zero_()
+            loss, *_ = model(inputs, labels)
+            loss.backward()
+            bucket.all_reduce_()
+            opt.step()
+            steps += 1
+        q.put((rank, steps))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_odd_day_count_two_ranks_no_hang():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_odd_day_worker, args=(r, 2, 29534, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, steps = q.get(timeout=100)
+        results[rank] = steps
+    for p in procs:
+        p.join(timeout=30)
+    assert results[0] == results[1] == 2
