@@ -334,7 +334,10 @@ class FusedTrainer:
         w["hpart"] = f(K * (2 * H + 2))
         cb = (C + 63) // 64
         ty = (2048 + cb - 1) // cb
-        rpb = max(64, (R + ty - 1) // ty)
+        # mirror the launcher's rpb clamp to [64, 8192]
+        # (extractor.hip fv_ln_bwd_params): without the upper clamp,
+        # yblocks at very large R exceeds the partial buffer
+        rpb = min(8192, max(64, (R + ty - 1) // ty))
         w["ln_part"] = f(((R + rpb - 1) // rpb + 1) * 2 * C)
         w["dzx"] = f(R, C)
         w["dxln"] = f(R, C)
@@ -671,19 +674,34 @@ class FusedTrainer:
         if self.training:
             self.ws["mask"].bernoulli_(1.0 - self.DROPOUT_P)
 
+    # max distinct (N, T) shapes whose workspaces + captured graphs stay
+    # resident; beyond this the least-recently-used shape is dropped
+    # (long multi-year runs see hundreds of distinct daily stock counts)
+    WS_CACHE_MAX = 64
+
     def _ensure_ws(self, N: int, T: int):
         """Workspaces (and captured graphs) are cached per (N, T): real
         universes have a different stock count N every day, and 288 GB
         HBM3E makes keeping one ~60 MB workspace per distinct shape far
-        cheaper than re-capturing the step graph each day."""
+        cheaper than re-capturing the step graph each day. An LRU cap
+        bounds total growth."""
         if self._ws_n == N and getattr(self, "_ws_t", None) == T:
             return
+        lru = self.__dict__.setdefault("_ws_lru", [])
+        if (N, T) in lru:
+            lru.remove((N, T))
+        lru.append((N, T))
         w = self._ws_cache.get((N, T))
         if w is not None:
             self.ws = w
             self._ws_n = N
             self._ws_t = T
         else:
+            while len(lru) > self.WS_CACHE_MAX:
+                old = lru.pop(0)
+                self._ws_cache.pop(old, None)
+                for kind in ("train", "predict"):
+                    self._graphs.pop((kind,) + old, None)
             self._alloc_ws(N, T)
 
     # -------------------------------------------------- graph capability probe

@@ -38,8 +38,10 @@ def build_argparser() -> argparse.ArgumentParser:
     p.add_argument("--start", type=str, default=None)
     p.add_argument("--end", type=str, default=None)
     p.add_argument("--out_dir", type=str, default="./scores")
-    p.add_argument("--normalize", action="store_true", default=True)
-    p.add_argument("--select_feature", action="store_true", default=False)
+    p.add_argument("--normalize", action=argparse.BooleanOptionalAction,
+                   default=True)
+    p.add_argument("--select_feature", action=argparse.BooleanOptionalAction,
+                   default=False)
     p.add_argument("--backtest", action="store_true",
                    help="run the top-k dropout backtest + risk report")
     p.add_argument("--topk", type=int, default=50)

@@ -141,7 +141,8 @@ def RankIC(df: pd.DataFrame, column1: str = "LABEL0", column2: str = "Pred"):
         ric_values.append(ric)
 
     if not ric_values:
-        return np.nan, np.nan
+        # same type as the populated case (callers index/print a frame)
+        return pd.DataFrame({"RankIC": [np.nan], "RankIC_IR": [np.nan]})
 
     ric = np.mean(ric_values)
     std = np.std(ric_values)

@@ -86,14 +86,21 @@ def test_date_range_slicing():
 
 
 def test_day_sharding_partition():
-    """DP sharding: ranks' day sets partition the epoch exactly."""
+    """DP sharding: equal per-rank counts (padded), union covers every
+    day; at most world_size-1 days repeat (wrap-around padding keeps the
+    per-step collectives in lockstep across ranks)."""
     df = make_synthetic_frame(n_days=23, n_stocks=6, seed=3)
     seen = []
+    per_rank = []
     for r in range(3):
         dl = init_data_loader(df, step_len=4, shuffle=True, start=None,
                               end=None, rank=r, world_size=3, seed=9)
         dl.batch_sampler.set_epoch(2)
+        cnt = 0
         for _, idx in dl:
             seen.append(tuple(sorted(map(tuple, idx[0]))))
-    assert len(seen) == 23
-    assert len(set(seen)) == 23  # disjoint, complete
+            cnt += 1
+        per_rank.append(cnt)
+    assert per_rank == [8, 8, 8]  # ceil(23/3) on every rank
+    assert len(seen) == 24
+    assert len(set(seen)) == 23  # complete; exactly 1 padded repeat
