@@ -74,9 +74,8 @@ def main():
     model = build_factorvae(num_latent=C, hidden_size=H, num_portfolio=M,
                             num_factor=K).to(device)
 
-    engine_name = args.engine
-    if engine_name == "auto":
-        engine_name = "fused" if device.type == "cuda" else "eager"
+    from factorvae_amd.engine.trainer import resolve_engine
+    engine_name = resolve_engine(args.engine, H, device.type)
 
     # synthetic device-resident day tensors (weak scaling: each rank its own days)
     g = torch.Generator(device=device)

@@ -37,6 +37,11 @@ from ..ops import get_extension
 from ..parallel.ddp import get_world_size, is_distributed
 
 
+class UnsupportedShapeError(ValueError):
+    """Model shape outside the fused kernels' tiling envelope; callers
+    fall back to the eager (PyTorch-ROCm) engine."""
+
+
 class _Arena:
     """Flat fp32 buffer + named views with a custom packing order."""
 
@@ -98,7 +103,13 @@ class FusedTrainer:
         enc = model.factor_encoder
         self.M = enc.linear.out_features
         self.K = model.factor_predictor.num_factor
-        assert self.H <= 64, "fused engine supports hidden_size <= 64"
+        if self.H > 64:
+            # the GRU/attention kernel register/LDS tilings are sized for
+            # H <= 64 (all reference checkpoints: H in {20, 32, 64});
+            # callers catch this and fall back to the eager engine
+            raise UnsupportedShapeError(
+                f"fused engine supports hidden_size <= 64, got {self.H}; "
+                f"use engine='eager'")
 
         self._build_param_arena()
         if self.bf16:

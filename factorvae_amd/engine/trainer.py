@@ -33,6 +33,26 @@ from ..parallel.ddp import (
 from ..utils import checkpoint_path, set_seed
 
 
+def resolve_engine(engine: str, hidden_size: int, device_type: str,
+                   warn: bool = True) -> str:
+    """Pick the execution engine. "auto" = fused on GPU, eager on CPU;
+    shapes outside the fused kernels' envelope (hidden_size > 64) fall
+    back to eager with a warning instead of failing (the public CLI
+    accepts any --hidden_size)."""
+    engine = engine or "auto"
+    if engine == "auto":
+        engine = "fused" if device_type == "cuda" else "eager"
+    if engine == "fused" and hidden_size > 64:
+        if warn:
+            import warnings
+            warnings.warn(
+                f"hidden_size={hidden_size} exceeds the fused engine's "
+                f"H<=64 kernel envelope; falling back to the eager engine",
+                RuntimeWarning, stacklevel=2)
+        engine = "eager"
+    return engine
+
+
 def _split_batch(char_with_label: torch.Tensor, device: torch.device):
     """Slice the (N,T,C+1) block into features (N,T,C) and the last-step
     label (N,1) (/root/reference/train_model.py:18-24)."""
@@ -156,9 +176,8 @@ def train_main(args, data_args, df=None) -> float:
         df = df.rename(columns={df.columns[-1]: "LABEL0"})
 
     device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
-    engine = getattr(args, "engine", "auto") or "auto"
-    if engine == "auto":
-        engine = "fused" if device.type == "cuda" else "eager"
+    engine = resolve_engine(getattr(args, "engine", "auto"),
+                            args.hidden_size, device.type)
 
     logger = MetricsLogger(args.run_name, out_dir=args.save_dir,
                            use_wandb=getattr(args, "wandb", False),

@@ -104,3 +104,23 @@ def test_day_sharding_partition():
     assert per_rank == [8, 8, 8]  # ceil(23/3) on every rank
     assert len(seen) == 24
     assert len(set(seen)) == 23  # complete; exactly 1 padded repeat
+
+
+def test_resolve_engine_h_gt_64_falls_back_to_eager():
+    """hidden_size > 64 exceeds the fused kernels' tiling envelope: the
+    engine resolver falls back to eager with a warning instead of the
+    fused engine raising mid-run (the CLI accepts any --hidden_size)."""
+    import warnings
+
+    from factorvae_amd.engine.trainer import resolve_engine
+
+    assert resolve_engine("auto", 64, "cpu") == "eager"
+    assert resolve_engine("auto", 64, "cuda") == "fused"
+    with warnings.catch_warnings(record=True) as rec:
+        warnings.simplefilter("always")
+        assert resolve_engine("fused", 128, "cuda") == "eager"
+    assert any("H<=64" in str(w.message) for w in rec)
+    assert resolve_engine("fused", 64, "cuda") == "fused"
+    # the engine itself raises the typed error (callers can catch it)
+    from factorvae_amd.engine.fused import UnsupportedShapeError
+    assert issubclass(UnsupportedShapeError, ValueError)

@@ -295,6 +295,11 @@ def eager_forward_explicit_noise(model, x, y, eps, mask, training):
     (300, 20, 158, 64, 128, 20, True),
     (300, 20, 158, 64, 128, 20, False),
     (37, 7, 33, 48, 24, 12, True),
+    # megakernel cutover boundaries: N=385 (first non-fused attn bwd),
+    # N=449 (first non-fused enc/attn fwd) — guards the N<=448 / N<=384
+    # threshold logic in engine/fused.py
+    (385, 20, 158, 64, 128, 20, True),
+    (449, 20, 158, 64, 128, 20, True),
     # full A-share shape (BASELINE.json config 4): N=3500, T=60, K=96
     (3500, 60, 158, 64, 128, 96, True),
 ])
