@@ -95,7 +95,11 @@ class FusedTrainer:
         self.eta_min = eta_min
         self.t_max = t_max
         self.training = train
-        self.use_graph = use_graph
+        import os as _os
+        # FV_GRAPH=0: eager-launch every kernel (no hipGraph capture at
+        # all) — the last-resort guard rail for capture-hostile boxes
+        self.use_graph = use_graph and _os.environ.get("FV_GRAPH",
+                                                       "1") != "0"
 
         fe = model.feature_extractor
         self.C = fe.num_latent
@@ -720,23 +724,34 @@ class FusedTrainer:
 
     def _probe_caps(self):
         """Can torch RNG ops / RCCL collectives be captured in a hipGraph?
-        Probed once; capture plans adapt (fallbacks keep correctness)."""
+        Probed once; capture plans adapt (fallbacks keep correctness).
+
+        Env guard rails (hardware de-risk for multi-GPU runs):
+        - FV_COMM_GRAPH=0 forces the split plan (never attempts to
+          capture an RCCL collective into a hipGraph — the safe path if
+          a ROCm build hangs at comm capture);
+        - FV_RNG_GRAPH=0 keeps RNG fills outside the graph.
+        """
         if self._caps is not None:
             return self._caps
-        rng_ok = True
-        try:
-            t_ = torch.zeros(8, device=self.device)
-            torch.cuda.synchronize()
-            gp = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(gp):
-                t_.normal_()
-            gp.replay()
-            torch.cuda.synchronize()
-        except Exception:
-            rng_ok = False
-            torch.cuda.synchronize()
+        import os as _os
+        allow_comm = _os.environ.get("FV_COMM_GRAPH", "1") != "0"
+        allow_rng = _os.environ.get("FV_RNG_GRAPH", "1") != "0"
+        rng_ok = allow_rng
+        if rng_ok:
+            try:
+                t_ = torch.zeros(8, device=self.device)
+                torch.cuda.synchronize()
+                gp = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(gp):
+                    t_.normal_()
+                gp.replay()
+                torch.cuda.synchronize()
+            except Exception:
+                rng_ok = False
+                torch.cuda.synchronize()
         comm_ok = False
-        if is_distributed():
+        if is_distributed() and allow_comm:
             try:
                 t_ = torch.ones(8, device=self.device)
                 torch.distributed.all_reduce(t_)  # eager warmup of the PG

@@ -23,17 +23,29 @@ import torch.distributed as dist
 def init_distributed(backend: Optional[str] = None) -> int:
     """Initialize torch.distributed from torchrun env vars; returns rank.
 
-    No-op (returns 0) when WORLD_SIZE is absent or 1.
+    No-op (returns 0) when WORLD_SIZE is absent or 1, unless
+    FV_FORCE_DIST=1 (hardware smoke of the RCCL path at world_size=1).
+
+    The process-group timeout defaults to FV_PG_TIMEOUT (300 s) so a
+    collective mismatch aborts the job with a clear error instead of
+    hanging until the outer driver kills it.
     """
+    from datetime import timedelta
+
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
-    if world_size <= 1:
+    force = os.environ.get("FV_FORCE_DIST", "0") == "1"
+    if world_size <= 1 and not force:
         return 0
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     if not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29500")
-        dist.init_process_group(backend=backend)
+        os.environ.setdefault("WORLD_SIZE", "1")
+        os.environ.setdefault("RANK", "0")
+        timeout = timedelta(seconds=int(os.environ.get("FV_PG_TIMEOUT",
+                                                       "300")))
+        dist.init_process_group(backend=backend, timeout=timeout)
     rank = dist.get_rank()
     if torch.cuda.is_available():
         local_rank = int(os.environ.get("LOCAL_RANK", rank % max(torch.cuda.device_count(), 1)))

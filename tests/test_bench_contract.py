@@ -58,3 +58,24 @@ def test_bench_torchrun_two_ranks_gloo():
     assert d["config"]["parallelism"] == "dp2"
     assert d["config"]["global_batch"] == 48
     assert d["scaling"] == "weak"
+
+
+@pytest.mark.timeout(300)
+def test_bench_forced_dist_world_size_1():
+    """FV_FORCE_DIST=1: a single-process run initializes the real
+    collective backend at world_size=1 (gloo on CPU; RCCL on a GPU box —
+    the hardware smoke for the DP path) and still reports n_gpus=1."""
+    env = dict(os.environ)
+    env.update({"FV_FORCE_DIST": "1", "MASTER_ADDR": "127.0.0.1",
+                "MASTER_PORT": "29572", "WORLD_SIZE": "1", "RANK": "0",
+                "LOCAL_RANK": "0"})
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "4", "--warmup", "1",
+         "--n_stocks", "24", "--seq_len", "5", "--num_factor", "4",
+         "--hidden_size", "16", "--num_portfolio", "8", "--n_days", "2",
+         "--engine", "eager"],
+        cwd=REPO, capture_output=True, text=True, timeout=240, env=env)
+    assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-2000:])
+    d = _parse_last_json(r.stdout)
+    assert d["n_gpus"] == 1
+    assert d["config"]["parallelism"] == "dp1"
