@@ -20,6 +20,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 import torch
@@ -97,6 +98,12 @@ def main():
             if args.steps % d == 0 and args.warmup % d == 0:
                 G = d
                 break
+        if is_distributed() or os.environ.get("FV_PRINT_CAPS") == "1":
+            rng_ok, comm_ok = trainer._probe_caps()
+            print(f"[bench rank {rank}] probe_caps: rng_in_graph={rng_ok} "
+                  f"comm_in_graph={comm_ok} world={world} "
+                  f"backend={torch.distributed.get_backend() if is_distributed() else 'none'}",
+                  file=sys.stderr, flush=True)
         runner, _ = trainer.make_bench_runner(days[:G])
 
         class _Multi:
