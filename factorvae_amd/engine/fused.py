@@ -722,6 +722,46 @@ class FusedTrainer:
     # -------------------------------------------------- graph capability probe
     _caps = None
 
+    class _abort_watchdog:
+        """Fail-fast guard around hipGraph captures that include RCCL
+        collectives at world_size > 1: a capture hang cannot be
+        interrupted in-process, so if it exceeds FV_CAPTURE_WATCHDOG_S
+        (default 180 s) the rank hard-exits — torchrun then fails the
+        whole job quickly instead of a multi-GPU scaling run hanging
+        until the outer driver's budget is gone. No-op at ws <= 1
+        (ws=1 hardware smoke: comm capture verified OK on MI355X,
+        profiles/r2_rccl_ws1.md)."""
+
+        def __init__(self, tag: str):
+            self.tag = tag
+
+        def __enter__(self):
+            import os as _os
+            import threading
+            self.evt = threading.Event()
+            self.armed = is_distributed() and get_world_size() > 1
+            if not self.armed:
+                return self
+            limit = float(_os.environ.get("FV_CAPTURE_WATCHDOG_S", "180"))
+
+            def _wd():
+                if not self.evt.wait(limit):
+                    import sys as _sys
+                    print(f"[fused] {self.tag} exceeded {limit:.0f}s — "
+                          f"aborting rank (set FV_COMM_GRAPH=0 to force "
+                          f"the split no-comm-capture plan)",
+                          file=_sys.stderr, flush=True)
+                    _os._exit(17)
+
+            self.thr = threading.Thread(target=_wd, daemon=True)
+            self.thr.start()
+            return self
+
+        def __exit__(self, *a):
+            if self.armed:
+                self.evt.set()
+            return False
+
     def _probe_caps(self):
         """Can torch RNG ops / RCCL collectives be captured in a hipGraph?
         Probed once; capture plans adapt (fallbacks keep correctness).
@@ -753,14 +793,15 @@ class FusedTrainer:
         comm_ok = False
         if is_distributed() and allow_comm:
             try:
-                t_ = torch.ones(8, device=self.device)
-                torch.distributed.all_reduce(t_)  # eager warmup of the PG
-                torch.cuda.synchronize()
-                gp = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(gp):
-                    torch.distributed.all_reduce(t_)
-                gp.replay()
-                torch.cuda.synchronize()
+                with self._abort_watchdog("RCCL comm-capture probe"):
+                    t_ = torch.ones(8, device=self.device)
+                    torch.distributed.all_reduce(t_)  # eager warmup of the PG
+                    torch.cuda.synchronize()
+                    gp = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(gp):
+                        torch.distributed.all_reduce(t_)
+                    gp.replay()
+                    torch.cuda.synchronize()
                 comm_ok = True
             except Exception:
                 comm_ok = False
@@ -867,10 +908,11 @@ class FusedTrainer:
                 self._launch_optimizer()
             self._graphs[key] = {"split": True, "g_fb": g1, "g_opt": g2}
         else:
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                self._graph_step_body(None, None, N, T, rng_ok,
-                                      is_distributed() and comm_ok)
+            with self._abort_watchdog("train-step graph capture"):
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._graph_step_body(None, None, N, T, rng_ok,
+                                          is_distributed() and comm_ok)
             self._graphs[key] = {"split": False, "g": g}
 
     # -------------------------------------------------------- bench fast path
@@ -908,15 +950,16 @@ class FusedTrainer:
                         for _ in range(G)]
 
         try:
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                for i, (x, y) in enumerate(days):
-                    if rng_bufs is not None:
-                        self.ws["eps"].copy_(rng_bufs[i][0])
-                        if self.training:
-                            self.ws["mask"].copy_(rng_bufs[i][1])
-                    self._graph_step_body(x, y, N, T, rng_ok,
-                                          is_distributed() and comm_ok)
+            with self._abort_watchdog("bench multi-step graph capture"):
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    for i, (x, y) in enumerate(days):
+                        if rng_bufs is not None:
+                            self.ws["eps"].copy_(rng_bufs[i][0])
+                            if self.training:
+                                self.ws["mask"].copy_(rng_bufs[i][1])
+                        self._graph_step_body(x, y, N, T, rng_ok,
+                                              is_distributed() and comm_ok)
         except Exception:
             torch.cuda.synchronize()
 
