@@ -342,6 +342,9 @@ class FusedTrainer:
         w["tn_part_u"] = f(small_mn)
         w["tn_part_s"] = f(small_mn)
         w["tn_partb_s"] = f(32 * max_m)
+        # attention-branch partials: that branch runs on its own stream
+        # concurrently with the dec/enc wgrads that use tn_part_s
+        w["tn_part_a"] = f(small_mn)
         # deterministic shared-grad partials (no float atomics anywhere)
         dec_iters = min(8, max(1, (N + 511) // 512))
         dec_nblk = (N + 4 * dec_iters - 1) // (4 * dec_iters) + 1
@@ -376,6 +379,46 @@ class FusedTrainer:
         self._ws_n = N
         self._ws_t = T
 
+    # ------------------------------------------------------ stream helpers
+    def _streams(self):
+        return (self.s_side, self.s_side2)
+
+    def _fork(self, si: int = 0):
+        """Make side stream si wait for everything issued on main."""
+        sides = self._streams()
+        if sides[si] is None:
+            return
+        e = torch.cuda.Event()
+        e.record(torch.cuda.current_stream(self.device))
+        sides[si].wait_event(e)
+
+    def _join(self, si: int = 0):
+        """Make main wait for everything issued on side stream si."""
+        sides = self._streams()
+        if sides[si] is None:
+            return
+        e = torch.cuda.Event()
+        e.record(sides[si])
+        torch.cuda.current_stream(self.device).wait_event(e)
+
+    class _OnSide:
+        """Context manager: run enqueues on a side stream (no-op when the
+        stream is unavailable, e.g. CPU)."""
+
+        def __init__(self, trainer, si: int = 0):
+            self.stream = trainer._streams()[si]
+
+        def __enter__(self):
+            if self.stream is not None:
+                self.ctx = torch.cuda.stream(self.stream)
+                self.ctx.__enter__()
+            return self
+
+        def __exit__(self, *a):
+            if self.stream is not None:
+                self.ctx.__exit__(*a)
+            return False
+
     # ------------------------------------------------------------ the step
     def _launch_forward(self, N: int, T: int, with_loss: bool = True,
                         x=None, y=None):
@@ -385,6 +428,12 @@ class FusedTrainer:
         x2d = (w["x"] if x is None else x).view(R, C)
         yv = w["y"] if y is None else y
         alpha = 1.0 / math.sqrt(float(H) + 1e-6)
+
+        # the K-head query/key projection depends only on parameters:
+        # issue it on the side stream overlapped with the extractor
+        self._fork(0)
+        with self._OnSide(self, 0):
+            ext.attn_qk_fwd(self.p_q, self.p_Wk, self.p_bk, w["qk"], w["cb"])
 
         if self.fp8:
             # fwd in e4m3 (weights+activations); also emits the bf16
@@ -424,6 +473,36 @@ class FusedTrainer:
                                   w["h_prev_bf"].view(-1))
         elif self.bf16:
             ext.cast_f32_bf16(w["h_prev"].view(-1), w["h_prev_bf"].view(-1))
+        # h is ready: the K-head attention branch (prior path) runs on the
+        # side stream, overlapped with the encoder + decoder branches on
+        # main — they are independent until the loss joins pmu/psig with
+        # the posterior-decoded recon
+        mask = w["mask"] if self.training else None
+        keep_inv = 1.0 / (1.0 - self.DROPOUT_P)
+        self._fork(0)
+        with self._OnSide(self, 0):
+            if N <= 448:
+                # whole per-head chain in ONE kernel (h staged in LDS)
+                ext.attn_fused_fwd(w["h"], w["qk"], w["cb"], mask, self.p_Wv,
+                                   self.p_bv, p("Wl"), p("bl"), p("wmu_p"),
+                                   p("bmu_p"), p("wsig_p"), p("bsig_p"),
+                                   w["a_att"], w["sd"], w["guard"], w["u"],
+                                   w["ctx"], w["hm2"], w["pmu"],
+                                   w["psig_pre"], w["psig"], w["psig_c"],
+                                   alpha, keep_inv)
+            else:
+                ext.gemm_nt(w["h"], w["qk"], w["cb"], w["s_att"], alpha,
+                            False, False)
+                ext.attn_softmax_fwd(w["s_att"], mask, w["a_att"], w["sd"],
+                                     w["guard"], keep_inv)
+                ext.gemm_tn(w["a_att"], w["h"], w["u"], w["tn_part_u"], 2,
+                            False)
+                ext.attn_ctx_fwd(w["u"], self.p_Wv, self.p_bv, w["guard"],
+                                 w["ctx"])
+                ext.pred_mlp_fwd(w["ctx"], p("Wl"), p("bl"), p("wmu_p"),
+                                 p("bmu_p"), p("wsig_p"), p("bsig_p"),
+                                 w["hm2"], w["pmu"], w["psig_pre"], w["psig"],
+                                 w["psig_c"])
         if N <= 448:
             ext.enc_fused_fwd(w["h"], p("Wenc"), p("benc"), yv,
                               w["scores_enc"], w["a_enc"], w["yp"])
@@ -434,43 +513,30 @@ class FusedTrainer:
         ext.enc_heads_fwd(w["yp"], p("Wmu_e"), p("bmu_e"), p("Wsig_e"),
                           p("bsig_e"), w["fmu"], w["fsig_pre"], w["fsig"],
                           w["fsig_c"])
-        ext.attn_qk_fwd(self.p_q, self.p_Wk, self.p_bk, w["qk"], w["cb"])
-        mask = w["mask"] if self.training else None
-        keep_inv = 1.0 / (1.0 - self.DROPOUT_P)
-        if N <= 448:
-            # whole per-head chain in ONE kernel (h staged in LDS)
-            ext.attn_fused_fwd(w["h"], w["qk"], w["cb"], mask, self.p_Wv,
-                               self.p_bv, p("Wl"), p("bl"), p("wmu_p"),
-                               p("bmu_p"), p("wsig_p"), p("bsig_p"),
-                               w["a_att"], w["sd"], w["guard"], w["u"],
-                               w["ctx"], w["hm2"], w["pmu"], w["psig_pre"],
-                               w["psig"], w["psig_c"], alpha, keep_inv)
-        else:
-            ext.gemm_nt(w["h"], w["qk"], w["cb"], w["s_att"], alpha, False,
-                        False)
-            ext.attn_softmax_fwd(w["s_att"], mask, w["a_att"], w["sd"],
-                                 w["guard"], keep_inv)
-            ext.gemm_tn(w["a_att"], w["h"], w["u"], w["tn_part_u"], 2, False)
-            ext.attn_ctx_fwd(w["u"], self.p_Wv, self.p_bv, w["guard"],
-                             w["ctx"])
-            ext.pred_mlp_fwd(w["ctx"], p("Wl"), p("bl"), p("wmu_p"),
-                             p("bmu_p"), p("wsig_p"), p("bsig_p"), w["hm2"],
-                             w["pmu"], w["psig_pre"], w["psig"], w["psig_c"])
         ext.dec_fwd(w["h"], p("W1d"), p("b1d"), p("wmu_d"), p("bmu_d"),
                     p("wsig_d"), p("bsig_d"), p("Wb"), p("bb"), w["fmu"],
                     w["fsig_c"], w["eps"], w["recon"], w["a1"], w["beta"],
                     w["asig_pre"], w["sigma"])
+        self._join(0)
         if with_loss:
-            ext.loss_fwd(w["recon"], yv, w["fmu"], w["fsig_c"], w["pmu"],
-                         w["psig_c"], w["loss"], w["mse"], w["kl"])
+            # one kernel: loss scalars AND the five loss input-gradients
+            # (the gradient writes cost nothing extra; validation simply
+            # ignores them)
+            ext.loss_fused(w["recon"], yv, w["fmu"], w["fsig_c"], w["pmu"],
+                           w["psig_c"], w["loss"], w["mse"], w["kl"],
+                           w["drecon"], w["dfmu"], w["dfsig_c"], w["dpmu"],
+                           w["dpsig_c"], 1.0)
 
     def _launch_backward(self, N: int, T: int, x=None, y=None,
                          comm_overlap: bool = False):
-        """Backward pass. The activation-gradient chain runs on the main
-        stream; every weight-gradient GEMM / colsum forks onto the side
-        stream as soon as its producer is done (fork/join events become
-        hipGraph edges under capture), so wgrad work overlaps the dgrad
-        critical path and joins before the optimizer."""
+        """Backward pass. Three independent gradient branches run
+        concurrently: the decoder+encoder chain on the main stream, the
+        K-head attention backward on side stream 2, and every
+        weight-gradient GEMM / colsum forked onto side stream 1 as soon
+        as its producer is done (fork/join events become hipGraph edges
+        under capture). The loss input-gradients were already produced
+        by forward's loss_fused kernel; the three dh contributions are
+        assembled by ONE dh_combine kernel at the join."""
         ext, w, p, g = self.ext, self.ws, self.p, self.g
         C, H, M, K = self.C, self.H, self.M, self.K
         R = N * T
@@ -483,105 +549,81 @@ class FusedTrainer:
                   or max(1, min(32, R // 1024)))
 
         main = torch.cuda.current_stream(self.device) if self.s_side else None
-        sides = (self.s_side, self.s_side2)
+        fork, _on_side = self._fork, self._OnSide
+        sides = self._streams()
 
-        def fork(si=0):
-            """Make side stream si wait for everything issued on main."""
-            if sides[si] is None:
-                return
-            e = torch.cuda.Event()
-            e.record(main)
-            sides[si].wait_event(e)
-
-        class _on_side:
-            def __init__(_s, si=0):
-                _s.si = si
-
-            def __enter__(_s):
-                if sides[_s.si] is not None:
-                    _s.ctx = torch.cuda.stream(sides[_s.si])
-                    _s.ctx.__enter__()
-                return _s
-
-            def __exit__(_s, *a):
-                if sides[_s.si] is not None:
-                    _s.ctx.__exit__(*a)
-
-        ext.loss_bwd(w["recon"], yv, w["fmu"], w["fsig_c"], w["pmu"],
-                     w["psig_c"], w["drecon"], w["dfmu"], w["dfsig_c"],
-                     w["dpmu"], w["dpsig_c"], 1.0)
-        ext.dec_bwd(w["drecon"], w["h"], w["a1"], w["beta"], w["asig_pre"],
-                    w["sigma"], w["eps"], w["fmu"], w["fsig_c"], p("W1d"),
-                    p("wmu_d"), p("wsig_d"), p("Wb"), w["dh"], w["dz1"],
-                    w["dbeta"], w["dec_part"], w["dfmu"], w["dfsig_c"],
-                    g("wmu_d"), g("bmu_d"), g("wsig_d"), g("bsig_d"))
-        fork()
-        with _on_side():
-            ext.gemm_tn(w["dz1"], w["h"], g("W1d"), w["tn_part_s"], 2, True,
-                        g("b1d"), w["tn_partb_s"])
-            ext.gemm_tn(w["dbeta"], w["h"], g("Wb"), w["tn_part_s"], 2, True,
-                        g("bb"), w["tn_partb_s"])
-
-        # predictor MLP + attention backward
+        # ---- attention backward branch (side stream 2): independent of
+        # the decoder/encoder chain given the loss grads from forward
         mask = w["mask"] if self.training else None
         gWv = self._gstack("Wv.0", (K, H, H))
         gbv = self._gstack("bv.0", (K, H))
         gq = self._gstack("q_att.0", (K, H))
         gWk = self._gstack("Wk.0", (K, H, H))
         gbk = self._gstack("bk.0", (K, H))
-        if N <= 384:
-            # whole per-head backward chain in ONE kernel (incl. the
-            # query/key/value wgrads); only the Wl wgrad (cross-head)
-            # stays on the side stream and dh accumulation stays as the
-            # two deterministic GEMMs below
-            ext.attn_fused_bwd(w["dpmu"], w["dpsig_c"], w["psig"],
-                               w["psig_pre"], w["hm2"], p("wmu_p"),
-                               p("wsig_p"), p("Wl"), w["h"], w["a_att"],
-                               w["sd"], mask, w["guard"], w["u"], self.p_Wv,
-                               self.p_q, self.p_Wk, self.p_bk, w["dz2"],
-                               w["du"], w["ds"], w["dc"], gWv, gbv, gq, gWk,
-                               gbk, w["hpart"], g("wmu_p"), g("bmu_p"),
-                               g("wsig_p"), g("bsig_p"), alpha, keep_inv)
-            fork()
-            with _on_side():
-                ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True,
-                            g("bl"))
-        else:
-            ext.pred_mlp_bwd(w["dpmu"], w["dpsig_c"], w["psig"],
-                             w["psig_pre"], w["hm2"], p("wmu_p"),
-                             p("wsig_p"), w["dz2"], w["hpart"], g("wmu_p"),
-                             g("bmu_p"), g("wsig_p"), g("bsig_p"))
-            fork()
-            with _on_side():
-                ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True,
-                            g("bl"))
-            ext.gemm_nn(w["dz2"], p("Wl"), None, w["dctx"], 1.0, False,
-                        False)
-            ext.attn_head_bwd(w["dctx"], w["u"], self.p_Wv, w["guard"],
-                              w["du"], gWv, gbv)
-            ext.gemm_nt(w["h"], w["du"], None, w["da"], 1.0, False, False)
-            ext.attn_softmax_bwd(w["da"], w["a_att"], w["sd"], mask,
-                                 w["guard"], w["ds"], w["dc"], keep_inv,
-                                 alpha)
-            fork()
-            with _on_side():
-                ext.gemm_tn(w["ds"], w["h"], w["dqk"], w["tn_part_s"], 2,
+        fork(1)
+        with _on_side(self, 1):
+            if N <= 384:
+                # whole per-head backward chain in ONE kernel (incl. the
+                # query/key/value wgrads)
+                ext.attn_fused_bwd(w["dpmu"], w["dpsig_c"], w["psig"],
+                                   w["psig_pre"], w["hm2"], p("wmu_p"),
+                                   p("wsig_p"), p("Wl"), w["h"], w["a_att"],
+                                   w["sd"], mask, w["guard"], w["u"],
+                                   self.p_Wv, self.p_q, self.p_Wk, self.p_bk,
+                                   w["dz2"], w["du"], w["ds"], w["dc"], gWv,
+                                   gbv, gq, gWk, gbk, w["hpart"], g("wmu_p"),
+                                   g("bmu_p"), g("wsig_p"), g("bsig_p"),
+                                   alpha, keep_inv)
+            else:
+                ext.pred_mlp_bwd(w["dpmu"], w["dpsig_c"], w["psig"],
+                                 w["psig_pre"], w["hm2"], p("wmu_p"),
+                                 p("wsig_p"), w["dz2"], w["hpart"],
+                                 g("wmu_p"), g("bmu_p"), g("wsig_p"),
+                                 g("bsig_p"))
+                ext.gemm_nn(w["dz2"], p("Wl"), None, w["dctx"], 1.0, False,
+                            False)
+                ext.attn_head_bwd(w["dctx"], w["u"], self.p_Wv, w["guard"],
+                                  w["du"], gWv, gbv)
+                ext.gemm_nt(w["h"], w["du"], None, w["da"], 1.0, False,
+                            False)
+                ext.attn_softmax_bwd(w["da"], w["a_att"], w["sd"], mask,
+                                     w["guard"], w["ds"], w["dc"], keep_inv,
+                                     alpha)
+                ext.gemm_tn(w["ds"], w["h"], w["dqk"], w["tn_part_a"], 2,
                             False)
                 ext.attn_qk_bwd(w["dqk"], w["dc"], self.p_q, self.p_Wk,
                                 self.p_bk, gq, gWk, gbk)
-        ext.gemm_nn(w["ds"], w["qk"], None, w["dh"], 1.0, True, False)
-        ext.gemm_nn(w["a_att"], w["du"], None, w["dh"], 1.0, True, False)
+            # cross-head Wl wgrad (dz2 is final in both branches)
+            ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True, g("bl"))
 
-        # encoder backward (heads + stock-axis softmax bwd, one kernel)
+        # ---- decoder backward (main)
+        ext.dec_bwd(w["drecon"], w["h"], w["a1"], w["beta"], w["asig_pre"],
+                    w["sigma"], w["eps"], w["fmu"], w["fsig_c"], p("W1d"),
+                    p("wmu_d"), p("wsig_d"), p("Wb"), w["dh"], w["dz1"],
+                    w["dbeta"], w["dec_part"], w["dfmu"], w["dfsig_c"],
+                    g("wmu_d"), g("bmu_d"), g("wsig_d"), g("bsig_d"))
+        fork()
+        with _on_side(self):
+            ext.gemm_tn(w["dz1"], w["h"], g("W1d"), w["tn_part_s"], 2, True,
+                        g("b1d"), w["tn_partb_s"])
+            ext.gemm_tn(w["dbeta"], w["h"], g("Wb"), w["tn_part_s"], 2, True,
+                        g("bb"), w["tn_partb_s"])
+
+        # ---- encoder backward (heads + stock-axis softmax bwd, main)
         ext.enc_bwd_fused(w["dfmu"], w["dfsig_c"], w["fsig"], w["fsig_pre"],
                           w["yp"], p("Wmu_e"), p("Wsig_e"), w["a_enc"], yv,
                           w["dscores"], g("Wmu_e"), g("bmu_e"), g("Wsig_e"),
                           g("bsig_e"))
         fork()
-        with _on_side():
+        with _on_side(self):
             ext.gemm_tn(w["dscores"], w["h"], g("Wenc"), w["tn_part_s"], 2,
                         True, g("benc"), w["tn_partb_s"])
-        ext.gemm_nn(w["dscores"], p("Wenc"), None, w["dh"], 1.0, True, False)
+
+        # ---- join attention branch; assemble all three dh contributions
+        # (attention ds@qk + a@du, encoder dscores@Wenc) in ONE kernel
+        self._join(1)
+        ext.dh_combine(w["dh"], w["ds"], w["qk"], w["a_att"], w["du"],
+                       w["dscores"], p("Wenc"))
 
         if comm_overlap:
             # every non-extractor gradient (arena tail from Wenc on) is
@@ -614,7 +656,7 @@ class FusedTrainer:
                         w["dgi"], w["dgh"], N, T, H)
         if self.bf16:
             fork(1)
-            with _on_side(1):
+            with _on_side(self, 1):
                 ext.gemm_tn_bf16(w["dgh_bf"].view(R, 3 * H), w["h_prev_bf"],
                                  g("Whh"), w["tn_part"], chunks, True,
                                  g("bhh"), w["tn_partb"])
@@ -625,7 +667,7 @@ class FusedTrainer:
                              None, w["dzx_bf"], 1.0, False, False,
                              w["xp_bf"])
             fork(1)
-            with _on_side(1):
+            with _on_side(self, 1):
                 ext.gemm_tn_bf16(w["dzx_bf"], w["xln_bf"], g("W1x"),
                                  w["tn_part3"], chunks, True, g("b1x"),
                                  w["tn_partb3"])
@@ -633,7 +675,7 @@ class FusedTrainer:
                              None, 1.0, False, False)
         else:
             fork(1)
-            with _on_side(1):
+            with _on_side(self, 1):
                 ext.gemm_tn(w["dgh"].view(R, 3 * H), w["h_prev"].view(R, H),
                             g("Whh"), w["tn_part"], chunks, True,
                             g("bhh"), w["tn_partb"])
@@ -644,12 +686,12 @@ class FusedTrainer:
                         False, False)
             ext.lrelu_bwd(w["dxp"], w["xp"], w["dzx"])
             fork(1)
-            with _on_side(1):
+            with _on_side(self, 1):
                 ext.gemm_tn(w["dzx"], w["xln"], g("W1x"), w["tn_part3"], chunks,
                             True, g("b1x"), w["tn_partb3"])
             ext.gemm_nn(w["dzx"], p("W1x"), None, w["dxln"], 1.0, False, False)
         fork()
-        with _on_side():
+        with _on_side(self):
             ext.ln_bwd_params(x2d, w["dxln"], w["mean"], w["rstd"],
                               w["ln_part"], g("ln_g"), g("ln_b"), chunks)
         # join: main waits for all side-stream wgrad work (+ the early

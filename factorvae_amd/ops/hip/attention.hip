@@ -655,7 +655,65 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_kernel(
   }
 }
 
+// dh assembly: dh[n][j] += ds[n,:]·qk[:,j] + a[n,:]·du[:,j]
+//                        + dscores[n,:]·Wenc[:,j]
+// One kernel replaces the three latency-serialized small NN GEMMs that
+// accumulate the attention and encoder contributions into the hidden
+// gradient on the backward critical path (engine/fused.py). The three
+// small B-matrices (qk/du: (K,H), Wenc: (M,H)) are staged in LDS once
+// per workgroup; each lane owns one H column, each wave one stock row
+// per iteration. Deterministic: fixed summation order per element, no
+// atomics. H <= 64.
+__global__ __launch_bounds__(256) void dh_combine_kernel(
+    float* __restrict__ dh, const float* __restrict__ ds,
+    const float* __restrict__ qk, const float* __restrict__ a,
+    const float* __restrict__ du, const float* __restrict__ dscores,
+    const float* __restrict__ Wenc, int N, int K, int M, int H) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* qkS = (float*)smem;         // [K][H]
+  float* duS = qkS + (size_t)K * H;  // [K][H]
+  float* weS = duS + (size_t)K * H;  // [M][H]
+  const int tid = threadIdx.x;
+  for (int i = tid; i < K * H; i += 256) {
+    qkS[i] = qk[i];
+    duS[i] = du[i];
+  }
+  for (int i = tid; i < M * H; i += 256) weS[i] = Wenc[i];
+  __syncthreads();
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  if (lane >= H) return;  // no further barriers below
+  const int stride = gridDim.x * 4;
+  for (int n = blockIdx.x * 4 + w; n < N; n += stride) {
+    float acc = dh[(size_t)n * H + lane];
+    const float* dsr = ds + (size_t)n * K;
+    const float* ar = a + (size_t)n * K;
+    for (int k = 0; k < K; ++k) {
+      acc = fmaf(dsr[k], qkS[k * H + lane], acc);
+      acc = fmaf(ar[k], duS[k * H + lane], acc);
+    }
+    const float* dr = dscores + (size_t)n * M;
+    for (int m = 0; m < M; ++m) acc = fmaf(dr[m], weS[m * H + lane], acc);
+    dh[(size_t)n * H + lane] = acc;
+  }
+}
+
 extern "C" {
+
+hipError_t fv_dh_combine(float* dh, const float* ds, const float* qk,
+                         const float* a, const float* du,
+                         const float* dscores, const float* Wenc, int N,
+                         int K, int M, int H, hipStream_t s) {
+  if (H > 64) return hipErrorInvalidValue;
+  const size_t lds = ((size_t)2 * K * H + (size_t)M * H) * sizeof(float);
+  if (lds > 160 * 1024) return hipErrorInvalidValue;
+  int grid = (N + 3) / 4;
+  if (grid > 224) grid = 224;  // bound LDS re-staging traffic at large N
+  hipLaunchKernelGGL(dh_combine_kernel, dim3(grid), dim3(256), lds, s, dh,
+                     ds, qk, a, du, dscores, Wenc, N, K, M, H);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
 
 hipError_t fv_attn_fused_bwd(const float* dpmu, const float* dpsig_c,
                              const float* psig, const float* psig_pre,

@@ -128,6 +128,13 @@ hipError_t fv_loss_fwd(const float*, const float*, const float*, const float*,
 hipError_t fv_loss_bwd(const float*, const float*, const float*, const float*,
                        const float*, const float*, float*, float*, float*,
                        float*, float*, int, int, float, hipStream_t);
+hipError_t fv_loss_fused(const float*, const float*, const float*,
+                         const float*, const float*, const float*, float*,
+                         float*, float*, float*, float*, float*, float*,
+                         float*, int, int, float, hipStream_t);
+hipError_t fv_dh_combine(float*, const float*, const float*, const float*,
+                         const float*, const float*, const float*, int, int,
+                         int, int, hipStream_t);
 hipError_t fv_step_inc(int*, hipStream_t);
 hipError_t fv_adam(float*, const float*, float*, float*, const int*, long,
                    float, float, float, float, float, float, hipStream_t);
@@ -737,6 +744,35 @@ void loss_bwd(torch::Tensor recon, torch::Tensor y, torch::Tensor fmu,
                   cur_stream()));
 }
 
+void loss_fused(torch::Tensor recon, torch::Tensor y, torch::Tensor fmu,
+                torch::Tensor fsig_c, torch::Tensor pmu, torch::Tensor psig_c,
+                torch::Tensor loss, torch::Tensor mse, torch::Tensor kl,
+                torch::Tensor drecon, torch::Tensor dfmu,
+                torch::Tensor dfsig_c, torch::Tensor dpmu,
+                torch::Tensor dpsig_c, double gscale) {
+  CK(recon); CK(y); CK(fmu); CK(fsig_c); CK(pmu); CK(psig_c); CK(loss);
+  CK(mse); CK(kl); CK(drecon); CK(dfmu); CK(dfsig_c); CK(dpmu); CK(dpsig_c);
+  RUN(fv_loss_fused(fp(recon), fp(y), fp(fmu), fp(fsig_c), fp(pmu),
+                    fp(psig_c), fpm(loss), fpm(mse), fpm(kl), fpm(drecon),
+                    fpm(dfmu), fpm(dfsig_c), fpm(dpmu), fpm(dpsig_c),
+                    recon.numel(), fmu.numel(), (float)gscale, cur_stream()));
+}
+
+void dh_combine(torch::Tensor dh, torch::Tensor ds, torch::Tensor qk,
+                torch::Tensor a, torch::Tensor du, torch::Tensor dscores,
+                torch::Tensor Wenc) {
+  CK(dh); CK(ds); CK(qk); CK(a); CK(du); CK(dscores); CK(Wenc);
+  const int N_ = dh.size(0), H_ = dh.size(1);
+  const int K_ = qk.size(0), M_ = Wenc.size(0);
+  TORCH_CHECK(ds.size(0) == N_ && ds.size(1) == K_, "ds shape");
+  TORCH_CHECK(a.sizes() == ds.sizes(), "a shape");
+  TORCH_CHECK(du.sizes() == qk.sizes(), "du shape");
+  TORCH_CHECK(dscores.size(0) == N_ && dscores.size(1) == M_, "dscores");
+  TORCH_CHECK(Wenc.size(1) == H_ && qk.size(1) == H_, "B cols");
+  RUN(fv_dh_combine(fpm(dh), fp(ds), fp(qk), fp(a), fp(du), fp(dscores),
+                    fp(Wenc), N_, K_, M_, H_, cur_stream()));
+}
+
 void step_inc(torch::Tensor step_t) {
   TORCH_CHECK(step_t.scalar_type() == torch::kInt32 && step_t.is_cuda());
   RUN(fv_step_inc(step_t.data_ptr<int>(), cur_stream()));
@@ -812,6 +848,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_head_bwd", &attn_head_bwd);
   mod.def("attn_softmax_bwd", &attn_softmax_bwd);
   mod.def("attn_qk_bwd", &attn_qk_bwd);
+  mod.def("loss_fused", &loss_fused);
+  mod.def("dh_combine", &dh_combine);
   mod.def("pred_mlp_fwd", &pred_mlp_fwd);
   mod.def("pred_mlp_bwd", &pred_mlp_bwd);
   mod.def("dec_fwd", &dec_fwd);

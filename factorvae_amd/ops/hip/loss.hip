@@ -61,7 +61,67 @@ __global__ __launch_bounds__(256) void loss_bwd_kernel(
   }
 }
 
+// Training-step loss: forward scalars AND all five input gradients in
+// ONE launch (the backward math of loss_bwd_kernel needs nothing the
+// forward didn't already read, so splitting them was pure dispatch +
+// re-read cost on the critical path). Single workgroup; N and K loops
+// strided by 256.
+__global__ __launch_bounds__(256) void loss_fused_kernel(
+    const float* __restrict__ recon, const float* __restrict__ y,
+    const float* __restrict__ fmu, const float* __restrict__ fsig_c,
+    const float* __restrict__ pmu, const float* __restrict__ psig_c,
+    float* __restrict__ loss_out, float* __restrict__ mse_out,
+    float* __restrict__ kl_out, float* __restrict__ drecon,
+    float* __restrict__ dfmu, float* __restrict__ dfsig_c,
+    float* __restrict__ dpmu, float* __restrict__ dpsig_c, int N, int K,
+    float gscale) {
+  __shared__ float scratch[8];
+  const int tid = threadIdx.x;
+  const float dscale = gscale * 2.0f / N;
+
+  float se = 0.0f;
+  for (int n = tid; n < N; n += 256) {
+    const float d = recon[n] - y[n];
+    se = fmaf(d, d, se);
+    drecon[n] = dscale * d;
+  }
+  se = block_reduce_sum(se, scratch);
+  const float mse = se / N;
+
+  float kl = 0.0f;
+  for (int k = tid; k < K; k += 256) {
+    const float fs = fsig_c[k], ps = psig_c[k];
+    const float dmu = fmu[k] - pmu[k];
+    const float ps2 = ps * ps;
+    kl += __logf(ps / fs) + (fs * fs + dmu * dmu) / (2.0f * ps2) - 0.5f;
+    dfmu[k] = gscale * dmu / ps2;
+    dfsig_c[k] = gscale * (-1.0f / fs + fs / ps2);
+    dpmu[k] = gscale * (-dmu / ps2);
+    dpsig_c[k] = gscale * (1.0f / ps - (fs * fs + dmu * dmu) / (ps2 * ps));
+  }
+  kl = block_reduce_sum(kl, scratch);
+
+  if (tid == 0) {
+    mse_out[0] = mse;
+    kl_out[0] = kl;
+    loss_out[0] = mse + kl;
+  }
+}
+
 extern "C" {
+
+hipError_t fv_loss_fused(const float* recon, const float* y, const float* fmu,
+                         const float* fsig_c, const float* pmu,
+                         const float* psig_c, float* loss, float* mse,
+                         float* kl, float* drecon, float* dfmu,
+                         float* dfsig_c, float* dpmu, float* dpsig_c, int N,
+                         int K, float gscale, hipStream_t s) {
+  hipLaunchKernelGGL(loss_fused_kernel, dim3(1), dim3(256), 0, s, recon, y,
+                     fmu, fsig_c, pmu, psig_c, loss, mse, kl, drecon, dfmu,
+                     dfsig_c, dpmu, dpsig_c, N, K, gscale);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
 
 hipError_t fv_loss_fwd(const float* recon, const float* y, const float* fmu,
                        const float* fsig_c, const float* pmu,
