@@ -289,6 +289,244 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
   }
 }
 
+// --------------------------------------------------- weight-stationary
+// NT/NN variants for the model's actual extractor shapes (k = Ci <= 192,
+// Co <= 192, R = N*T large). The 64x64x64-tile kernels above run only
+// ~3 k-iterations per workgroup: the pipeline is too short to hide
+// global-load latency and the whole weight gets re-staged per row tile.
+// Here the WEIGHT BLOCK STAYS RESIDENT IN LDS for the kernel's lifetime
+// and each workgroup streams RB row-blocks (full k per block, A
+// double-buffered): a long pipeline with one global stream.
+#define WSKP 192            // max padded k (3H)
+#define WSA (WSKP + 8)      // A row stride (16B-aligned b128 reads)
+#define WSB 72              // NN B row stride (tr16 conflict-free)
+
+// NT: W (Co,Ci) row-major (k-contiguous) — direct b128 fragment reads.
+__global__ __launch_bounds__(256) void gemm_nt_bf16_ws_kernel(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ W,
+    const float* __restrict__ bias, float* __restrict__ out_f32,
+    __bf16* __restrict__ out_bf16, int R, int Ci, int Co, float alpha,
+    int flags, int rb_per_wg) {
+  __shared__ __bf16 Ws[64][WSA];
+  __shared__ __bf16 As[2][64][WSA];
+
+  const int c0 = blockIdx.x * BBC;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const int fi = lane & 15;
+  const int fk = lane >> 4;
+
+  const int KP = (Ci + 31) & ~31;   // padded k (multiple of 32)
+  const int KD = KP >> 1;           // dwords per staged row
+  const int nu = (64 * KD) >> 8;    // staging dwords per thread
+
+  // stage the weight block once (guarded, zero-padded)
+  for (int idx = tid; idx < 64 * KD; idx += 256) {
+    const int row = idx / KD;
+    const int cp = (idx % KD) * 2;
+    *(unsigned int*)&Ws[row][cp] = load_dw_guard(W, (long)c0 + row, cp, Co,
+                                                 Ci, Ci);
+  }
+
+  const int rt0 = blockIdx.y * rb_per_wg;
+  const int rtiles = (R + BBR - 1) / BBR;
+  const int rt_end = min(rt0 + rb_per_wg, rtiles);
+  if (rt0 >= rtiles) return;
+
+  unsigned int pa[24];
+  auto stage_regs = [&](int rt) {
+    const long r0 = (long)rt * BBR;
+    if (r0 + BBR <= R && KP == Ci) {
+      for (int u = 0; u < nu; ++u) {
+        const int idx = tid + u * 256;
+        const int row = idx / KD;
+        const int cp = (idx % KD) * 2;
+        pa[u] = *(const unsigned int*)(A + (r0 + row) * Ci + cp);
+      }
+    } else {
+      for (int u = 0; u < nu; ++u) {
+        const int idx = tid + u * 256;
+        const int row = idx / KD;
+        const int cp = (idx % KD) * 2;
+        pa[u] = load_dw_guard(A, r0 + row, cp, R, Ci, Ci);
+      }
+    }
+  };
+  auto regs_to_lds = [&](int buf) {
+    for (int u = 0; u < nu; ++u) {
+      const int idx = tid + u * 256;
+      const int row = idx / KD;
+      const int cp = (idx % KD) * 2;
+      *(unsigned int*)&As[buf][row][cp] = pa[u];
+    }
+  };
+
+  stage_regs(rt0);
+  regs_to_lds(0);
+
+  for (int rt = rt0; rt < rt_end; ++rt) {
+    __syncthreads();
+    if (rt + 1 < rt_end) stage_regs(rt + 1);
+    const int buf = (rt - rt0) & 1;
+    f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+    for (int k32 = 0; k32 < KP; k32 += 32) {
+      const bf16x8 a = *(const bf16x8*)&As[buf][wv * 16 + fi][k32 + fk * 8];
+#pragma unroll
+      for (int jt = 0; jt < 4; ++jt) {
+        const bf16x8 b = *(const bf16x8*)&Ws[jt * 16 + fi][k32 + fk * 8];
+        acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[jt],
+                                                          0, 0, 0);
+      }
+    }
+    if (rt + 1 < rt_end) {
+      __syncthreads();
+      regs_to_lds(1 - buf);
+    }
+    const long r0 = (long)rt * BBR;
+#pragma unroll
+    for (int jt = 0; jt < 4; ++jt) {
+      const int gc = c0 + jt * 16 + fi;
+      if (gc >= Co) continue;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const long gr = r0 + wv * 16 + fk * 4 + rr;
+        if (gr >= R) continue;
+        float v = acc[jt][rr];
+        if (flags & 4) v += bias[gc];
+        v *= alpha;
+        if (flags & 2) v = lrelu_(v);
+        if (out_f32) {
+          float* o = &out_f32[gr * Co + gc];
+          if (flags & 1) v += *o;
+          *o = v;
+        }
+        if (out_bf16) out_bf16[gr * Co + gc] = to_bf16(v);
+      }
+    }
+  }
+}
+
+// NN: B (Ci,Co) row-major (k over rows) — whole B resident in LDS,
+// fragments via ds_read_b64_tr_b16 (same supplier mapping as below).
+__global__ __launch_bounds__(256) void gemm_nn_bf16_ws_kernel(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    const float* __restrict__ bias, float* __restrict__ out_f32,
+    __bf16* __restrict__ out_bf16, const __bf16* __restrict__ Y,
+    int R, int Ci, int Co, float alpha, int flags, int rb_per_wg) {
+  __shared__ __bf16 Bs[WSKP][WSB];
+  __shared__ __bf16 As[2][64][WSA];
+
+  const int c0 = blockIdx.x * BBC;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const int fi = lane & 15;
+  const int fk = lane >> 4;
+  const int b_kofs = fi >> 2;
+  const int b_nq = (lane & 3) * 4;
+
+  const int KP = (Ci + 31) & ~31;
+  const int KD = KP >> 1;
+  const int nu = (64 * KD) >> 8;
+
+  // stage all of B for this column block (zero rows beyond Ci)
+  for (int idx = tid; idx < KP * 32; idx += 256) {
+    const int row = idx >> 5;
+    const int cp = (idx & 31) * 2;
+    *(unsigned int*)&Bs[row][cp] = load_dw_guard(B, row, c0 + cp, Ci, Co,
+                                                 Co);
+  }
+
+  const int rt0 = blockIdx.y * rb_per_wg;
+  const int rtiles = (R + BBR - 1) / BBR;
+  const int rt_end = min(rt0 + rb_per_wg, rtiles);
+  if (rt0 >= rtiles) return;
+
+  unsigned int pa[24];
+  auto stage_regs = [&](int rt) {
+    const long r0 = (long)rt * BBR;
+    if (r0 + BBR <= R && KP == Ci) {
+      for (int u = 0; u < nu; ++u) {
+        const int idx = tid + u * 256;
+        const int row = idx / KD;
+        const int cp = (idx % KD) * 2;
+        pa[u] = *(const unsigned int*)(A + (r0 + row) * Ci + cp);
+      }
+    } else {
+      for (int u = 0; u < nu; ++u) {
+        const int idx = tid + u * 256;
+        const int row = idx / KD;
+        const int cp = (idx % KD) * 2;
+        pa[u] = load_dw_guard(A, r0 + row, cp, R, Ci, Ci);
+      }
+    }
+  };
+  auto regs_to_lds = [&](int buf) {
+    for (int u = 0; u < nu; ++u) {
+      const int idx = tid + u * 256;
+      const int row = idx / KD;
+      const int cp = (idx % KD) * 2;
+      *(unsigned int*)&As[buf][row][cp] = pa[u];
+    }
+  };
+
+  stage_regs(rt0);
+  regs_to_lds(0);
+
+  for (int rt = rt0; rt < rt_end; ++rt) {
+    __syncthreads();
+    if (rt + 1 < rt_end) stage_regs(rt + 1);
+    const int buf = (rt - rt0) & 1;
+    f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+    for (int k32 = 0; k32 < KP; k32 += 32) {
+      const bf16x8 a = *(const bf16x8*)&As[buf][wv * 16 + fi][k32 + fk * 8];
+      const int kb = k32 + fk * 8 + b_kofs;
+#pragma unroll
+      for (int jt = 0; jt < 4; ++jt) {
+        s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4p)&Bs[kb][jt * 16 + b_nq]);
+        s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4p)&Bs[kb + 4][jt * 16 + b_nq]);
+        bf16x8 b;
+        *(bf16x4*)&b = *(bf16x4*)&b0;
+        *(((bf16x4*)&b) + 1) = *(bf16x4*)&b1;
+        acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[jt],
+                                                          0, 0, 0);
+      }
+    }
+    if (rt + 1 < rt_end) {
+      __syncthreads();
+      regs_to_lds(1 - buf);
+    }
+    const long r0 = (long)rt * BBR;
+#pragma unroll
+    for (int jt = 0; jt < 4; ++jt) {
+      const int gc = c0 + jt * 16 + fi;
+      if (gc >= Co) continue;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const long gr = r0 + wv * 16 + fk * 4 + rr;
+        if (gr >= R) continue;
+        float v = acc[jt][rr];
+        if (flags & 4) v += bias[gc];
+        v *= alpha;
+        if (flags & 2) v = lrelu_(v);
+        if (flags & 8) {
+          const float y = (float)Y[gr * Co + gc];
+          v *= (y > 0.0f ? 1.0f : 0.01f);
+        }
+        if (out_f32) {
+          float* o = &out_f32[gr * Co + gc];
+          if (flags & 1) v += *o;
+          *o = v;
+        }
+        if (out_bf16) out_bf16[gr * Co + gc] = to_bf16(v);
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------- TN
 // out(M,N) (+)= A(R,M)^T @ B(R,N), k = R. Both operands arrive k-major
 // (row-major over R), and MFMA fragments need 8 k-contiguous elements
@@ -555,7 +793,21 @@ hipError_t fv_gemm_nt_bf16(const void* A, const void* W, const float* bias,
                            int Co, float alpha, int accumulate, int act_lrelu,
                            hipStream_t stream) {
   int flags = (accumulate ? 1 : 0) | (act_lrelu ? 2 : 0) | (bias ? 4 : 0);
-  dim3 grid((Co + BBC - 1) / BBC, (R + BBR - 1) / BBR);
+  const int rtiles = (R + BBR - 1) / BBR;
+  const int cblocks = (Co + BBC - 1) / BBC;
+  if (Ci <= WSKP) {
+    // weight-stationary streaming path (all model shapes): ~2 WGs/CU,
+    // long row-block pipeline per WG
+    int rb = (rtiles * cblocks + 511) / 512;
+    if (rb < 1) rb = 1;
+    dim3 grid(cblocks, (rtiles + rb - 1) / rb);
+    hipLaunchKernelGGL(gemm_nt_bf16_ws_kernel, grid, dim3(256), 0, stream,
+                       (const __bf16*)A, (const __bf16*)W, bias, out_f32,
+                       (__bf16*)out_bf16, R, Ci, Co, alpha, flags, rb);
+    HIP_CHECK_LAST();
+    return hipSuccess;
+  }
+  dim3 grid(cblocks, rtiles);
   hipLaunchKernelGGL(gemm_nt_bf16_kernel, grid, dim3(256), 0, stream,
                      (const __bf16*)A, (const __bf16*)W, bias, out_f32,
                      (__bf16*)out_bf16, R, Ci, Co, alpha, flags);
@@ -569,7 +821,20 @@ hipError_t fv_gemm_nn_bf16(const void* A, const void* B, const float* bias,
                            int act_lrelu, hipStream_t stream) {
   int flags = (accumulate ? 1 : 0) | (act_lrelu ? 2 : 0) | (bias ? 4 : 0) |
               (Y ? 8 : 0);
-  dim3 grid((Co + BBC - 1) / BBC, (R + BBR - 1) / BBR);
+  const int rtiles = (R + BBR - 1) / BBR;
+  const int cblocks = (Co + BBC - 1) / BBC;
+  if (Ci <= WSKP) {
+    int rb = (rtiles * cblocks + 511) / 512;
+    if (rb < 1) rb = 1;
+    dim3 grid(cblocks, (rtiles + rb - 1) / rb);
+    hipLaunchKernelGGL(gemm_nn_bf16_ws_kernel, grid, dim3(256), 0, stream,
+                       (const __bf16*)A, (const __bf16*)B, bias, out_f32,
+                       (__bf16*)out_bf16, (const __bf16*)Y, R, Ci, Co, alpha,
+                       flags, rb);
+    HIP_CHECK_LAST();
+    return hipSuccess;
+  }
+  dim3 grid(cblocks, rtiles);
   hipLaunchKernelGGL(gemm_nn_bf16_kernel, grid, dim3(256), 0, stream,
                      (const __bf16*)A, (const __bf16*)B, bias, out_f32,
                      (__bf16*)out_bf16, (const __bf16*)Y, R, Ci, Co, alpha,
