@@ -118,10 +118,18 @@ class FusedTrainer:
         self._build_param_arena()
         if self.bf16:
             C3 = 3 * self.H
-            self.w1x_bf = torch.empty(self.C, self.C, dtype=torch.bfloat16,
-                                      device=self.device)
-            self.wih_bf = torch.empty(C3, self.C, dtype=torch.bfloat16,
-                                      device=self.device)
+            # padded (+ transposed-padded) weight shadows for the
+            # register-stationary NT kernel: k dim zero-padded to a
+            # multiple of 32 so the MFMA k-tail is exact (pads are
+            # zeroed here once and never written by the refresh kernel)
+            KPc = (self.C + 31) & ~31
+            KP3 = (C3 + 31) & ~31
+            zb = lambda *s: torch.zeros(*s, dtype=torch.bfloat16,
+                                        device=self.device)
+            self.w1x_p = zb(self.C, KPc)    # W1x (C,C) k-padded
+            self.w1xT_p = zb(self.C, KPc)   # W1x^T
+            self.wih_p = zb(C3, KPc)        # Wih (3H,C) k-padded
+            self.wihT_p = zb(self.C, KP3)   # Wih^T (C,3H) k-padded
             self.whh_bf = torch.empty(C3, self.H, dtype=torch.bfloat16,
                                       device=self.device)
             if self.fp8:
@@ -362,10 +370,17 @@ class FusedTrainer:
         if self.bf16:
             fb = lambda *shape: torch.zeros(*shape, device=d,
                                             dtype=torch.bfloat16)
-            w["xln_bf"] = fb(R, C)
-            w["xp_bf"] = fb(R, C)
-            w["dgi_bf"] = fb(R, 3 * H)
-            w["dzx_bf"] = fb(R, C)
+            # A-side operands of the register-stationary NT kernel get
+            # 8 elements of tail slack (its k-tail b128 fragment may
+            # read up to 4 bytes past the last row; the zero-padded
+            # weight makes the contribution exact)
+            fbs = lambda r, c: torch.zeros(r * c + 8, device=d,
+                                           dtype=torch.bfloat16
+                                           )[:r * c].view(r, c)
+            w["xln_bf"] = fbs(R, C)
+            w["xp_bf"] = fbs(R, C)
+            w["dgi_bf"] = fbs(R, 3 * H)
+            w["dzx_bf"] = fbs(R, C)
             w["h_prev_bf"] = fb(R, H)
             w["dgh_bf"] = fb(R, 3 * H)
             if self.fp8:
@@ -449,10 +464,10 @@ class FusedTrainer:
         elif self.bf16:
             ext.ln_fwd(x2d, p("ln_g"), p("ln_b"), None, w["mean"],
                        w["rstd"], 1e-5, w["xln_bf"])
-            ext.gemm_nt_bf16(w["xln_bf"], self.w1x_bf, p("b1x"), None,
-                             w["xp_bf"], 1.0, False, True)
-            ext.gemm_nt_bf16(w["xp_bf"], self.wih_bf, p("bih"), w["gi"],
-                             None, 1.0, False, False)
+            ext.gemm_nt_bf16_rs(w["xln_bf"], self.w1x_p, p("b1x"), None,
+                                w["xp_bf"], None, 1.0, True)
+            ext.gemm_nt_bf16_rs(w["xp_bf"], self.wih_p, p("bih"), w["gi"],
+                                None, None, 1.0, False)
         else:
             ext.ln_fwd(x2d, p("ln_g"), p("ln_b"), w["xln"],
                        w["mean"], w["rstd"], 1e-5)
@@ -663,16 +678,16 @@ class FusedTrainer:
                 ext.gemm_tn_bf16(w["dgi_bf"].view(R, 3 * H), w["xp_bf"],
                                  g("Wih"), w["tn_part2"], chunks, True,
                                  g("bih"), w["tn_partb2"])
-            ext.gemm_nn_bf16(w["dgi_bf"].view(R, 3 * H), self.wih_bf, None,
-                             None, w["dzx_bf"], 1.0, False, False,
-                             w["xp_bf"])
+            ext.gemm_nt_bf16_rs(w["dgi_bf"].view(R, 3 * H), self.wihT_p,
+                                None, None, w["dzx_bf"], w["xp_bf"], 1.0,
+                                False)
             fork(1)
             with _on_side(self, 1):
                 ext.gemm_tn_bf16(w["dzx_bf"], w["xln_bf"], g("W1x"),
                                  w["tn_part3"], chunks, True, g("b1x"),
                                  w["tn_partb3"])
-            ext.gemm_nn_bf16(w["dzx_bf"], self.w1x_bf, None, w["dxln"],
-                             None, 1.0, False, False)
+            ext.gemm_nt_bf16_rs(w["dzx_bf"], self.w1xT_p, None, w["dxln"],
+                                None, None, 1.0, False)
         else:
             fork(1)
             with _on_side(self, 1):
@@ -706,9 +721,9 @@ class FusedTrainer:
                 main.wait_event(e)
 
     def _refresh_bf16_shadows(self):
-        self.ext.cast3_f32_bf16(self.p("W1x"), self.w1x_bf,
-                                self.p("Wih"), self.wih_bf,
-                                self.p("Whh"), self.whh_bf)
+        self.ext.cast_shadows(self.p("W1x"), self.w1x_p, self.w1xT_p,
+                              self.p("Wih"), self.wih_p, self.wihT_p,
+                              self.p("Whh"), self.whh_bf)
         if self.fp8:
             self.ext.absmax_scale(self.p("W1x"), self.s_w1x, self.is_w1x)
             self.ext.cast_f32_fp8_scaled(self.p("W1x"), self.w1x_f8,

@@ -21,6 +21,12 @@ hipError_t fv_gemm_nt_bf16(const void*, const void*, const float*, float*,
 hipError_t fv_gemm_nn_bf16(const void*, const void*, const float*, float*,
                            void*, const void*, int, int, int, float, int,
                            int, hipStream_t);
+hipError_t fv_gemm_nt_bf16_rs(const void*, const void*, const float*,
+                              float*, void*, const void*, int, int, int,
+                              int, float, int, hipStream_t);
+hipError_t fv_cast_shadows(const float*, void*, void*, int, int, int, int,
+                           const float*, void*, void*, int, int, int, int,
+                           const float*, void*, long, hipStream_t);
 hipError_t fv_gemm_tn_bf16(const void*, const void*, float*, float*, float*,
                            float*, int, int, int, int, int, hipStream_t);
 hipError_t fv_cast_f32_bf16(const float*, void*, long, hipStream_t);
@@ -281,6 +287,57 @@ void gemm_tn_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor out,
   }
   RUN(fv_gemm_tn_bf16(bfpc(A), bfpc(B), fpm(out), pp, dbp, dbpp, R, M, N,
                       (int)r_chunks, accumulate, cur_stream()));
+}
+
+void gemm_nt_bf16_rs(torch::Tensor A, torch::Tensor Wp,
+                     c10::optional<torch::Tensor> bias,
+                     c10::optional<torch::Tensor> out_f32,
+                     c10::optional<torch::Tensor> out_bf16,
+                     c10::optional<torch::Tensor> lrelu_bwd_of,
+                     double alpha, bool act_lrelu) {
+  CKB(A); CKB(Wp);
+  const int R = A.size(0), Ci = A.size(1), KP = Wp.size(1);
+  TORCH_CHECK(KP == ((Ci + 31) / 32) * 32, "Wp must be k-padded to 32");
+  int Co = -1;
+  const float* b = nullptr;
+  if (bias.has_value()) { CK(*bias); b = fp(*bias); Co = bias->numel(); }
+  float* of = nullptr; void* ob = nullptr;
+  if (out_f32.has_value()) {
+    CK(*out_f32);
+    Co = out_f32->size(1);
+    TORCH_CHECK(out_f32->size(0) == R);
+    of = fpm(*out_f32);
+  }
+  if (out_bf16.has_value()) {
+    CKB(*out_bf16);
+    Co = out_bf16->size(1);
+    TORCH_CHECK(out_bf16->size(0) == R);
+    ob = bfp(*out_bf16);
+  }
+  TORCH_CHECK(of || ob, "need at least one output");
+  TORCH_CHECK(Wp.size(0) >= Co, "Wp rows < Co");
+  const void* yp = nullptr;
+  if (lrelu_bwd_of.has_value()) {
+    CKB(*lrelu_bwd_of);
+    TORCH_CHECK(lrelu_bwd_of->size(0) == R && lrelu_bwd_of->size(1) == Co);
+    yp = bfpc(*lrelu_bwd_of);
+  }
+  RUN(fv_gemm_nt_bf16_rs(bfpc(A), bfpc(Wp), b, of, ob, yp, R, Ci, Co, KP,
+                         (float)alpha, act_lrelu, cur_stream()));
+}
+
+void cast_shadows(torch::Tensor s1, torch::Tensor d1, torch::Tensor d1t,
+                  torch::Tensor s2, torch::Tensor d2, torch::Tensor d2t,
+                  torch::Tensor s3, torch::Tensor d3) {
+  CK(s1); CKB(d1); CKB(d1t); CK(s2); CKB(d2); CKB(d2t); CK(s3); CKB(d3);
+  const int M1 = s1.size(0), N1 = s1.size(1);
+  const int M2 = s2.size(0), N2 = s2.size(1);
+  TORCH_CHECK(d1.size(0) == M1 && d1t.size(0) == N1);
+  TORCH_CHECK(d2.size(0) == M2 && d2t.size(0) == N2);
+  RUN(fv_cast_shadows(fp(s1), bfp(d1), bfp(d1t), M1, N1, (int)d1.size(1),
+                      (int)d1t.size(1), fp(s2), bfp(d2), bfp(d2t), M2, N2,
+                      (int)d2.size(1), (int)d2t.size(1), fp(s3), bfp(d3),
+                      s3.numel(), cur_stream()));
 }
 
 void cast3_f32_bf16(torch::Tensor s0, torch::Tensor d0, torch::Tensor s1,
@@ -850,6 +907,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_qk_bwd", &attn_qk_bwd);
   mod.def("loss_fused", &loss_fused);
   mod.def("dh_combine", &dh_combine);
+  mod.def("gemm_nt_bf16_rs", &gemm_nt_bf16_rs, py::arg("A"), py::arg("Wp"),
+          py::arg("bias") = py::none(), py::arg("out_f32") = py::none(),
+          py::arg("out_bf16") = py::none(),
+          py::arg("lrelu_bwd_of") = py::none(), py::arg("alpha") = 1.0,
+          py::arg("act_lrelu") = false);
+  mod.def("cast_shadows", &cast_shadows);
   mod.def("pred_mlp_fwd", &pred_mlp_fwd);
   mod.def("pred_mlp_bwd", &pred_mlp_bwd);
   mod.def("dec_fwd", &dec_fwd);

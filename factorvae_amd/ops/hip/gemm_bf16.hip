@@ -289,224 +289,94 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
   }
 }
 
-// --------------------------------------------------- weight-stationary
-// NT/NN variants for the model's actual extractor shapes (k = Ci <= 192,
-// Co <= 192, R = N*T large). The 64x64x64-tile kernels above run only
-// ~3 k-iterations per workgroup: the pipeline is too short to hide
-// global-load latency and the whole weight gets re-staged per row tile.
-// Here the WEIGHT BLOCK STAYS RESIDENT IN LDS for the kernel's lifetime
-// and each workgroup streams RB row-blocks (full k per block, A
-// double-buffered): a long pipeline with one global stream.
-#define WSKP 192            // max padded k (3H)
-#define WSA (WSKP + 8)      // A row stride (16B-aligned b128 reads)
-#define WSB 72              // NN B row stride (tr16 conflict-free)
+// ------------------------------------------------- register-stationary
+// NT with the whole 64-column weight block RESIDENT IN REGISTERS and no
+// LDS / no barriers at all: every wave independently streams 16-row
+// strips of A straight from global memory into MFMA A-fragments (A is
+// k-contiguous, so a lane's b128 load IS the fragment), two strips in
+// flight. The 64x64-tile kernels above spend their time staging LDS
+// tiles and waiting at barriers for k-pipelines only ~3 deep; here the
+// only synchronization is the wave's own s_waitcnt.
+//
+// Wp is the weight block PRE-PADDED to KP = ceil(k/32)*32 columns with
+// ZEROS (the engine keeps padded bf16 shadows, refreshed after Adam):
+// the zero pad makes the k-tail MFMA contribution exact even though the
+// A-side tail fragment reads 2 elements of the next row (callers
+// guarantee >= 4 bytes of slack after A's last row — engine workspaces
+// are allocated with slack). For NN-shaped products (B k-major) the
+// engine passes the TRANSPOSED padded shadow, so this one kernel covers
+// the extractor's forward and dgrad GEMMs.
+// flags: bit1 lrelu, bit2 bias, bit3 multiply by lrelu'(Y).
+#define RSK 6   // max k32 groups (k <= 192)
 
-// NT: W (Co,Ci) row-major (k-contiguous) — direct b128 fragment reads.
-__global__ __launch_bounds__(256) void gemm_nt_bf16_ws_kernel(
-    const __bf16* __restrict__ A, const __bf16* __restrict__ W,
-    const float* __restrict__ bias, float* __restrict__ out_f32,
-    __bf16* __restrict__ out_bf16, int R, int Ci, int Co, float alpha,
-    int flags, int rb_per_wg) {
-  __shared__ __bf16 Ws[64][WSA];
-  __shared__ __bf16 As[2][64][WSA];
-
-  const int c0 = blockIdx.x * BBC;
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wv = tid >> 6;
-  const int fi = lane & 15;
-  const int fk = lane >> 4;
-
-  const int KP = (Ci + 31) & ~31;   // padded k (multiple of 32)
-  const int KD = KP >> 1;           // dwords per staged row
-  const int nu = (64 * KD) >> 8;    // staging dwords per thread
-
-  // stage the weight block once (guarded, zero-padded)
-  for (int idx = tid; idx < 64 * KD; idx += 256) {
-    const int row = idx / KD;
-    const int cp = (idx % KD) * 2;
-    *(unsigned int*)&Ws[row][cp] = load_dw_guard(W, (long)c0 + row, cp, Co,
-                                                 Ci, Ci);
-  }
-
-  const int rt0 = blockIdx.y * rb_per_wg;
-  const int rtiles = (R + BBR - 1) / BBR;
-  const int rt_end = min(rt0 + rb_per_wg, rtiles);
-  if (rt0 >= rtiles) return;
-
-  unsigned int pa[24];
-  auto stage_regs = [&](int rt) {
-    const long r0 = (long)rt * BBR;
-    if (r0 + BBR <= R && KP == Ci) {
-      for (int u = 0; u < nu; ++u) {
-        const int idx = tid + u * 256;
-        const int row = idx / KD;
-        const int cp = (idx % KD) * 2;
-        pa[u] = *(const unsigned int*)(A + (r0 + row) * Ci + cp);
-      }
-    } else {
-      for (int u = 0; u < nu; ++u) {
-        const int idx = tid + u * 256;
-        const int row = idx / KD;
-        const int cp = (idx % KD) * 2;
-        pa[u] = load_dw_guard(A, r0 + row, cp, R, Ci, Ci);
-      }
-    }
-  };
-  auto regs_to_lds = [&](int buf) {
-    for (int u = 0; u < nu; ++u) {
-      const int idx = tid + u * 256;
-      const int row = idx / KD;
-      const int cp = (idx % KD) * 2;
-      *(unsigned int*)&As[buf][row][cp] = pa[u];
-    }
-  };
-
-  stage_regs(rt0);
-  regs_to_lds(0);
-
-  for (int rt = rt0; rt < rt_end; ++rt) {
-    __syncthreads();
-    if (rt + 1 < rt_end) stage_regs(rt + 1);
-    const int buf = (rt - rt0) & 1;
-    f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
-    for (int k32 = 0; k32 < KP; k32 += 32) {
-      const bf16x8 a = *(const bf16x8*)&As[buf][wv * 16 + fi][k32 + fk * 8];
-#pragma unroll
-      for (int jt = 0; jt < 4; ++jt) {
-        const bf16x8 b = *(const bf16x8*)&Ws[jt * 16 + fi][k32 + fk * 8];
-        acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[jt],
-                                                          0, 0, 0);
-      }
-    }
-    if (rt + 1 < rt_end) {
-      __syncthreads();
-      regs_to_lds(1 - buf);
-    }
-    const long r0 = (long)rt * BBR;
-#pragma unroll
-    for (int jt = 0; jt < 4; ++jt) {
-      const int gc = c0 + jt * 16 + fi;
-      if (gc >= Co) continue;
-#pragma unroll
-      for (int rr = 0; rr < 4; ++rr) {
-        const long gr = r0 + wv * 16 + fk * 4 + rr;
-        if (gr >= R) continue;
-        float v = acc[jt][rr];
-        if (flags & 4) v += bias[gc];
-        v *= alpha;
-        if (flags & 2) v = lrelu_(v);
-        if (out_f32) {
-          float* o = &out_f32[gr * Co + gc];
-          if (flags & 1) v += *o;
-          *o = v;
-        }
-        if (out_bf16) out_bf16[gr * Co + gc] = to_bf16(v);
-      }
-    }
-  }
-}
-
-// NN: B (Ci,Co) row-major (k over rows) — whole B resident in LDS,
-// fragments via ds_read_b64_tr_b16 (same supplier mapping as below).
-__global__ __launch_bounds__(256) void gemm_nn_bf16_ws_kernel(
-    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+__global__ __launch_bounds__(256) void gemm_nt_bf16_rs_kernel(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ Wp,
     const float* __restrict__ bias, float* __restrict__ out_f32,
     __bf16* __restrict__ out_bf16, const __bf16* __restrict__ Y,
-    int R, int Ci, int Co, float alpha, int flags, int rb_per_wg) {
-  __shared__ __bf16 Bs[WSKP][WSB];
-  __shared__ __bf16 As[2][64][WSA];
-
-  const int c0 = blockIdx.x * BBC;
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wv = tid >> 6;
+    int R, int Ci, int Co, int KP, float alpha, int flags, int spw) {
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
   const int fi = lane & 15;
   const int fk = lane >> 4;
-  const int b_kofs = fi >> 2;
-  const int b_nq = (lane & 3) * 4;
+  const int c0 = blockIdx.x * 64;
+  const int nk32 = KP >> 5;
 
-  const int KP = (Ci + 31) & ~31;
-  const int KD = KP >> 1;
-  const int nu = (64 * KD) >> 8;
-
-  // stage all of B for this column block (zero rows beyond Ci)
-  for (int idx = tid; idx < KP * 32; idx += 256) {
-    const int row = idx >> 5;
-    const int cp = (idx & 31) * 2;
-    *(unsigned int*)&Bs[row][cp] = load_dw_guard(B, row, c0 + cp, Ci, Co,
-                                                 Co);
+  // weight fragments: wave-invariant, one-time guarded load
+  bf16x8 wfr[4][RSK];
+#pragma unroll
+  for (int jt = 0; jt < 4; ++jt) {
+    const int gc = c0 + jt * 16 + fi;
+#pragma unroll
+    for (int k32 = 0; k32 < RSK; ++k32) {
+      if (k32 < nk32 && gc < Co) {
+        wfr[jt][k32] = *(const bf16x8*)&Wp[(long)gc * KP + k32 * 32 + fk * 8];
+      } else {
+        bf16x8 z;
+#pragma unroll
+        for (int u = 0; u < 8; ++u) z[u] = (__bf16)0.0f;
+        wfr[jt][k32] = z;
+      }
+    }
   }
 
-  const int rt0 = blockIdx.y * rb_per_wg;
-  const int rtiles = (R + BBR - 1) / BBR;
-  const int rt_end = min(rt0 + rb_per_wg, rtiles);
-  if (rt0 >= rtiles) return;
+  const int strips = (R + 15) >> 4;
+  const int s0 = (blockIdx.y * 4 + wv) * spw;
+  const int s_end = min(s0 + spw, strips);
+  if (s0 >= strips) return;
 
-  unsigned int pa[24];
-  auto stage_regs = [&](int rt) {
-    const long r0 = (long)rt * BBR;
-    if (r0 + BBR <= R && KP == Ci) {
-      for (int u = 0; u < nu; ++u) {
-        const int idx = tid + u * 256;
-        const int row = idx / KD;
-        const int cp = (idx % KD) * 2;
-        pa[u] = *(const unsigned int*)(A + (r0 + row) * Ci + cp);
-      }
-    } else {
-      for (int u = 0; u < nu; ++u) {
-        const int idx = tid + u * 256;
-        const int row = idx / KD;
-        const int cp = (idx % KD) * 2;
-        pa[u] = load_dw_guard(A, r0 + row, cp, R, Ci, Ci);
-      }
-    }
-  };
-  auto regs_to_lds = [&](int buf) {
-    for (int u = 0; u < nu; ++u) {
-      const int idx = tid + u * 256;
-      const int row = idx / KD;
-      const int cp = (idx % KD) * 2;
-      *(unsigned int*)&As[buf][row][cp] = pa[u];
-    }
-  };
-
-  stage_regs(rt0);
-  regs_to_lds(0);
-
-  for (int rt = rt0; rt < rt_end; ++rt) {
-    __syncthreads();
-    if (rt + 1 < rt_end) stage_regs(rt + 1);
-    const int buf = (rt - rt0) & 1;
-    f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
-    for (int k32 = 0; k32 < KP; k32 += 32) {
-      const bf16x8 a = *(const bf16x8*)&As[buf][wv * 16 + fi][k32 + fk * 8];
-      const int kb = k32 + fk * 8 + b_kofs;
+  auto loadA = [&](bf16x8 (&fr)[RSK], int s) {
+    const long row = (long)s * 16 + fi;
+    const bool live = row < R;
 #pragma unroll
-      for (int jt = 0; jt < 4; ++jt) {
-        s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (lds_v4p)&Bs[kb][jt * 16 + b_nq]);
-        s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (lds_v4p)&Bs[kb + 4][jt * 16 + b_nq]);
-        bf16x8 b;
-        *(bf16x4*)&b = *(bf16x4*)&b0;
-        *(((bf16x4*)&b) + 1) = *(bf16x4*)&b1;
-        acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[jt],
-                                                          0, 0, 0);
+    for (int k32 = 0; k32 < RSK; ++k32) {
+      if (k32 < nk32 && live) {
+        fr[k32] = *(const bf16x8*)&A[row * Ci + k32 * 32 + fk * 8];
+      } else {
+        bf16x8 z;
+#pragma unroll
+        for (int u = 0; u < 8; ++u) z[u] = (__bf16)0.0f;
+        fr[k32] = z;
       }
     }
-    if (rt + 1 < rt_end) {
-      __syncthreads();
-      regs_to_lds(1 - buf);
+  };
+  auto compute_store = [&](bf16x8 (&fr)[RSK], int s) {
+    f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+#pragma unroll
+    for (int k32 = 0; k32 < RSK; ++k32) {
+      if (k32 < nk32) {
+#pragma unroll
+        for (int jt = 0; jt < 4; ++jt)
+          acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              fr[k32], wfr[jt][k32], acc[jt], 0, 0, 0);
+      }
     }
-    const long r0 = (long)rt * BBR;
 #pragma unroll
     for (int jt = 0; jt < 4; ++jt) {
       const int gc = c0 + jt * 16 + fi;
       if (gc >= Co) continue;
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
-        const long gr = r0 + wv * 16 + fk * 4 + rr;
+        const long gr = (long)s * 16 + fk * 4 + rr;
         if (gr >= R) continue;
         float v = acc[jt][rr];
         if (flags & 4) v += bias[gc];
@@ -516,15 +386,54 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_ws_kernel(
           const float y = (float)Y[gr * Co + gc];
           v *= (y > 0.0f ? 1.0f : 0.01f);
         }
-        if (out_f32) {
-          float* o = &out_f32[gr * Co + gc];
-          if (flags & 1) v += *o;
-          *o = v;
-        }
+        if (out_f32) out_f32[gr * Co + gc] = v;
         if (out_bf16) out_bf16[gr * Co + gc] = to_bf16(v);
       }
     }
+  };
+
+  // two strips in flight (register double buffer, unroll-by-2)
+  bf16x8 fr0[RSK], fr1[RSK];
+  int s = s0;
+  loadA(fr0, s);
+  while (true) {
+    if (s + 1 < s_end) loadA(fr1, s + 1);
+    compute_store(fr0, s);
+    if (++s >= s_end) break;
+    if (s + 1 < s_end) loadA(fr0, s + 1);
+    compute_store(fr1, s);
+    if (++s >= s_end) break;
   }
+}
+
+// padded + transposed-padded bf16 weight shadow refresh, ONE launch for
+// both extractor weights plus the flat Whh copy (runs after every Adam
+// step; pads were zeroed at allocation and are never written here)
+__global__ __launch_bounds__(256) void cast_shadows_kernel(
+    const float* __restrict__ s1, __bf16* __restrict__ d1,
+    __bf16* __restrict__ d1t, int M1, int N1, int KPn1, int KPm1,
+    const float* __restrict__ s2, __bf16* __restrict__ d2,
+    __bf16* __restrict__ d2t, int M2, int N2, int KPn2, int KPm2,
+    const float* __restrict__ s3, __bf16* __restrict__ d3, long n3) {
+  long i = (long)blockIdx.x * 256 + threadIdx.x;
+  const long e1 = (long)M1 * N1, e2 = (long)M2 * N2;
+  if (i < e1) {
+    const int m = (int)(i / N1), n = (int)(i % N1);
+    const __bf16 v = (__bf16)s1[i];
+    d1[(long)m * KPn1 + n] = v;
+    d1t[(long)n * KPm1 + m] = v;
+    return;
+  }
+  i -= e1;
+  if (i < e2) {
+    const int m = (int)(i / N2), n = (int)(i % N2);
+    const __bf16 v = (__bf16)s2[i];
+    d2[(long)m * KPn2 + n] = v;
+    d2t[(long)n * KPm2 + m] = v;
+    return;
+  }
+  i -= e2;
+  if (i < n3) d3[i] = (__bf16)s3[i];
 }
 
 // ---------------------------------------------------------------- TN
@@ -793,21 +702,7 @@ hipError_t fv_gemm_nt_bf16(const void* A, const void* W, const float* bias,
                            int Co, float alpha, int accumulate, int act_lrelu,
                            hipStream_t stream) {
   int flags = (accumulate ? 1 : 0) | (act_lrelu ? 2 : 0) | (bias ? 4 : 0);
-  const int rtiles = (R + BBR - 1) / BBR;
-  const int cblocks = (Co + BBC - 1) / BBC;
-  if (Ci <= WSKP) {
-    // weight-stationary streaming path (all model shapes): ~2 WGs/CU,
-    // long row-block pipeline per WG
-    int rb = (rtiles * cblocks + 511) / 512;
-    if (rb < 1) rb = 1;
-    dim3 grid(cblocks, (rtiles + rb - 1) / rb);
-    hipLaunchKernelGGL(gemm_nt_bf16_ws_kernel, grid, dim3(256), 0, stream,
-                       (const __bf16*)A, (const __bf16*)W, bias, out_f32,
-                       (__bf16*)out_bf16, R, Ci, Co, alpha, flags, rb);
-    HIP_CHECK_LAST();
-    return hipSuccess;
-  }
-  dim3 grid(cblocks, rtiles);
+  dim3 grid((Co + BBC - 1) / BBC, (R + BBR - 1) / BBR);
   hipLaunchKernelGGL(gemm_nt_bf16_kernel, grid, dim3(256), 0, stream,
                      (const __bf16*)A, (const __bf16*)W, bias, out_f32,
                      (__bf16*)out_bf16, R, Ci, Co, alpha, flags);
@@ -821,20 +716,7 @@ hipError_t fv_gemm_nn_bf16(const void* A, const void* B, const float* bias,
                            int act_lrelu, hipStream_t stream) {
   int flags = (accumulate ? 1 : 0) | (act_lrelu ? 2 : 0) | (bias ? 4 : 0) |
               (Y ? 8 : 0);
-  const int rtiles = (R + BBR - 1) / BBR;
-  const int cblocks = (Co + BBC - 1) / BBC;
-  if (Ci <= WSKP) {
-    int rb = (rtiles * cblocks + 511) / 512;
-    if (rb < 1) rb = 1;
-    dim3 grid(cblocks, (rtiles + rb - 1) / rb);
-    hipLaunchKernelGGL(gemm_nn_bf16_ws_kernel, grid, dim3(256), 0, stream,
-                       (const __bf16*)A, (const __bf16*)B, bias, out_f32,
-                       (__bf16*)out_bf16, (const __bf16*)Y, R, Ci, Co, alpha,
-                       flags, rb);
-    HIP_CHECK_LAST();
-    return hipSuccess;
-  }
-  dim3 grid(cblocks, rtiles);
+  dim3 grid((Co + BBC - 1) / BBC, (R + BBR - 1) / BBR);
   hipLaunchKernelGGL(gemm_nn_bf16_kernel, grid, dim3(256), 0, stream,
                      (const __bf16*)A, (const __bf16*)B, bias, out_f32,
                      (__bf16*)out_bf16, (const __bf16*)Y, R, Ci, Co, alpha,
@@ -872,6 +754,43 @@ hipError_t fv_gemm_tn_bf16(const void* A, const void* B, float* out,
                        accumulate);
     HIP_CHECK_LAST();
   }
+  return hipSuccess;
+}
+
+hipError_t fv_gemm_nt_bf16_rs(const void* A, const void* Wp,
+                              const float* bias, float* out_f32,
+                              void* out_bf16, const void* Y, int R, int Ci,
+                              int Co, int KP, float alpha, int act_lrelu,
+                              hipStream_t stream) {
+  if (KP > 32 * RSK || KP < Ci || (KP & 31)) return hipErrorInvalidValue;
+  int flags = (act_lrelu ? 2 : 0) | (bias ? 4 : 0) | (Y ? 8 : 0);
+  const int cblocks = (Co + 63) / 64;
+  const int strips = (R + 15) / 16;
+  // strips per wave: target ~4k waves so every SIMD holds ~2 chunks
+  int spw = (strips * cblocks) / 4096;
+  if (spw < 1) spw = 1;
+  const int yblocks = (strips + spw * 4 - 1) / (spw * 4);
+  dim3 grid(cblocks, yblocks);
+  hipLaunchKernelGGL(gemm_nt_bf16_rs_kernel, grid, dim3(256), 0, stream,
+                     (const __bf16*)A, (const __bf16*)Wp, bias, out_f32,
+                     (__bf16*)out_bf16, (const __bf16*)Y, R, Ci, Co, KP,
+                     alpha, flags, spw);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_cast_shadows(const float* s1, void* d1, void* d1t, int M1,
+                           int N1, int KPn1, int KPm1, const float* s2,
+                           void* d2, void* d2t, int M2, int N2, int KPn2,
+                           int KPm2, const float* s3, void* d3, long n3,
+                           hipStream_t stream) {
+  const long total = (long)M1 * N1 + (long)M2 * N2 + n3;
+  dim3 grid((unsigned)((total + 255) / 256));
+  hipLaunchKernelGGL(cast_shadows_kernel, grid, dim3(256), 0, stream, s1,
+                     (__bf16*)d1, (__bf16*)d1t, M1, N1, KPn1, KPm1, s2,
+                     (__bf16*)d2, (__bf16*)d2t, M2, N2, KPn2, KPm2, s3,
+                     (__bf16*)d3, n3);
+  HIP_CHECK_LAST();
   return hipSuccess;
 }
 

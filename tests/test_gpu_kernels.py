@@ -768,3 +768,59 @@ def test_full_step_bit_determinism():
     p2 = run()
     diff = (p1 != p2).sum().item()
     assert diff == 0, f"{diff}/{p1.numel()} parameter words differ between identically-seeded runs"
+
+
+@pytest.mark.parametrize("R,Ci,Co", [(300, 158, 192), (37, 33, 20),
+                                     (6000, 158, 158), (210000, 192, 158)])
+def test_gemm_nt_bf16_rs(R, Ci, Co):
+    """Register-stationary NT: pre-padded weight, LDS-free streaming."""
+    KP = (Ci + 31) & ~31
+    A, W = bt(R, Ci, seed=21), bt(Co, Ci, seed=22)
+    # A needs >=4B of tail slack (engine workspaces provide it)
+    Abuf = torch.zeros(R * Ci + 8, device=DEV, dtype=torch.bfloat16)
+    Abuf[:R * Ci] = A.reshape(-1)
+    As = Abuf[:R * Ci].view(R, Ci)
+    Wp = torch.zeros(Co, KP, device=DEV, dtype=torch.bfloat16)
+    Wp[:, :Ci] = W
+    b = t(Co, seed=23)
+    out = torch.empty(R, Co, device=DEV)
+    outb = torch.empty(R, Co, device=DEV, dtype=torch.bfloat16)
+    ext.gemm_nt_bf16_rs(As, Wp, b, out, outb, None, 1.0, False)
+    torch.cuda.synchronize()
+    ref = A.float() @ W.float().t() + b
+    tol = 3e-2 * math.sqrt(Ci / 64)
+    assert_close(out, ref, atol=tol, rtol=tol, what="gemm_nt_bf16_rs f32")
+    assert_close(outb.float(), ref, atol=4e-1, rtol=2e-2,
+                 what="gemm_nt_bf16_rs bf16")
+    # lrelu epilogue + Y-gated (lrelu-backward fusion) path
+    Y = bt(R, Co, seed=24)
+    out2 = torch.empty(R, Co, device=DEV)
+    ext.gemm_nt_bf16_rs(As, Wp, b, out2, None, Y, 0.5, True)
+    torch.cuda.synchronize()
+    ref2 = F.leaky_relu(0.5 * ref, 0.01)
+    ref2 = ref2 * torch.where(Y.float() > 0, 1.0, 0.01)
+    assert_close(out2, ref2, atol=tol, rtol=tol, what="rs lrelu+Y")
+
+
+def test_cast_shadows_padded_transposed():
+    C, C3, H = 158, 192, 64
+    KPc = 160
+    W1x = t(C, C, seed=31)
+    Wih = t(C3, C, seed=32)
+    Whh = t(C3, H, seed=33)
+    d1 = torch.zeros(C, KPc, device=DEV, dtype=torch.bfloat16)
+    d1t = torch.zeros(C, KPc, device=DEV, dtype=torch.bfloat16)
+    d2 = torch.zeros(C3, KPc, device=DEV, dtype=torch.bfloat16)
+    d2t = torch.zeros(C, C3, device=DEV, dtype=torch.bfloat16)
+    d3 = torch.zeros(C3, H, device=DEV, dtype=torch.bfloat16)
+    ext.cast_shadows(W1x, d1, d1t, Wih, d2, d2t, Whh, d3)
+    torch.cuda.synchronize()
+    assert_close(d1[:, :C].float(), W1x.to(torch.bfloat16).float(),
+                 atol=0, rtol=0, what="d1")
+    assert d1[:, C:].abs().sum().item() == 0
+    assert_close(d1t[:, :C].float(), W1x.t().to(torch.bfloat16).float()
+                 .contiguous(), atol=0, rtol=0, what="d1t")
+    assert_close(d2t.float(), Wih.t().to(torch.bfloat16).float()
+                 .contiguous(), atol=0, rtol=0, what="d2t")
+    assert_close(d3.float(), Whh.to(torch.bfloat16).float(), atol=0,
+                 rtol=0, what="d3")
