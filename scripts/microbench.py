@@ -158,6 +158,17 @@ def gemm_bench():
         t = timeit(lambda: ext.gemm_nn_bf16(A, B, None, None, out, 1.0, False, False), 20)
         fl = 2.0 * R * Ci * Co
         print(f"gemm_nn_bf16 {R}x{Ci}x{Co}: {t*1e6:.1f} us  {fl/t/1e12:.1f} TF/s")
+    for R, Ci, Co in [(6000, 158, 192), (210000, 158, 192), (210000, 192, 158)]:
+        KP = (Ci + 31) & ~31
+        Abuf = torch.zeros(R * Ci + 8, device=dev, dtype=torch.bfloat16)
+        Abuf[:R * Ci] = torch.randn(R * Ci, device=dev).bfloat16()
+        A = Abuf[:R * Ci].view(R, Ci)
+        Wp = torch.zeros(Co, KP, device=dev, dtype=torch.bfloat16)
+        Wp[:, :Ci] = torch.randn(Co, Ci, device=dev).bfloat16()
+        out = torch.empty(R, Co, device=dev, dtype=torch.bfloat16)
+        t = timeit(lambda: ext.gemm_nt_bf16_rs(A, Wp, None, None, out, None, 1.0, False), 20)
+        fl = 2.0 * R * Ci * Co
+        print(f"gemm_nt_bf16_rs {R}x{Ci}x{Co}: {t*1e6:.1f} us  {fl/t/1e12:.1f} TF/s")
     for R, M, N, chunks in [(6000, 192, 158, 8), (210000, 192, 158, 8)]:
         A = torch.randn(R, M, device=dev).bfloat16()
         B = torch.randn(R, N, device=dev).bfloat16()
