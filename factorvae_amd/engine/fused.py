@@ -133,11 +133,17 @@ class FusedTrainer:
             self.whh_bf = torch.empty(C3, self.H, dtype=torch.bfloat16,
                                       device=self.device)
             if self.fp8:
-                ldp = (self.C + 3) & ~3
+                # row stride padded to 128 for the MX-scaled K=128 MFMA
+                # path (the only fp8 form at the ~5 PF/s rate); pads are
+                # zeroed here and never written (casts write Ci cols)
+                ldp = ((self.C + 127) & ~127)
+                self._fp8_rs = ldp <= 256
+                if not self._fp8_rs:  # fallback: K=32 fp8 MFMA path
+                    ldp = (self.C + 3) & ~3
                 f8 = torch.float8_e4m3fn
-                self.w1x_f8 = torch.empty(self.C, ldp, dtype=f8,
+                self.w1x_f8 = torch.zeros(self.C, ldp, dtype=f8,
                                           device=self.device)
-                self.wih_f8 = torch.empty(C3, ldp, dtype=f8,
+                self.wih_f8 = torch.zeros(C3, ldp, dtype=f8,
                                           device=self.device)
                 self.s_w1x = torch.ones(1, device=self.device)
                 self.is_w1x = torch.ones(1, device=self.device)
@@ -384,7 +390,7 @@ class FusedTrainer:
             w["h_prev_bf"] = fb(R, H)
             w["dgh_bf"] = fb(R, 3 * H)
             if self.fp8:
-                ldp = (C + 3) & ~3
+                ldp = self.w1x_f8.size(1)
                 f8t = lambda *shape: torch.zeros(
                     *shape, device=d, dtype=torch.float8_e4m3fn)
                 w["xln_f8"] = f8t(R, ldp)
@@ -455,12 +461,21 @@ class FusedTrainer:
             # activation copies the bf16 backward consumes
             ext.ln_fwd(x2d, p("ln_g"), p("ln_b"), None, w["mean"],
                        w["rstd"], 1e-5, w["xln_bf"], w["xln_f8"])
-            ext.gemm_nt_fp8(w["xln_f8"], self.w1x_f8, p("b1x"), self.is_w1x,
-                            None, w["xp_bf"], w["xp_f8"], R, self.C, self.C,
-                            1.0, True)
-            ext.gemm_nt_fp8(w["xp_f8"], self.wih_f8, p("bih"), self.is_wih,
-                            w["gi"].view(R, 3 * H), None, None, R, self.C,
-                            3 * H, 1.0, False)
+            if self._fp8_rs:
+                # MX-scaled K=128 MFMA (double the K=32 form's rate)
+                ext.gemm_nt_fp8_rs(w["xln_f8"], self.w1x_f8, p("b1x"),
+                                   self.is_w1x, None, w["xp_bf"],
+                                   w["xp_f8"], R, self.C, self.C, 1.0, True)
+                ext.gemm_nt_fp8_rs(w["xp_f8"], self.wih_f8, p("bih"),
+                                   self.is_wih, w["gi"].view(R, 3 * H),
+                                   None, None, R, self.C, 3 * H, 1.0, False)
+            else:
+                ext.gemm_nt_fp8(w["xln_f8"], self.w1x_f8, p("b1x"),
+                                self.is_w1x, None, w["xp_bf"], w["xp_f8"],
+                                R, self.C, self.C, 1.0, True)
+                ext.gemm_nt_fp8(w["xp_f8"], self.wih_f8, p("bih"),
+                                self.is_wih, w["gi"].view(R, 3 * H), None,
+                                None, R, self.C, 3 * H, 1.0, False)
         elif self.bf16:
             ext.ln_fwd(x2d, p("ln_g"), p("ln_b"), None, w["mean"],
                        w["rstd"], 1e-5, w["xln_bf"])

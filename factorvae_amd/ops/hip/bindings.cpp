@@ -42,6 +42,9 @@ hipError_t fv_gemm_nt_fp8(const void*, const void*, const float*,
                           const float*, float*, void*, void*, int, int, int,
                           int, int, int, float, int, int, hipStream_t);
 hipError_t fv_absmax_scale(const float*, long, float*, float*, hipStream_t);
+hipError_t fv_gemm_nt_fp8_rs(const void*, const void*, const float*,
+                             const float*, float*, void*, void*, int, int,
+                             int, int, int, float, int, int, hipStream_t);
 hipError_t fv_cast_f32_fp8_scaled(const float*, void*, const float*, long,
                                   int, int, hipStream_t);
 hipError_t fv_ln_bwd_params(const float*, const float*, const float*,
@@ -463,6 +466,36 @@ void gemm_nt_fp8(torch::Tensor A, torch::Tensor W,
   RUN(fv_gemm_nt_fp8(A.data_ptr(), W.data_ptr(), b, isw, of, ob, o8, ldo,
                      (int)R, (int)Ci, (int)Co, lda, ldw, (float)alpha,
                      act_lrelu, b != nullptr, cur_stream()));
+}
+
+void gemm_nt_fp8_rs(torch::Tensor A, torch::Tensor Wp,
+                    c10::optional<torch::Tensor> bias,
+                    c10::optional<torch::Tensor> inv_sw,
+                    c10::optional<torch::Tensor> out_f32,
+                    c10::optional<torch::Tensor> out_bf16,
+                    c10::optional<torch::Tensor> out_fp8,
+                    long R, long Ci, long Co, double alpha, bool act_lrelu) {
+  CK8(A); CK8(Wp);
+  const int KP = A.size(1);
+  TORCH_CHECK(Wp.size(1) == KP && (KP & 127) == 0 && KP >= Ci,
+              "A/Wp must share a 128-padded k stride");
+  TORCH_CHECK(A.size(0) >= R && Wp.size(0) >= Co);
+  const float* b = nullptr;
+  if (bias.has_value()) { CK(*bias); b = fp(*bias); }
+  const float* isw = nullptr;
+  if (inv_sw.has_value()) { CK(*inv_sw); isw = fp(*inv_sw); }
+  float* of = nullptr; void* ob = nullptr; void* o8 = nullptr; int ldo = 0;
+  if (out_f32.has_value()) { CK(*out_f32); of = fpm(*out_f32); }
+  if (out_bf16.has_value()) { CKB(*out_bf16); ob = bfp(*out_bf16); }
+  if (out_fp8.has_value()) {
+    CK8(*out_fp8);
+    o8 = out_fp8->data_ptr();
+    ldo = out_fp8->size(-1);
+  }
+  TORCH_CHECK(of || ob || o8, "need at least one output");
+  RUN(fv_gemm_nt_fp8_rs(A.data_ptr(), Wp.data_ptr(), b, isw, of, ob, o8,
+                        ldo, (int)R, (int)Ci, (int)Co, KP, (float)alpha,
+                        act_lrelu, b != nullptr, cur_stream()));
 }
 
 void absmax_scale(torch::Tensor src, torch::Tensor scale,
@@ -913,6 +946,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           py::arg("lrelu_bwd_of") = py::none(), py::arg("alpha") = 1.0,
           py::arg("act_lrelu") = false);
   mod.def("cast_shadows", &cast_shadows);
+  mod.def("gemm_nt_fp8_rs", &gemm_nt_fp8_rs, py::arg("A"), py::arg("Wp"),
+          py::arg("bias") = py::none(), py::arg("inv_sw") = py::none(),
+          py::arg("out_f32") = py::none(), py::arg("out_bf16") = py::none(),
+          py::arg("out_fp8") = py::none(), py::arg("R"), py::arg("Ci"),
+          py::arg("Co"), py::arg("alpha") = 1.0,
+          py::arg("act_lrelu") = false);
   mod.def("pred_mlp_fwd", &pred_mlp_fwd);
   mod.def("pred_mlp_bwd", &pred_mlp_bwd);
   mod.def("dec_fwd", &dec_fwd);

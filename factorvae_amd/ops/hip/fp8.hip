@@ -137,6 +137,109 @@ __global__ __launch_bounds__(256) void gemm_nt_fp8_kernel(
   }
 }
 
+// ------------------------------------------------- register-stationary
+// MX-scaled NT (k <= 256): v_mfma_scale_f32_16x16x128_f8f6f4 at K=128 —
+// the ONLY fp8 MFMA form that runs at the ~5 PF/s dense fp8 rate (the
+// 16x16x32 form above runs at the bf16 rate; guide §4). Block scales
+// are passed as unit (E8M0 bias 127): dequant stays the per-tensor fp32
+// software scale in the epilogue, so numerics match the K=32 path
+// exactly. Same wave-independent streaming structure as
+// gemm_nt_bf16_rs: W block resident in registers, A rows ARE the
+// fragments (32 k-contiguous fp8 bytes per lane), no LDS, no barriers.
+// A and Wp must be row-padded to KP (multiple of 128) with ZEROS.
+typedef __attribute__((ext_vector_type(8))) int i32x8;
+#define FRSK 2  // max k128 groups (k <= 256)
+
+__global__ __launch_bounds__(256) void gemm_nt_fp8_rs_kernel(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ Wp,
+    const float* __restrict__ bias, const float* __restrict__ inv_sw,
+    float* __restrict__ out_f32, __bf16* __restrict__ out_bf16,
+    unsigned char* __restrict__ out_fp8, int ldo, int R, int Ci, int Co,
+    int KP, float alpha, int flags, int spw) {
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  const int fi = lane & 15;
+  const int fk = lane >> 4;
+  const int c0 = blockIdx.x * 64;
+  const int nk = KP >> 7;  // k128 groups
+
+  i32x8 wfr[4][FRSK];
+#pragma unroll
+  for (int jt = 0; jt < 4; ++jt) {
+    const int gc = c0 + jt * 16 + fi;
+#pragma unroll
+    for (int k = 0; k < FRSK; ++k) {
+      if (k < nk && gc < Co) {
+        wfr[jt][k] = *(const i32x8*)&Wp[(long)gc * KP + k * 128 + fk * 32];
+      } else {
+        i32x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+        wfr[jt][k] = z;
+      }
+    }
+  }
+
+  const int strips = (R + 15) >> 4;
+  const int s0 = (blockIdx.y * 4 + wv) * spw;
+  const int s_end = min(s0 + spw, strips);
+  if (s0 >= strips) return;
+
+  auto loadA = [&](i32x8 (&fr)[FRSK], int s) {
+    const long row = (long)s * 16 + fi;
+    const bool live = row < R;
+#pragma unroll
+    for (int k = 0; k < FRSK; ++k) {
+      if (k < nk && live) {
+        fr[k] = *(const i32x8*)&A[row * KP + k * 128 + fk * 32];
+      } else {
+        i32x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+        fr[k] = z;
+      }
+    }
+  };
+  const float sw = inv_sw ? *inv_sw : 1.0f;
+  auto compute_store = [&](i32x8 (&fr)[FRSK], int s) {
+    f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+#pragma unroll
+    for (int k = 0; k < FRSK; ++k) {
+      if (k < nk) {
+#pragma unroll
+        for (int jt = 0; jt < 4; ++jt)
+          acc[jt] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              fr[k], wfr[jt][k], acc[jt], 0, 0, 0, 127, 0, 127);
+      }
+    }
+#pragma unroll
+    for (int jt = 0; jt < 4; ++jt) {
+      const int gc = c0 + jt * 16 + fi;
+      if (gc >= Co) continue;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const long gr = (long)s * 16 + fk * 4 + rr;
+        if (gr >= R) continue;
+        float v = acc[jt][rr] * sw;
+        if (flags & 4) v += bias[gc];
+        v *= alpha;
+        if (flags & 2) v = lrelu_(v);
+        if (out_f32) out_f32[gr * Co + gc] = v;
+        if (out_bf16) out_bf16[gr * Co + gc] = (__bf16)v;
+        if (out_fp8) out_fp8[gr * ldo + gc] = f32_to_e4m3(v);
+      }
+    }
+  };
+
+  i32x8 fr0[FRSK], fr1[FRSK];
+  int s = s0;
+  loadA(fr0, s);
+  while (true) {
+    if (s + 1 < s_end) loadA(fr1, s + 1);
+    compute_store(fr0, s);
+    if (++s >= s_end) break;
+    if (s + 1 < s_end) loadA(fr0, s + 1);
+    compute_store(fr1, s);
+    if (++s >= s_end) break;
+  }
+}
+
 // scale[0] = 448 / max(absmax(src), 1e-8): per-tensor e4m3 full-scale.
 // Single workgroup (params are <=128k elements); deterministic.
 __global__ __launch_bounds__(1024) void absmax_scale_kernel(
@@ -186,6 +289,29 @@ hipError_t fv_gemm_nt_fp8(const void* A, const void* W, const float* bias,
                      inv_sw, out_f32, (__bf16*)out_bf16,
                      (unsigned char*)out_fp8, ldo, R, Ci, Co, lda, ldw,
                      alpha, flags);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_gemm_nt_fp8_rs(const void* A, const void* Wp,
+                             const float* bias, const float* inv_sw,
+                             float* out_f32, void* out_bf16, void* out_fp8,
+                             int ldo, int R, int Ci, int Co, int KP,
+                             float alpha, int act_lrelu, int has_bias,
+                             hipStream_t stream) {
+  if (KP > 128 * FRSK || KP < Ci || (KP & 127)) return hipErrorInvalidValue;
+  int flags = (act_lrelu ? 2 : 0) | (has_bias ? 4 : 0);
+  const int cblocks = (Co + 63) / 64;
+  const int strips = (R + 15) / 16;
+  int spw = (strips * cblocks) / 4096;
+  if (spw < 1) spw = 1;
+  const int yblocks = (strips + spw * 4 - 1) / (spw * 4);
+  dim3 grid(cblocks, yblocks);
+  hipLaunchKernelGGL(gemm_nt_fp8_rs_kernel, grid, dim3(256), 0, stream,
+                     (const unsigned char*)A, (const unsigned char*)Wp, bias,
+                     inv_sw, out_f32, (__bf16*)out_bf16,
+                     (unsigned char*)out_fp8, ldo, R, Ci, Co, KP, alpha,
+                     flags, spw);
   HIP_CHECK_LAST();
   return hipSuccess;
 }

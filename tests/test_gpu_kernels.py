@@ -824,3 +824,32 @@ def test_cast_shadows_padded_transposed():
                  .contiguous(), atol=0, rtol=0, what="d2t")
     assert_close(d3.float(), Whh.to(torch.bfloat16).float(), atol=0,
                  rtol=0, what="d3")
+
+
+@pytest.mark.parametrize("R,Ci,Co", [(300, 158, 192), (6000, 158, 158),
+                                     (210000, 192, 158)])
+def test_gemm_nt_fp8_rs_mx(R, Ci, Co):
+    """MX-scaled K=128 fp8 MFMA path vs fp32 reference (unit block
+    scales; per-tensor software dequant in the epilogue)."""
+    KP = (Ci + 127) & ~127
+    g = torch.Generator(device="cpu").manual_seed(41)
+    Af = (torch.randn(R, Ci, generator=g)).to(DEV)
+    Wf = (torch.randn(Co, Ci, generator=g)).to(DEV)
+    A8 = torch.zeros(R, KP, device=DEV, dtype=torch.float8_e4m3fn)
+    A8[:, :Ci] = Af.to(torch.float8_e4m3fn)
+    scale = torch.zeros(1, device=DEV)
+    inv = torch.zeros(1, device=DEV)
+    ext.absmax_scale(Wf, scale, inv)
+    W8 = torch.zeros(Co, KP, device=DEV, dtype=torch.float8_e4m3fn)
+    ext.cast_f32_fp8_scaled(Wf, W8[:, :Ci].contiguous(), scale)
+    # write into padded buffer via the strided cast (ldp from dst)
+    W82 = torch.zeros(Co, KP, device=DEV, dtype=torch.float8_e4m3fn)
+    ext.cast_f32_fp8_scaled(Wf, W82, scale)
+    out = torch.empty(R, Co, device=DEV)
+    ext.gemm_nt_fp8_rs(A8, W82, None, inv, out, None, None, R, Ci, Co,
+                       1.0, False)
+    torch.cuda.synchronize()
+    ref = A8[:, :Ci].float() @ (W82[:, :Ci].float() * inv) .t()
+    # e4m3 inputs: compare against the e4m3-quantized operands exactly
+    assert_close(out, ref, atol=2e-2 * math.sqrt(Ci), rtol=2e-2,
+                 what="gemm_nt_fp8_rs")
