@@ -149,6 +149,24 @@ class FusedTrainer:
                 self.is_w1x = torch.ones(1, device=self.device)
                 self.s_wih = torch.ones(1, device=self.device)
                 self.is_wih = torch.ones(1, device=self.device)
+                if self._fp8_rs:
+                    # fp8 dgrad path: transposed padded weight shadows +
+                    # delayed-scaling state for the two dgrad operands
+                    # (amax collected this step -> scale next step)
+                    KPt3 = (C3 + 127) & ~127
+                    KPtC = (self.C + 127) & ~127
+                    self.wihT_f8 = torch.zeros(self.C, KPt3, dtype=f8,
+                                               device=self.device)
+                    self.w1xT_f8 = torch.zeros(self.C, KPtC, dtype=f8,
+                                               device=self.device)
+                    one = lambda v: torch.full((1,), float(v),
+                                               device=self.device)
+                    self.amax_dgi = one(1.0)
+                    self.s_dgi = one(448.0)
+                    self.is_dgi = one(1.0 / 448.0)
+                    self.amax_dzx = one(1.0)
+                    self.s_dzx = one(448.0)
+                    self.is_dzx = one(1.0 / 448.0)
             self._refresh_bf16_shadows()
         self.grads = torch.zeros_like(self.params.flat)
         self.adam_m = torch.zeros_like(self.params.flat)
@@ -395,6 +413,9 @@ class FusedTrainer:
                     *shape, device=d, dtype=torch.float8_e4m3fn)
                 w["xln_f8"] = f8t(R, ldp)
                 w["xp_f8"] = f8t(R, ldp)
+                if self._fp8_rs:
+                    w["dgi_f8"] = f8t(R, self.wihT_f8.size(1))
+                    w["dzx_f8"] = f8t(R, self.w1xT_f8.size(1))
         self._ws_cache[(N, T)] = w
         self.ws = w
         self._ws_n = N
@@ -693,16 +714,38 @@ class FusedTrainer:
                 ext.gemm_tn_bf16(w["dgi_bf"].view(R, 3 * H), w["xp_bf"],
                                  g("Wih"), w["tn_part2"], chunks, True,
                                  g("bih"), w["tn_partb2"])
-            ext.gemm_nt_bf16_rs(w["dgi_bf"].view(R, 3 * H), self.wihT_p,
-                                None, None, w["dzx_bf"], w["xp_bf"], 1.0,
-                                False)
+            if self.fp8 and self._fp8_rs:
+                # fp8 dgrads on the MX K=128 path (delayed per-tensor
+                # scaling; the bf16 dzx copy still feeds the W1x wgrad)
+                ext.scale_from_amax2(self.amax_dgi, self.s_dgi,
+                                     self.is_dgi, self.amax_dzx,
+                                     self.s_dzx, self.is_dzx)
+                ext.cast_f32_fp8_damax(w["dgi"].view(R, 3 * H), w["dgi_f8"],
+                                       self.s_dgi, self.amax_dgi)
+                ext.gemm_nt_fp8_rs(w["dgi_f8"], self.wihT_f8, None,
+                                   self.is_wih, None, w["dzx_bf"],
+                                   w["dzx_f8"], R, 3 * H, self.C, 1.0,
+                                   False, inv_sa=self.is_dgi,
+                                   lrelu_bwd_of=w["xp_bf"],
+                                   s_out=self.s_dzx,
+                                   amax_out=self.amax_dzx)
+            else:
+                ext.gemm_nt_bf16_rs(w["dgi_bf"].view(R, 3 * H), self.wihT_p,
+                                    None, None, w["dzx_bf"], w["xp_bf"],
+                                    1.0, False)
             fork(1)
             with _on_side(self, 1):
                 ext.gemm_tn_bf16(w["dzx_bf"], w["xln_bf"], g("W1x"),
                                  w["tn_part3"], chunks, True, g("b1x"),
                                  w["tn_partb3"])
-            ext.gemm_nt_bf16_rs(w["dzx_bf"], self.w1xT_p, None, w["dxln"],
-                                None, None, 1.0, False)
+            if self.fp8 and self._fp8_rs:
+                ext.gemm_nt_fp8_rs(w["dzx_f8"], self.w1xT_f8, None,
+                                   self.is_w1x, w["dxln"], None, None, R,
+                                   self.C, self.C, 1.0, False,
+                                   inv_sa=self.is_dzx)
+            else:
+                ext.gemm_nt_bf16_rs(w["dzx_bf"], self.w1xT_p, None,
+                                    w["dxln"], None, None, 1.0, False)
         else:
             fork(1)
             with _on_side(self, 1):
@@ -746,6 +789,11 @@ class FusedTrainer:
             self.ext.absmax_scale(self.p("Wih"), self.s_wih, self.is_wih)
             self.ext.cast_f32_fp8_scaled(self.p("Wih"), self.wih_f8,
                                          self.s_wih)
+            if self._fp8_rs:
+                self.ext.cast_f32_fp8_scaled_t(self.p("Wih"), self.wihT_f8,
+                                               self.s_wih)
+                self.ext.cast_f32_fp8_scaled_t(self.p("W1x"), self.w1xT_f8,
+                                               self.s_w1x)
 
     def _launch_optimizer(self, inc: bool = True):
         if inc:

@@ -43,8 +43,16 @@ hipError_t fv_gemm_nt_fp8(const void*, const void*, const float*,
                           int, int, int, float, int, int, hipStream_t);
 hipError_t fv_absmax_scale(const float*, long, float*, float*, hipStream_t);
 hipError_t fv_gemm_nt_fp8_rs(const void*, const void*, const float*,
-                             const float*, float*, void*, void*, int, int,
-                             int, int, int, float, int, int, hipStream_t);
+                             const float*, const float*, const void*,
+                             const float*, float*, float*, void*, void*,
+                             int, int, int, int, int, float, int, int,
+                             hipStream_t);
+hipError_t fv_scale_from_amax2(float*, float*, float*, float*, float*,
+                               float*, hipStream_t);
+hipError_t fv_cast_f32_fp8_damax(const float*, void*, const float*, float*,
+                                 long, int, int, hipStream_t);
+hipError_t fv_cast_f32_fp8_scaled_t(const float*, void*, const float*, int,
+                                    int, int, hipStream_t);
 hipError_t fv_cast_f32_fp8_scaled(const float*, void*, const float*, long,
                                   int, int, hipStream_t);
 hipError_t fv_ln_bwd_params(const float*, const float*, const float*,
@@ -474,7 +482,11 @@ void gemm_nt_fp8_rs(torch::Tensor A, torch::Tensor Wp,
                     c10::optional<torch::Tensor> out_f32,
                     c10::optional<torch::Tensor> out_bf16,
                     c10::optional<torch::Tensor> out_fp8,
-                    long R, long Ci, long Co, double alpha, bool act_lrelu) {
+                    long R, long Ci, long Co, double alpha, bool act_lrelu,
+                    c10::optional<torch::Tensor> inv_sa = c10::nullopt,
+                    c10::optional<torch::Tensor> lrelu_bwd_of = c10::nullopt,
+                    c10::optional<torch::Tensor> s_out = c10::nullopt,
+                    c10::optional<torch::Tensor> amax_out = c10::nullopt) {
   CK8(A); CK8(Wp);
   const int KP = A.size(1);
   TORCH_CHECK(Wp.size(1) == KP && (KP & 127) == 0 && KP >= Ci,
@@ -484,6 +496,17 @@ void gemm_nt_fp8_rs(torch::Tensor A, torch::Tensor Wp,
   if (bias.has_value()) { CK(*bias); b = fp(*bias); }
   const float* isw = nullptr;
   if (inv_sw.has_value()) { CK(*inv_sw); isw = fp(*inv_sw); }
+  const float* isa = nullptr;
+  if (inv_sa.has_value()) { CK(*inv_sa); isa = fp(*inv_sa); }
+  const void* yp = nullptr;
+  if (lrelu_bwd_of.has_value()) {
+    CKB(*lrelu_bwd_of);
+    yp = bfpc(*lrelu_bwd_of);
+  }
+  const float* sop = nullptr;
+  if (s_out.has_value()) { CK(*s_out); sop = fp(*s_out); }
+  float* amp = nullptr;
+  if (amax_out.has_value()) { CK(*amax_out); amp = fpm(*amax_out); }
   float* of = nullptr; void* ob = nullptr; void* o8 = nullptr; int ldo = 0;
   if (out_f32.has_value()) { CK(*out_f32); of = fpm(*out_f32); }
   if (out_bf16.has_value()) { CKB(*out_bf16); ob = bfp(*out_bf16); }
@@ -493,9 +516,37 @@ void gemm_nt_fp8_rs(torch::Tensor A, torch::Tensor Wp,
     ldo = out_fp8->size(-1);
   }
   TORCH_CHECK(of || ob || o8, "need at least one output");
-  RUN(fv_gemm_nt_fp8_rs(A.data_ptr(), Wp.data_ptr(), b, isw, of, ob, o8,
-                        ldo, (int)R, (int)Ci, (int)Co, KP, (float)alpha,
-                        act_lrelu, b != nullptr, cur_stream()));
+  RUN(fv_gemm_nt_fp8_rs(A.data_ptr(), Wp.data_ptr(), b, isw, isa, yp, sop,
+                        amp, of, ob, o8, ldo, (int)R, (int)Ci, (int)Co, KP,
+                        (float)alpha, act_lrelu, b != nullptr,
+                        cur_stream()));
+}
+
+void scale_from_amax2(torch::Tensor amax1, torch::Tensor s1,
+                      torch::Tensor is1, torch::Tensor amax2,
+                      torch::Tensor s2, torch::Tensor is2) {
+  CK(amax1); CK(s1); CK(is1); CK(amax2); CK(s2); CK(is2);
+  RUN(fv_scale_from_amax2(fpm(amax1), fpm(s1), fpm(is1), fpm(amax2),
+                          fpm(s2), fpm(is2), cur_stream()));
+}
+
+void cast_f32_fp8_damax(torch::Tensor src, torch::Tensor dst,
+                        torch::Tensor s, torch::Tensor amax_out) {
+  CK(src); CK8(dst); CK(s); CK(amax_out);
+  const long rows = src.numel() / src.size(-1);
+  const int cols = src.size(-1);
+  TORCH_CHECK(dst.size(-1) >= cols && dst.numel() / dst.size(-1) >= rows);
+  RUN(fv_cast_f32_fp8_damax(fp(src), dst.data_ptr(), fp(s), fpm(amax_out),
+                            rows, cols, (int)dst.size(-1), cur_stream()));
+}
+
+void cast_f32_fp8_scaled_t(torch::Tensor src, torch::Tensor dstT,
+                           torch::Tensor scale) {
+  CK(src); CK8(dstT); CK(scale);
+  const int M = src.size(0), N = src.size(1);
+  TORCH_CHECK(dstT.size(0) >= N && dstT.size(1) >= M);
+  RUN(fv_cast_f32_fp8_scaled_t(fp(src), dstT.data_ptr(), fp(scale), M, N,
+                               (int)dstT.size(1), cur_stream()));
 }
 
 void absmax_scale(torch::Tensor src, torch::Tensor scale,
@@ -951,7 +1002,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           py::arg("out_f32") = py::none(), py::arg("out_bf16") = py::none(),
           py::arg("out_fp8") = py::none(), py::arg("R"), py::arg("Ci"),
           py::arg("Co"), py::arg("alpha") = 1.0,
-          py::arg("act_lrelu") = false);
+          py::arg("act_lrelu") = false, py::arg("inv_sa") = py::none(),
+          py::arg("lrelu_bwd_of") = py::none(),
+          py::arg("s_out") = py::none(), py::arg("amax_out") = py::none());
+  mod.def("scale_from_amax2", &scale_from_amax2);
+  mod.def("cast_f32_fp8_damax", &cast_f32_fp8_damax);
+  mod.def("cast_f32_fp8_scaled_t", &cast_f32_fp8_scaled_t);
   mod.def("pred_mlp_fwd", &pred_mlp_fwd);
   mod.def("pred_mlp_bwd", &pred_mlp_bwd);
   mod.def("dec_fwd", &dec_fwd);

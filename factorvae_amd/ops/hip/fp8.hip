@@ -150,9 +150,20 @@ __global__ __launch_bounds__(256) void gemm_nt_fp8_kernel(
 typedef __attribute__((ext_vector_type(8))) int i32x8;
 #define FRSK 2  // max k128 groups (k <= 256)
 
+// Extras for the fp8 DGRAD path (delayed per-tensor scaling):
+// - inv_sa: the A operand's inverse scale (folded with inv_sw);
+// - Y (flags bit3): multiply by lrelu'(Y) — fuses the activation
+//   backward into the dgrad GEMM;
+// - s_out: scale applied before the e4m3 out cast (the NEXT consumer's
+//   operand scale);
+// - amax_out: running |v| max of the true results, collected via
+//   atomicMax (max is order-independent -> still deterministic); the
+//   next step's scale_from_amax turns it into s/is.
 __global__ __launch_bounds__(256) void gemm_nt_fp8_rs_kernel(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ Wp,
     const float* __restrict__ bias, const float* __restrict__ inv_sw,
+    const float* __restrict__ inv_sa, const __bf16* __restrict__ Y,
+    const float* __restrict__ s_out, float* __restrict__ amax_out,
     float* __restrict__ out_f32, __bf16* __restrict__ out_bf16,
     unsigned char* __restrict__ out_fp8, int ldo, int R, int Ci, int Co,
     int KP, float alpha, int flags, int spw) {
@@ -196,7 +207,10 @@ __global__ __launch_bounds__(256) void gemm_nt_fp8_rs_kernel(
       }
     }
   };
-  const float sw = inv_sw ? *inv_sw : 1.0f;
+  float sw = inv_sw ? *inv_sw : 1.0f;
+  if (inv_sa) sw *= *inv_sa;
+  const float so = s_out ? *s_out : 1.0f;
+  float amax_l = 0.0f;
   auto compute_store = [&](i32x8 (&fr)[FRSK], int s) {
     f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
 #pragma unroll
@@ -220,9 +234,14 @@ __global__ __launch_bounds__(256) void gemm_nt_fp8_rs_kernel(
         if (flags & 4) v += bias[gc];
         v *= alpha;
         if (flags & 2) v = lrelu_(v);
+        if (flags & 8) {
+          const float y = (float)Y[gr * Co + gc];
+          v *= (y > 0.0f ? 1.0f : 0.01f);
+        }
+        if (amax_out) amax_l = fmaxf(amax_l, fabsf(v));
         if (out_f32) out_f32[gr * Co + gc] = v;
         if (out_bf16) out_bf16[gr * Co + gc] = (__bf16)v;
-        if (out_fp8) out_fp8[gr * ldo + gc] = f32_to_e4m3(v);
+        if (out_fp8) out_fp8[gr * ldo + gc] = f32_to_e4m3(v * so);
       }
     }
   };
@@ -237,6 +256,68 @@ __global__ __launch_bounds__(256) void gemm_nt_fp8_rs_kernel(
     if (s + 1 < s_end) loadA(fr0, s + 1);
     compute_store(fr1, s);
     if (++s >= s_end) break;
+  }
+  if (amax_out) {
+    amax_l = wave_reduce_max(amax_l);
+    if (lane == 0 && amax_l > 0.0f)
+      atomicMax((int*)amax_out, __float_as_int(amax_l));
+  }
+}
+
+// delayed-scaling bookkeeping for the two fp8 dgrad operands: turn the
+// previous step's collected amax into (s = 448/amax, is = amax/448) and
+// reset amax for this step's collection. One tiny launch per step.
+__global__ void scale_from_amax2_kernel(
+    float* __restrict__ amax1, float* __restrict__ s1, float* __restrict__ is1,
+    float* __restrict__ amax2, float* __restrict__ s2,
+    float* __restrict__ is2) {
+  if (threadIdx.x == 0) {
+    const float m1 = fmaxf(amax1[0], 1e-8f);
+    s1[0] = 448.0f / m1;
+    is1[0] = m1 / 448.0f;
+    amax1[0] = 0.0f;
+  } else if (threadIdx.x == 1) {
+    const float m2 = fmaxf(amax2[0], 1e-8f);
+    s2[0] = 448.0f / m2;
+    is2[0] = m2 / 448.0f;
+    amax2[0] = 0.0f;
+  }
+}
+
+// dst_f8(R, ldp) = e4m3(src * s[0]) with fused running-amax collection
+// of |src| (delayed scaling: s comes from the PREVIOUS step's amax).
+__global__ __launch_bounds__(256) void cast_f32_fp8_damax_kernel(
+    const float* __restrict__ src, unsigned char* __restrict__ dst,
+    const float* __restrict__ s, float* __restrict__ amax_out, long rows,
+    int cols, int ldp) {
+  __shared__ float red[4];
+  const float sc = s[0];
+  float m = 0.0f;
+  for (long i = (long)blockIdx.x * 256 + threadIdx.x; i < rows * cols;
+       i += (long)gridDim.x * 256) {
+    const long r = i / cols;
+    const int c = (int)(i % cols);
+    const float v = src[i];
+    m = fmaxf(m, fabsf(v));
+    dst[r * (long)ldp + c] = f32_to_e4m3(v * sc);
+  }
+  m = wave_reduce_max(m);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = m;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    m = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+    if (m > 0.0f) atomicMax((int*)amax_out, __float_as_int(m));
+  }
+}
+
+// transposed scaled cast: dstT(N, ldp) = e4m3(src(M,N)^T * scale[0])
+__global__ __launch_bounds__(256) void cast_f32_fp8_scaled_t_kernel(
+    const float* __restrict__ src, unsigned char* __restrict__ dstT,
+    const float* __restrict__ scale, int M, int N, int ldp) {
+  const long i = (long)blockIdx.x * 256 + threadIdx.x;
+  if (i < (long)M * N) {
+    const int m = (int)(i / N), n = (int)(i % N);
+    dstT[(long)n * ldp + m] = f32_to_e4m3(src[i] * scale[0]);
   }
 }
 
@@ -295,12 +376,14 @@ hipError_t fv_gemm_nt_fp8(const void* A, const void* W, const float* bias,
 
 hipError_t fv_gemm_nt_fp8_rs(const void* A, const void* Wp,
                              const float* bias, const float* inv_sw,
+                             const float* inv_sa, const void* Y,
+                             const float* s_out, float* amax_out,
                              float* out_f32, void* out_bf16, void* out_fp8,
                              int ldo, int R, int Ci, int Co, int KP,
                              float alpha, int act_lrelu, int has_bias,
                              hipStream_t stream) {
   if (KP > 128 * FRSK || KP < Ci || (KP & 127)) return hipErrorInvalidValue;
-  int flags = (act_lrelu ? 2 : 0) | (has_bias ? 4 : 0);
+  int flags = (act_lrelu ? 2 : 0) | (has_bias ? 4 : 0) | (Y ? 8 : 0);
   const int cblocks = (Co + 63) / 64;
   const int strips = (R + 15) / 16;
   int spw = (strips * cblocks) / 4096;
@@ -309,9 +392,40 @@ hipError_t fv_gemm_nt_fp8_rs(const void* A, const void* Wp,
   dim3 grid(cblocks, yblocks);
   hipLaunchKernelGGL(gemm_nt_fp8_rs_kernel, grid, dim3(256), 0, stream,
                      (const unsigned char*)A, (const unsigned char*)Wp, bias,
-                     inv_sw, out_f32, (__bf16*)out_bf16,
-                     (unsigned char*)out_fp8, ldo, R, Ci, Co, KP, alpha,
-                     flags, spw);
+                     inv_sw, inv_sa, (const __bf16*)Y, s_out, amax_out,
+                     out_f32, (__bf16*)out_bf16, (unsigned char*)out_fp8,
+                     ldo, R, Ci, Co, KP, alpha, flags, spw);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_scale_from_amax2(float* amax1, float* s1, float* is1,
+                               float* amax2, float* s2, float* is2,
+                               hipStream_t stream) {
+  hipLaunchKernelGGL(scale_from_amax2_kernel, dim3(1), dim3(64), 0, stream,
+                     amax1, s1, is1, amax2, s2, is2);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_cast_f32_fp8_damax(const float* src, void* dst, const float* s,
+                                 float* amax_out, long rows, int cols,
+                                 int ldp, hipStream_t stream) {
+  long blocks = (rows * cols + 255) / 256;
+  if (blocks > 4096) blocks = 4096;
+  hipLaunchKernelGGL(cast_f32_fp8_damax_kernel, dim3((unsigned)blocks),
+                     dim3(256), 0, stream, src, (unsigned char*)dst, s,
+                     amax_out, rows, cols, ldp);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t fv_cast_f32_fp8_scaled_t(const float* src, void* dstT,
+                                    const float* scale, int M, int N,
+                                    int ldp, hipStream_t stream) {
+  dim3 grid((unsigned)(((long)M * N + 255) / 256));
+  hipLaunchKernelGGL(cast_f32_fp8_scaled_t_kernel, grid, dim3(256), 0,
+                     stream, src, (unsigned char*)dstT, scale, M, N, ldp);
   HIP_CHECK_LAST();
   return hipSuccess;
 }
