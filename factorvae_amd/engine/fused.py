@@ -692,11 +692,25 @@ class FusedTrainer:
                 torch.distributed.all_reduce(tail)
 
         # extractor backward
+        fp8_rs = self.fp8 and getattr(self, "_fp8_rs", False)
+        fp8_gru_fused = fp8_rs and H == 64
+        if fp8_rs:
+            # turn last step's collected amaxes into this step's operand
+            # scales (and reset); must precede the producers below
+            ext.scale_from_amax2(self.amax_dgi, self.s_dgi, self.is_dgi,
+                                 self.amax_dzx, self.s_dzx, self.is_dzx)
         if self.bf16 and H == 64:
+            # fp32 dgi/dgh images are dead in bf16 mode (the wgrads use
+            # the bf16 copies); in fp8 mode the kernel also emits the
+            # scaled e4m3 dgrad operand directly from registers
             ext.gru_bwd_mfma(w["dh"], w["h_prev"], w["gates4"], self.whh_bf,
-                             w["dgi"], w["dgh"], N, T, H,
+                             None, None, N, T, H,
                              w["dgi_bf"].view(N, T, 3 * H),
-                             w["dgh_bf"].view(N, T, 3 * H))
+                             w["dgh_bf"].view(N, T, 3 * H),
+                             dgi_f8=w["dgi_f8"] if fp8_gru_fused else None,
+                             s_dgi=self.s_dgi if fp8_gru_fused else None,
+                             amax_dgi=(self.amax_dgi if fp8_gru_fused
+                                       else None))
         elif self.bf16:
             ext.gru_bwd(w["dh"], w["h_prev"], w["gates4"], p("Whh"),
                         w["dgi"], w["dgh"], N, T, H,
@@ -714,14 +728,13 @@ class FusedTrainer:
                 ext.gemm_tn_bf16(w["dgi_bf"].view(R, 3 * H), w["xp_bf"],
                                  g("Wih"), w["tn_part2"], chunks, True,
                                  g("bih"), w["tn_partb2"])
-            if self.fp8 and self._fp8_rs:
+            if fp8_rs:
                 # fp8 dgrads on the MX K=128 path (delayed per-tensor
                 # scaling; the bf16 dzx copy still feeds the W1x wgrad)
-                ext.scale_from_amax2(self.amax_dgi, self.s_dgi,
-                                     self.is_dgi, self.amax_dzx,
-                                     self.s_dzx, self.is_dzx)
-                ext.cast_f32_fp8_damax(w["dgi"].view(R, 3 * H), w["dgi_f8"],
-                                       self.s_dgi, self.amax_dgi)
+                if not fp8_gru_fused:
+                    ext.cast_f32_fp8_damax(w["dgi"].view(R, 3 * H),
+                                           w["dgi_f8"], self.s_dgi,
+                                           self.amax_dgi)
                 ext.gemm_nt_fp8_rs(w["dgi_f8"], self.wihT_f8, None,
                                    self.is_wih, None, w["dzx_bf"],
                                    w["dzx_f8"], R, 3 * H, self.C, 1.0,
@@ -738,7 +751,7 @@ class FusedTrainer:
                 ext.gemm_tn_bf16(w["dzx_bf"], w["xln_bf"], g("W1x"),
                                  w["tn_part3"], chunks, True, g("b1x"),
                                  w["tn_partb3"])
-            if self.fp8 and self._fp8_rs:
+            if fp8_rs:
                 ext.gemm_nt_fp8_rs(w["dzx_f8"], self.w1xT_f8, None,
                                    self.is_w1x, w["dxln"], None, None, R,
                                    self.C, self.C, 1.0, False,

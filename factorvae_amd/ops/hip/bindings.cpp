@@ -67,8 +67,9 @@ hipError_t fv_gru_fwd_mfma(const float*, const void*, const float*, float*,
                            float*, float*, float*, int, int, int,
                            hipStream_t);
 hipError_t fv_gru_bwd_mfma(const float*, const float*, const float*,
-                           const void*, float*, float*, void*, void*, int,
-                           int, int, hipStream_t);
+                           const void*, float*, float*, void*, void*,
+                           void*, int, const float*, float*, int, int, int,
+                           hipStream_t);
 hipError_t fv_enc_softmax_fwd(const float*, const float*, float*, float*, int,
                               int, hipStream_t);
 hipError_t fv_enc_fused_fwd(const float*, const float*, const float*,
@@ -614,17 +615,37 @@ void gru_fwd_mfma(torch::Tensor gi, torch::Tensor whh_bf, torch::Tensor bhh,
 
 void gru_bwd_mfma(torch::Tensor dh_final, torch::Tensor h_prev,
                   torch::Tensor gates4, torch::Tensor whh_bf,
-                  torch::Tensor dgi, torch::Tensor dgh, long N, long T,
+                  c10::optional<torch::Tensor> dgi,
+                  c10::optional<torch::Tensor> dgh, long N, long T,
                   long H,
                   c10::optional<torch::Tensor> dgi_bf = c10::nullopt,
-                  c10::optional<torch::Tensor> dgh_bf = c10::nullopt) {
-  CK(dh_final); CK(h_prev); CK(gates4); CKB(whh_bf); CK(dgi); CK(dgh);
+                  c10::optional<torch::Tensor> dgh_bf = c10::nullopt,
+                  c10::optional<torch::Tensor> dgi_f8 = c10::nullopt,
+                  c10::optional<torch::Tensor> s_dgi = c10::nullopt,
+                  c10::optional<torch::Tensor> amax_dgi = c10::nullopt) {
+  CK(dh_final); CK(h_prev); CK(gates4); CKB(whh_bf);
+  float* gi = nullptr; float* gh = nullptr;
+  if (dgi.has_value()) { CK(*dgi); gi = fpm(*dgi); }
+  if (dgh.has_value()) { CK(*dgh); gh = fpm(*dgh); }
   void* gib = nullptr; void* ghb = nullptr;
   if (dgi_bf.has_value()) { CKB(*dgi_bf); gib = bfp(*dgi_bf); }
   if (dgh_bf.has_value()) { CKB(*dgh_bf); ghb = bfp(*dgh_bf); }
+  void* g8 = nullptr; int ld8 = 0;
+  const float* sd = nullptr; float* am = nullptr;
+  if (dgi_f8.has_value()) {
+    CK8(*dgi_f8);
+    g8 = dgi_f8->data_ptr();
+    ld8 = dgi_f8->size(-1);
+    TORCH_CHECK(s_dgi.has_value() && amax_dgi.has_value(),
+                "dgi_f8 needs s_dgi + amax_dgi");
+    CK(*s_dgi); CK(*amax_dgi);
+    sd = fp(*s_dgi);
+    am = fpm(*amax_dgi);
+  }
+  TORCH_CHECK(gi || gib || g8, "gru_bwd_mfma needs a dgi output");
   RUN(fv_gru_bwd_mfma(fp(dh_final), fp(h_prev), fp(gates4), bfpc(whh_bf),
-                      fpm(dgi), fpm(dgh), gib, ghb, (int)N, (int)T, (int)H,
-                      cur_stream()));
+                      gi, gh, gib, ghb, g8, ld8, sd, am, (int)N, (int)T,
+                      (int)H, cur_stream()));
 }
 
 void attn_fused_fwd(torch::Tensor h, torch::Tensor qk, torch::Tensor cb,
@@ -974,7 +995,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           py::arg("h_prev"), py::arg("gates4"), py::arg("whh_bf"),
           py::arg("dgi"), py::arg("dgh"), py::arg("N"), py::arg("T"),
           py::arg("H"), py::arg("dgi_bf") = py::none(),
-          py::arg("dgh_bf") = py::none());
+          py::arg("dgh_bf") = py::none(), py::arg("dgi_f8") = py::none(),
+          py::arg("s_dgi") = py::none(), py::arg("amax_dgi") = py::none());
   mod.def("gru_bwd", &gru_bwd, py::arg("dh_final"), py::arg("h_prev"),
           py::arg("gates4"), py::arg("Whh"), py::arg("dgi"), py::arg("dgh"),
           py::arg("N"), py::arg("T"), py::arg("H"),

@@ -19,6 +19,13 @@ DEVINL float softplus_gradf_(float x) {
   return x > 20.0f ? 1.0f : sigmoidf_(x);
 }
 
+// fp32 -> OCP e4m3 (saturating), shared by the fp8 kernels
+DEVINL unsigned char f32_to_e4m3_(float x) {
+  unsigned int v = 0;
+  v = __builtin_amdgcn_cvt_pk_fp8_f32(x, 0.0f, v, false);
+  return (unsigned char)(v & 0xff);
+}
+
 DEVINL float lrelu_(float x) { return x > 0.0f ? x : 0.01f * x; }
 DEVINL float lrelu_grad_from_out_(float y) { return y > 0.0f ? 1.0f : 0.01f; }
 

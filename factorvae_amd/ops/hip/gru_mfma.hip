@@ -151,14 +151,23 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_kernel(
   }
 }
 
+// dgi/dgh fp32 outputs are OPTIONAL (the bf16 engine only consumes the
+// bf16 copies — the fp32 images were ~320 MB of dead writes per A-share
+// step). dgi_f8/s_dgi/amax_dgi: optional fused e4m3 emit for the fp8
+// dgrad path (delayed per-tensor scale s_dgi, running |dgi| amax
+// collected via order-independent atomicMax) — the values are already
+// in registers here, so the fp8 operand costs three extra dword stores
+// per thread-step instead of a separate 160 MB cast pass.
 __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
     const float* __restrict__ dh_final,   // (N,64)
     const float* __restrict__ h_prev_in,  // (N,T,64)
     const float* __restrict__ gates4,     // (N,T,256)
     const void* __restrict__ whh_bf_,     // (192,64) bf16
-    float* __restrict__ dgi,              // (N,T,192)
-    float* __restrict__ dgh,              // (N,T,192)
+    float* __restrict__ dgi,              // (N,T,192) optional
+    float* __restrict__ dgh,              // (N,T,192) optional
     __bf16* __restrict__ dgi_bf, __bf16* __restrict__ dgh_bf,
+    unsigned char* __restrict__ dgi_f8, int ld8,
+    const float* __restrict__ s_dgi, float* __restrict__ amax_dgi,
     int N, int T) {
   const __bf16* whh_bf = (const __bf16*)whh_bf_;
   __shared__ __bf16 dgB[GM_S][GM_GB];    // bf16 dgh image (A-frags)
@@ -188,6 +197,7 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
   const int ej = (tid & 15) * 4;
   const bool elive = (s0 + es) < N;
   const long erow = (long)(s0 + es) * T;
+  float amax_l = 0.0f;
 
   // init dh = dh_final; zero dgB pad rows
   for (int idx = tid; idx < GM_S * GM_GB; idx += 256)
@@ -230,14 +240,35 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
         dgB[es][64 + j] = (__bf16)dgate_z;
         dgB[es][128 + j] = (__bf16)dgh_n;
       }
-      float* di = &dgi[tb * 192];
-      float* dg = &dgh[tb * 192];
-      *(f32x4*)&di[ej] = *(f32x4*)dgr4;
-      *(f32x4*)&di[64 + ej] = *(f32x4*)dgz4;
-      *(f32x4*)&di[128 + ej] = *(f32x4*)da4;
-      *(f32x4*)&dg[ej] = *(f32x4*)dgr4;
-      *(f32x4*)&dg[64 + ej] = *(f32x4*)dgz4;
-      *(f32x4*)&dg[128 + ej] = *(f32x4*)dghn4;
+      if (dgi) {
+        float* di = &dgi[tb * 192];
+        *(f32x4*)&di[ej] = *(f32x4*)dgr4;
+        *(f32x4*)&di[64 + ej] = *(f32x4*)dgz4;
+        *(f32x4*)&di[128 + ej] = *(f32x4*)da4;
+      }
+      if (dgh) {
+        float* dg = &dgh[tb * 192];
+        *(f32x4*)&dg[ej] = *(f32x4*)dgr4;
+        *(f32x4*)&dg[64 + ej] = *(f32x4*)dgz4;
+        *(f32x4*)&dg[128 + ej] = *(f32x4*)dghn4;
+      }
+      if (dgi_f8) {
+        const float s8 = s_dgi[0];
+        union { unsigned int u; unsigned char b[4]; } p0, p1, p2;
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          p0.b[u] = f32_to_e4m3_(dgr4[u] * s8);
+          p1.b[u] = f32_to_e4m3_(dgz4[u] * s8);
+          p2.b[u] = f32_to_e4m3_(da4[u] * s8);
+          amax_l = fmaxf(amax_l,
+                         fmaxf(fabsf(dgr4[u]),
+                               fmaxf(fabsf(dgz4[u]), fabsf(da4[u]))));
+        }
+        unsigned char* d8 = &dgi_f8[tb * (long)ld8];
+        *(unsigned int*)&d8[ej] = p0.u;
+        *(unsigned int*)&d8[64 + ej] = p1.u;
+        *(unsigned int*)&d8[128 + ej] = p2.u;
+      }
       if (dgi_bf) {
 #pragma unroll
         for (int u = 0; u < 4; ++u) {
@@ -271,6 +302,11 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
       dhS[m][i] = acc[rr] + zdh[m][i];
     }
     __syncthreads();
+  }
+  if (dgi_f8 && amax_dgi) {
+    amax_l = wave_reduce_max(amax_l);
+    if ((tid & 63) == 0 && amax_l > 0.0f)
+      atomicMax((int*)amax_dgi, __float_as_int(amax_l));
   }
 }
 
@@ -522,13 +558,15 @@ hipError_t fv_gru_bwd_mfma_f32(const float* dh_final, const float* h_prev,
 hipError_t fv_gru_bwd_mfma(const float* dh_final, const float* h_prev,
                            const float* gates4, const void* whh_bf,
                            float* dgi, float* dgh, void* dgi_bf,
-                           void* dgh_bf, int N, int T, int H,
-                           hipStream_t stream) {
+                           void* dgh_bf, void* dgi_f8, int ld8,
+                           const float* s_dgi, float* amax_dgi, int N,
+                           int T, int H, hipStream_t stream) {
   if (H != 64) return hipErrorInvalidValue;
   dim3 grid((N + GM_S - 1) / GM_S);
   hipLaunchKernelGGL(gru_bwd_mfma_kernel, grid, dim3(256), 0, stream,
                      dh_final, h_prev, gates4, whh_bf, dgi, dgh,
-                     (__bf16*)dgi_bf, (__bf16*)dgh_bf, N, T);
+                     (__bf16*)dgi_bf, (__bf16*)dgh_bf,
+                     (unsigned char*)dgi_f8, ld8, s_dgi, amax_dgi, N, T);
   HIP_CHECK_LAST();
   return hipSuccess;
 }
