@@ -246,15 +246,20 @@ __global__ __launch_bounds__(256) void gemm_nt_fp8_rs_kernel(
     }
   };
 
-  i32x8 fr0[FRSK], fr1[FRSK];
+  // three strips in flight (register rotation, unroll-by-3)
+  i32x8 fr0[FRSK], fr1[FRSK], fr2[FRSK];
   int s = s0;
   loadA(fr0, s);
+  if (s + 1 < s_end) loadA(fr1, s + 1);
   while (true) {
-    if (s + 1 < s_end) loadA(fr1, s + 1);
+    if (s + 2 < s_end) loadA(fr2, s + 2);
     compute_store(fr0, s);
     if (++s >= s_end) break;
-    if (s + 1 < s_end) loadA(fr0, s + 1);
+    if (s + 2 < s_end) loadA(fr0, s + 2);
     compute_store(fr1, s);
+    if (++s >= s_end) break;
+    if (s + 2 < s_end) loadA(fr1, s + 2);
+    compute_store(fr2, s);
     if (++s >= s_end) break;
   }
   if (amax_out) {
