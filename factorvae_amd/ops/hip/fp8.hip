@@ -246,20 +246,15 @@ __global__ __launch_bounds__(256) void gemm_nt_fp8_rs_kernel(
     }
   };
 
-  // three strips in flight (register rotation, unroll-by-3)
-  i32x8 fr0[FRSK], fr1[FRSK], fr2[FRSK];
+  i32x8 fr0[FRSK], fr1[FRSK];
   int s = s0;
   loadA(fr0, s);
-  if (s + 1 < s_end) loadA(fr1, s + 1);
   while (true) {
-    if (s + 2 < s_end) loadA(fr2, s + 2);
+    if (s + 1 < s_end) loadA(fr1, s + 1);
     compute_store(fr0, s);
     if (++s >= s_end) break;
-    if (s + 2 < s_end) loadA(fr0, s + 2);
+    if (s + 1 < s_end) loadA(fr0, s + 1);
     compute_store(fr1, s);
-    if (++s >= s_end) break;
-    if (s + 2 < s_end) loadA(fr1, s + 2);
-    compute_store(fr2, s);
     if (++s >= s_end) break;
   }
   if (amax_out) {

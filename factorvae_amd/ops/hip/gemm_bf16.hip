@@ -393,22 +393,15 @@ __global__ __launch_bounds__(256) void gemm_nt_bf16_rs_kernel(
   };
 
   // two strips in flight (register double buffer, unroll-by-2)
-  // three strips in flight (register rotation, unroll-by-3): the MFMA +
-  // epilogue of one strip is shorter than a global-load round trip, so
-  // two-deep left each wave stalled on its next fragment set
-  bf16x8 fr0[RSK], fr1[RSK], fr2[RSK];
+  bf16x8 fr0[RSK], fr1[RSK];
   int s = s0;
   loadA(fr0, s);
-  if (s + 1 < s_end) loadA(fr1, s + 1);
   while (true) {
-    if (s + 2 < s_end) loadA(fr2, s + 2);
+    if (s + 1 < s_end) loadA(fr1, s + 1);
     compute_store(fr0, s);
     if (++s >= s_end) break;
-    if (s + 2 < s_end) loadA(fr0, s + 2);
+    if (s + 1 < s_end) loadA(fr0, s + 1);
     compute_store(fr1, s);
-    if (++s >= s_end) break;
-    if (s + 2 < s_end) loadA(fr1, s + 2);
-    compute_store(fr2, s);
     if (++s >= s_end) break;
   }
 }
