@@ -159,13 +159,25 @@ def risk_analysis(r: pd.Series, N: int = TRADING_DAYS_PER_YEAR) -> pd.DataFrame:
     )
 
 
+SIMPLIFICATIONS_NOTE = (
+    "simulator: equal-weight top-k dropout with open/close costs only — "
+    "unlike the reference's qlib SimulatorExecutor it does NOT model the "
+    "9.5% price limit, per-trade minimum cost, or account-level share "
+    "rounding; numbers are comparable across models run through THIS "
+    "simulator, not directly against qlib-engine outputs"
+)
+
+
 def backtest_report(result: BacktestResult,
                     N: int = TRADING_DAYS_PER_YEAR) -> Dict[str, pd.DataFrame]:
     """The two tables the reference notebook prints (cell 8): risk
-    analysis of excess return without and with cost."""
+    analysis of excess return without and with cost. The 'note' entry
+    names this simulator's simplifications vs the reference's qlib
+    engine (backtest.ipynb cell 6)."""
     return {
         "excess_return_without_cost": risk_analysis(result.excess_no_cost, N),
         "excess_return_with_cost": risk_analysis(result.excess_with_cost, N),
+        "note": SIMPLIFICATIONS_NOTE,
     }
 
 

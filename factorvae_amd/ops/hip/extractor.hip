@@ -42,16 +42,18 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
     // single pass, vectorized: lane l owns the f32x4 chunk at 4*l (plus
     // a strided tail for C > 256); the row stays in registers for the
     // normalize write — one HBM read of x instead of three
+    // variance uses the TWO-PASS form sum((x-mu)^2) over the
+    // register-resident row (a second wave reduce, no extra memory
+    // traffic): the single-pass E[x^2]-mu^2 form cancels
+    // catastrophically in fp32 when |mean| >> std — plausible for
+    // un-normalized financial features.
     float xv[8];
-    float s = 0.0f, sq = 0.0f;
+    float s = 0.0f;
     if (c4 + 4 <= C) {
       const f32x4 v = *(const f32x4*)&xr[c4];
       xv[0] = v.x; xv[1] = v.y; xv[2] = v.z; xv[3] = v.w;
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        s += xv[i];
-        sq = fmaf(xv[i], xv[i], sq);
-      }
+      for (int i = 0; i < 4; ++i) s += xv[i];
     } else {
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
@@ -59,7 +61,6 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
         const float xx = (c < C) ? xr[c] : 0.0f;
         xv[i] = xx;
         s += xx;
-        sq = fmaf(xx, xx, sq);
       }
     }
     int tail = 0;
@@ -67,14 +68,26 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
       const float xx = xr[c];
       if (tail < 4) xv[4 + tail] = xx;
       s += xx;
-      sq = fmaf(xx, xx, sq);
     }
     s = wave_reduce_sum(s);
-    sq = wave_reduce_sum(sq);
     s = __shfl(s, 0, 64);
-    sq = __shfl(sq, 0, 64);
     const float mu = s / C;
-    const float var = fmaxf(sq / C - mu * mu, 0.0f);
+    float sq = 0.0f;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      if (c4 + i < C) {
+        const float d = xv[i] - mu;
+        sq = fmaf(d, d, sq);
+      }
+    }
+    tail = 0;
+    for (int c = 256 + lane; c < C; c += 64, ++tail) {
+      const float d = ((tail < 4) ? xv[4 + tail] : xr[c]) - mu;
+      sq = fmaf(d, d, sq);
+    }
+    sq = wave_reduce_sum(sq);
+    sq = __shfl(sq, 0, 64);
+    const float var = fmaxf(sq / C, 0.0f);
     const float rs = rsqrtf(var + eps);
 
     if (lane == 0) {

@@ -83,6 +83,7 @@ def main(argv=None):
         print(report["excess_return_without_cost"])
         print("\nexcess_return_with_cost:")
         print(report["excess_return_with_cost"])
+        print(f"\nNOTE: {report['note']}")
         print("\nRankIC:")
         print(RankIC(merged, column1="LABEL0", column2="score"))
     return scores

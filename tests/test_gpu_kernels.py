@@ -853,3 +853,19 @@ def test_gemm_nt_fp8_rs_mx(R, Ci, Co):
     # e4m3 inputs: compare against the e4m3-quantized operands exactly
     assert_close(out, ref, atol=2e-2 * math.sqrt(Ci), rtol=2e-2,
                  what="gemm_nt_fp8_rs")
+
+
+def test_ln_fwd_large_mean_numerics():
+    """ADVICE round-1: fp32 E[x^2]-mu^2 cancels catastrophically when
+    |mean| >> std. The two-pass form must match F.layer_norm even for
+    un-normalized financial-scale inputs (mean ~1e3, std ~1)."""
+    R, C = 512, 158
+    x = t(R, C, seed=77) + 4096.0  # |mean| >> std
+    g_, b_ = t(C, seed=78), t(C, seed=79)
+    xln = torch.empty(R, C, device=DEV)
+    mean = torch.empty(R, device=DEV)
+    rstd = torch.empty(R, device=DEV)
+    ext.ln_fwd(x, g_, b_, xln, mean, rstd, 1e-5)
+    torch.cuda.synchronize()
+    ref = F.layer_norm(x, (C,), g_, b_, 1e-5)
+    assert_close(xln, ref, atol=2e-3, rtol=2e-3, what="ln_fwd large-mean")
