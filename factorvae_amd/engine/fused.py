@@ -848,8 +848,8 @@ class FusedTrainer:
             while len(lru) > self.WS_CACHE_MAX:
                 old = lru.pop(0)
                 self._ws_cache.pop(old, None)
-                for kind in ("train", "predict"):
-                    self._graphs.pop((kind,) + old, None)
+                for gk in [k for k in self._graphs if k[1:3] == old]:
+                    del self._graphs[gk]
             self._alloc_ws(N, T)
 
     # -------------------------------------------------- graph capability probe
@@ -1112,14 +1112,37 @@ class FusedTrainer:
         return run, G
 
     def forward_only(self, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
-        """Fused validation forward (no grad); returns device loss."""
+        """Fused validation forward (no grad); returns device loss.
+        hipGraph-captured per (N, T, dropout-mode) like the train step —
+        the per-epoch validation pass otherwise pays full launch
+        latency for every kernel."""
         N, T, C = x.shape
         self._ensure_ws(N, T)
         w = self.ws
         w["x"].copy_(x)
         w["y"].copy_(y.view(N, 1))
         self._fill_rng(N)
-        self._launch_forward(N, T)
+        if not self.use_graph:
+            self._launch_forward(N, T)
+            return w["loss"]
+        key = ("val", N, T, self.training)
+        if key not in self._graphs:
+            try:
+                torch.cuda.synchronize(self.device)
+                self._launch_forward(N, T)  # warmup
+                torch.cuda.synchronize(self.device)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._launch_forward(N, T)
+                self._graphs[key] = {"g": g}
+            except Exception:
+                torch.cuda.synchronize(self.device)
+                self._graphs[key] = {"g": None}
+        g = self._graphs[key]["g"]
+        if g is None:
+            self._launch_forward(N, T)
+        else:
+            g.replay()
         return w["loss"]
 
     def _launch_predict(self, N: int, T: int):

@@ -3,7 +3,9 @@
 // extern "C" launchers defined in the .hip translation units.
 
 #include <torch/extension.h>
-#include <ATen/cuda/CUDAContext.h>
+// Direct ATen/hip include (torch-ROCm's native header; no CUDA-named
+// compatibility path — the hipify build pass is a no-op on this file).
+#include <ATen/hip/HIPContext.h>
 
 #include <hip/hip_runtime_api.h>
 
@@ -161,7 +163,7 @@ hipError_t fv_adam(float*, const float*, float*, float*, const int*, long,
 namespace {
 
 inline hipStream_t cur_stream() {
-  return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+  return c10::hip::getCurrentHIPStream().stream();
 }
 
 inline const float* fp(const torch::Tensor& t) { return t.data_ptr<float>(); }
