@@ -569,6 +569,12 @@ class FusedTrainer:
                     w["fsig_c"], w["eps"], w["recon"], w["a1"], w["beta"],
                     w["asig_pre"], w["sigma"])
         self._join(0)
+        if self.bf16:
+            # the h_prev bf16 cast was forked onto side stream 2: it must
+            # join before a forward-only hipGraph capture ends (an
+            # unjoined captured branch invalidates the capture; in the
+            # training step the backward's joins covered it)
+            self._join(1)
         if with_loss:
             # one kernel: loss scalars AND the five loss input-gradients
             # (the gradient writes cost nothing extra; validation simply
