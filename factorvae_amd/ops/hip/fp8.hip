@@ -10,6 +10,7 @@
 // to a multiple of 4 bytes so staging can use dword-coalesced loads.
 
 #include "common.h"
+#include <stdlib.h>
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
@@ -386,7 +387,14 @@ hipError_t fv_gemm_nt_fp8_rs(const void* A, const void* Wp,
   int flags = (act_lrelu ? 2 : 0) | (has_bias ? 4 : 0) | (Y ? 8 : 0);
   const int cblocks = (Co + 63) / 64;
   const int strips = (R + 15) / 16;
-  int spw = (strips * cblocks) / 4096;
+  // FV_RS_TGT: target wave count for the strip split (tuning knob;
+  // default 4096 ~= 2 waves per SIMD slot)
+  static int rs_tgt = 0;
+  if (rs_tgt == 0) {
+    const char* e = getenv("FV_RS_TGT");
+    rs_tgt = (e && atoi(e) > 0) ? atoi(e) : 4096;
+  }
+  int spw = (strips * cblocks) / rs_tgt;
   if (spw < 1) spw = 1;
   const int yblocks = (strips + spw * 4 - 1) / (spw * 4);
   dim3 grid(cblocks, yblocks);

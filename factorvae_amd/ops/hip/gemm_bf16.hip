@@ -16,6 +16,7 @@
 //                 reduced in fixed order (deterministic)
 
 #include "common.h"
+#include <stdlib.h>
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
@@ -767,7 +768,14 @@ hipError_t fv_gemm_nt_bf16_rs(const void* A, const void* Wp,
   const int cblocks = (Co + 63) / 64;
   const int strips = (R + 15) / 16;
   // strips per wave: target ~4k waves so every SIMD holds ~2 chunks
-  int spw = (strips * cblocks) / 4096;
+  // FV_RS_TGT: target wave count for the strip split (tuning knob;
+  // default 4096 ~= 2 waves per SIMD slot)
+  static int rs_tgt = 0;
+  if (rs_tgt == 0) {
+    const char* e = getenv("FV_RS_TGT");
+    rs_tgt = (e && atoi(e) > 0) ? atoi(e) : 4096;
+  }
+  int spw = (strips * cblocks) / rs_tgt;
   if (spw < 1) spw = 1;
   const int yblocks = (strips + spw * 4 - 1) / (spw * 4);
   dim3 grid(cblocks, yblocks);
