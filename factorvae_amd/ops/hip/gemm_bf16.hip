@@ -310,6 +310,12 @@ __global__ __launch_bounds__(256) void gemm_nn_bf16_kernel(
 // flags: bit1 lrelu, bit2 bias, bit3 multiply by lrelu'(Y).
 #define RSK 6   // max k32 groups (k <= 192)
 
+// JT = column sub-tiles per wave. JT=4: a wave owns all 64 columns of
+// its strip (96 VGPR of weight fragments -> ~2 waves/SIMD resident).
+// JT=2: waves pair up on a strip, each owning 32 columns — half the
+// weight registers, twice the resident waves, A fragments loaded twice
+// (L2-served: the paired waves run in lockstep).
+template <int JT>
 __global__ __launch_bounds__(256) void gemm_nt_bf16_rs_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ Wp,
     const float* __restrict__ bias, float* __restrict__ out_f32,
@@ -319,13 +325,13 @@ __global__ __launch_bounds__(256) void gemm_nt_bf16_rs_kernel(
   const int wv = threadIdx.x >> 6;
   const int fi = lane & 15;
   const int fk = lane >> 4;
-  const int c0 = blockIdx.x * 64;
+  const int c0 = blockIdx.x * 64 + (JT == 2 ? (wv & 1) * 32 : 0);
   const int nk32 = KP >> 5;
 
   // weight fragments: wave-invariant, one-time guarded load
-  bf16x8 wfr[4][RSK];
+  bf16x8 wfr[JT][RSK];
 #pragma unroll
-  for (int jt = 0; jt < 4; ++jt) {
+  for (int jt = 0; jt < JT; ++jt) {
     const int gc = c0 + jt * 16 + fi;
 #pragma unroll
     for (int k32 = 0; k32 < RSK; ++k32) {
@@ -341,7 +347,9 @@ __global__ __launch_bounds__(256) void gemm_nt_bf16_rs_kernel(
   }
 
   const int strips = (R + 15) >> 4;
-  const int s0 = (blockIdx.y * 4 + wv) * spw;
+  const int wslot = (JT == 2) ? (blockIdx.y * 2 + (wv >> 1))
+                              : (blockIdx.y * 4 + wv);
+  const int s0 = wslot * spw;
   const int s_end = min(s0 + spw, strips);
   if (s0 >= strips) return;
 
@@ -361,18 +369,20 @@ __global__ __launch_bounds__(256) void gemm_nt_bf16_rs_kernel(
     }
   };
   auto compute_store = [&](bf16x8 (&fr)[RSK], int s) {
-    f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+    f32x4 acc[JT];
+#pragma unroll
+    for (int jt = 0; jt < JT; ++jt) acc[jt] = (f32x4){0, 0, 0, 0};
 #pragma unroll
     for (int k32 = 0; k32 < RSK; ++k32) {
       if (k32 < nk32) {
 #pragma unroll
-        for (int jt = 0; jt < 4; ++jt)
+        for (int jt = 0; jt < JT; ++jt)
           acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               fr[k32], wfr[jt][k32], acc[jt], 0, 0, 0);
       }
     }
 #pragma unroll
-    for (int jt = 0; jt < 4; ++jt) {
+    for (int jt = 0; jt < JT; ++jt) {
       const int gc = c0 + jt * 16 + fi;
       if (gc >= Co) continue;
 #pragma unroll
@@ -775,14 +785,28 @@ hipError_t fv_gemm_nt_bf16_rs(const void* A, const void* Wp,
     const char* e = getenv("FV_RS_TGT");
     rs_tgt = (e && atoi(e) > 0) ? atoi(e) : 4096;
   }
+  // FV_RS_JT=2: half-column waves (more resident waves, fewer weight
+  // registers); default full-column
+  static int rs_jt = 0;
+  if (rs_jt == 0) {
+    const char* e2 = getenv("FV_RS_JT");
+    rs_jt = (e2 && atoi(e2) == 2) ? 2 : 4;
+  }
+  const int wps = (rs_jt == 2) ? 2 : 4;  // strip slots per 256-thread WG
   int spw = (strips * cblocks) / rs_tgt;
   if (spw < 1) spw = 1;
-  const int yblocks = (strips + spw * 4 - 1) / (spw * 4);
+  const int yblocks = (strips + spw * wps - 1) / (spw * wps);
   dim3 grid(cblocks, yblocks);
-  hipLaunchKernelGGL(gemm_nt_bf16_rs_kernel, grid, dim3(256), 0, stream,
-                     (const __bf16*)A, (const __bf16*)Wp, bias, out_f32,
-                     (__bf16*)out_bf16, (const __bf16*)Y, R, Ci, Co, KP,
-                     alpha, flags, spw);
+  if (rs_jt == 2)
+    hipLaunchKernelGGL(gemm_nt_bf16_rs_kernel<2>, grid, dim3(256), 0, stream,
+                       (const __bf16*)A, (const __bf16*)Wp, bias, out_f32,
+                       (__bf16*)out_bf16, (const __bf16*)Y, R, Ci, Co, KP,
+                       alpha, flags, spw);
+  else
+    hipLaunchKernelGGL(gemm_nt_bf16_rs_kernel<4>, grid, dim3(256), 0, stream,
+                       (const __bf16*)A, (const __bf16*)Wp, bias, out_f32,
+                       (__bf16*)out_bf16, (const __bf16*)Y, R, Ci, Co, KP,
+                       alpha, flags, spw);
   HIP_CHECK_LAST();
   return hipSuccess;
 }
