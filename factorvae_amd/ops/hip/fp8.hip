@@ -160,6 +160,7 @@ typedef __attribute__((ext_vector_type(8))) int i32x8;
 // - amax_out: running |v| max of the true results, collected via
 //   atomicMax (max is order-independent -> still deterministic); the
 //   next step's scale_from_amax turns it into s/is.
+template <int JT>
 __global__ __launch_bounds__(256) void gemm_nt_fp8_rs_kernel(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ Wp,
     const float* __restrict__ bias, const float* __restrict__ inv_sw,
@@ -172,12 +173,12 @@ __global__ __launch_bounds__(256) void gemm_nt_fp8_rs_kernel(
   const int wv = threadIdx.x >> 6;
   const int fi = lane & 15;
   const int fk = lane >> 4;
-  const int c0 = blockIdx.x * 64;
+  const int c0 = blockIdx.x * 64 + (JT == 2 ? (wv & 1) * 32 : 0);
   const int nk = KP >> 7;  // k128 groups
 
-  i32x8 wfr[4][FRSK];
+  i32x8 wfr[JT][FRSK];
 #pragma unroll
-  for (int jt = 0; jt < 4; ++jt) {
+  for (int jt = 0; jt < JT; ++jt) {
     const int gc = c0 + jt * 16 + fi;
 #pragma unroll
     for (int k = 0; k < FRSK; ++k) {
@@ -191,7 +192,9 @@ __global__ __launch_bounds__(256) void gemm_nt_fp8_rs_kernel(
   }
 
   const int strips = (R + 15) >> 4;
-  const int s0 = (blockIdx.y * 4 + wv) * spw;
+  const int wslot = (JT == 2) ? (blockIdx.y * 2 + (wv >> 1))
+                              : (blockIdx.y * 4 + wv);
+  const int s0 = wslot * spw;
   const int s_end = min(s0 + spw, strips);
   if (s0 >= strips) return;
 
@@ -213,18 +216,20 @@ __global__ __launch_bounds__(256) void gemm_nt_fp8_rs_kernel(
   const float so = s_out ? *s_out : 1.0f;
   float amax_l = 0.0f;
   auto compute_store = [&](i32x8 (&fr)[FRSK], int s) {
-    f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+    f32x4 acc[JT];
+#pragma unroll
+    for (int jt = 0; jt < JT; ++jt) acc[jt] = (f32x4){0, 0, 0, 0};
 #pragma unroll
     for (int k = 0; k < FRSK; ++k) {
       if (k < nk) {
 #pragma unroll
-        for (int jt = 0; jt < 4; ++jt)
+        for (int jt = 0; jt < JT; ++jt)
           acc[jt] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
               fr[k], wfr[jt][k], acc[jt], 0, 0, 0, 127, 0, 127);
       }
     }
 #pragma unroll
-    for (int jt = 0; jt < 4; ++jt) {
+    for (int jt = 0; jt < JT; ++jt) {
       const int gc = c0 + jt * 16 + fi;
       if (gc >= Co) continue;
 #pragma unroll
@@ -394,15 +399,30 @@ hipError_t fv_gemm_nt_fp8_rs(const void* A, const void* Wp,
     const char* e = getenv("FV_RS_TGT");
     rs_tgt = (e && atoi(e) > 0) ? atoi(e) : 4096;
   }
+  static int rs_jt = 0;
+  if (rs_jt == 0) {
+    const char* e2 = getenv("FV_RS_JT");
+    rs_jt = (e2 && atoi(e2) == 4) ? 4 : 2;  // JT=2 measured faster
+  }
+  const int wps = (rs_jt == 2) ? 2 : 4;
   int spw = (strips * cblocks) / rs_tgt;
   if (spw < 1) spw = 1;
-  const int yblocks = (strips + spw * 4 - 1) / (spw * 4);
+  const int yblocks = (strips + spw * wps - 1) / (spw * wps);
   dim3 grid(cblocks, yblocks);
-  hipLaunchKernelGGL(gemm_nt_fp8_rs_kernel, grid, dim3(256), 0, stream,
-                     (const unsigned char*)A, (const unsigned char*)Wp, bias,
-                     inv_sw, inv_sa, (const __bf16*)Y, s_out, amax_out,
-                     out_f32, (__bf16*)out_bf16, (unsigned char*)out_fp8,
-                     ldo, R, Ci, Co, KP, alpha, flags, spw);
+  if (rs_jt == 2)
+    hipLaunchKernelGGL(gemm_nt_fp8_rs_kernel<2>, grid, dim3(256), 0, stream,
+                       (const unsigned char*)A, (const unsigned char*)Wp,
+                       bias, inv_sw, inv_sa, (const __bf16*)Y, s_out,
+                       amax_out, out_f32, (__bf16*)out_bf16,
+                       (unsigned char*)out_fp8, ldo, R, Ci, Co, KP, alpha,
+                       flags, spw);
+  else
+    hipLaunchKernelGGL(gemm_nt_fp8_rs_kernel<4>, grid, dim3(256), 0, stream,
+                       (const unsigned char*)A, (const unsigned char*)Wp,
+                       bias, inv_sw, inv_sa, (const __bf16*)Y, s_out,
+                       amax_out, out_f32, (__bf16*)out_bf16,
+                       (unsigned char*)out_fp8, ldo, R, Ci, Co, KP, alpha,
+                       flags, spw);
   HIP_CHECK_LAST();
   return hipSuccess;
 }
