@@ -403,28 +403,9 @@ __global__ __launch_bounds__(256) void gemm_nt_bf16_rs_kernel(
     }
   };
 
-  // strips in flight: JT=4 keeps 2 (register pressure), JT=2 runs a
-  // 3-deep rotation (the halved weight-register footprint leaves room,
-  // and one strip of MFMA+epilogue is shorter than a load round trip)
+  // two strips in flight (register double buffer, unroll-by-2)
   bf16x8 fr0[RSK], fr1[RSK];
   int s = s0;
-  if (JT == 2) {
-    bf16x8 fr2[RSK];
-    loadA(fr0, s);
-    if (s + 1 < s_end) loadA(fr1, s + 1);
-    while (true) {
-      if (s + 2 < s_end) loadA(fr2, s + 2);
-      compute_store(fr0, s);
-      if (++s >= s_end) break;
-      if (s + 2 < s_end) loadA(fr0, s + 2);
-      compute_store(fr1, s);
-      if (++s >= s_end) break;
-      if (s + 2 < s_end) loadA(fr1, s + 2);
-      compute_store(fr2, s);
-      if (++s >= s_end) break;
-    }
-    return;
-  }
   loadA(fr0, s);
   while (true) {
     if (s + 1 < s_end) loadA(fr1, s + 1);
