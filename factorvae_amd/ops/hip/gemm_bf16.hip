@@ -465,13 +465,14 @@ __global__ __launch_bounds__(256) void cast_shadows_kernel(
 #define TSA (TBM + 8)   // row stride: 36 dwords -> conflict-free tr16 reads
                         // (34 dw collided k-row offsets 0/2 and 1/3)
 
+template <int TK>
 __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     float* __restrict__ out, float* __restrict__ part,
     float* __restrict__ db, float* __restrict__ db_part,
     int R, int M, int N, int accumulate) {
-  __shared__ __bf16 As[2][TBK][TSA];
-  __shared__ __bf16 Bs[2][TBK][TSA];
+  __shared__ __bf16 As[2][TK][TSA];
+  __shared__ __bf16 Bs[2][TK][TSA];
   __shared__ float bred[8][TBM];
 
   const int m0 = blockIdx.x * TBM;
@@ -503,22 +504,22 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
   f32x4 acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
 
   const int span = rend - rbeg;
-  const int ktiles = (span + TBK - 1) / TBK;
-  unsigned int pa[8], pb[8];
+  const int ktiles = (span + TK - 1) / TK;
+  unsigned int pa[TK / 8], pb[TK / 8];
 
   auto stage_regs = [&](int r0_) {
-    const bool interior = (r0_ + TBK <= rend) && (m0 + TBM <= M) &&
+    const bool interior = (r0_ + TK <= rend) && (m0 + TBM <= M) &&
                           (n0 + TBN <= N);
     if (interior) {
 #pragma unroll
-      for (int u = 0; u < 8; ++u) {
+      for (int u = 0; u < TK / 8; ++u) {
         const long gr = (long)r0_ + s_kr0 + 8 * u;
         pa[u] = *(const unsigned int*)(A + gr * M + m0 + s_cp);
         pb[u] = *(const unsigned int*)(B + gr * N + n0 + s_cp);
       }
     } else {
 #pragma unroll
-      for (int u = 0; u < 8; ++u) {
+      for (int u = 0; u < TK / 8; ++u) {
         const long gr = (long)r0_ + s_kr0 + 8 * u;
         pa[u] = load_dw_guard(A, gr, m0 + s_cp, rend, M, M);
         pb[u] = load_dw_guard(B, gr, n0 + s_cp, rend, N, N);
@@ -527,7 +528,7 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
   };
   auto regs_to_lds = [&](int buf) {
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
+    for (int u = 0; u < TK / 8; ++u) {
       const int kr = s_kr0 + 8 * u;
       *(unsigned int*)&As[buf][kr][s_cp] = pa[u];
       *(unsigned int*)&Bs[buf][kr][s_cp] = pb[u];
@@ -535,7 +536,7 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
   };
   auto bias_from_regs = [&]() {
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
+    for (int u = 0; u < TK / 8; ++u) {
       dw_bf2 d;
       d.u = pa[u];
       bsum0 += (float)d.h[0];
@@ -552,12 +553,12 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
   for (int kt = 0; kt < ktiles; ++kt) {
     __syncthreads();
     if (kt + 1 < ktiles) {
-      stage_regs(rbeg + (kt + 1) * TBK);
+      stage_regs(rbeg + (kt + 1) * TK);
       if (do_bias) bias_from_regs();
     }
     const int buf = kt & 1;
 #pragma unroll
-    for (int k32 = 0; k32 < TBK; k32 += 32) {
+    for (int k32 = 0; k32 < TK; k32 += 32) {
       const int kb = k32 + fk * 8 + qm;
       s16x4 a0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
           (lds_v4p)&As[buf][kb][wv * 16 + nq]);
@@ -752,9 +753,19 @@ hipError_t fv_gemm_tn_bf16(const void* A, const void* B, float* out,
   }
   if (r_chunks > 1 && db && !db_part) r_chunks = 1;
   dim3 grid((M + TBM - 1) / TBM, (N + TBN - 1) / TBN, r_chunks);
-  hipLaunchKernelGGL(gemm_tn_bf16_kernel, grid, dim3(256), 0, stream,
-                     (const __bf16*)A, (const __bf16*)B, out, part, db,
-                     db_part, R, M, N, accumulate);
+  static int tn_tk = 0;
+  if (tn_tk == 0) {
+    const char* e = getenv("FV_TN_TK");
+    tn_tk = (e && atoi(e) == 128) ? 128 : 64;
+  }
+  if (tn_tk == 128)
+    hipLaunchKernelGGL(gemm_tn_bf16_kernel<128>, grid, dim3(256), 0, stream,
+                       (const __bf16*)A, (const __bf16*)B, out, part, db,
+                       db_part, R, M, N, accumulate);
+  else
+    hipLaunchKernelGGL(gemm_tn_bf16_kernel<64>, grid, dim3(256), 0, stream,
+                       (const __bf16*)A, (const __bf16*)B, out, part, db,
+                       db_part, R, M, N, accumulate);
   HIP_CHECK_LAST();
   if (r_chunks > 1) {
     const long elems = (long)M * N;
