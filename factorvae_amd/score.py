@@ -44,6 +44,10 @@ def build_argparser() -> argparse.ArgumentParser:
                    default=False)
     p.add_argument("--backtest", action="store_true",
                    help="run the top-k dropout backtest + risk report")
+    p.add_argument("--report", action="store_true",
+                   help="with --backtest: also write backtest.png and the "
+                        "plotly HTML report (the reference notebook's "
+                        "backtest.png / backtest_plotly/ artifacts)")
     p.add_argument("--topk", type=int, default=50)
     p.add_argument("--n_drop", type=int, default=10)
     return p
@@ -86,6 +90,15 @@ def main(argv=None):
         print(f"\nNOTE: {report['note']}")
         print("\nRankIC:")
         print(RankIC(merged, column1="LABEL0", column2="score"))
+        if args.report:
+            from .report import write_backtest_png, write_plotly_report
+
+            png = write_backtest_png(
+                result, os.path.join(args.out_dir, "backtest.png"),
+                title=f"{args.run_name} backtest")
+            html = write_plotly_report(
+                result, os.path.join(args.out_dir, "backtest_plotly"))
+            print(f"wrote {png} and {html}")
     return scores
 
 

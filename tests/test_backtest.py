@@ -1,5 +1,6 @@
 """L6 backtest/reporting layer tests (reference backtest.ipynb cells 6-9)."""
 
+import os
 import numpy as np
 import pandas as pd
 import pytest
@@ -155,3 +156,18 @@ def test_score_cli_end_to_end(tmp_path):
     back = pd.read_csv(csvs[0])
     assert list(back.columns) == ["datetime", "instrument", "score"]
     assert len(back) == len(out)
+
+
+def test_report_artifacts(tmp_path):
+    """backtest.png + plotly HTML report (the reference notebook's
+    artifacts, VERDICT round-1 missing item 4)."""
+    from factorvae_amd.report import write_backtest_png, write_plotly_report
+
+    df = _frame()
+    res = topk_dropout_backtest(df, config=BacktestConfig(topk=10, n_drop=5))
+    png = write_backtest_png(res, str(tmp_path / "backtest.png"))
+    assert os.path.exists(png) and os.path.getsize(png) > 10000
+    html = write_plotly_report(res, str(tmp_path / "backtest_plotly"))
+    assert os.path.exists(html)
+    body = open(html).read()
+    assert "cumulative excess return" in body
