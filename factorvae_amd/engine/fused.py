@@ -651,7 +651,7 @@ class FusedTrainer:
                 ext.attn_qk_bwd(w["dqk"], w["dc"], self.p_q, self.p_Wk,
                                 self.p_bk, gq, gWk, gbk)
             # cross-head Wl wgrad (dz2 is final in both branches)
-            ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, True, g("bl"))
+            ext.gemm_tn(w["dz2"], w["ctx"], g("Wl"), None, 1, False, g("bl"))
 
         # ---- decoder backward (main)
         ext.dec_bwd(w["drecon"], w["h"], w["a1"], w["beta"], w["asig_pre"],
@@ -661,9 +661,9 @@ class FusedTrainer:
                     g("wmu_d"), g("bmu_d"), g("wsig_d"), g("bsig_d"))
         fork()
         with _on_side(self):
-            ext.gemm_tn(w["dz1"], w["h"], g("W1d"), w["tn_part_s"], 2, True,
+            ext.gemm_tn(w["dz1"], w["h"], g("W1d"), w["tn_part_s"], 2, False,
                         g("b1d"), w["tn_partb_s"])
-            ext.gemm_tn(w["dbeta"], w["h"], g("Wb"), w["tn_part_s"], 2, True,
+            ext.gemm_tn(w["dbeta"], w["h"], g("Wb"), w["tn_part_s"], 2, False,
                         g("bb"), w["tn_partb_s"])
 
         # ---- encoder backward (heads + stock-axis softmax bwd, main)
@@ -674,7 +674,7 @@ class FusedTrainer:
         fork()
         with _on_side(self):
             ext.gemm_tn(w["dscores"], w["h"], g("Wenc"), w["tn_part_s"], 2,
-                        True, g("benc"), w["tn_partb_s"])
+                        False, g("benc"), w["tn_partb_s"])
 
         # ---- join attention branch; assemble all three dh contributions
         # (attention ds@qk + a@du, encoder dscores@Wenc) in ONE kernel
@@ -729,10 +729,10 @@ class FusedTrainer:
             fork(1)
             with _on_side(self, 1):
                 ext.gemm_tn_bf16(w["dgh_bf"].view(R, 3 * H), w["h_prev_bf"],
-                                 g("Whh"), w["tn_part"], chunks, True,
+                                 g("Whh"), w["tn_part"], chunks, False,
                                  g("bhh"), w["tn_partb"])
                 ext.gemm_tn_bf16(w["dgi_bf"].view(R, 3 * H), w["xp_bf"],
-                                 g("Wih"), w["tn_part2"], chunks, True,
+                                 g("Wih"), w["tn_part2"], chunks, False,
                                  g("bih"), w["tn_partb2"])
             if fp8_rs:
                 # fp8 dgrads on the MX K=128 path (delayed per-tensor
@@ -755,7 +755,7 @@ class FusedTrainer:
             fork(1)
             with _on_side(self, 1):
                 ext.gemm_tn_bf16(w["dzx_bf"], w["xln_bf"], g("W1x"),
-                                 w["tn_part3"], chunks, True, g("b1x"),
+                                 w["tn_part3"], chunks, False, g("b1x"),
                                  w["tn_partb3"])
             if fp8_rs:
                 ext.gemm_nt_fp8_rs(w["dzx_f8"], self.w1xT_f8, None,
@@ -769,10 +769,10 @@ class FusedTrainer:
             fork(1)
             with _on_side(self, 1):
                 ext.gemm_tn(w["dgh"].view(R, 3 * H), w["h_prev"].view(R, H),
-                            g("Whh"), w["tn_part"], chunks, True,
+                            g("Whh"), w["tn_part"], chunks, False,
                             g("bhh"), w["tn_partb"])
                 ext.gemm_tn(w["dgi"].view(R, 3 * H), w["xp"], g("Wih"),
-                            w["tn_part2"], chunks, True, g("bih"),
+                            w["tn_part2"], chunks, False, g("bih"),
                             w["tn_partb2"])
             ext.gemm_nn(w["dgi"].view(R, 3 * H), p("Wih"), None, w["dxp"], 1.0,
                         False, False)
@@ -780,7 +780,7 @@ class FusedTrainer:
             fork(1)
             with _on_side(self, 1):
                 ext.gemm_tn(w["dzx"], w["xln"], g("W1x"), w["tn_part3"], chunks,
-                            True, g("b1x"), w["tn_partb3"])
+                            False, g("b1x"), w["tn_partb3"])
             ext.gemm_nn(w["dzx"], p("W1x"), None, w["dxln"], 1.0, False, False)
         fork()
         with _on_side(self):
@@ -950,8 +950,9 @@ class FusedTrainer:
 
     def _graph_step_body(self, x, y, N, T, rng_in_graph: bool,
                          comm_in_graph: bool, with_opt: bool = True):
-        """The full training step as a capturable kernel sequence."""
-        self.grads.zero_()
+        """The full training step as a capturable kernel sequence.
+        No grads.zero_(): every gradient-arena slot has exactly one
+        plain-writing producer kernel per step."""
         inc_early = False
         if with_opt and self.s_side is not None:
             # LR-step counter increment runs on the side stream, overlapped
@@ -989,7 +990,6 @@ class FusedTrainer:
             from ..observability import roctx_range
 
             self._fill_rng(N)
-            self.grads.zero_()
             with roctx_range("fv_forward"):
                 self._launch_forward(N, T)
             with roctx_range("fv_backward"):

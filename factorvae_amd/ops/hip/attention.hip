@@ -133,8 +133,8 @@ __global__ __launch_bounds__(256) void attn_head_bwd_kernel(
   // dWv[j][i] += dctx[j]*u[i] — flat coalesced sweep
   float* dW = dWv + (long)k * H * H;
   for (int idx = tid; idx < H * H; idx += 256)
-    dW[idx] += dcS[idx / H] * uS[idx % H];
-  if (tid < H) dbv[(long)k * H + tid] += dcS[tid];
+    dW[idx] = dcS[idx / H] * uS[idx % H];
+  if (tid < H) dbv[(long)k * H + tid] = dcS[tid];
 
   // du[i] = sum_j Wv[j][i]*dctx[j] — 4 waves split j, lane <-> i
   float acc = 0.0f;
@@ -212,15 +212,15 @@ __global__ __launch_bounds__(256) void attn_qk_bwd_kernel(
   // dWk[j][i] += q[j]*dqk[i] — flat coalesced sweep
   float* dW = dWk + (long)k * H * H;
   for (int idx = tid; idx < H * H; idx += 256)
-    dW[idx] += qS[idx / H] * dqkS[idx % H];
-  if (tid < H) dbk[(long)k * H + tid] += dck * qS[tid];
+    dW[idx] = qS[idx / H] * dqkS[idx % H];
+  if (tid < H) dbk[(long)k * H + tid] = dck * qS[tid];
 
   // dq[j] = dc*bk[j] + sum_i Wk[j][i]*dqk[i] — wave per j-chunk, lane <-> i
   const int jpw = (H + 3) / 4;
   for (int j = w * jpw; j < min((w + 1) * jpw, H); ++j) {
     float v = (lane < H) ? Wk[((long)k * H + j) * H + lane] * dqkS[lane] : 0.0f;
     v = wave_reduce_sum(v);
-    if (lane == 0) dq[(long)k * H + j] += dck * bk[(long)k * H + j] + v;
+    if (lane == 0) dq[(long)k * H + j] = dck * bk[(long)k * H + j] + v;
   }
 }
 
@@ -321,10 +321,10 @@ __global__ __launch_bounds__(256) void pred_head_reduce_kernel(
   if (k < K) s0 += hpart[(long)k * E + e];
   const float s = wave_reduce_sum(s0 + s1);
   if (lane != 0) return;
-  if (e < H) dwmu[e] += s;
-  else if (e < 2 * H) dwsig[e - H] += s;
-  else if (e == 2 * H) dbmu[0] += s;
-  else dbsig[0] += s;
+  if (e < H) dwmu[e] = s;
+  else if (e < 2 * H) dwsig[e - H] = s;
+  else if (e == 2 * H) dbmu[0] = s;
+  else dbsig[0] = s;
 }
 
 // ---------------------------------------------------------------------
@@ -562,9 +562,9 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_kernel(
   for (int idx = tid; idx < H * H; idx += 256) {
     const float wv_ = Wvk[idx];
     WlS[(idx / H) * SH + (idx % H)] = wv_;
-    dW[idx] += dcS[idx / H] * uS[idx % H];
+    dW[idx] = dcS[idx / H] * uS[idx % H];
   }
-  if (tid < H) dbv[(long)k * H + tid] += dcS[tid];
+  if (tid < H) dbv[(long)k * H + tid] = dcS[tid];
   __syncthreads();
   {
     float acc = 0.0f;
@@ -643,15 +643,15 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_kernel(
   float* dWkh = dWk + (long)k * H * H;
   for (int idx = tid; idx < H * H; idx += 256) {
     WlS[(idx / H) * SH + (idx % H)] = Wkk[idx];
-    dWkh[idx] += qh[idx / H] * dqkS[idx % H];
+    dWkh[idx] = qh[idx / H] * dqkS[idx % H];
   }
-  if (tid < H) dbk[(long)k * H + tid] += dck * qh[tid];
+  if (tid < H) dbk[(long)k * H + tid] = dck * qh[tid];
   __syncthreads();
   for (int j = w * jpw; j < min((w + 1) * jpw, H); ++j) {
     float v = (lane < H) ? WlS[(size_t)j * SH + lane] * dqkS[lane] : 0.0f;
     v = wave_reduce_sum(v);
     if (lane == 0)
-      dq[(long)k * H + j] += dck * bk[(long)k * H + j] + v;
+      dq[(long)k * H + j] = dck * bk[(long)k * H + j] + v;
   }
 }
 

@@ -275,10 +275,10 @@ __global__ __launch_bounds__(256) void dec_bwd_reduce_kernel(
   if (lane != 0) return;
   if (e < K) dfmu[e] += s;
   else if (e < 2 * K) dfsig_c[e - K] += s;
-  else if (e < 2 * K + H) dwmu[e - 2 * K] += s;
-  else if (e < 2 * K + 2 * H) dwsig[e - 2 * K - H] += s;
-  else if (e == 2 * K + 2 * H) dbmu[0] += s;
-  else dbsig[0] += s;
+  else if (e < 2 * K + H) dwmu[e - 2 * K] = s;
+  else if (e < 2 * K + 2 * H) dwsig[e - 2 * K - H] = s;
+  else if (e == 2 * K + 2 * H) dbmu[0] = s;
+  else dbsig[0] = s;
 }
 
 extern "C" {

@@ -146,11 +146,11 @@ __global__ __launch_bounds__(256) void enc_bwd_fused_kernel(
     dmuS[k] = dm;
     dsigS[k] = dsg;
     // column m of the head weight grads (plain +=, column-owned)
-    dWmu[(long)k * M + m] += dm * yp[m];
-    dWsig[(long)k * M + m] += dsg * yp[m];
+    dWmu[(long)k * M + m] = dm * yp[m];
+    dWsig[(long)k * M + m] = dsg * yp[m];
     if (m == 0) {
-      dbmu[k] += dm;
-      dbsig[k] += dsg;
+      dbmu[k] = dm;
+      dbsig[k] = dsg;
     }
   }
   __syncthreads();
@@ -234,12 +234,12 @@ __global__ __launch_bounds__(256) void enc_heads_bwd_kernel(
   }
   for (long idx = threadIdx.x; idx < (long)K * M; idx += 256) {
     const int k = idx / M, m = idx % M;
-    dWmu[idx] += dmuS[k] * yp[m];
-    dWsig[idx] += dsigS[k] * yp[m];
+    dWmu[idx] = dmuS[k] * yp[m];
+    dWsig[idx] = dsigS[k] * yp[m];
   }
   for (int k = threadIdx.x; k < K; k += 256) {
-    dbmu[k] += dmuS[k];
-    dbsig[k] += dsigS[k];
+    dbmu[k] = dmuS[k];
+    dbsig[k] = dsigS[k];
   }
 }
 

@@ -189,8 +189,8 @@ __global__ __launch_bounds__(256) void ln_bwd_reduce_kernel(
   if (z < yblocks) s0 += pe[z];
   float s = wave_reduce_sum(s0 + s1);
   if (lane == 0) {
-    if (e < C) dgamma[e] += s;
-    else dbeta[e - C] += s;
+    if (e < C) dgamma[e] = s;
+    else dbeta[e - C] = s;
   }
 }
 
