@@ -324,6 +324,7 @@ class FusedTrainer:
         w["a_att"] = f(N, K)
         w["sd"] = f(N, K)
         w["guard"] = torch.zeros(K, device=d, dtype=torch.int32)
+        w["enc_done"] = torch.zeros(1, device=d, dtype=torch.int32)
         w["u"] = f(K, H)
         w["ctx"] = f(K, H)
         w["hm2"] = f(K, H)
@@ -555,15 +556,20 @@ class FusedTrainer:
                                  w["hm2"], w["pmu"], w["psig_pre"], w["psig"],
                                  w["psig_c"])
         if N <= 448:
+            # the mu/sigma heads ride the portfolio kernel (its last
+            # workgroup computes them after an agent-scope yp handoff)
             ext.enc_fused_fwd(w["h"], p("Wenc"), p("benc"), yv,
-                              w["scores_enc"], w["a_enc"], w["yp"])
+                              w["scores_enc"], w["a_enc"], w["yp"],
+                              p("Wmu_e"), p("bmu_e"), p("Wsig_e"),
+                              p("bsig_e"), w["fmu"], w["fsig_pre"],
+                              w["fsig"], w["fsig_c"], w["enc_done"])
         else:
             ext.gemm_nt(w["h"], p("Wenc"), p("benc"), w["scores_enc"], 1.0,
                         False, False)
             ext.enc_softmax_fwd(w["scores_enc"], yv, w["a_enc"], w["yp"])
-        ext.enc_heads_fwd(w["yp"], p("Wmu_e"), p("bmu_e"), p("Wsig_e"),
-                          p("bsig_e"), w["fmu"], w["fsig_pre"], w["fsig"],
-                          w["fsig_c"])
+            ext.enc_heads_fwd(w["yp"], p("Wmu_e"), p("bmu_e"), p("Wsig_e"),
+                              p("bsig_e"), w["fmu"], w["fsig_pre"],
+                              w["fsig"], w["fsig_c"])
         ext.dec_fwd(w["h"], p("W1d"), p("b1d"), p("wmu_d"), p("bmu_d"),
                     p("wsig_d"), p("bsig_d"), p("Wb"), p("bb"), w["fmu"],
                     w["fsig_c"], w["eps"], w["recon"], w["a1"], w["beta"],

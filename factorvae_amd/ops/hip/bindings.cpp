@@ -75,8 +75,10 @@ hipError_t fv_gru_bwd_mfma(const float*, const float*, const float*,
 hipError_t fv_enc_softmax_fwd(const float*, const float*, float*, float*, int,
                               int, hipStream_t);
 hipError_t fv_enc_fused_fwd(const float*, const float*, const float*,
-                            const float*, float*, float*, float*, int, int,
-                            int, hipStream_t);
+                            const float*, float*, float*, float*,
+                            const float*, const float*, const float*,
+                            const float*, float*, float*, float*, float*,
+                            int*, int, int, int, int, hipStream_t);
 hipError_t fv_enc_bwd_fused(const float*, const float*, const float*,
                             const float*, const float*, const float*,
                             const float*, const float*, const float*,
@@ -707,11 +709,38 @@ void attn_fused_bwd(torch::Tensor dpmu, torch::Tensor dpsig_c,
 
 void enc_fused_fwd(torch::Tensor h, torch::Tensor Wenc, torch::Tensor benc,
                    torch::Tensor y, torch::Tensor scores, torch::Tensor a,
-                   torch::Tensor yp) {
+                   torch::Tensor yp,
+                   c10::optional<torch::Tensor> Wmu = c10::nullopt,
+                   c10::optional<torch::Tensor> bmu = c10::nullopt,
+                   c10::optional<torch::Tensor> Wsig = c10::nullopt,
+                   c10::optional<torch::Tensor> bsig = c10::nullopt,
+                   c10::optional<torch::Tensor> fmu = c10::nullopt,
+                   c10::optional<torch::Tensor> fsig_pre = c10::nullopt,
+                   c10::optional<torch::Tensor> fsig = c10::nullopt,
+                   c10::optional<torch::Tensor> fsig_c = c10::nullopt,
+                   c10::optional<torch::Tensor> done = c10::nullopt) {
   CK(h); CK(Wenc); CK(benc); CK(y); CK(scores); CK(a); CK(yp);
   const int N = h.size(0), H = h.size(1), M = Wenc.size(0);
+  const float* wm = nullptr; const float* bm = nullptr;
+  const float* ws = nullptr; const float* bs = nullptr;
+  float* fm = nullptr; float* fsp = nullptr; float* fs = nullptr;
+  float* fsc = nullptr; int* dn = nullptr; int K = 0;
+  if (fmu.has_value()) {
+    CK(*Wmu); CK(*bmu); CK(*Wsig); CK(*bsig); CK(*fmu); CK(*fsig_pre);
+    CK(*fsig); CK(*fsig_c);
+    TORCH_CHECK(done.has_value() &&
+                done->scalar_type() == torch::kInt32 && done->is_cuda(),
+                "fused heads need an int32 done counter");
+    wm = fp(*Wmu); bm = fp(*bmu); ws = fp(*Wsig); bs = fp(*bsig);
+    fm = fpm(*fmu); fsp = fpm(*fsig_pre); fs = fpm(*fsig);
+    fsc = fpm(*fsig_c);
+    dn = done->data_ptr<int>();
+    K = fmu->numel();
+    TORCH_CHECK(Wmu->size(0) == K && Wmu->size(1) == M);
+  }
   RUN(fv_enc_fused_fwd(fp(h), fp(Wenc), fp(benc), fp(y), fpm(scores),
-                       fpm(a), fpm(yp), N, M, H, cur_stream()));
+                       fpm(a), fpm(yp), wm, bm, ws, bs, fm, fsp, fs, fsc,
+                       dn, K, N, M, H, cur_stream()));
 }
 
 void enc_bwd_fused(torch::Tensor dfmu, torch::Tensor dfsig_c,
@@ -990,7 +1019,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gru_fwd", &gru_fwd);
   mod.def("gru_fwd_mfma", &gru_fwd_mfma);
   mod.def("attn_fused_fwd", &attn_fused_fwd);
-  mod.def("enc_fused_fwd", &enc_fused_fwd);
+  mod.def("enc_fused_fwd", &enc_fused_fwd, py::arg("h"),
+          py::arg("Wenc"), py::arg("benc"), py::arg("y"),
+          py::arg("scores"), py::arg("a"), py::arg("yp"),
+          py::arg("Wmu") = py::none(), py::arg("bmu") = py::none(),
+          py::arg("Wsig") = py::none(), py::arg("bsig") = py::none(),
+          py::arg("fmu") = py::none(),
+          py::arg("fsig_pre") = py::none(),
+          py::arg("fsig") = py::none(),
+          py::arg("fsig_c") = py::none(),
+          py::arg("done") = py::none());
   mod.def("enc_bwd_fused", &enc_bwd_fused);
   mod.def("attn_fused_bwd", &attn_fused_bwd);
   mod.def("gru_bwd_mfma", &gru_bwd_mfma, py::arg("dh_final"),

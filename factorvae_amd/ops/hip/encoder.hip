@@ -41,6 +41,13 @@ __global__ __launch_bounds__(256) void enc_softmax_fwd_kernel(
 // softmax + portfolio return in ONE kernel — one workgroup per portfolio
 // m, h staged in LDS. Folds the (N,H)x(H,M) gemm_nt into the softmax
 // pass (the column read of scores becomes a LDS-resident dot).
+// When the head pointers are non-null, the LAST workgroup to finish its
+// yp[m] also computes the mu/sigma heads (the former enc_heads_fwd
+// launch): producer WGs agent-release their yp[m] via the done counter,
+// the last WG acquires and reads all of yp (guide: inter-workgroup
+// visibility needs agent-scope release -> counter -> acquire). done
+// must be zero at launch; the last WG resets it (stream-ordered), so
+// the kernel is hipGraph-replay-safe.
 __global__ __launch_bounds__(256) void enc_fused_fwd_kernel(
     const float* __restrict__ h,      // (N,H)
     const float* __restrict__ Wenc,   // (M,H)
@@ -49,6 +56,11 @@ __global__ __launch_bounds__(256) void enc_fused_fwd_kernel(
     float* __restrict__ scores_out,   // (N,M) saved for backward
     float* __restrict__ a,            // (N,M)
     float* __restrict__ yp,           // (M)
+    const float* __restrict__ Wmu, const float* __restrict__ bmu,
+    const float* __restrict__ Wsig, const float* __restrict__ bsig,
+    float* __restrict__ fmu, float* __restrict__ fsig_pre,
+    float* __restrict__ fsig, float* __restrict__ fsig_c,
+    int* __restrict__ done, int K,
     int N, int M, int H) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* hS = (float*)smem;               // [N][H+1]
@@ -89,7 +101,40 @@ __global__ __launch_bounds__(256) void enc_fused_fwd_kernel(
     wy = fmaf(an, y[n], wy);
   }
   wy = block_reduce_sum(wy, scratch);
-  if (tid == 0) yp[m] = wy;
+  __shared__ int lastS;
+  if (tid == 0) {
+    yp[m] = wy;
+    if (fmu) {
+      const int prev = __hip_atomic_fetch_add(done, 1, __ATOMIC_ACQ_REL,
+                                              __HIP_MEMORY_SCOPE_AGENT);
+      lastS = (prev == M - 1) ? 1 : 0;
+      if (lastS)
+        __hip_atomic_store(done, 0, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+    } else {
+      lastS = 0;
+    }
+  }
+  if (!fmu) return;
+  __syncthreads();
+  if (!lastS) return;
+
+  // heads tail (same math as enc_heads_fwd_kernel); the yp reads are
+  // ordered by the acquiring fetch_add above + the barrier
+  for (int k = tid; k < K; k += 256) {
+    float sm = bmu[k], ss = bsig[k];
+    const float* wm = &Wmu[(long)k * M];
+    const float* ws = &Wsig[(long)k * M];
+    for (int i = 0; i < M; ++i) {
+      sm = fmaf(yp[i], wm[i], sm);
+      ss = fmaf(yp[i], ws[i], ss);
+    }
+    fmu[k] = sm;
+    fsig_pre[k] = ss;
+    const float s = softplusf_(ss);
+    fsig[k] = s;
+    fsig_c[k] = (s == 0.0f) ? 1e-6f : s;
+  }
 }
 
 // dyp (M) -> dscores (N,M):  da[n] = dyp[m]*y[n];
@@ -262,13 +307,18 @@ hipError_t fv_enc_bwd_fused(const float* dfmu, const float* dfsig_c,
 
 hipError_t fv_enc_fused_fwd(const float* h, const float* Wenc,
                             const float* benc, const float* y,
-                            float* scores, float* a, float* yp, int N, int M,
+                            float* scores, float* a, float* yp,
+                            const float* Wmu, const float* bmu,
+                            const float* Wsig, const float* bsig,
+                            float* fmu, float* fsig_pre, float* fsig,
+                            float* fsig_c, int* done, int K, int N, int M,
                             int H, hipStream_t s) {
   if (H > 64) return hipErrorInvalidValue;
   const size_t lds = ((size_t)N * (H + 1) + N + 8 + 64) * sizeof(float);
   if (lds > 128 * 1024) return hipErrorInvalidValue;
   hipLaunchKernelGGL(enc_fused_fwd_kernel, dim3(M), dim3(256), lds, s,
-                     h, Wenc, benc, y, scores, a, yp, N, M, H);
+                     h, Wenc, benc, y, scores, a, yp, Wmu, bmu, Wsig, bsig,
+                     fmu, fsig_pre, fsig, fsig_c, done, K, N, M, H);
   HIP_CHECK_LAST();
   return hipSuccess;
 }
