@@ -10,6 +10,17 @@
 
 DEVINL float sigmoidf_(float x) { return 1.0f / (1.0f + __expf(-x)); }
 
+// fast tanh via __expf (same fast-math family as sigmoidf_ above; the
+// libm tanhf lowers to a multi-branch ocml path that dominates the
+// GRU's per-step latency chain). Computed on |x| so the exp never
+// overflows; ~2 ulp of __expf, well inside the GRU parity tolerances.
+DEVINL float tanhf_(float x) {
+  const float ax = fabsf(x);
+  const float t = __expf(-2.0f * ax);
+  const float r = (1.0f - t) / (1.0f + t);
+  return copysignf(r, x);
+}
+
 // softplus matching torch.nn.functional.softplus (beta=1, threshold=20)
 DEVINL float softplusf_(float x) {
   return x > 20.0f ? x : log1pf(__expf(x));

@@ -70,7 +70,7 @@ __global__ __launch_bounds__(256) void gru_fwd_generic_kernel(
       const float gin = gi[base + 2 * H + lane];
       const float r = sigmoidf_(gir + ghr);
       const float z = sigmoidf_(giz + ghz);
-      const float n = tanhf(fmaf(r, q, gin));
+      const float n = tanhf_(fmaf(r, q, gin));
       const float hp_l = hprev[w * H + lane];
       hn = fmaf(z, hp_l - n, n);  // (1-z)*n + z*hp
 
@@ -249,7 +249,7 @@ __global__ __launch_bounds__(512) void gru_fwd_fast_kernel(
       }
       const float r = sigmoidf_(gir + ghr);
       const float z = sigmoidf_(giz + ghz);
-      const float n = tanhf(fmaf(r, q, gin));
+      const float n = tanhf_(fmaf(r, q, gin));
       const float hp_l = hprev[w * 64 + lane];
       hn = fmaf(z, hp_l - n, n);
 

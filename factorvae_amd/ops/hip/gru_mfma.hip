@@ -124,7 +124,7 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_kernel(
         const float r = sigmoidf_(pgr[u] + ghS[es][j] + bh_r[u]);
         const float z = sigmoidf_(pgz[u] + ghS[es][64 + j] + bh_z[u]);
         const float q = ghS[es][128 + j] + bh_n[u];
-        const float n = tanhf(fmaf(r, q, pgn[u]));
+        const float n = tanhf_(fmaf(r, q, pgn[u]));
         hn[u] = fmaf(z, hp - n, n);
         gr4[u] = r; gz4[u] = z; gn4[u] = n; gq4[u] = q; hp4[u] = hp;
       }
@@ -402,7 +402,7 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_f32_kernel(
         const float r = sigmoidf_(pgr[u] + ghS[es][j] + bh_r[u]);
         const float z = sigmoidf_(pgz[u] + ghS[es][64 + j] + bh_z[u]);
         const float q = ghS[es][128 + j] + bh_n[u];
-        const float n = tanhf(fmaf(r, q, pgn[u]));
+        const float n = tanhf_(fmaf(r, q, pgn[u]));
         hn[u] = fmaf(z, hp - n, n);
         gr4[u] = r; gz4[u] = z; gn4[u] = n; gq4[u] = q; hp4[u] = hp;
       }
