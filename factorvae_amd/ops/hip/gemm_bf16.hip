@@ -325,7 +325,8 @@ __global__ __launch_bounds__(256) void gemm_nt_bf16_rs_kernel(
   const int wv = threadIdx.x >> 6;
   const int fi = lane & 15;
   const int fk = lane >> 4;
-  const int c0 = blockIdx.x * 64 + (JT == 2 ? (wv & 1) * 32 : 0);
+  const int c0 = blockIdx.x * 64 +
+                 (JT == 2 ? (wv & 1) * 32 : (JT == 1 ? (wv & 3) * 16 : 0));
   const int nk32 = KP >> 5;
 
   // weight fragments: wave-invariant, one-time guarded load
@@ -348,7 +349,8 @@ __global__ __launch_bounds__(256) void gemm_nt_bf16_rs_kernel(
 
   const int strips = (R + 15) >> 4;
   const int wslot = (JT == 2) ? (blockIdx.y * 2 + (wv >> 1))
-                              : (blockIdx.y * 4 + wv);
+                              : (JT == 1 ? blockIdx.y
+                                         : (blockIdx.y * 4 + wv));
   const int s0 = wslot * spw;
   const int s_end = min(s0 + spw, strips);
   if (s0 >= strips) return;
@@ -801,14 +803,20 @@ hipError_t fv_gemm_nt_bf16_rs(const void* A, const void* Wp,
   static int rs_jt = 0;
   if (rs_jt == 0) {
     const char* e2 = getenv("FV_RS_JT");
-    rs_jt = (e2 && atoi(e2) == 4) ? 4 : 2;  // JT=2 measured faster (140 vs 108 TF/s isolated)
+    const int v = e2 ? atoi(e2) : 2;
+    rs_jt = (v == 4 || v == 1) ? v : 2;  // JT=2 measured fastest so far
   }
-  const int wps = (rs_jt == 2) ? 2 : 4;  // strip slots per 256-thread WG
+  const int wps = (rs_jt == 4) ? 4 : (rs_jt == 2 ? 2 : 1);
   int spw = (strips * cblocks) / rs_tgt;
   if (spw < 1) spw = 1;
   const int yblocks = (strips + spw * wps - 1) / (spw * wps);
   dim3 grid(cblocks, yblocks);
-  if (rs_jt == 2)
+  if (rs_jt == 1)
+    hipLaunchKernelGGL(gemm_nt_bf16_rs_kernel<1>, grid, dim3(256), 0, stream,
+                       (const __bf16*)A, (const __bf16*)Wp, bias, out_f32,
+                       (__bf16*)out_bf16, (const __bf16*)Y, R, Ci, Co, KP,
+                       alpha, flags, spw);
+  else if (rs_jt == 2)
     hipLaunchKernelGGL(gemm_nt_bf16_rs_kernel<2>, grid, dim3(256), 0, stream,
                        (const __bf16*)A, (const __bf16*)Wp, bias, out_f32,
                        (__bf16*)out_bf16, (const __bf16*)Y, R, Ci, Co, KP,
