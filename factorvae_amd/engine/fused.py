@@ -116,23 +116,8 @@ class FusedTrainer:
                 f"use engine='eager'")
 
         self._build_param_arena()
-        C3 = 3 * self.H
-        # fp32 engine mode: the extractor GEMMs run as split-bf16
-        # 3-product MFMA (hi/lo weight shadows, ~2^-16 relative error,
-        # bf16-MFMA rate) unless FV_F32_GEMM=exact forces the exact
-        # 16x16x4-f32 path
-        self._f32s = (not self.bf16 and self.C <= 192 and C3 <= 192
-                      and _os.environ.get("FV_F32_GEMM", "") != "exact")
-        if self._f32s:
-            KPc = (self.C + 31) & ~31
-            KP3 = (C3 + 31) & ~31
-            zb = lambda *s: torch.zeros(*s, dtype=torch.bfloat16,
-                                        device=self.device)
-            self.w1x_hi, self.w1x_lo = zb(self.C, KPc), zb(self.C, KPc)
-            self.w1xT_hi, self.w1xT_lo = zb(self.C, KPc), zb(self.C, KPc)
-            self.wih_hi, self.wih_lo = zb(C3, KPc), zb(C3, KPc)
-            self.wihT_hi, self.wihT_lo = zb(self.C, KP3), zb(self.C, KP3)
         if self.bf16:
+            C3 = 3 * self.H
             # padded (+ transposed-padded) weight shadows for the
             # register-stationary NT kernel: k dim zero-padded to a
             # multiple of 32 so the MFMA k-tail is exact (pads are
@@ -183,8 +168,6 @@ class FusedTrainer:
                     self.s_dzx = one(448.0)
                     self.is_dzx = one(1.0 / 448.0)
             self._refresh_bf16_shadows()
-        if self._f32s:
-            self._refresh_f32s_shadows()
         self.grads = torch.zeros_like(self.params.flat)
         self.adam_m = torch.zeros_like(self.params.flat)
         self.adam_v = torch.zeros_like(self.params.flat)
@@ -315,18 +298,13 @@ class FusedTrainer:
         C, H, M, K = self.C, self.H, self.M, self.K
         R = N * T
         f = lambda *shape: torch.zeros(*shape, device=d, dtype=torch.float32)
-        # fp32 A-side operands of the split-bf16 GEMMs get 8 floats of
-        # tail slack (k-tail b128 fragments may read past the last row;
-        # the zero-padded weights make the contribution exact)
-        fs = lambda r, c: torch.zeros(r * c + 8, device=d,
-                                      dtype=torch.float32)[:r * c].view(r, c)
         w = {}
         w["x"] = f(N, T, C)
         w["y"] = f(N, 1)
-        w["xln"] = fs(R, C)
+        w["xln"] = f(R, C)
         w["mean"] = f(R)
         w["rstd"] = f(R)
-        w["xp"] = fs(R, C)
+        w["xp"] = f(R, C)
         w["gi"] = f(R, 3 * H)
         w["h"] = f(N, H)
         w["h_seq"] = f(N, T, H)
@@ -412,7 +390,7 @@ class FusedTrainer:
         # yblocks at very large R exceeds the partial buffer
         rpb = min(8192, max(64, (R + ty - 1) // ty))
         w["ln_part"] = f(((R + rpb - 1) // rpb + 1) * 2 * C)
-        w["dzx"] = fs(R, C)
+        w["dzx"] = f(R, C)
         w["dxln"] = f(R, C)
         if self.bf16:
             fb = lambda *shape: torch.zeros(*shape, device=d,
@@ -530,17 +508,8 @@ class FusedTrainer:
         else:
             ext.ln_fwd(x2d, p("ln_g"), p("ln_b"), w["xln"],
                        w["mean"], w["rstd"], 1e-5)
-            if self._f32s:
-                ext.gemm_nt_f32s_rs(w["xln"], self.w1x_hi, self.w1x_lo,
-                                    p("b1x"), w["xp"], None, 1.0, True)
-                ext.gemm_nt_f32s_rs(w["xp"], self.wih_hi, self.wih_lo,
-                                    p("bih"), w["gi"].view(R, 3 * H), None,
-                                    1.0, False)
-            else:
-                ext.gemm_nt(w["xln"], p("W1x"), p("b1x"), w["xp"], 1.0,
-                            False, True)
-                ext.gemm_nt(w["xp"], p("Wih"), p("bih"), w["gi"], 1.0,
-                            False, False)
+            ext.gemm_nt(w["xln"], p("W1x"), p("b1x"), w["xp"], 1.0, False, True)
+            ext.gemm_nt(w["xp"], p("Wih"), p("bih"), w["gi"], 1.0, False, False)
         if self.bf16 and H == 64:
             ext.gru_fwd_mfma(w["gi"], self.whh_bf, p("bhh"), w["h"],
                              w["h_seq"], w["h_prev"], w["gates4"], N, T, H)
@@ -811,25 +780,14 @@ class FusedTrainer:
                 ext.gemm_tn(w["dgi"].view(R, 3 * H), w["xp"], g("Wih"),
                             w["tn_part2"], chunks, False, g("bih"),
                             w["tn_partb2"])
-            if self._f32s:
-                # dzx = lrelu'(xp) * (dgi @ Wih) in one split-GEMM call
-                ext.gemm_nt_f32s_rs(w["dgi"].view(R, 3 * H), self.wihT_hi,
-                                    self.wihT_lo, None, w["dzx"], w["xp"],
-                                    1.0, False)
-            else:
-                ext.gemm_nn(w["dgi"].view(R, 3 * H), p("Wih"), None,
-                            w["dxp"], 1.0, False, False)
-                ext.lrelu_bwd(w["dxp"], w["xp"], w["dzx"])
+            ext.gemm_nn(w["dgi"].view(R, 3 * H), p("Wih"), None, w["dxp"], 1.0,
+                        False, False)
+            ext.lrelu_bwd(w["dxp"], w["xp"], w["dzx"])
             fork(1)
             with _on_side(self, 1):
                 ext.gemm_tn(w["dzx"], w["xln"], g("W1x"), w["tn_part3"], chunks,
                             False, g("b1x"), w["tn_partb3"])
-            if self._f32s:
-                ext.gemm_nt_f32s_rs(w["dzx"], self.w1xT_hi, self.w1xT_lo,
-                                    None, w["dxln"], None, 1.0, False)
-            else:
-                ext.gemm_nn(w["dzx"], p("W1x"), None, w["dxln"], 1.0, False,
-                            False)
+            ext.gemm_nn(w["dzx"], p("W1x"), None, w["dxln"], 1.0, False, False)
         fork()
         with _on_side(self):
             ext.ln_bwd_params(x2d, w["dxln"], w["mean"], w["rstd"],
@@ -844,12 +802,6 @@ class FusedTrainer:
                 e = torch.cuda.Event()
                 e.record(sstream)
                 main.wait_event(e)
-
-    def _refresh_f32s_shadows(self):
-        self.ext.cast_shadows_f32s(self.p("W1x"), self.w1x_hi, self.w1x_lo,
-                                   self.w1xT_hi, self.w1xT_lo,
-                                   self.p("Wih"), self.wih_hi, self.wih_lo,
-                                   self.wihT_hi, self.wihT_lo)
 
     def _refresh_bf16_shadows(self):
         self.ext.cast_shadows(self.p("W1x"), self.w1x_p, self.w1xT_p,
@@ -876,8 +828,6 @@ class FusedTrainer:
                       0.9, 0.999, 1e-8)
         if self.bf16:
             self._refresh_bf16_shadows()
-        elif self._f32s:
-            self._refresh_f32s_shadows()
 
     def _fill_rng(self, N: int):
         self.ws["eps"].normal_()

@@ -29,13 +29,6 @@ hipError_t fv_gemm_nt_bf16_rs(const void*, const void*, const float*,
 hipError_t fv_cast_shadows(const float*, void*, void*, int, int, int, int,
                            const float*, void*, void*, int, int, int, int,
                            const float*, void*, long, hipStream_t);
-hipError_t fv_gemm_nt_f32s_rs(const float*, const void*, const void*,
-                              const float*, float*, const float*, int, int,
-                              int, int, float, int, hipStream_t);
-hipError_t fv_cast_shadows_f32s(const float*, void*, void*, void*, void*,
-                                int, int, int, int, const float*, void*,
-                                void*, void*, void*, int, int, int, int,
-                                hipStream_t);
 hipError_t fv_gemm_tn_bf16(const void*, const void*, float*, float*, float*,
                            float*, int, int, int, int, int, hipStream_t);
 hipError_t fv_cast_f32_bf16(const float*, void*, long, hipStream_t);
@@ -347,45 +340,6 @@ void gemm_nt_bf16_rs(torch::Tensor A, torch::Tensor Wp,
   }
   RUN(fv_gemm_nt_bf16_rs(bfpc(A), bfpc(Wp), b, of, ob, yp, R, Ci, Co, KP,
                          (float)alpha, act_lrelu, cur_stream()));
-}
-
-void gemm_nt_f32s_rs(torch::Tensor A, torch::Tensor Whi, torch::Tensor Wlo,
-                     c10::optional<torch::Tensor> bias,
-                     torch::Tensor out_f32,
-                     c10::optional<torch::Tensor> lrelu_bwd_of,
-                     double alpha, bool act_lrelu) {
-  CK(A); CKB(Whi); CKB(Wlo); CK(out_f32);
-  const int R = A.size(0), Ci = A.size(1), KP = Whi.size(1);
-  const int Co = out_f32.size(1);
-  TORCH_CHECK(KP == ((Ci + 31) / 32) * 32, "Whi must be k-padded to 32");
-  TORCH_CHECK(Wlo.sizes() == Whi.sizes() && Whi.size(0) >= Co);
-  TORCH_CHECK(out_f32.size(0) == R);
-  const float* b = nullptr;
-  if (bias.has_value()) { CK(*bias); b = fp(*bias); }
-  const float* yp = nullptr;
-  if (lrelu_bwd_of.has_value()) {
-    CK(*lrelu_bwd_of);
-    TORCH_CHECK(lrelu_bwd_of->size(0) == R && lrelu_bwd_of->size(1) == Co);
-    yp = fp(*lrelu_bwd_of);
-  }
-  RUN(fv_gemm_nt_f32s_rs(fp(A), bfpc(Whi), bfpc(Wlo), b, fpm(out_f32), yp,
-                         R, Ci, Co, KP, (float)alpha, act_lrelu,
-                         cur_stream()));
-}
-
-void cast_shadows_f32s(torch::Tensor s1, torch::Tensor h1, torch::Tensor l1,
-                       torch::Tensor h1t, torch::Tensor l1t,
-                       torch::Tensor s2, torch::Tensor h2, torch::Tensor l2,
-                       torch::Tensor h2t, torch::Tensor l2t) {
-  CK(s1); CKB(h1); CKB(l1); CKB(h1t); CKB(l1t);
-  CK(s2); CKB(h2); CKB(l2); CKB(h2t); CKB(l2t);
-  const int M1 = s1.size(0), N1 = s1.size(1);
-  const int M2 = s2.size(0), N2 = s2.size(1);
-  RUN(fv_cast_shadows_f32s(fp(s1), bfp(h1), bfp(l1), bfp(h1t), bfp(l1t),
-                           M1, N1, (int)h1.size(1), (int)h1t.size(1),
-                           fp(s2), bfp(h2), bfp(l2), bfp(h2t), bfp(l2t),
-                           M2, N2, (int)h2.size(1), (int)h2t.size(1),
-                           cur_stream()));
 }
 
 void cast_shadows(torch::Tensor s1, torch::Tensor d1, torch::Tensor d1t,
@@ -1105,11 +1059,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           py::arg("lrelu_bwd_of") = py::none(), py::arg("alpha") = 1.0,
           py::arg("act_lrelu") = false);
   mod.def("cast_shadows", &cast_shadows);
-  mod.def("gemm_nt_f32s_rs", &gemm_nt_f32s_rs, py::arg("A"), py::arg("Whi"),
-          py::arg("Wlo"), py::arg("bias") = py::none(), py::arg("out_f32"),
-          py::arg("lrelu_bwd_of") = py::none(), py::arg("alpha") = 1.0,
-          py::arg("act_lrelu") = false);
-  mod.def("cast_shadows_f32s", &cast_shadows_f32s);
   mod.def("gemm_nt_fp8_rs", &gemm_nt_fp8_rs, py::arg("A"), py::arg("Wp"),
           py::arg("bias") = py::none(), py::arg("inv_sw") = py::none(),
           py::arg("out_f32") = py::none(), py::arg("out_bf16") = py::none(),

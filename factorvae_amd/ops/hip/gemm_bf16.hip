@@ -419,169 +419,6 @@ __global__ __launch_bounds__(256) void gemm_nt_bf16_rs_kernel(
   }
 }
 
-// ---------------------------------------------- split-bf16 fp32 GEMM
-// fp32 NT at the bf16 MFMA rate via the 3-product split ("bf16x3",
-// Ootomo/Yokota-style): a = ah + al (ah = bf16(a), al = bf16(a - ah)),
-// b likewise pre-split into hi/lo weight shadows; a·b ≈ ah·bh + ah·bl
-// + al·bh with fp32 accumulation. Dropped al·bl and the al rounding
-// give ~2^-16 relative error — fp32-grade at this model's k (<= 192),
-// vs the exact-but-16x-slower v_mfma_16x16x4_f32 path. Same JT=2
-// register-stationary structure as gemm_nt_bf16_rs: per-strip fp32 A
-// rows are loaded and split in registers; Whi/Wlo live in registers.
-// flags: bit1 lrelu, bit2 bias, bit3 multiply by lrelu'(Y) (fp32 Y).
-// A needs >= 32 bytes of row slack when Ci % 32 != 0 (zero-padded
-// weight columns make the k-tail exact).
-__global__ __launch_bounds__(256) void gemm_nt_f32s_rs_kernel(
-    const float* __restrict__ A, const __bf16* __restrict__ Whi,
-    const __bf16* __restrict__ Wlo, const float* __restrict__ bias,
-    float* __restrict__ out_f32, const float* __restrict__ Y,
-    int R, int Ci, int Co, int KP, float alpha, int flags, int spw) {
-  const int lane = threadIdx.x & 63;
-  const int wv = threadIdx.x >> 6;
-  const int fi = lane & 15;
-  const int fk = lane >> 4;
-  const int c0 = blockIdx.x * 64 + (wv & 1) * 32;
-  const int nk32 = KP >> 5;
-
-  bf16x8 whi[2][RSK], wlo[2][RSK];
-#pragma unroll
-  for (int jt = 0; jt < 2; ++jt) {
-    const int gc = c0 + jt * 16 + fi;
-#pragma unroll
-    for (int k32 = 0; k32 < RSK; ++k32) {
-      if (k32 < nk32 && gc < Co) {
-        whi[jt][k32] = *(const bf16x8*)&Whi[(long)gc * KP + k32 * 32 + fk * 8];
-        wlo[jt][k32] = *(const bf16x8*)&Wlo[(long)gc * KP + k32 * 32 + fk * 8];
-      } else {
-        bf16x8 z;
-#pragma unroll
-        for (int u = 0; u < 8; ++u) z[u] = (__bf16)0.0f;
-        whi[jt][k32] = z;
-        wlo[jt][k32] = z;
-      }
-    }
-  }
-
-  const int strips = (R + 15) >> 4;
-  const int wslot = blockIdx.y * 2 + (wv >> 1);
-  const int s0 = wslot * spw;
-  const int s_end = min(s0 + spw, strips);
-  if (s0 >= strips) return;
-
-  struct FragPair { bf16x8 hi[RSK]; bf16x8 lo[RSK]; };
-  auto loadA = [&](FragPair& fr, int s) {
-    const long row = (long)s * 16 + fi;
-    const bool live = row < R;
-#pragma unroll
-    for (int k32 = 0; k32 < RSK; ++k32) {
-      if (k32 < nk32 && live) {
-        const float* p = &A[row * Ci + k32 * 32 + fk * 8];
-        f32x4 v0 = *(const f32x4*)p;
-        f32x4 v1 = *(const f32x4*)(p + 4);
-#pragma unroll
-        for (int u = 0; u < 4; ++u) {
-          const __bf16 h0 = (__bf16)v0[u];
-          const __bf16 h1 = (__bf16)v1[u];
-          fr.hi[k32][u] = h0;
-          fr.hi[k32][4 + u] = h1;
-          fr.lo[k32][u] = (__bf16)(v0[u] - (float)h0);
-          fr.lo[k32][4 + u] = (__bf16)(v1[u] - (float)h1);
-        }
-      } else {
-        bf16x8 z;
-#pragma unroll
-        for (int u = 0; u < 8; ++u) z[u] = (__bf16)0.0f;
-        fr.hi[k32] = z;
-        fr.lo[k32] = z;
-      }
-    }
-  };
-  auto compute_store = [&](FragPair& fr, int s) {
-    f32x4 acc[2] = {{0, 0, 0, 0}, {0, 0, 0, 0}};
-#pragma unroll
-    for (int k32 = 0; k32 < RSK; ++k32) {
-      if (k32 < nk32) {
-#pragma unroll
-        for (int jt = 0; jt < 2; ++jt) {
-          acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              fr.hi[k32], wlo[jt][k32], acc[jt], 0, 0, 0);
-          acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              fr.lo[k32], whi[jt][k32], acc[jt], 0, 0, 0);
-          acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              fr.hi[k32], whi[jt][k32], acc[jt], 0, 0, 0);
-        }
-      }
-    }
-#pragma unroll
-    for (int jt = 0; jt < 2; ++jt) {
-      const int gc = c0 + jt * 16 + fi;
-      if (gc >= Co) continue;
-#pragma unroll
-      for (int rr = 0; rr < 4; ++rr) {
-        const long gr = (long)s * 16 + fk * 4 + rr;
-        if (gr >= R) continue;
-        float v = acc[jt][rr];
-        if (flags & 4) v += bias[gc];
-        v *= alpha;
-        if (flags & 2) v = lrelu_(v);
-        if (flags & 8) {
-          const float y = Y[gr * Co + gc];
-          v *= (y > 0.0f ? 1.0f : 0.01f);
-        }
-        out_f32[gr * Co + gc] = v;
-      }
-    }
-  };
-
-  FragPair fr0, fr1;
-  int s = s0;
-  loadA(fr0, s);
-  while (true) {
-    if (s + 1 < s_end) loadA(fr1, s + 1);
-    compute_store(fr0, s);
-    if (++s >= s_end) break;
-    if (s + 1 < s_end) loadA(fr0, s + 1);
-    compute_store(fr1, s);
-    if (++s >= s_end) break;
-  }
-}
-
-// hi/lo split shadow refresh for the fp32 engine: each weight is
-// emitted as padded bf16 hi/lo pairs AND their transposed pairs in one
-// launch (two weights per call).
-__global__ __launch_bounds__(256) void cast_shadows_f32s_kernel(
-    const float* __restrict__ s1, __bf16* __restrict__ h1,
-    __bf16* __restrict__ l1, __bf16* __restrict__ h1t,
-    __bf16* __restrict__ l1t, int M1, int N1, int KPn1, int KPm1,
-    const float* __restrict__ s2, __bf16* __restrict__ h2,
-    __bf16* __restrict__ l2, __bf16* __restrict__ h2t,
-    __bf16* __restrict__ l2t, int M2, int N2, int KPn2, int KPm2) {
-  long i = (long)blockIdx.x * 256 + threadIdx.x;
-  const long e1 = (long)M1 * N1;
-  if (i < e1) {
-    const int m = (int)(i / N1), n = (int)(i % N1);
-    const float v = s1[i];
-    const __bf16 hi = (__bf16)v;
-    const __bf16 lo = (__bf16)(v - (float)hi);
-    h1[(long)m * KPn1 + n] = hi;
-    l1[(long)m * KPn1 + n] = lo;
-    h1t[(long)n * KPm1 + m] = hi;
-    l1t[(long)n * KPm1 + m] = lo;
-    return;
-  }
-  i -= e1;
-  if (i < (long)M2 * N2) {
-    const int m = (int)(i / N2), n = (int)(i % N2);
-    const float v = s2[i];
-    const __bf16 hi = (__bf16)v;
-    const __bf16 lo = (__bf16)(v - (float)hi);
-    h2[(long)m * KPn2 + n] = hi;
-    l2[(long)m * KPn2 + n] = lo;
-    h2t[(long)n * KPm2 + m] = hi;
-    l2t[(long)n * KPm2 + m] = lo;
-  }
-}
-
 // padded + transposed-padded bf16 weight shadow refresh, ONE launch for
 // both extractor weights plus the flat Whh copy (runs after every Adam
 // step; pads were zeroed at allocation and are never written here)
@@ -989,43 +826,6 @@ hipError_t fv_gemm_nt_bf16_rs(const void* A, const void* Wp,
                        (const __bf16*)A, (const __bf16*)Wp, bias, out_f32,
                        (__bf16*)out_bf16, (const __bf16*)Y, R, Ci, Co, KP,
                        alpha, flags, spw);
-  HIP_CHECK_LAST();
-  return hipSuccess;
-}
-
-hipError_t fv_gemm_nt_f32s_rs(const float* A, const void* Whi,
-                              const void* Wlo, const float* bias,
-                              float* out_f32, const float* Y, int R, int Ci,
-                              int Co, int KP, float alpha, int act_lrelu,
-                              hipStream_t stream) {
-  if (KP > 32 * RSK || KP < Ci || (KP & 31)) return hipErrorInvalidValue;
-  int flags = (act_lrelu ? 2 : 0) | (bias ? 4 : 0) | (Y ? 8 : 0);
-  const int cblocks = (Co + 63) / 64;
-  const int strips = (R + 15) / 16;
-  int spw = (strips * cblocks) / 4096;
-  if (spw < 1) spw = 1;
-  const int yblocks = (strips + spw * 2 - 1) / (spw * 2);
-  dim3 grid(cblocks, yblocks);
-  hipLaunchKernelGGL(gemm_nt_f32s_rs_kernel, grid, dim3(256), 0, stream, A,
-                     (const __bf16*)Whi, (const __bf16*)Wlo, bias, out_f32,
-                     Y, R, Ci, Co, KP, alpha, flags, spw);
-  HIP_CHECK_LAST();
-  return hipSuccess;
-}
-
-hipError_t fv_cast_shadows_f32s(const float* s1, void* h1, void* l1,
-                                void* h1t, void* l1t, int M1, int N1,
-                                int KPn1, int KPm1, const float* s2,
-                                void* h2, void* l2, void* h2t, void* l2t,
-                                int M2, int N2, int KPn2, int KPm2,
-                                hipStream_t stream) {
-  const long total = (long)M1 * N1 + (long)M2 * N2;
-  dim3 grid((unsigned)((total + 255) / 256));
-  hipLaunchKernelGGL(cast_shadows_f32s_kernel, grid, dim3(256), 0, stream,
-                     s1, (__bf16*)h1, (__bf16*)l1, (__bf16*)h1t,
-                     (__bf16*)l1t, M1, N1, KPn1, KPm1, s2, (__bf16*)h2,
-                     (__bf16*)l2, (__bf16*)h2t, (__bf16*)l2t, M2, N2, KPn2,
-                     KPm2);
   HIP_CHECK_LAST();
   return hipSuccess;
 }
