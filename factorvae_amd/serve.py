@@ -82,6 +82,7 @@ class ScoringEngine:
 
 
 try:  # module-level request models (FastAPI resolves annotations here)
+    from fastapi import Request, Response
     from pydantic import BaseModel
 
     class ScoreRequest(BaseModel):
@@ -90,7 +91,7 @@ try:  # module-level request models (FastAPI resolves annotations here)
     class BatchRequest(BaseModel):
         days: List[List[List[List[float]]]]
 except ImportError:  # serving extras absent: ScoringEngine still usable
-    ScoreRequest = BatchRequest = None
+    ScoreRequest = BatchRequest = Request = Response = None
 
 
 def build_app(engine: ScoringEngine):
@@ -124,6 +125,32 @@ def build_app(engine: ScoringEngine):
         except ValueError as e:
             raise HTTPException(status_code=422, detail=str(e))
         return {"scores": outs}
+
+    @app.post("/score_raw")
+    async def score_raw(request: Request):
+        """Binary fast path for production clients: the body is the
+        (N, T, C) cross-section as little-endian float32 bytes (N
+        derived from the length); the response body is the N scores as
+        float32 bytes. Avoids the JSON float parse/serialize that
+        dominates /score (28 MB of text per CSI300 request)."""
+        import numpy as np
+
+        body = await request.body()
+        per_row = engine.seq_length * engine.num_latent * 4
+        if len(body) == 0 or len(body) % per_row != 0:
+            raise HTTPException(
+                status_code=422,
+                detail=f"body must be N*{per_row} bytes of float32")
+        n = len(body) // per_row
+        try:
+            x = torch.from_numpy(
+                np.frombuffer(body, dtype="<f4").copy().reshape(
+                    n, engine.seq_length, engine.num_latent))
+            out = engine.score(x)
+        except ValueError as e:
+            raise HTTPException(status_code=422, detail=str(e))
+        return Response(content=out.numpy().astype("<f4").tobytes(),
+                        media_type="application/octet-stream")
 
     return app
 

@@ -52,3 +52,24 @@ def test_score_batch_and_validation(client):
     bad = rng.standard_normal((4, 6, 12)).tolist()
     r = client.post("/score", json={"x": bad})
     assert r.status_code == 422
+
+
+def test_score_raw_binary_endpoint():
+    """Binary fast path: float32 body in, float32 scores out."""
+    import numpy as np
+
+    from factorvae_amd.serve import ScoringEngine, build_app
+    from fastapi.testclient import TestClient
+
+    eng = ScoringEngine(None, num_latent=12, hidden_size=8,
+                        num_portfolio=6, num_factor=4, seq_length=5,
+                        device="cpu")
+    client = TestClient(build_app(eng))
+    x = np.random.randn(7, 5, 12).astype("<f4")
+    r = client.post("/score_raw", content=x.tobytes())
+    assert r.status_code == 200
+    out = np.frombuffer(r.content, dtype="<f4")
+    assert out.shape == (7,) and np.isfinite(out).all()
+    # malformed length rejected
+    r2 = client.post("/score_raw", content=b"abc")
+    assert r2.status_code == 422
