@@ -869,3 +869,32 @@ def test_ln_fwd_large_mean_numerics():
     torch.cuda.synchronize()
     ref = F.layer_norm(x, (C,), g_, b_, 1e-5)
     assert_close(xln, ref, atol=2e-3, rtol=2e-3, what="ln_fwd large-mean")
+
+
+def test_forward_only_graph_capture_bf16():
+    """Regression: a forward-only hipGraph capture must join every
+    side-stream branch (the bf16 h_prev cast once leaked un-joined,
+    invalidating the capture and poisoning the stream)."""
+    from factorvae_amd.engine.fused import FusedTrainer
+    from factorvae_amd.models.modules import build_factorvae
+    from factorvae_amd.utils import set_seed
+
+    set_seed(0)
+    N, T, C, H, M, K = 128, 6, 158, 64, 32, 8
+    model = build_factorvae(num_latent=C, hidden_size=H, num_portfolio=M,
+                            num_factor=K).to(DEV)
+    tr = FusedTrainer(model, lr=1e-4, t_max=10, device=DEV, dtype="bf16",
+                      train=False)
+    x = t(N, T, C, seed=90)
+    y = t(N, 1, seed=91)
+    l1 = float(tr.forward_only(x, y)[0])
+    l2 = float(tr.forward_only(x, y)[0])  # replay path
+    torch.cuda.synchronize()
+    assert l1 == l1 and l2 == l2  # finite, stream not poisoned
+    key = ("val", N, T, False)
+    assert key in tr._graphs and tr._graphs[key]["g"] is not None, \
+        "validation forward was not captured"
+    # predict (with_loss=False) capture exercises the same join
+    p1 = tr.predict(x)
+    torch.cuda.synchronize()
+    assert torch.isfinite(p1).all()
