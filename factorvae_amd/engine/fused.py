@@ -188,6 +188,14 @@ class FusedTrainer:
         # exceeds the main path, so one side stream becomes the bottleneck
         self.s_side2 = (torch.cuda.Stream(device=self.device)
                         if self.device.type == "cuda" else None)
+        # main-chain stream: the step's critical path runs on a
+        # HIGH-priority stream (hip greatest = -1) so the side-stream
+        # weight-gradient kernels yield CU arbitration to it — measured
+        # knob FV_MAIN_PRIO=0 disables
+        self.s_main = None
+        if (self.device.type == "cuda"
+                and _os.environ.get("FV_MAIN_PRIO", "1") != "0"):
+            self.s_main = torch.cuda.Stream(device=self.device, priority=-1)
         # comm stream: in DP runs the attention/encoder/decoder slice of
         # the gradient arena (everything packed after the extractor) is
         # all-reduced here as soon as its last producer finishes,
@@ -421,6 +429,26 @@ class FusedTrainer:
         self.ws = w
         self._ws_n = N
         self._ws_t = T
+
+    def _main_ctx(self):
+        """Context manager entering the high-priority main stream (or a
+        no-op when unavailable/disabled)."""
+        if self.s_main is not None:
+            return torch.cuda.stream(self.s_main)
+        import contextlib
+        return contextlib.nullcontext()
+
+    def _main_entry(self):
+        """Order s_main after work the caller enqueued on its stream
+        (input tensors)."""
+        if self.s_main is not None:
+            self.s_main.wait_stream(torch.cuda.current_stream(self.device))
+
+    def _main_exit(self):
+        """Order the caller's stream after the step's s_main work (the
+        caller reads the device loss/scores)."""
+        if self.s_main is not None:
+            torch.cuda.current_stream(self.device).wait_stream(self.s_main)
 
     # ------------------------------------------------------ stream helpers
     def _streams(self):
@@ -989,23 +1017,27 @@ class FusedTrainer:
         assert C == self.C
         self._ensure_ws(N, T)
         w = self.ws
-        w["x"].copy_(x)
-        w["y"].copy_(y.view(N, 1))
+        self._main_entry()
+        with self._main_ctx():
+            w["x"].copy_(x)
+            w["y"].copy_(y.view(N, 1))
 
         if not self.use_graph:
             from ..observability import roctx_range
 
-            self._fill_rng(N)
-            with roctx_range("fv_forward"):
-                self._launch_forward(N, T)
-            with roctx_range("fv_backward"):
-                self._launch_backward(N, T)
-            if is_distributed():
-                with roctx_range("fv_allreduce"):
-                    self.grads.div_(get_world_size())
-                    torch.distributed.all_reduce(self.grads)
-            with roctx_range("fv_adam"):
-                self._launch_optimizer()
+            with self._main_ctx():
+                self._fill_rng(N)
+                with roctx_range("fv_forward"):
+                    self._launch_forward(N, T)
+                with roctx_range("fv_backward"):
+                    self._launch_backward(N, T)
+                if is_distributed():
+                    with roctx_range("fv_allreduce"):
+                        self.grads.div_(get_world_size())
+                        torch.distributed.all_reduce(self.grads)
+                with roctx_range("fv_adam"):
+                    self._launch_optimizer()
+            self._main_exit()
             return w["loss"]
 
         rng_ok, comm_ok = self._probe_caps()
@@ -1022,40 +1054,44 @@ class FusedTrainer:
                 return self.step(x, y)
         plan = self._graphs[key]
         if not rng_ok:
-            self._fill_rng(N)
-        if plan["split"]:
-            plan["g_fb"].replay()
-            self.grads.div_(get_world_size())
-            torch.distributed.all_reduce(self.grads)
-            plan["g_opt"].replay()
-        else:
-            plan["g"].replay()
+            with self._main_ctx():
+                self._fill_rng(N)
+        with self._main_ctx():
+            if plan["split"]:
+                plan["g_fb"].replay()
+                self.grads.div_(get_world_size())
+                torch.distributed.all_reduce(self.grads)
+                plan["g_opt"].replay()
+            else:
+                plan["g"].replay()
+        self._main_exit()
         return w["loss"]
 
     def _capture(self, key, N: int, T: int, rng_ok: bool, comm_ok: bool):
         # warmup fwd+bwd (settles lazy state; params/step counter untouched)
         torch.cuda.synchronize()
-        if not rng_ok:
-            self._fill_rng(N)
-        self.grads.zero_()
-        self._launch_forward(N, T)
-        self._launch_backward(N, T)
+        with self._main_ctx():
+            if not rng_ok:
+                self._fill_rng(N)
+            self.grads.zero_()
+            self._launch_forward(N, T)
+            self._launch_backward(N, T)
         torch.cuda.synchronize()
 
         split = is_distributed() and not comm_ok
         if split:
             g1 = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g1):
+            with torch.cuda.graph(g1, stream=self.s_main):
                 self._graph_step_body(None, None, N, T, rng_ok, False,
                                       with_opt=False)
             g2 = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g2):
+            with torch.cuda.graph(g2, stream=self.s_main):
                 self._launch_optimizer()
             self._graphs[key] = {"split": True, "g_fb": g1, "g_opt": g2}
         else:
             with self._abort_watchdog("train-step graph capture"):
                 g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
+                with torch.cuda.graph(g, stream=self.s_main):
                     self._graph_step_body(None, None, N, T, rng_ok,
                                           is_distributed() and comm_ok)
             self._graphs[key] = {"split": False, "g": g}
@@ -1080,11 +1116,12 @@ class FusedTrainer:
 
         # warmup
         torch.cuda.synchronize()
-        if not rng_ok:
-            self._fill_rng(N)
-        self.grads.zero_()
-        self._launch_forward(N, T, x=days[0][0], y=days[0][1])
-        self._launch_backward(N, T, x=days[0][0], y=days[0][1])
+        with self._main_ctx():
+            if not rng_ok:
+                self._fill_rng(N)
+            self.grads.zero_()
+            self._launch_forward(N, T, x=days[0][0], y=days[0][1])
+            self._launch_backward(N, T, x=days[0][0], y=days[0][1])
         torch.cuda.synchronize()
 
         # per-day RNG buffers when RNG can't live in the graph
@@ -1097,7 +1134,7 @@ class FusedTrainer:
         try:
             with self._abort_watchdog("bench multi-step graph capture"):
                 g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
+                with torch.cuda.graph(g, stream=self.s_main):
                     for i, (x, y) in enumerate(days):
                         if rng_bufs is not None:
                             self.ws["eps"].copy_(rng_bufs[i][0])
@@ -1119,7 +1156,10 @@ class FusedTrainer:
                     e.normal_()
                     if self.training:
                         m.bernoulli_(1.0 - self.DROPOUT_P)
-            g.replay()
+            self._main_entry()  # order replay after the RNG fills
+            with self._main_ctx():
+                g.replay()
+            self._main_exit()   # next call's fills wait for this replay
 
         return run, G
 
@@ -1131,30 +1171,37 @@ class FusedTrainer:
         N, T, C = x.shape
         self._ensure_ws(N, T)
         w = self.ws
-        w["x"].copy_(x)
-        w["y"].copy_(y.view(N, 1))
-        self._fill_rng(N)
+        self._main_entry()
+        with self._main_ctx():
+            w["x"].copy_(x)
+            w["y"].copy_(y.view(N, 1))
+            self._fill_rng(N)
         if not self.use_graph:
-            self._launch_forward(N, T)
+            with self._main_ctx():
+                self._launch_forward(N, T)
+            self._main_exit()
             return w["loss"]
         key = ("val", N, T, self.training)
         if key not in self._graphs:
             try:
                 torch.cuda.synchronize(self.device)
-                self._launch_forward(N, T)  # warmup
+                with self._main_ctx():
+                    self._launch_forward(N, T)  # warmup
                 torch.cuda.synchronize(self.device)
                 g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
+                with torch.cuda.graph(g, stream=self.s_main):
                     self._launch_forward(N, T)
                 self._graphs[key] = {"g": g}
             except Exception:
                 torch.cuda.synchronize(self.device)
                 self._graphs[key] = {"g": None}
         g = self._graphs[key]["g"]
-        if g is None:
-            self._launch_forward(N, T)
-        else:
-            g.replay()
+        with self._main_ctx():
+            if g is None:
+                self._launch_forward(N, T)
+            else:
+                g.replay()
+        self._main_exit()
         return w["loss"]
 
     def _launch_predict(self, N: int, T: int):
@@ -1173,34 +1220,42 @@ class FusedTrainer:
         N, T, C = x.shape
         self._ensure_ws(N, T)
         w = self.ws
-        w["x"].copy_(x)
+        self._main_entry()
+        with self._main_ctx():
+            w["x"].copy_(x)
         was_training = self.training
         self.training = False  # prediction: dropout off
         try:
-            self._fill_rng(N)
+            with self._main_ctx():
+                self._fill_rng(N)
             if not self.use_graph:
-                w["y"].zero_()
-                self._launch_predict(N, T)
+                with self._main_ctx():
+                    w["y"].zero_()
+                    self._launch_predict(N, T)
+                self._main_exit()
                 return w["recon"].view(N, 1).clone()
             key = ("predict", N, T)
             if key not in self._graphs:
                 try:
                     w["y"].zero_()
                     torch.cuda.synchronize(self.device)
-                    self._launch_predict(N, T)  # warmup
+                    with self._main_ctx():
+                        self._launch_predict(N, T)  # warmup
                     torch.cuda.synchronize(self.device)
                     gp = torch.cuda.CUDAGraph()
-                    with torch.cuda.graph(gp):
+                    with torch.cuda.graph(gp, stream=self.s_main):
                         self._launch_predict(N, T)
                     self._graphs[key] = {"g": gp}
                 except Exception:
                     torch.cuda.synchronize(self.device)
                     self._graphs[key] = {"g": None}
             gp = self._graphs[key]["g"]
-            if gp is None:
-                self._launch_predict(N, T)
-            else:
-                gp.replay()
+            with self._main_ctx():
+                if gp is None:
+                    self._launch_predict(N, T)
+                else:
+                    gp.replay()
+            self._main_exit()
             return w["recon"].view(N, 1).clone()
         finally:
             self.training = was_training
