@@ -188,13 +188,14 @@ class FusedTrainer:
         # exceeds the main path, so one side stream becomes the bottleneck
         self.s_side2 = (torch.cuda.Stream(device=self.device)
                         if self.device.type == "cuda" else None)
-        # main-chain stream: the step's critical path runs on a
-        # HIGH-priority stream (hip greatest = -1) so the side-stream
-        # weight-gradient kernels yield CU arbitration to it — measured
-        # knob FV_MAIN_PRIO=0 disables
+        # FV_MAIN_PRIO=1 (opt-in, default OFF): run the critical path on
+        # a HIGH-priority stream. Measured MUCH slower on gfx950
+        # (CSI300 2804 -> 1338 cs/s; A-share 591 -> 531): replaying the
+        # graph from a priority queue appears to serialize the captured
+        # side branches. Kept as a documented negative-result knob.
         self.s_main = None
         if (self.device.type == "cuda"
-                and _os.environ.get("FV_MAIN_PRIO", "1") != "0"):
+                and _os.environ.get("FV_MAIN_PRIO", "0") == "1"):
             self.s_main = torch.cuda.Stream(device=self.device, priority=-1)
         # comm stream: in DP runs the attention/encoder/decoder slice of
         # the gradient arena (everything packed after the extractor) is
