@@ -316,7 +316,6 @@ class FusedTrainer:
         w["xp"] = f(R, C)
         w["gi"] = f(R, 3 * H)
         w["h"] = f(N, H)
-        w["h_seq"] = f(N, T, H)
         w["h_prev"] = f(N, T, H)
         w["gates4"] = f(N, T, 4 * H)
         w["scores_enc"] = f(N, M)
@@ -540,10 +539,12 @@ class FusedTrainer:
             ext.gemm_nt(w["xln"], p("W1x"), p("b1x"), w["xp"], 1.0, False, True)
             ext.gemm_nt(w["xp"], p("Wih"), p("bih"), w["gi"], 1.0, False, False)
         if self.bf16 and H == 64:
+            # h_seq is a dead output in the engine (backward consumes
+            # h_prev + gates4); skip its (N,T,H) write entirely
             ext.gru_fwd_mfma(w["gi"], self.whh_bf, p("bhh"), w["h"],
-                             w["h_seq"], w["h_prev"], w["gates4"], N, T, H)
+                             None, w["h_prev"], w["gates4"], N, T, H)
         else:
-            ext.gru_fwd(w["gi"], p("Whh"), p("bhh"), w["h"], w["h_seq"],
+            ext.gru_fwd(w["gi"], p("Whh"), p("bhh"), w["h"], None,
                         w["h_prev"], w["gates4"], N, T, H)
         if self.bf16 and self.s_side2 is not None:
             e_ = torch.cuda.Event()

@@ -585,10 +585,13 @@ void ln_bwd_params(torch::Tensor x, torch::Tensor dxln, torch::Tensor mean,
 }
 
 void gru_fwd(torch::Tensor gi, torch::Tensor Whh, torch::Tensor bhh,
-             torch::Tensor h_final, torch::Tensor h_seq, torch::Tensor h_prev,
-             torch::Tensor gates4, long N, long T, long H) {
-  CK(gi); CK(Whh); CK(bhh); CK(h_final); CK(h_seq); CK(h_prev); CK(gates4);
-  RUN(fv_gru_fwd(fp(gi), fp(Whh), fp(bhh), fpm(h_final), fpm(h_seq),
+             torch::Tensor h_final, c10::optional<torch::Tensor> h_seq,
+             torch::Tensor h_prev, torch::Tensor gates4, long N, long T,
+             long H) {
+  CK(gi); CK(Whh); CK(bhh); CK(h_final); CK(h_prev); CK(gates4);
+  float* hs = nullptr;
+  if (h_seq.has_value()) { CK(*h_seq); hs = fpm(*h_seq); }
+  RUN(fv_gru_fwd(fp(gi), fp(Whh), fp(bhh), fpm(h_final), hs,
                  fpm(h_prev), fpm(gates4), (int)N, (int)T, (int)H,
                  cur_stream()));
 }
@@ -607,13 +610,15 @@ void gru_bwd(torch::Tensor dh_final, torch::Tensor h_prev, torch::Tensor gates4,
 }
 
 void gru_fwd_mfma(torch::Tensor gi, torch::Tensor whh_bf, torch::Tensor bhh,
-                  torch::Tensor h_final, torch::Tensor h_seq,
+                  torch::Tensor h_final, c10::optional<torch::Tensor> h_seq,
                   torch::Tensor h_prev, torch::Tensor gates4, long N, long T,
                   long H) {
-  CK(gi); CKB(whh_bf); CK(bhh); CK(h_final); CK(h_seq); CK(h_prev);
+  CK(gi); CKB(whh_bf); CK(bhh); CK(h_final); CK(h_prev);
   CK(gates4);
+  float* hs = nullptr;
+  if (h_seq.has_value()) { CK(*h_seq); hs = fpm(*h_seq); }
   RUN(fv_gru_fwd_mfma(fp(gi), bfpc(whh_bf), fp(bhh), fpm(h_final),
-                      fpm(h_seq), fpm(h_prev), fpm(gates4), (int)N, (int)T,
+                      hs, fpm(h_prev), fpm(gates4), (int)N, (int)T,
                       (int)H, cur_stream()));
 }
 

@@ -75,7 +75,7 @@ __global__ __launch_bounds__(256) void gru_fwd_generic_kernel(
       hn = fmaf(z, hp_l - n, n);  // (1-z)*n + z*hp
 
       const long ob = ((long)s * T + t) * H + lane;
-      h_seq[ob] = hn;
+      if (h_seq) h_seq[ob] = hn;
       h_prev_out[ob] = hp_l;
       const long gb = ((long)s * T + t) * 4 * H + lane;
       gates4[gb] = r;
@@ -254,7 +254,7 @@ __global__ __launch_bounds__(512) void gru_fwd_fast_kernel(
       hn = fmaf(z, hp_l - n, n);
 
       const long ob = ((long)s * T + t) * H + lane;
-      h_seq[ob] = hn;
+      if (h_seq) h_seq[ob] = hn;
       h_prev_out[ob] = hp_l;
       const long gb = ((long)s * T + t) * 4 * H + lane;
       gates4[gb] = r;

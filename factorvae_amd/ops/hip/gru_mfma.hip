@@ -130,7 +130,7 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_kernel(
       }
       // global saves (b128-shaped: ej is a multiple of 4)
       const long tb = (erow + t);
-      *(f32x4*)&h_seq[tb * 64 + ej] = *(f32x4*)hn;
+      if (h_seq) *(f32x4*)&h_seq[tb * 64 + ej] = *(f32x4*)hn;
       *(f32x4*)&h_prev_out[tb * 64 + ej] = *(f32x4*)hp4;
       float* g4 = &gates4[tb * 256];
       *(f32x4*)&g4[ej] = *(f32x4*)gr4;
@@ -407,7 +407,7 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_f32_kernel(
         gr4[u] = r; gz4[u] = z; gn4[u] = n; gq4[u] = q; hp4[u] = hp;
       }
       const long tb = (erow + t);
-      *(f32x4*)&h_seq[tb * 64 + ej] = *(f32x4*)hn;
+      if (h_seq) *(f32x4*)&h_seq[tb * 64 + ej] = *(f32x4*)hn;
       *(f32x4*)&h_prev_out[tb * 64 + ej] = *(f32x4*)hp4;
       float* g4 = &gates4[tb * 256];
       *(f32x4*)&g4[ej] = *(f32x4*)gr4;
