@@ -84,3 +84,40 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def union_gap_attribution(db, tail_frac=0.3, top=12):
+    """Largest ALL-queues-idle gaps with the kernels ending before and
+    starting after each gap."""
+    con = sqlite3.connect(db)
+    cur = con.cursor()
+    tables = [r[0] for r in cur.execute(
+        "SELECT name FROM sqlite_master WHERE type='table'")]
+    disp = next(t for t in tables if "kernel_dispatch" in t)
+    sym_t = next(t for t in tables if "info_kernel_symbol" in t)
+    sym = dict(cur.execute(f"SELECT id, display_name FROM {sym_t}"))
+    rows = list(cur.execute(
+        f"SELECT kernel_id, start, end FROM {disp} ORDER BY start"))
+    t1 = max(r[2] for r in rows)
+    t0 = t1 - (t1 - min(r[1] for r in rows)) * tail_frac
+    tail = [r for r in rows if r[1] >= t0]
+    # union timeline
+    iv = sorted((s, e, k) for k, s, e in tail)
+    gaps = []
+    ce = iv[0][1]
+    last_k = iv[0][2]
+    for s, e, k in iv[1:]:
+        if s > ce:
+            gaps.append((s - ce, last_k, k, ce))
+        if e > ce:
+            ce = e
+            last_k = k
+    gaps.sort(reverse=True)
+    nm = lambda k: (sym.get(k) or "?").split("(")[0][:40]
+    print("largest union-idle gaps (us, before -> after):")
+    for g, kb, ka, at in gaps[:top]:
+        print(f"  {g/1e3:8.2f}  {nm(kb)} -> {nm(ka)}")
+
+
+if len(sys.argv) > 2 and sys.argv[2] == "--attr":
+    union_gap_attribution(sys.argv[1])
