@@ -82,10 +82,6 @@ def main():
         print(f"  {g/1e3:9.2f}  @{at/1e6:8.3f}")
 
 
-if __name__ == "__main__":
-    main()
-
-
 def union_gap_attribution(db, tail_frac=0.3, top=12):
     """Largest ALL-queues-idle gaps with the kernels ending before and
     starting after each gap."""
@@ -119,5 +115,8 @@ def union_gap_attribution(db, tail_frac=0.3, top=12):
         print(f"  {g/1e3:8.2f}  {nm(kb)} -> {nm(ka)}")
 
 
-if len(sys.argv) > 2 and sys.argv[2] == "--attr":
-    union_gap_attribution(sys.argv[1])
+if __name__ == "__main__":
+    if "--attr" in sys.argv:
+        union_gap_attribution(sys.argv[1])
+    else:
+        main()
