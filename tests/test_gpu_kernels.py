@@ -300,6 +300,10 @@ def eager_forward_explicit_noise(model, x, y, eps, mask, training):
     # threshold logic in engine/fused.py
     (385, 20, 158, 64, 128, 20, True),
     (449, 20, 158, 64, 128, 20, True),
+    # the reference's RELEASED checkpoint configs: the backtested
+    # VAE-Revision (K=64,H=32,M=100) and free20 (T=20,K=20,H=20)
+    (300, 20, 158, 32, 100, 64, True),
+    (300, 20, 158, 20, 128, 20, True),
     # full A-share shape (BASELINE.json config 4): N=3500, T=60, K=96
     (3500, 60, 158, 64, 128, 96, True),
 ])
@@ -898,3 +902,30 @@ def test_forward_only_graph_capture_bf16():
     p1 = tr.predict(x)
     torch.cuda.synchronize()
     assert torch.isfinite(p1).all()
+
+
+
+def test_ws_cache_lru_eviction():
+    """WS_CACHE_MAX eviction: long ragged runs must bound workspace +
+    graph growth and keep producing correct losses after re-allocation
+    of an evicted shape."""
+    from factorvae_amd.engine.fused import FusedTrainer
+    from factorvae_amd.models.modules import build_factorvae
+    from factorvae_amd.utils import set_seed
+
+    set_seed(0)
+    C, H, M, K, T = 158, 64, 32, 8, 5
+    model = build_factorvae(num_latent=C, hidden_size=H, num_portfolio=M,
+                            num_factor=K).to(DEV)
+    tr = FusedTrainer(model, lr=1e-4, t_max=100, device=DEV)
+    tr.WS_CACHE_MAX = 3
+    sizes = [60, 70, 80, 90, 100, 110]
+    for rounds in range(2):  # second pass re-allocates evicted shapes
+        for n in sizes:
+            x = t(n, T, C, seed=n)
+            y = t(n, 1, seed=n + 1)
+            loss = float(tr.step(x, y)[0])
+            assert loss == loss, (rounds, n)
+    torch.cuda.synchronize()
+    assert len(tr._ws_cache) <= 3
+    assert len(tr._graphs) <= 3 * 3  # train/val/predict kinds per shape
