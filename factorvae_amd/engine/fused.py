@@ -1263,13 +1263,94 @@ class FusedTrainer:
             self.training = was_training
 
     # ------------------------------------------------------------- epochs
+    # steps per multi-step training graph (real-epoch batching)
+    TRAIN_GRAPH_STEPS = 8
+
+    def _get_multi_graph(self, N: int, T: int, g_len: int):
+        """A captured graph of g_len full training steps reading from
+        per-slot input buffers, with in-graph per-step loss
+        accumulation into `lsum` — one replay per g_len days removes
+        the per-day replay/launch host overhead from real epochs."""
+        key = ("trainG", N, T, g_len)
+        plan = self._graphs.get(key)
+        if plan is not None:
+            return plan
+        w = self.ws
+        xb = [torch.empty_like(w["x"]) for _ in range(g_len)]
+        yb = [torch.empty_like(w["y"]) for _ in range(g_len)]
+        lsum = torch.zeros(1, device=self.device)
+        # warmup (params/step counter untouched)
+        torch.cuda.synchronize(self.device)
+        with self._main_ctx():
+            self._fill_rng(N)
+            self._launch_forward(N, T, x=xb[0], y=yb[0])
+            self._launch_backward(N, T, x=xb[0], y=yb[0])
+        torch.cuda.synchronize(self.device)
+        try:
+            with self._abort_watchdog("train multi-step graph capture"):
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g, stream=self.s_main):
+                    for i in range(g_len):
+                        self._graph_step_body(xb[i], yb[i], N, T, True,
+                                              is_distributed())
+                        lsum += w["loss"]
+        except Exception:
+            torch.cuda.synchronize(self.device)
+            plan = {"g": None}
+            self._graphs[key] = plan
+            return plan
+        plan = {"g": g, "xb": xb, "yb": yb, "lsum": lsum}
+        self._graphs[key] = plan
+        return plan
+
     def train_epoch(self, days, shuffle_order=None) -> float:
+        """One epoch over device-resident day tensors. Runs of
+        TRAIN_GRAPH_STEPS consecutive same-shape days execute as ONE
+        multi-step graph replay (in-graph RNG + loss accumulation);
+        ragged leftovers fall back to the per-day step graph."""
         total = torch.zeros((), device=self.device)
         n = 0
-        for x, y in days:
+        days = list(days)
+        G = self.TRAIN_GRAPH_STEPS
+        rng_ok, comm_ok = (self._probe_caps()
+                           if (self.use_graph and self.device.type == "cuda")
+                           else (False, False))
+        use_multi = (self.use_graph and rng_ok and G > 1
+                     and not (is_distributed() and not comm_ok)
+                     and len(days) > 0
+                     and days[0][0].numel() * 4 <= 16 << 20)
+        # (size gate: the per-slot D2D input copies must stay well under
+        # the ~20 us/step host replay overhead the batching removes —
+        # big A-share days fall back to per-day replays)
+        i = 0
+        while i < len(days):
+            x, y = days[i]
+            run = 1
+            if use_multi:
+                while (run < G and i + run < len(days)
+                       and days[i + run][0].shape == x.shape):
+                    run += 1
+            if use_multi and run == G:
+                N, T, C = x.shape
+                self._ensure_ws(N, T)
+                plan = self._get_multi_graph(N, T, G)
+                if plan["g"] is not None:
+                    self._main_entry()
+                    with self._main_ctx():
+                        for j in range(G):
+                            plan["xb"][j].copy_(days[i + j][0])
+                            plan["yb"][j].copy_(days[i + j][1].view(-1, 1))
+                        plan["lsum"].zero_()
+                        plan["g"].replay()
+                        total += plan["lsum"][0]
+                    self._main_exit()
+                    n += G
+                    i += G
+                    continue
             loss = self.step(x, y)
             total += loss[0]
             n += 1
+            i += 1
         return (total / max(n, 1)).item()
 
     # ------------------------------------------------- optimizer side-car
