@@ -79,3 +79,22 @@ def test_bench_forced_dist_world_size_1():
     d = _parse_last_json(r.stdout)
     assert d["n_gpus"] == 1
     assert d["config"]["parallelism"] == "dp1"
+
+
+@pytest.mark.timeout(600)
+def test_bench_torchrun_four_ranks_gloo():
+    """4-rank ladder rung of the driver's SCALE launch (CPU/gloo)."""
+    env = dict(os.environ)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", "29573", "bench.py", "--gpus", "4",
+         "--steps", "4", "--warmup", "1", "--n_stocks", "16",
+         "--seq_len", "4", "--num_factor", "4", "--hidden_size", "16",
+         "--num_portfolio", "8", "--n_days", "2", "--engine", "eager"],
+        cwd=REPO, capture_output=True, text=True, timeout=540, env=env)
+    assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-2000:])
+    d = _parse_last_json(r.stdout)
+    assert d["n_gpus"] == 4
+    assert d["config"]["parallelism"] == "dp4"
+    assert d["config"]["global_batch"] == 64
