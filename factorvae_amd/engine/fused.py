@@ -116,6 +116,10 @@ class FusedTrainer:
                 f"use engine='eager'")
 
         self._build_param_arena()
+        # bf16/fp8 @ H=64: Whh wgrad computed inside the GRU backward
+        # kernel (FV_WHH_FUSED=0 restores the standalone TN call)
+        self._whh_fused = (self.bf16 and self.H == 64
+                           and _os.environ.get("FV_WHH_FUSED", "1") != "0")
         if self.bf16:
             C3 = 3 * self.H
             # padded (+ transposed-padded) weight shadows for the
@@ -414,8 +418,15 @@ class FusedTrainer:
             w["xp_bf"] = fbs(R, C)
             w["dgi_bf"] = fbs(R, 3 * H)
             w["dzx_bf"] = fbs(R, C)
-            w["h_prev_bf"] = fb(R, H)
-            w["dgh_bf"] = fb(R, 3 * H)
+            if self._whh_fused:
+                # the in-GRU Whh wgrad replaces the dgh_bf operand image
+                # + h_prev cast with per-block partials
+                nblk = (N + 15) // 16
+                w["whh_part"] = f(nblk * 3 * H * H)
+                w["bhh_part"] = f(nblk * 3 * H)
+            else:
+                w["h_prev_bf"] = fb(R, H)
+                w["dgh_bf"] = fb(R, 3 * H)
             if self.fp8:
                 ldp = self.w1x_f8.size(1)
                 f8t = lambda *shape: torch.zeros(
@@ -546,15 +557,18 @@ class FusedTrainer:
         else:
             ext.gru_fwd(w["gi"], p("Whh"), p("bhh"), w["h"], None,
                         w["h_prev"], w["gates4"], N, T, H)
-        if self.bf16 and self.s_side2 is not None:
-            e_ = torch.cuda.Event()
-            e_.record(torch.cuda.current_stream(self.device))
-            self.s_side2.wait_event(e_)
-            with torch.cuda.stream(self.s_side2):
+        if self.bf16 and not self._whh_fused:
+            # h_prev bf16 image (Whh TN wgrad operand) on side stream 2
+            if self.s_side2 is not None:
+                e_ = torch.cuda.Event()
+                e_.record(torch.cuda.current_stream(self.device))
+                self.s_side2.wait_event(e_)
+                with torch.cuda.stream(self.s_side2):
+                    ext.cast_f32_bf16(w["h_prev"].view(-1),
+                                      w["h_prev_bf"].view(-1))
+            else:
                 ext.cast_f32_bf16(w["h_prev"].view(-1),
                                   w["h_prev_bf"].view(-1))
-        elif self.bf16:
-            ext.cast_f32_bf16(w["h_prev"].view(-1), w["h_prev_bf"].view(-1))
         # h is ready: the K-head attention branch (prior path) runs on the
         # side stream, overlapped with the encoder + decoder branches on
         # main — they are independent until the loss joins pmu/psig with
@@ -744,14 +758,21 @@ class FusedTrainer:
         if self.bf16 and H == 64:
             # fp32 dgi/dgh images are dead in bf16 mode (the wgrads use
             # the bf16 copies); in fp8 mode the kernel also emits the
-            # scaled e4m3 dgrad operand directly from registers
+            # scaled e4m3 dgrad operand directly from registers; with
+            # _whh_fused the Whh/bhh wgrad partials come out of the same
+            # kernel (no dgh_bf image, no standalone TN call)
             ext.gru_bwd_mfma(w["dh"], w["h_prev"], w["gates4"], self.whh_bf,
                              None, None, N, T, H,
                              w["dgi_bf"].view(N, T, 3 * H),
-                             w["dgh_bf"].view(N, T, 3 * H),
+                             (None if self._whh_fused
+                              else w["dgh_bf"].view(N, T, 3 * H)),
                              dgi_f8=w["dgi_f8"] if fp8_gru_fused else None,
                              s_dgi=self.s_dgi if fp8_gru_fused else None,
                              amax_dgi=(self.amax_dgi if fp8_gru_fused
+                                       else None),
+                             whh_part=(w["whh_part"] if self._whh_fused
+                                       else None),
+                             bhh_part=(w["bhh_part"] if self._whh_fused
                                        else None))
         elif self.bf16:
             ext.gru_bwd(w["dh"], w["h_prev"], w["gates4"], p("Whh"),
@@ -764,9 +785,15 @@ class FusedTrainer:
         if self.bf16:
             fork(1)
             with _on_side(self, 1):
-                ext.gemm_tn_bf16(w["dgh_bf"].view(R, 3 * H), w["h_prev_bf"],
-                                 g("Whh"), w["tn_part"], chunks, False,
-                                 g("bhh"), w["tn_partb"])
+                if self._whh_fused:
+                    ext.wgrad_reduce(w["whh_part"], g("Whh"),
+                                     w["bhh_part"], g("bhh"),
+                                     (N + 15) // 16, False)
+                else:
+                    ext.gemm_tn_bf16(w["dgh_bf"].view(R, 3 * H),
+                                     w["h_prev_bf"], g("Whh"), w["tn_part"],
+                                     chunks, False, g("bhh"),
+                                     w["tn_partb"])
                 ext.gemm_tn_bf16(w["dgi_bf"].view(R, 3 * H), w["xp_bf"],
                                  g("Wih"), w["tn_part2"], chunks, False,
                                  g("bih"), w["tn_partb2"])

@@ -70,8 +70,10 @@ hipError_t fv_gru_fwd_mfma(const float*, const void*, const float*, float*,
                            hipStream_t);
 hipError_t fv_gru_bwd_mfma(const float*, const float*, const float*,
                            const void*, float*, float*, void*, void*,
-                           void*, int, const float*, float*, int, int, int,
-                           hipStream_t);
+                           void*, int, const float*, float*, float*,
+                           float*, int, int, int, hipStream_t);
+hipError_t fv_wgrad_reduce(const float*, float*, long, const float*,
+                           float*, long, int, int, hipStream_t);
 hipError_t fv_enc_softmax_fwd(const float*, const float*, float*, float*, int,
                               int, hipStream_t);
 hipError_t fv_enc_fused_fwd(const float*, const float*, const float*,
@@ -631,7 +633,9 @@ void gru_bwd_mfma(torch::Tensor dh_final, torch::Tensor h_prev,
                   c10::optional<torch::Tensor> dgh_bf = c10::nullopt,
                   c10::optional<torch::Tensor> dgi_f8 = c10::nullopt,
                   c10::optional<torch::Tensor> s_dgi = c10::nullopt,
-                  c10::optional<torch::Tensor> amax_dgi = c10::nullopt) {
+                  c10::optional<torch::Tensor> amax_dgi = c10::nullopt,
+                  c10::optional<torch::Tensor> whh_part = c10::nullopt,
+                  c10::optional<torch::Tensor> bhh_part = c10::nullopt) {
   CK(dh_final); CK(h_prev); CK(gates4); CKB(whh_bf);
   float* gi = nullptr; float* gh = nullptr;
   if (dgi.has_value()) { CK(*dgi); gi = fpm(*dgi); }
@@ -652,9 +656,29 @@ void gru_bwd_mfma(torch::Tensor dh_final, torch::Tensor h_prev,
     am = fpm(*amax_dgi);
   }
   TORCH_CHECK(gi || gib || g8, "gru_bwd_mfma needs a dgi output");
+  float* wp = nullptr; float* bp = nullptr;
+  if (whh_part.has_value()) {
+    CK(*whh_part);
+    TORCH_CHECK(bhh_part.has_value(), "whh_part needs bhh_part");
+    CK(*bhh_part);
+    const long nblk = (N + 15) / 16;
+    TORCH_CHECK(whh_part->numel() >= nblk * 192 * 64 &&
+                bhh_part->numel() >= nblk * 192, "wgrad partials too small");
+    wp = fpm(*whh_part);
+    bp = fpm(*bhh_part);
+  }
   RUN(fv_gru_bwd_mfma(fp(dh_final), fp(h_prev), fp(gates4), bfpc(whh_bf),
-                      gi, gh, gib, ghb, g8, ld8, sd, am, (int)N, (int)T,
-                      (int)H, cur_stream()));
+                      gi, gh, gib, ghb, g8, ld8, sd, am, wp, bp, (int)N,
+                      (int)T, (int)H, cur_stream()));
+}
+
+void wgrad_reduce(torch::Tensor part, torch::Tensor out,
+                  torch::Tensor db_part, torch::Tensor db, long z,
+                  bool accumulate) {
+  CK(part); CK(out); CK(db_part); CK(db);
+  RUN(fv_wgrad_reduce(fp(part), fpm(out), out.numel(), fp(db_part),
+                      fpm(db), db.numel(), (int)z, accumulate,
+                      cur_stream()));
 }
 
 void attn_fused_fwd(torch::Tensor h, torch::Tensor qk, torch::Tensor cb,
@@ -1041,7 +1065,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           py::arg("dgi"), py::arg("dgh"), py::arg("N"), py::arg("T"),
           py::arg("H"), py::arg("dgi_bf") = py::none(),
           py::arg("dgh_bf") = py::none(), py::arg("dgi_f8") = py::none(),
-          py::arg("s_dgi") = py::none(), py::arg("amax_dgi") = py::none());
+          py::arg("s_dgi") = py::none(), py::arg("amax_dgi") = py::none(),
+          py::arg("whh_part") = py::none(),
+          py::arg("bhh_part") = py::none());
+  mod.def("wgrad_reduce", &wgrad_reduce, py::arg("part"), py::arg("out"),
+          py::arg("db_part"), py::arg("db"), py::arg("z"),
+          py::arg("accumulate") = false);
   mod.def("gru_bwd", &gru_bwd, py::arg("dh_final"), py::arg("h_prev"),
           py::arg("gates4"), py::arg("Whh"), py::arg("dgi"), py::arg("dgh"),
           py::arg("N"), py::arg("T"), py::arg("H"),

@@ -845,6 +845,18 @@ hipError_t fv_cast_shadows(const float* s1, void* d1, void* d1t, int M1,
   return hipSuccess;
 }
 
+// fixed-order reduce of per-block wgrad partials (the in-GRU Whh
+// wgrad path): out(elems) = sum_z part[z]; db(m) = sum_z db_part[z].
+hipError_t fv_wgrad_reduce(const float* part, float* out, long elems,
+                           const float* db_part, float* db, long m_elems,
+                           int z, int accumulate, hipStream_t stream) {
+  dim3 rgrid((unsigned)((elems + m_elems + 63) / 64));
+  hipLaunchKernelGGL(tn_reduce_bf16_kernel, rgrid, dim3(256), 0, stream,
+                     part, out, elems, db_part, db, m_elems, z, accumulate);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
 hipError_t fv_cast3_f32_bf16(const float* s0, void* d0, long n0,
                              const float* s1, void* d1, long n1,
                              const float* s2, void* d2, long n2,

@@ -22,6 +22,8 @@
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) short s16x4_g;
+typedef __attribute__((address_space(3))) s16x4_g* lds_v4p_g;
 
 #define GM_S 16       // stocks per workgroup
 #define GM_HB 72      // bf16 h image row stride (16B-aligned frags)
@@ -158,6 +160,13 @@ __global__ __launch_bounds__(256) void gru_fwd_mfma_kernel(
 // collected via order-independent atomicMax) — the values are already
 // in registers here, so the fp8 operand costs three extra dword stores
 // per thread-step instead of a separate 160 MB cast pass.
+// whh_part/bhh_part (optional): per-workgroup Whh weight-gradient and
+// bhh bias-gradient partials computed IN-KERNEL — each step's
+// dgh_t^T @ h_prev_t accumulates into register tiles via tr16-read
+// MFMA on the dgh image that the dh_prev matmul already stages, so the
+// standalone TN wgrad call, its dgh_bf operand image (81 MB of writes
+// + 108 MB of reads at A-share) and the h_prev bf16 cast all disappear.
+// Deterministic: per-block partials, fixed-order reduce kernel.
 __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
     const float* __restrict__ dh_final,   // (N,64)
     const float* __restrict__ h_prev_in,  // (N,T,64)
@@ -168,9 +177,14 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
     __bf16* __restrict__ dgi_bf, __bf16* __restrict__ dgh_bf,
     unsigned char* __restrict__ dgi_f8, int ld8,
     const float* __restrict__ s_dgi, float* __restrict__ amax_dgi,
+    float* __restrict__ whh_part,         // (nblk, 192*64) optional
+    float* __restrict__ bhh_part,         // (nblk, 192)
     int N, int T) {
   const __bf16* whh_bf = (const __bf16*)whh_bf_;
-  __shared__ __bf16 dgB[GM_S][GM_GB];    // bf16 dgh image (A-frags)
+  // dgB padded to 32 k-rows: the in-kernel wgrad's tr16 A-fragments
+  // read k-row offsets up to 27 (stocks beyond GM_S stay zero)
+  __shared__ __bf16 dgB[2 * GM_S][GM_GB];  // bf16 dgh image
+  __shared__ __bf16 hpB[2 * GM_S][72];     // bf16 h_prev image (wgrad B)
   __shared__ float dhS[GM_S][64];
   __shared__ float zdh[GM_S][64];
 
@@ -198,10 +212,22 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
   const bool elive = (s0 + es) < N;
   const long erow = (long)(s0 + es) * T;
   float amax_l = 0.0f;
+  // in-kernel wgrad state: bias-grad partials + weight-grad register
+  // tiles (wave wv owns gate rows [wv*48, wv*48+48) x all 64 h-cols)
+  float dbr[4] = {0, 0, 0, 0}, dbz[4] = {0, 0, 0, 0}, dbn[4] = {0, 0, 0, 0};
+  f32x4 accw[3][4];
+#pragma unroll
+  for (int m3 = 0; m3 < 3; ++m3)
+#pragma unroll
+    for (int n4 = 0; n4 < 4; ++n4) accw[m3][n4] = (f32x4){0, 0, 0, 0};
+  const int qm = (lane & 15) >> 2;   // tr16 supplier k-row offset
+  const int nq = (lane & 3) * 4;     // tr16 supplier column-quad base
 
-  // init dh = dh_final; zero dgB pad rows
-  for (int idx = tid; idx < GM_S * GM_GB; idx += 256)
+  // init dh = dh_final; zero dgB/hpB pad rows
+  for (int idx = tid; idx < 2 * GM_S * GM_GB; idx += 256)
     dgB[idx / GM_GB][idx % GM_GB] = (__bf16)0.0f;
+  for (int idx = tid; idx < 2 * GM_S * 72; idx += 256)
+    hpB[idx / 72][idx % 72] = (__bf16)0.0f;
   if (elive) {
 #pragma unroll
     for (int u = 0; u < 4; ++u)
@@ -239,6 +265,12 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
         dgB[es][j] = (__bf16)dgate_r;
         dgB[es][64 + j] = (__bf16)dgate_z;
         dgB[es][128 + j] = (__bf16)dgh_n;
+        if (whh_part) {
+          hpB[es][j] = (__bf16)hp;
+          dbr[u] += dgate_r;
+          dbz[u] += dgate_z;
+          dbn[u] += dgh_n;
+        }
       }
       if (dgi) {
         float* di = &dgi[tb * 192];
@@ -295,6 +327,37 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
       const bf16x8 a = *(const bf16x8*)&dgB[fi][c * 32 + fk * 8];
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr[c], acc, 0, 0, 0);
     }
+    if (whh_part) {
+      // dWhh partial += dgh_t^T (192 x 16stk) @ h_prev_t (16stk x 64):
+      // both operands tr16-read from the staged images; k = 32 with the
+      // zero-padded stock rows making the tail exact
+      const int kb = fk * 8 + qm;
+      bf16x8 bw[4];
+#pragma unroll
+      for (int n4 = 0; n4 < 4; ++n4) {
+        s16x4_g b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4p_g)&hpB[kb][n4 * 16 + nq]);
+        s16x4_g b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4p_g)&hpB[kb + 4][n4 * 16 + nq]);
+        *(s16x4_g*)&bw[n4] = b0;
+        *(((s16x4_g*)&bw[n4]) + 1) = b1;
+      }
+#pragma unroll
+      for (int m3 = 0; m3 < 3; ++m3) {
+        const int mb = wv * 48 + m3 * 16;
+        s16x4_g a0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4p_g)&dgB[kb][mb + nq]);
+        s16x4_g a1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4p_g)&dgB[kb + 4][mb + nq]);
+        bf16x8 aw;
+        *(s16x4_g*)&aw = a0;
+        *(((s16x4_g*)&aw) + 1) = a1;
+#pragma unroll
+        for (int n4 = 0; n4 < 4; ++n4)
+          accw[m3][n4] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              aw, bw[n4], accw[m3][n4], 0, 0, 0);
+      }
+    }
 #pragma unroll
     for (int rr = 0; rr < 4; ++rr) {
       const int m = fk * 4 + rr;
@@ -307,6 +370,35 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
     amax_l = wave_reduce_max(amax_l);
     if ((tid & 63) == 0 && amax_l > 0.0f)
       atomicMax((int*)amax_dgi, __float_as_int(amax_l));
+  }
+  if (whh_part) {
+    float* wp = whh_part + (long)blockIdx.x * (192 * 64);
+#pragma unroll
+    for (int m3 = 0; m3 < 3; ++m3) {
+      const int mrow = wv * 48 + m3 * 16 + fk * 4;
+#pragma unroll
+      for (int n4 = 0; n4 < 4; ++n4) {
+        const int col = n4 * 16 + fi;
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr)
+          wp[(mrow + rr) * 64 + col] = accw[m3][n4][rr];
+      }
+    }
+    // bias-grad partials: 16-stock reduce per column through zdh
+    float* bp = bhh_part + (long)blockIdx.x * 192;
+    const float* dbg[3] = {dbr, dbz, dbn};
+    for (int grp = 0; grp < 3; ++grp) {
+      __syncthreads();
+#pragma unroll
+      for (int u = 0; u < 4; ++u) zdh[es][ej + u] = dbg[grp][u];
+      __syncthreads();
+      if (tid < 64) {
+        float s = 0.0f;
+#pragma unroll
+        for (int k = 0; k < GM_S; ++k) s += zdh[k][tid];
+        bp[grp * 64 + tid] = s;
+      }
+    }
   }
 }
 
@@ -559,14 +651,16 @@ hipError_t fv_gru_bwd_mfma(const float* dh_final, const float* h_prev,
                            const float* gates4, const void* whh_bf,
                            float* dgi, float* dgh, void* dgi_bf,
                            void* dgh_bf, void* dgi_f8, int ld8,
-                           const float* s_dgi, float* amax_dgi, int N,
+                           const float* s_dgi, float* amax_dgi,
+                           float* whh_part, float* bhh_part, int N,
                            int T, int H, hipStream_t stream) {
   if (H != 64) return hipErrorInvalidValue;
   dim3 grid((N + GM_S - 1) / GM_S);
   hipLaunchKernelGGL(gru_bwd_mfma_kernel, grid, dim3(256), 0, stream,
                      dh_final, h_prev, gates4, whh_bf, dgi, dgh,
                      (__bf16*)dgi_bf, (__bf16*)dgh_bf,
-                     (unsigned char*)dgi_f8, ld8, s_dgi, amax_dgi, N, T);
+                     (unsigned char*)dgi_f8, ld8, s_dgi, amax_dgi,
+                     whh_part, bhh_part, N, T);
   HIP_CHECK_LAST();
   return hipSuccess;
 }

@@ -929,3 +929,38 @@ def test_ws_cache_lru_eviction():
     torch.cuda.synchronize()
     assert len(tr._ws_cache) <= 3
     assert len(tr._graphs) <= 3 * 3  # train/val/predict kinds per shape
+
+
+def test_whh_fused_wgrad_matches_tn(monkeypatch):
+    """The in-GRU Whh/bhh wgrad (per-block MFMA partials) must match
+    the standalone TN wgrad path on the same bf16 operands."""
+    from factorvae_amd.engine.fused import FusedTrainer
+    from factorvae_amd.models.modules import build_factorvae
+    from factorvae_amd.utils import set_seed
+
+    N, T, C, H, M, K = 300, 20, 158, 64, 32, 8
+    grads = {}
+    for mode, env in (("fused", "1"), ("tn", "0")):
+        monkeypatch.setenv("FV_WHH_FUSED", env)
+        set_seed(0)
+        model = build_factorvae(num_latent=C, hidden_size=H,
+                                num_portfolio=M, num_factor=K).to(DEV)
+        tr = FusedTrainer(model, lr=1e-4, t_max=100, device=DEV,
+                          use_graph=False, train=True, dtype="bf16")
+        assert tr._whh_fused == (env == "1")
+        x = t(N, T, C, seed=60)
+        y = t(N, 1, seed=61)
+        tr._ensure_ws(N, T)
+        tr.ws["x"].copy_(x)
+        tr.ws["y"].copy_(y)
+        set_seed(1)
+        tr._fill_rng(N)
+        tr.grads.zero_()
+        tr._launch_forward(N, T)
+        tr._launch_backward(N, T)
+        torch.cuda.synchronize()
+        grads[mode] = (tr.g("Whh").clone(), tr.g("bhh").clone())
+    assert_close(grads["fused"][0], grads["tn"][0], atol=2e-3, rtol=2e-3,
+                 what="Whh wgrad fused-vs-TN")
+    assert_close(grads["fused"][1], grads["tn"][1], atol=2e-3, rtol=2e-3,
+                 what="bhh wgrad fused-vs-TN")
