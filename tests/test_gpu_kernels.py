@@ -964,3 +964,28 @@ def test_whh_fused_wgrad_matches_tn(monkeypatch):
                  what="Whh wgrad fused-vs-TN")
     assert_close(grads["fused"][1], grads["tn"][1], atol=2e-3, rtol=2e-3,
                  what="bhh wgrad fused-vs-TN")
+
+
+def test_train_epoch_uses_multi_step_graphs():
+    """train_epoch must batch runs of uniform-shape days into the
+    multi-step graph (and still produce finite, decreasing losses)."""
+    from factorvae_amd.engine.fused import FusedTrainer
+    from factorvae_amd.models.modules import build_factorvae
+    from factorvae_amd.utils import set_seed
+
+    set_seed(0)
+    N, T, C, H, M, K = 96, 6, 158, 64, 32, 8
+    model = build_factorvae(num_latent=C, hidden_size=H, num_portfolio=M,
+                            num_factor=K).to(DEV)
+    tr = FusedTrainer(model, lr=5e-3, t_max=200, device=DEV, dtype="bf16")
+    g = torch.Generator(device=DEV).manual_seed(3)
+    days = [(torch.randn(N, T, C, device=DEV, generator=g),
+             torch.randn(N, 1, device=DEV, generator=g))
+            for _ in range(19)]  # 2 chunks of 8 + 3 per-day leftovers
+    l0 = tr.train_epoch(days)
+    l1 = tr.train_epoch(days)
+    torch.cuda.synchronize()
+    key = ("trainG", N, T, tr.TRAIN_GRAPH_STEPS)
+    assert key in tr._graphs and tr._graphs[key]["g"] is not None
+    assert ("train", N, T) in tr._graphs  # leftover per-day path used too
+    assert l0 == l0 and l1 == l1 and l1 < l0  # finite + learning
