@@ -182,9 +182,12 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
     int N, int T) {
   const __bf16* whh_bf = (const __bf16*)whh_bf_;
   // dgB padded to 32 k-rows: the in-kernel wgrad's tr16 A-fragments
-  // read k-row offsets up to 27 (stocks beyond GM_S stay zero)
-  __shared__ __bf16 dgB[2 * GM_S][GM_GB];  // bf16 dgh image
-  __shared__ __bf16 hpB[2 * GM_S][72];     // bf16 h_prev image (wgrad B)
+  // read k-row offsets up to 27 (stocks beyond GM_S stay zero).
+  // DOUBLE-BUFFERED: step t's wgrad MFMAs issue during step t-1's
+  // elementwise phase, overlapping the matrix pipe with the gate VALU
+  // instead of serializing after the dh matmul.
+  __shared__ __bf16 dgB[2][2 * GM_S][GM_GB];  // bf16 dgh image
+  __shared__ __bf16 hpB[2][2 * GM_S][72];     // bf16 h_prev image
   __shared__ float dhS[GM_S][64];
   __shared__ float zdh[GM_S][64];
 
@@ -223,11 +226,13 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
   const int qm = (lane & 15) >> 2;   // tr16 supplier k-row offset
   const int nq = (lane & 3) * 4;     // tr16 supplier column-quad base
 
-  // init dh = dh_final; zero dgB/hpB pad rows
-  for (int idx = tid; idx < 2 * GM_S * GM_GB; idx += 256)
-    dgB[idx / GM_GB][idx % GM_GB] = (__bf16)0.0f;
-  for (int idx = tid; idx < 2 * GM_S * 72; idx += 256)
-    hpB[idx / 72][idx % 72] = (__bf16)0.0f;
+  // init dh = dh_final; zero dgB/hpB pad rows (both buffers)
+  for (int idx = tid; idx < 2 * 2 * GM_S * GM_GB; idx += 256)
+    dgB[idx / (2 * GM_S * GM_GB)][(idx / GM_GB) % (2 * GM_S)]
+       [idx % GM_GB] = (__bf16)0.0f;
+  for (int idx = tid; idx < 2 * 2 * GM_S * 72; idx += 256)
+    hpB[idx / (2 * GM_S * 72)][(idx / 72) % (2 * GM_S)]
+       [idx % 72] = (__bf16)0.0f;
   if (elive) {
 #pragma unroll
     for (int u = 0; u < 4; ++u)
@@ -238,7 +243,42 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
   }
   __syncthreads();
 
+  // wgrad MFMA accumulation from a completed image buffer
+  auto wgrad_acc = [&](int buf) {
+    const int kb = fk * 8 + qm;
+    bf16x8 bw[4];
+#pragma unroll
+    for (int n4 = 0; n4 < 4; ++n4) {
+      s16x4_g b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+          (lds_v4p_g)&hpB[buf][kb][n4 * 16 + nq]);
+      s16x4_g b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+          (lds_v4p_g)&hpB[buf][kb + 4][n4 * 16 + nq]);
+      *(s16x4_g*)&bw[n4] = b0;
+      *(((s16x4_g*)&bw[n4]) + 1) = b1;
+    }
+#pragma unroll
+    for (int m3 = 0; m3 < 3; ++m3) {
+      const int mb = wv * 48 + m3 * 16;
+      s16x4_g a0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+          (lds_v4p_g)&dgB[buf][kb][mb + nq]);
+      s16x4_g a1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+          (lds_v4p_g)&dgB[buf][kb + 4][mb + nq]);
+      bf16x8 aw;
+      *(s16x4_g*)&aw = a0;
+      *(((s16x4_g*)&aw) + 1) = a1;
+#pragma unroll
+      for (int n4 = 0; n4 < 4; ++n4)
+        accw[m3][n4] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            aw, bw[n4], accw[m3][n4], 0, 0, 0);
+    }
+  };
+
   for (int t = T - 1; t >= 0; --t) {
+    const int buf = t & 1;
+    // step t+1's wgrad (its images are complete and in the OTHER
+    // buffer): issued here so the MFMA pipe overlaps this step's gate
+    // VALU below
+    if (whh_part && t + 1 <= T - 1) wgrad_acc((t + 1) & 1);
     // ---- elementwise phase: gate grads for this step
     if (elive) {
       const long tb = erow + t;
@@ -262,11 +302,11 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
         const float dgate_z = dz * z * (1.0f - z);
         dgr4[u] = dgate_r; dgz4[u] = dgate_z; da4[u] = da; dghn4[u] = dgh_n;
         zdh[es][j] = dh * z;
-        dgB[es][j] = (__bf16)dgate_r;
-        dgB[es][64 + j] = (__bf16)dgate_z;
-        dgB[es][128 + j] = (__bf16)dgh_n;
+        dgB[buf][es][j] = (__bf16)dgate_r;
+        dgB[buf][es][64 + j] = (__bf16)dgate_z;
+        dgB[buf][es][128 + j] = (__bf16)dgh_n;
         if (whh_part) {
-          hpB[es][j] = (__bf16)hp;
+          hpB[buf][es][j] = (__bf16)hp;
           dbr[u] += dgate_r;
           dbz[u] += dgate_z;
           dbn[u] += dgh_n;
@@ -324,40 +364,10 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
     f32x4 acc = {0, 0, 0, 0};
 #pragma unroll
     for (int c = 0; c < 6; ++c) {
-      const bf16x8 a = *(const bf16x8*)&dgB[fi][c * 32 + fk * 8];
+      const bf16x8 a = *(const bf16x8*)&dgB[buf][fi][c * 32 + fk * 8];
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr[c], acc, 0, 0, 0);
     }
-    if (whh_part) {
-      // dWhh partial += dgh_t^T (192 x 16stk) @ h_prev_t (16stk x 64):
-      // both operands tr16-read from the staged images; k = 32 with the
-      // zero-padded stock rows making the tail exact
-      const int kb = fk * 8 + qm;
-      bf16x8 bw[4];
-#pragma unroll
-      for (int n4 = 0; n4 < 4; ++n4) {
-        s16x4_g b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (lds_v4p_g)&hpB[kb][n4 * 16 + nq]);
-        s16x4_g b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (lds_v4p_g)&hpB[kb + 4][n4 * 16 + nq]);
-        *(s16x4_g*)&bw[n4] = b0;
-        *(((s16x4_g*)&bw[n4]) + 1) = b1;
-      }
-#pragma unroll
-      for (int m3 = 0; m3 < 3; ++m3) {
-        const int mb = wv * 48 + m3 * 16;
-        s16x4_g a0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (lds_v4p_g)&dgB[kb][mb + nq]);
-        s16x4_g a1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (lds_v4p_g)&dgB[kb + 4][mb + nq]);
-        bf16x8 aw;
-        *(s16x4_g*)&aw = a0;
-        *(((s16x4_g*)&aw) + 1) = a1;
-#pragma unroll
-        for (int n4 = 0; n4 < 4; ++n4)
-          accw[m3][n4] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              aw, bw[n4], accw[m3][n4], 0, 0, 0);
-      }
-    }
+
 #pragma unroll
     for (int rr = 0; rr < 4; ++rr) {
       const int m = fk * 4 + rr;
@@ -372,6 +382,7 @@ __global__ __launch_bounds__(256) void gru_bwd_mfma_kernel(
       atomicMax((int*)amax_dgi, __float_as_int(amax_l));
   }
   if (whh_part) {
+    wgrad_acc(0);  // step t=0's images (barrier'd by its M phase)
     float* wp = whh_part + (long)blockIdx.x * (192 * 64);
 #pragma unroll
     for (int m3 = 0; m3 < 3; ++m3) {
