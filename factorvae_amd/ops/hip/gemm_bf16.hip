@@ -748,9 +748,16 @@ hipError_t fv_gemm_tn_bf16(const void* A, const void* B, float* out,
   if (r_chunks > 1) {
     // fill the chip: at TBM=TBN=64 the x/y grid is tiny (<=9 blocks for
     // this model), so the z split is the only parallelism lever (4+
-    // resident blocks/CU hide the staging latency)
+    // resident blocks/CU hide the staging latency). FV_TN_Z caps the
+    // split — a SMALLER grid leaves more CUs to the concurrent
+    // main-chain GEMMs (contention shaping; the wgrads have slack)
+    static int zcap = 0;
+    if (zcap == 0) {
+      const char* e = getenv("FV_TN_Z");
+      zcap = (e && atoi(e) > 0) ? atoi(e) : 128;
+    }
     r_chunks = (R + 511) / 512;
-    if (r_chunks > 128) r_chunks = 128;
+    if (r_chunks > zcap) r_chunks = zcap;
     if (r_chunks < 1) r_chunks = 1;
   }
   if (r_chunks > 1 && db && !db_part) r_chunks = 1;
