@@ -507,56 +507,75 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
 
   const int span = rend - rbeg;
   const int ktiles = (span + TK - 1) / TK;
+  // TWO register staging sets: global loads are issued two k-tiles
+  // ahead (a load set stays in flight across two compute phases — one
+  // tile of lead time was shorter than the DRAM round trip, leaving
+  // the waves parked ~70% on s_waitcnt)
   unsigned int pa[TK / 8], pb[TK / 8];
+  unsigned int pa2[TK / 8], pb2[TK / 8];
 
-  auto stage_regs = [&](int r0_) {
+  auto stage_regs_to = [&](int r0_, unsigned int (&qa)[TK / 8],
+                           unsigned int (&qb)[TK / 8]) {
     const bool interior = (r0_ + TK <= rend) && (m0 + TBM <= M) &&
                           (n0 + TBN <= N);
     if (interior) {
 #pragma unroll
       for (int u = 0; u < TK / 8; ++u) {
         const long gr = (long)r0_ + s_kr0 + 8 * u;
-        pa[u] = *(const unsigned int*)(A + gr * M + m0 + s_cp);
-        pb[u] = *(const unsigned int*)(B + gr * N + n0 + s_cp);
+        qa[u] = *(const unsigned int*)(A + gr * M + m0 + s_cp);
+        qb[u] = *(const unsigned int*)(B + gr * N + n0 + s_cp);
       }
     } else {
 #pragma unroll
       for (int u = 0; u < TK / 8; ++u) {
         const long gr = (long)r0_ + s_kr0 + 8 * u;
-        pa[u] = load_dw_guard(A, gr, m0 + s_cp, rend, M, M);
-        pb[u] = load_dw_guard(B, gr, n0 + s_cp, rend, N, N);
+        qa[u] = load_dw_guard(A, gr, m0 + s_cp, rend, M, M);
+        qb[u] = load_dw_guard(B, gr, n0 + s_cp, rend, N, N);
       }
     }
   };
-  auto regs_to_lds = [&](int buf) {
+  auto regs_to_lds_from = [&](int buf, unsigned int (&qa)[TK / 8],
+                              unsigned int (&qb)[TK / 8]) {
 #pragma unroll
     for (int u = 0; u < TK / 8; ++u) {
       const int kr = s_kr0 + 8 * u;
-      *(unsigned int*)&As[buf][kr][s_cp] = pa[u];
-      *(unsigned int*)&Bs[buf][kr][s_cp] = pb[u];
+      *(unsigned int*)&As[buf][kr][s_cp] = qa[u];
+      *(unsigned int*)&Bs[buf][kr][s_cp] = qb[u];
     }
   };
-  auto bias_from_regs = [&]() {
+  auto bias_from = [&](unsigned int (&qa)[TK / 8]) {
 #pragma unroll
     for (int u = 0; u < TK / 8; ++u) {
       dw_bf2 d;
-      d.u = pa[u];
+      d.u = qa[u];
       bsum0 += (float)d.h[0];
       bsum1 += (float)d.h[1];
     }
   };
 
   if (ktiles > 0) {
-    stage_regs(rbeg);
-    if (do_bias) bias_from_regs();
-    regs_to_lds(0);
+    stage_regs_to(rbeg, pa, pb);
+    if (do_bias) bias_from(pa);
+    regs_to_lds_from(0, pa, pb);
+    if (ktiles > 1) {
+      stage_regs_to(rbeg + TK, pa2, pb2);  // kt=1 loads, issued early
+      if (do_bias) bias_from(pa2);
+    }
   }
 
   for (int kt = 0; kt < ktiles; ++kt) {
     __syncthreads();
-    if (kt + 1 < ktiles) {
-      stage_regs(rbeg + (kt + 1) * TK);
-      if (do_bias) bias_from_regs();
+    const bool even = (kt & 1) == 0;
+    if (kt + 2 < ktiles) {
+      // issue kt+2's loads into the set consumed at the END of the
+      // NEXT iteration: two compute phases of latency cover
+      if (even) {
+        stage_regs_to(rbeg + (kt + 2) * TK, pa, pb);
+        if (do_bias) bias_from(pa);
+      } else {
+        stage_regs_to(rbeg + (kt + 2) * TK, pa2, pb2);
+        if (do_bias) bias_from(pa2);
+      }
     }
     const int buf = kt & 1;
 #pragma unroll
@@ -583,7 +602,10 @@ __global__ __launch_bounds__(256) void gemm_tn_bf16_kernel(
     }
     if (kt + 1 < ktiles) {
       __syncthreads();
-      regs_to_lds(1 - buf);
+      if (even)
+        regs_to_lds_from(1 - buf, pa2, pb2);
+      else
+        regs_to_lds_from(1 - buf, pa, pb);
     }
   }
 
