@@ -3,9 +3,10 @@ path.
 
 The reference has no serving story (scores are produced offline by a
 notebook); this daemon keeps a checkpoint resident on the GPU and scores
-cross-sections on demand at the fused engine's inference rate
-(~5.2k cross-sections/s measured on 1×MI355X; eager CPU fallback when no
-GPU is present).
+cross-sections on demand: 612 req/s / 184k stock-scores/s at p50
+1.6 ms over loopback HTTP on CSI300-shaped days via /score_raw
+(measured on 1×MI355X, profiles/r2_serving.md; eager CPU fallback when
+no GPU is present).
 
 Run:  python -m factorvae_amd.serve --checkpoint best_models/x.pt \
           [--num_factor 96 --hidden_size 64 ...] [--port 8321]
