@@ -32,6 +32,9 @@ def main():
     p.add_argument("--reqs", type=int, default=200)
     p.add_argument("--warmup", type=int, default=20)
     p.add_argument("--port", type=int, default=8441)
+    p.add_argument("--skip-json", action="store_true",
+                   help="only bench /score_raw (JSON is ~300x slower at "
+                        "large N; skips a multi-minute wait)")
     args = p.parse_args()
 
     from factorvae_amd.serve import ScoringEngine, build_app
@@ -76,14 +79,16 @@ def main():
             print(f"{name}: {args.reqs/el:8.1f} req/s "
                   f"({args.n*args.reqs/el:9.0f} stock-scores/s)  "
                   f"p50 {lat[len(lat)//2]*1e3:6.2f} ms  "
-                  f"p95 {lat[int(len(lat)*0.95)]*1e3:6.2f} ms")
+                  f"p95 {lat[int(len(lat)*0.95)]*1e3:6.2f} ms",
+                  flush=True)
 
         body = x.tobytes()
         run("/score_raw (binary)",
             lambda: client.post("/score_raw", content=body))
-        payload = {"x": x.tolist()}
-        run("/score     (json)  ",
-            lambda: client.post("/score", json=payload))
+        if not args.skip_json:
+            payload = {"x": x.tolist()}
+            run("/score     (json)  ",
+                lambda: client.post("/score", json=payload))
     server.should_exit = True
     th.join(timeout=5)
 
