@@ -21,11 +21,22 @@ from torch.utils.data import DataLoader, Dataset, Sampler
 
 
 def np_ffill(arr: np.ndarray) -> np.ndarray:
-    """Forward-fill NaN along axis 0 (/root/reference/dataset.py:24-39)."""
+    """Forward-fill NaN along axis 0 (/root/reference/dataset.py:24-39).
+
+    The reference version broadcasts incorrectly for 2-D inputs (it only
+    ever calls it on 1-D data); here the 2-D case fills each column
+    independently, matching ``pd.DataFrame(arr).ffill()`` — leading NaNs
+    stay NaN. 1-D semantics are identical to the reference.
+    """
     mask = np.isnan(arr)
-    idx = np.where(~mask, np.arange(mask.shape[0]), 0)
+    rows = np.arange(mask.shape[0])
+    if arr.ndim == 1:
+        idx = np.where(~mask, rows, 0)
+        np.maximum.accumulate(idx, axis=0, out=idx)
+        return arr[idx]
+    idx = np.where(~mask, rows[:, None], 0)
     np.maximum.accumulate(idx, axis=0, out=idx)
-    return arr[idx]
+    return arr[idx, np.arange(arr.shape[1])[None, :]]
 
 
 class TSDataSampler:
