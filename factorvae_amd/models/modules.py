@@ -278,6 +278,8 @@ class FactorVAE(nn.Module):
                 + (sigma1 ** 2 + (mu1 - mu2) ** 2) / (2 * sigma2 ** 2) - 0.5).sum()
 
     def forward(self, x: torch.Tensor, returns: torch.Tensor):
+        if returns.dim() == 1:  # accept (N,) like the encoder does; the
+            returns = returns.unsqueeze(1)  # mse would silently broadcast (N,N)
         stock_latent = self.feature_extractor(x)
         factor_mu, factor_sigma = self.factor_encoder(stock_latent, returns)
         reconstruction = self.factor_decoder(stock_latent, factor_mu, factor_sigma)
