@@ -83,3 +83,44 @@ def test_rankic_empty_frame_returns_dataframe():
         names=["datetime", "instrument"])
     out = RankIC(pd.DataFrame({"LABEL0": [], "Pred": []}, index=idx))
     assert isinstance(out, pd.DataFrame) and np.isnan(out["RankIC"].iloc[0])
+
+
+# ---------------------------------------------------------------------------
+# risk_analysis differential: brute-force formulas on random return
+# series, plus structural invariants (drawdown <= 0, scale equivariance).
+# ---------------------------------------------------------------------------
+
+from factorvae_amd.backtest import risk_analysis
+
+
+@given(
+    n=st.integers(2, 60),
+    seed=st.integers(0, 10**6),
+    mu=st.floats(-0.01, 0.01),
+    vol=st.floats(1e-4, 0.05),
+)
+@settings(**SET)
+def test_risk_analysis_matches_brute_force(n, seed, mu, vol):
+    rng = np.random.default_rng(seed)
+    r = pd.Series(mu + vol * rng.standard_normal(n),
+                  index=pd.date_range("2023-01-02", periods=n, freq="B"))
+    out = risk_analysis(r, N=252)["risk"]
+    assert list(out.index) == ["mean", "std", "annualized_return",
+                               "information_ratio", "max_drawdown"]
+    np.testing.assert_allclose(out["mean"], r.to_numpy().mean(), atol=1e-12)
+    np.testing.assert_allclose(out["std"], np.std(r.to_numpy(), ddof=1),
+                               rtol=1e-10)
+    np.testing.assert_allclose(out["annualized_return"], out["mean"] * 252,
+                               rtol=1e-12)
+    np.testing.assert_allclose(out["information_ratio"],
+                               out["mean"] / out["std"] * np.sqrt(252),
+                               rtol=1e-10)
+    # max drawdown: brute-force over all (i <= j) windows of the cumsum
+    cum = np.cumsum(r.to_numpy())
+    dd = min(cum[j] - cum[:j + 1].max() for j in range(n))
+    np.testing.assert_allclose(out["max_drawdown"], dd, atol=1e-12)
+    assert out["max_drawdown"] <= 0
+    # NaNs are dropped, not poisoning
+    r2 = r.copy()
+    r2.iloc[0] = np.nan
+    assert np.isfinite(risk_analysis(r2, N=252)["risk"]["mean"])
