@@ -124,3 +124,70 @@ def test_risk_analysis_matches_brute_force(n, seed, mu, vol):
     r2 = r.copy()
     r2.iloc[0] = np.nan
     assert np.isfinite(risk_analysis(r2, N=252)["risk"]["mean"])
+
+
+# ---------------------------------------------------------------------------
+# Top-k dropout simulator invariants across random score/return panels.
+# ---------------------------------------------------------------------------
+
+from factorvae_amd.backtest import BacktestConfig, topk_dropout_backtest
+
+
+@given(
+    n_days=st.integers(1, 8),
+    n_stocks=st.integers(1, 15),
+    topk=st.integers(1, 8),
+    n_drop=st.integers(0, 4),
+    seed=st.integers(0, 10**6),
+)
+@settings(**SET)
+def test_topk_dropout_invariants(n_days, n_stocks, topk, n_drop, seed):
+    rng = np.random.default_rng(seed)
+    dates = pd.date_range("2023-03-01", periods=n_days, freq="B")
+    idx = pd.MultiIndex.from_product(
+        [dates, [f"S{i}" for i in range(n_stocks)]],
+        names=["datetime", "instrument"])
+    df = pd.DataFrame({
+        "score": rng.standard_normal(len(idx)),
+        "LABEL0": 0.02 * rng.standard_normal(len(idx)),
+    }, index=idx)
+    cfg = BacktestConfig(topk=topk, n_drop=n_drop)
+    res = topk_dropout_backtest(df, config=cfg)
+
+    assert len(res.holdings) == n_days
+    held_prev = []
+    for d, held in enumerate(res.holdings):
+        day_scores = df.loc[dates[d], "score"]
+        # book size: never above topk; equals min(topk, universe) when
+        # enough candidates exist
+        assert len(held) <= topk
+        assert len(set(held)) == len(held)  # no duplicate names
+        assert set(held) <= set(day_scores.index)
+        if d == 0:
+            assert held == list(
+                day_scores.sort_values(ascending=False).index[:topk])
+        else:
+            # at most n_drop names leave the book per day
+            dropped = set(held_prev) - set(held)
+            assert len(dropped) <= n_drop
+        held_prev = held
+
+    # turnover/cost bounds: first day all buys; after that at most
+    # (n_drop sells + n_drop buys)/k of the book
+    assert (res.daily_turnover >= 0).all()
+    assert (res.daily_cost >= 0).all()
+    assert res.daily_turnover.iloc[0] <= 1.0 + 1e-12
+    if n_days > 1:
+        k = max(1, min(topk, n_stocks))
+        assert (res.daily_turnover.iloc[1:] <= 2 * n_drop / k + 1e-12).all()
+
+    # returns: equal-weight mean of held names' LABEL0
+    for d, held in enumerate(res.holdings):
+        exp = (df.loc[dates[d], "LABEL0"].reindex(held).fillna(0).mean()
+               if held else 0.0)
+        np.testing.assert_allclose(res.daily_return.iloc[d], exp, atol=1e-12)
+
+    # with-cost excess = no-cost excess - cost (no benchmark given)
+    np.testing.assert_allclose(
+        res.excess_with_cost.to_numpy(),
+        (res.excess_no_cost - res.daily_cost).to_numpy(), atol=1e-15)
