@@ -129,3 +129,35 @@ def test_train_h128_eager_fallback_target():
     loss = train(model, loader, opt, None, device=device)
     assert np.isfinite(loss)
     assert np.isfinite(validate(model, loader, device=device))
+
+
+def test_root_entrypoint_shims(tmp_path):
+    """Reference-layout drop-ins: `python main.py` and
+    `from train_model import train` work from the repo root
+    (/root/reference/main.py, train_model.py)."""
+    import os
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    import importlib.util as iu
+
+    spec = iu.spec_from_file_location("train_model",
+                                     os.path.join(root, "train_model.py"))
+    tm = iu.module_from_spec(spec)
+    spec.loader.exec_module(tm)
+    assert callable(tm.train) and callable(tm.validate) and callable(tm.test)
+
+    df = _frame()
+    data = tmp_path / "d.pkl"
+    df.to_pickle(data)
+    r = subprocess.run(
+        [sys.executable, "main.py", "--num_epochs", "1", "--run_name", "sh",
+         "--dataset", str(data), "--num_latent", "10", "--hidden_size", "8",
+         "--num_portfolio", "12", "--num_factor", "4", "--seq_len", "5",
+         "--num_workers", "0", "--save_dir", str(tmp_path),
+         "--start_time", "2015-01-01", "--fit_end_time", "2015-02-10",
+         "--val_start_time", "2015-02-11", "--end_time", "2015-03-01",
+         "--val_end_time", "2015-03-01"],
+        cwd=root, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
