@@ -161,3 +161,24 @@ def test_shuffle_is_epoch_seeded(frame):
     o3 = [g[0] for g in s]
     assert o1 == o2
     assert o1 != o3
+
+
+def test_make_dataset_cli(tmp_path):
+    """Reference-layout ETL entrypoint (data/make_dataset.py) writes a
+    loadable dataset pickle with the reference output contract."""
+    import os
+    import subprocess
+    import sys
+
+    import pandas as pd
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = tmp_path / "mk.pkl"
+    r = subprocess.run(
+        [sys.executable, os.path.join(root, "data", "make_dataset.py"),
+         "--out", str(out), "--n_days", "10", "--n_stocks", "6"],
+        capture_output=True, text=True, timeout=180)
+    assert r.returncode == 0, r.stderr[-1500:]
+    df = pd.read_pickle(out)
+    assert list(df.index.names) == ["datetime", "instrument"]
+    assert df.shape == (60, 159) and "LABEL0" in df.columns
