@@ -98,3 +98,25 @@ def test_bench_torchrun_four_ranks_gloo():
     assert d["n_gpus"] == 4
     assert d["config"]["parallelism"] == "dp4"
     assert d["config"]["global_batch"] == 64
+
+
+@pytest.mark.timeout(600)
+def test_bench_torchrun_eight_ranks_gloo():
+    """Full-node rung of the driver's SCALE launch (nproc-per-node 8,
+    CPU/gloo): the exact width the round-end 8×MI355X run uses, incl.
+    n_days(2) < world_size(8) — wrap-padding must keep all 8 ranks in
+    lockstep."""
+    env = dict(os.environ)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", "29574", "bench.py", "--gpus", "8",
+         "--steps", "3", "--warmup", "1", "--n_stocks", "12",
+         "--seq_len", "4", "--num_factor", "3", "--hidden_size", "16",
+         "--num_portfolio", "6", "--n_days", "2", "--engine", "eager"],
+        cwd=REPO, capture_output=True, text=True, timeout=540, env=env)
+    assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-2000:])
+    d = _parse_last_json(r.stdout)
+    assert d["n_gpus"] == 8
+    assert d["config"]["parallelism"] == "dp8"
+    assert d["config"]["global_batch"] == 96
