@@ -28,6 +28,8 @@ Design:
 from __future__ import annotations
 
 import math
+import os
+import warnings
 from typing import Dict, Optional
 
 import torch
@@ -894,14 +896,28 @@ class FusedTrainer:
     # max distinct (N, T) shapes whose workspaces + captured graphs stay
     # resident; beyond this the least-recently-used shape is dropped
     # (long multi-year runs see hundreds of distinct daily stock counts)
-    WS_CACHE_MAX = 64
+    WS_CACHE_MAX = 256
 
     def _ensure_ws(self, N: int, T: int):
         """Workspaces (and captured graphs) are cached per (N, T): real
         universes have a different stock count N every day, and 288 GB
         HBM3E makes keeping one ~60 MB workspace per distinct shape far
-        cheaper than re-capturing the step graph each day. An LRU cap
-        bounds total growth."""
+        cheaper than re-capturing the step graph each day (256 shapes
+        ≈ a few tens of GB, sized for MI355X HBM).
+
+        Overflow policy is a synchronized FULL reset — drop every
+        workspace and every captured graph at once, never individual
+        shapes. Destroying single hipGraphExec objects while dozens of
+        sibling graphs stay live was observed to corrupt later replays
+        of the SURVIVORS on ROCm 7.2 (host segfault inside
+        hipGraphLaunch) once the churn got big enough: a 96-shape
+        ragged-universe run crashed in epoch 2 with per-shape LRU
+        eviction (profiles/r2_final_validation.md), while an
+        all-at-once rebuild from an empty runtime state is the pattern
+        the small-cap eviction test and every fixed-shape run already
+        exercise safely. The reset costs one recapture pass (~30 ms per
+        shape) and fires only when a run exceeds WS_CACHE_MAX
+        (`FV_WS_CACHE` overrides) distinct day shapes."""
         if self._ws_n == N and getattr(self, "_ws_t", None) == T:
             return
         lru = self.__dict__.setdefault("_ws_lru", [])
@@ -914,11 +930,20 @@ class FusedTrainer:
             self._ws_n = N
             self._ws_t = T
         else:
-            while len(lru) > self.WS_CACHE_MAX:
-                old = lru.pop(0)
-                self._ws_cache.pop(old, None)
-                for gk in [k for k in self._graphs if k[1:3] == old]:
-                    del self._graphs[gk]
+            cap = int(os.environ.get("FV_WS_CACHE", "0")) or self.WS_CACHE_MAX
+            if len(lru) > cap:
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize(self.device)
+                self._graphs.clear()
+                self._ws_cache.clear()
+                lru.clear()
+                lru.append((N, T))
+                if not getattr(self, "_ws_reset_warned", False):
+                    self._ws_reset_warned = True
+                    warnings.warn(
+                        f"shape-cache overflow (> {cap} distinct (N, T) "
+                        "day shapes): all workspaces/graphs reset and "
+                        "rebuilt; raise FV_WS_CACHE to avoid recaptures")
             self._alloc_ws(N, T)
 
     # -------------------------------------------------- graph capability probe

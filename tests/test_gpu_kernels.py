@@ -906,9 +906,11 @@ def test_forward_only_graph_capture_bf16():
 
 
 def test_ws_cache_lru_eviction():
-    """WS_CACHE_MAX eviction: long ragged runs must bound workspace +
-    graph growth and keep producing correct losses after re-allocation
-    of an evicted shape."""
+    """WS_CACHE_MAX overflow: long ragged runs must bound workspace +
+    graph growth and keep producing correct losses after the
+    synchronized full reset (individual hipGraphExec destroys are
+    avoided — see _ensure_ws docstring) re-allocates a dropped
+    shape."""
     from factorvae_amd.engine.fused import FusedTrainer
     from factorvae_amd.models.modules import build_factorvae
     from factorvae_amd.utils import set_seed
