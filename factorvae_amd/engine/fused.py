@@ -97,10 +97,9 @@ class FusedTrainer:
         self.eta_min = eta_min
         self.t_max = t_max
         self.training = train
-        import os as _os
         # FV_GRAPH=0: eager-launch every kernel (no hipGraph capture at
         # all) — the last-resort guard rail for capture-hostile boxes
-        self.use_graph = use_graph and _os.environ.get("FV_GRAPH",
+        self.use_graph = use_graph and os.environ.get("FV_GRAPH",
                                                        "1") != "0"
 
         fe = model.feature_extractor
@@ -121,7 +120,7 @@ class FusedTrainer:
         # bf16/fp8 @ H=64: Whh wgrad computed inside the GRU backward
         # kernel (FV_WHH_FUSED=0 restores the standalone TN call)
         self._whh_fused = (self.bf16 and self.H == 64
-                           and _os.environ.get("FV_WHH_FUSED", "1") != "0")
+                           and os.environ.get("FV_WHH_FUSED", "1") != "0")
         if self.bf16:
             C3 = 3 * self.H
             # padded (+ transposed-padded) weight shadows for the
@@ -201,7 +200,7 @@ class FusedTrainer:
         # side branches. Kept as a documented negative-result knob.
         self.s_main = None
         if (self.device.type == "cuda"
-                and _os.environ.get("FV_MAIN_PRIO", "0") == "1"):
+                and os.environ.get("FV_MAIN_PRIO", "0") == "1"):
             self.s_main = torch.cuda.Stream(device=self.device, priority=-1)
         # comm stream: in DP runs the attention/encoder/decoder slice of
         # the gradient arena (everything packed after the extractor) is
@@ -653,8 +652,7 @@ class FusedTrainer:
         yv = w["y"] if y is None else y
         alpha = 1.0 / math.sqrt(float(H) + 1e-6)
         keep_inv = 1.0 / (1.0 - self.DROPOUT_P)
-        import os as _os
-        chunks = (int(_os.environ.get("FV_TN_CHUNKS", "0"))
+        chunks = (int(os.environ.get("FV_TN_CHUNKS", "0"))
                   or max(1, min(32, R // 1024)))
 
         main = torch.cuda.current_stream(self.device) if self.s_side else None
@@ -963,13 +961,13 @@ class FusedTrainer:
             self.tag = tag
 
         def __enter__(self):
-            import os as _os
             import threading
+
             self.evt = threading.Event()
             self.armed = is_distributed() and get_world_size() > 1
             if not self.armed:
                 return self
-            limit = float(_os.environ.get("FV_CAPTURE_WATCHDOG_S", "180"))
+            limit = float(os.environ.get("FV_CAPTURE_WATCHDOG_S", "180"))
 
             def _wd():
                 if not self.evt.wait(limit):
@@ -1001,9 +999,8 @@ class FusedTrainer:
         """
         if self._caps is not None:
             return self._caps
-        import os as _os
-        allow_comm = _os.environ.get("FV_COMM_GRAPH", "1") != "0"
-        allow_rng = _os.environ.get("FV_RNG_GRAPH", "1") != "0"
+        allow_comm = os.environ.get("FV_COMM_GRAPH", "1") != "0"
+        allow_rng = os.environ.get("FV_RNG_GRAPH", "1") != "0"
         rng_ok = allow_rng
         if rng_ok:
             try:
