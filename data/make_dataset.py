@@ -13,9 +13,10 @@ If you DO have qlib and a data dump, run the reference's script to get
 real CSI300/S&P500 data — the resulting pickle loads here unchanged.
 """
 import argparse
+import os
 import sys
 
-sys.path.insert(0, __file__.rsplit("/", 2)[0])
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from factorvae_amd.data.synthetic import make_synthetic_frame
 
