@@ -1,5 +1,6 @@
+import os
 import sys, time
-sys.path.insert(0, "/root/repo")
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 from factorvae_amd.ops import get_extension
 ext = get_extension()

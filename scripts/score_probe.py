@@ -1,5 +1,6 @@
+import os
 import sys, time
-sys.path.insert(0, "/root/repo")
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 from factorvae_amd.data.sampler import init_data_loader
 from factorvae_amd.data.synthetic import make_synthetic_frame
