@@ -73,3 +73,41 @@ def test_score_raw_binary_endpoint():
     # malformed length rejected
     r2 = client.post("/score_raw", content=b"abc")
     assert r2.status_code == 422
+
+
+def test_daemon_real_socket_binary_roundtrip():
+    """Full daemon path over a real loopback socket (uvicorn + httpx),
+    not just the in-process ASGI TestClient: binary /score_raw
+    roundtrip at CPU-eager speed."""
+    import threading
+    import time
+
+    uvicorn = pytest.importorskip("uvicorn")
+    httpx = pytest.importorskip("httpx")
+
+    eng = ScoringEngine(None, num_latent=6, hidden_size=8, num_portfolio=4,
+                        num_factor=3, seq_length=4, device="cpu")
+    app = build_app(eng)
+    cfg = uvicorn.Config(app, host="127.0.0.1", port=8473,
+                         log_level="error")
+    server = uvicorn.Server(cfg)
+    th = threading.Thread(target=server.run, daemon=True)
+    th.start()
+    try:
+        for _ in range(100):
+            if server.started:
+                break
+            time.sleep(0.05)
+        else:
+            pytest.skip("uvicorn did not start (port in use?)")
+        x = np.random.default_rng(0).standard_normal((5, 4, 6)).astype("<f4")
+        with httpx.Client(base_url="http://127.0.0.1:8473",
+                          timeout=30.0) as client:
+            assert client.get("/health").json()["status"] == "ok"
+            r = client.post("/score_raw", content=x.tobytes())
+            assert r.status_code == 200
+            out = np.frombuffer(r.content, dtype="<f4")
+            assert out.shape == (5,) and np.isfinite(out).all()
+    finally:
+        server.should_exit = True
+        th.join(timeout=5)
