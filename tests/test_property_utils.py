@@ -191,3 +191,53 @@ def test_topk_dropout_invariants(n_days, n_stocks, topk, n_drop, seed):
     np.testing.assert_allclose(
         res.excess_with_cost.to_numpy(),
         (res.excess_no_cost - res.daily_cost).to_numpy(), atol=1e-15)
+
+
+# ---------------------------------------------------------------------------
+# Synthetic generator invariants — it underpins most other tests.
+# ---------------------------------------------------------------------------
+
+from factorvae_amd.data.synthetic import make_synthetic_frame
+
+
+@given(
+    n_days=st.integers(1, 10),
+    n_stocks=st.integers(5, 40),
+    n_features=st.integers(1, 12),
+    seed=st.integers(0, 10**6),
+    ragged=st.booleans(),
+    lff=st.booleans(),
+)
+@settings(**SET)
+def test_synthetic_frame_invariants(n_days, n_stocks, n_features, seed,
+                                    ragged, lff):
+    df = make_synthetic_frame(n_days=n_days, n_stocks=n_stocks,
+                              n_features=n_features, seed=seed,
+                              ragged=ragged, label_from_features=lff)
+    assert list(df.index.names) == ["datetime", "instrument"]
+    assert df.index.is_monotonic_increasing  # sorted as loaders expect
+    assert not df.index.duplicated().any()
+    assert list(df.columns[-1:]) == ["LABEL0"]
+    assert df.shape[1] == n_features + 1
+    assert np.isfinite(df.to_numpy()).all()  # no NaN/Inf anywhere
+    assert df.dtypes.eq(np.float32).all()
+    sizes = df.groupby(level=0).size()
+    assert len(sizes) == n_days
+    if ragged and n_stocks > 10:
+        assert (sizes >= 5).all() and (sizes <= n_stocks).all()
+    else:
+        assert (sizes == n_stocks).all()
+    # CSRankNorm label: bounded by ±0.5*sqrt(12); pct-rank gives an
+    # exact per-day mean of sqrt(12)/(2N) when there are no ties
+    lab = df["LABEL0"]
+    bound = 0.5 * np.sqrt(12.0) + 1e-5
+    assert (lab.abs() <= bound).all()
+    for _, day in lab.groupby(level=0):
+        np.testing.assert_allclose(day.mean(),
+                                   np.sqrt(12.0) / (2 * len(day)),
+                                   atol=1e-3)
+    # determinism: same seed -> identical frame
+    df2 = make_synthetic_frame(n_days=n_days, n_stocks=n_stocks,
+                               n_features=n_features, seed=seed,
+                               ragged=ragged, label_from_features=lff)
+    pd.testing.assert_frame_equal(df, df2)
