@@ -64,3 +64,43 @@ def test_forward_invariants_any_cross_section(n, t, seed, training):
     model.eval()
     p = model.prediction(x)
     assert p.reshape(-1).shape == (n,) and torch.isfinite(p).all()
+
+
+@given(
+    c=st.integers(4, 24),
+    h=st.integers(4, 24),
+    m=st.integers(2, 12),
+    k=st.integers(1, 8),
+    seed=st.integers(0, 10**5),
+)
+@settings(derandomize=True, max_examples=25, deadline=None)
+def test_checkpoint_roundtrip_any_hyperparams(tmp_path_factory, c, h, m, k,
+                                              seed):
+    """state_dict save/load roundtrips bitwise for arbitrary (C, H, M, K)
+    — key namespace and shapes must not depend on the specific released
+    configs the fixed tests use."""
+    import io
+
+    torch.manual_seed(seed)
+    a = build_factorvae(num_latent=c, hidden_size=h, num_portfolio=m,
+                        num_factor=k)
+    buf = io.BytesIO()
+    torch.save(a.state_dict(), buf)
+    buf.seek(0)
+    b = build_factorvae(num_latent=c, hidden_size=h, num_portfolio=m,
+                        num_factor=k)
+    missing, unexpected = b.load_state_dict(
+        torch.load(buf, weights_only=True))
+    assert not missing and not unexpected
+    for (ka_, va), (kb_, vb) in zip(a.state_dict().items(),
+                                    b.state_dict().items()):
+        assert ka_ == kb_
+        assert torch.equal(va, vb), ka_
+    # loaded model computes identically
+    x = torch.randn(6, 3, c)
+    y = torch.randn(6, 1)
+    torch.manual_seed(0)
+    la = a(x, y)[0]
+    torch.manual_seed(0)
+    lb = b(x, y)[0]
+    assert torch.equal(la, lb)
