@@ -42,6 +42,14 @@ def resolve_engine(engine: str, hidden_size: int, device_type: str,
     engine = engine or "auto"
     if engine == "auto":
         engine = "fused" if device_type == "cuda" else "eager"
+    if engine == "fused" and device_type != "cuda":
+        if warn:
+            import warnings
+            warnings.warn(
+                "--engine fused requires a GPU (HIP streams/graphs); "
+                "falling back to the eager engine on this CPU-only "
+                "machine", RuntimeWarning, stacklevel=2)
+        engine = "eager"
     if engine == "fused" and hidden_size > 64:
         if warn:
             import warnings

@@ -116,6 +116,9 @@ def test_resolve_engine_h_gt_64_falls_back_to_eager():
 
     assert resolve_engine("auto", 64, "cpu") == "eager"
     assert resolve_engine("auto", 64, "cuda") == "fused"
+    with warnings.catch_warnings(record=True):
+        warnings.simplefilter("always")
+        assert resolve_engine("fused", 64, "cpu") == "eager"
     with warnings.catch_warnings(record=True) as rec:
         warnings.simplefilter("always")
         assert resolve_engine("fused", 128, "cuda") == "eager"
