@@ -186,6 +186,11 @@ def train_main(args, data_args, df=None) -> float:
     device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
     engine = resolve_engine(getattr(args, "engine", "auto"),
                             args.hidden_size, device.type)
+    if engine != "fused" and getattr(args, "dtype", "fp32") != "fp32":
+        import warnings
+        warnings.warn(
+            f"--dtype {args.dtype} only applies to the fused engine; the "
+            "eager path runs fp32", RuntimeWarning, stacklevel=2)
 
     logger = MetricsLogger(args.run_name, out_dir=args.save_dir,
                            use_wandb=getattr(args, "wandb", False),
