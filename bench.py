@@ -193,7 +193,8 @@ def main():
                           "unnamed CUDA GPU (backtest.ipynb cell 4) — "
                           "vs_baseline is training-vs-inference, not "
                           "like-for-like"),
-        "dtype": args.dtype,
+        # effective compute dtype: the eager oracle path always runs fp32
+        "dtype": args.dtype if engine_name == "fused" else "fp32",
         "data": "synthetic",
         "config": {
             "model": "FactorVAE",
