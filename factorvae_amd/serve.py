@@ -166,11 +166,14 @@ def main(argv=None):
     p.add_argument("--seq_length", type=int, default=20)
     p.add_argument("--host", type=str, default="127.0.0.1")
     p.add_argument("--port", type=int, default=8321)
+    p.add_argument("--device", type=str, default=None,
+                   help="cuda / cpu (default: cuda if available)")
     args = p.parse_args(argv)
 
     engine = ScoringEngine(args.checkpoint, args.num_latent,
                            args.hidden_size, args.num_portfolio,
-                           args.num_factor, args.seq_length)
+                           args.num_factor, args.seq_length,
+                           device=args.device)
     app = build_app(engine)
     import uvicorn
 
