@@ -42,6 +42,10 @@ def build_argparser() -> argparse.ArgumentParser:
                    default=True)
     p.add_argument("--select_feature", action=argparse.BooleanOptionalAction,
                    default=False)
+    p.add_argument("--engine", type=str, default="auto",
+                   choices=["auto", "fused", "eager"],
+                   help="auto = fused HIP predict path on GPU, eager "
+                        "module path otherwise")
     p.add_argument("--backtest", action="store_true",
                    help="run the top-k dropout backtest + risk report")
     p.add_argument("--report", action="store_true",
@@ -68,7 +72,8 @@ def main(argv=None):
     df = pd.read_pickle(args.dataset)
     loader = init_data_loader(df, step_len=args.seq_length, shuffle=False,
                               start=args.start, end=args.end)
-    scores = generate_prediction_scores(model, loader, loader.dataset, targs)
+    scores = generate_prediction_scores(model, loader, loader.dataset,
+                                        targs, engine=args.engine)
 
     os.makedirs(args.out_dir, exist_ok=True)
     # reference artifact name schema (scores/readme.md)
