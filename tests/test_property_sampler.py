@@ -231,3 +231,36 @@ def test_dp_shard_invariants(n_days, world_size, shuffle, epoch, sizes):
                                  world_size=world_size, seed=5)
     s2.set_epoch(epoch)
     assert list(s2) == per_rank[0]
+
+
+@given(
+    n_dates=st.integers(2, 6),
+    n_inst=st.integers(2, 4),
+    flt_bits=st.integers(0, 2**24 - 1),
+    T=st.integers(1, 4),
+)
+@settings(**SET)
+def test_flt_data_row_filter(n_dates, n_inst, flt_bits, T):
+    """flt_data restricts WHICH rows are sampled, but windows still
+    gather history from the FULL frame (the reference's fltdata
+    semantics): index/len reflect the filtered rows, window content
+    matches the unfiltered sampler at the same (date, inst)."""
+    df = make_frame(n_dates, n_inst, 2**28 - 1, 0)  # full presence
+    flt = pd.Series(
+        [(flt_bits >> (i % 24)) & 1 == 1 for i in range(len(df))],
+        index=df.index)
+    if not flt.any():
+        return
+    smp = TSDataSampler(df, None, None, step_len=T, fillna_type="ffill",
+                        flt_data=flt)
+    full = TSDataSampler(df, None, None, step_len=T, fillna_type="ffill")
+    expect_index = df.sort_index().index[flt.to_numpy()]
+    assert smp.get_index().equals(expect_index)
+    assert len(smp) == int(flt.sum())
+    # each filtered position's window equals the unfiltered sampler's
+    # window at the corresponding full-frame position
+    full_pos = {key: p for p, key in enumerate(df.sort_index().index)}
+    for p in range(len(smp)):
+        got, actual = smp[p]
+        exp, _ = full[full_pos[actual[0]]]
+        np.testing.assert_array_equal(got, exp)
