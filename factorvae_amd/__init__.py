@@ -16,7 +16,7 @@ cross-sectional stock returns) for AMD Instinct MI355X (gfx950, CDNA4):
 - Device-resident epoch cache sized for 288 GB HBM3E per GPU.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from .models.modules import (
     FeatureExtractor,
